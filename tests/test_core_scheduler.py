@@ -1,0 +1,324 @@
+"""Unit tests for the scheduler core: state machine, ETA/MPE, seed plan,
+optimize_jobs (equal + weighted split, deferral, complementary production,
+step scaling, pixel caps), requeue-on-failure, config round-trip.
+
+Mirrors the test strategy SURVEY.md §4 prescribes for the missing reference
+suite; everything here is pure logic and runs on CPU.
+"""
+import os
+
+import pytest
+
+from sdwd_amd.config import ConfigModel, SettingsModel, WorkerModel, load_config, save_config
+from sdwd_amd.core import (
+    GenRequest,
+    IllegalTransition,
+    Job,
+    State,
+    StateMachine,
+    Worker,
+    World,
+    fix_seed,
+    shard_seeds,
+)
+from sdwd_amd.core.eta import EtaPredictor
+
+
+# -- state machine (ref C14) -------------------------------------------------
+class TestStateMachine:
+    def test_legal_cycle(self):
+        sm = StateMachine()
+        sm.set(State.WORKING)
+        sm.set(State.INTERRUPTED)
+        sm.set(State.IDLE)
+        assert sm.state is State.IDLE
+
+    def test_illegal_raises(self):
+        sm = StateMachine(State.UNAVAILABLE)
+        with pytest.raises(IllegalTransition):
+            sm.set(State.WORKING, strict=True)
+
+    def test_illegal_nonstrict_refuses(self):
+        sm = StateMachine(State.DISABLED)
+        assert not sm.set(State.WORKING, strict=False)
+        assert sm.state is State.DISABLED
+
+    def test_hooks_fire(self):
+        sm = StateMachine()
+        seen = []
+        sm.on_transition(lambda a, b: seen.append((a, b)))
+        sm.set(State.WORKING)
+        assert seen == [(State.IDLE, State.WORKING)]
+
+
+# -- ETA predictor (ref C6) ----------------------------------------------------
+class TestEta:
+    def test_canonical_identity(self):
+        e = EtaPredictor(avg_ipm=60.0)  # 1 image per second
+        assert e.eta(1) == pytest.approx(1.0)
+        assert e.eta(10) == pytest.approx(10.0)
+
+    def test_scaling(self):
+        e = EtaPredictor(avg_ipm=60.0)
+        assert e.eta(1, steps=40) == pytest.approx(2.0)
+        assert e.eta(1, width=1024, height=1024) == pytest.approx(4.0)
+        assert e.eta(1, sampler_name="Heun") == pytest.approx(2.0)
+
+    def test_hr_pass(self):
+        e = EtaPredictor(avg_ipm=60.0)
+        # base 20 steps at 512 + hr 10 steps at 2x scale (4x pixels)
+        assert e.eta(1, hr_scale=2.0, hr_steps=10) == pytest.approx(1.0 + 2.0)
+
+    def test_mpe_correction(self):
+        e = EtaPredictor(avg_ipm=60.0)
+        for _ in range(5):
+            e.record_outcome(predicted=1.0, actual=1.5)  # +50% error
+        assert e.mpe() == pytest.approx(50.0)
+        assert e.eta(1) == pytest.approx(1.5)
+
+    def test_mpe_outlier_discarded(self):
+        e = EtaPredictor(avg_ipm=60.0)
+        e.record_outcome(predicted=1.0, actual=100.0)  # +9900% -> discarded
+        assert e.mpe() == 0.0
+
+    def test_mpe_window(self):
+        e = EtaPredictor(avg_ipm=60.0)
+        for i in range(10):
+            e.record_outcome(1.0, 1.0 + 0.1 * i)
+        assert len(e.percent_errors) == 5
+
+
+# -- seed plan (ref C22) -------------------------------------------------------
+class TestSeeds:
+    def test_fix_seed_stable(self):
+        assert fix_seed(1234) == 1234
+
+    def test_fix_seed_random(self):
+        s = fix_seed(-1)
+        assert 0 <= s < 2**32
+
+    def test_contiguous_offsets(self):
+        sp = shard_seeds(100, offset=3, count=4)
+        assert sp.seeds == [103, 104, 105, 106]
+
+    def test_subseed_mode(self):
+        sp = shard_seeds(100, offset=2, count=3, subseed=7, subseed_strength=0.5)
+        assert sp.seeds == [100, 100, 100]
+        assert sp.subseeds == [9, 10, 11]
+
+    def test_gallery_equals_single_gpu(self):
+        """N shards concatenated == the 1-GPU seed sequence."""
+        whole = shard_seeds(42, 0, 8).seeds
+        parts = shard_seeds(42, 0, 3).seeds + shard_seeds(42, 3, 5).seeds
+        assert parts == whole
+
+
+# -- optimizer (ref C3) --------------------------------------------------------
+def make_world(ipms, **settings):
+    world = World(settings=SettingsModel(**settings))
+    for i, ipm in enumerate(ipms):
+        w = Worker(label=f"gpu{i}", device=i, avg_ipm=ipm, is_master=(i == 0))
+        world.add_worker(w)
+    return world
+
+
+class TestOptimizer:
+    def test_homogeneous_equal_split(self):
+        world = make_world([30.0] * 8)
+        jobs = world.make_jobs(GenRequest(batch_size=64))
+        assert len(jobs) == 8
+        assert all(j.batch_size == 8 for j in jobs)
+        assert not any(j.complementary for j in jobs)
+
+    def test_remainder_round_robin(self):
+        world = make_world([30.0] * 8)
+        jobs = world.make_jobs(GenRequest(batch_size=67))
+        sizes = sorted(j.batch_size for j in jobs)
+        assert sizes == [8, 8, 8, 8, 8, 9, 9, 9]
+        assert sum(sizes) == 67
+
+    def test_unbenchmarked_equal_split(self):
+        world = make_world([0.0, 0.0, 0.0, 0.0])
+        jobs = world.make_jobs(GenRequest(batch_size=10))
+        assert sorted(j.batch_size for j in jobs) == [2, 2, 3, 3]
+
+    def test_weighted_split(self):
+        world = make_world([60.0, 30.0], complement_production=False)
+        jobs = world.make_jobs(GenRequest(batch_size=9))
+        by = {j.worker_label: j.batch_size for j in jobs}
+        assert by["gpu0"] == 6 and by["gpu1"] == 3
+
+    def test_slow_rank_deferred(self):
+        # gpu2 is 100x slower: even one image stalls the others beyond
+        # job_timeout -> deferred to complementary with bonus images.
+        world = make_world([60.0, 60.0, 0.6], job_timeout=1.0)
+        jobs = world.make_jobs(GenRequest(batch_size=16))
+        comp = [j for j in jobs if j.complementary]
+        real = [j for j in jobs if not j.complementary]
+        assert {j.worker_label for j in real} == {"gpu0", "gpu1"}
+        assert sum(j.batch_size for j in real) == 16
+        # the slow rank got nothing or a small bonus shard
+        for j in comp:
+            assert j.worker_label == "gpu2"
+
+    def test_step_scaling_for_hopeless_rank(self):
+        world = make_world(
+            [60.0, 60.0, 0.06],
+            job_timeout=1.0,
+            step_scaling=True,
+        )
+        jobs = world.make_jobs(GenRequest(batch_size=16, steps=20))
+        comp = [j for j in jobs if j.complementary]
+        if comp:  # the hopeless rank only participates with scaled steps
+            assert comp[0].step_override is not None
+            assert comp[0].step_override < 20
+
+    def test_pixel_cap_redistribution(self):
+        world = make_world([30.0, 30.0], complement_production=False)
+        world.get_worker("gpu1").pixel_cap = 2 * 512 * 512  # max 2 images
+        jobs = world.make_jobs(GenRequest(batch_size=8))
+        by = {j.worker_label: j.batch_size for j in jobs}
+        assert by["gpu1"] == 2
+        assert by["gpu0"] == 6
+
+    def test_thin_client_excludes_master(self):
+        world = make_world([30.0] * 3, thin_client=True)
+        jobs = world.make_jobs(GenRequest(batch_size=4))
+        assert all(j.worker_label != "gpu0" for j in jobs)
+        assert sum(j.batch_size for j in jobs) == 4
+
+    def test_gallery_offsets_contiguous_and_seeded(self):
+        world = make_world([30.0, 60.0, 30.0])
+        req = GenRequest(batch_size=10, seed=1000)
+        jobs = world.make_jobs(req)
+        real = sorted(
+            (j for j in jobs if not j.complementary),
+            key=lambda j: j.gallery_offset,
+        )
+        expect_off = 0
+        all_seeds = []
+        for j in real:
+            assert j.gallery_offset == expect_off
+            expect_off += j.batch_size
+            all_seeds.extend(j.seeds)
+        assert all_seeds == [1000 + i for i in range(10)]
+
+    def test_disabled_rank_excluded(self):
+        world = make_world([30.0, 30.0])
+        world.get_worker("gpu1").set_state(State.DISABLED)
+        jobs = world.make_jobs(GenRequest(batch_size=4))
+        assert len(jobs) == 1 and jobs[0].worker_label == "gpu0"
+
+
+class TestRequeue:
+    def test_failed_shard_requeued_with_seeds(self):
+        world = make_world([30.0, 30.0, 30.0], complement_production=False)
+        req = GenRequest(batch_size=9, seed=500)
+        jobs = world.make_jobs(req)
+        victim = jobs[1]
+        replacement = world.requeue_failed(victim, req)
+        assert world.get_worker(victim.worker_label).state is State.UNAVAILABLE
+        assert sum(j.batch_size for j in replacement) == victim.batch_size
+        seeds = []
+        for j in sorted(replacement, key=lambda j: j.gallery_offset):
+            seeds.extend(j.seeds)
+        assert seeds == victim.seeds
+
+    def test_recovery_flips_idle(self):
+        world = make_world([30.0, 30.0])
+        w = world.get_worker("gpu1")
+        w.set_state(State.UNAVAILABLE)
+        # no GPU here: reachable() is False, stays UNAVAILABLE
+        world.ping()
+        assert w.state is State.UNAVAILABLE
+        # simulate a successful probe
+        w.reachable = lambda: True  # type: ignore
+        world.ping()
+        assert w.state is State.IDLE
+
+
+# -- interrupt (ref C20) -------------------------------------------------------
+class TestInterrupt:
+    def test_interrupt_propagates(self):
+        world = make_world([30.0, 30.0])
+        for w in world.workers:
+            w.set_state(State.WORKING)
+        world.interrupt_all()
+        assert world.interrupted.is_set()
+        assert all(w.state is State.INTERRUPTED for w in world.workers)
+        assert all(w.interrupt_event.is_set() for w in world.workers)
+        world.clear_interrupt()
+        assert all(w.state is State.IDLE for w in world.workers)
+
+
+# -- config round-trip (ref C12) -----------------------------------------------
+class TestConfig:
+    def test_round_trip(self, tmp_path):
+        path = str(tmp_path / "cfg.json")
+        world = make_world([30.0, 45.0])
+        world.get_worker("gpu1").pixel_cap = 123456
+        world.get_worker("gpu0").eta.record_outcome(1.0, 1.2)
+        world.save(path)
+
+        world2 = World()
+        world2.load(path)
+        assert len(world2.workers) == 2
+        w1 = world2.get_worker("gpu1")
+        assert w1.eta.avg_ipm == 45.0
+        assert w1.pixel_cap == 123456
+        w0 = world2.get_worker("gpu0")
+        assert w0.eta.mpe() == pytest.approx(20.0)
+        assert w0.is_master
+
+    def test_missing_file_defaults(self, tmp_path):
+        cfg = load_config(str(tmp_path / "nope.json"))
+        assert isinstance(cfg, ConfigModel)
+        assert cfg.workers == []
+
+    def test_corrupt_file_defaults(self, tmp_path):
+        p = tmp_path / "bad.json"
+        p.write_text("{not json")
+        cfg = load_config(str(p))
+        assert cfg.workers == []
+
+    def test_legacy_migration(self, tmp_path):
+        p = tmp_path / "old.json"
+        p.write_text(
+            '{"workers": {"gpu0": {"avg_ipm": 12.0}}, "version": 0}'
+        )
+        cfg = load_config(str(p))
+        assert cfg.workers[0].label == "gpu0"
+        assert cfg.workers[0].avg_ipm == 12.0
+
+
+# -- benchmark engine (ref C8) --------------------------------------------------
+class TestBenchmark:
+    def test_benchmark_updates_ipm(self):
+        world = make_world([0.0, 0.0])
+
+        def fake_runner(worker, payload):
+            return 2.0 if worker.label == "gpu0" else 4.0  # seconds/sample
+
+        speeds = world.benchmark(fake_runner)
+        assert speeds["gpu0"] == pytest.approx(30.0)  # 1 img / 2 s = 30 ipm
+        assert speeds["gpu1"] == pytest.approx(15.0)
+
+    def test_no_rebenchmark_when_fresh(self):
+        world = make_world([30.0])
+        calls = []
+
+        def runner(worker, payload):
+            calls.append(worker.label)
+            return 1.0
+
+        world.benchmark(runner, rebenchmark=False)
+        assert calls == []
+        world.benchmark(runner, rebenchmark=True)
+        assert calls  # 2 warmup + 3 timed
+
+
+class TestJob:
+    def test_add_work_pixel_cap(self):
+        j = Job(worker_label="x")
+        took = j.add_work(10, 512, 512, pixel_cap=3 * 512 * 512)
+        assert took == 3 and j.batch_size == 3
