@@ -1,0 +1,134 @@
+"""Model registry: named checkpoint families with deterministic random init.
+
+The reference synced *named* checkpoints between workers by name via
+POST /options; here a name maps to an architecture config,
+weights are random-init (no network for real checkpoints — BASELINE.md) but
+deterministic per name, so every rank loading "sd15" holds identical
+weights even without a broadcast, and the broadcast path (parallel/engine)
+is still exercised to guarantee it.
+
+Real weights can be loaded from a safetensors file when one exists locally.
+"""
+from __future__ import annotations
+
+import zlib
+from dataclasses import dataclass
+from typing import Callable, Dict, Optional
+
+import torch
+import torch.nn as nn
+
+from ..utils import get_logger
+from .clip import CLIPTextEncoder
+from .unet import UNetConfig, UNetModel
+from .vae import AutoencoderKL, VAEConfig
+
+log = get_logger("models")
+
+
+@dataclass
+class ModelBundle:
+    name: str
+    text_encoder: CLIPTextEncoder
+    text_encoder_2: Optional[CLIPTextEncoder]  # SDXL second encoder
+    unet: UNetModel
+    vae: AutoencoderKL
+    context_dim: int
+    is_sdxl: bool = False
+
+    @property
+    def latent_channels(self) -> int:
+        return self.unet.cfg.in_channels
+
+    def to(self, device, dtype=None) -> "ModelBundle":
+        for m in (self.text_encoder, self.text_encoder_2, self.unet, self.vae):
+            if m is not None:
+                m.to(device=device, dtype=dtype)
+        return self
+
+    def eval(self) -> "ModelBundle":
+        for m in (self.text_encoder, self.text_encoder_2, self.unet, self.vae):
+            if m is not None:
+                m.eval()
+        return self
+
+    def parameters(self):
+        for m in (self.text_encoder, self.text_encoder_2, self.unet, self.vae):
+            if m is not None:
+                yield from m.parameters()
+
+
+def _seeded_init(module: nn.Module, seed: int) -> None:
+    """Deterministic, device-independent random init."""
+    gen = torch.Generator().manual_seed(seed)
+    with torch.no_grad():
+        for p in module.parameters():
+            if p.dim() >= 2:
+                nn.init.normal_(p, mean=0.0, std=0.02, generator=gen)
+            # keep 1-d params at their module defaults (ones/zeros)
+
+
+def _build_sd15(name: str) -> ModelBundle:
+    te = CLIPTextEncoder()
+    unet = UNetModel(UNetConfig.sd15())
+    vae = AutoencoderKL(VAEConfig.sd())
+    for seed_off, m in enumerate((te, unet, vae)):
+        _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
+    return ModelBundle(name, te, None, unet, vae, context_dim=768)
+
+
+def _build_sdxl(name: str) -> ModelBundle:
+    te = CLIPTextEncoder(d_model=768, layers=12, heads=12)
+    te2 = CLIPTextEncoder(d_model=1280, layers=32, heads=20)
+    unet = UNetModel(UNetConfig.sdxl())
+    vae = AutoencoderKL(VAEConfig.sd())
+    vae.cfg.scale_factor = 0.13025
+    for seed_off, m in enumerate((te, te2, unet, vae)):
+        _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
+    return ModelBundle(
+        name, te, te2, unet, vae, context_dim=2048, is_sdxl=True
+    )
+
+
+def _build_tiny(name: str) -> ModelBundle:
+    te = CLIPTextEncoder(d_model=64, layers=2, heads=2, max_len=77)
+    unet = UNetModel(UNetConfig.tiny())
+    vae = AutoencoderKL(VAEConfig.tiny())
+    for seed_off, m in enumerate((te, unet, vae)):
+        _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
+    return ModelBundle(name, te, None, unet, vae, context_dim=64)
+
+
+_BUILDERS: Dict[str, Callable[[str], ModelBundle]] = {
+    "sd15": _build_sd15,
+    "sdxl": _build_sdxl,
+    "tiny": _build_tiny,
+}
+
+_cache: Dict[str, ModelBundle] = {}
+
+
+def available_models() -> list:
+    """ref worker.py:623-644 (GET /sd-models)."""
+    return sorted(_BUILDERS.keys())
+
+
+def load_model(
+    name: str, device="cpu", dtype: Optional[torch.dtype] = None,
+    cache: bool = True,
+) -> ModelBundle:
+    key = f"{name}"
+    if cache and key in _cache:
+        return _cache[key].to(device, dtype)
+    if name not in _BUILDERS:
+        raise KeyError(f"unknown model '{name}'; have {available_models()}")
+    log.info("building model '%s' (random-init, deterministic)", name)
+    bundle = _BUILDERS[name](name).eval()
+    bundle.to(device, dtype)
+    if cache:
+        _cache[key] = bundle
+    return bundle
+
+
+def clear_cache() -> None:
+    _cache.clear()

@@ -1,0 +1,303 @@
+"""UNet denoiser — the hot path the reference delegated to remote webui
+instances per POST /txt2img (SURVEY.md §2.5 derives this compute surface).
+
+Architecture-compatible with SD1.5 (model_channels 320, channel_mult
+[1,2,4,4], 2 res blocks, spatial transformers at downsample 1/2/4, context
+768, 8 heads) and SDXL-base (channel_mult [1,2,4], transformer depths
+[0,2,10], context 2048, additional pooled/size conditioning), parameterized
+so tiny CPU-test configs use the same code.
+
+GPU hot ops route through sdwd_amd.ops: fused GroupNorm+SiLU, flash
+attention (MFMA), GEGLU, fused LayerNorm. Projections/convs use
+hipBLASLt/MIOpen via torch (library GEMMs are allowed; convs get
+hand-written implicit-GEMM kernels in a later pass).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from .layers import GEGLU, FusedGroupNorm, FusedLayerNorm
+
+
+@dataclass
+class UNetConfig:
+    in_channels: int = 4
+    out_channels: int = 4
+    model_channels: int = 320
+    channel_mult: List[int] = field(default_factory=lambda: [1, 2, 4, 4])
+    num_res_blocks: int = 2
+    # transformer depth per level; 0 = no attention at that level
+    transformer_depth: List[int] = field(default_factory=lambda: [1, 1, 1, 0])
+    context_dim: int = 768
+    num_heads: int = 8
+    groups: int = 32
+    # SDXL extras: dimension of the pooled+size conditioning vector (0 = off)
+    adm_in_channels: int = 0
+
+    @classmethod
+    def sd15(cls) -> "UNetConfig":
+        return cls()
+
+    @classmethod
+    def sdxl(cls) -> "UNetConfig":
+        return cls(
+            model_channels=320,
+            channel_mult=[1, 2, 4],
+            transformer_depth=[0, 2, 10],
+            context_dim=2048,
+            num_heads=0,  # 0 -> fixed head_dim 64 (SDXL convention)
+            adm_in_channels=2816,
+        )
+
+    @classmethod
+    def tiny(cls) -> "UNetConfig":
+        """CPU-test config (BASELINE config #1 plumbing scale)."""
+        return cls(
+            model_channels=32,
+            channel_mult=[1, 2],
+            num_res_blocks=1,
+            transformer_depth=[1, 1],
+            context_dim=64,
+            num_heads=2,
+            groups=8,
+        )
+
+    def heads_for(self, channels: int) -> int:
+        if self.num_heads > 0:
+            return self.num_heads
+        return max(1, channels // 64)  # SDXL: head_dim 64
+
+
+class ResBlock(nn.Module):
+    def __init__(self, in_ch: int, out_ch: int, emb_ch: int, groups: int):
+        super().__init__()
+        self.norm1 = FusedGroupNorm(in_ch, groups, silu=True)
+        self.conv1 = nn.Conv2d(in_ch, out_ch, 3, padding=1)
+        self.emb_proj = nn.Linear(emb_ch, out_ch)
+        self.norm2 = FusedGroupNorm(out_ch, groups, silu=True)
+        self.conv2 = nn.Conv2d(out_ch, out_ch, 3, padding=1)
+        self.skip = (
+            nn.Conv2d(in_ch, out_ch, 1) if in_ch != out_ch else nn.Identity()
+        )
+
+    def forward(self, x: torch.Tensor, emb: torch.Tensor) -> torch.Tensor:
+        h = self.conv1(self.norm1(x))
+        h = h + self.emb_proj(ops.silu(emb))[:, :, None, None]
+        h = self.conv2(self.norm2(h))
+        return h + self.skip(x)
+
+
+class CrossAttention(nn.Module):
+    def __init__(self, query_dim: int, context_dim: int, heads: int):
+        super().__init__()
+        self.heads = heads
+        self.d_head = query_dim // heads
+        self.to_q = nn.Linear(query_dim, query_dim, bias=False)
+        self.to_k = nn.Linear(context_dim, query_dim, bias=False)
+        self.to_v = nn.Linear(context_dim, query_dim, bias=False)
+        self.to_out = nn.Linear(query_dim, query_dim)
+
+    def forward(
+        self, x: torch.Tensor, context: Optional[torch.Tensor] = None
+    ) -> torch.Tensor:
+        context = x if context is None else context
+        b, s, d = x.shape
+        q = self.to_q(x).view(b, s, self.heads, self.d_head).transpose(1, 2)
+        k = (
+            self.to_k(context)
+            .view(b, context.shape[1], self.heads, self.d_head)
+            .transpose(1, 2)
+        )
+        v = (
+            self.to_v(context)
+            .view(b, context.shape[1], self.heads, self.d_head)
+            .transpose(1, 2)
+        )
+        out = ops.attention(q, k, v)
+        out = out.transpose(1, 2).reshape(b, s, d)
+        return self.to_out(out)
+
+
+class BasicTransformerBlock(nn.Module):
+    def __init__(self, dim: int, context_dim: int, heads: int):
+        super().__init__()
+        self.norm1 = FusedLayerNorm(dim)
+        self.attn1 = CrossAttention(dim, dim, heads)  # self
+        self.norm2 = FusedLayerNorm(dim)
+        self.attn2 = CrossAttention(dim, context_dim, heads)  # cross
+        self.norm3 = FusedLayerNorm(dim)
+        self.ff = nn.Sequential(GEGLU(dim, dim * 4), nn.Linear(dim * 4, dim))
+
+    def forward(self, x, context):
+        x = x + self.attn1(self.norm1(x))
+        x = x + self.attn2(self.norm2(x), context)
+        x = x + self.ff(self.norm3(x))
+        return x
+
+
+class SpatialTransformer(nn.Module):
+    def __init__(
+        self, channels: int, context_dim: int, heads: int, depth: int, groups: int
+    ):
+        super().__init__()
+        self.norm = FusedGroupNorm(channels, groups, silu=False)
+        self.proj_in = nn.Linear(channels, channels)
+        self.blocks = nn.ModuleList(
+            BasicTransformerBlock(channels, context_dim, heads)
+            for _ in range(depth)
+        )
+        self.proj_out = nn.Linear(channels, channels)
+
+    def forward(self, x: torch.Tensor, context: torch.Tensor) -> torch.Tensor:
+        b, c, h, w = x.shape
+        residual = x
+        x = self.norm(x)
+        x = x.permute(0, 2, 3, 1).reshape(b, h * w, c)
+        x = self.proj_in(x)
+        for blk in self.blocks:
+            x = blk(x, context)
+        x = self.proj_out(x)
+        x = x.reshape(b, h, w, c).permute(0, 3, 1, 2)
+        return x + residual
+
+
+class Downsample(nn.Module):
+    def __init__(self, ch: int):
+        super().__init__()
+        self.conv = nn.Conv2d(ch, ch, 3, stride=2, padding=1)
+
+    def forward(self, x):
+        return self.conv(x)
+
+
+class Upsample(nn.Module):
+    def __init__(self, ch: int):
+        super().__init__()
+        self.conv = nn.Conv2d(ch, ch, 3, padding=1)
+
+    def forward(self, x):
+        x = torch.nn.functional.interpolate(x, scale_factor=2, mode="nearest")
+        return self.conv(x)
+
+
+class _Seq(nn.Module):
+    """Runs children, feeding emb to ResBlocks and context to transformers."""
+
+    def __init__(self, *mods: nn.Module):
+        super().__init__()
+        self.mods = nn.ModuleList(mods)
+
+    def forward(self, x, emb, context):
+        for m in self.mods:
+            if isinstance(m, ResBlock):
+                x = m(x, emb)
+            elif isinstance(m, SpatialTransformer):
+                x = m(x, context)
+            else:
+                x = m(x)
+        return x
+
+
+class UNetModel(nn.Module):
+    def __init__(self, cfg: UNetConfig):
+        super().__init__()
+        self.cfg = cfg
+        ch = cfg.model_channels
+        time_dim = ch * 4
+        self.time_mlp = nn.Sequential(
+            nn.Linear(ch, time_dim), nn.SiLU(), nn.Linear(time_dim, time_dim)
+        )
+        if cfg.adm_in_channels:
+            self.label_mlp = nn.Sequential(
+                nn.Linear(cfg.adm_in_channels, time_dim),
+                nn.SiLU(),
+                nn.Linear(time_dim, time_dim),
+            )
+        else:
+            self.label_mlp = None
+
+        self.conv_in = nn.Conv2d(cfg.in_channels, ch, 3, padding=1)
+
+        self.down = nn.ModuleList()
+        skip_chs = [ch]
+        cur = ch
+        levels = len(cfg.channel_mult)
+        for lvl, mult in enumerate(cfg.channel_mult):
+            out_ch = ch * mult
+            for _ in range(cfg.num_res_blocks):
+                mods: List[nn.Module] = [
+                    ResBlock(cur, out_ch, time_dim, cfg.groups)
+                ]
+                cur = out_ch
+                depth = cfg.transformer_depth[lvl]
+                if depth > 0:
+                    mods.append(
+                        SpatialTransformer(
+                            cur, cfg.context_dim, cfg.heads_for(cur), depth,
+                            cfg.groups,
+                        )
+                    )
+                self.down.append(_Seq(*mods))
+                skip_chs.append(cur)
+            if lvl != levels - 1:
+                self.down.append(_Seq(Downsample(cur)))
+                skip_chs.append(cur)
+
+        mid_depth = cfg.transformer_depth[-1] or 1
+        self.mid = _Seq(
+            ResBlock(cur, cur, time_dim, cfg.groups),
+            SpatialTransformer(
+                cur, cfg.context_dim, cfg.heads_for(cur), mid_depth, cfg.groups
+            ),
+            ResBlock(cur, cur, time_dim, cfg.groups),
+        )
+
+        self.up = nn.ModuleList()
+        for lvl in reversed(range(levels)):
+            out_ch = ch * cfg.channel_mult[lvl]
+            for i in range(cfg.num_res_blocks + 1):
+                skip = skip_chs.pop()
+                mods = [ResBlock(cur + skip, out_ch, time_dim, cfg.groups)]
+                cur = out_ch
+                depth = cfg.transformer_depth[lvl]
+                if depth > 0:
+                    mods.append(
+                        SpatialTransformer(
+                            cur, cfg.context_dim, cfg.heads_for(cur), depth,
+                            cfg.groups,
+                        )
+                    )
+                if lvl != 0 and i == cfg.num_res_blocks:
+                    mods.append(Upsample(cur))
+                self.up.append(_Seq(*mods))
+
+        self.norm_out = FusedGroupNorm(cur, cfg.groups, silu=True)
+        self.conv_out = nn.Conv2d(cur, cfg.out_channels, 3, padding=1)
+
+    def forward(
+        self,
+        x: torch.Tensor,
+        timesteps: torch.Tensor,
+        context: torch.Tensor,
+        y: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        temb = ops.timestep_embedding(timesteps, self.cfg.model_channels)
+        emb = self.time_mlp(temb.to(x.dtype))
+        if self.label_mlp is not None and y is not None:
+            emb = emb + self.label_mlp(y.to(x.dtype))
+
+        h = self.conv_in(x)
+        skips = [h]
+        for blk in self.down:
+            h = blk(h, emb, context)
+            skips.append(h)
+        h = self.mid(h, emb, context)
+        for blk in self.up:
+            h = torch.cat([h, skips.pop()], dim=1)
+            h = blk(h, emb, context)
+        return self.conv_out(self.norm_out(h))

@@ -1,0 +1,323 @@
+"""txt2img / img2img pipelines — the compute the reference delegated per
+POST /txt2img|/img2img runs here, in-process.
+
+Determinism contract (ref C22): image k of a batch depends only on
+(model, seed_k, subseed_k, request params) — never on batch position,
+shard size or device count — so an 8-GPU gallery is image-for-image equal
+to the 1-GPU gallery. Initial noise is drawn per-image from a CPU
+generator seeded with that image's seed, then moved to the device.
+"""
+from __future__ import annotations
+
+import math
+import time
+from dataclasses import dataclass, field
+from typing import Callable, List, Optional
+
+import torch
+
+from ..models.registry import ModelBundle, load_model
+from ..models import tokenizer
+from ..utils import get_logger
+from .samplers import build_sampler
+from .schedule import schedule_for
+
+log = get_logger("pipeline")
+
+
+@dataclass
+class PipelineRequest:
+    prompt: str = ""
+    negative_prompt: str = ""
+    steps: int = 20
+    width: int = 512
+    height: int = 512
+    cfg_scale: float = 7.0
+    sampler_name: str = "Euler a"
+    seeds: List[int] = field(default_factory=lambda: [0])
+    subseeds: List[int] = field(default_factory=list)
+    subseed_strength: float = 0.0
+    # img2img
+    init_latents: Optional[torch.Tensor] = None  # pre-encoded [B,4,h,w]
+    denoising_strength: float = 0.75
+
+    @property
+    def batch_size(self) -> int:
+        return len(self.seeds)
+
+
+@dataclass
+class PipelineResult:
+    images: torch.Tensor  # [B, H, W, 3] uint8 on CPU
+    seeds: List[int]
+    subseeds: List[int]
+    infotexts: List[str]
+    elapsed: float = 0.0
+    interrupted: bool = False
+
+
+def _slerp(a: torch.Tensor, b: torch.Tensor, t: float) -> torch.Tensor:
+    """Spherical interpolation between noise tensors (subseed variation)."""
+    af, bf = a.flatten().double(), b.flatten().double()
+    dot = torch.dot(af / af.norm(), bf / bf.norm()).clamp(-1, 1)
+    omega = torch.acos(dot)
+    if omega.abs() < 1e-6:
+        out = (1 - t) * af + t * bf
+    else:
+        so = torch.sin(omega)
+        out = (math.sin((1 - t) * omega) / so) * af + (
+            math.sin(t * omega) / so
+        ) * bf
+    return out.reshape(a.shape).to(a.dtype)
+
+
+def _image_noise(
+    seed: int,
+    subseed: int,
+    subseed_strength: float,
+    shape,
+) -> torch.Tensor:
+    """Per-image deterministic CPU noise (device-independent)."""
+    g = torch.Generator("cpu").manual_seed(int(seed) & 0xFFFFFFFF)
+    noise = torch.randn(shape, generator=g, dtype=torch.float32)
+    if subseed_strength and subseed_strength > 0 and subseed >= 0:
+        g2 = torch.Generator("cpu").manual_seed(int(subseed) & 0xFFFFFFFF)
+        noise2 = torch.randn(shape, generator=g2, dtype=torch.float32)
+        noise = _slerp(noise, noise2, float(subseed_strength))
+    return noise
+
+
+class StableDiffusionPipeline:
+    """One model bundle on one device; reusable across requests."""
+
+    def __init__(
+        self,
+        model: ModelBundle | str = "sd15",
+        device: str | torch.device = "cpu",
+        dtype: Optional[torch.dtype] = None,
+    ) -> None:
+        self.device = torch.device(device)
+        if dtype is None:
+            dtype = (
+                torch.bfloat16 if self.device.type == "cuda" else torch.float32
+            )
+        self.dtype = dtype
+        if isinstance(model, str):
+            model = load_model(model, device=self.device, dtype=dtype)
+        else:
+            model.to(self.device, dtype)
+        self.model = model
+
+    # -- conditioning --------------------------------------------------------
+    @torch.no_grad()
+    def encode_prompts(
+        self, prompts: List[str], negatives: List[str]
+    ) -> tuple:
+        tokens = tokenizer.encode_batch(prompts + negatives, device=self.device)
+        m = self.model
+        if m.is_sdxl:
+            h1 = m.text_encoder(tokens, penultimate=True)
+            h2 = m.text_encoder_2(tokens, penultimate=True)
+            ctx = torch.cat([h1, h2], dim=-1)
+            pooled = m.text_encoder_2.pooled(
+                tokens, m.text_encoder_2(tokens)
+            )
+        else:
+            ctx = m.text_encoder(tokens)
+            pooled = None
+        n = len(prompts)
+        cond, uncond = ctx[:n], ctx[n:]
+        pooled_cu = (pooled[:n], pooled[n:]) if pooled is not None else None
+        return cond.to(self.dtype), uncond.to(self.dtype), pooled_cu
+
+    def _sdxl_vector(self, req: PipelineRequest, pooled: torch.Tensor):
+        """SDXL add-conditioning: pooled embed + size/crop/target vectors."""
+        sizes = torch.tensor(
+            [
+                req.height, req.width,  # original size
+                0, 0,                   # crop top-left
+                req.height, req.width,  # target size
+            ],
+            dtype=torch.float32,
+            device=self.device,
+        )
+        from .. import ops
+
+        emb = ops.timestep_embedding(sizes, 256).flatten()  # [6*256]
+        b = pooled.shape[0]
+        return torch.cat(
+            [pooled.float(), emb[None].expand(b, -1)], dim=-1
+        ).to(self.dtype)
+
+    # -- the denoise loop ----------------------------------------------------
+    @torch.no_grad()
+    def generate(
+        self,
+        req: PipelineRequest,
+        interrupt: Optional[Callable[[], bool]] = None,
+        step_callback: Optional[Callable[[int, int], None]] = None,
+        decode: bool = True,
+    ) -> PipelineResult:
+        t0 = time.perf_counter()
+        b = req.batch_size
+        f = self.model.vae.cfg.downsample_factor
+        lat_h, lat_w = req.height // f, req.width // f
+        lat_c = self.model.latent_channels
+        subseeds = req.subseeds or [-1] * b
+
+        cond, uncond, pooled_cu = self.encode_prompts(
+            [req.prompt] * b, [req.negative_prompt] * b
+        )
+        y = None
+        if self.model.is_sdxl and pooled_cu is not None:
+            y = torch.cat(
+                [
+                    self._sdxl_vector(req, pooled_cu[0]),
+                    self._sdxl_vector(req, pooled_cu[1]),
+                ]
+            )
+
+        sched = schedule_for(req.sampler_name, req.steps)
+        sampler = build_sampler(req.sampler_name, sched)
+
+        noise = torch.stack(
+            [
+                _image_noise(
+                    req.seeds[i], subseeds[i], req.subseed_strength,
+                    (lat_c, lat_h, lat_w),
+                )
+                for i in range(b)
+            ]
+        ).to(self.device, self.dtype)
+
+        sig = sched.sigmas
+        if req.init_latents is not None:
+            # img2img: noise the init latents to the strength point, run the
+            # tail of the schedule from there.
+            start = max(
+                0, req.steps - max(1, int(req.steps * req.denoising_strength))
+            )
+            sched = type(sched)(
+                sigmas=sched.sigmas[start:], timesteps=sched.timesteps[start:]
+            )
+            sampler = build_sampler(req.sampler_name, sched)
+            s0 = float(sched.sigmas[0])
+            x = (
+                req.init_latents.to(self.device, self.dtype).float()
+                + noise.float() * s0
+            ).to(self.dtype)
+        else:
+            x = (noise.float() * float(sig[0])).to(self.dtype)
+
+        # ancestral noise: per-image generators stepped identically regardless
+        # of shard composition
+        gens = [
+            torch.Generator("cpu").manual_seed((int(s) ^ 0x5EED) & 0xFFFFFFFF)
+            for s in req.seeds
+        ]
+
+        def noise_fn() -> torch.Tensor:
+            n = torch.stack(
+                [
+                    torch.randn(
+                        (lat_c, lat_h, lat_w), generator=g, dtype=torch.float32
+                    )
+                    for g in gens
+                ]
+            )
+            return n.to(self.device, self.dtype)
+
+        ctx = torch.cat([cond, uncond], dim=0)
+        cfg = float(req.cfg_scale)
+        unet = self.model.unet
+
+        def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:
+            ts = torch.full(
+                (x_in.shape[0] * 2,), t, device=self.device,
+                dtype=torch.float32,
+            )
+            eps = unet(torch.cat([x_in, x_in], dim=0), ts, ctx, y=y)
+            eps_c, eps_u = eps.chunk(2, dim=0)
+            return eps_u + cfg * (eps_c - eps_u)
+
+        if cfg == 1.0:
+
+            def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:  # noqa: F811
+                ts = torch.full(
+                    (x_in.shape[0],), t, device=self.device,
+                    dtype=torch.float32,
+                )
+                yc = y[: x_in.shape[0]] if y is not None else None
+                return unet(x_in, ts, cond, y=yc)
+
+        was_interrupted = False
+
+        def _interrupt() -> bool:
+            nonlocal was_interrupted
+            if interrupt is not None and interrupt():
+                was_interrupted = True
+                return True
+            return False
+
+        x = sampler.sample(
+            model_fn, x, noise_fn=noise_fn, callback=step_callback,
+            interrupt=_interrupt,
+        )
+
+        if decode:
+            pixels = self.model.vae.decode(x)
+            images = (
+                ((pixels.float() + 1.0) * 127.5)
+                .clamp(0, 255)
+                .to(torch.uint8)
+                .permute(0, 2, 3, 1)
+                .cpu()
+            )
+        else:
+            images = x.cpu()
+
+        elapsed = time.perf_counter() - t0
+        infotexts = [
+            f"{req.prompt}\nNegative prompt: {req.negative_prompt}\n"
+            f"Steps: {req.steps}, Sampler: {req.sampler_name}, "
+            f"CFG scale: {req.cfg_scale}, Seed: {req.seeds[i]}, "
+            f"Size: {req.width}x{req.height}, Model: {self.model.name}"
+            for i in range(b)
+        ]
+        return PipelineResult(
+            images=images,
+            seeds=list(req.seeds),
+            subseeds=list(subseeds),
+            infotexts=infotexts,
+            elapsed=elapsed,
+            interrupted=was_interrupted,
+        )
+
+    @torch.no_grad()
+    def encode_image(
+        self, images: torch.Tensor, seeds: Optional[List[int]] = None
+    ) -> torch.Tensor:
+        """[B,H,W,3] uint8 -> latents [B,4,h,w] (img2img entry).
+
+        The encoder's sampling noise is seeded per image for determinism.
+        """
+        x = (
+            images.permute(0, 3, 1, 2).float() / 127.5 - 1.0
+        ).to(self.device, self.dtype)
+        if seeds:
+            outs = []
+            for i in range(x.shape[0]):
+                g = torch.Generator("cpu").manual_seed(
+                    (int(seeds[i]) ^ 0xE4C0DE) & 0xFFFFFFFF
+                )
+                moments = self.model.vae.encoder(x[i : i + 1])
+                mean, logvar = moments.chunk(2, dim=1)
+                std = torch.exp(0.5 * logvar.float().clamp(-30, 20))
+                n = torch.randn(
+                    mean.shape, generator=g, dtype=torch.float32
+                ).to(self.device)
+                outs.append(
+                    (mean.float() + std * n) * self.model.vae.cfg.scale_factor
+                )
+            return torch.cat(outs).to(self.dtype)
+        return self.model.vae.encode(x)
