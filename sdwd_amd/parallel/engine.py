@@ -1,0 +1,478 @@
+"""Generation engines: the reference's dispatch/gather path (C2/C9/C10/C20)
+re-designed for one node.
+
+Two engines share the scheduler (core.World) and pipeline:
+
+* LocalEngine — one process drives N devices with one thread per rank
+  (the reference's threading model, distributed.py:316-318), full recovery
+  semantics: a failing rank's shard is requeued onto survivors
+  (improving on ref worker.py:498-500 which dropped it).
+
+* DistributedEngine — one process per GPU under torchrun, RCCL over xGMI:
+  plan broadcast from rank 0, per-rank shard execution, padded all_gather
+  of uint8 images, weight broadcast at load (C13), TCPStore out-of-band
+  control plane for interrupt (C20) — collectives never carry control.
+"""
+from __future__ import annotations
+
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+
+from ..core import GenRequest, Job, State, World, Worker
+from ..pipeline import PipelineRequest, StableDiffusionPipeline
+from ..utils import get_logger
+from ..utils.images import make_grid
+from . import group as pg
+
+log = get_logger("engine")
+
+
+@dataclass
+class GenerationRequest:
+    """User-facing request = scheduling fields + pipeline fields."""
+
+    prompt: str = ""
+    negative_prompt: str = ""
+    batch_size: int = 1
+    width: int = 512
+    height: int = 512
+    steps: int = 20
+    cfg_scale: float = 7.0
+    sampler_name: str = "Euler a"
+    seed: int = -1
+    subseed: int = -1
+    subseed_strength: float = 0.0
+    init_images: Optional[torch.Tensor] = None  # [B,H,W,3] uint8 (img2img)
+    denoising_strength: float = 0.75
+
+    def sched(self) -> GenRequest:
+        return GenRequest(
+            batch_size=self.batch_size,
+            width=self.width,
+            height=self.height,
+            steps=self.steps,
+            sampler_name=self.sampler_name,
+            seed=self.seed,
+            subseed=self.subseed,
+            subseed_strength=self.subseed_strength,
+        )
+
+
+@dataclass
+class GalleryResult:
+    images: torch.Tensor  # [N,H,W,3] uint8 cpu, requested batch first
+    seeds: List[int]
+    infotexts: List[str]
+    grid: Optional[torch.Tensor] = None
+    elapsed: float = 0.0
+    job_summary: List[str] = field(default_factory=list)
+    interrupted: bool = False
+
+
+def _job_pipeline_request(
+    gen: GenerationRequest, job: Job, init_latents=None
+) -> PipelineRequest:
+    return PipelineRequest(
+        prompt=gen.prompt,
+        negative_prompt=gen.negative_prompt,
+        steps=job.step_override or gen.steps,
+        width=gen.width,
+        height=gen.height,
+        cfg_scale=gen.cfg_scale,
+        sampler_name=gen.sampler_name,
+        seeds=list(job.seeds),
+        subseeds=list(job.subseeds),
+        subseed_strength=gen.subseed_strength,
+        init_latents=init_latents,
+        denoising_strength=gen.denoising_strength,
+    )
+
+
+class _EngineBase:
+    def _assemble(
+        self,
+        gen: GenerationRequest,
+        jobs: List[Job],
+        shards: Dict[str, torch.Tensor],
+        infos: Dict[str, List[str]],
+        elapsed: float,
+        interrupted: bool = False,
+    ) -> GalleryResult:
+        """Order shards by gallery offset, fix up infotexts with the worker
+        label (ref distributed.py:343-349), build the grid."""
+        total = sum(j.batch_size for j in jobs)
+        h, w = gen.height, gen.width
+        images = torch.zeros(total, h, w, 3, dtype=torch.uint8)
+        seeds = [0] * total
+        infotexts = [""] * total
+        summary = []
+        for job in sorted(jobs, key=lambda j: j.gallery_offset):
+            shard = shards.get(id(job), shards.get(job.worker_label))
+            o = job.gallery_offset
+            n = job.batch_size
+            if shard is not None and shard.shape[0] >= n:
+                images[o : o + n] = shard[:n]
+            job_infos = infos.get(id(job), infos.get(job.worker_label)) or []
+            for i in range(n):
+                seeds[o + i] = job.seeds[i] if i < len(job.seeds) else 0
+                base = job_infos[i] if i < len(job_infos) else ""
+                infotexts[o + i] = f"{base}\nWorker: {job.worker_label}"
+            tag = " (complementary)" if job.complementary else ""
+            summary.append(
+                f"{job.worker_label}: {n} image(s){tag}, "
+                f"predicted {job.predicted_eta:.2f}s, actual {job.elapsed:.2f}s"
+            )
+        grid = make_grid(images) if total > 1 else None
+        return GalleryResult(
+            images=images,
+            seeds=seeds,
+            infotexts=infotexts,
+            grid=grid,
+            elapsed=elapsed,
+            job_summary=summary,
+            interrupted=interrupted,
+        )
+
+    def _record_eta(self, world: World, job: Job) -> None:
+        w = world.get_worker(job.worker_label)
+        if w is not None and job.predicted_eta > 0 and job.elapsed > 0:
+            w.eta.record_outcome(job.predicted_eta, job.elapsed)
+
+
+class LocalEngine(_EngineBase):
+    """One process, one thread per rank (device)."""
+
+    def __init__(
+        self,
+        model: str = "sd15",
+        devices: Optional[List[str]] = None,
+        world: Optional[World] = None,
+        dtype: Optional[torch.dtype] = None,
+    ) -> None:
+        if devices is None:
+            if torch.cuda.is_available():
+                devices = [f"cuda:{i}" for i in range(torch.cuda.device_count())]
+            else:
+                devices = ["cpu"]
+        self.devices = devices
+        self.model_name = model
+        self.world = world or World.from_devices(len(devices))
+        self.pipes: Dict[str, StableDiffusionPipeline] = {}
+        self.dtype = dtype
+        for i, dev in enumerate(devices):
+            self.pipes[f"gpu{i}"] = StableDiffusionPipeline(
+                model, device=dev, dtype=dtype
+            )
+        self._fail_injection: Dict[str, bool] = {}
+
+    # benchmark runner wired into core.Worker.benchmark (ref C8)
+    def _bench_runner(self, worker: Worker, payload) -> float:
+        pipe = self.pipes[worker.label]
+        req = PipelineRequest(
+            prompt=payload.prompt,
+            negative_prompt=payload.negative_prompt,
+            steps=payload.steps,
+            width=payload.width,
+            height=payload.height,
+            sampler_name=payload.sampler_name,
+            seeds=list(range(payload.batch_size)),
+        )
+        t0 = time.perf_counter()
+        pipe.generate(req, decode=True)
+        if pipe.device.type == "cuda":
+            torch.cuda.synchronize(pipe.device)
+        return time.perf_counter() - t0
+
+    def benchmark(self, rebenchmark: bool = False) -> Dict[str, float]:
+        return self.world.benchmark(self._bench_runner, rebenchmark=rebenchmark)
+
+    def inject_failure(self, label: str) -> None:
+        """Test hook: make the next shard on this rank raise."""
+        self._fail_injection[label] = True
+
+    def _run_job(
+        self,
+        gen: GenerationRequest,
+        job: Job,
+        shards: Dict[str, torch.Tensor],
+        infos: Dict[str, List[str]],
+        errors: Dict[str, Exception],
+    ) -> None:
+        worker = self.world.get_worker(job.worker_label)
+        pipe = self.pipes[job.worker_label]
+        try:
+            if self._fail_injection.pop(job.worker_label, False):
+                raise RuntimeError("injected failure")
+            worker.set_state(State.WORKING)
+            init_latents = None
+            if gen.init_images is not None:
+                src = gen.init_images
+                idx = [
+                    (job.gallery_offset + i) % src.shape[0]
+                    for i in range(job.batch_size)
+                ]
+                init_latents = pipe.encode_image(src[idx], seeds=job.seeds)
+            t0 = time.perf_counter()
+            res = pipe.generate(
+                _job_pipeline_request(gen, job, init_latents),
+                interrupt=worker.interrupt_event.is_set,
+            )
+            job.elapsed = time.perf_counter() - t0
+            shards[id(job)] = res.images
+            infos[id(job)] = res.infotexts
+            self._record_eta(self.world, job)
+            worker.set_state(
+                State.INTERRUPTED if res.interrupted else State.IDLE
+            )
+        except Exception as exc:  # noqa: BLE001 - recovery path
+            log.warning("rank %s failed: %s", job.worker_label, exc)
+            errors[job.worker_label] = exc
+            worker.set_state(State.UNAVAILABLE)
+
+    def generate(self, gen: GenerationRequest) -> GalleryResult:
+        t0 = time.perf_counter()
+        self.world.clear_interrupt()
+        jobs = self.world.make_jobs(gen.sched())
+        shards: Dict[str, torch.Tensor] = {}
+        infos: Dict[str, List[str]] = {}
+        errors: Dict[str, Exception] = {}
+
+        threads = [
+            threading.Thread(
+                target=self._run_job, args=(gen, j, shards, infos, errors)
+            )
+            for j in jobs
+        ]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join()  # the gather barrier (ref distributed.py:137-142)
+
+        # recovery: requeue failed shards onto survivors (seeds preserved)
+        for job in [j for j in jobs if j.worker_label in errors]:
+            if job.complementary:
+                jobs.remove(job)
+                continue
+            try:
+                retries = self.world.requeue_failed(job, gen.sched())
+            except RuntimeError:
+                log.error("no survivors; shard lost")
+                continue
+            jobs.remove(job)
+            jobs.extend(retries)
+            rts = [
+                threading.Thread(
+                    target=self._run_job, args=(gen, j, shards, infos, errors)
+                )
+                for j in retries
+            ]
+            for t in rts:
+                t.start()
+            for t in rts:
+                t.join()
+
+        interrupted = self.world.interrupted.is_set()
+        result = self._assemble(
+            gen, jobs, shards, infos, time.perf_counter() - t0, interrupted
+        )
+        return result
+
+    def interrupt(self) -> None:
+        self.world.interrupt_all()
+
+
+class DistributedEngine(_EngineBase):
+    """One process per GPU (torchrun), RCCL over xGMI. Every rank calls
+    generate() with the same request; rank 0's plan is broadcast; images are
+    all-gathered; every rank returns the full gallery (rank 0 saves it)."""
+
+    def __init__(
+        self,
+        model: str = "sd15",
+        dtype: Optional[torch.dtype] = None,
+        backend: Optional[str] = None,
+    ) -> None:
+        self.is_dist = pg.init_group(backend=backend)
+        import torch.distributed as dist
+
+        self.rank = dist.get_rank() if self.is_dist else 0
+        self.world_size = dist.get_world_size() if self.is_dist else 1
+        if torch.cuda.is_available():
+            self.device = torch.device("cuda", pg.env_local_rank())
+            torch.cuda.set_device(self.device)
+        else:
+            self.device = torch.device("cpu")
+        self.label = f"gpu{self.rank}"
+        self.pipe = StableDiffusionPipeline(model, device=self.device, dtype=dtype)
+        # weight sync from rank 0 (C13): registry init is deterministic, the
+        # broadcast makes identity unconditional.
+        for m in (
+            self.pipe.model.text_encoder,
+            self.pipe.model.text_encoder_2,
+            self.pipe.model.unet,
+            self.pipe.model.vae,
+        ):
+            if m is not None:
+                pg.sync_weights(m)
+        self.world = World.from_devices(self.world_size)
+        self._store = None
+        if self.is_dist:
+            try:
+                from torch.distributed.distributed_c10d import _get_default_store
+
+                self._store = _get_default_store()
+            except Exception:  # pragma: no cover
+                self._store = None
+        self._interrupt_epoch = 0
+
+    # -- out-of-band interrupt (C20): TCPStore control plane -----------------
+    def interrupt(self) -> None:
+        self._interrupt_epoch += 1
+        if self._store is not None:
+            self._store.set("sdwd_interrupt", str(self._interrupt_epoch))
+
+    def _interrupted(self) -> bool:
+        if self._store is None:
+            return False
+        try:
+            if self._store.check(["sdwd_interrupt"]):
+                val = int(self._store.get("sdwd_interrupt"))
+                return val > 0
+        except Exception:
+            return False
+        return False
+
+    def _clear_interrupt(self) -> None:
+        if self._store is not None and self.rank == 0:
+            try:
+                self._store.set("sdwd_interrupt", "0")
+            except Exception:
+                pass
+
+    # -- benchmark (C8): each rank times itself, speeds are all-gathered -----
+    def benchmark(self, rebenchmark: bool = False) -> Dict[str, float]:
+        me = self.world.get_worker(self.label)
+        if rebenchmark or me.eta.avg_ipm <= 0:
+            payload = self.world.benchmark_payload
+            req = PipelineRequest(
+                prompt=payload.prompt,
+                steps=payload.steps,
+                width=payload.width,
+                height=payload.height,
+                sampler_name=payload.sampler_name,
+                seeds=list(range(payload.batch_size)),
+            )
+            from ..config import BENCHMARK_TIMED_SAMPLES, BENCHMARK_WARMUP_SAMPLES
+
+            for _ in range(BENCHMARK_WARMUP_SAMPLES):
+                self.pipe.generate(req)
+            samples = []
+            for _ in range(BENCHMARK_TIMED_SAMPLES):
+                t0 = time.perf_counter()
+                self.pipe.generate(req)
+                if self.device.type == "cuda":
+                    torch.cuda.synchronize(self.device)
+                samples.append(
+                    payload.batch_size * 60.0 / (time.perf_counter() - t0)
+                )
+            my_ipm = sum(samples) / len(samples)
+        else:
+            my_ipm = me.eta.avg_ipm
+        speeds = pg.allgather_floats([my_ipm], self.device)
+        for r, (ipm,) in enumerate(speeds):
+            self.world.get_worker(f"gpu{r}").eta.avg_ipm = ipm
+        return {w.label: w.eta.avg_ipm for w in self.world.workers}
+
+    def generate(self, gen: GenerationRequest) -> GalleryResult:
+        t0 = time.perf_counter()
+        self._clear_interrupt()
+        if self.rank == 0:
+            jobs = self.world.make_jobs(gen.sched())
+        else:
+            jobs = None
+        jobs = pg.broadcast_object(jobs)
+
+        mine = next((j for j in jobs if j.worker_label == self.label), None)
+        h, w = gen.height, gen.width
+        my_images = torch.zeros(0, h, w, 3, dtype=torch.uint8)
+        my_ok, my_elapsed = 1.0, 0.0
+        my_infos: List[str] = []
+        interrupted = False
+        if mine is not None and mine.batch_size > 0:
+            try:
+                init_latents = None
+                if gen.init_images is not None:
+                    src = gen.init_images
+                    idx = [
+                        (mine.gallery_offset + i) % src.shape[0]
+                        for i in range(mine.batch_size)
+                    ]
+                    init_latents = self.pipe.encode_image(
+                        src[idx], seeds=mine.seeds
+                    )
+                ts = time.perf_counter()
+                res = self.pipe.generate(
+                    _job_pipeline_request(gen, mine, init_latents),
+                    interrupt=self._interrupted,
+                )
+                my_elapsed = time.perf_counter() - ts
+                my_images = res.images
+                my_infos = res.infotexts
+                interrupted = res.interrupted
+            except Exception as exc:  # noqa: BLE001
+                log.warning("rank %d shard failed: %s", self.rank, exc)
+                my_ok = 0.0
+
+        # status + images to everyone (rank 0 consumes)
+        sizes = [
+            next(
+                (
+                    j.batch_size
+                    for j in jobs
+                    if j.worker_label == f"gpu{r}" and j.batch_size > 0
+                ),
+                0,
+            )
+            for r in range(self.world_size)
+        ]
+        status = pg.allgather_floats([my_ok, my_elapsed], self.device)
+        shard_dev = my_images.to(self.device)
+        gathered = pg.gather_images(shard_dev, sizes, self.device)
+        all_images = gathered.cpu() if gathered is not None else my_images
+
+        # slice the gathered stack back into per-rank shards
+        shards: Dict[str, torch.Tensor] = {}
+        infos: Dict[str, List[str]] = {}
+        pos = 0
+        for r in range(self.world_size):
+            n = sizes[r]
+            shards[f"gpu{r}"] = all_images[pos : pos + n]
+            pos += n
+        infos[self.label] = my_infos
+
+        # recovery: rank 0 re-runs failed shards locally
+        for r in range(self.world_size):
+            ok, elapsed = status[r][0], status[r][1]
+            job = next(
+                (j for j in jobs if j.worker_label == f"gpu{r}"), None
+            )
+            if job is not None:
+                job.elapsed = elapsed
+                if job.predicted_eta > 0 and elapsed > 0:
+                    self._record_eta(self.world, job)
+            if ok < 0.5 and job is not None:
+                self.world.get_worker(f"gpu{r}").set_state(State.UNAVAILABLE)
+                if self.rank == 0 and not job.complementary:
+                    log.warning(
+                        "re-running failed shard of %s on rank 0", job.worker_label
+                    )
+                    res = self.pipe.generate(_job_pipeline_request(gen, job))
+                    shards[job.worker_label] = res.images
+
+        return self._assemble(
+            gen, jobs, shards, infos, time.perf_counter() - t0, interrupted
+        )

@@ -1,0 +1,114 @@
+"""Image utilities: grid assembly (ref world.py:588-591) and PNG encoding
+(the reference shipped base64 PNGs over HTTP, distributed.py:103-106; here
+PNG is only an output format). Pure python + zlib — no Pillow in the image.
+"""
+from __future__ import annotations
+
+import math
+import struct
+import zlib
+from typing import List, Optional
+
+import torch
+
+
+def make_grid(images: torch.Tensor, rows: Optional[int] = None) -> torch.Tensor:
+    """[N,H,W,3] uint8 -> one [GH,GW,3] grid tensor (sdwui grid layout:
+    near-square, row-major)."""
+    n, h, w, c = images.shape
+    if rows is None:
+        rows = int(math.sqrt(n))
+        rows = max(1, min(rows, n))
+    cols = math.ceil(n / rows)
+    grid = torch.zeros(rows * h, cols * w, c, dtype=torch.uint8)
+    for i in range(n):
+        r, col = divmod(i, cols)
+        grid[r * h : (r + 1) * h, col * w : (col + 1) * w] = images[i]
+    return grid
+
+
+def encode_png(image: torch.Tensor) -> bytes:
+    """[H,W,3] uint8 -> PNG bytes (filter 0, zlib level 6)."""
+    h, w, c = image.shape
+    assert c == 3 and image.dtype == torch.uint8
+    raw = image.contiguous().numpy().tobytes()
+    stride = w * 3
+    scanlines = b"".join(
+        b"\x00" + raw[y * stride : (y + 1) * stride] for y in range(h)
+    )
+
+    def chunk(tag: bytes, payload: bytes) -> bytes:
+        return (
+            struct.pack(">I", len(payload))
+            + tag
+            + payload
+            + struct.pack(">I", zlib.crc32(tag + payload) & 0xFFFFFFFF)
+        )
+
+    ihdr = struct.pack(">IIBBBBB", w, h, 8, 2, 0, 0, 0)
+    return (
+        b"\x89PNG\r\n\x1a\n"
+        + chunk(b"IHDR", ihdr)
+        + chunk(b"IDAT", zlib.compress(scanlines, 6))
+        + chunk(b"IEND", b"")
+    )
+
+
+def save_png(image: torch.Tensor, path: str) -> str:
+    with open(path, "wb") as fh:
+        fh.write(encode_png(image))
+    return path
+
+
+def decode_png(data: bytes) -> torch.Tensor:
+    """PNG (8-bit RGB, filter 0/1/2/3/4) -> [H,W,3] uint8. Minimal reader
+    for round-tripping our own encoder's output and API uploads."""
+    assert data[:8] == b"\x89PNG\r\n\x1a\n", "not a PNG"
+    pos = 8
+    idat = b""
+    w = h = 0
+    while pos < len(data):
+        (length,) = struct.unpack(">I", data[pos : pos + 4])
+        tag = data[pos + 4 : pos + 8]
+        payload = data[pos + 8 : pos + 8 + length]
+        if tag == b"IHDR":
+            w, h, depth, color = struct.unpack(">IIBB", payload[:10])
+            if depth != 8 or color != 2:
+                raise ValueError("only 8-bit RGB PNGs supported")
+        elif tag == b"IDAT":
+            idat += payload
+        elif tag == b"IEND":
+            break
+        pos += 12 + length
+    raw = zlib.decompress(idat)
+    stride = w * 3
+    out = bytearray(h * stride)
+    prev = bytearray(stride)
+    for y in range(h):
+        f = raw[y * (stride + 1)]
+        line = bytearray(raw[y * (stride + 1) + 1 : (y + 1) * (stride + 1)])
+        if f == 1:  # sub
+            for i in range(3, stride):
+                line[i] = (line[i] + line[i - 3]) & 0xFF
+        elif f == 2:  # up
+            for i in range(stride):
+                line[i] = (line[i] + prev[i]) & 0xFF
+        elif f == 3:  # average
+            for i in range(stride):
+                a = line[i - 3] if i >= 3 else 0
+                line[i] = (line[i] + ((a + prev[i]) >> 1)) & 0xFF
+        elif f == 4:  # paeth
+            for i in range(stride):
+                a = line[i - 3] if i >= 3 else 0
+                b = prev[i]
+                cdiag = prev[i - 3] if i >= 3 else 0
+                p = a + b - cdiag
+                pa, pb, pc = abs(p - a), abs(p - b), abs(p - cdiag)
+                pred = a if (pa <= pb and pa <= pc) else (b if pb <= pc else cdiag)
+                line[i] = (line[i] + pred) & 0xFF
+        out[y * stride : (y + 1) * stride] = line
+        prev = line
+    import numpy as np
+
+    arr = np.frombuffer(bytes(out), dtype=np.uint8).reshape(h, w, 3)
+    return torch.from_numpy(arr.copy())

@@ -1,0 +1,171 @@
+"""Engine tests: LocalEngine (threads over devices) on CPU at tiny scale,
+and DistributedEngine over a 2-rank gloo group (the RCCL path's CPU stand-in,
+SURVEY.md §4 'distributed' tier).
+"""
+import os
+import sys
+
+import pytest
+import torch
+
+from sdwd_amd.core import State
+from sdwd_amd.parallel import GenerationRequest, LocalEngine
+from sdwd_amd.utils.images import decode_png, encode_png, make_grid
+
+
+def make_engine(n=2, ipm=60.0):
+    eng = LocalEngine(model="tiny", devices=["cpu"] * n)
+    for w in eng.world.workers:
+        w.eta.avg_ipm = ipm
+    return eng
+
+
+class TestLocalEngine:
+    def test_generate_gallery(self):
+        eng = make_engine(2)
+        res = eng.generate(
+            GenerationRequest(
+                prompt="cows", batch_size=4, width=64, height=64, steps=2,
+                seed=123,
+            )
+        )
+        assert res.images.shape == (4, 64, 64, 3)
+        assert res.seeds == [123, 124, 125, 126]
+        assert res.grid is not None
+        assert len(res.job_summary) == 2
+        assert all("Worker: gpu" in t for t in res.infotexts)
+
+    def test_matches_single_rank(self):
+        """2-rank gallery ~= 1-rank gallery (<=1 uint8 LSB, see pipeline
+        shard test)."""
+        req = dict(
+            prompt="same", batch_size=4, width=64, height=64, steps=2,
+            seed=77,
+        )
+        one = make_engine(1).generate(GenerationRequest(**req))
+        two = make_engine(2).generate(GenerationRequest(**req))
+        assert one.seeds == two.seeds
+        diff = (one.images.float() - two.images.float()).abs()
+        assert diff.max() <= 1.0
+
+    def test_failure_requeued(self):
+        eng = make_engine(3)
+        eng.inject_failure("gpu1")
+        res = eng.generate(
+            GenerationRequest(
+                prompt="f", batch_size=6, width=64, height=64, steps=2,
+                seed=10,
+            )
+        )
+        assert res.images.shape == (6, 64, 64, 3)
+        assert res.seeds == list(range(10, 16))
+        # every image was produced (no zero frames)
+        assert all(
+            res.images[i].float().std() > 0 for i in range(6)
+        )
+        assert eng.world.get_worker("gpu1").state is State.UNAVAILABLE
+
+    def test_benchmark_sets_speeds(self):
+        eng = LocalEngine(model="tiny", devices=["cpu"])
+        eng.world.benchmark_payload.width = 64
+        eng.world.benchmark_payload.height = 64
+        eng.world.benchmark_payload.steps = 2
+        speeds = eng.benchmark()
+        assert speeds["gpu0"] > 0
+
+    def test_img2img_through_engine(self):
+        eng = make_engine(2)
+        init = torch.randint(0, 255, (4, 64, 64, 3), dtype=torch.uint8)
+        res = eng.generate(
+            GenerationRequest(
+                prompt="re", batch_size=4, width=64, height=64, steps=2,
+                seed=5, init_images=init, denoising_strength=0.6,
+            )
+        )
+        assert res.images.shape == (4, 64, 64, 3)
+
+
+class TestImages:
+    def test_grid_layout(self):
+        imgs = torch.arange(4, dtype=torch.uint8)[:, None, None, None].expand(
+            4, 8, 8, 3
+        )
+        grid = make_grid(imgs)
+        assert grid.shape == (16, 16, 3)
+        assert grid[0, 0, 0] == 0 and grid[0, 8, 0] == 1
+        assert grid[8, 0, 0] == 2 and grid[8, 8, 0] == 3
+
+    def test_png_round_trip(self):
+        img = torch.randint(0, 255, (16, 24, 3), dtype=torch.uint8)
+        data = encode_png(img)
+        assert data[:4] == b"\x89PNG"
+        back = decode_png(data)
+        assert torch.equal(back, img)
+
+
+# ---------------------------------------------------------------------------
+# 2-rank gloo DistributedEngine (multi-process, CPU)
+# ---------------------------------------------------------------------------
+def _dist_worker(rank, world_size, port, tmpdir, fail_rank):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from sdwd_amd.parallel import DistributedEngine, GenerationRequest
+
+    eng = DistributedEngine(model="tiny", backend="gloo")
+    for w in eng.world.workers:
+        w.eta.avg_ipm = 60.0
+    if fail_rank == rank:
+        # simulate a failing shard on this rank
+        def boom(*a, **k):
+            raise RuntimeError("injected shard failure")
+
+        eng.pipe.generate = boom
+    res = eng.generate(
+        GenerationRequest(
+            prompt="dist", batch_size=4, width=64, height=64, steps=2,
+            seed=900,
+        )
+    )
+    if rank == 0:
+        assert res.images.shape == (4, 64, 64, 3)
+        assert res.seeds == [900, 901, 902, 903]
+        for i in range(4):
+            assert res.images[i].float().std() > 0, f"image {i} empty"
+        torch.save(res.images, os.path.join(tmpdir, "gallery.pt"))
+    from sdwd_amd.parallel import destroy_group
+
+    destroy_group()
+
+
+def _spawn(world_size, tmpdir, fail_rank=-1):
+    import torch.multiprocessing as mp
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    mp.start_processes(
+        _dist_worker,
+        args=(world_size, port, tmpdir, fail_rank),
+        nprocs=world_size,
+        start_method="spawn",
+        join=True,
+    )
+
+
+@pytest.mark.timeout(300)
+class TestDistributedEngine:
+    def test_two_rank_gloo(self, tmp_path):
+        _spawn(2, str(tmp_path))
+        gallery = torch.load(tmp_path / "gallery.pt")
+        assert gallery.shape == (4, 64, 64, 3)
+
+    def test_two_rank_failure_recovery(self, tmp_path):
+        """Rank 1's shard fails; rank 0 re-runs it so the gallery is full."""
+        _spawn(2, str(tmp_path), fail_rank=1)
+        gallery = torch.load(tmp_path / "gallery.pt")
+        assert gallery.shape == (4, 64, 64, 3)
+        assert all(gallery[i].float().std() > 0 for i in range(4))
