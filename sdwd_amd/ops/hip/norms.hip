@@ -1,0 +1,200 @@
+// Fused normalisation kernels.
+//
+// group_norm_silu: NCHW GroupNorm with the SiLU folded into the second pass
+//   (one kernel instead of GN + act: saves one full HBM round trip of the
+//   activation tensor — the UNet/VAE call this before every conv).
+//   One 256-thread block per (sample, group); a group's elements are
+//   contiguous in NCHW so both passes stream 16 B/lane.
+//
+// layer_norm: row-wise LN over the last dim (transformer blocks).
+//   One block per row batch; fp32 statistics.
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// GroupNorm(+SiLU), bf16
+// grid.x = N * G   block = 256
+// ---------------------------------------------------------------------------
+template <bool SILU, bool VEC>
+__global__ void group_norm_silu_bf16_kernel(
+    const __hip_bfloat16 *__restrict__ x, const float *__restrict__ w,
+    const float *__restrict__ b, __hip_bfloat16 *__restrict__ out, int N,
+    int C, long HW, int G, float eps) {
+  __shared__ float lds[16];
+  const int ng = blockIdx.x;
+  const int n = ng / G, g = ng % G;
+  const int cg = C / G;            // channels per group
+  const long len = (long)cg * HW;  // elements per (n, g), contiguous
+  const __hip_bfloat16 *base = x + (long)n * C * HW + (long)g * len;
+  __hip_bfloat16 *obase = out + (base - x);
+
+  float sum = 0.f, sumsq = 0.f;
+  if (VEC) {
+    const long nv = len / 8;
+    for (long i = threadIdx.x; i < nv; i += blockDim.x) {
+      bf16x8 v = ((const bf16x8 *)base)[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = (float)v[j];
+        sum += f;
+        sumsq += f * f;
+      }
+    }
+  } else {
+    for (long i = threadIdx.x; i < len; i += blockDim.x) {
+      float f = bf2f(base[i]);
+      sum += f;
+      sumsq += f * f;
+    }
+  }
+  block_reduce2<256>(sum, sumsq, lds);
+  const float mean = sum / (float)len;
+  const float var = sumsq / (float)len - mean * mean;
+  const float rstd = rsqrtf(var + eps);
+
+  if (VEC) {
+    const long nv = len / 8;
+    const long hv = HW / 8;  // vectors per channel (HW % 8 == 0 in VEC mode)
+    for (long i = threadIdx.x; i < nv; i += blockDim.x) {
+      const int c = g * cg + (int)(i / hv);
+      const float scale = rstd * w[c];
+      const float shift = b[c] - mean * scale;
+      bf16x8 v = ((const bf16x8 *)base)[i];
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = (float)v[j] * scale + shift;
+        o[j] = (__bf16)(SILU ? silu_f(f) : f);
+      }
+      ((bf16x8 *)obase)[i] = o;
+    }
+  } else {
+    for (long i = threadIdx.x; i < len; i += blockDim.x) {
+      const int c = g * cg + (int)(i / HW);
+      float f = (bf2f(base[i]) - mean) * rstd * w[c] + b[c];
+      obase[i] = f2bf(SILU ? silu_f(f) : f);
+    }
+  }
+}
+
+// fp32 variant (debug / CPU-parity checks on GPU)
+template <bool SILU>
+__global__ void group_norm_silu_f32_kernel(const float *__restrict__ x,
+                                           const float *__restrict__ w,
+                                           const float *__restrict__ b,
+                                           float *__restrict__ out, int N,
+                                           int C, long HW, int G, float eps) {
+  __shared__ float lds[16];
+  const int ng = blockIdx.x;
+  const int n = ng / G, g = ng % G;
+  const int cg = C / G;
+  const long len = (long)cg * HW;
+  const float *base = x + (long)n * C * HW + (long)g * cg * HW;
+  float *obase = out + (base - x);
+  float sum = 0.f, sumsq = 0.f;
+  for (long i = threadIdx.x; i < len; i += blockDim.x) {
+    float f = base[i];
+    sum += f;
+    sumsq += f * f;
+  }
+  block_reduce2<256>(sum, sumsq, lds);
+  const float mean = sum / (float)len;
+  const float rstd = rsqrtf(sumsq / (float)len - mean * mean + eps);
+  for (long i = threadIdx.x; i < len; i += blockDim.x) {
+    const int c = g * cg + (int)(i / HW);
+    float f = (base[i] - mean) * rstd * w[c] + b[c];
+    obase[i] = SILU ? silu_f(f) : f;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// LayerNorm over rows [R, D], bf16. One wave per row (D <= a few K).
+// block = 256 = 4 waves -> 4 rows per block.
+// ---------------------------------------------------------------------------
+__global__ void layer_norm_bf16_kernel(const __hip_bfloat16 *__restrict__ x,
+                                       const float *__restrict__ w,
+                                       const float *__restrict__ b,
+                                       __hip_bfloat16 *__restrict__ out,
+                                       long R, int D, float eps) {
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const long row = (long)blockIdx.x * 4 + wid;
+  if (row >= R) return;
+  const __hip_bfloat16 *base = x + row * D;
+  __hip_bfloat16 *obase = out + row * D;
+
+  float sum = 0.f, sumsq = 0.f;
+  const bool vec = (D % 8) == 0;
+  if (vec) {
+    const int nv = D / 8;
+    for (int i = lane; i < nv; i += WAVE) {
+      bf16x8 v = ((const bf16x8 *)base)[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = (float)v[j];
+        sum += f;
+        sumsq += f * f;
+      }
+    }
+  } else {
+    for (int i = lane; i < D; i += WAVE) {
+      float f = bf2f(base[i]);
+      sum += f;
+      sumsq += f * f;
+    }
+  }
+#pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    sum += __shfl_down(sum, off, WAVE);
+    sumsq += __shfl_down(sumsq, off, WAVE);
+  }
+  sum = __shfl(sum, 0, WAVE);
+  sumsq = __shfl(sumsq, 0, WAVE);
+  const float mean = sum / D;
+  const float rstd = rsqrtf(sumsq / D - mean * mean + eps);
+
+  if (vec) {
+    const int nv = D / 8;
+    for (int i = lane; i < nv; i += WAVE) {
+      bf16x8 v = ((const bf16x8 *)base)[i];
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d = i * 8 + j;
+        o[j] = (__bf16)(((float)v[j] - mean) * rstd * w[d] + b[d]);
+      }
+      ((bf16x8 *)obase)[i] = o;
+    }
+  } else {
+    for (int i = lane; i < D; i += WAVE)
+      obase[i] = f2bf((bf2f(base[i]) - mean) * rstd * w[i] + b[i]);
+  }
+}
+
+__global__ void layer_norm_f32_kernel(const float *__restrict__ x,
+                                      const float *__restrict__ w,
+                                      const float *__restrict__ b,
+                                      float *__restrict__ out, long R, int D,
+                                      float eps) {
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const long row = (long)blockIdx.x * 4 + wid;
+  if (row >= R) return;
+  const float *base = x + row * D;
+  float sum = 0.f, sumsq = 0.f;
+  for (int i = lane; i < D; i += WAVE) {
+    float f = base[i];
+    sum += f;
+    sumsq += f * f;
+  }
+#pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    sum += __shfl_down(sum, off, WAVE);
+    sumsq += __shfl_down(sumsq, off, WAVE);
+  }
+  sum = __shfl(sum, 0, WAVE);
+  sumsq = __shfl(sumsq, 0, WAVE);
+  const float mean = sum / D;
+  const float rstd = rsqrtf(sumsq / D - mean * mean + eps);
+  for (int i = lane; i < D; i += WAVE)
+    out[row * D + i] = (base[i] - mean) * rstd * w[i] + b[i];
+}
