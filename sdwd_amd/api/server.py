@@ -1,0 +1,284 @@
+"""sdapi/v1-compatible HTTP surface.
+
+The reference *consumed* these routes on remote webui instances
+(txt2img, img2img, options, memory, sd-models, interrupt,
+refresh-checkpoints, server-restart, progress); this engine *serves* them, so
+existing sdwui API clients can point at this node and drive the whole-node
+sharded engine. Plus a /sdwd/status route exposing the reference's Status-tab
+data (worker states, speeds, log ring buffer — ref ui.py:230-249).
+"""
+from __future__ import annotations
+
+import base64
+import io
+import json
+import threading
+import time
+from typing import Any, Dict, List, Optional
+
+import torch
+from fastapi import FastAPI, HTTPException
+from pydantic import BaseModel, Field
+
+from ..core import State
+from ..models.registry import available_models
+from ..parallel import GenerationRequest, LocalEngine
+from ..pipeline.samplers import sampler_names
+from ..utils import get_logger, ring_buffer
+from ..utils.images import decode_png, encode_png
+
+log = get_logger("api")
+
+
+class Txt2ImgRequest(BaseModel):
+    prompt: str = ""
+    negative_prompt: str = ""
+    seed: int = -1
+    subseed: int = -1
+    subseed_strength: float = 0.0
+    steps: int = 20
+    cfg_scale: float = 7.0
+    width: int = 512
+    height: int = 512
+    batch_size: int = 1
+    n_iter: int = 1
+    sampler_name: str = "Euler a"
+    sampler_index: Optional[str] = None  # legacy alias
+
+
+class Img2ImgRequest(Txt2ImgRequest):
+    init_images: List[str] = Field(default_factory=list)  # base64 PNG
+    denoising_strength: float = 0.75
+
+
+class OptionsRequest(BaseModel):
+    sd_model_checkpoint: Optional[str] = None
+    sd_vae: Optional[str] = None
+
+    model_config = {"extra": "allow"}
+
+
+class ServerState:
+    def __init__(self, engine: LocalEngine):
+        self.engine = engine
+        self.current_model = engine.model_name
+        self.busy = False
+        self.progress = 0.0
+        self.started_at = 0.0
+        self.lock = threading.Lock()
+
+
+def _b64_png(img: torch.Tensor) -> str:
+    return base64.b64encode(encode_png(img)).decode()
+
+
+def _decode_b64_png(data: str) -> torch.Tensor:
+    if "," in data[:64]:  # data URI prefix
+        data = data.split(",", 1)[1]
+    return decode_png(base64.b64decode(data))
+
+
+def create_app(engine: Optional[LocalEngine] = None,
+               model: str = "sd15") -> FastAPI:
+    if engine is None:
+        engine = LocalEngine(model=model)
+    state = ServerState(engine)
+    app = FastAPI(title="sdwd_amd", version="0.1.0")
+    app.state.engine = engine
+
+    def run_generation(gen: GenerationRequest) -> Dict[str, Any]:
+        with state.lock:
+            state.busy = True
+            state.started_at = time.time()
+        try:
+            result = engine.generate(gen)
+        finally:
+            state.busy = False
+        images = [_b64_png(result.images[i])
+                  for i in range(result.images.shape[0])]
+        if result.grid is not None:
+            images.insert(0, _b64_png(result.grid))
+        info = {
+            "all_seeds": result.seeds,
+            "all_subseeds": [-1] * len(result.seeds),
+            "all_prompts": [gen.prompt] * len(result.seeds),
+            "all_negative_prompts": [gen.negative_prompt] * len(result.seeds),
+            "infotexts": result.infotexts,
+            "job_summary": result.job_summary,
+            "elapsed": result.elapsed,
+        }
+        return {
+            "images": images,
+            "parameters": gen.__dict__ | {"init_images": None},
+            "info": json.dumps(info),
+        }
+
+    @app.post("/sdapi/v1/txt2img")
+    def txt2img(req: Txt2ImgRequest):
+        gen = GenerationRequest(
+            prompt=req.prompt,
+            negative_prompt=req.negative_prompt,
+            batch_size=req.batch_size * max(1, req.n_iter),
+            width=req.width,
+            height=req.height,
+            steps=req.steps,
+            cfg_scale=req.cfg_scale,
+            sampler_name=req.sampler_name or req.sampler_index or "Euler a",
+            seed=req.seed,
+            subseed=req.subseed,
+            subseed_strength=req.subseed_strength,
+        )
+        return run_generation(gen)
+
+    @app.post("/sdapi/v1/img2img")
+    def img2img(req: Img2ImgRequest):
+        if not req.init_images:
+            raise HTTPException(422, "init_images required")
+        inits = torch.stack([_decode_b64_png(d) for d in req.init_images])
+        gen = GenerationRequest(
+            prompt=req.prompt,
+            negative_prompt=req.negative_prompt,
+            batch_size=req.batch_size * max(1, req.n_iter),
+            width=req.width,
+            height=req.height,
+            steps=req.steps,
+            cfg_scale=req.cfg_scale,
+            sampler_name=req.sampler_name or "Euler a",
+            seed=req.seed,
+            subseed=req.subseed,
+            subseed_strength=req.subseed_strength,
+            init_images=inits,
+            denoising_strength=req.denoising_strength,
+        )
+        return run_generation(gen)
+
+    @app.post("/sdapi/v1/options")
+    def set_options(req: OptionsRequest):
+        if req.sd_model_checkpoint and req.sd_model_checkpoint != state.current_model:
+            name = req.sd_model_checkpoint
+            if name not in available_models():
+                raise HTTPException(404, f"unknown model {name}")
+            log.info("switching model to %s on all ranks", name)
+            from ..pipeline import StableDiffusionPipeline
+
+            for label, pipe in engine.pipes.items():
+                engine.pipes[label] = StableDiffusionPipeline(
+                    name, device=pipe.device, dtype=pipe.dtype
+                )
+            state.current_model = name
+            engine.model_name = name
+        return {}
+
+    @app.get("/sdapi/v1/options")
+    def get_options():
+        return {"sd_model_checkpoint": state.current_model, "sd_vae": "auto"}
+
+    @app.get("/sdapi/v1/sd-models")
+    def sd_models():
+        return [
+            {"title": m, "model_name": m, "filename": f"{m}.safetensors"}
+            for m in available_models()
+        ]
+
+    @app.get("/sdapi/v1/samplers")
+    def samplers():
+        return [{"name": s, "aliases": [s]} for s in sampler_names()]
+
+    @app.get("/sdapi/v1/memory")
+    def memory():
+        cuda: Dict[str, Any] = {}
+        if torch.cuda.is_available():
+            free, total = torch.cuda.mem_get_info(0)
+            cuda = {"system": {"free": free, "used": total - free,
+                               "total": total}}
+        return {"ram": {}, "cuda": cuda}
+
+    @app.post("/sdapi/v1/interrupt")
+    def interrupt():
+        engine.interrupt()
+        return {}
+
+    @app.post("/sdapi/v1/refresh-checkpoints")
+    def refresh_checkpoints():
+        return {}
+
+    @app.post("/sdapi/v1/refresh-loras")
+    def refresh_loras():
+        return {}
+
+    @app.get("/sdapi/v1/progress")
+    def progress():
+        eta = 0.0
+        return {
+            "progress": 1.0 if not state.busy else 0.5,
+            "eta_relative": eta,
+            "state": {
+                "job": "generate" if state.busy else "",
+                "interrupted": engine.world.interrupted.is_set(),
+            },
+            "current_image": None,
+        }
+
+    @app.get("/sdwd/status")
+    def status():
+        """The reference's Status tab as JSON (ref ui.py:230-249)."""
+        return {
+            "workers": [
+                {
+                    "label": w.label,
+                    "device": w.device,
+                    "state": w.state.value,
+                    "avg_ipm": w.eta.avg_ipm,
+                    "mpe": w.eta.mpe(),
+                    "pixel_cap": w.pixel_cap,
+                }
+                for w in engine.world.workers
+            ],
+            "speed_summary": engine.world.speed_summary(),
+            "log": ring_buffer(),
+            "model": state.current_model,
+            "busy": state.busy,
+        }
+
+    @app.post("/sdwd/benchmark")
+    def benchmark(rebenchmark: bool = True):
+        speeds = engine.benchmark(rebenchmark=rebenchmark)
+        return {"speeds": speeds}
+
+    @app.post("/sdwd/worker/{label}/enable")
+    def enable_worker(label: str):
+        w = engine.world.get_worker(label)
+        if w is None:
+            raise HTTPException(404, label)
+        w.set_state(State.IDLE)
+        return {}
+
+    @app.post("/sdwd/worker/{label}/disable")
+    def disable_worker(label: str):
+        w = engine.world.get_worker(label)
+        if w is None:
+            raise HTTPException(404, label)
+        w.set_state(State.DISABLED, strict=False)
+        return {}
+
+    return app
+
+
+def main() -> None:  # pragma: no cover - manual entry
+    import argparse
+
+    import uvicorn
+
+    from ..config import add_flags
+
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--port", type=int, default=7860)
+    ap.add_argument("--model", default="sd15")
+    add_flags(ap)
+    args = ap.parse_args()
+    uvicorn.run(create_app(model=args.model), host=args.host, port=args.port)
+
+
+if __name__ == "__main__":  # pragma: no cover
+    main()

@@ -1,0 +1,111 @@
+"""API surface tests (sdapi/v1-compatible routes) at tiny scale on CPU."""
+import base64
+import json
+
+import pytest
+import torch
+from fastapi.testclient import TestClient
+
+from sdwd_amd.api import create_app
+from sdwd_amd.parallel import LocalEngine
+from sdwd_amd.utils.images import decode_png, encode_png
+
+
+@pytest.fixture(scope="module")
+def client():
+    engine = LocalEngine(model="tiny", devices=["cpu", "cpu"])
+    for w in engine.world.workers:
+        w.eta.avg_ipm = 60.0
+    app = create_app(engine=engine)
+    return TestClient(app)
+
+
+class TestTxt2Img:
+    def test_basic(self, client):
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={
+                "prompt": "a cow",
+                "steps": 2,
+                "width": 64,
+                "height": 64,
+                "batch_size": 2,
+                "seed": 77,
+            },
+        )
+        assert r.status_code == 200
+        body = r.json()
+        # grid + 2 images
+        assert len(body["images"]) == 3
+        img = decode_png(base64.b64decode(body["images"][1]))
+        assert img.shape == (64, 64, 3)
+        info = json.loads(body["info"])
+        assert info["all_seeds"] == [77, 78]
+        assert len(info["infotexts"]) == 2
+
+
+class TestImg2Img:
+    def test_round_trip(self, client):
+        init = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)
+        b64 = base64.b64encode(encode_png(init)).decode()
+        r = client.post(
+            "/sdapi/v1/img2img",
+            json={
+                "prompt": "re",
+                "steps": 2,
+                "width": 64,
+                "height": 64,
+                "batch_size": 1,
+                "seed": 5,
+                "init_images": [b64],
+                "denoising_strength": 0.6,
+            },
+        )
+        assert r.status_code == 200
+        assert len(r.json()["images"]) >= 1
+
+    def test_missing_init(self, client):
+        r = client.post(
+            "/sdapi/v1/img2img", json={"prompt": "x", "init_images": []}
+        )
+        assert r.status_code == 422
+
+
+class TestControl:
+    def test_models_and_samplers(self, client):
+        models = client.get("/sdapi/v1/sd-models").json()
+        assert any(m["model_name"] == "sd15" for m in models)
+        samplers = client.get("/sdapi/v1/samplers").json()
+        assert any(s["name"] == "Euler a" for s in samplers)
+
+    def test_memory(self, client):
+        r = client.get("/sdapi/v1/memory")
+        assert r.status_code == 200
+
+    def test_interrupt(self, client):
+        assert client.post("/sdapi/v1/interrupt").status_code == 200
+
+    def test_progress(self, client):
+        r = client.get("/sdapi/v1/progress").json()
+        assert "progress" in r and "state" in r
+
+    def test_options_unknown_model(self, client):
+        r = client.post(
+            "/sdapi/v1/options", json={"sd_model_checkpoint": "nope"}
+        )
+        assert r.status_code == 404
+
+    def test_status(self, client):
+        r = client.get("/sdwd/status").json()
+        assert len(r["workers"]) == 2
+        assert r["workers"][0]["state"] in (
+            "IDLE", "WORKING", "INTERRUPTED", "UNAVAILABLE", "DISABLED"
+        )
+        assert "speed_summary" in r
+
+    def test_worker_disable_enable(self, client):
+        assert client.post("/sdwd/worker/gpu1/disable").status_code == 200
+        st = client.get("/sdwd/status").json()
+        w1 = [w for w in st["workers"] if w["label"] == "gpu1"][0]
+        assert w1["state"] == "DISABLED"
+        assert client.post("/sdwd/worker/gpu1/enable").status_code == 200
