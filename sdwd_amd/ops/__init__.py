@@ -91,7 +91,9 @@ def group_norm_silu(
     HBM-bound sweep with ushort8-vectorised bf16 loads.
     """
     if x.is_cuda:
-        return ext().group_norm_silu(x, weight, bias, groups, eps, silu)
+        return ext().group_norm_silu(
+            x.contiguous(), weight, bias, groups, eps, silu
+        )
     out = F.group_norm(x.float(), groups, weight.float(), bias.float(), eps)
     if silu:
         out = F.silu(out)
@@ -109,7 +111,7 @@ def layer_norm(
     x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor, eps: float = 1e-5
 ) -> torch.Tensor:
     if x.is_cuda:
-        return ext().layer_norm(x, weight, bias, eps)
+        return ext().layer_norm(x.contiguous(), weight, bias, eps)
     return F.layer_norm(
         x.float(), (x.shape[-1],), weight.float(), bias.float(), eps
     ).to(x.dtype)
@@ -145,14 +147,14 @@ def attention(
 # ---------------------------------------------------------------------------
 def geglu(h: torch.Tensor) -> torch.Tensor:
     if h.is_cuda:
-        return ext().geglu(h)
+        return ext().geglu(h.contiguous())
     x, gate = h.float().chunk(2, dim=-1)
     return (x * F.gelu(gate)).to(h.dtype)
 
 
 def silu(x: torch.Tensor) -> torch.Tensor:
     if x.is_cuda:
-        return ext().silu(x)
+        return ext().silu(x.contiguous())
     return F.silu(x.float()).to(x.dtype)
 
 
@@ -164,7 +166,9 @@ def euler_step(
 ) -> torch.Tensor:
     """x + (x - denoised)/sigma * (sigma_next - sigma), one fused kernel."""
     if x.is_cuda:
-        return ext().euler_step(x, denoised, sigma, sigma_next)
+        return ext().euler_step(
+            x.contiguous(), denoised.contiguous(), sigma, sigma_next
+        )
     d = (x.float() - denoised.float()) / sigma
     return (x.float() + d * (sigma_next - sigma)).to(x.dtype)
 
@@ -174,7 +178,7 @@ def add_noise(
 ) -> torch.Tensor:
     """a*x + b*noise in one pass (ancestral samplers, img2img noising)."""
     if x.is_cuda:
-        return ext().axpby(x, noise, a, b)
+        return ext().axpby(x.contiguous(), noise.contiguous(), a, b)
     return (a * x.float() + b * noise.float()).to(x.dtype)
 
 

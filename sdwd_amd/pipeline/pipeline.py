@@ -19,6 +19,7 @@ import torch
 from ..models.registry import ModelBundle, load_model
 from ..models import tokenizer
 from ..utils import get_logger
+from .graphs import GraphedDenoiser
 from .samplers import build_sampler
 from .schedule import schedule_for
 
@@ -107,6 +108,10 @@ class StableDiffusionPipeline:
         else:
             model.to(self.device, dtype)
         self.model = model
+        self._denoiser = GraphedDenoiser(
+            lambda x, ts, ctx, y: self.model.unet(x, ts, ctx, y=y),
+            self.device,
+        )
 
     # -- conditioning --------------------------------------------------------
     @torch.no_grad()
@@ -231,12 +236,14 @@ class StableDiffusionPipeline:
         cfg = float(req.cfg_scale)
         unet = self.model.unet
 
+        denoiser = self._denoiser
+
         def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:
             ts = torch.full(
                 (x_in.shape[0] * 2,), t, device=self.device,
                 dtype=torch.float32,
             )
-            eps = unet(torch.cat([x_in, x_in], dim=0), ts, ctx, y=y)
+            eps = denoiser(torch.cat([x_in, x_in], dim=0), ts, ctx, y)
             eps_c, eps_u = eps.chunk(2, dim=0)
             return eps_u + cfg * (eps_c - eps_u)
 
@@ -248,7 +255,7 @@ class StableDiffusionPipeline:
                     dtype=torch.float32,
                 )
                 yc = y[: x_in.shape[0]] if y is not None else None
-                return unet(x_in, ts, cond, y=yc)
+                return denoiser(x_in, ts, cond, yc)
 
         was_interrupted = False
 
