@@ -240,6 +240,14 @@ def create_app(engine: Optional[LocalEngine] = None,
             "busy": state.busy,
         }
 
+    @app.post("/sdwd/sync-script")
+    def sync_script():
+        """Run the user's sync* script (ref C21, ui.py:26-55)."""
+        from ..utils.sync_scripts import run_sync_script
+
+        rc, output = run_sync_script()
+        return {"returncode": rc, "output": output[-4000:]}
+
     @app.post("/sdwd/benchmark")
     def benchmark(rebenchmark: bool = True):
         speeds = engine.benchmark(rebenchmark=rebenchmark)

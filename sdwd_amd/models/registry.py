@@ -132,3 +132,49 @@ def load_model(
 
 def clear_cache() -> None:
     _cache.clear()
+
+
+# -- checkpoint files (ref C13 synced checkpoints by NAME over /options;
+# here weights also round-trip to local safetensors files) -------------------
+def save_checkpoint(bundle: ModelBundle, path: str) -> str:
+    from safetensors.torch import save_file
+
+    tensors = {}
+    for prefix, mod in (
+        ("text_encoder", bundle.text_encoder),
+        ("text_encoder_2", bundle.text_encoder_2),
+        ("unet", bundle.unet),
+        ("vae", bundle.vae),
+    ):
+        if mod is None:
+            continue
+        for k, v in mod.state_dict().items():
+            tensors[f"{prefix}.{k}"] = v.contiguous().cpu()
+    save_file(tensors, path, metadata={"arch": bundle.name})
+    return path
+
+
+def load_checkpoint(path: str, device="cpu", dtype=None) -> ModelBundle:
+    """Load a safetensors file saved by save_checkpoint; the architecture
+    name is read from metadata (falls back to sd15)."""
+    from safetensors import safe_open
+
+    with safe_open(path, framework="pt") as f:
+        meta = f.metadata() or {}
+        arch = meta.get("arch", "sd15")
+        keys = list(f.keys())
+        bundle = _BUILDERS[arch.split("/")[0] if arch in _BUILDERS else "sd15"](arch)
+        by_prefix: Dict[str, Dict[str, torch.Tensor]] = {}
+        for k in keys:
+            prefix, rest = k.split(".", 1)
+            by_prefix.setdefault(prefix, {})[rest] = f.get_tensor(k)
+    for prefix, mod in (
+        ("text_encoder", bundle.text_encoder),
+        ("text_encoder_2", bundle.text_encoder_2),
+        ("unet", bundle.unet),
+        ("vae", bundle.vae),
+    ):
+        if mod is not None and prefix in by_prefix:
+            mod.load_state_dict(by_prefix[prefix], strict=False)
+    bundle.eval().to(device, dtype)
+    return bundle

@@ -48,6 +48,9 @@ class GenerationRequest:
     subseed_strength: float = 0.0
     init_images: Optional[torch.Tensor] = None  # [B,H,W,3] uint8 (img2img)
     denoising_strength: float = 0.75
+    enable_hr: bool = False
+    hr_scale: float = 2.0
+    hr_steps: int = 0
 
     def sched(self) -> GenRequest:
         return GenRequest(
@@ -59,6 +62,8 @@ class GenerationRequest:
             seed=self.seed,
             subseed=self.subseed,
             subseed_strength=self.subseed_strength,
+            hr_scale=self.hr_scale if self.enable_hr else 0.0,
+            hr_steps=(self.hr_steps or self.steps) if self.enable_hr else 0,
         )
 
 
@@ -89,6 +94,9 @@ def _job_pipeline_request(
         subseed_strength=gen.subseed_strength,
         init_latents=init_latents,
         denoising_strength=gen.denoising_strength,
+        enable_hr=gen.enable_hr,
+        hr_scale=gen.hr_scale,
+        hr_steps=gen.hr_steps,
     )
 
 
@@ -105,7 +113,12 @@ class _EngineBase:
         """Order shards by gallery offset, fix up infotexts with the worker
         label (ref distributed.py:343-349), build the grid."""
         total = sum(j.batch_size for j in jobs)
-        h, w = gen.height, gen.width
+        f = gen.hr_scale if gen.enable_hr else 1.0
+        h, w = int(gen.height * f), int(gen.width * f)
+        for sh in shards.values():  # trust actual shard shape (fractional hr)
+            if sh is not None and sh.numel():
+                h, w = sh.shape[1], sh.shape[2]
+                break
         images = torch.zeros(total, h, w, 3, dtype=torch.uint8)
         seeds = [0] * total
         infotexts = [""] * total
@@ -397,7 +410,8 @@ class DistributedEngine(_EngineBase):
         jobs = pg.broadcast_object(jobs)
 
         mine = next((j for j in jobs if j.worker_label == self.label), None)
-        h, w = gen.height, gen.width
+        hf = gen.hr_scale if gen.enable_hr else 1.0
+        h, w = int(gen.height * hf), int(gen.width * hf)
         my_images = torch.zeros(0, h, w, 3, dtype=torch.uint8)
         my_ok, my_elapsed = 1.0, 0.0
         my_infos: List[str] = []

@@ -41,6 +41,10 @@ class PipelineRequest:
     # img2img
     init_latents: Optional[torch.Tensor] = None  # pre-encoded [B,4,h,w]
     denoising_strength: float = 0.75
+    # hires fix (sdwui two-pass: base gen -> latent upscale -> img2img pass)
+    enable_hr: bool = False
+    hr_scale: float = 2.0
+    hr_steps: int = 0  # 0 = same as steps
 
     @property
     def batch_size(self) -> int:
@@ -270,6 +274,53 @@ class StableDiffusionPipeline:
             model_fn, x, noise_fn=noise_fn, callback=step_callback,
             interrupt=_interrupt,
         )
+
+        # hires fix: latent-upscale the base result and run a second,
+        # strength-limited denoise pass at the scaled resolution
+        # (ref eta_hr, worker.py:205-228 predicts exactly this shape).
+        if req.enable_hr and req.hr_scale > 1.0 and not was_interrupted:
+            hr_steps = req.hr_steps or req.steps
+            x = torch.nn.functional.interpolate(
+                x.float(), scale_factor=req.hr_scale, mode="nearest"
+            ).to(self.dtype)
+            hsched = schedule_for(req.sampler_name, hr_steps)
+            start = max(
+                0, hr_steps - max(1, int(hr_steps * req.denoising_strength))
+            )
+            hsched = type(hsched)(
+                sigmas=hsched.sigmas[start:], timesteps=hsched.timesteps[start:]
+            )
+            hsampler = build_sampler(req.sampler_name, hsched)
+            hh, hw = x.shape[2], x.shape[3]
+            hr_noise = torch.stack(
+                [
+                    _image_noise(
+                        (int(req.seeds[i]) ^ 0x12E50),
+                        subseeds[i],
+                        0.0,
+                        (lat_c, hh, hw),
+                    )
+                    for i in range(b)
+                ]
+            ).to(self.device, self.dtype)
+            s0 = float(hsched.sigmas[0])
+            x = (x.float() + hr_noise.float() * s0).to(self.dtype)
+
+            def hr_noise_fn() -> torch.Tensor:
+                n = torch.stack(
+                    [
+                        torch.randn(
+                            (lat_c, hh, hw), generator=g, dtype=torch.float32
+                        )
+                        for g in gens
+                    ]
+                )
+                return n.to(self.device, self.dtype)
+
+            x = hsampler.sample(
+                model_fn, x, noise_fn=hr_noise_fn, callback=step_callback,
+                interrupt=_interrupt,
+            )
 
         if decode:
             pixels = self.model.vae.decode(x)

@@ -322,3 +322,42 @@ class TestJob:
         j = Job(worker_label="x")
         took = j.add_work(10, 512, 512, pixel_cap=3 * 512 * 512)
         assert took == 3 and j.batch_size == 3
+
+
+class TestCheckpointFiles:
+    def test_safetensors_round_trip(self, tmp_path):
+        import torch
+        from sdwd_amd.models.registry import (
+            load_checkpoint,
+            load_model,
+            save_checkpoint,
+        )
+
+        b = load_model("tiny", cache=False)
+        with torch.no_grad():
+            list(b.unet.parameters())[0].add_(1.23)
+        p = str(tmp_path / "m.safetensors")
+        save_checkpoint(b, p)
+        b2 = load_checkpoint(p)
+        assert torch.equal(
+            list(b.unet.parameters())[0], list(b2.unet.parameters())[0]
+        )
+
+
+class TestSyncScript:
+    def test_runs_user_script(self, tmp_path):
+        from sdwd_amd.utils.sync_scripts import run_sync_script
+
+        d = tmp_path / "user"
+        d.mkdir()
+        script = d / "sync.sh"
+        script.write_text("#!/bin/sh\necho synced-ok\n")
+        rc, out = run_sync_script(str(d))
+        assert rc == 0
+        assert "synced-ok" in out
+
+    def test_missing_script(self, tmp_path):
+        from sdwd_amd.utils.sync_scripts import run_sync_script
+
+        rc, _ = run_sync_script(str(tmp_path))
+        assert rc == 127

@@ -184,3 +184,32 @@ class TestRegistry:
     def test_unknown_model_raises(self):
         with pytest.raises(KeyError):
             load_model("nope")
+
+
+class TestHiresFix:
+    def test_two_pass_upscale(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        req = PipelineRequest(
+            prompt="hr", steps=3, width=64, height=64, seeds=[9],
+            enable_hr=True, hr_scale=2.0, hr_steps=2,
+            denoising_strength=0.6,
+        )
+        res = pipe.generate(req)
+        assert res.images.shape == (1, 128, 128, 3)
+        assert torch.isfinite(res.images.float()).all()
+
+    def test_engine_hires(self):
+        import torch as t
+        from sdwd_amd.parallel import GenerationRequest, LocalEngine
+
+        eng = LocalEngine(model="tiny", devices=["cpu", "cpu"])
+        for w in eng.world.workers:
+            w.eta.avg_ipm = 60.0
+        res = eng.generate(
+            GenerationRequest(
+                prompt="hr", batch_size=2, width=64, height=64, steps=2,
+                seed=4, enable_hr=True, hr_scale=2.0, hr_steps=2,
+            )
+        )
+        assert res.images.shape == (2, 128, 128, 3)
