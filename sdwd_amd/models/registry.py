@@ -130,8 +130,35 @@ def load_model(
     return bundle
 
 
+_cn_cache: Dict[str, "object"] = {}
+
+
+def load_controlnet(name: str = "controlnet-sd15", device="cpu", dtype=None):
+    """ControlNet for a base arch (deterministic random init, like models).
+    name: controlnet-<arch>, e.g. controlnet-sd15 / controlnet-tiny."""
+    from .controlnet import ControlNetModel
+    from .unet import UNetConfig
+
+    if name in _cn_cache:
+        mod = _cn_cache[name]
+    else:
+        arch = name.split("-", 1)[1] if "-" in name else "sd15"
+        cfg = {
+            "sd15": UNetConfig.sd15,
+            "sdxl": UNetConfig.sdxl,
+            "tiny": UNetConfig.tiny,
+        }[arch]()
+        mod = ControlNetModel(cfg, hint_factor=2 if arch == "tiny" else 8)
+        _seeded_init(mod, zlib.crc32(name.encode()) % (2**31))
+        mod.eval()
+        _cn_cache[name] = mod
+    mod.to(device=device, dtype=dtype)
+    return mod
+
+
 def clear_cache() -> None:
     _cache.clear()
+    _cn_cache.clear()
 
 
 # -- checkpoint files (ref C13 synced checkpoints by NAME over /options;

@@ -285,7 +285,10 @@ class UNetModel(nn.Module):
         timesteps: torch.Tensor,
         context: torch.Tensor,
         y: Optional[torch.Tensor] = None,
+        control: Optional[dict] = None,
     ) -> torch.Tensor:
+        """control: {"down": [residual per skip], "mid": residual} from a
+        ControlNetModel (models/controlnet.py)."""
         temb = ops.timestep_embedding(timesteps, self.cfg.model_channels)
         emb = self.time_mlp(temb.to(x.dtype))
         if self.label_mlp is not None and y is not None:
@@ -296,7 +299,12 @@ class UNetModel(nn.Module):
         for blk in self.down:
             h = blk(h, emb, context)
             skips.append(h)
+        if control is not None:
+            skips = [s + c for s, c in zip(skips, control["down"])]
+            h = skips[-1]
         h = self.mid(h, emb, context)
+        if control is not None:
+            h = h + control["mid"]
         for blk in self.up:
             h = torch.cat([h, skips.pop()], dim=1)
             h = blk(h, emb, context)

@@ -51,6 +51,9 @@ class GenerationRequest:
     enable_hr: bool = False
     hr_scale: float = 2.0
     hr_steps: int = 0
+    control_image: Optional[torch.Tensor] = None
+    control_model: str = ""
+    control_scale: float = 1.0
 
     def sched(self) -> GenRequest:
         return GenRequest(
@@ -97,6 +100,9 @@ def _job_pipeline_request(
         enable_hr=gen.enable_hr,
         hr_scale=gen.hr_scale,
         hr_steps=gen.hr_steps,
+        control_image=gen.control_image,
+        control_model=gen.control_model,
+        control_scale=gen.control_scale,
     )
 
 

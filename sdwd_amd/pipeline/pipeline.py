@@ -45,6 +45,10 @@ class PipelineRequest:
     enable_hr: bool = False
     hr_scale: float = 2.0
     hr_steps: int = 0  # 0 = same as steps
+    # controlnet (ref C17 executed natively)
+    control_image: Optional[torch.Tensor] = None  # [B,H,W,3] uint8
+    control_model: str = ""  # e.g. "controlnet-sd15"; "" = off
+    control_scale: float = 1.0
 
     @property
     def batch_size(self) -> int:
@@ -242,12 +246,32 @@ class StableDiffusionPipeline:
 
         denoiser = self._denoiser
 
+        controlnet = None
+        hint = None
+        if req.control_model and req.control_image is not None:
+            from ..models.registry import load_controlnet
+
+            controlnet = load_controlnet(
+                req.control_model, device=self.device, dtype=self.dtype
+            )
+            hint = (
+                req.control_image.permute(0, 3, 1, 2).float() / 255.0
+            ).to(self.device, self.dtype)
+            if hint.shape[0] == 1 and b > 1:
+                hint = hint.expand(b, -1, -1, -1)
+
         def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:
             ts = torch.full(
                 (x_in.shape[0] * 2,), t, device=self.device,
                 dtype=torch.float32,
             )
-            eps = denoiser(torch.cat([x_in, x_in], dim=0), ts, ctx, y)
+            x2 = torch.cat([x_in, x_in], dim=0)
+            if controlnet is not None:
+                h2 = torch.cat([hint, hint], dim=0)
+                ctrl = controlnet(x2, h2, ts, ctx, req.control_scale)
+                eps = unet(x2, ts, ctx, y=y, control=ctrl)
+            else:
+                eps = denoiser(x2, ts, ctx, y)
             eps_c, eps_u = eps.chunk(2, dim=0)
             return eps_u + cfg * (eps_c - eps_u)
 

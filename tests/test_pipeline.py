@@ -213,3 +213,31 @@ class TestHiresFix:
             )
         )
         assert res.images.shape == (2, 128, 128, 3)
+
+
+class TestControlNet:
+    def test_control_changes_output(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="c", steps=2, width=64, height=64, seeds=[3])
+        plain = pipe.generate(PipelineRequest(**base)).images
+        hint = torch.randint(0, 255, (1, 64, 64, 3), dtype=torch.uint8)
+        ctrl = pipe.generate(
+            PipelineRequest(
+                **base, control_image=hint, control_model="controlnet-tiny"
+            )
+        ).images
+        assert ctrl.shape == plain.shape
+        assert not torch.equal(ctrl, plain)  # residuals actually flowed
+
+    def test_control_deterministic(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        hint = torch.randint(0, 255, (1, 64, 64, 3), dtype=torch.uint8)
+        req = PipelineRequest(
+            prompt="c", steps=2, width=64, height=64, seeds=[3],
+            control_image=hint, control_model="controlnet-tiny",
+        )
+        a = pipe.generate(req).images
+        b = pipe.generate(req).images
+        assert torch.equal(a, b)
