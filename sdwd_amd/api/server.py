@@ -44,6 +44,15 @@ class Txt2ImgRequest(BaseModel):
     n_iter: int = 1
     sampler_name: str = "Euler a"
     sampler_index: Optional[str] = None  # legacy alias
+    # hires fix (sdwui fields)
+    enable_hr: bool = False
+    hr_scale: float = 2.0
+    hr_second_pass_steps: int = 0
+    denoising_strength: float = 0.75
+    # alwayson scripts (ref C17/C18: the reference forwarded these; we
+    # execute the controlnet unit natively, other scripts are ignored
+    # with a warning as the reference's compat filter did)
+    alwayson_scripts: Dict[str, Any] = Field(default_factory=dict)
 
 
 class Img2ImgRequest(Txt2ImgRequest):
@@ -76,6 +85,31 @@ def _decode_b64_png(data: str) -> torch.Tensor:
     if "," in data[:64]:  # data URI prefix
         data = data.split(",", 1)[1]
     return decode_png(base64.b64decode(data))
+
+
+def _parse_controlnet(alwayson: Dict[str, Any]):
+    """sdwui controlnet payload: {"controlnet": {"args": [unit, ...]}}
+    (ref control_net.py:20-79 packed this; we consume unit 0 natively).
+    Unknown scripts are logged and skipped (ref C18 compat filter)."""
+    control_image = None
+    control_model = ""
+    control_scale = 1.0
+    for name, body in (alwayson or {}).items():
+        if name.lower() != "controlnet":
+            log.warning("ignoring unsupported alwayson script '%s'", name)
+            continue
+        args = (body or {}).get("args", [])
+        if not args:
+            continue
+        unit = args[0]
+        img_b64 = unit.get("input_image") or unit.get("image")
+        if img_b64:
+            control_image = _decode_b64_png(img_b64)[None]
+        control_model = unit.get("model", "controlnet-sd15")
+        control_scale = float(unit.get("weight", 1.0))
+        if len(args) > 1:
+            log.warning("only the first controlnet unit is applied")
+    return control_image, control_model, control_scale
 
 
 def create_app(engine: Optional[LocalEngine] = None,
@@ -115,6 +149,9 @@ def create_app(engine: Optional[LocalEngine] = None,
 
     @app.post("/sdapi/v1/txt2img")
     def txt2img(req: Txt2ImgRequest):
+        control_image, control_model, control_scale = _parse_controlnet(
+            req.alwayson_scripts
+        )
         gen = GenerationRequest(
             prompt=req.prompt,
             negative_prompt=req.negative_prompt,
@@ -127,6 +164,13 @@ def create_app(engine: Optional[LocalEngine] = None,
             seed=req.seed,
             subseed=req.subseed,
             subseed_strength=req.subseed_strength,
+            enable_hr=req.enable_hr,
+            hr_scale=req.hr_scale,
+            hr_steps=req.hr_second_pass_steps,
+            denoising_strength=req.denoising_strength,
+            control_image=control_image,
+            control_model=control_model,
+            control_scale=control_scale,
         )
         return run_generation(gen)
 

@@ -109,3 +109,41 @@ class TestControl:
         w1 = [w for w in st["workers"] if w["label"] == "gpu1"][0]
         assert w1["state"] == "DISABLED"
         assert client.post("/sdwd/worker/gpu1/enable").status_code == 200
+
+
+class TestAlwaysonControlNet:
+    def test_controlnet_unit_applied(self, client):
+        hint = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)
+        b64 = base64.b64encode(encode_png(hint)).decode()
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={
+                "prompt": "cn",
+                "steps": 2,
+                "width": 64,
+                "height": 64,
+                "seed": 3,
+                "alwayson_scripts": {
+                    "controlnet": {
+                        "args": [
+                            {"input_image": b64, "model": "controlnet-tiny",
+                             "weight": 0.8}
+                        ]
+                    }
+                },
+            },
+        )
+        assert r.status_code == 200
+
+    def test_hires_fields(self, client):
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={
+                "prompt": "hr", "steps": 2, "width": 64, "height": 64,
+                "seed": 3, "enable_hr": True, "hr_scale": 2.0,
+                "hr_second_pass_steps": 2,
+            },
+        )
+        assert r.status_code == 200
+        img = decode_png(base64.b64decode(r.json()["images"][0]))
+        assert img.shape[0] == 128  # grid of one 128x128 image
