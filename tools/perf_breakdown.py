@@ -4,7 +4,11 @@ hipGraph), VAE decode, flash attention vs composed fallback. Guides kernel
 optimization (run under gpurun; writes JSON to stdout)."""
 import argparse
 import json
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
