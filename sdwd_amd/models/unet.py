@@ -107,20 +107,13 @@ class CrossAttention(nn.Module):
     ) -> torch.Tensor:
         context = x if context is None else context
         b, s, d = x.shape
-        q = self.to_q(x).view(b, s, self.heads, self.d_head).transpose(1, 2)
-        k = (
-            self.to_k(context)
-            .view(b, context.shape[1], self.heads, self.d_head)
-            .transpose(1, 2)
-        )
-        v = (
-            self.to_v(context)
-            .view(b, context.shape[1], self.heads, self.d_head)
-            .transpose(1, 2)
-        )
-        out = ops.attention(q, k, v)
-        out = out.transpose(1, 2).reshape(b, s, d)
-        return self.to_out(out)
+        sk = context.shape[1]
+        # natural [B,S,H,D] layout end-to-end: the flash kernel reads strided
+        q = self.to_q(x).view(b, s, self.heads, self.d_head)
+        k = self.to_k(context).view(b, sk, self.heads, self.d_head)
+        v = self.to_v(context).view(b, sk, self.heads, self.d_head)
+        out = ops.attention_bshd(q, k, v)
+        return self.to_out(out.reshape(b, s, d))
 
 
 class BasicTransformerBlock(nn.Module):

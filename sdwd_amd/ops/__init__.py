@@ -142,6 +142,31 @@ def attention(
     return torch.matmul(p, vf).to(q.dtype)
 
 
+def attention_bshd(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    """q,k,v: [B, S, H, D] (natural projection layout) -> [B, Sq, H, D].
+
+    GPU: strided flash kernel, zero transpose/pad copies. CPU (or an
+    unsupported head dim): permuted reference path.
+    """
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if q.is_cuda:
+        mod = ext()
+        d = q.shape[-1]
+        if mod.flash_supported(d):
+            return mod.attention_fwd_bshd(q, k, v, scale)
+    out = attention(
+        q.permute(0, 2, 1, 3), k.permute(0, 2, 1, 3), v.permute(0, 2, 1, 3),
+        scale,
+    )
+    return out.permute(0, 2, 1, 3)
+
+
 # ---------------------------------------------------------------------------
 # GEGLU activation: x, gate = split(h); x * gelu(gate)
 # ---------------------------------------------------------------------------

@@ -1,6 +1,6 @@
 // Flash-style fused attention forward for gfx950 (MFMA 32x32x16 bf16).
 //
-// Structure (v1, correctness-first with the known-good idioms):
+// Structure (v2):
 //  * workgroup = 4 waves; each wave owns 32 q-rows, the WG shares K/V tiles
 //    of 32 keys staged in LDS (K row-major padded, V transposed for
 //    contiguous B-fragment reads).
@@ -10,11 +10,14 @@
 //  * P (bf16-packed) is redistributed to the PV A-fragment layout with
 //    v_permlane32_swap pairs (guide T12), then PV accumulates fp32 via MFMA.
 //  * online softmax with per-tile rescale; the previous tile's PV completes
-//    before the rescale decision (textbook order, no defer threshold yet).
+//    before the rescale decision (textbook order).
+//  * STRIDED access: Q/K/V/O are addressed with (batch, head, row) strides,
+//    so both [B,H,S,D] and [B,S,H,D] layouts run with NO transpose or pad
+//    copies; the head dim D only needs D % 8 == 0 (8-element groups beyond
+//    D are masked in-kernel, DPAD = next multiple of 16).
 //
-// Contract: D == DPAD (multiple of 16, <= 192); callers pad the head dim.
-// Q,K,V,O: [BH, S, DPAD] bf16 contiguous. UNet shapes: D 40->48, 80, 160;
-// SDXL 64; VAE (D=512) takes the composed fallback path in ext.hip.
+// UNet shapes: D 40(->48), 80, 160; SDXL 64; VAE (D=512) takes the composed
+// fallback path in ext.hip.
 #include "common.h"
 
 __device__ __forceinline__ unsigned pack_bf16(float lo, float hi) {
@@ -26,31 +29,39 @@ __device__ __forceinline__ unsigned pack_bf16(float lo, float hi) {
   return cvt.u;
 }
 
+struct AttnStrides {
+  long qb, qh, qr;  // batch, head, row strides (elements)
+  long kb, kh, kr;
+  long vb, vh, vr;
+  long ob, oh, or_;
+};
+
 template <int DPAD>
 __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
     const __hip_bfloat16 *__restrict__ Q, const __hip_bfloat16 *__restrict__ K,
     const __hip_bfloat16 *__restrict__ V, __hip_bfloat16 *__restrict__ O,
-    long Sq, long Sk, float scale) {
+    int H, long Sq, long Sk, int D, float scale, AttnStrides st) {
   constexpr int KVB = 32;
   constexpr int PADK = 8;   // bf16 per-row pad: breaks ds_read_b128 conflicts
-  constexpr int NC = DPAD / 16;          // QK^T k-chunks
+  constexpr int NC = DPAD / 16;              // QK^T k-chunks
   constexpr int DV = (DPAD + 31) / 32 * 32;  // PV d extent (32-col O tiles)
-  constexpr int ND = DV / 32;            // PV d-chunks (O accum tiles)
+  constexpr int ND = DV / 32;                // PV d-chunks (O accum tiles)
 
   __shared__ __align__(16) __bf16 kt[KVB][DPAD + PADK];
   __shared__ __align__(16) __bf16 vt[DV][KVB + PADK];
 
   const int bh = blockIdx.y;
+  const int bb = bh / H, hh = bh % H;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
   const int lq = lane % 32;   // q-row (softmax) / d-col (PV C) index
   const int half = lane / 32; // half-wave id
   const long q0 = (long)blockIdx.x * 128 + wid * 32;
 
-  const __hip_bfloat16 *Qb = Q + (long)bh * Sq * DPAD;
-  const __hip_bfloat16 *Kb = K + (long)bh * Sk * DPAD;
-  const __hip_bfloat16 *Vb = V + (long)bh * Sk * DPAD;
-  __hip_bfloat16 *Ob = O + (long)bh * Sq * DPAD;
+  const __hip_bfloat16 *Qb = Q + bb * st.qb + hh * st.qh;
+  const __hip_bfloat16 *Kb = K + bb * st.kb + hh * st.kh;
+  const __hip_bfloat16 *Vb = V + bb * st.vb + hh * st.vh;
+  __hip_bfloat16 *Ob = O + bb * st.ob + hh * st.oh;
 
   // Q fragments for the whole row-block, read once:
   // B-frag of mfma(K,Q): lane holds Q[q=lq][d = c*16 + 8*half + i]
@@ -58,8 +69,12 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
   {
     const long qrow = (q0 + lq < Sq) ? (q0 + lq) : (Sq - 1);
 #pragma unroll
-    for (int c = 0; c < NC; ++c)
-      qf[c] = *(const bf16x8 *)(Qb + qrow * DPAD + c * 16 + 8 * half);
+    for (int c = 0; c < NC; ++c) {
+      const int d0 = c * 16 + 8 * half;
+      qf[c] = (d0 + 8 <= D)
+                  ? *(const bf16x8 *)(Qb + qrow * st.qr + d0)
+                  : (bf16x8){};
+    }
   }
 
   f32x16 o[ND];
@@ -67,37 +82,35 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
   for (int d = 0; d < ND; ++d) o[d] = (f32x16){};
   float m = -1e30f, l = 0.f;
 
-  // zero vt's pad rows once (DPAD..DV); they are never re-staged
-  if (DV != DPAD) {
-    for (int idx = threadIdx.x; idx < (DV - DPAD) * (KVB + PADK); idx += 256)
-      vt[DPAD + idx / (KVB + PADK)][idx % (KVB + PADK)] = (__bf16)0.0f;
-  }
+  // zero vt's pad rows once (D..DV); they are never re-staged
+  for (int idx = threadIdx.x + (D / 8) * 8 * (KVB + PADK);
+       idx < DV * (KVB + PADK); idx += 256)
+    ((__bf16 *)vt)[idx] = (__bf16)0.0f;
 
   for (long kv = 0; kv < Sk; kv += KVB) {
     __syncthreads();  // previous tile's LDS reads complete
     // cooperative K/V stage: 256 threads, 8 bf16 each per step
-    for (int idx = threadIdx.x; idx < KVB * NC * 2; idx += 256) {
-      const int r = idx / (NC * 2);       // key row in tile
-      const int c8 = idx % (NC * 2);      // 8-elem column group
+    for (int idx = threadIdx.x; idx < KVB * (DPAD / 8); idx += 256) {
+      const int r = idx / (DPAD / 8);     // key row in tile
+      const int c8 = idx % (DPAD / 8);    // 8-elem column group
       bf16x8 kvec = (bf16x8){};
       bf16x8 vvec = (bf16x8){};
-      if (kv + r < Sk) {
-        kvec = *(const bf16x8 *)(Kb + (kv + r) * DPAD + c8 * 8);
-        vvec = *(const bf16x8 *)(Vb + (kv + r) * DPAD + c8 * 8);
+      if (kv + r < Sk && c8 * 8 + 8 <= D) {
+        kvec = *(const bf16x8 *)(Kb + (kv + r) * st.kr + c8 * 8);
+        vvec = *(const bf16x8 *)(Vb + (kv + r) * st.vr + c8 * 8);
       }
       *(bf16x8 *)&kt[r][c8 * 8] = kvec;
 #pragma unroll
       for (int j = 0; j < 8; ++j) vt[c8 * 8 + j][r] = vvec[j];
-      // (vt rows c8*8+j < DPAD only; pad rows stay zero)
     }
     __syncthreads();
 
     // S^T[32k, 32q] = sum_c K[.,c] x Q^T[c,.]
-    f32x16 st = (f32x16){};
+    f32x16 stile = (f32x16){};
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
       bf16x8 kf = *(const bf16x8 *)&kt[lq][c * 16 + 8 * half];
-      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], st, 0, 0, 0);
+      stile = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], stile, 0, 0, 0);
     }
 
     // online softmax for q-row lq; lane has k = (r&3)+8*(r>>2)+4*half
@@ -106,7 +119,7 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int kk = (r & 3) + 8 * (r >> 2) + 4 * half;
-      float s = (kv + kk < Sk) ? st[r] * scale : -1e30f;
+      float s = (kv + kk < Sk) ? stile[r] * scale : -1e30f;
       p[r] = s;
       pmax = fmaxf(pmax, s);
     }
@@ -124,8 +137,7 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
     m = mnew;
 
     // O *= alpha: O's q-row layout differs from P's (lane-local) layout, so
-    // broadcast alpha[qrow] from the lane that owns that q-row (lane==qrow,
-    // both halves hold identical alpha after the shfl_xor above).
+    // broadcast alpha[qrow] from the lane that owns that q-row.
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
@@ -142,8 +154,7 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
     {
       auto r0 = __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
       auto r1 = __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
-      unsigned w0 = r0[0], w1 = r1[0], w2 = r0[1], w3 = r1[1];
-      unsigned frag[4] = {w0, w1, w2, w3};
+      unsigned frag[4] = {r0[0], r1[0], r0[1], r1[1]};
       pa0 = *(bf16x8 *)frag;
       auto r2 = __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
       auto r3 = __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
@@ -170,46 +181,59 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
     if (q0 + qrow >= Sq) continue;
 #pragma unroll
     for (int d = 0; d < ND; ++d)
-      if (d * 32 + lq < DPAD)
-        Ob[(q0 + qrow) * DPAD + d * 32 + lq] = f2bf(o[d][r] * inv);
+      if (d * 32 + lq < D)
+        Ob[(q0 + qrow) * st.or_ + d * 32 + lq] = f2bf(o[d][r] * inv);
   }
 }
 
 // ---------------------------------------------------------------------------
 // host-side dispatch (torch API lives in ext.hip which includes this file)
 // ---------------------------------------------------------------------------
-bool flash_supported(long d_head) { return d_head > 16 && d_head <= 192; }
+bool flash_supported(long d_head) {
+  return d_head > 16 && d_head <= 192 && (d_head % 8) == 0;
+}
 
 #ifdef __HIP_PLATFORM_AMD__
 static inline long round16(long d) { return (d + 15) / 16 * 16; }
 
-torch::Tensor flash_attention_bf16(torch::Tensor q, torch::Tensor k,
-                                   torch::Tensor v, double scale) {
-  const long B = q.size(0), H = q.size(1);
-  const long Sq = q.size(2), Sk = k.size(2), D = q.size(3);
-  const long DP = round16(D);
-  auto qp = q, kp = k, vp = v;
-  if (DP != D) {
-    namespace F = torch::nn::functional;
-    auto opts = F::PadFuncOptions({0, DP - D});
-    qp = F::pad(q, opts).contiguous();
-    kp = F::pad(k, opts).contiguous();
-    vp = F::pad(v, opts).contiguous();
+// qkv layout: "bhsd" (q.size = [B,H,S,D]) or "bshd" ([B,S,H,D]); tensors
+// need unit stride in D and 16-byte-aligned row starts, NOT full contiguity.
+static torch::Tensor flash_attention_raw(torch::Tensor q, torch::Tensor k,
+                                         torch::Tensor v, double scale,
+                                         bool bshd) {
+  const long B = q.size(0);
+  const long H = bshd ? q.size(2) : q.size(1);
+  const long Sq = bshd ? q.size(1) : q.size(2);
+  const long Sk = bshd ? k.size(1) : k.size(2);
+  const long D = q.size(3);
+  TORCH_CHECK(q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1,
+              "flash: D must be unit-stride");
+  auto out = bshd ? torch::empty({B, Sq, H, D}, q.options())
+                  : torch::empty({B, H, Sq, D}, q.options());
+  AttnStrides st;
+  if (bshd) {
+    st.qb = q.stride(0); st.qh = q.stride(2); st.qr = q.stride(1);
+    st.kb = k.stride(0); st.kh = k.stride(2); st.kr = k.stride(1);
+    st.vb = v.stride(0); st.vh = v.stride(2); st.vr = v.stride(1);
+    st.ob = out.stride(0); st.oh = out.stride(2); st.or_ = out.stride(1);
+  } else {
+    st.qb = q.stride(0); st.qh = q.stride(1); st.qr = q.stride(2);
+    st.kb = k.stride(0); st.kh = k.stride(1); st.kr = k.stride(2);
+    st.vb = v.stride(0); st.vh = v.stride(1); st.vr = v.stride(2);
+    st.ob = out.stride(0); st.oh = out.stride(1); st.or_ = out.stride(2);
   }
-  auto q3 = qp.view({B * H, Sq, DP});
-  auto k3 = kp.view({B * H, Sk, DP});
-  auto v3 = vp.view({B * H, Sk, DP});
-  auto out = torch::empty_like(q3);
+  const long DP = round16(D);
   dim3 grid((unsigned)((Sq + 127) / 128), (unsigned)(B * H));
   dim3 block(256);
   auto stream = cur_stream();
 
 #define LAUNCH_FLASH(DP_)                                                   \
   hipLaunchKernelGGL(flash_fwd_bf16_kernel<DP_>, grid, block, 0, stream,    \
-                     (const __hip_bfloat16 *)q3.data_ptr(),                 \
-                     (const __hip_bfloat16 *)k3.data_ptr(),                 \
-                     (const __hip_bfloat16 *)v3.data_ptr(),                 \
-                     (__hip_bfloat16 *)out.data_ptr(), Sq, Sk, (float)scale)
+                     (const __hip_bfloat16 *)q.data_ptr(),                  \
+                     (const __hip_bfloat16 *)k.data_ptr(),                  \
+                     (const __hip_bfloat16 *)v.data_ptr(),                  \
+                     (__hip_bfloat16 *)out.data_ptr(), (int)H, Sq, Sk,      \
+                     (int)D, (float)scale, st)
 
   switch (DP) {
     case 32: LAUNCH_FLASH(32); break;
@@ -226,8 +250,11 @@ torch::Tensor flash_attention_bf16(torch::Tensor q, torch::Tensor k,
     default: TORCH_CHECK(false, "flash: unsupported padded head dim ", DP);
   }
 #undef LAUNCH_FLASH
-  auto o4 = out.view({B, H, Sq, DP});
-  if (DP != D) o4 = o4.slice(3, 0, D).contiguous();
-  return o4;
+  return out;
+}
+
+torch::Tensor flash_attention_bf16(torch::Tensor q, torch::Tensor k,
+                                   torch::Tensor v, double scale) {
+  return flash_attention_raw(q, k, v, scale, /*bshd=*/false);
 }
 #endif

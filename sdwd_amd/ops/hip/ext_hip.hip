@@ -189,14 +189,14 @@ void row_softmax_(torch::Tensor x, double scale) {
 
 torch::Tensor attention_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                             double scale) {
-  // [B,H,S,D]
+  // [B,H,S,D]; the flash path reads strided (no transpose/pad copies)
+  if (q.scalar_type() == torch::kBFloat16 && flash_supported(q.size(3)) &&
+      q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1) {
+    return flash_attention_raw(q, k, v, scale, /*bshd=*/false);
+  }
   auto qc = q.contiguous(), kc = k.contiguous(), vc = v.contiguous();
   const long B = qc.size(0), H = qc.size(1);
   const long Sq = qc.size(2), Sk = kc.size(2), D = qc.size(3);
-  if (qc.scalar_type() == torch::kBFloat16 && (D % 8) == 0 &&
-      flash_supported(D)) {
-    return flash_attention_bf16(qc, kc, vc, scale);
-  }
   // fallback composition: hipBLASLt GEMMs + fused-softmax kernel, chunked
   // over B*H to bound the score buffer.
   auto q3 = qc.view({B * H, Sq, D});
@@ -235,6 +235,15 @@ torch::Tensor probe_mfma16(torch::Tensor A, torch::Tensor B) {
   return C;
 }
 
+torch::Tensor attention_fwd_bshd(torch::Tensor q, torch::Tensor k,
+                                 torch::Tensor v, double scale) {
+  // [B,S,H,D] (the natural projection layout; avoids transposes entirely)
+  TORCH_CHECK(q.scalar_type() == torch::kBFloat16 &&
+                  flash_supported(q.size(3)),
+              "attention_fwd_bshd: bf16 with supported head dim only");
+  return flash_attention_raw(q, k, v, scale, /*bshd=*/true);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("silu", &silu);
   m.def("geglu", &geglu);
@@ -244,6 +253,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm", &layer_norm);
   m.def("row_softmax_", &row_softmax_);
   m.def("attention_fwd", &attention_fwd);
+  m.def("attention_fwd_bshd", &attention_fwd_bshd);
   m.def("probe_mfma32", &probe_mfma32);
   m.def("probe_mfma16", &probe_mfma16);
   m.def("flash_supported", &flash_supported);

@@ -191,3 +191,41 @@ class TestPipelineGPU:
             eps = m.unet(x, t, ctx)
         assert eps.shape == x.shape
         assert torch.isfinite(eps.float()).all()
+
+
+class TestAttentionBSHD:
+    @pytest.mark.parametrize("B,S,H,D", [
+        (2, 256, 8, 40),
+        (2, 1024, 8, 80),
+        (1, 64, 8, 160),
+        (3, 100, 2, 64),
+    ])
+    def test_bshd_vs_reference(self, dev, B, S, H, D):
+        torch.manual_seed(7)
+        q = torch.randn(B, S, H, D, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(B, S, H, D, device=dev, dtype=torch.bfloat16)
+        v = torch.randn(B, S, H, D, device=dev, dtype=torch.bfloat16)
+        from sdwd_amd import ops as O
+
+        out = O.attention_bshd(q, k, v)
+        scale = 1.0 / math.sqrt(D)
+        qp = q.permute(0, 2, 1, 3).float()
+        kp = k.permute(0, 2, 1, 3).float()
+        vp = v.permute(0, 2, 1, 3).float()
+        s = torch.matmul(qp, kp.transpose(-1, -2)) * scale
+        ref = torch.matmul(s.softmax(-1), vp).permute(0, 2, 1, 3)
+        assert relerr(out, ref) < 0.04
+
+    def test_bhsd_strided_no_copy(self, dev):
+        """The [B,H,S,D] entry works on a transposed (non-contiguous) view."""
+        B, S, H, D = 2, 128, 4, 80
+        q = torch.randn(B, S, H, D, device=dev, dtype=torch.bfloat16)
+        k = torch.randn_like(q)
+        v = torch.randn_like(q)
+        qt = q.transpose(1, 2)  # [B,H,S,D] non-contiguous
+        out = ops.attention(qt, k.transpose(1, 2), v.transpose(1, 2))
+        ref = ops.attention(
+            qt.contiguous(), k.transpose(1, 2).contiguous(),
+            v.transpose(1, 2).contiguous(),
+        )
+        assert relerr(out, ref) < 1e-4
