@@ -207,6 +207,19 @@ def add_noise(
     return (a * x.float() + b * noise.float()).to(x.dtype)
 
 
+def lincomb(x: torch.Tensor, y: torch.Tensor, a: float, b: float):
+    """a*x + b*y, fp32 math, one kernel (sampler updates, CFG combine)."""
+    return add_noise(x, y, a, b)
+
+
+def scale(x: torch.Tensor, c: float) -> torch.Tensor:
+    """c*x in one kernel (sampler input scaling)."""
+    if x.is_cuda:
+        xc = x.contiguous()
+        return ext().axpby(xc, xc, c, 0.0)
+    return (c * x.float()).to(x.dtype)
+
+
 # ---------------------------------------------------------------------------
 # timestep embedding (sinusoidal)
 # ---------------------------------------------------------------------------

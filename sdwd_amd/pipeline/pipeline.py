@@ -273,7 +273,9 @@ class StableDiffusionPipeline:
             else:
                 eps = denoiser(x2, ts, ctx, y)
             eps_c, eps_u = eps.chunk(2, dim=0)
-            return eps_u + cfg * (eps_c - eps_u)
+            from .. import ops as _ops
+
+            return _ops.lincomb(eps_c, eps_u, cfg, 1.0 - cfg)
 
         if cfg == 1.0:
 

@@ -25,13 +25,13 @@ StepCallback = Optional[Callable[[int, int], None]]
 
 
 def _denoised(x: torch.Tensor, eps: torch.Tensor, sigma: float) -> torch.Tensor:
-    return (x.float() - sigma * eps.float()).to(x.dtype)
+    return ops.lincomb(x, eps, 1.0, -sigma)
 
 
 def _eval(model_fn: ModelFn, x: torch.Tensor, sigma: float, t: float):
     """Scale input to unit variance, call the eps-model, return denoised."""
     c_in = 1.0 / math.sqrt(sigma * sigma + 1.0)
-    eps = model_fn((x.float() * c_in).to(x.dtype), t)
+    eps = model_fn(ops.scale(x, c_in), t)
     return _denoised(x, eps, sigma)
 
 
@@ -108,10 +108,10 @@ class Heun(Sampler):
             return ops.euler_step(x, denoised, sigma, sigma_next)
         x1 = ops.euler_step(x, denoised, sigma, sigma_next)
         denoised2 = _eval(model_fn, x1, sigma_next, t)
-        d1 = (x.float() - denoised.float()) / sigma
-        d2 = (x1.float() - denoised2.float()) / sigma_next
-        d = (d1 + d2) / 2
-        return (x.float() + d * (sigma_next - sigma)).to(x.dtype)
+        d1 = ops.lincomb(x, denoised, 1.0 / sigma, -1.0 / sigma)
+        d2 = ops.lincomb(x1, denoised2, 1.0 / sigma_next, -1.0 / sigma_next)
+        d = ops.lincomb(d1, d2, 0.5, 0.5)
+        return ops.lincomb(x, d, 1.0, sigma_next - sigma)
 
 
 class DPMpp2M(Sampler):
@@ -128,16 +128,16 @@ class DPMpp2M(Sampler):
         tn = -math.log(sigma_next)
         h = tn - tt
         if self.old_denoised is None or self.h_last is None:
-            d = denoised.float()
+            d = denoised
         else:
             r = self.h_last / h
-            d = (1 + 1 / (2 * r)) * denoised.float() - (
-                1 / (2 * r)
-            ) * self.old_denoised.float()
-        x = (sigma_next / sigma) * x.float() - math.expm1(-h) * d
+            d = ops.lincomb(
+                denoised, self.old_denoised, 1 + 1 / (2 * r), -1 / (2 * r)
+            )
+        x = ops.lincomb(x, d, sigma_next / sigma, -math.expm1(-h))
         self.old_denoised = denoised
         self.h_last = h
-        return x.to(denoised.dtype)
+        return x
 
 
 class DPMppSDE(Sampler):
