@@ -90,6 +90,30 @@ def _build_sdxl(name: str) -> ModelBundle:
     )
 
 
+def _build_tiny_xl(name: str) -> ModelBundle:
+    """Tiny model exercising the SDXL code path (dual encoders, penultimate
+    hidden states, pooled+size ADM conditioning) at CPU-test scale."""
+    from .unet import UNetConfig
+
+    te = CLIPTextEncoder(d_model=32, layers=2, heads=2)
+    te2 = CLIPTextEncoder(d_model=32, layers=2, heads=2)
+    cfg = UNetConfig(
+        model_channels=32,
+        channel_mult=[1, 2],
+        num_res_blocks=1,
+        transformer_depth=[0, 1],
+        context_dim=64,
+        num_heads=0,  # head_dim 64 convention
+        groups=8,
+        adm_in_channels=32 + 6 * 256,
+    )
+    unet = UNetModel(cfg)
+    vae = AutoencoderKL(VAEConfig.tiny())
+    for seed_off, m in enumerate((te, te2, unet, vae)):
+        _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
+    return ModelBundle(name, te, te2, unet, vae, context_dim=64, is_sdxl=True)
+
+
 def _build_tiny(name: str) -> ModelBundle:
     te = CLIPTextEncoder(d_model=64, layers=2, heads=2, max_len=77)
     unet = UNetModel(UNetConfig.tiny())
@@ -103,6 +127,7 @@ _BUILDERS: Dict[str, Callable[[str], ModelBundle]] = {
     "sd15": _build_sd15,
     "sdxl": _build_sdxl,
     "tiny": _build_tiny,
+    "tiny-xl": _build_tiny_xl,
 }
 
 _cache: Dict[str, ModelBundle] = {}

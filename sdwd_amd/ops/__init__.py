@@ -91,9 +91,9 @@ def group_norm_silu(
     HBM-bound sweep with ushort8-vectorised bf16 loads.
     """
     if x.is_cuda:
-        return ext().group_norm_silu(
-            x.contiguous(), weight, bias, groups, eps, silu
-        )
+        if not x.is_contiguous(memory_format=torch.channels_last):
+            x = x.contiguous()  # keep channels_last when already NHWC
+        return ext().group_norm_silu(x, weight, bias, groups, eps, silu)
     out = F.group_norm(x.float(), groups, weight.float(), bias.float(), eps)
     if silu:
         out = F.silu(out)

@@ -103,11 +103,54 @@ torch::Tensor euler_step(torch::Tensor x, torch::Tensor denoised, double sigma,
 }
 
 // ---------------------------------------------------------------------------
+torch::Tensor group_norm_silu_nhwc(torch::Tensor x, torch::Tensor w,
+                                   torch::Tensor b, long groups, double eps,
+                                   bool silu_act) {
+  // x: NCHW sizes, channels_last memory (= NHWC rows of C)
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "nhwc GN: bf16 only");
+  const long N = x.size(0);
+  const int C = x.size(1);
+  const long HW = x.size(2) * (long)x.size(3);
+  TORCH_CHECK(C % 8 == 0 && C <= 2048, "nhwc GN: C%8==0 and C<=2048");
+  auto wf = w.to(torch::kFloat).contiguous();
+  auto bf = b.to(torch::kFloat).contiguous();
+  auto out = torch::empty_like(x);  // keeps channels_last strides
+  const int S = (int)std::min<long>(
+      std::max<long>(1, 2048 / std::max<long>(N, 1)), (HW + 255) / 256);
+  auto opts = torch::TensorOptions()
+                  .dtype(torch::kFloat)
+                  .device(x.device());
+  auto partial = torch::empty({N * S, 2L * C}, opts);
+  auto stats = torch::empty({N * (long)groups, 2L}, opts);
+  auto stream = cur_stream();
+  hipLaunchKernelGGL(gn_nhwc_partial_bf16, dim3((unsigned)(N * S)), dim3(256),
+                     0, stream, (const __hip_bfloat16 *)x.data_ptr(),
+                     partial.data_ptr<float>(), C, HW, S);
+  hipLaunchKernelGGL(gn_nhwc_stats, dim3((unsigned)(N * groups)), dim3(WAVE),
+                     0, stream, partial.data_ptr<float>(),
+                     stats.data_ptr<float>(), C, HW, (int)groups, S,
+                     (float)eps);
+  const long total = N * HW * (C / 8);
+  auto kern =
+      silu_act ? gn_nhwc_norm_bf16<true> : gn_nhwc_norm_bf16<false>;
+  hipLaunchKernelGGL(kern, dim3(ew_grid(total)), dim3(256), 0, stream,
+                     (const __hip_bfloat16 *)x.data_ptr(),
+                     stats.data_ptr<float>(), wf.data_ptr<float>(),
+                     bf.data_ptr<float>(),
+                     (__hip_bfloat16 *)out.data_ptr(), C, HW, (int)groups, N);
+  return out;
+}
+
 torch::Tensor group_norm_silu(torch::Tensor x, torch::Tensor w,
                               torch::Tensor b, long groups, double eps,
                               bool silu_act) {
-  CHECK_IN(x);
   TORCH_CHECK(x.dim() == 4, "expect NCHW");
+  if (x.is_contiguous(torch::MemoryFormat::ChannelsLast) &&
+      x.scalar_type() == torch::kBFloat16 && x.size(1) % 8 == 0 &&
+      x.size(1) <= 2048 && x.size(1) > 8) {
+    return group_norm_silu_nhwc(x, w, b, groups, eps, silu_act);
+  }
+  CHECK_IN(x);
   const int N = x.size(0), C = x.size(1);
   const long HW = x.size(2) * (long)x.size(3);
   TORCH_CHECK(C % groups == 0);

@@ -116,6 +116,11 @@ class StableDiffusionPipeline:
         else:
             model.to(self.device, dtype)
         self.model = model
+        if self.device.type == "cuda":
+            # channels_last: MIOpen NHWC conv solvers are ~30% faster on
+            # gfx950 and skip the NCHW<->NHWC transposes (conv_ab probe).
+            self.model.unet.to(memory_format=torch.channels_last)
+            self.model.vae.to(memory_format=torch.channels_last)
         self._denoiser = GraphedDenoiser(
             lambda x, ts, ctx, y: self.model.unet(x, ts, ctx, y=y),
             self.device,
@@ -254,6 +259,8 @@ class StableDiffusionPipeline:
             controlnet = load_controlnet(
                 req.control_model, device=self.device, dtype=self.dtype
             )
+            if self.device.type == "cuda":
+                controlnet.to(memory_format=torch.channels_last)
             hint = (
                 req.control_image.permute(0, 3, 1, 2).float() / 255.0
             ).to(self.device, self.dtype)

@@ -229,3 +229,33 @@ class TestAttentionBSHD:
             v.transpose(1, 2).contiguous(),
         )
         assert relerr(out, ref) < 1e-4
+
+
+class TestGroupNormNHWC:
+    @pytest.mark.parametrize("shape,groups", [
+        ((2, 320, 64, 64), 32),
+        ((4, 1280, 8, 8), 32),
+        ((2, 128, 512, 512), 32),
+        ((2, 32, 16, 16), 8),
+    ])
+    def test_channels_last_matches_fp32(self, dev, shape, groups):
+        x = torch.randn(*shape, device=dev, dtype=torch.bfloat16)
+        xc = x.to(memory_format=torch.channels_last)
+        w = torch.randn(shape[1], device=dev)
+        b = torch.randn(shape[1], device=dev)
+        out = ops.group_norm_silu(xc, w, b, groups)
+        assert out.is_contiguous(memory_format=torch.channels_last)
+        ref = torch.nn.functional.silu(
+            torch.nn.functional.group_norm(x.float(), groups, w, b, 1e-5)
+        )
+        assert relerr(out.contiguous(), ref) < 0.05
+
+    def test_nhwc_equals_nchw_path(self, dev):
+        x = torch.randn(3, 64, 32, 32, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(64, device=dev)
+        b = torch.randn(64, device=dev)
+        a = ops.group_norm_silu(x, w, b, 32)
+        c = ops.group_norm_silu(
+            x.to(memory_format=torch.channels_last), w, b, 32
+        ).contiguous()
+        assert relerr(a, c) < 0.02

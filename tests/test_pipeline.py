@@ -241,3 +241,28 @@ class TestControlNet:
         a = pipe.generate(req).images
         b = pipe.generate(req).images
         assert torch.equal(a, b)
+
+
+class TestSDXLPath:
+    def test_tiny_xl_generate(self):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny-xl", device="cpu")
+        assert pipe.model.is_sdxl
+        res = pipe.generate(
+            PipelineRequest(
+                prompt="xl cow", steps=2, width=64, height=64, seeds=[1, 2]
+            )
+        )
+        assert res.images.shape == (2, 64, 64, 3)
+        assert torch.isfinite(res.images.float()).all()
+
+    def test_tiny_xl_deterministic(self):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny-xl", device="cpu")
+        req = PipelineRequest(prompt="d", steps=2, width=64, height=64,
+                              seeds=[5])
+        a = pipe.generate(req).images
+        b = pipe.generate(req).images
+        assert torch.equal(a, b)
