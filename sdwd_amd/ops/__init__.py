@@ -91,8 +91,17 @@ def group_norm_silu(
     HBM-bound sweep with ushort8-vectorised bf16 loads.
     """
     if x.is_cuda:
-        if not x.is_contiguous(memory_format=torch.channels_last):
-            x = x.contiguous()  # keep channels_last when already NHWC
+        # keep channels_last only when the NHWC kernel accepts the shape;
+        # otherwise force plain NCHW contiguity (mirrors ext.hip dispatch)
+        c = x.shape[1]
+        nhwc_ok = (
+            x.is_contiguous(memory_format=torch.channels_last)
+            and x.dtype == torch.bfloat16
+            and c % 8 == 0
+            and 8 < c <= 3072
+        )
+        if not nhwc_ok and not x.is_contiguous():
+            x = x.contiguous()
         return ext().group_norm_silu(x, weight, bias, groups, eps, silu)
     out = F.group_norm(x.float(), groups, weight.float(), bias.float(), eps)
     if silu:

@@ -111,7 +111,7 @@ torch::Tensor group_norm_silu_nhwc(torch::Tensor x, torch::Tensor w,
   const long N = x.size(0);
   const int C = x.size(1);
   const long HW = x.size(2) * (long)x.size(3);
-  TORCH_CHECK(C % 8 == 0 && C <= 2048, "nhwc GN: C%8==0 and C<=2048");
+  TORCH_CHECK(C % 8 == 0 && C <= 3072, "nhwc GN: C%8==0 and C<=3072");
   auto wf = w.to(torch::kFloat).contiguous();
   auto bf = b.to(torch::kFloat).contiguous();
   auto out = torch::empty_like(x);  // keeps channels_last strides
@@ -147,7 +147,7 @@ torch::Tensor group_norm_silu(torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(x.dim() == 4, "expect NCHW");
   if (x.is_contiguous(torch::MemoryFormat::ChannelsLast) &&
       x.scalar_type() == torch::kBFloat16 && x.size(1) % 8 == 0 &&
-      x.size(1) <= 2048 && x.size(1) > 8) {
+      x.size(1) <= 3072 && x.size(1) > 8) {
     return group_norm_silu_nhwc(x, w, b, groups, eps, silu_act);
   }
   CHECK_IN(x);

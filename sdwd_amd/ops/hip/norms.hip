@@ -216,9 +216,9 @@ __global__ void gn_nhwc_partial_bf16(const __hip_bfloat16 *__restrict__ x,
   const __hip_bfloat16 *base = x + (long)n * HW * C;
 
   const int cpt = (C + 255) / 256;  // channels owned per thread
-  float sum[8], sumsq[8];           // cpt <= 8 (C <= 2048)
+  float sum[12], sumsq[12];         // cpt <= 12 (C <= 3072; 2560 after cat)
 #pragma unroll
-  for (int j = 0; j < 8; ++j) sum[j] = sumsq[j] = 0.f;
+  for (int j = 0; j < 12; ++j) sum[j] = sumsq[j] = 0.f;
 
   for (long hw = hw0; hw < hw1; ++hw) {
     const __hip_bfloat16 *row = base + hw * C;

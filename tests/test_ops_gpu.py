@@ -236,6 +236,7 @@ class TestGroupNormNHWC:
         ((2, 320, 64, 64), 32),
         ((4, 1280, 8, 8), 32),
         ((2, 128, 512, 512), 32),
+        ((1, 2560, 16, 16), 32),
         ((2, 32, 16, 16), 8),
     ])
     def test_channels_last_matches_fp32(self, dev, shape, groups):
