@@ -37,6 +37,9 @@ def main() -> int:
     ap.add_argument("--denoise-steps", type=int, default=20)
     ap.add_argument("--sampler", type=str, default="Euler a")
     ap.add_argument("--cfg", type=float, default=7.0)
+    ap.add_argument("--task", choices=["txt2img", "img2img"], default="txt2img",
+                    help="img2img adds VAE encode + strength-limited denoise")
+    ap.add_argument("--strength", type=float, default=0.75)
     args = ap.parse_args()
 
     from sdwd_amd.config import add_flags  # noqa: F401  (flag surface)
@@ -53,6 +56,13 @@ def main() -> int:
     for w in engine.world.workers:
         w.eta.avg_ipm = 60.0
 
+    init_images = None
+    if args.task == "img2img":
+        g = torch.Generator().manual_seed(7)
+        init_images = torch.randint(
+            0, 255, (args.global_batch, args.height, args.width, 3),
+            generator=g, dtype=torch.uint8,
+        )
     req = GenerationRequest(
         prompt="A herd of cows grazing at the bottom of a sunny valley",
         negative_prompt="blurry, low quality",
@@ -63,6 +73,8 @@ def main() -> int:
         cfg_scale=args.cfg,
         sampler_name=args.sampler,
         seed=1234,
+        init_images=init_images,
+        denoising_strength=args.strength,
     )
 
     def sync():
@@ -107,6 +119,7 @@ def main() -> int:
             "dtype": "bf16" if have_gpu else "fp32",
             "data": "synthetic prompts, random-init weights",
             "config": {
+                "task": args.task,
                 "model": args.model,
                 "global_batch": args.global_batch,
                 "seq_len": args.width,
