@@ -10,6 +10,7 @@ generator seeded with that image's seed, then moved to the device.
 from __future__ import annotations
 
 import math
+import os
 import time
 from dataclasses import dataclass, field
 from typing import Callable, List, Optional
@@ -356,14 +357,20 @@ class StableDiffusionPipeline:
             )
 
         if decode:
-            pixels = self.model.vae.decode(x)
-            images = (
-                ((pixels.float() + 1.0) * 127.5)
-                .clamp(0, 255)
-                .to(torch.uint8)
-                .permute(0, 2, 3, 1)
-                .cpu()
-            )
+            # chunked decode: bounds decoder activation memory at large
+            # batches and keeps per-chunk conv shapes in MIOpen's sweet spot
+            chunk = int(os.environ.get("SDWD_DECODE_CHUNK", "16"))
+            outs = []
+            for i in range(0, x.shape[0], chunk):
+                pixels = self.model.vae.decode(x[i : i + chunk])
+                outs.append(
+                    ((pixels.float() + 1.0) * 127.5)
+                    .clamp(0, 255)
+                    .to(torch.uint8)
+                    .permute(0, 2, 3, 1)
+                    .cpu()
+                )
+            images = torch.cat(outs) if len(outs) > 1 else outs[0]
         else:
             images = x.cpu()
 
