@@ -28,6 +28,9 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--batch", type=int, default=64, help="images (pre-CFG)")
     ap.add_argument("--res", type=int, default=512)
+    ap.add_argument("--only", choices=["all", "unet", "vae"], default="all")
+    ap.add_argument("--nchw", action="store_true",
+                    help="keep NCHW (disable channels_last)")
     args = ap.parse_args()
     assert torch.cuda.is_available()
     dev = torch.device("cuda:0")
@@ -36,6 +39,9 @@ def main():
     from sdwd_amd import ops
 
     m = load_model("sd15", device=dev, dtype=dt)
+    if not args.nchw:
+        m.unet.to(memory_format=torch.channels_last)
+        m.vae.to(memory_format=torch.channels_last)
     b2 = args.batch * 2  # CFG doubling
     lat = args.res // 8
     x = torch.randn(b2, 4, lat, lat, device=dev, dtype=dt)
@@ -45,6 +51,16 @@ def main():
 
     with torch.no_grad():
         out["unet_eager_ms"] = timeit(lambda: m.unet(x, ts, ctx)) * 1000
+        if args.only == "unet":
+            print(json.dumps(out, indent=1))
+            return
+        if args.only == "vae":
+            z = torch.randn(args.batch, 4, args.res // 8, args.res // 8,
+                            device=dev, dtype=dt)
+            out["vae_decode_ms"] = timeit(
+                lambda: m.vae.decode(z), iters=3) * 1000
+            print(json.dumps(out, indent=1))
+            return
 
         from sdwd_amd.pipeline.graphs import GraphedDenoiser
 
