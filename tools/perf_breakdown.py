@@ -40,6 +40,7 @@ def main():
 
     m = load_model("sd15", device=dev, dtype=dt)
     if not args.nchw:
+        torch.backends.cudnn.benchmark = True
         m.unet.to(memory_format=torch.channels_last)
         m.vae.to(memory_format=torch.channels_last)
     b2 = args.batch * 2  # CFG doubling
