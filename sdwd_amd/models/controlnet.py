@@ -14,11 +14,12 @@ import torch
 import torch.nn as nn
 
 from .. import ops
+from .layers import SDConv2d
 from .unet import ResBlock, SpatialTransformer, Downsample, UNetConfig, _Seq
 
 
 def zero_conv(ch: int) -> nn.Conv2d:
-    conv = nn.Conv2d(ch, ch, 1)
+    conv = SDConv2d(ch, ch, 1)
     nn.init.zeros_(conv.weight)
     nn.init.zeros_(conv.bias)
     return conv
@@ -30,13 +31,13 @@ class HintEncoder(nn.Module):
     def __init__(self, model_channels: int, factor: int = 8):
         super().__init__()
         chans = [16, 32, 96, 256]
-        layers: List[nn.Module] = [nn.Conv2d(3, chans[0], 3, padding=1)]
+        layers: List[nn.Module] = [SDConv2d(3, chans[0], 3, padding=1)]
         cur = chans[0]
         n_stride = {8: 3, 4: 2, 2: 1, 1: 0}.get(factor, 3)
         for i in range(3):
             nxt = chans[min(i + 1, len(chans) - 1)]
             stride = 2 if i < n_stride else 1
-            layers += [nn.SiLU(), nn.Conv2d(cur, nxt, 3, padding=1, stride=stride)]
+            layers += [nn.SiLU(), SDConv2d(cur, nxt, 3, padding=1, stride=stride)]
             cur = nxt
         layers += [nn.SiLU(), zero_conv_out(cur, model_channels)]
         self.body = nn.Sequential(*layers)
@@ -46,7 +47,7 @@ class HintEncoder(nn.Module):
 
 
 def zero_conv_out(in_ch: int, out_ch: int) -> nn.Conv2d:
-    conv = nn.Conv2d(in_ch, out_ch, 3, padding=1)
+    conv = SDConv2d(in_ch, out_ch, 3, padding=1)
     nn.init.zeros_(conv.weight)
     nn.init.zeros_(conv.bias)
     return conv
@@ -62,7 +63,7 @@ class ControlNetModel(nn.Module):
         self.time_mlp = nn.Sequential(
             nn.Linear(ch, time_dim), nn.SiLU(), nn.Linear(time_dim, time_dim)
         )
-        self.conv_in = nn.Conv2d(cfg.in_channels, ch, 3, padding=1)
+        self.conv_in = SDConv2d(cfg.in_channels, ch, 3, padding=1)
         self.hint_encoder = HintEncoder(ch, hint_factor)
 
         self.down = nn.ModuleList()

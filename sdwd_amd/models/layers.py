@@ -43,3 +43,79 @@ class GEGLU(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return ops.geglu(self.proj(x))
+
+
+class SDConv2d(nn.Conv2d):
+    """Conv2d dispatching to the MI355X-native paths:
+
+    * 3x3 pad-1 stride-1/2 with Cin % 32 == 0, bf16 channels_last on GPU ->
+      the implicit-GEMM MFMA kernel (ops/hip/conv.hip), with bias and an
+      optional residual add fused into the epilogue;
+    * 1x1 stride-1 on GPU -> a hipBLASLt GEMM over the NHWC rows;
+    * everything else (stem convs, CPU) -> F.conv2d (MIOpen / native).
+    """
+
+    def __init__(self, *args, **kwargs):
+        super().__init__(*args, **kwargs)
+        self._wprep_cache = None
+
+    def _wprep(self) -> torch.Tensor:
+        w = self.weight
+        cache = self._wprep_cache
+        if (
+            cache is None
+            or cache[0] != w.data_ptr()
+            or cache[1] != w.dtype
+        ):
+            prep = w.permute(0, 2, 3, 1).contiguous()  # [Cout,3,3,Cin]
+            self._wprep_cache = (w.data_ptr(), w.dtype, prep)
+            return prep
+        return cache[2]
+
+    def forward(  # type: ignore[override]
+        self, x: torch.Tensor, residual: torch.Tensor | None = None
+    ) -> torch.Tensor:
+        import torch.nn.functional as F
+
+        from .. import ops
+
+        k = self.kernel_size
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            if (
+                k == (3, 3)
+                and self.padding == (1, 1)
+                and self.stride[0] in (1, 2)
+                and self.stride[0] == self.stride[1]
+                and self.in_channels % 32 == 0
+                and self.groups == 1
+            ):
+                xc = x.contiguous(memory_format=torch.channels_last)
+                res = None
+                if residual is not None:
+                    res = residual.contiguous(
+                        memory_format=torch.channels_last
+                    )
+                return ops.conv3x3(
+                    xc, self._wprep(), self.bias, res, self.stride[0]
+                )
+            if k == (1, 1) and self.stride == (1, 1) and self.groups == 1:
+                xc = x.contiguous(memory_format=torch.channels_last)
+                n, c, h, w = xc.shape
+                rows = xc.permute(0, 2, 3, 1).reshape(n * h * w, c)
+                out = F.linear(
+                    rows, self.weight.reshape(self.out_channels, c), self.bias
+                )
+                out = (
+                    out.reshape(n, h, w, self.out_channels)
+                    .permute(0, 3, 1, 2)
+                )
+                if residual is not None:
+                    out = out + residual
+                return out
+        out = F.conv2d(
+            x, self.weight, self.bias, self.stride, self.padding,
+            self.dilation, self.groups,
+        )
+        if residual is not None:
+            out = out + residual
+        return out

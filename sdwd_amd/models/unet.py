@@ -21,7 +21,7 @@ import torch
 import torch.nn as nn
 
 from .. import ops
-from .layers import GEGLU, FusedGroupNorm, FusedLayerNorm
+from .layers import GEGLU, FusedGroupNorm, FusedLayerNorm, SDConv2d
 
 
 @dataclass
@@ -77,19 +77,20 @@ class ResBlock(nn.Module):
     def __init__(self, in_ch: int, out_ch: int, emb_ch: int, groups: int):
         super().__init__()
         self.norm1 = FusedGroupNorm(in_ch, groups, silu=True)
-        self.conv1 = nn.Conv2d(in_ch, out_ch, 3, padding=1)
+        self.conv1 = SDConv2d(in_ch, out_ch, 3, padding=1)
         self.emb_proj = nn.Linear(emb_ch, out_ch)
         self.norm2 = FusedGroupNorm(out_ch, groups, silu=True)
-        self.conv2 = nn.Conv2d(out_ch, out_ch, 3, padding=1)
+        self.conv2 = SDConv2d(out_ch, out_ch, 3, padding=1)
         self.skip = (
-            nn.Conv2d(in_ch, out_ch, 1) if in_ch != out_ch else nn.Identity()
+            SDConv2d(in_ch, out_ch, 1) if in_ch != out_ch else nn.Identity()
         )
 
     def forward(self, x: torch.Tensor, emb: torch.Tensor) -> torch.Tensor:
         h = self.conv1(self.norm1(x))
         h = h + self.emb_proj(ops.silu(emb))[:, :, None, None]
-        h = self.conv2(self.norm2(h))
-        return h + self.skip(x)
+        skip = self.skip(x)
+        # the residual add is fused into conv2's epilogue on GPU
+        return self.conv2(self.norm2(h), residual=skip)
 
 
 class CrossAttention(nn.Module):
@@ -162,7 +163,7 @@ class SpatialTransformer(nn.Module):
 class Downsample(nn.Module):
     def __init__(self, ch: int):
         super().__init__()
-        self.conv = nn.Conv2d(ch, ch, 3, stride=2, padding=1)
+        self.conv = SDConv2d(ch, ch, 3, stride=2, padding=1)
 
     def forward(self, x):
         return self.conv(x)
@@ -171,7 +172,7 @@ class Downsample(nn.Module):
 class Upsample(nn.Module):
     def __init__(self, ch: int):
         super().__init__()
-        self.conv = nn.Conv2d(ch, ch, 3, padding=1)
+        self.conv = SDConv2d(ch, ch, 3, padding=1)
 
     def forward(self, x):
         x = torch.nn.functional.interpolate(x, scale_factor=2, mode="nearest")
@@ -214,7 +215,7 @@ class UNetModel(nn.Module):
         else:
             self.label_mlp = None
 
-        self.conv_in = nn.Conv2d(cfg.in_channels, ch, 3, padding=1)
+        self.conv_in = SDConv2d(cfg.in_channels, ch, 3, padding=1)
 
         self.down = nn.ModuleList()
         skip_chs = [ch]
@@ -270,7 +271,7 @@ class UNetModel(nn.Module):
                 self.up.append(_Seq(*mods))
 
         self.norm_out = FusedGroupNorm(cur, cfg.groups, silu=True)
-        self.conv_out = nn.Conv2d(cur, cfg.out_channels, 3, padding=1)
+        self.conv_out = SDConv2d(cur, cfg.out_channels, 3, padding=1)
 
     def forward(
         self,

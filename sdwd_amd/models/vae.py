@@ -14,7 +14,7 @@ import torch
 import torch.nn as nn
 
 from .. import ops
-from .layers import FusedGroupNorm
+from .layers import FusedGroupNorm, SDConv2d
 
 SD_VAE_SCALE = 0.18215
 SDXL_VAE_SCALE = 0.13025
@@ -51,17 +51,16 @@ class VAEResBlock(nn.Module):
     def __init__(self, in_ch: int, out_ch: int, groups: int):
         super().__init__()
         self.norm1 = FusedGroupNorm(in_ch, groups, silu=True)
-        self.conv1 = nn.Conv2d(in_ch, out_ch, 3, padding=1)
+        self.conv1 = SDConv2d(in_ch, out_ch, 3, padding=1)
         self.norm2 = FusedGroupNorm(out_ch, groups, silu=True)
-        self.conv2 = nn.Conv2d(out_ch, out_ch, 3, padding=1)
+        self.conv2 = SDConv2d(out_ch, out_ch, 3, padding=1)
         self.skip = (
-            nn.Conv2d(in_ch, out_ch, 1) if in_ch != out_ch else nn.Identity()
+            SDConv2d(in_ch, out_ch, 1) if in_ch != out_ch else nn.Identity()
         )
 
     def forward(self, x):
         h = self.conv1(self.norm1(x))
-        h = self.conv2(self.norm2(h))
-        return h + self.skip(x)
+        return self.conv2(self.norm2(h), residual=self.skip(x))
 
 
 class VAEAttention(nn.Module):
@@ -90,7 +89,7 @@ class VAEEncoder(nn.Module):
     def __init__(self, cfg: VAEConfig):
         super().__init__()
         ch = cfg.base_channels
-        self.conv_in = nn.Conv2d(3, ch, 3, padding=1)
+        self.conv_in = SDConv2d(3, ch, 3, padding=1)
         blocks: List[nn.Module] = []
         cur = ch
         for lvl, mult in enumerate(cfg.channel_mult):
@@ -99,7 +98,7 @@ class VAEEncoder(nn.Module):
                 blocks.append(VAEResBlock(cur, out_ch, cfg.groups))
                 cur = out_ch
             if lvl != len(cfg.channel_mult) - 1:
-                blocks.append(nn.Conv2d(cur, cur, 3, stride=2, padding=1))
+                blocks.append(SDConv2d(cur, cur, 3, stride=2, padding=1))
         self.blocks = nn.ModuleList(blocks)
         self.mid = nn.ModuleList(
             [
@@ -109,7 +108,7 @@ class VAEEncoder(nn.Module):
             ]
         )
         self.norm_out = FusedGroupNorm(cur, cfg.groups, silu=True)
-        self.conv_out = nn.Conv2d(cur, 2 * cfg.latent_channels, 3, padding=1)
+        self.conv_out = SDConv2d(cur, 2 * cfg.latent_channels, 3, padding=1)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         """[B,3,H,W] -> moments [B, 2*latent, H/8, W/8]."""
@@ -126,7 +125,7 @@ class VAEDecoder(nn.Module):
         super().__init__()
         ch = cfg.base_channels
         cur = ch * cfg.channel_mult[-1]
-        self.conv_in = nn.Conv2d(cfg.latent_channels, cur, 3, padding=1)
+        self.conv_in = SDConv2d(cfg.latent_channels, cur, 3, padding=1)
         self.mid = nn.ModuleList(
             [
                 VAEResBlock(cur, cur, cfg.groups),
@@ -144,7 +143,7 @@ class VAEDecoder(nn.Module):
                 blocks.append(_Up(cur))
         self.blocks = nn.ModuleList(blocks)
         self.norm_out = FusedGroupNorm(cur, cfg.groups, silu=True)
-        self.conv_out = nn.Conv2d(cur, 3, 3, padding=1)
+        self.conv_out = SDConv2d(cur, 3, 3, padding=1)
 
     def forward(self, z: torch.Tensor) -> torch.Tensor:
         """latents [B,4,h,w] -> pixels [B,3,8h,8w] in [-1,1]."""
@@ -159,7 +158,7 @@ class VAEDecoder(nn.Module):
 class _Up(nn.Module):
     def __init__(self, ch: int):
         super().__init__()
-        self.conv = nn.Conv2d(ch, ch, 3, padding=1)
+        self.conv = SDConv2d(ch, ch, 3, padding=1)
 
     def forward(self, x):
         return self.conv(

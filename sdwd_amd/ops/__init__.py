@@ -229,6 +229,18 @@ def scale(x: torch.Tensor, c: float) -> torch.Tensor:
     return (c * x.float()).to(x.dtype)
 
 
+def conv3x3(
+    x: torch.Tensor,
+    w_prep: torch.Tensor,
+    bias: Optional[torch.Tensor],
+    residual: Optional[torch.Tensor],
+    stride: int = 1,
+) -> torch.Tensor:
+    """NHWC implicit-GEMM 3x3 conv (pad 1), bias + residual fused into the
+    epilogue. GPU-only entry (callers fall back to F.conv2d on CPU)."""
+    return ext().conv3x3_nhwc(x, w_prep, bias, residual, stride)
+
+
 # ---------------------------------------------------------------------------
 # timestep embedding (sinusoidal)
 # ---------------------------------------------------------------------------

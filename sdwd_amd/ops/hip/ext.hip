@@ -15,6 +15,7 @@ static inline hipStream_t cur_stream() {
 #include "softmax.hip"
 #include "probe.hip"
 #include "attention.hip"
+#include "conv.hip"
 
 #define CHECK_IN(x)                                                     \
   TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                        \
@@ -299,4 +300,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("probe_mfma32", &probe_mfma32);
   m.def("probe_mfma16", &probe_mfma16);
   m.def("flash_supported", &flash_supported);
+  m.def("conv3x3_nhwc", &conv3x3_nhwc);
+  m.def("conv3x3_supported", &conv3x3_supported);
 }

@@ -120,12 +120,10 @@ class StableDiffusionPipeline:
         if self.device.type == "cuda":
             # channels_last: MIOpen NHWC conv solvers are ~30% faster on
             # gfx950 and skip the NCHW<->NHWC transposes (conv_ab probe).
-            # benchmark find is REQUIRED with channels_last: immediate-mode
-            # solver selection falls back to slow no-workspace kernels on
-            # the UNet's concat shapes (2560/1920/960 ch) — measured 3x
-            # whole-step regression without it.
+            # MIOpen benchmark-find is too slow on fresh machines (~10 min
+            # of solver tuning per process); immediate FAST find + our own
+            # implicit-GEMM convs for the hot 3x3/1x1 shapes instead.
             os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
-            torch.backends.cudnn.benchmark = True
             self.model.unet.to(memory_format=torch.channels_last)
             self.model.vae.to(memory_format=torch.channels_last)
         self._denoiser = GraphedDenoiser(

@@ -260,3 +260,64 @@ class TestGroupNormNHWC:
             x.to(memory_format=torch.channels_last), w, b, 32
         ).contiguous()
         assert relerr(a, c) < 0.02
+
+
+class TestConv3x3:
+    @pytest.mark.parametrize("N,Cin,H,W,Cout,stride", [
+        (2, 32, 16, 16, 64, 1),
+        (2, 320, 32, 32, 320, 1),
+        (1, 320, 64, 64, 320, 1),
+        (2, 640, 17, 17, 320, 1),   # odd spatial
+        (2, 320, 32, 32, 320, 2),   # downsample
+        (1, 2560, 8, 8, 1280, 1),   # concat shape
+        (1, 128, 96, 96, 40, 1),    # Cout not multiple of tile
+    ])
+    def test_vs_conv2d(self, dev, N, Cin, H, W, Cout, stride):
+        torch.manual_seed(3)
+        x = torch.randn(N, Cin, H, W, device=dev, dtype=torch.bfloat16)
+        conv = torch.nn.Conv2d(Cin, Cout, 3, stride=stride, padding=1)
+        conv = conv.to(dev, torch.bfloat16)
+        ref = torch.nn.functional.conv2d(
+            x.float(), conv.weight.float(), conv.bias.float(),
+            stride=stride, padding=1,
+        )
+        xc = x.contiguous(memory_format=torch.channels_last)
+        wprep = conv.weight.permute(0, 2, 3, 1).contiguous()
+        out = ops.conv3x3(xc, wprep, conv.bias, None, stride)
+        err = relerr(out.contiguous(), ref)
+        assert err < 0.05, f"conv err {err}"
+
+    def test_fused_residual(self, dev):
+        x = torch.randn(2, 64, 16, 16, device=dev, dtype=torch.bfloat16)
+        res = torch.randn(2, 64, 16, 16, device=dev, dtype=torch.bfloat16)
+        conv = torch.nn.Conv2d(64, 64, 3, padding=1).to(dev, torch.bfloat16)
+        ref = torch.nn.functional.conv2d(
+            x.float(), conv.weight.float(), conv.bias.float(), padding=1
+        ) + res.float()
+        xc = x.contiguous(memory_format=torch.channels_last)
+        rc = res.contiguous(memory_format=torch.channels_last)
+        wprep = conv.weight.permute(0, 2, 3, 1).contiguous()
+        out = ops.conv3x3(xc, wprep, conv.bias, rc, 1)
+        assert relerr(out.contiguous(), ref) < 0.05
+
+    def test_sdconv_module_dispatch(self, dev):
+        from sdwd_amd.models.layers import SDConv2d
+
+        conv = SDConv2d(320, 320, 3, padding=1).to(dev, torch.bfloat16)
+        x = torch.randn(2, 320, 32, 32, device=dev, dtype=torch.bfloat16)
+        out = conv(x.contiguous(memory_format=torch.channels_last))
+        ref = torch.nn.functional.conv2d(
+            x.float(), conv.weight.float(), conv.bias.float(), padding=1
+        )
+        assert relerr(out.contiguous(), ref) < 0.05
+
+    def test_sdconv_1x1_gemm(self, dev):
+        from sdwd_amd.models.layers import SDConv2d
+
+        conv = SDConv2d(320, 640, 1).to(dev, torch.bfloat16)
+        x = torch.randn(2, 320, 16, 16, device=dev, dtype=torch.bfloat16)
+        out = conv(x)
+        ref = torch.nn.functional.conv2d(
+            x.float(), conv.weight.float(), conv.bias.float()
+        )
+        assert relerr(out.contiguous(), ref) < 0.05
