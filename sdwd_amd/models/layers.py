@@ -73,7 +73,10 @@ class SDConv2d(nn.Conv2d):
         return cache[2]
 
     def forward(  # type: ignore[override]
-        self, x: torch.Tensor, residual: torch.Tensor | None = None
+        self,
+        x: torch.Tensor,
+        residual: torch.Tensor | None = None,
+        chan_bias: torch.Tensor | None = None,
     ) -> torch.Tensor:
         import torch.nn.functional as F
 
@@ -96,7 +99,8 @@ class SDConv2d(nn.Conv2d):
                         memory_format=torch.channels_last
                     )
                 return ops.conv3x3(
-                    xc, self._wprep(), self.bias, res, self.stride[0]
+                    xc, self._wprep(), self.bias, res, self.stride[0],
+                    chan_bias,
                 )
             if k == (1, 1) and self.stride == (1, 1) and self.groups == 1:
                 xc = x.contiguous(memory_format=torch.channels_last)
@@ -116,6 +120,8 @@ class SDConv2d(nn.Conv2d):
             x, self.weight, self.bias, self.stride, self.padding,
             self.dilation, self.groups,
         )
+        if chan_bias is not None:
+            out = out + chan_bias.to(out.dtype)[:, :, None, None]
         if residual is not None:
             out = out + residual
         return out

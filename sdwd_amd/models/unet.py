@@ -86,10 +86,11 @@ class ResBlock(nn.Module):
         )
 
     def forward(self, x: torch.Tensor, emb: torch.Tensor) -> torch.Tensor:
-        h = self.conv1(self.norm1(x))
-        h = h + self.emb_proj(ops.silu(emb))[:, :, None, None]
+        # the time-embedding channel bias rides conv1's epilogue; the
+        # skip-residual add rides conv2's (both fused on GPU)
+        emb_b = self.emb_proj(ops.silu(emb))
+        h = self.conv1(self.norm1(x), chan_bias=emb_b)
         skip = self.skip(x)
-        # the residual add is fused into conv2's epilogue on GPU
         return self.conv2(self.norm2(h), residual=skip)
 
 
@@ -128,10 +129,16 @@ class BasicTransformerBlock(nn.Module):
         self.ff = nn.Sequential(GEGLU(dim, dim * 4), nn.Linear(dim * 4, dim))
 
     def forward(self, x, context):
-        x = x + self.attn1(self.norm1(x))
-        x = x + self.attn2(self.norm2(x), context)
-        x = x + self.ff(self.norm3(x))
-        return x
+        h = self.attn1(self.norm1(x))
+        # fused residual-add + pre-norm (one HBM round trip saved per hop)
+        x, n2 = ops.add_layer_norm(
+            x, h, self.norm2.weight, self.norm2.bias, self.norm2.eps
+        )
+        h = self.attn2(n2, context)
+        x, n3 = ops.add_layer_norm(
+            x, h, self.norm3.weight, self.norm3.bias, self.norm3.eps
+        )
+        return x + self.ff(n3)
 
 
 class SpatialTransformer(nn.Module):

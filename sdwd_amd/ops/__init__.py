@@ -126,6 +126,16 @@ def layer_norm(
     ).to(x.dtype)
 
 
+def add_layer_norm(x, res, weight, bias, eps: float = 1e-5):
+    """(x + res, LayerNorm(x + res)) in one kernel (transformer residuals)."""
+    if x.is_cuda and x.dtype == torch.bfloat16 and x.shape[-1] % 8 == 0 \
+            and x.shape[-1] <= 1536:
+        out = ext().add_layer_norm(x, res, weight, bias, eps)
+        return out[0], out[1]
+    s = (x.float() + res.float()).to(x.dtype)
+    return s, layer_norm(s, weight, bias, eps)
+
+
 # ---------------------------------------------------------------------------
 # attention (flash-style, fused softmax(QK^T/sqrt(d))·V)
 # ---------------------------------------------------------------------------
@@ -235,10 +245,12 @@ def conv3x3(
     bias: Optional[torch.Tensor],
     residual: Optional[torch.Tensor],
     stride: int = 1,
+    chan_bias: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
-    """NHWC implicit-GEMM 3x3 conv (pad 1), bias + residual fused into the
-    epilogue. GPU-only entry (callers fall back to F.conv2d on CPU)."""
-    return ext().conv3x3_nhwc(x, w_prep, bias, residual, stride)
+    """NHWC implicit-GEMM 3x3 conv (pad 1); bias, an optional residual and
+    an optional per-(sample, channel) bias (the ResBlock time-embedding
+    projection) all fused into the epilogue. GPU-only entry."""
+    return ext().conv3x3_nhwc(x, w_prep, bias, residual, chan_bias, stride)
 
 
 # ---------------------------------------------------------------------------

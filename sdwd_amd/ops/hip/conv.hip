@@ -32,12 +32,13 @@ __device__ __forceinline__ int swz(int row, int byte_in_row) {
   return byte_in_row ^ (((row >> 2) & 3) << 4);
 }
 
-template <bool HAS_BIAS, bool HAS_RES>
+template <bool HAS_BIAS, bool HAS_RES, bool HAS_CB>
 __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
     const __hip_bfloat16 *__restrict__ X,   // [N,H,W,Cin]
     const __hip_bfloat16 *__restrict__ Wt,  // [Cout,3,3,Cin]
     const float *__restrict__ bias,         // [Cout] or null
     const __hip_bfloat16 *__restrict__ Res, // [N,Ho,Wo,Cout] or null
+    const __hip_bfloat16 *__restrict__ CB,  // [N,Cout] per-sample chan bias
     __hip_bfloat16 *__restrict__ Y,         // [N,Ho,Wo,Cout]
     int Nn, int H, int W, int Cin, int Cout, int Ho, int Wo, int stride) {
   __shared__ __align__(16) __bf16 at[CONV_BM][CONV_BK];
@@ -157,6 +158,10 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
         const long m = m0 + wm + i * 16 + r4 + r;
         if (m >= M) continue;
         float v = acc[i][j][r] + bv;
+        if (HAS_CB) {
+          const int ni = (int)(m / ((long)Ho * Wo));
+          v += (float)CB[(long)ni * Cout + co];
+        }
         if (HAS_RES) v += (float)Res[m * Cout + co];
         Y[m * Cout + co] = f2bf(v);
       }
@@ -171,6 +176,7 @@ bool conv3x3_supported(long cin) { return cin % 32 == 0 && cin >= 32; }
 torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
                            c10::optional<torch::Tensor> bias,
                            c10::optional<torch::Tensor> residual,
+                           c10::optional<torch::Tensor> chan_bias,
                            long stride) {
   // x: [N,C,H,W] channels_last; w_prep: [Cout,3,3,Cin] contiguous
   TORCH_CHECK(x.is_contiguous(torch::MemoryFormat::ChannelsLast),
@@ -202,16 +208,30 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
         residual->is_contiguous(torch::MemoryFormat::ChannelsLast));
     rptr = (const __hip_bfloat16 *)residual->data_ptr();
   }
+  const bool has_cb = chan_bias.has_value();
+  torch::Tensor cbt;
+  const __hip_bfloat16 *cbptr = nullptr;
+  if (has_cb) {
+    cbt = chan_bias->to(torch::kBFloat16).contiguous();
+    cbptr = (const __hip_bfloat16 *)cbt.data_ptr();
+  }
   auto stream = cur_stream();
-  auto kern = has_b ? (has_r ? conv3x3_nhwc_bf16_kernel<true, true>
-                             : conv3x3_nhwc_bf16_kernel<true, false>)
-                    : (has_r ? conv3x3_nhwc_bf16_kernel<false, true>
-                             : conv3x3_nhwc_bf16_kernel<false, false>);
+#define PICK(B_, R_, C_) conv3x3_nhwc_bf16_kernel<B_, R_, C_>
+  auto kern =
+      has_b ? (has_r ? (has_cb ? PICK(true, true, true)
+                               : PICK(true, true, false))
+                     : (has_cb ? PICK(true, false, true)
+                               : PICK(true, false, false)))
+            : (has_r ? (has_cb ? PICK(false, true, true)
+                               : PICK(false, true, false))
+                     : (has_cb ? PICK(false, false, true)
+                               : PICK(false, false, false)));
+#undef PICK
   hipLaunchKernelGGL(kern, grid, block, 0, stream,
                      (const __hip_bfloat16 *)x.data_ptr(),
                      (const __hip_bfloat16 *)w_prep.data_ptr(), bptr, rptr,
-                     (__hip_bfloat16 *)y.data_ptr(), N, H, W, Cin, Cout, Ho,
-                     Wo, (int)stride);
+                     cbptr, (__hip_bfloat16 *)y.data_ptr(), N, H, W, Cin,
+                     Cout, Ho, Wo, (int)stride);
   return y;
 }
 #endif

@@ -211,6 +211,30 @@ torch::Tensor layer_norm(torch::Tensor x, torch::Tensor w, torch::Tensor b,
   return out;
 }
 
+std::vector<torch::Tensor> add_layer_norm(torch::Tensor x, torch::Tensor r,
+                                          torch::Tensor w, torch::Tensor b,
+                                          double eps) {
+  auto xc = x.contiguous();
+  auto rc = r.contiguous();
+  const int D = xc.size(-1);
+  const long R = xc.numel() / D;
+  TORCH_CHECK(xc.scalar_type() == torch::kBFloat16 && D % 8 == 0 &&
+                  D <= 1536,
+              "add_layer_norm: bf16, D%8==0, D<=1536");
+  auto wf = w.to(torch::kFloat).contiguous();
+  auto bf = b.to(torch::kFloat).contiguous();
+  auto out_sum = torch::empty_like(xc);
+  auto out_ln = torch::empty_like(xc);
+  dim3 grid((unsigned)((R + 3) / 4)), block(256);
+  hipLaunchKernelGGL(add_layer_norm_bf16_kernel, grid, block, 0, cur_stream(),
+                     (const __hip_bfloat16 *)xc.data_ptr(),
+                     (const __hip_bfloat16 *)rc.data_ptr(),
+                     wf.data_ptr<float>(), bf.data_ptr<float>(),
+                     (__hip_bfloat16 *)out_sum.data_ptr(),
+                     (__hip_bfloat16 *)out_ln.data_ptr(), R, D, (float)eps);
+  return {out_sum, out_ln};
+}
+
 // ---------------------------------------------------------------------------
 // attention
 // ---------------------------------------------------------------------------
@@ -294,6 +318,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("euler_step", &euler_step);
   m.def("group_norm_silu", &group_norm_silu);
   m.def("layer_norm", &layer_norm);
+  m.def("add_layer_norm", &add_layer_norm);
   m.def("row_softmax_", &row_softmax_);
   m.def("attention_fwd", &attention_fwd);
   m.def("attention_fwd_bshd", &attention_fwd_bshd);

@@ -297,3 +297,61 @@ __global__ void gn_nhwc_norm_bf16(const __hip_bfloat16 *__restrict__ x,
     ((bf16x8 *)out)[i] = o;
   }
 }
+
+// ---------------------------------------------------------------------------
+// Fused residual-add + LayerNorm: sum = x + res; ln = LN(sum)*w + b.
+// The transformer pre-norm chain needs both tensors; fusing saves one full
+// HBM round trip per residual. One wave per row, row held in registers
+// (D <= 1536).
+// ---------------------------------------------------------------------------
+__global__ void add_layer_norm_bf16_kernel(
+    const __hip_bfloat16 *__restrict__ x, const __hip_bfloat16 *__restrict__ r,
+    const float *__restrict__ w, const float *__restrict__ b,
+    __hip_bfloat16 *__restrict__ out_sum, __hip_bfloat16 *__restrict__ out_ln,
+    long R, int D, float eps) {
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const long row = (long)blockIdx.x * 4 + wid;
+  if (row >= R) return;
+  const int nv = D / 8;  // D % 8 == 0 guaranteed by dispatch
+  const __hip_bfloat16 *xb = x + row * D;
+  const __hip_bfloat16 *rb = r + row * D;
+
+  float vals[24 * 8];  // up to 3 vec8 per lane at D=1536
+  float sum = 0.f, sumsq = 0.f;
+  int cnt = 0;
+  for (int i = lane; i < nv; i += WAVE, ++cnt) {
+    bf16x8 vx = ((const bf16x8 *)xb)[i];
+    bf16x8 vr = ((const bf16x8 *)rb)[i];
+    bf16x8 vs;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = (float)vx[j] + (float)vr[j];
+      vals[cnt * 8 + j] = f;
+      sum += f;
+      sumsq += f * f;
+      vs[j] = (__bf16)f;
+    }
+    ((bf16x8 *)(out_sum + row * D))[i] = vs;
+  }
+#pragma unroll
+  for (int off = WAVE / 2; off > 0; off >>= 1) {
+    sum += __shfl_down(sum, off, WAVE);
+    sumsq += __shfl_down(sumsq, off, WAVE);
+  }
+  sum = __shfl(sum, 0, WAVE);
+  sumsq = __shfl(sumsq, 0, WAVE);
+  const float mean = sum / D;
+  const float rstd = rsqrtf(sumsq / D - mean * mean + eps);
+
+  cnt = 0;
+  for (int i = lane; i < nv; i += WAVE, ++cnt) {
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int d = i * 8 + j;
+      o[j] = (__bf16)((vals[cnt * 8 + j] - mean) * rstd * w[d] + b[d]);
+    }
+    ((bf16x8 *)(out_ln + row * D))[i] = o;
+  }
+}

@@ -321,3 +321,30 @@ class TestConv3x3:
             x.float(), conv.weight.float(), conv.bias.float()
         )
         assert relerr(out.contiguous(), ref) < 0.05
+
+    def test_fused_chan_bias(self, dev):
+        """Time-embedding projection as per-(sample,channel) epilogue bias."""
+        x = torch.randn(3, 64, 16, 16, device=dev, dtype=torch.bfloat16)
+        cb = torch.randn(3, 64, device=dev, dtype=torch.bfloat16)
+        conv = torch.nn.Conv2d(64, 64, 3, padding=1).to(dev, torch.bfloat16)
+        ref = torch.nn.functional.conv2d(
+            x.float(), conv.weight.float(), conv.bias.float(), padding=1
+        ) + cb.float()[:, :, None, None]
+        xc = x.contiguous(memory_format=torch.channels_last)
+        wprep = conv.weight.permute(0, 2, 3, 1).contiguous()
+        out = ops.conv3x3(xc, wprep, conv.bias, None, 1, chan_bias=cb)
+        assert relerr(out.contiguous(), ref) < 0.05
+
+
+class TestAddLayerNorm:
+    @pytest.mark.parametrize("D", [320, 640, 1280])
+    def test_fused_matches(self, dev, D):
+        x = torch.randn(4, 64, D, device=dev, dtype=torch.bfloat16)
+        r = torch.randn_like(x)
+        w = torch.randn(D, device=dev)
+        b = torch.randn(D, device=dev)
+        s, ln = ops.add_layer_norm(x, r, w, b)
+        ref_s = x.float() + r.float()
+        ref_ln = torch.nn.functional.layer_norm(ref_s, (D,), w, b, 1e-5)
+        assert relerr(s, ref_s) < 0.05
+        assert relerr(ln, ref_ln) < 0.05
