@@ -348,6 +348,28 @@ class DistributedEngine(_EngineBase):
                 self._store = None
         self._interrupt_epoch = 0
 
+    # -- heartbeat (C11): ranks stamp the store; anyone can read liveness ----
+    def heartbeat(self) -> None:
+        if self._store is not None:
+            try:
+                self._store.set(f"sdwd_hb_{self.rank}", str(time.time()))
+            except Exception:
+                pass
+
+    def heartbeats(self) -> Dict[int, float]:
+        """age (seconds) of each rank's last heartbeat; missing = never."""
+        out: Dict[int, float] = {}
+        if self._store is None:
+            return out
+        now = time.time()
+        for r in range(self.world_size):
+            try:
+                if self._store.check([f"sdwd_hb_{r}"]):
+                    out[r] = now - float(self._store.get(f"sdwd_hb_{r}"))
+            except Exception:
+                continue
+        return out
+
     # -- out-of-band interrupt (C20): TCPStore control plane -----------------
     def interrupt(self) -> None:
         self._interrupt_epoch += 1
@@ -408,6 +430,7 @@ class DistributedEngine(_EngineBase):
 
     def generate(self, gen: GenerationRequest) -> GalleryResult:
         t0 = time.perf_counter()
+        self.heartbeat()
         self._clear_interrupt()
         if self.rank == 0:
             jobs = self.world.make_jobs(gen.sched())

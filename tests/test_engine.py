@@ -169,3 +169,44 @@ class TestDistributedEngine:
         gallery = torch.load(tmp_path / "gallery.pt")
         assert gallery.shape == (4, 64, 64, 3)
         assert all(gallery[i].float().std() > 0 for i in range(4))
+
+
+def _hb_worker(rank, world_size, port, tmpdir):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from sdwd_amd.parallel import DistributedEngine, GenerationRequest, destroy_group
+
+    eng = DistributedEngine(model="tiny", backend="gloo")
+    for w in eng.world.workers:
+        w.eta.avg_ipm = 60.0
+    eng.generate(
+        GenerationRequest(prompt="hb", batch_size=2, width=64, height=64,
+                          steps=1, seed=1)
+    )
+    if rank == 0:
+        hb = eng.heartbeats()
+        assert 0 in hb and 1 in hb, f"missing heartbeats: {hb}"
+        assert all(age < 120 for age in hb.values())
+        with open(os.path.join(tmpdir, "hb_ok"), "w") as fh:
+            fh.write("ok")
+    destroy_group()
+
+
+@pytest.mark.timeout(300)
+class TestHeartbeat:
+    def test_ranks_stamp_store(self, tmp_path):
+        import socket
+
+        import torch.multiprocessing as mp
+
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        mp.start_processes(
+            _hb_worker, args=(2, port, str(tmp_path)), nprocs=2,
+            start_method="spawn", join=True,
+        )
+        assert (tmp_path / "hb_ok").exists()
