@@ -89,7 +89,7 @@ class SDConv2d(nn.Conv2d):
                 and self.padding == (1, 1)
                 and self.stride[0] in (1, 2)
                 and self.stride[0] == self.stride[1]
-                and self.in_channels % 32 == 0
+                and self.in_channels % 64 == 0
                 and self.groups == 1
             ):
                 xc = x.contiguous(memory_format=torch.channels_last)

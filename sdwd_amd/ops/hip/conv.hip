@@ -1,35 +1,44 @@
 // Implicit-GEMM 3x3 convolution for gfx950, NHWC (channels_last), bf16.
 //
-// GEMM view: C[M=N*Ho*Wo, N'=Cout] = A[M, K=9*Cin] x B[K, N'], where A is
-// the implicit im2col of the activations (never materialised: per-plane
-// (dy,dx) address offsets are constant across pixels, so each staged row
-// caches its base address + (ho,wo) once and adds a plane offset) and
-// B is the weight in [Cout][3][3][Cin] (torch channels_last) layout read
-// as rows of K.
+// GEMM view: C[M=N*Ho*Wo, Cout] = A[M, K=9*Cin] x B[K, Cout]; A is the
+// implicit im2col of the activations (per-(dy,dx)-plane address offsets are
+// constant across pixels, so each staged row caches its base address once),
+// B is the weight in [Cout][3][3][Cin] (torch channels_last) layout.
 //
-// Structure = the canonical 128x128-tile MFMA GEMM (guide §5): 4 waves in
-// 2x2, each computing a 64x64 sub-tile as 4x4 mfma_f32_16x16x32_bf16
-// fragments; A/B tiles staged in LDS as [128][BK=32] rows of 64 B with a
-// ((row&3)<<4) XOR byte swizzle so the b128 fragment reads are
-// conflict-free; bias and an optional residual add are fused into the
-// epilogue (the ResBlock skip-add never touches HBM separately).
+// v2 structure (guide §5 step-3 + 'glds, 2 LDS buffers' row):
+//  * 128x128 tile, BK=64; 4 waves in 2x2, each wave a 64x64 sub-tile as
+//    4x4(x2 k-sub) mfma_f32_16x16x32_bf16;
+//  * A/B tiles staged by global_load_lds (16 B/lane direct-to-LDS DMA),
+//    double-buffered in ONE __shared__ array (a second __shared__ object
+//    makes hipcc drain vmcnt before every ds_read - guide §5 trap (a));
+//  * tile t+1's 8 DMA instructions stay in flight across the barrier:
+//    raw s_barrier + counted `s_waitcnt vmcnt(8)` (a plain __syncthreads
+//    would drain the pipeline);
+//  * LDS rows are 128 B; the 8 16-B chunks are XOR-permuted by (row>>1)&7
+//    applied to the DMA *source* address (rule 21: the glds destination is
+//    lane-linear) so the b128 fragment reads are bank-conflict-free;
+//  * padding rows ride a 64-B zero page (glds is wave-uniform, lanes with
+//    out-of-image taps just read zeros);
+//  * bias, per-(sample,channel) bias (time embedding) and the ResBlock
+//    residual add are fused into the epilogue.
 //
-// Requirements: Cin % 32 == 0 (all SD/VAE hot shapes; small-Cin stem convs
-// stay on MIOpen), any Cout, stride 1 or 2, pad 1, kernel 3x3.
+// Requirements: Cin % 64 == 0 (every SD1.5/SDXL/VAE hot shape; small-Cin
+// stem convs stay on MIOpen), any Cout, stride 1 or 2, pad 1, kernel 3x3.
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(4))) float f32x4c;
 
 #define CONV_BM 128
 #define CONV_BN 128
-#define CONV_BK 32
+#define CONV_BK 64
+#define TILE_ELEMS (CONV_BM * CONV_BK)  // per operand per buffer
 
-// XOR-permute the 4 16-byte chunks of a 64-byte LDS row by bits 2-3 of the
-// row index: a b128 fragment read's 16-lane group touches 16 consecutive
-// rows at one chunk, and rows r/r+4/r+8/r+12 would share a bank with a
-// plain layout (row stride 16 dwords); (row>>2)&3 separates them.
-__device__ __forceinline__ int swz(int row, int byte_in_row) {
-  return byte_in_row ^ (((row >> 2) & 3) << 4);
+// XOR-permute the 8 16-byte chunks of a 128-byte LDS row: a b128 fragment
+// read's 16-lane group touches 16 consecutive rows at one chunk; rows of
+// equal parity would collide every 2 rows at a 32-dword stride, so spread
+// by (row>>1)&7.
+__device__ __forceinline__ int cswz(int row, int chunk) {
+  return chunk ^ ((row >> 1) & 7);
 }
 
 template <bool HAS_BIAS, bool HAS_RES, bool HAS_CB>
@@ -38,11 +47,12 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
     const __hip_bfloat16 *__restrict__ Wt,  // [Cout,3,3,Cin]
     const float *__restrict__ bias,         // [Cout] or null
     const __hip_bfloat16 *__restrict__ Res, // [N,Ho,Wo,Cout] or null
-    const __hip_bfloat16 *__restrict__ CB,  // [N,Cout] per-sample chan bias
+    const __hip_bfloat16 *__restrict__ CB,  // [N,Cout] or null
+    const __hip_bfloat16 *__restrict__ Zero,// >=16B of zeros
     __hip_bfloat16 *__restrict__ Y,         // [N,Ho,Wo,Cout]
     int Nn, int H, int W, int Cin, int Cout, int Ho, int Wo, int stride) {
-  __shared__ __align__(16) __bf16 at[CONV_BM][CONV_BK];
-  __shared__ __align__(16) __bf16 bt[CONV_BN][CONV_BK];
+  // one __shared__ object: [buf0: A(8K elems) B(8K)][buf1: A B]
+  __shared__ __align__(16) __bf16 smem[4 * TILE_ELEMS];
 
   const long M = (long)Nn * Ho * Wo;
   const long m0 = (long)blockIdx.x * CONV_BM;
@@ -51,27 +61,66 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
   const int tid = threadIdx.x;
   const int lane = tid % WAVE;
   const int wid = tid / WAVE;
-  const int wm = (wid >> 1) * 64;  // wave row offset in tile
-  const int wn = (wid & 1) * 64;   // wave col offset
+  const int wm = (wid >> 1) * 64;
+  const int wn = (wid & 1) * 64;
 
-  // ---- per-thread staged-row bookkeeping --------------------------------
-  // staging: 128 rows x 32 cols = 4096 bf16 = 512 vec8; 256 threads -> 2
-  // vec8 each; thread t stages rows r = t/2 (A) with col half t%2.
-  const int arow = tid >> 1;        // 0..127: two threads stage one row
-  const int ehalf = (tid & 1) * 16; // element offset of this half-row
-  long abase;
-  int ho_s, wo_s;
-  bool mvalid;
-  {
-    const long m = m0 + arow;
-    mvalid = m < M;
-    const long mm = mvalid ? m : (M - 1);
+  // ---- per-lane DMA row bookkeeping -------------------------------------
+  // stage instr i (0..3) per wave writes LDS bytes
+  // [i*4096 + wid*1024 + lane*16]; linear row = byte/128:
+  //   row_i = i*32 + wid*8 + lane/8, chunk = lane%8 (then source-swizzled)
+  long abase[4];
+  int hs[4], ws[4];   // ho*stride, wo*stride per staged A row
+  int bco[4];         // cout row per staged B row
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int row = i * 32 + wid * 8 + lane / 8;
+    const long m = m0 + row;
+    const long mm = (m < M) ? m : (M - 1);
     const int n_img = (int)(mm / ((long)Ho * Wo));
     const int rem = (int)(mm % ((long)Ho * Wo));
-    ho_s = rem / Wo;
-    wo_s = rem % Wo;
-    abase = (((long)n_img * H + ho_s * stride) * W + wo_s * stride) * Cin;
+    hs[i] = (rem / Wo) * stride;
+    ws[i] = (rem % Wo) * stride;
+    abase[i] = (((long)n_img * H + hs[i]) * W + ws[i]) * Cin;
+    bco[i] = n0 + row;
   }
+  const int schunk = lane % 8;
+
+  const int kc_per_plane = Cin / CONV_BK;
+  const int NT = 9 * kc_per_plane;
+
+  // stage tile t into buffer b (8 glds: 4 A + 4 B)
+  auto stage = [&](int t, int b) {
+    const int plane = t / kc_per_plane;
+    const int kc = t % kc_per_plane;
+    const int dy = plane / 3 - 1, dx = plane % 3 - 1;
+    const long poff = ((long)dy * W + dx) * Cin + (long)kc * CONV_BK;
+    __bf16 *abuf = smem + b * 2 * TILE_ELEMS;
+    __bf16 *bbuf = abuf + TILE_ELEMS;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int row = i * 32 + wid * 8 + lane / 8;
+      const int sc = cswz(row, schunk);
+      const bool av = (hs[i] + dy >= 0) && (hs[i] + dy < H) &&
+                      (ws[i] + dx >= 0) && (ws[i] + dx < W);
+      const __hip_bfloat16 *asrc =
+          av ? (X + abase[i] + poff + sc * 8) : Zero;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)asrc,
+          (__attribute__((address_space(3))) unsigned int
+               *)(abuf + i * 2048 + wid * 512),
+          16, 0, 0);
+      const bool bv = bco[i] < Cout;
+      const __hip_bfloat16 *bsrc =
+          bv ? (Wt + (long)bco[i] * 9 * Cin + plane * Cin + kc * CONV_BK +
+                sc * 8)
+             : Zero;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int *)bsrc,
+          (__attribute__((address_space(3))) unsigned int
+               *)(bbuf + i * 2048 + wid * 512),
+          16, 0, 0);
+    }
+  };
 
   f32x4c acc[4][4];
 #pragma unroll
@@ -79,72 +128,51 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4c){};
 
-  const int kc_per_plane = Cin / CONV_BK;
-  for (int plane = 0; plane < 9; ++plane) {
-    const int dy = plane / 3 - 1, dx = plane % 3 - 1;
-    const int hi = ho_s * stride + dy;
-    const int wi = wo_s * stride + dx;
-    const bool pvalid =
-        mvalid && hi >= 0 && hi < H && wi >= 0 && wi < W;
-    const long poff = ((long)dy * W + dx) * Cin;
+  stage(0, 0);
+  if (NT > 1) stage(1, 1);
 
-    for (int kc = 0; kc < kc_per_plane; ++kc) {
-      __syncthreads();
-      // stage A: rows = output pixels, 32 ci of this plane/chunk
-      {
-        bf16x8 v0 = (bf16x8){}, v1 = (bf16x8){};
-        if (pvalid) {
-          const __hip_bfloat16 *src = X + abase + poff + kc * CONV_BK;
-          v0 = *(const bf16x8 *)(src + ehalf);
-          v1 = *(const bf16x8 *)(src + ehalf + 8);
-        }
-        char *arow_p = (char *)&at[arow][0];
-        *(bf16x8 *)(arow_p + swz(arow, ehalf * 2)) = v0;
-        *(bf16x8 *)(arow_p + swz(arow, ehalf * 2 + 16)) = v1;
-      }
-      // stage B: rows = cout, same 32-k chunk from [Cout][9*Cin]
-      {
-        const int co = n0 + arow;
-        bf16x8 v0 = (bf16x8){}, v1 = (bf16x8){};
-        if (co < Cout) {
-          const __hip_bfloat16 *src =
-              Wt + (long)co * 9 * Cin + plane * Cin + kc * CONV_BK;
-          v0 = *(const bf16x8 *)(src + ehalf);
-          v1 = *(const bf16x8 *)(src + ehalf + 8);
-        }
-        char *brow_p = (char *)&bt[arow][0];
-        *(bf16x8 *)(brow_p + swz(arow, ehalf * 2)) = v0;
-        *(bf16x8 *)(brow_p + swz(arow, ehalf * 2 + 16)) = v1;
-      }
-      __syncthreads();
+  const int l16 = lane % 16;
+  const int kq = (lane / 16) * 8;  // k offset of this lane's 8 elements
 
-      // compute: each wave 4x4 fragments of 16x16x32 over this k-chunk
-      const int l16 = lane % 16;
-      const int k8 = (lane / 16) * 8;  // 8 consecutive k per lane
+  for (int t = 0; t < NT; ++t) {
+    // current tile's 8 DMAs landed; the next tile's 8 stay in flight
+    if (t + 1 < NT)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+
+    const __bf16 *abuf = smem + (t & 1) * 2 * TILE_ELEMS;
+    const __bf16 *bbuf = abuf + TILE_ELEMS;
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      bf16x8 af[4], bf[4];
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const int am = wm + i * 16 + l16;
-        bf16x8 af;
-        {
-          const char *p = (const char *)&at[am][0];
-          af = *(const bf16x8 *)(p + swz(am, k8 * 2));
-        }
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          const int bn = wn + j * 16 + l16;
-          bf16x8 bf;
-          const char *p = (const char *)&bt[bn][0];
-          bf = *(const bf16x8 *)(p + swz(bn, k8 * 2));
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af, bf, acc[i][j], 0, 0, 0);
-        }
+        const int ck = cswz(am, (s * 32 + kq) / 8);
+        af[i] = *(const bf16x8 *)((const char *)(abuf + am * CONV_BK) +
+                                  ck * 16);
       }
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int bn = wn + j * 16 + l16;
+        const int ck = cswz(bn, (s * 32 + kq) / 8);
+        bf[j] = *(const bf16x8 *)((const char *)(bbuf + bn * CONV_BK) +
+                                  ck * 16);
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
     }
+    __builtin_amdgcn_s_barrier();  // everyone done reading this buffer
+    if (t + 2 < NT) stage(t + 2, t & 1);
   }
 
-  // ---- epilogue: bias, residual, store ----------------------------------
-  // C frag 16x16: col = lane%16, row = 4*(lane/16) + reg
-  const int l16 = lane % 16;
+  // ---- epilogue: bias, channel-bias, residual, store --------------------
   const int r4 = (lane / 16) * 4;
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
@@ -171,7 +199,7 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
 
 // ---------------------------------------------------------------------------
 #ifdef __HIP_PLATFORM_AMD__
-bool conv3x3_supported(long cin) { return cin % 32 == 0 && cin >= 32; }
+bool conv3x3_supported(long cin) { return cin % 64 == 0 && cin >= 64; }
 
 torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
                            c10::optional<torch::Tensor> bias,
@@ -184,12 +212,15 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
   const int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
   const int Cout = w_prep.size(0);
-  TORCH_CHECK(conv3x3_supported(Cin), "conv3x3: Cin % 32 != 0");
+  TORCH_CHECK(conv3x3_supported(Cin), "conv3x3: Cin % 64 != 0");
   const int Ho = (H + 2 - 3) / stride + 1;
   const int Wo = (W + 2 - 3) / stride + 1;
   auto y = torch::empty({N, Cout, Ho, Wo},
                         x.options().memory_format(
                             torch::MemoryFormat::ChannelsLast));
+  static torch::Tensor zero_page;
+  if (!zero_page.defined() || zero_page.device() != x.device())
+    zero_page = torch::zeros({64}, x.options());
   const long M = (long)N * Ho * Wo;
   dim3 grid((unsigned)((M + CONV_BM - 1) / CONV_BM),
             (unsigned)((Cout + CONV_BN - 1) / CONV_BN));
@@ -230,8 +261,10 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
   hipLaunchKernelGGL(kern, grid, block, 0, stream,
                      (const __hip_bfloat16 *)x.data_ptr(),
                      (const __hip_bfloat16 *)w_prep.data_ptr(), bptr, rptr,
-                     cbptr, (__hip_bfloat16 *)y.data_ptr(), N, H, W, Cin,
-                     Cout, Ho, Wo, (int)stride);
+                     cbptr,
+                     (const __hip_bfloat16 *)zero_page.data_ptr(),
+                     (__hip_bfloat16 *)y.data_ptr(), N, H, W, Cin, Cout, Ho,
+                     Wo, (int)stride);
   return y;
 }
 #endif

@@ -121,15 +121,17 @@ torch::Tensor group_norm_silu_nhwc(torch::Tensor x, torch::Tensor w,
   auto opts = torch::TensorOptions()
                   .dtype(torch::kFloat)
                   .device(x.device());
-  auto partial = torch::empty({N * S, 2L * C}, opts);
+  const int VC = C / 8;
+  const int RP = VC >= 256 ? 1 : std::max(1, 256 / VC);
+  auto partial = torch::empty({(long)N * S * RP, 2L * C}, opts);
   auto stats = torch::empty({N * (long)groups, 2L}, opts);
   auto stream = cur_stream();
   hipLaunchKernelGGL(gn_nhwc_partial_bf16, dim3((unsigned)(N * S)), dim3(256),
                      0, stream, (const __hip_bfloat16 *)x.data_ptr(),
-                     partial.data_ptr<float>(), C, HW, S);
+                     partial.data_ptr<float>(), C, HW, S, RP);
   hipLaunchKernelGGL(gn_nhwc_stats, dim3((unsigned)(N * groups)), dim3(WAVE),
                      0, stream, partial.data_ptr<float>(),
-                     stats.data_ptr<float>(), C, HW, (int)groups, S,
+                     stats.data_ptr<float>(), C, HW, (int)groups, S * RP,
                      (float)eps);
   const long total = N * HW * (C / 8);
   auto kern =

@@ -205,9 +205,14 @@ __global__ void layer_norm_f32_kernel(const float *__restrict__ x,
 // 3 phases: per-channel partial sums (coalesced C-contiguous loads),
 // per-(n,g) stat finalize, fused normalize+SiLU sweep.
 // ---------------------------------------------------------------------------
+// Threads own vec8 channel columns: thread (ro, vc) reads rows hw0+ro,
+// hw0+ro+RP, ... at 16 B per lane (consecutive vc -> consecutive 16 B:
+// fully coalesced). RP rows are reduced per block pass; the stats kernel
+// sums over S*RP partial slots.
 __global__ void gn_nhwc_partial_bf16(const __hip_bfloat16 *__restrict__ x,
-                                     float *__restrict__ partial,  // [N*S][2C]
-                                     int C, long HW, int S) {
+                                     float *__restrict__ partial,
+                                     // [N*S*RP][2C]
+                                     int C, long HW, int S, int RP) {
   const int ns = blockIdx.x;      // n * S + s
   const int n = ns / S, sc = ns % S;
   const long chunk = (HW + S - 1) / S;
@@ -215,28 +220,44 @@ __global__ void gn_nhwc_partial_bf16(const __hip_bfloat16 *__restrict__ x,
   const long hw1 = min(HW, hw0 + chunk);
   const __hip_bfloat16 *base = x + (long)n * HW * C;
 
-  const int cpt = (C + 255) / 256;  // channels owned per thread
-  float sum[12], sumsq[12];         // cpt <= 12 (C <= 3072; 2560 after cat)
+  const int VC = C / 8;                       // vec8 columns
+  const int cpt8 = (VC + 255) / 256;          // vec cols per thread (<=2)
+  const int vc0 = (int)threadIdx.x % (cpt8 == 1 ? VC : 256);
+  const int ro = cpt8 == 1 ? (int)threadIdx.x / VC : 0;
+  float sum[2][8], sumsq[2][8];
 #pragma unroll
-  for (int j = 0; j < 12; ++j) sum[j] = sumsq[j] = 0.f;
+  for (int j2 = 0; j2 < 2; ++j2)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) sum[j2][j] = sumsq[j2][j] = 0.f;
 
-  for (long hw = hw0; hw < hw1; ++hw) {
-    const __hip_bfloat16 *row = base + hw * C;
-    for (int j = 0; j < cpt; ++j) {
-      const int c = j * 256 + threadIdx.x;
-      if (c < C) {
-        float f = bf2f(row[c]);
-        sum[j] += f;
-        sumsq[j] += f * f;
+  if (ro < RP) {
+    for (long hw = hw0 + ro; hw < hw1; hw += RP) {
+      const __hip_bfloat16 *row = base + hw * C;
+      for (int j2 = 0; j2 < cpt8; ++j2) {
+        const int vc = j2 * 256 + vc0;
+        if (vc < VC) {
+          bf16x8 v = ((const bf16x8 *)row)[vc];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            float f = (float)v[j];
+            sum[j2][j] += f;
+            sumsq[j2][j] += f * f;
+          }
+        }
       }
     }
   }
-  float *out = partial + (long)ns * 2 * C;
-  for (int j = 0; j < cpt; ++j) {
-    const int c = j * 256 + threadIdx.x;
-    if (c < C) {
-      out[c] = sum[j];
-      out[C + c] = sumsq[j];
+  float *out = partial + ((long)ns * RP + ro) * 2 * C;
+  if (ro < RP) {
+    for (int j2 = 0; j2 < cpt8; ++j2) {
+      const int vc = j2 * 256 + vc0;
+      if (vc < VC) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          out[vc * 8 + j] = sum[j2][j];
+          out[C + vc * 8 + j] = sumsq[j2][j];
+        }
+      }
     }
   }
 }

@@ -264,13 +264,14 @@ class TestGroupNormNHWC:
 
 class TestConv3x3:
     @pytest.mark.parametrize("N,Cin,H,W,Cout,stride", [
-        (2, 32, 16, 16, 64, 1),
+        (2, 64, 16, 16, 64, 1),
         (2, 320, 32, 32, 320, 1),
         (1, 320, 64, 64, 320, 1),
         (2, 640, 17, 17, 320, 1),   # odd spatial
         (2, 320, 32, 32, 320, 2),   # downsample
         (1, 2560, 8, 8, 1280, 1),   # concat shape
         (1, 128, 96, 96, 40, 1),    # Cout not multiple of tile
+        (2, 320, 15, 15, 192, 1),   # M remainder + small Cout
     ])
     def test_vs_conv2d(self, dev, N, Cin, H, W, Cout, stride):
         torch.manual_seed(3)
