@@ -41,7 +41,7 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
     const __hip_bfloat16 *__restrict__ Q, const __hip_bfloat16 *__restrict__ K,
     const __hip_bfloat16 *__restrict__ V, __hip_bfloat16 *__restrict__ O,
     int H, long Sq, long Sk, int D, float scale, AttnStrides st) {
-  constexpr int KVB = 32;
+  constexpr int KVB = 64;  // two 32-key S^T sub-tiles per staging phase
   constexpr int PADK = 8;   // bf16 per-row pad: breaks ds_read_b128 conflicts
   constexpr int NC = DPAD / 16;              // QK^T k-chunks
   constexpr int DV = (DPAD + 31) / 32 * 32;  // PV d extent (32-col O tiles)
@@ -105,33 +105,42 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
     }
     __syncthreads();
 
-    // S^T[32k, 32q] = sum_c K[.,c] x Q^T[c,.]
-    f32x16 stile = (f32x16){};
+    // two S^T[32k, 32q] sub-tiles = sum_c K[.,c] x Q^T[c,.]
+    f32x16 stile[2] = {(f32x16){}, (f32x16){}};
 #pragma unroll
-    for (int c = 0; c < NC; ++c) {
-      bf16x8 kf = *(const bf16x8 *)&kt[lq][c * 16 + 8 * half];
-      stile = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[c], stile, 0, 0, 0);
-    }
+    for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+      for (int c = 0; c < NC; ++c) {
+        bf16x8 kf =
+            *(const bf16x8 *)&kt[sub * 32 + lq][c * 16 + 8 * half];
+        stile[sub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            kf, qf[c], stile[sub], 0, 0, 0);
+      }
 
-    // online softmax for q-row lq; lane has k = (r&3)+8*(r>>2)+4*half
-    float p[16];
+    // merged online softmax over all 64 keys: ONE m/l update and ONE
+    // O-rescale per staging phase (halves the softmax overhead per key)
+    float p[2][16];
     float pmax = -1e30f;
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int kk = (r & 3) + 8 * (r >> 2) + 4 * half;
-      float s = (kv + kk < Sk) ? stile[r] * scale : -1e30f;
-      p[r] = s;
-      pmax = fmaxf(pmax, s);
-    }
+    for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kk = sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+        float sv = (kv + kk < Sk) ? stile[sub][r] * scale : -1e30f;
+        p[sub][r] = sv;
+        pmax = fmaxf(pmax, sv);
+      }
     pmax = fmaxf(pmax, __shfl_xor(pmax, 32, WAVE));
     const float mnew = fmaxf(m, pmax);
     const float alpha = __expf(m - mnew);
     float rowsum = 0.f;
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      p[r] = __expf(p[r] - mnew);
-      rowsum += p[r];
-    }
+    for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        p[sub][r] = __expf(p[sub][r] - mnew);
+        rowsum += p[sub][r];
+      }
     rowsum += __shfl_xor(rowsum, 32, WAVE);
     l = l * alpha + rowsum;
     m = mnew;
@@ -146,29 +155,40 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
       for (int d = 0; d < ND; ++d) o[d][r] *= a;
     }
 
-    // pack P -> PV A-fragments via permlane32_swap (chunk k0..15, k16..31)
-    unsigned pk[8];
+    // pack P -> PV A-fragments via permlane32_swap (per 16-k chunk)
 #pragma unroll
-    for (int t = 0; t < 8; ++t) pk[t] = pack_bf16(p[2 * t], p[2 * t + 1]);
-    bf16x8 pa0, pa1;
-    {
-      auto r0 = __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
-      auto r1 = __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
-      unsigned frag[4] = {r0[0], r1[0], r0[1], r1[1]};
-      pa0 = *(bf16x8 *)frag;
-      auto r2 = __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
-      auto r3 = __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
-      unsigned frag1[4] = {r2[0], r3[0], r2[1], r3[1]};
-      pa1 = *(bf16x8 *)frag1;
-    }
-
-    // PV: O[32q, 32d] += P[32q,16k] x V[16k,32d] per d-block
+    for (int sub = 0; sub < 2; ++sub) {
+      unsigned pk[8];
 #pragma unroll
-    for (int d = 0; d < ND; ++d) {
-      bf16x8 v0 = *(const bf16x8 *)&vt[d * 32 + lq][8 * half];
-      o[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa0, v0, o[d], 0, 0, 0);
-      bf16x8 v1 = *(const bf16x8 *)&vt[d * 32 + lq][16 + 8 * half];
-      o[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, v1, o[d], 0, 0, 0);
+      for (int t = 0; t < 8; ++t)
+        pk[t] = pack_bf16(p[sub][2 * t], p[sub][2 * t + 1]);
+      bf16x8 pa0, pa1;
+      {
+        auto r0 =
+            __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
+        auto r1 =
+            __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
+        unsigned frag[4] = {r0[0], r1[0], r0[1], r1[1]};
+        pa0 = *(bf16x8 *)frag;
+        auto r2 =
+            __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
+        auto r3 =
+            __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
+        unsigned frag1[4] = {r2[0], r3[0], r2[1], r3[1]};
+        pa1 = *(bf16x8 *)frag1;
+      }
+      // PV: O[32q, 32d] += P x V per d-block (keys sub*32 .. sub*32+31)
+#pragma unroll
+      for (int d = 0; d < ND; ++d) {
+        bf16x8 v0 =
+            *(const bf16x8 *)&vt[d * 32 + lq][sub * 32 + 8 * half];
+        o[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa0, v0, o[d], 0,
+                                                       0, 0);
+        bf16x8 v1 =
+            *(const bf16x8 *)&vt[d * 32 + lq][sub * 32 + 16 + 8 * half];
+        o[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, v1, o[d], 0,
+                                                       0, 0);
+      }
     }
   }
 
