@@ -99,42 +99,21 @@ class CrossAttention(nn.Module):
         super().__init__()
         self.heads = heads
         self.d_head = query_dim // heads
-        self.is_self = query_dim == context_dim
         self.to_q = nn.Linear(query_dim, query_dim, bias=False)
         self.to_k = nn.Linear(context_dim, query_dim, bias=False)
         self.to_v = nn.Linear(context_dim, query_dim, bias=False)
         self.to_out = nn.Linear(query_dim, query_dim)
-        self._fused_w = None  # (weight key, concatenated qkv weight)
-
-    def _qkv_weight(self) -> torch.Tensor:
-        # self-attention: one [3d, d] GEMM reads the HBM-bound input once
-        # instead of three times
-        key = (self.to_q.weight.data_ptr(), self.to_q.weight.dtype)
-        if self._fused_w is None or self._fused_w[0] != key:
-            w = torch.cat(
-                [self.to_q.weight, self.to_k.weight, self.to_v.weight], dim=0
-            ).contiguous()
-            self._fused_w = (key, w)
-        return self._fused_w[1]
 
     def forward(
         self, x: torch.Tensor, context: Optional[torch.Tensor] = None
     ) -> torch.Tensor:
-        is_self = context is None
         context = x if context is None else context
         b, s, d = x.shape
         sk = context.shape[1]
         # natural [B,S,H,D] layout end-to-end: the flash kernel reads strided
-        if is_self and self.is_self:
-            qkv = torch.nn.functional.linear(x, self._qkv_weight())
-            q, k, v = qkv.chunk(3, dim=-1)
-            q = q.reshape(b, s, self.heads, self.d_head)
-            k = k.reshape(b, sk, self.heads, self.d_head)
-            v = v.reshape(b, sk, self.heads, self.d_head)
-        else:
-            q = self.to_q(x).view(b, s, self.heads, self.d_head)
-            k = self.to_k(context).view(b, sk, self.heads, self.d_head)
-            v = self.to_v(context).view(b, sk, self.heads, self.d_head)
+        q = self.to_q(x).view(b, s, self.heads, self.d_head)
+        k = self.to_k(context).view(b, sk, self.heads, self.d_head)
+        v = self.to_v(context).view(b, sk, self.heads, self.d_head)
         out = ops.attention_bshd(q, k, v)
         return self.to_out(out.reshape(b, s, d))
 
