@@ -87,51 +87,21 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
        idx < DV * (KVB + PADK); idx += 256)
     ((__bf16 *)vt)[idx] = (__bf16)0.0f;
 
-  // T14 async staging: each thread covers <= 2 (row, col-group) slots of
-  // the K/V tile; the NEXT tile's global loads are issued before the MFMA
-  // phase so their ~500-cycle HBM latency hides under the compute, and the
-  // register set is written to LDS after the barrier.
-  constexpr int NSLOT = (KVB * (DPAD / 8) + 255) / 256;
-  bf16x8 kreg[NSLOT], vreg[NSLOT];
-  auto load_tile = [&](long kv) {
-#pragma unroll
-    for (int pft = 0; pft < NSLOT; ++pft) {
-      const int idx = (int)threadIdx.x + pft * 256;
-      kreg[pft] = (bf16x8){};
-      vreg[pft] = (bf16x8){};
-      if (idx < KVB * (DPAD / 8)) {
-        const int r = idx / (DPAD / 8);
-        const int c8 = idx % (DPAD / 8);
-        if (kv + r < Sk && c8 * 8 + 8 <= D) {
-          kreg[pft] = *(const bf16x8 *)(Kb + (kv + r) * st.kr + c8 * 8);
-          vreg[pft] = *(const bf16x8 *)(Vb + (kv + r) * st.vr + c8 * 8);
-        }
-      }
-    }
-  };
-  load_tile(0);
-
   for (long kv = 0; kv < Sk; kv += KVB) {
     __syncthreads();  // previous tile's LDS reads complete
-    // write the prefetched registers, then immediately issue the next
-    // tile's loads (first consumed at the next iteration's write)
-    bf16x8 kw[NSLOT], vw[NSLOT];
-#pragma unroll
-    for (int pft = 0; pft < NSLOT; ++pft) {
-      kw[pft] = kreg[pft];
-      vw[pft] = vreg[pft];
-    }
-    if (kv + KVB < Sk) load_tile(kv + KVB);
-#pragma unroll
-    for (int pft = 0; pft < NSLOT; ++pft) {
-      const int idx = (int)threadIdx.x + pft * 256;
-      if (idx < KVB * (DPAD / 8)) {
-        const int r = idx / (DPAD / 8);
-        const int c8 = idx % (DPAD / 8);
-        *(bf16x8 *)&kt[r][c8 * 8] = kw[pft];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) vt[c8 * 8 + j][r] = vw[pft][j];
+    // cooperative K/V stage: 256 threads, 8 bf16 each per step
+    for (int idx = threadIdx.x; idx < KVB * (DPAD / 8); idx += 256) {
+      const int r = idx / (DPAD / 8);     // key row in tile
+      const int c8 = idx % (DPAD / 8);    // 8-elem column group
+      bf16x8 kvec = (bf16x8){};
+      bf16x8 vvec = (bf16x8){};
+      if (kv + r < Sk && c8 * 8 + 8 <= D) {
+        kvec = *(const bf16x8 *)(Kb + (kv + r) * st.kr + c8 * 8);
+        vvec = *(const bf16x8 *)(Vb + (kv + r) * st.vr + c8 * 8);
       }
+      *(bf16x8 *)&kt[r][c8 * 8] = kvec;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vt[c8 * 8 + j][r] = vvec[j];
     }
     __syncthreads();
 
