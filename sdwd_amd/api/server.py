@@ -147,6 +147,15 @@ def create_app(engine: Optional[LocalEngine] = None,
             "info": json.dumps(info),
         }
 
+    @app.get("/")
+    def index():
+        """Built-in control surface (ref C19)."""
+        from fastapi.responses import HTMLResponse
+
+        from .ui import PAGE
+
+        return HTMLResponse(PAGE)
+
     @app.post("/sdapi/v1/txt2img")
     def txt2img(req: Txt2ImgRequest):
         control_image, control_model, control_scale = _parse_controlnet(

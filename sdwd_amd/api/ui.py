@@ -1,0 +1,83 @@
+"""Minimal built-in web UI (ref C19, ui.py:16-404 as a single page).
+
+The reference injected a Gradio accordion with Status / Utils / Worker
+Config / Settings tabs into the host webui; this serves an equivalent
+single-page control surface at ``/`` on top of the JSON API: live status
+(1.5 s auto-refresh like distributed.js:7-23), a txt2img form, benchmark /
+interrupt / sync-script buttons and per-worker enable/disable toggles.
+"""
+
+PAGE = """<!DOCTYPE html>
+<html><head><title>sdwd_amd</title>
+<style>
+ body{font-family:monospace;margin:2em;background:#111;color:#ddd}
+ h2{color:#e8a33d} table{border-collapse:collapse}
+ td,th{border:1px solid #444;padding:4px 10px}
+ button{margin:2px;background:#333;color:#ddd;border:1px solid #666;
+        padding:4px 10px;cursor:pointer}
+ input,select{background:#222;color:#ddd;border:1px solid #555;padding:3px}
+ #log{white-space:pre-wrap;background:#000;padding:8px;max-height:220px;
+      overflow-y:auto;font-size:11px}
+ img{max-width:256px;margin:4px;border:1px solid #444}
+</style></head>
+<body>
+<h2>sdwd_amd — distributed SD engine</h2>
+<div id="status">loading…</div>
+<h2>generate</h2>
+<form onsubmit="gen(event)">
+ prompt <input id="prompt" size="50" value="a herd of cows"/>
+ batch <input id="batch" size="3" value="4"/>
+ steps <input id="steps" size="3" value="20"/>
+ size <input id="size" size="4" value="512"/>
+ seed <input id="seed" size="8" value="-1"/>
+ <button>generate</button>
+ <button type="button" onclick="fetch('/sdapi/v1/interrupt',{method:'POST'})">
+   interrupt</button>
+</form>
+<div id="gallery"></div>
+<h2>utils</h2>
+<button onclick="fetch('/sdwd/benchmark',{method:'POST'})">re-benchmark</button>
+<button onclick="fetch('/sdwd/sync-script',{method:'POST'})">run sync script</button>
+<h2>log</h2><div id="log"></div>
+<script>
+async function refresh(){
+  try{
+    const s = await (await fetch('/sdwd/status')).json();
+    let h = '<table><tr><th>rank</th><th>device</th><th>state</th>'+
+            '<th>ipm</th><th>mpe%</th><th></th></tr>';
+    for(const w of s.workers){
+      h += `<tr><td>${w.label}</td><td>${w.device}</td><td>${w.state}</td>`+
+           `<td>${w.avg_ipm.toFixed(1)}</td><td>${w.mpe.toFixed(1)}</td>`+
+           `<td><button onclick="tog('${w.label}','${w.state}')">`+
+           `${w.state==='DISABLED'?'enable':'disable'}</button></td></tr>`;
+    }
+    h += `</table><p>model: ${s.model} — busy: ${s.busy}</p>`;
+    document.getElementById('status').innerHTML = h;
+    document.getElementById('log').textContent = s.log.slice(-16).join('\\n');
+  }catch(e){}
+}
+async function tog(label, state){
+  const act = state==='DISABLED' ? 'enable' : 'disable';
+  await fetch(`/sdwd/worker/${label}/${act}`,{method:'POST'});
+  refresh();
+}
+async function gen(ev){
+  ev.preventDefault();
+  const body = {
+    prompt: document.getElementById('prompt').value,
+    batch_size: parseInt(document.getElementById('batch').value),
+    steps: parseInt(document.getElementById('steps').value),
+    width: parseInt(document.getElementById('size').value),
+    height: parseInt(document.getElementById('size').value),
+    seed: parseInt(document.getElementById('seed').value),
+  };
+  const r = await fetch('/sdapi/v1/txt2img', {method:'POST',
+    headers:{'Content-Type':'application/json'}, body: JSON.stringify(body)});
+  const out = await r.json();
+  document.getElementById('gallery').innerHTML =
+    out.images.map(b => `<img src="data:image/png;base64,${b}"/>`).join('');
+}
+setInterval(refresh, 1500);  // ref distributed.js:7-23 auto-refresh cadence
+refresh();
+</script>
+</body></html>"""

@@ -147,3 +147,11 @@ class TestAlwaysonControlNet:
         assert r.status_code == 200
         img = decode_png(base64.b64decode(r.json()["images"][0]))
         assert img.shape[0] == 128  # grid of one 128x128 image
+
+
+class TestWebUI:
+    def test_index_page(self, client):
+        r = client.get("/")
+        assert r.status_code == 200
+        assert "sdwd_amd" in r.text
+        assert "/sdwd/status" in r.text

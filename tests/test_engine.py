@@ -210,3 +210,43 @@ class TestHeartbeat:
             start_method="spawn", join=True,
         )
         assert (tmp_path / "hb_ok").exists()
+
+
+class TestEdgeCases:
+    def test_batch_smaller_than_ranks(self):
+        eng = make_engine(3)
+        res = eng.generate(
+            GenerationRequest(prompt="s", batch_size=1, width=64, height=64,
+                              steps=2, seed=9)
+        )
+        assert res.images.shape == (1, 64, 64, 3)
+        assert res.seeds == [9]
+
+    def test_batch_one_per_rank(self):
+        eng = make_engine(2)
+        res = eng.generate(
+            GenerationRequest(prompt="s", batch_size=2, width=64, height=64,
+                              steps=2, seed=3)
+        )
+        assert res.images.shape == (2, 64, 64, 3)
+
+    def test_large_remainder(self):
+        eng = make_engine(3)
+        res = eng.generate(
+            GenerationRequest(prompt="s", batch_size=7, width=64, height=64,
+                              steps=1, seed=1)
+        )
+        assert res.images.shape == (7, 64, 64, 3)
+        assert res.seeds == list(range(1, 8))
+
+    def test_interrupt_mid_generation(self):
+        import threading as th
+
+        eng = make_engine(2)
+        req = GenerationRequest(prompt="i", batch_size=2, width=64,
+                                height=64, steps=30, seed=2)
+        timer = th.Timer(0.5, eng.interrupt)
+        timer.start()
+        res = eng.generate(req)
+        timer.cancel()
+        assert res.images.shape == (2, 64, 64, 3)
