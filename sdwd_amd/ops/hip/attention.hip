@@ -56,7 +56,7 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
   const int lane = threadIdx.x % WAVE;
   const int lq = lane % 32;   // q-row (softmax) / d-col (PV C) index
   const int half = lane / 32; // half-wave id
-  const long q0 = (long)blockIdx.x * 128 + wid * 32;
+  const long q0 = (long)blockIdx.x * 256 + wid * 64;
 
   const __hip_bfloat16 *Qb = Q + bb * st.qb + hh * st.qh;
   const __hip_bfloat16 *Kb = K + bb * st.kb + hh * st.kh;
@@ -65,22 +65,26 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
 
   // Q fragments for the whole row-block, read once:
   // B-frag of mfma(K,Q): lane holds Q[q=lq][d = c*16 + 8*half + i]
-  bf16x8 qf[NC];
-  {
-    const long qrow = (q0 + lq < Sq) ? (q0 + lq) : (Sq - 1);
+  bf16x8 qf[2][NC];
+#pragma unroll
+  for (int qs = 0; qs < 2; ++qs) {
+    const long qq = q0 + qs * 32 + lq;
+    const long qrow = (qq < Sq) ? qq : (Sq - 1);
 #pragma unroll
     for (int c = 0; c < NC; ++c) {
       const int d0 = c * 16 + 8 * half;
-      qf[c] = (d0 + 8 <= D)
-                  ? *(const bf16x8 *)(Qb + qrow * st.qr + d0)
-                  : (bf16x8){};
+      qf[qs][c] = (d0 + 8 <= D)
+                      ? *(const bf16x8 *)(Qb + qrow * st.qr + d0)
+                      : (bf16x8){};
     }
   }
 
-  f32x16 o[ND];
+  f32x16 o[2][ND];
 #pragma unroll
-  for (int d = 0; d < ND; ++d) o[d] = (f32x16){};
-  float m = -1e30f, l = 0.f;
+  for (int qs = 0; qs < 2; ++qs)
+#pragma unroll
+    for (int d = 0; d < ND; ++d) o[qs][d] = (f32x16){};
+  float m[2] = {-1e30f, -1e30f}, l[2] = {0.f, 0.f};
 
   // zero vt's pad rows once (D..DV); they are never re-staged
   for (int idx = threadIdx.x + (D / 8) * 8 * (KVB + PADK);
@@ -105,104 +109,105 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
     }
     __syncthreads();
 
-    // two S^T[32k, 32q] sub-tiles = sum_c K[.,c] x Q^T[c,.]
-    f32x16 stile[2] = {(f32x16){}, (f32x16){}};
+    // per q-subtile: two S^T[32k, 32q] k-sub-tiles, merged softmax, PV
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub)
+    for (int qs = 0; qs < 2; ++qs) {
+      f32x16 stile[2] = {(f32x16){}, (f32x16){}};
 #pragma unroll
-      for (int c = 0; c < NC; ++c) {
-        bf16x8 kf =
-            *(const bf16x8 *)&kt[sub * 32 + lq][c * 16 + 8 * half];
-        stile[sub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            kf, qf[c], stile[sub], 0, 0, 0);
-      }
+      for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+        for (int c = 0; c < NC; ++c) {
+          bf16x8 kf =
+              *(const bf16x8 *)&kt[sub * 32 + lq][c * 16 + 8 * half];
+          stile[sub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              kf, qf[qs][c], stile[sub], 0, 0, 0);
+        }
 
-    // merged online softmax over all 64 keys: ONE m/l update and ONE
-    // O-rescale per staging phase (halves the softmax overhead per key)
-    float p[2][16];
-    float pmax = -1e30f;
+      float p[2][16];
+      float pmax = -1e30f;
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub)
+      for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kk = sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+          float sv = (kv + kk < Sk) ? stile[sub][r] * scale : -1e30f;
+          p[sub][r] = sv;
+          pmax = fmaxf(pmax, sv);
+        }
+      pmax = fmaxf(pmax, __shfl_xor(pmax, 32, WAVE));
+      const float mnew = fmaxf(m[qs], pmax);
+      const float alpha = __expf(m[qs] - mnew);
+      float rowsum = 0.f;
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          p[sub][r] = __expf(p[sub][r] - mnew);
+          rowsum += p[sub][r];
+        }
+      rowsum += __shfl_xor(rowsum, 32, WAVE);
+      l[qs] = l[qs] * alpha + rowsum;
+      m[qs] = mnew;
+
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const int kk = sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
-        float sv = (kv + kk < Sk) ? stile[sub][r] * scale : -1e30f;
-        p[sub][r] = sv;
-        pmax = fmaxf(pmax, sv);
-      }
-    pmax = fmaxf(pmax, __shfl_xor(pmax, 32, WAVE));
-    const float mnew = fmaxf(m, pmax);
-    const float alpha = __expf(m - mnew);
-    float rowsum = 0.f;
+        const int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
+        const float a = __shfl(alpha, qrow, WAVE);
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub)
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        p[sub][r] = __expf(p[sub][r] - mnew);
-        rowsum += p[sub][r];
+        for (int d = 0; d < ND; ++d) o[qs][d][r] *= a;
       }
-    rowsum += __shfl_xor(rowsum, 32, WAVE);
-    l = l * alpha + rowsum;
-    m = mnew;
 
-    // O *= alpha: O's q-row layout differs from P's (lane-local) layout, so
-    // broadcast alpha[qrow] from the lane that owns that q-row.
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
-      const float a = __shfl(alpha, qrow, WAVE);
+      for (int sub = 0; sub < 2; ++sub) {
+        unsigned pk[8];
 #pragma unroll
-      for (int d = 0; d < ND; ++d) o[d][r] *= a;
-    }
-
-    // pack P -> PV A-fragments via permlane32_swap (per 16-k chunk)
+        for (int t = 0; t < 8; ++t)
+          pk[t] = pack_bf16(p[sub][2 * t], p[sub][2 * t + 1]);
+        bf16x8 pa0, pa1;
+        {
+          auto r0 =
+              __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
+          auto r1 =
+              __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
+          unsigned frag[4] = {r0[0], r1[0], r0[1], r1[1]};
+          pa0 = *(bf16x8 *)frag;
+          auto r2 =
+              __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
+          auto r3 =
+              __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
+          unsigned frag1[4] = {r2[0], r3[0], r2[1], r3[1]};
+          pa1 = *(bf16x8 *)frag1;
+        }
 #pragma unroll
-    for (int sub = 0; sub < 2; ++sub) {
-      unsigned pk[8];
-#pragma unroll
-      for (int t = 0; t < 8; ++t)
-        pk[t] = pack_bf16(p[sub][2 * t], p[sub][2 * t + 1]);
-      bf16x8 pa0, pa1;
-      {
-        auto r0 =
-            __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
-        auto r1 =
-            __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
-        unsigned frag[4] = {r0[0], r1[0], r0[1], r1[1]};
-        pa0 = *(bf16x8 *)frag;
-        auto r2 =
-            __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
-        auto r3 =
-            __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
-        unsigned frag1[4] = {r2[0], r3[0], r2[1], r3[1]};
-        pa1 = *(bf16x8 *)frag1;
-      }
-      // PV: O[32q, 32d] += P x V per d-block (keys sub*32 .. sub*32+31)
-#pragma unroll
-      for (int d = 0; d < ND; ++d) {
-        bf16x8 v0 =
-            *(const bf16x8 *)&vt[d * 32 + lq][sub * 32 + 8 * half];
-        o[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa0, v0, o[d], 0,
-                                                       0, 0);
-        bf16x8 v1 =
-            *(const bf16x8 *)&vt[d * 32 + lq][sub * 32 + 16 + 8 * half];
-        o[d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, v1, o[d], 0,
-                                                       0, 0);
+        for (int d = 0; d < ND; ++d) {
+          bf16x8 v0 =
+              *(const bf16x8 *)&vt[d * 32 + lq][sub * 32 + 8 * half];
+          o[qs][d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              pa0, v0, o[qs][d], 0, 0, 0);
+          bf16x8 v1 =
+              *(const bf16x8 *)&vt[d * 32 + lq][sub * 32 + 16 + 8 * half];
+          o[qs][d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              pa1, v1, o[qs][d], 0, 0, 0);
+        }
       }
     }
   }
 
   // epilogue: O /= l, store (column-per-lane scatter; widen later, T21)
-  const float linv = 1.0f / fmaxf(l, 1e-30f);
 #pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    const int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
-    const float inv = __shfl(linv, qrow, WAVE);
-    if (q0 + qrow >= Sq) continue;
+  for (int qs = 0; qs < 2; ++qs) {
+    const float linv = 1.0f / fmaxf(l[qs], 1e-30f);
 #pragma unroll
-    for (int d = 0; d < ND; ++d)
-      if (d * 32 + lq < D)
-        Ob[(q0 + qrow) * st.or_ + d * 32 + lq] = f2bf(o[d][r] * inv);
+    for (int r = 0; r < 16; ++r) {
+      const int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
+      const float inv = __shfl(linv, qrow, WAVE);
+      const long qq = q0 + qs * 32 + qrow;
+      if (qq >= Sq) continue;
+#pragma unroll
+      for (int d = 0; d < ND; ++d)
+        if (d * 32 + lq < D)
+          Ob[qq * st.or_ + d * 32 + lq] = f2bf(o[qs][d][r] * inv);
+    }
   }
 }
 
@@ -243,7 +248,7 @@ static torch::Tensor flash_attention_raw(torch::Tensor q, torch::Tensor k,
     st.ob = out.stride(0); st.oh = out.stride(1); st.or_ = out.stride(2);
   }
   const long DP = round16(D);
-  dim3 grid((unsigned)((Sq + 127) / 128), (unsigned)(B * H));
+  dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)(B * H));
   dim3 block(256);
   auto stream = cur_stream();
 
