@@ -16,6 +16,7 @@ static inline hipStream_t cur_stream() {
 #include "probe.hip"
 #include "attention.hip"
 #include "conv.hip"
+#include "gemm.hip"
 
 #define CHECK_IN(x)                                                     \
   TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                        \
@@ -329,4 +330,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_supported", &flash_supported);
   m.def("conv3x3_nhwc", &conv3x3_nhwc);
   m.def("conv3x3_supported", &conv3x3_supported);
+  m.def("linear_bf16", &linear_bf16);
+  m.def("linear_supported", &linear_supported);
 }
