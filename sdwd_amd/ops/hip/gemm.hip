@@ -3,8 +3,10 @@
 // exactly like conv.hip's weight rows. Same pipelined structure as the
 // conv kernel (128x128 tile, BK=64, global_load_lds double buffering, raw
 // barrier + counted vmcnt, source-side XOR chunk swizzle).
-// Used for the UNet's HBM-bound projection GEMMs (qkv/out/FF) where
-// hipBLASLt's generic pick measures well under the roofline at SD shapes.
+// STATUS: experimental / unused by the models — measured on MI355X,
+// hipBLASLt beats this 2-phase structure on the SD projection shapes
+// (666-1360 TF vs 480-630; tools/gemm_perf.py). The models keep F.linear;
+// this stays as the baseline for an 8-phase rewrite (guide §5 template).
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(4))) float f32x4g;

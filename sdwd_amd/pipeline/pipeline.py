@@ -407,19 +407,23 @@ class StableDiffusionPipeline:
             images.permute(0, 3, 1, 2).float() / 127.5 - 1.0
         ).to(self.device, self.dtype)
         if seeds:
-            outs = []
-            for i in range(x.shape[0]):
-                g = torch.Generator("cpu").manual_seed(
-                    (int(seeds[i]) ^ 0xE4C0DE) & 0xFFFFFFFF
-                )
-                moments = self.model.vae.encoder(x[i : i + 1])
-                mean, logvar = moments.chunk(2, dim=1)
-                std = torch.exp(0.5 * logvar.float().clamp(-30, 20))
-                n = torch.randn(
-                    mean.shape, generator=g, dtype=torch.float32
-                ).to(self.device)
-                outs.append(
-                    (mean.float() + std * n) * self.model.vae.cfg.scale_factor
-                )
-            return torch.cat(outs).to(self.dtype)
+            # one batched encoder forward; only the sampling noise is drawn
+            # per-image (deterministic per seed, device-independent)
+            moments = self.model.vae.encoder(x)
+            mean, logvar = moments.chunk(2, dim=1)
+            std = torch.exp(0.5 * logvar.float().clamp(-30, 20))
+            noise = torch.stack(
+                [
+                    torch.randn(
+                        mean.shape[1:],
+                        generator=torch.Generator("cpu").manual_seed(
+                            (int(sd) ^ 0xE4C0DE) & 0xFFFFFFFF
+                        ),
+                        dtype=torch.float32,
+                    )
+                    for sd in seeds
+                ]
+            ).to(self.device)
+            lat = (mean.float() + std * noise) * self.model.vae.cfg.scale_factor
+            return lat.to(self.dtype)
         return self.model.vae.encode(x)
