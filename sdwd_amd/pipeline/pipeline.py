@@ -362,8 +362,13 @@ class StableDiffusionPipeline:
 
         if decode:
             # chunked decode: bounds decoder activation memory at large
-            # batches and keeps per-chunk conv shapes in MIOpen's sweet spot
-            chunk = int(os.environ.get("SDWD_DECODE_CHUNK", "16"))
+            # batches (and sidesteps a torch NHWC-upsample grid-size limit
+            # seen at batch 64 x 512^2); scale the chunk down with pixels
+            px = req.width * req.height * (
+                req.hr_scale**2 if req.enable_hr else 1.0
+            )
+            default = max(1, int(16 * (512 * 512) / max(px, 1)))
+            chunk = int(os.environ.get("SDWD_DECODE_CHUNK", str(default)))
             outs = []
             for i in range(0, x.shape[0], chunk):
                 pixels = self.model.vae.decode(x[i : i + chunk])
