@@ -55,8 +55,10 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
   __shared__ __align__(16) __bf16 smem[4 * TILE_ELEMS];
 
   const long M = (long)Nn * Ho * Wo;
-  const long m0 = (long)blockIdx.x * CONV_BM;
-  const int n0 = blockIdx.y * CONV_BN;
+  // grid: x = Cout tiles, y = M tiles, so consecutively-dispatched blocks
+  // are the column tiles of ONE row tile and share its A reads in L2/L3
+  const long m0 = (long)blockIdx.y * CONV_BM;
+  const int n0 = blockIdx.x * CONV_BN;
 
   const int tid = threadIdx.x;
   const int lane = tid % WAVE;
@@ -161,12 +163,14 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
         bf[j] = *(const bf16x8 *)((const char *)(bbuf + bn * CONV_BK) +
                                   ck * 16);
       }
+      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int i = 0; i < 4; ++i)
 #pragma unroll
         for (int j = 0; j < 4; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[i], bf[j], acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
     }
     __builtin_amdgcn_s_barrier();  // everyone done reading this buffer
     if (t + 2 < NT) stage(t + 2, t & 1);
@@ -222,8 +226,10 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
   if (!zero_page.defined() || zero_page.device() != x.device())
     zero_page = torch::zeros({64}, x.options());
   const long M = (long)N * Ho * Wo;
-  dim3 grid((unsigned)((M + CONV_BM - 1) / CONV_BM),
-            (unsigned)((Cout + CONV_BN - 1) / CONV_BN));
+  dim3 grid((unsigned)((Cout + CONV_BN - 1) / CONV_BN),
+            (unsigned)((M + CONV_BM - 1) / CONV_BM));
+  TORCH_CHECK((M + CONV_BM - 1) / CONV_BM <= 65535,
+              "conv3x3: M tile count exceeds grid.y");
   dim3 block(256);
   const bool has_b = bias.has_value();
   const bool has_r = residual.has_value();
