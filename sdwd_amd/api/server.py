@@ -121,13 +121,15 @@ def create_app(engine: Optional[LocalEngine] = None,
     app.state.engine = engine
 
     def run_generation(gen: GenerationRequest) -> Dict[str, Any]:
+        # one generation at a time (the reference serialized on the host's
+        # queue_lock, world.py:244,273); concurrent requests queue here
         with state.lock:
             state.busy = True
             state.started_at = time.time()
-        try:
-            result = engine.generate(gen)
-        finally:
-            state.busy = False
+            try:
+                result = engine.generate(gen)
+            finally:
+                state.busy = False
         images = [_b64_png(result.images[i])
                   for i in range(result.images.shape[0])]
         if result.grid is not None:

@@ -349,3 +349,17 @@ class TestAddLayerNorm:
         ref_ln = torch.nn.functional.layer_norm(ref_s, (D,), w, b, 1e-5)
         assert relerr(s, ref_s) < 0.05
         assert relerr(ln, ref_ln) < 0.05
+
+
+class TestCrossDeviceConsistency:
+    def test_tiny_pipeline_gpu_matches_cpu(self, dev):
+        """bf16 GPU pipeline vs fp32 CPU pipeline on the same request: same
+        trajectory class (loose tolerance for precision differences)."""
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        req = PipelineRequest(prompt="xdev", steps=2, width=64, height=64,
+                              seeds=[21], sampler_name="Euler")
+        cpu = StableDiffusionPipeline("tiny", device="cpu").generate(req)
+        gpu = StableDiffusionPipeline("tiny", device=dev).generate(req)
+        diff = (cpu.images.float() - gpu.images.float()).abs()
+        assert diff.mean() < 8.0, f"mean abs diff {diff.mean()} too high"
