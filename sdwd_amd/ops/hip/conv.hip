@@ -253,6 +253,31 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
     cbptr = (const __hip_bfloat16 *)cbt.data_ptr();
   }
   auto stream = cur_stream();
+  static const bool use_v2 = getenv("SDWD_CONV_V2") != nullptr;
+  if (!use_v2) {
+    // v3: deep-pipelined 256x128 tile (conv_v3.hip)
+    dim3 g3((unsigned)((Cout + 127) / 128), (unsigned)((M + 255) / 256));
+    TORCH_CHECK((M + 255) / 256 <= 65535, "conv3x3: M tiles exceed grid.y");
+    dim3 b3(512);
+#define PICK3(B_, R_, C_) conv3x3_v3_kernel<B_, R_, C_>
+    auto k3 = has_b ? (has_r ? (has_cb ? PICK3(true, true, true)
+                                       : PICK3(true, true, false))
+                             : (has_cb ? PICK3(true, false, true)
+                                       : PICK3(true, false, false)))
+                    : (has_r ? (has_cb ? PICK3(false, true, true)
+                                       : PICK3(false, true, false))
+                             : (has_cb ? PICK3(false, false, true)
+                                       : PICK3(false, false, false)));
+#undef PICK3
+    hipLaunchKernelGGL(k3, g3, b3, 0, stream,
+                       (const __hip_bfloat16 *)x.data_ptr(),
+                       (const __hip_bfloat16 *)w_prep.data_ptr(), bptr, rptr,
+                       cbptr,
+                       (const __hip_bfloat16 *)zero_page.data_ptr(),
+                       (__hip_bfloat16 *)y.data_ptr(), N, H, W, Cin, Cout,
+                       Ho, Wo, (int)stride);
+    return y;
+  }
 #define PICK(B_, R_, C_) conv3x3_nhwc_bf16_kernel<B_, R_, C_>
   auto kern =
       has_b ? (has_r ? (has_cb ? PICK(true, true, true)

@@ -15,6 +15,7 @@ static inline hipStream_t cur_stream() {
 #include "softmax.hip"
 #include "probe.hip"
 #include "attention.hip"
+#include "conv_v3.hip"
 #include "conv.hip"
 #include "gemm.hip"
 
