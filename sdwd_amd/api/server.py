@@ -263,16 +263,35 @@ def create_app(engine: Optional[LocalEngine] = None,
 
     @app.get("/sdapi/v1/progress")
     def progress():
-        eta = 0.0
+        frac = engine.progress() if state.busy else 1.0
         return {
-            "progress": 1.0 if not state.busy else 0.5,
-            "eta_relative": eta,
+            "progress": frac,
+            "eta_relative": 0.0,
             "state": {
                 "job": "generate" if state.busy else "",
                 "interrupted": engine.world.interrupted.is_set(),
             },
             "current_image": None,
         }
+
+    @app.post("/sdapi/v1/server-restart")
+    def server_restart():
+        """Soft restart (ref worker.py:690-717): rebuild every rank's
+        pipeline, clear caches, reset worker states."""
+        from ..models.registry import clear_cache
+        from ..pipeline import StableDiffusionPipeline
+
+        log.info("soft restart: rebuilding pipelines")
+        engine.world.clear_interrupt()
+        clear_cache()
+        for label, pipe in list(engine.pipes.items()):
+            engine.pipes[label] = StableDiffusionPipeline(
+                state.current_model, device=pipe.device, dtype=pipe.dtype
+            )
+        for w in engine.world.workers:
+            if w.state is not State.DISABLED:
+                w.sm.force(State.IDLE)
+        return {}
 
     @app.get("/sdwd/status")
     def status():

@@ -23,6 +23,9 @@ class Job:
     # measured on completion:
     predicted_eta: float = 0.0
     elapsed: float = 0.0
+    # live progress (denoise steps done / total), updated by the executor
+    steps_done: int = 0
+    steps_total: int = 0
 
     def add_work(self, images: int, width: int, height: int,
                  pixel_cap: int = 0) -> int:

@@ -181,6 +181,7 @@ class LocalEngine(_EngineBase):
         self.model_name = model
         self.world = world or World.from_devices(len(devices))
         self.pipes: Dict[str, StableDiffusionPipeline] = {}
+        self._live_jobs = []
         self.dtype = dtype
         for i, dev in enumerate(devices):
             self.pipes[f"gpu{i}"] = StableDiffusionPipeline(
@@ -236,9 +237,15 @@ class LocalEngine(_EngineBase):
                 ]
                 init_latents = pipe.encode_image(src[idx], seeds=job.seeds)
             t0 = time.perf_counter()
+
+            def on_step(i, n, _job=job):
+                _job.steps_done = i
+                _job.steps_total = n
+
             res = pipe.generate(
                 _job_pipeline_request(gen, job, init_latents),
                 interrupt=worker.interrupt_event.is_set,
+                step_callback=on_step,
             )
             job.elapsed = time.perf_counter() - t0
             shards[id(job)] = res.images
@@ -252,10 +259,22 @@ class LocalEngine(_EngineBase):
             errors[job.worker_label] = exc
             worker.set_state(State.UNAVAILABLE)
 
+    def progress(self) -> float:
+        """0..1 across the jobs of the generation in flight (ref /progress)."""
+        jobs = self._live_jobs
+        if not jobs:
+            return 1.0
+        fracs = [
+            (j.steps_done / j.steps_total) if j.steps_total else 0.0
+            for j in jobs
+        ]
+        return sum(fracs) / len(fracs)
+
     def generate(self, gen: GenerationRequest) -> GalleryResult:
         t0 = time.perf_counter()
         self.world.clear_interrupt()
         jobs = self.world.make_jobs(gen.sched())
+        self._live_jobs = jobs
         shards: Dict[str, torch.Tensor] = {}
         infos: Dict[str, List[str]] = {}
         errors: Dict[str, Exception] = {}

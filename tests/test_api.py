@@ -155,3 +155,16 @@ class TestWebUI:
         assert r.status_code == 200
         assert "sdwd_amd" in r.text
         assert "/sdwd/status" in r.text
+
+
+class TestRestart:
+    def test_soft_restart(self, client):
+        r = client.post("/sdapi/v1/server-restart")
+        assert r.status_code == 200
+        # still serves after restart
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={"prompt": "post-restart", "steps": 1, "width": 64,
+                  "height": 64, "seed": 1},
+        )
+        assert r.status_code == 200
