@@ -396,17 +396,27 @@ class DistributedEngine(_EngineBase):
             self._store.set("sdwd_interrupt", str(self._interrupt_epoch))
 
     def _interrupted(self) -> bool:
+        """Polled between denoise steps; the store check is a TCP round
+        trip, so throttle to the reference's 0.5 s cadence
+        (ref worker.py:444-448)."""
         if self._store is None:
             return False
+        now = time.monotonic()
+        if now - getattr(self, "_int_polled", 0.0) < 0.5:
+            return getattr(self, "_int_cached", False)
+        self._int_polled = now
+        val = False
         try:
             if self._store.check(["sdwd_interrupt"]):
-                val = int(self._store.get("sdwd_interrupt"))
-                return val > 0
+                val = int(self._store.get("sdwd_interrupt")) > 0
         except Exception:
-            return False
-        return False
+            val = False
+        self._int_cached = val
+        return val
 
     def _clear_interrupt(self) -> None:
+        self._int_cached = False
+        self._int_polled = 0.0
         if self._store is not None and self.rank == 0:
             try:
                 self._store.set("sdwd_interrupt", "0")
