@@ -50,13 +50,30 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
   __shared__ __align__(16) __bf16 kt[KVB][DPAD + PADK];
   __shared__ __align__(16) __bf16 vt[DV][KVB + PADK];
 
-  const int bh = blockIdx.y;
+  // XCD-aware remap (T1): dispatch places block b on XCD b%8, so give a
+  // head's q-blocks ids congruent mod 8 — they then share K/V in one
+  // XCD's L2 instead of re-fetching per q-block. Bijective for BH%8==0.
+  int bh, qblk;
+  {
+    const int nq = gridDim.x;
+    const int id = blockIdx.x + blockIdx.y * nq;
+    const long BH = (long)gridDim.y;
+    if (BH % 8 == 0) {
+      const int grp = id / (8 * nq);
+      const int rem = id % (8 * nq);
+      bh = grp * 8 + rem % 8;
+      qblk = rem / 8;
+    } else {
+      bh = blockIdx.y;
+      qblk = blockIdx.x;
+    }
+  }
   const int bb = bh / H, hh = bh % H;
   const int wid = threadIdx.x / WAVE;
   const int lane = threadIdx.x % WAVE;
   const int lq = lane % 32;   // q-row (softmax) / d-col (PV C) index
   const int half = lane / 32; // half-wave id
-  const long q0 = (long)blockIdx.x * 256 + wid * 64;
+  const long q0 = (long)qblk * 256 + wid * 64;
 
   const __hip_bfloat16 *Qb = Q + bb * st.qb + hh * st.qh;
   const __hip_bfloat16 *Kb = K + bb * st.kb + hh * st.kh;
@@ -123,6 +140,9 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
               kf, qf[qs][c], stile[sub], 0, 0, 0);
         }
 
+      // softmax in base-2: scores carry scale*log2(e) once, so the hot
+      // exponentials are bare v_exp2 with no per-element multiply
+      const float l2scale = scale * 1.4426950408889634f;
       float p[2][16];
       float pmax = -1e30f;
 #pragma unroll
@@ -130,19 +150,19 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int kk = sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
-          float sv = (kv + kk < Sk) ? stile[sub][r] * scale : -1e30f;
+          float sv = (kv + kk < Sk) ? stile[sub][r] * l2scale : -1e30f;
           p[sub][r] = sv;
           pmax = fmaxf(pmax, sv);
         }
       pmax = fmaxf(pmax, __shfl_xor(pmax, 32, WAVE));
       const float mnew = fmaxf(m[qs], pmax);
-      const float alpha = __expf(m[qs] - mnew);
+      const float alpha = exp2f(m[qs] - mnew);
       float rowsum = 0.f;
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub)
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          p[sub][r] = __expf(p[sub][r] - mnew);
+          p[sub][r] = exp2f(p[sub][r] - mnew);
           rowsum += p[sub][r];
         }
       rowsum += __shfl_xor(rowsum, 32, WAVE);
