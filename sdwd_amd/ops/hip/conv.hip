@@ -55,9 +55,11 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
   __shared__ __align__(16) __bf16 smem[4 * TILE_ELEMS];
 
   const long M = (long)Nn * Ho * Wo;
-  // grid: x = Cout tiles, y = M tiles, so consecutively-dispatched blocks
-  // are the column tiles of ONE row tile and share its A reads in L2/L3
-  const long m0 = (long)blockIdx.y * CONV_BM;
+  // grid: x = Cout tiles, y (+z overflow) = M tiles, so consecutively-
+  // dispatched blocks are the column tiles of ONE row tile and share its
+  // A reads in L2/L3
+  const long m0 =
+      ((long)blockIdx.y + (long)blockIdx.z * 32768) * CONV_BM;
   const int n0 = blockIdx.x * CONV_BN;
 
   const int tid = threadIdx.x;
@@ -226,10 +228,10 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
   if (!zero_page.defined() || zero_page.device() != x.device())
     zero_page = torch::zeros({64}, x.options());
   const long M = (long)N * Ho * Wo;
+  const long mtiles = (M + CONV_BM - 1) / CONV_BM;
   dim3 grid((unsigned)((Cout + CONV_BN - 1) / CONV_BN),
-            (unsigned)((M + CONV_BM - 1) / CONV_BM));
-  TORCH_CHECK((M + CONV_BM - 1) / CONV_BM <= 65535,
-              "conv3x3: M tile count exceeds grid.y");
+            (unsigned)std::min<long>(mtiles, 32768),
+            (unsigned)((mtiles + 32767) / 32768));
   dim3 block(256);
   const bool has_b = bias.has_value();
   const bool has_r = residual.has_value();
@@ -259,8 +261,10 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
   static const bool use_v3 = getenv("SDWD_CONV_V3") != nullptr;
   if (use_v3) {
     // v3: deep-pipelined 256x128 tile (conv_v3.hip)
-    dim3 g3((unsigned)((Cout + 127) / 128), (unsigned)((M + 255) / 256));
-    TORCH_CHECK((M + 255) / 256 <= 65535, "conv3x3: M tiles exceed grid.y");
+    const long mt3 = (M + 255) / 256;
+    dim3 g3((unsigned)((Cout + 127) / 128),
+            (unsigned)std::min<long>(mt3, 32768),
+            (unsigned)((mt3 + 32767) / 32768));
     dim3 b3(512);
 #define PICK3(B_, R_, C_) conv3x3_v3_kernel<B_, R_, C_>
     auto k3 = has_b ? (has_r ? (has_cb ? PICK3(true, true, true)

@@ -39,7 +39,8 @@ __launch_bounds__(512, 2) __global__ void conv3x3_v3_kernel(
   __shared__ __align__(16) __bf16 smem[3 * V3_BUF];
 
   const long M = (long)Nn * Ho * Wo;
-  const long m0 = (long)blockIdx.y * V3_BM;
+  const long m0 = ((long)blockIdx.y + (long)blockIdx.z * 32768) * V3_BM;
+  if (m0 >= M) return;
   const int n0 = blockIdx.x * V3_BN;
 
   const int tid = threadIdx.x;
