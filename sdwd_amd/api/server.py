@@ -10,7 +10,6 @@ data (worker states, speeds, log ring buffer — ref ui.py:230-249).
 from __future__ import annotations
 
 import base64
-import io
 import json
 import threading
 import time

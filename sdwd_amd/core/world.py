@@ -30,7 +30,7 @@ from ..config import (
     save_config,
 )
 from ..utils import get_logger
-from .eta import sampler_cost
+
 from .job import Job
 from .seeds import fix_seed, shard_seeds
 from .state import State

@@ -13,7 +13,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from .. import ops
+
 
 
 class CLIPAttention(nn.Module):

@@ -6,7 +6,6 @@ epsilon-predictor taking the *scaled* input x/sqrt(sigma^2+1) and the
 """
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 
 import torch
