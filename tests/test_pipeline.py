@@ -301,3 +301,33 @@ class TestSamplerOracle:
         out = sampler.sample(model_fn, x, noise_fn=noise_fn)
         err = (out - x0).abs().max().item()
         assert err < 2e-2, f"{name} landed {err} away from the point mass"
+
+
+class TestKarrasOracle:
+    @pytest.mark.parametrize("name", ["DPM++ 2M Karras", "DPM++ SDE Karras"])
+    def test_karras_also_converges(self, name):
+        """Same point-mass oracle through the Karras sigma schedule (uses
+        fractional timesteps -> sigma via exp-interp of the log table)."""
+        from sdwd_amd.pipeline.samplers import build_sampler
+        from sdwd_amd.pipeline.schedule import karras_schedule, make_sigmas_full
+
+        table = make_sigmas_full()
+        logt = table.log()
+        x0 = torch.full((1, 4, 8, 8), -0.3)
+        sched = karras_schedule(12)
+        sampler = build_sampler(name, sched)
+
+        def model_fn(x_scaled, t):
+            lo = int(t)
+            frac = t - lo
+            hi = min(lo + 1, len(table) - 1)
+            sigma = float((logt[lo] * (1 - frac) + logt[hi] * frac).exp())
+            c_in = 1.0 / (sigma * sigma + 1.0) ** 0.5
+            x = x_scaled.float() / c_in
+            return (x - x0) / sigma
+
+        g = torch.Generator().manual_seed(11)
+        x = torch.randn(x0.shape, generator=g) * float(sched.sigmas[0])
+        out = sampler.sample(model_fn, x,
+                             noise_fn=lambda: torch.randn(x0.shape, generator=g))
+        assert (out - x0).abs().max().item() < 5e-2
