@@ -12,7 +12,7 @@ from __future__ import annotations
 import math
 import os
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass, field, replace
 from typing import Callable, List, Optional
 
 import torch
@@ -126,6 +126,9 @@ class StableDiffusionPipeline:
             os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
             self.model.unet.to(memory_format=torch.channels_last)
             self.model.vae.to(memory_format=torch.channels_last)
+        from ..models.lora import LoraManager
+
+        self.lora = LoraManager(self.model.unet)
         self._denoiser = GraphedDenoiser(
             lambda x, ts, ctx, y: self.model.unet(x, ts, ctx, y=y),
             self.device,
@@ -182,6 +185,14 @@ class StableDiffusionPipeline:
         decode: bool = True,
     ) -> PipelineResult:
         t0 = time.perf_counter()
+        # sdwui <lora:name:scale> prompt tags: merge the adapter set into
+        # the weights for this request (reversible; no per-step cost)
+        from ..models.lora import parse_prompt_loras
+
+        prompt, loras = parse_prompt_loras(req.prompt)
+        if loras or self.lora.active:
+            req = replace(req, prompt=prompt)
+            self.lora.set_active(loras)
         b = req.batch_size
         f = self.model.vae.cfg.downsample_factor
         lat_h, lat_w = req.height // f, req.width // f
