@@ -57,6 +57,7 @@ class Txt2ImgRequest(BaseModel):
 class Img2ImgRequest(Txt2ImgRequest):
     init_images: List[str] = Field(default_factory=list)  # base64 PNG
     denoising_strength: float = 0.75
+    mask: Optional[str] = None  # base64 PNG, white = repaint
 
 
 class OptionsRequest(BaseModel):
@@ -189,6 +190,11 @@ def create_app(engine: Optional[LocalEngine] = None,
         if not req.init_images:
             raise HTTPException(422, "init_images required")
         inits = torch.stack([_decode_b64_png(d) for d in req.init_images])
+        mask_image = None
+        if req.mask:
+            mask_image = _decode_b64_png(req.mask).float().mean(-1).to(
+                torch.uint8
+            )
         gen = GenerationRequest(
             prompt=req.prompt,
             negative_prompt=req.negative_prompt,
@@ -203,6 +209,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             subseed_strength=req.subseed_strength,
             init_images=inits,
             denoising_strength=req.denoising_strength,
+            mask_image=mask_image,
         )
         return run_generation(gen)
 

@@ -48,6 +48,7 @@ class GenerationRequest:
     subseed_strength: float = 0.0
     init_images: Optional[torch.Tensor] = None  # [B,H,W,3] uint8 (img2img)
     denoising_strength: float = 0.75
+    mask_image: Optional[torch.Tensor] = None   # [H,W] uint8, 255=repaint
     enable_hr: bool = False
     hr_scale: float = 2.0
     hr_steps: int = 0
@@ -97,6 +98,7 @@ def _job_pipeline_request(
         subseed_strength=gen.subseed_strength,
         init_latents=init_latents,
         denoising_strength=gen.denoising_strength,
+        mask_image=gen.mask_image,
         enable_hr=gen.enable_hr,
         hr_scale=gen.hr_scale,
         hr_steps=gen.hr_steps,

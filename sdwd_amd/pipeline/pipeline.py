@@ -42,6 +42,8 @@ class PipelineRequest:
     # img2img
     init_latents: Optional[torch.Tensor] = None  # pre-encoded [B,4,h,w]
     denoising_strength: float = 0.75
+    # inpainting: uint8 mask, 255 = repaint region (sdwui convention)
+    mask_image: Optional[torch.Tensor] = None    # [H,W] or [B,H,W]
     # hires fix (sdwui two-pass: base gen -> latent upscale -> img2img pass)
     enable_hr: bool = False
     hr_scale: float = 2.0
@@ -319,9 +321,32 @@ class StableDiffusionPipeline:
                 return True
             return False
 
+        post_step = None
+        if req.mask_image is not None and req.init_latents is not None:
+            # latent-space inpainting: outside the mask the trajectory is
+            # pinned to the init re-noised at the current sigma (sdwui
+            # masked-img2img semantics; deterministic via the init noise)
+            mk = req.mask_image
+            if mk.dim() == 2:
+                mk = mk[None]
+            lat_mask = torch.nn.functional.interpolate(
+                (mk.float() / 255.0)[:, None], size=(lat_h, lat_w),
+                mode="area",
+            ).clamp(0, 1).to(self.device)
+            if lat_mask.shape[0] == 1 and b > 1:
+                lat_mask = lat_mask.expand(b, -1, -1, -1)
+            init_lat = req.init_latents.to(self.device).float()
+            noise_f32 = noise.float()
+
+            def post_step(xc, sigma_next):
+                keep = init_lat + noise_f32 * sigma_next
+                return (
+                    lat_mask * xc.float() + (1.0 - lat_mask) * keep
+                ).to(xc.dtype)
+
         x = sampler.sample(
             model_fn, x, noise_fn=noise_fn, callback=step_callback,
-            interrupt=_interrupt,
+            interrupt=_interrupt, post_step=post_step,
         )
 
         # hires fix: latent-upscale the base result and run a second,

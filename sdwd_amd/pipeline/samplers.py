@@ -64,7 +64,10 @@ class Sampler:
         noise_fn: Optional[Callable[[], torch.Tensor]] = None,
         callback: StepCallback = None,
         interrupt: Optional[Callable[[], bool]] = None,
+        post_step: Optional[Callable[[torch.Tensor, float], torch.Tensor]] = None,
     ) -> torch.Tensor:
+        """post_step(x, sigma_next) transforms the state after each update
+        (inpainting re-imposes the init outside the mask there)."""
         sig = self.schedule.sigmas.tolist()
         ts = self.schedule.timesteps.tolist()
         self.reset()
@@ -72,6 +75,8 @@ class Sampler:
             if interrupt is not None and interrupt():
                 break
             x = self.step(model_fn, x, sig[i], sig[i + 1], ts[i], noise_fn)
+            if post_step is not None:
+                x = post_step(x, sig[i + 1])
             if callback is not None:
                 callback(i + 1, len(ts))
         return x
