@@ -363,3 +363,37 @@ class TestCrossDeviceConsistency:
         gpu = StableDiffusionPipeline("tiny", device=dev).generate(req)
         diff = (cpu.images.float() - gpu.images.float()).abs()
         assert diff.mean() < 8.0, f"mean abs diff {diff.mean()} too high"
+
+
+class TestLoRAGPU:
+    def test_lora_on_gpu_pipeline(self, dev):
+        """LoRA merge/unmerge through the GPU pipeline (also exercises the
+        SDConv2d prepped-weight cache invalidation)."""
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny", device=dev)
+        base = dict(steps=2, width=64, height=64, seeds=[3])
+        plain = pipe.generate(PipelineRequest(prompt="cow", **base)).images
+        lora = pipe.generate(
+            PipelineRequest(prompt="cow <lora:gstyle:1.0>", **base)
+        ).images
+        plain2 = pipe.generate(PipelineRequest(prompt="cow", **base)).images
+        assert not torch.equal(plain, lora)
+        assert torch.equal(plain, plain2)
+
+
+class TestInpaintingGPU:
+    def test_mask_runs_on_gpu(self, dev):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny", device=dev)
+        init = torch.randint(0, 255, (1, 64, 64, 3), dtype=torch.uint8)
+        lat = pipe.encode_image(init, seeds=[5])
+        mask = torch.zeros(64, 64, dtype=torch.uint8)
+        mask[:, 32:] = 255
+        res = pipe.generate(
+            PipelineRequest(prompt="gpu paint", steps=3, width=64, height=64,
+                            seeds=[5], init_latents=lat, mask_image=mask)
+        )
+        assert res.images.shape == (1, 64, 64, 3)
+        assert torch.isfinite(res.images.float()).all()
