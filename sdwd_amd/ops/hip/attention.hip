@@ -143,13 +143,13 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
         }
       pmax = fmaxf(pmax, __shfl_xor(pmax, 32, WAVE));
       const float mnew = fmaxf(m[qs], pmax);
-      const float alpha = exp2f(m[qs] - mnew);
+      const float alpha = __builtin_amdgcn_exp2f(m[qs] - mnew);
       float rowsum = 0.f;
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub)
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          p[sub][r] = exp2f(p[sub][r] - mnew);
+          p[sub][r] = __builtin_amdgcn_exp2f(p[sub][r] - mnew);
           rowsum += p[sub][r];
         }
       rowsum += __shfl_xor(rowsum, 32, WAVE);
