@@ -397,3 +397,15 @@ class TestInpaintingGPU:
         )
         assert res.images.shape == (1, 64, 64, 3)
         assert torch.isfinite(res.images.float()).all()
+
+
+class TestDeterminismGPU:
+    def test_same_request_same_images(self, dev):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny", device=dev)
+        req = PipelineRequest(prompt="det", steps=3, width=64, height=64,
+                              seeds=[13, 14])
+        a = pipe.generate(req).images
+        b = pipe.generate(req).images
+        assert torch.equal(a, b), "GPU pipeline must be run-to-run exact"
