@@ -190,6 +190,10 @@ class LocalEngine(_EngineBase):
                 model, device=dev, dtype=dtype
             )
         self._fail_injection: Dict[str, bool] = {}
+        if torch.cuda.is_available():
+            # startup liveness sweep (ref distributed.py:48-52 pinged all
+            # remotes at init); CPU ranks have no memory probe to ping
+            self.world.ping(indiscriminate=True)
 
     # benchmark runner wired into core.Worker.benchmark (ref C8)
     def _bench_runner(self, worker: Worker, payload) -> float:
