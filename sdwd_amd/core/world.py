@@ -162,7 +162,6 @@ class World:
                 w.label: self._predict(w, shares[w.label], request)
                 for w in realtime
             }
-            t_max = max(etas.values())
             worst = max(realtime, key=lambda w: etas[w.label])
             one_img = self._predict(worst, 1, request)
             others = [w for w in realtime if w is not worst]
