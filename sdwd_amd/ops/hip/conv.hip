@@ -41,7 +41,10 @@ __device__ __forceinline__ int cswz(int row, int chunk) {
   return chunk ^ ((row >> 1) & 7);
 }
 
-template <bool HAS_BIAS, bool HAS_RES, bool HAS_CB>
+// BN = 128 (default) or 64 (exact tiling for Cout % 128 == 64, e.g. the
+// SD1.5 320-channel level: 5 exact 64-col tiles instead of 3 x 128 with a
+// 17% masked-FLOP tail). NJ = column fragments per wave.
+template <bool HAS_BIAS, bool HAS_RES, bool HAS_CB, int BN = 128>
 __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
     const __hip_bfloat16 *__restrict__ X,   // [N,H,W,Cin]
     const __hip_bfloat16 *__restrict__ Wt,  // [Cout,3,3,Cin]
@@ -58,15 +61,17 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
   // grid: x = Cout tiles, y (+z overflow) = M tiles, so consecutively-
   // dispatched blocks are the column tiles of ONE row tile and share its
   // A reads in L2/L3
+  constexpr int NJ = BN / 32;          // 4 or 2 column fragments per wave
+  constexpr int BTILE = BN * CONV_BK;  // B-tile elements
   const long m0 =
       ((long)blockIdx.y + (long)blockIdx.z * 32768) * CONV_BM;
-  const int n0 = blockIdx.x * CONV_BN;
+  const int n0 = blockIdx.x * BN;
 
   const int tid = threadIdx.x;
   const int lane = tid % WAVE;
   const int wid = tid / WAVE;
   const int wm = (wid >> 1) * 64;
-  const int wn = (wid & 1) * 64;
+  const int wn = (wid & 1) * (BN / 2);
 
   // ---- per-lane DMA row bookkeeping -------------------------------------
   // stage instr i (0..3) per wave writes LDS bytes
@@ -74,7 +79,8 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
   //   row_i = i*32 + wid*8 + lane/8, chunk = lane%8 (then source-swizzled)
   long abase[4];
   int hs[4], ws[4];   // ho*stride, wo*stride per staged A row
-  int bco[4];         // cout row per staged B row
+  int bco[4];         // cout row per staged B row (NB instrs used)
+  constexpr int NB = BN / 32;  // B stage instrs (rows BN over 4 waves x 8)
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     const int row = i * 32 + wid * 8 + lane / 8;
@@ -98,7 +104,7 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
     const int kc = t % kc_per_plane;
     const int dy = plane / 3 - 1, dx = plane % 3 - 1;
     const long poff = ((long)dy * W + dx) * Cin + (long)kc * CONV_BK;
-    __bf16 *abuf = smem + b * 2 * TILE_ELEMS;
+    __bf16 *abuf = smem + b * (TILE_ELEMS + BTILE);
     __bf16 *bbuf = abuf + TILE_ELEMS;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
@@ -113,24 +119,26 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
           (__attribute__((address_space(3))) unsigned int
                *)(abuf + i * 2048 + wid * 512),
           16, 0, 0);
-      const bool bv = bco[i] < Cout;
-      const __hip_bfloat16 *bsrc =
-          bv ? (Wt + (long)bco[i] * 9 * Cin + plane * Cin + kc * CONV_BK +
-                sc * 8)
-             : Zero;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int *)bsrc,
-          (__attribute__((address_space(3))) unsigned int
-               *)(bbuf + i * 2048 + wid * 512),
-          16, 0, 0);
+      if (i < NB) {
+        const bool bv = bco[i] < Cout;
+        const __hip_bfloat16 *bsrc =
+            bv ? (Wt + (long)bco[i] * 9 * Cin + plane * Cin +
+                  kc * CONV_BK + sc * 8)
+               : Zero;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int *)bsrc,
+            (__attribute__((address_space(3))) unsigned int
+                 *)(bbuf + i * 2048 + wid * 512),
+            16, 0, 0);
+      }
     }
   };
 
-  f32x4c acc[4][4];
+  f32x4c acc[4][NJ];
 #pragma unroll
   for (int i = 0; i < 4; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4c){};
+    for (int j = 0; j < NJ; ++j) acc[i][j] = (f32x4c){};
 
   stage(0, 0);
   if (NT > 1) stage(1, 1);
@@ -139,18 +147,18 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
   const int kq = (lane / 16) * 8;  // k offset of this lane's 8 elements
 
   for (int t = 0; t < NT; ++t) {
-    // current tile's 8 DMAs landed; the next tile's 8 stay in flight
+    // current tile's DMAs landed; the next tile's stay in flight
     if (t + 1 < NT)
-      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      asm volatile("s_waitcnt vmcnt(%0)" :: "i"(4 + NB) : "memory");
     else
       asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
 
-    const __bf16 *abuf = smem + (t & 1) * 2 * TILE_ELEMS;
+    const __bf16 *abuf = smem + (t & 1) * (TILE_ELEMS + BTILE);
     const __bf16 *bbuf = abuf + TILE_ELEMS;
 #pragma unroll
     for (int s = 0; s < 2; ++s) {
-      bf16x8 af[4], bf[4];
+      bf16x8 af[4], bf[NJ];
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const int am = wm + i * 16 + l16;
@@ -159,7 +167,7 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
                                   ck * 16);
       }
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
+      for (int j = 0; j < NJ; ++j) {
         const int bn = wn + j * 16 + l16;
         const int ck = cswz(bn, (s * 32 + kq) / 8);
         bf[j] = *(const bf16x8 *)((const char *)(bbuf + bn * CONV_BK) +
@@ -169,7 +177,7 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
 #pragma unroll
       for (int i = 0; i < 4; ++i)
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < NJ; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[i], bf[j], acc[i][j], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
@@ -183,7 +191,7 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < NJ; ++j) {
       const int co = n0 + wn + j * 16 + l16;
       if (co >= Cout) continue;
       const float bv = HAS_BIAS ? bias[co] : 0.0f;
@@ -285,7 +293,10 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
                        Ho, Wo, (int)stride);
     return y;
   }
-#define PICK(B_, R_, C_) conv3x3_nhwc_bf16_kernel<B_, R_, C_>
+  const bool bn64 = (Cout % 128) == 64;  // exact 64-col tiling (Cout 320..)
+#define PICK(B_, R_, C_)                                              \
+  (bn64 ? conv3x3_nhwc_bf16_kernel<B_, R_, C_, 64>                    \
+        : conv3x3_nhwc_bf16_kernel<B_, R_, C_, 128>)
   auto kern =
       has_b ? (has_r ? (has_cb ? PICK(true, true, true)
                                : PICK(true, true, false))
@@ -296,6 +307,8 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
                      : (has_cb ? PICK(false, false, true)
                                : PICK(false, false, false)));
 #undef PICK
+  if (bn64)
+    grid.x = (unsigned)((Cout + 63) / 64);
   hipLaunchKernelGGL(kern, grid, block, 0, stream,
                      (const __hip_bfloat16 *)x.data_ptr(),
                      (const __hip_bfloat16 *)w_prep.data_ptr(), bptr, rptr,
