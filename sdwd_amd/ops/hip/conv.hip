@@ -293,7 +293,10 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
                        Ho, Wo, (int)stride);
     return y;
   }
-  const bool bn64 = (Cout % 128) == 64;  // exact 64-col tiling (Cout 320..)
+  // BN=64 exact tiling for Cout=320 measured SLOWER (460 vs 557 TF: the
+  // halved B-reuse doubles A traffic and drops per-block efficiency more
+  // than the 17% masked-FLOP tail costs); template retained, 128 always.
+  const bool bn64 = false;
 #define PICK(B_, R_, C_)                                              \
   (bn64 ? conv3x3_nhwc_bf16_kernel<B_, R_, C_, 64>                    \
         : conv3x3_nhwc_bf16_kernel<B_, R_, C_, 128>)
