@@ -1,0 +1,82 @@
+"""Unit tests for the utility layers: logging ring buffer, CLI flags,
+process-group helpers in single-process mode, tokenizer determinism."""
+import argparse
+
+import torch
+
+from sdwd_amd.config import add_flags, export_env
+from sdwd_amd.models import tokenizer
+from sdwd_amd.parallel import broadcast_object, gather_images
+from sdwd_amd.parallel.group import allgather_floats
+from sdwd_amd.utils import get_logger, ring_buffer
+
+
+class TestLogging:
+    def test_ring_buffer_captures(self):
+        log = get_logger("ringtest")
+        log.info("ring-entry-%d", 42)
+        lines = ring_buffer()
+        assert any("ring-entry-42" in ln for ln in lines)
+
+    def test_ring_buffer_bounded(self):
+        log = get_logger("ringtest")
+        for i in range(100):
+            log.info("flood %d", i)
+        assert len(ring_buffer()) <= 64
+
+
+class TestFlags:
+    def test_defaults_and_export(self, monkeypatch, tmp_path):
+        monkeypatch.delenv("SDWD_CONFIG", raising=False)
+        ap = argparse.ArgumentParser()
+        add_flags(ap)
+        args = ap.parse_args(
+            ["--sdwd-config", str(tmp_path / "c.json"), "--sdwd-debug"]
+        )
+        assert args.sdwd_debug
+        export_env(args)
+        import os
+
+        assert os.environ["SDWD_CONFIG"].endswith("c.json")
+        assert os.environ["SDWD_DEBUG"] == "1"
+
+
+class TestGroupSingleProcess:
+    def test_broadcast_object_identity(self):
+        obj = {"a": [1, 2, 3]}
+        assert broadcast_object(obj) is obj
+
+    def test_gather_images_passthrough(self):
+        shard = torch.randint(0, 255, (3, 8, 8, 3), dtype=torch.uint8)
+        out = gather_images(shard, [3], "cpu")
+        assert torch.equal(out, shard)
+
+    def test_allgather_floats_single(self):
+        assert allgather_floats([1.5, 2.5], "cpu") == [[1.5, 2.5]]
+
+
+class TestTokenizer:
+    def test_deterministic_and_shaped(self):
+        a = tokenizer.encode("A herd of cows")
+        b = tokenizer.encode("A herd of cows")
+        assert a == b
+        assert len(a) == 77
+        assert a[0] == tokenizer.BOS
+        assert tokenizer.EOS in a
+
+    def test_batch(self):
+        t = tokenizer.encode_batch(["one", "two words here"])
+        assert t.shape == (2, 77)
+        assert t.dtype == torch.long
+
+    def test_ids_in_vocab(self):
+        ids = tokenizer.encode("punctuation, too! and-hyphens 123")
+        assert all(0 <= i < tokenizer.VOCAB_SIZE for i in ids)
+
+
+class TestEtaDefaults:
+    def test_unknown_sampler_cost(self):
+        from sdwd_amd.core import sampler_cost
+
+        assert sampler_cost("Some Future Sampler") == 1.0
+        assert sampler_cost("Heun") == 2.0
