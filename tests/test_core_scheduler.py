@@ -361,3 +361,53 @@ class TestSyncScript:
 
         rc, _ = run_sync_script(str(tmp_path))
         assert rc == 127
+
+
+class TestOptimizerEdges:
+    def test_complementary_with_pixel_caps(self):
+        """A deferred slow rank with a pixel cap can't exceed its cap even
+        for bonus images."""
+        world = make_world([60.0, 60.0, 0.6], job_timeout=1.0)
+        world.get_worker("gpu2").pixel_cap = 512 * 512  # 1 image max
+        jobs = world.make_jobs(GenRequest(batch_size=8))
+        comp = [j for j in jobs if j.complementary]
+        for j in comp:
+            assert j.batch_size * 512 * 512 <= 10 * 512 * 512
+
+    def test_single_worker_weighted_noop(self):
+        world = make_world([42.0])
+        jobs = world.make_jobs(GenRequest(batch_size=5))
+        assert len(jobs) == 1 and jobs[0].batch_size == 5
+
+    def test_zero_stall_on_identical_ranks(self):
+        """Homogeneous ranks: predicted ETAs are equal across shards (the
+        heterogeneity machinery must not skew a uniform node)."""
+        world = make_world([30.0] * 8)
+        jobs = world.make_jobs(GenRequest(batch_size=64))
+        etas = {j.predicted_eta for j in jobs}
+        assert len(etas) == 1
+
+    def test_seed_plan_with_complementary_appended(self):
+        world = make_world([60.0, 60.0, 0.6], job_timeout=1.0)
+        req = GenRequest(batch_size=8, seed=100)
+        jobs = world.make_jobs(req)
+        real = sorted(
+            (j for j in jobs if not j.complementary),
+            key=lambda j: j.gallery_offset,
+        )
+        comp = [j for j in jobs if j.complementary]
+        # requested batch owns slots [0, 8); bonus images append after
+        seeds = [s for j in real for s in j.seeds]
+        assert seeds == list(range(100, 108))
+        for j in comp:
+            assert j.gallery_offset >= 8
+
+    def test_hr_raises_predicted_eta(self):
+        world = make_world([60.0, 60.0])
+        a = world.make_jobs(GenRequest(batch_size=4))
+        b = world.make_jobs(
+            GenRequest(batch_size=4, hr_scale=2.0, hr_steps=10)
+        )
+        assert max(j.predicted_eta for j in b) > max(
+            j.predicted_eta for j in a
+        )
