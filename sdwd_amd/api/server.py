@@ -130,8 +130,12 @@ def create_app(engine: Optional[LocalEngine] = None,
                 result = engine.generate(gen)
             finally:
                 state.busy = False
-        images = [_b64_png(result.images[i])
-                  for i in range(result.images.shape[0])]
+        images = [
+            base64.b64encode(
+                encode_png(result.images[i], result.infotexts[i])
+            ).decode()
+            for i in range(result.images.shape[0])
+        ]
         if result.grid is not None:
             images.insert(0, _b64_png(result.grid))
         info = {

@@ -27,8 +27,12 @@ def make_grid(images: torch.Tensor, rows: Optional[int] = None) -> torch.Tensor:
     return grid
 
 
-def encode_png(image: torch.Tensor) -> bytes:
-    """[H,W,3] uint8 -> PNG bytes (filter 0, zlib level 6)."""
+def encode_png(image: torch.Tensor, parameters: Optional[str] = None) -> bytes:
+    """[H,W,3] uint8 -> PNG bytes (filter 0, zlib level 6).
+
+    ``parameters`` is embedded as a tEXt chunk under the "parameters" key —
+    the sdwui infotext convention, so downstream tools can read the
+    prompt/seed/sampler back out of the file."""
     h, w, c = image.shape
     assert c == 3 and image.dtype == torch.uint8
     raw = image.contiguous().numpy().tobytes()
@@ -46,17 +50,41 @@ def encode_png(image: torch.Tensor) -> bytes:
         )
 
     ihdr = struct.pack(">IIBBBBB", w, h, 8, 2, 0, 0, 0)
+    text = b""
+    if parameters:
+        payload = b"parameters\x00" + parameters.encode("latin-1", "replace")
+        text = chunk(b"tEXt", payload)
     return (
         b"\x89PNG\r\n\x1a\n"
         + chunk(b"IHDR", ihdr)
+        + text
         + chunk(b"IDAT", zlib.compress(scanlines, 6))
         + chunk(b"IEND", b"")
     )
 
 
-def save_png(image: torch.Tensor, path: str) -> str:
+def png_parameters(data: bytes) -> Optional[str]:
+    """Read back the tEXt "parameters" infotext, if present."""
+    pos = 8
+    while pos + 8 <= len(data):
+        (length,) = struct.unpack(">I", data[pos : pos + 4])
+        tag = data[pos + 4 : pos + 8]
+        if tag == b"tEXt":
+            payload = data[pos + 8 : pos + 8 + length]
+            key, _, value = payload.partition(b"\x00")
+            if key == b"parameters":
+                return value.decode("latin-1")
+        if tag == b"IEND":
+            break
+        pos += 12 + length
+    return None
+
+
+def save_png(
+    image: torch.Tensor, path: str, parameters: Optional[str] = None
+) -> str:
     with open(path, "wb") as fh:
-        fh.write(encode_png(image))
+        fh.write(encode_png(image, parameters))
     return path
 
 

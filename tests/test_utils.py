@@ -80,3 +80,23 @@ class TestEtaDefaults:
 
         assert sampler_cost("Some Future Sampler") == 1.0
         assert sampler_cost("Heun") == 2.0
+
+
+class TestPngMetadata:
+    def test_parameters_round_trip(self):
+        from sdwd_amd.utils.images import encode_png, png_parameters
+
+        img = torch.randint(0, 255, (8, 8, 3), dtype=torch.uint8)
+        info = "a cow\nSteps: 20, Sampler: Euler a, Seed: 7"
+        data = encode_png(img, parameters=info)
+        assert png_parameters(data) == info
+        # image still decodes
+        from sdwd_amd.utils.images import decode_png
+
+        assert torch.equal(decode_png(data), img)
+
+    def test_no_metadata_is_none(self):
+        from sdwd_amd.utils.images import encode_png, png_parameters
+
+        img = torch.zeros(4, 4, 3, dtype=torch.uint8)
+        assert png_parameters(encode_png(img)) is None

@@ -72,7 +72,7 @@ def main() -> int:
     os.makedirs(args.out, exist_ok=True)
     for i in range(res.images.shape[0]):
         path = os.path.join(args.out, f"{res.seeds[i]}_{i:03d}.png")
-        save_png(res.images[i], path)
+        save_png(res.images[i], path, parameters=res.infotexts[i])
         print(f"saved {path}")
     if res.grid is not None:
         save_png(res.grid, os.path.join(args.out, "grid.png"))
