@@ -250,3 +250,41 @@ class TestEdgeCases:
         res = eng.generate(req)
         timer.cancel()
         assert res.images.shape == (2, 64, 64, 3)
+
+
+def _bench_worker(rank, world_size, port, tmpdir):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from sdwd_amd.parallel import DistributedEngine, destroy_group
+
+    eng = DistributedEngine(model="tiny", backend="gloo")
+    eng.world.benchmark_payload.width = 64
+    eng.world.benchmark_payload.height = 64
+    eng.world.benchmark_payload.steps = 1
+    speeds = eng.benchmark(rebenchmark=True)  # each rank times itself
+    assert all(v > 0 for v in speeds.values()), speeds
+    assert len(speeds) == world_size
+    if rank == 0:
+        with open(os.path.join(tmpdir, "bench_ok"), "w") as fh:
+            fh.write(str(speeds))
+    destroy_group()
+
+
+@pytest.mark.timeout(300)
+class TestDistributedBenchmark:
+    def test_allgathered_speeds(self, tmp_path):
+        import socket
+
+        import torch.multiprocessing as mp
+
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        mp.start_processes(
+            _bench_worker, args=(2, port, str(tmp_path)), nprocs=2,
+            start_method="spawn", join=True,
+        )
+        assert (tmp_path / "bench_ok").exists()
