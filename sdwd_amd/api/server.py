@@ -356,12 +356,36 @@ def create_app(engine: Optional[LocalEngine] = None,
     return app
 
 
+def install_signal_handlers(engine: LocalEngine) -> None:
+    """Save config then chain the previous handler on SIGINT/SIGTERM
+    (ref distributed.py:359-375)."""
+    import signal
+
+    previous = {}
+
+    def handler(signum, frame):
+        try:
+            if engine.world.config_path:
+                engine.world.save()
+                log.info("config saved on signal %d", signum)
+        finally:
+            prev = previous.get(signum)
+            if callable(prev):
+                prev(signum, frame)
+            else:
+                raise SystemExit(0)
+
+    for sig in (signal.SIGINT, signal.SIGTERM):
+        previous[sig] = signal.getsignal(sig)
+        signal.signal(sig, handler)
+
+
 def main() -> None:  # pragma: no cover - manual entry
     import argparse
 
     import uvicorn
 
-    from ..config import add_flags
+    from ..config import add_flags, default_config_path, export_env
 
     ap = argparse.ArgumentParser()
     ap.add_argument("--host", default="127.0.0.1")
@@ -369,7 +393,18 @@ def main() -> None:  # pragma: no cover - manual entry
     ap.add_argument("--model", default="sd15")
     add_flags(ap)
     args = ap.parse_args()
-    uvicorn.run(create_app(model=args.model), host=args.host, port=args.port)
+    export_env(args)
+    engine = LocalEngine(model=args.model)
+    engine.world.config_path = args.sdwd_config or default_config_path()
+    engine.world.load(engine.world.config_path)  # persisted speeds/settings
+    if not engine.world.workers:  # fresh config: build from devices
+        from ..core import World
+
+        engine.world = World.from_devices(
+            len(engine.devices), config_path=engine.world.config_path
+        )
+    install_signal_handlers(engine)
+    uvicorn.run(create_app(engine=engine), host=args.host, port=args.port)
 
 
 if __name__ == "__main__":  # pragma: no cover

@@ -323,7 +323,21 @@ class LocalEngine(_EngineBase):
         result = self._assemble(
             gen, jobs, shards, infos, time.perf_counter() - t0, interrupted
         )
+        self._autosave()
         return result
+
+    def _autosave(self) -> None:
+        """Persist worker speeds/MPE after each run (ref distributed.py:357,
+        gated like --distributed-remotes-autosave)."""
+        import os as _os
+
+        if self.world.config_path and _os.environ.get(
+            "SDWD_AUTOSAVE", "1"
+        ) not in ("", "0", "false"):
+            try:
+                self.world.save()
+            except OSError as exc:
+                log.warning("config autosave failed: %s", exc)
 
     def interrupt(self) -> None:
         self.world.interrupt_all()

@@ -288,3 +288,29 @@ class TestDistributedBenchmark:
             start_method="spawn", join=True,
         )
         assert (tmp_path / "bench_ok").exists()
+
+
+class TestAutosave:
+    def test_config_saved_after_run(self, tmp_path, monkeypatch):
+        monkeypatch.setenv("SDWD_AUTOSAVE", "1")
+        eng = make_engine(2)
+        eng.world.config_path = str(tmp_path / "auto.json")
+        eng.generate(
+            GenerationRequest(prompt="a", batch_size=2, width=64, height=64,
+                              steps=1, seed=1)
+        )
+        assert (tmp_path / "auto.json").exists()
+        import json
+
+        cfg = json.loads((tmp_path / "auto.json").read_text())
+        assert len(cfg["workers"]) == 2
+
+    def test_autosave_disabled(self, tmp_path, monkeypatch):
+        monkeypatch.setenv("SDWD_AUTOSAVE", "0")
+        eng = make_engine(1)
+        eng.world.config_path = str(tmp_path / "no.json")
+        eng.generate(
+            GenerationRequest(prompt="a", batch_size=1, width=64, height=64,
+                              steps=1, seed=1)
+        )
+        assert not (tmp_path / "no.json").exists()
