@@ -193,12 +193,20 @@ def create_app(engine: Optional[LocalEngine] = None,
     def img2img(req: Img2ImgRequest):
         if not req.init_images:
             raise HTTPException(422, "init_images required")
-        inits = torch.stack([_decode_b64_png(d) for d in req.init_images])
+        try:
+            inits = torch.stack(
+                [_decode_b64_png(d) for d in req.init_images]
+            )
+        except Exception as exc:
+            raise HTTPException(422, f"bad init image: {exc}")
         mask_image = None
         if req.mask:
-            mask_image = _decode_b64_png(req.mask).float().mean(-1).to(
-                torch.uint8
-            )
+            try:
+                mask_image = _decode_b64_png(req.mask).float().mean(-1).to(
+                    torch.uint8
+                )
+            except Exception as exc:
+                raise HTTPException(422, f"bad mask image: {exc}")
         gen = GenerationRequest(
             prompt=req.prompt,
             negative_prompt=req.negative_prompt,

@@ -395,6 +395,19 @@ class DistributedEngine(_EngineBase):
             except Exception:
                 pass
 
+    def rank_progress(self) -> Dict[int, str]:
+        """Last published 'step/total' per rank (torchrun status surface)."""
+        out: Dict[int, str] = {}
+        if self._store is None:
+            return out
+        for r in range(self.world_size):
+            try:
+                if self._store.check([f"sdwd_prog_{r}"]):
+                    out[r] = self._store.get(f"sdwd_prog_{r}").decode()
+            except Exception:
+                continue
+        return out
+
     def heartbeats(self) -> Dict[int, float]:
         """age (seconds) of each rank's last heartbeat; missing = never."""
         out: Dict[int, float] = {}
@@ -507,9 +520,28 @@ class DistributedEngine(_EngineBase):
                         src[idx], seeds=mine.seeds
                     )
                 ts = time.perf_counter()
+
+                last_pub = [0.0]
+
+                def on_step(i, n):
+                    # publish rank progress out-of-band, throttled to the
+                    # interrupt-poll cadence (C19 status for torchrun mode)
+                    now = time.monotonic()
+                    if self._store is not None and (
+                        now - last_pub[0] > 0.5 or i == n
+                    ):
+                        last_pub[0] = now
+                        try:
+                            self._store.set(
+                                f"sdwd_prog_{self.rank}", f"{i}/{n}"
+                            )
+                        except Exception:
+                            pass
+
                 res = self.pipe.generate(
                     _job_pipeline_request(gen, mine, init_latents),
                     interrupt=self._interrupted,
+                    step_callback=on_step,
                 )
                 my_elapsed = time.perf_counter() - ts
                 my_images = res.images

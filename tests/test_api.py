@@ -168,3 +168,13 @@ class TestRestart:
                   "height": 64, "seed": 1},
         )
         assert r.status_code == 200
+
+
+class TestBadInput:
+    def test_malformed_init_image_422(self, client):
+        r = client.post(
+            "/sdapi/v1/img2img",
+            json={"prompt": "x", "init_images": ["bm90YXBuZw=="],
+                  "steps": 1, "width": 64, "height": 64},
+        )
+        assert r.status_code == 422

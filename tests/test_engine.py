@@ -190,6 +190,8 @@ def _hb_worker(rank, world_size, port, tmpdir):
         hb = eng.heartbeats()
         assert 0 in hb and 1 in hb, f"missing heartbeats: {hb}"
         assert all(age < 120 for age in hb.values())
+        prog = eng.rank_progress()
+        assert prog.get(0, "").endswith("/1"), prog
         with open(os.path.join(tmpdir, "hb_ok"), "w") as fh:
             fh.write("ok")
     destroy_group()
