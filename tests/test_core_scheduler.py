@@ -411,3 +411,23 @@ class TestOptimizerEdges:
         assert max(j.predicted_eta for j in b) > max(
             j.predicted_eta for j in a
         )
+
+
+class TestWorldSurfaces:
+    def test_default_batch_size(self):
+        world = make_world([30.0, 30.0, 30.0])
+        assert world.default_batch_size() == 3
+        world.get_worker("gpu2").set_state(State.DISABLED, strict=False)
+        assert world.default_batch_size() == 2
+
+    def test_speed_summary_format(self):
+        world = make_world([30.0, 60.0])
+        txt = world.speed_summary()
+        assert "gpu0" in txt and "gpu1" in txt
+        assert "total: 90.00 ipm" in txt
+
+    def test_worker_memory_without_gpu(self):
+        w = Worker(label="x", device=0)
+        free, total = w.memory()
+        assert (free, total) == (0, 0)
+        assert not w.reachable()
