@@ -74,7 +74,10 @@ class Sampler:
         for i in range(len(ts)):
             if interrupt is not None and interrupt():
                 break
-            x = self.step(model_fn, x, sig[i], sig[i + 1], ts[i], noise_fn)
+            t_next = ts[i + 1] if i + 1 < len(ts) else ts[i]
+            x = self.step(
+                model_fn, x, sig[i], sig[i + 1], ts[i], noise_fn, t_next
+            )
             if post_step is not None:
                 x = post_step(x, sig[i + 1])
             if callback is not None:
@@ -84,18 +87,32 @@ class Sampler:
     def reset(self) -> None:
         pass
 
-    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn):
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
         raise NotImplementedError
+
+    @staticmethod
+    def _t_for(sigma_mid, sigma, sigma_next, t, t_next):
+        """Interpolate the train-timestep for an intermediate sigma
+        (log-sigma linear, matching the schedule construction)."""
+        if sigma_next <= 0 or t_next is None or t_next == t:
+            return t
+        w = (math.log(sigma_mid) - math.log(sigma)) / (
+            math.log(sigma_next) - math.log(sigma)
+        )
+        return t + w * (t_next - t)
 
 
 class Euler(Sampler):
-    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn):
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
         denoised = _eval(model_fn, x, sigma, t)
         return ops.euler_step(x, denoised, sigma, sigma_next)
 
 
 class EulerAncestral(Sampler):
-    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn):
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
         denoised = _eval(model_fn, x, sigma, t)
         sd, su = _ancestral_sigmas(sigma, sigma_next)
         x = ops.euler_step(x, denoised, sigma, sd)
@@ -107,7 +124,8 @@ class EulerAncestral(Sampler):
 class Heun(Sampler):
     order = 2
 
-    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn):
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
         denoised = _eval(model_fn, x, sigma, t)
         if sigma_next <= 0:
             return ops.euler_step(x, denoised, sigma, sigma_next)
@@ -124,7 +142,8 @@ class DPMpp2M(Sampler):
         self.old_denoised = None
         self.h_last = None
 
-    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn):
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
         denoised = _eval(model_fn, x, sigma, t)
         tt = -math.log(sigma)
         if sigma_next <= 0:
@@ -150,19 +169,92 @@ class DPMppSDE(Sampler):
 
     order = 2
 
-    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn):
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
         denoised = _eval(model_fn, x, sigma, t)
         if sigma_next <= 0:
             return denoised
         # midpoint in log-sigma
         sigma_mid = math.exp((math.log(sigma) + math.log(sigma_next)) / 2)
         x_mid = ops.euler_step(x, denoised, sigma, sigma_mid)
-        denoised2 = _eval(model_fn, x_mid, sigma_mid, t)
+        t_mid = self._t_for(sigma_mid, sigma, sigma_next, t, t_next)
+        denoised2 = _eval(model_fn, x_mid, sigma_mid, t_mid)
         sd, su = _ancestral_sigmas(sigma, sigma_next)
         x = ops.euler_step(x, denoised2, sigma, sd)
         if su > 0 and noise_fn is not None:
             x = ops.add_noise(x, noise_fn(), 1.0, su)
         return x
+
+
+class DPM2(Sampler):
+    """k-diffusion sample_dpm_2: explicit midpoint in log-sigma."""
+
+    order = 2
+    eta = 0.0
+
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
+        denoised = _eval(model_fn, x, sigma, t)
+        sd, su = _ancestral_sigmas(sigma, sigma_next, self.eta)
+        if sd <= 0:
+            return ops.euler_step(x, denoised, sigma, sigma_next)
+        sigma_mid = math.exp((math.log(sigma) + math.log(sd)) / 2)
+        x_mid = ops.euler_step(x, denoised, sigma, sigma_mid)
+        t_mid = self._t_for(sigma_mid, sigma, sigma_next, t, t_next)
+        denoised2 = _eval(model_fn, x_mid, sigma_mid, t_mid)
+        # full step from sigma with the midpoint derivative:
+        # x + d2 * (sd - sigma), d2 = (x_mid - denoised2) / sigma_mid
+        d2 = ops.lincomb(x_mid, denoised2, 1.0 / sigma_mid, -1.0 / sigma_mid)
+        x = ops.lincomb(x, d2, 1.0, sd - sigma)
+        if su > 0 and noise_fn is not None:
+            x = ops.add_noise(x, noise_fn(), 1.0, su)
+        return x
+
+
+class DPM2Ancestral(DPM2):
+    eta = 1.0
+
+
+class LMS(Sampler):
+    """Linear multistep (order <= 4) with exactly integrated Adams
+    coefficients over each sigma interval (k-diffusion sample_lms)."""
+
+    max_order = 4
+
+    def reset(self):
+        self.ds: List[torch.Tensor] = []
+        self.step_idx = 0
+
+    def _coeff(self, order, j, sigmas, i):
+        from scipy import integrate
+
+        def fn(tau):
+            prod = 1.0
+            for k in range(order):
+                if j == k:
+                    continue
+                prod *= (tau - sigmas[i - k]) / (sigmas[i - j] - sigmas[i - k])
+            return prod
+
+        val, _ = integrate.quad(fn, sigmas[i], sigmas[i + 1], epsrel=1e-4)
+        return val
+
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
+        denoised = _eval(model_fn, x, sigma, t)
+        d = ops.lincomb(x, denoised, 1.0 / sigma, -1.0 / sigma)
+        self.ds.append(d)
+        if len(self.ds) > self.max_order:
+            self.ds.pop(0)
+        sigmas = self.schedule.sigmas.tolist()
+        i = self.step_idx
+        order = min(i + 1, self.max_order)
+        coeffs = [self._coeff(order, j, sigmas, i) for j in range(order)]
+        out = x.float()
+        for c, dj in zip(coeffs, reversed(self.ds)):
+            out = out + c * dj.float()
+        self.step_idx += 1
+        return out.to(x.dtype)
 
 
 SAMPLERS: Dict[str, type] = {
@@ -174,7 +266,11 @@ SAMPLERS: Dict[str, type] = {
     "DPM++ 2M Karras": DPMpp2M,
     "DPM++ SDE": DPMppSDE,
     "DPM++ SDE Karras": DPMppSDE,
-    "LMS": Euler,
+    "LMS": LMS,
+    "LMS Karras": LMS,
+    "DPM2": DPM2,
+    "DPM2 a": DPM2Ancestral,
+    "DDPM": EulerAncestral,
 }
 
 

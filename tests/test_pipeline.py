@@ -274,7 +274,8 @@ class TestSamplerOracle:
     validates each update rule's algebra end-to-end."""
 
     @pytest.mark.parametrize("name", ["Euler", "Euler a", "DDIM", "Heun",
-                                      "DPM++ 2M", "DPM++ SDE"])
+                                      "DPM++ 2M", "DPM++ SDE", "LMS",
+                                      "DPM2", "DPM2 a", "DDPM"])
     def test_converges_to_point_mass(self, name):
         from sdwd_amd.pipeline.samplers import build_sampler
         from sdwd_amd.pipeline.schedule import discrete_schedule, make_sigmas_full
