@@ -409,3 +409,20 @@ class TestDeterminismGPU:
         a = pipe.generate(req).images
         b = pipe.generate(req).images
         assert torch.equal(a, b), "GPU pipeline must be run-to-run exact"
+
+
+class TestAllSamplersGPU:
+    def test_every_sampler_runs_on_gpu(self, dev):
+        from sdwd_amd.pipeline import (
+            PipelineRequest,
+            StableDiffusionPipeline,
+            sampler_names,
+        )
+
+        pipe = StableDiffusionPipeline("tiny", device=dev)
+        for name in sampler_names():
+            res = pipe.generate(
+                PipelineRequest(prompt="s", steps=3, width=64, height=64,
+                                seeds=[1], sampler_name=name)
+            )
+            assert torch.isfinite(res.images.float()).all(), name
