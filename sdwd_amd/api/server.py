@@ -43,6 +43,7 @@ class Txt2ImgRequest(BaseModel):
     n_iter: int = 1
     sampler_name: str = "Euler a"
     sampler_index: Optional[str] = None  # legacy alias
+    clip_skip: int = 1
     # hires fix (sdwui fields)
     enable_hr: bool = False
     hr_scale: float = 2.0
@@ -183,6 +184,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             hr_scale=req.hr_scale,
             hr_steps=req.hr_second_pass_steps,
             denoising_strength=req.denoising_strength,
+            clip_skip=req.clip_skip,
             control_image=control_image,
             control_model=control_model,
             control_scale=control_scale,

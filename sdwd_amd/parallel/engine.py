@@ -55,6 +55,7 @@ class GenerationRequest:
     control_image: Optional[torch.Tensor] = None
     control_model: str = ""
     control_scale: float = 1.0
+    clip_skip: int = 1
 
     def sched(self) -> GenRequest:
         return GenRequest(
@@ -105,6 +106,7 @@ def _job_pipeline_request(
         control_image=gen.control_image,
         control_model=gen.control_model,
         control_scale=gen.control_scale,
+        clip_skip=gen.clip_skip,
     )
 
 

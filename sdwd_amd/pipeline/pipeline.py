@@ -52,6 +52,7 @@ class PipelineRequest:
     control_image: Optional[torch.Tensor] = None  # [B,H,W,3] uint8
     control_model: str = ""  # e.g. "controlnet-sd15"; "" = off
     control_scale: float = 1.0
+    clip_skip: int = 1  # 1 = final layer; 2 = penultimate (sdwui setting)
 
     @property
     def batch_size(self) -> int:
@@ -139,10 +140,13 @@ class StableDiffusionPipeline:
     # -- conditioning --------------------------------------------------------
     @torch.no_grad()
     def encode_prompts(
-        self, prompts: List[str], negatives: List[str]
+        self, prompts: List[str], negatives: List[str], clip_skip: int = 1
     ) -> tuple:
-        tokens = tokenizer.encode_batch(prompts + negatives, device=self.device)
+        tokens, weights = tokenizer.encode_batch_weighted(
+            prompts + negatives, device=self.device
+        )
         m = self.model
+        penult = clip_skip >= 2
         if m.is_sdxl:
             h1 = m.text_encoder(tokens, penultimate=True)
             h2 = m.text_encoder_2(tokens, penultimate=True)
@@ -151,8 +155,15 @@ class StableDiffusionPipeline:
                 tokens, m.text_encoder_2(tokens)
             )
         else:
-            ctx = m.text_encoder(tokens)
+            ctx = m.text_encoder(tokens, penultimate=penult)
             pooled = None
+        # prompt-attention weights (sdwui semantics): scale the hidden
+        # states of weighted tokens, then restore the original mean
+        if (weights != 1.0).any():
+            orig_mean = ctx.float().mean(dim=(1, 2), keepdim=True)
+            ctx = ctx * weights[:, :, None].to(ctx.dtype)
+            new_mean = ctx.float().mean(dim=(1, 2), keepdim=True)
+            ctx = ctx * (orig_mean / new_mean.clamp_min(1e-9)).to(ctx.dtype)
         n = len(prompts)
         cond, uncond = ctx[:n], ctx[n:]
         pooled_cu = (pooled[:n], pooled[n:]) if pooled is not None else None
@@ -202,7 +213,7 @@ class StableDiffusionPipeline:
         subseeds = req.subseeds or [-1] * b
 
         cond, uncond, pooled_cu = self.encode_prompts(
-            [req.prompt] * b, [req.negative_prompt] * b
+            [req.prompt] * b, [req.negative_prompt] * b, req.clip_skip
         )
         y = None
         if self.model.is_sdxl and pooled_cu is not None:

@@ -428,3 +428,42 @@ class TestInpainting:
         a = pipe.generate(req).images
         b = pipe.generate(req).images
         assert torch.equal(a, b)
+
+
+class TestPromptWeighting:
+    def test_parse_weighted(self):
+        from sdwd_amd.models.tokenizer import parse_weighted
+
+        assert parse_weighted("a (big:1.4) cow") == [
+            ("a ", 1.0), ("big", 1.4), (" cow", 1.0)
+        ]
+        fr = parse_weighted("((double)) and [down]")
+        weights = {f.strip(): w for f, w in fr}
+        assert abs(weights["double"] - 1.21) < 1e-9
+        assert abs(weights["down"] - 1 / 1.1) < 1e-9
+
+    def test_weight_changes_output(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(steps=2, width=64, height=64, seeds=[2])
+        a = pipe.generate(PipelineRequest(prompt="a big cow", **base)).images
+        b = pipe.generate(
+            PipelineRequest(prompt="a (big:1.5) cow", **base)
+        ).images
+        assert not torch.equal(a, b)
+
+    def test_plain_prompt_unaffected(self, pipe):
+        """Weight machinery must be a no-op for unweighted prompts."""
+        from sdwd_amd.models.tokenizer import encode, encode_weighted
+
+        ids, wts = encode_weighted("a herd of cows")
+        assert ids == encode("a herd of cows")
+        assert all(w == 1.0 for w in wts)
+
+    def test_clip_skip(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="skip", steps=2, width=64, height=64, seeds=[2])
+        a = pipe.generate(PipelineRequest(**base, clip_skip=1)).images
+        b = pipe.generate(PipelineRequest(**base, clip_skip=2)).images
+        assert not torch.equal(a, b)
