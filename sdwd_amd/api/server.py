@@ -39,8 +39,8 @@ class Txt2ImgRequest(BaseModel):
     cfg_scale: float = 7.0
     width: int = 512
     height: int = 512
-    batch_size: int = 1
-    n_iter: int = 1
+    batch_size: int = Field(default=1, ge=1, le=4096)
+    n_iter: int = Field(default=1, ge=1, le=64)
     sampler_name: str = "Euler a"
     sampler_index: Optional[str] = None  # legacy alias
     clip_skip: int = 1

@@ -212,9 +212,18 @@ class StableDiffusionPipeline:
         lat_c = self.model.latent_channels
         subseeds = req.subseeds or [-1] * b
 
-        cond, uncond, pooled_cu = self.encode_prompts(
-            [req.prompt] * b, [req.negative_prompt] * b, req.clip_skip
+        # one prompt per request: encode once, broadcast across the batch
+        cond1, uncond1, pooled1 = self.encode_prompts(
+            [req.prompt], [req.negative_prompt], req.clip_skip
         )
+        cond = cond1.expand(b, -1, -1)
+        uncond = uncond1.expand(b, -1, -1)
+        pooled_cu = None
+        if pooled1 is not None:
+            pooled_cu = (
+                pooled1[0].expand(b, -1),
+                pooled1[1].expand(b, -1),
+            )
         y = None
         if self.model.is_sdxl and pooled_cu is not None:
             y = torch.cat(
