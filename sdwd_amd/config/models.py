@@ -54,6 +54,10 @@ class SettingsModel(BaseModel):
     complement_production: bool = True  # slow ranks produce bonus images
     step_scaling: bool = False  # scale complementary jobs' steps down to fit
     thin_client: bool = False  # rank 0 only orchestrates, takes no shard
+    # per-task distribution toggles (ref CHANGELOG 2.3.0: separate txt2img /
+    # img2img enable states); disabled -> the whole batch runs on rank 0
+    distribute_txt2img: bool = True
+    distribute_img2img: bool = True
 
 
 class ConfigModel(BaseModel):

@@ -43,6 +43,7 @@ log = get_logger("world")
 class GenRequest:
     """The slice of a generation request the scheduler cares about."""
 
+    task: str = "txt2img"  # or "img2img" (per-task distribution toggles)
     batch_size: int = 1
     width: int = 512
     height: int = 512
@@ -118,6 +119,18 @@ class World:
         if not workers:
             raise RuntimeError("no available ranks")
         total = int(request.batch_size)
+
+        distribute = (
+            self.settings.distribute_img2img
+            if request.task == "img2img"
+            else self.settings.distribute_txt2img
+        )
+        if not distribute or len(workers) == 1:
+            # single-rank mode: the whole batch on the first available rank
+            job = Job(worker_label=workers[0].label, batch_size=total)
+            job.predicted_eta = self._predict(workers[0], total, request)
+            self._log_distribution([job], request)
+            return [job]
 
         benchmarked = all(w.eta.avg_ipm > 0 for w in workers)
         if not benchmarked or len(workers) == 1:

@@ -59,6 +59,7 @@ class GenerationRequest:
 
     def sched(self) -> GenRequest:
         return GenRequest(
+            task="img2img" if self.init_images is not None else "txt2img",
             batch_size=self.batch_size,
             width=self.width,
             height=self.height,

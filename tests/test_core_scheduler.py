@@ -431,3 +431,19 @@ class TestWorldSurfaces:
         free, total = w.memory()
         assert (free, total) == (0, 0)
         assert not w.reachable()
+
+
+class TestPerTaskToggles:
+    def test_txt2img_disabled_runs_on_one_rank(self):
+        world = make_world([30.0, 30.0, 30.0], distribute_txt2img=False)
+        jobs = world.make_jobs(GenRequest(batch_size=6))
+        assert len(jobs) == 1 and jobs[0].batch_size == 6
+
+    def test_img2img_toggle_independent(self):
+        world = make_world(
+            [30.0, 30.0], distribute_txt2img=False, distribute_img2img=True
+        )
+        t2i = world.make_jobs(GenRequest(batch_size=4, task="txt2img"))
+        i2i = world.make_jobs(GenRequest(batch_size=4, task="img2img"))
+        assert len(t2i) == 1
+        assert len(i2i) == 2
