@@ -19,6 +19,7 @@ import torch
 from fastapi import FastAPI, HTTPException
 from pydantic import BaseModel, Field
 
+from ..config.models import SettingsModel
 from ..core import State
 from ..models.registry import available_models
 from ..parallel import GenerationRequest, LocalEngine
@@ -346,6 +347,28 @@ def create_app(engine: Optional[LocalEngine] = None,
     def benchmark(rebenchmark: bool = True):
         speeds = engine.benchmark(rebenchmark=rebenchmark)
         return {"speeds": speeds}
+
+    @app.get("/sdwd/settings")
+    def get_settings():
+        """Current engine settings (the reference's Settings tab state,
+        ref ui.py:363-391)."""
+        return engine.world.settings.model_dump()
+
+    @app.post("/sdwd/settings")
+    def set_settings(body: Dict[str, Any]):
+        """Update engine settings live and persist them (ref ui.py Settings
+        tab + update_world handlers). Unknown keys are rejected."""
+        current = engine.world.settings.model_dump()
+        unknown = set(body) - set(current)
+        if unknown:
+            raise HTTPException(422, f"unknown settings: {sorted(unknown)}")
+        current.update(body)
+        try:
+            engine.world.settings = SettingsModel.model_validate(current)
+        except Exception as exc:
+            raise HTTPException(422, str(exc))
+        engine.world.save()
+        return engine.world.settings.model_dump()
 
     @app.post("/sdwd/worker/{label}/enable")
     def enable_worker(label: str):

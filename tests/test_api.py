@@ -12,7 +12,19 @@ from sdwd_amd.utils.images import decode_png, encode_png
 
 
 @pytest.fixture(scope="module")
-def client():
+def monkeypatch_module():
+    from _pytest.monkeypatch import MonkeyPatch
+
+    mp = MonkeyPatch()
+    yield mp
+    mp.undo()
+
+
+@pytest.fixture(scope="module")
+def client(tmp_path_factory, monkeypatch_module):
+    monkeypatch_module.setenv(
+        "SDWD_CONFIG", str(tmp_path_factory.mktemp("cfg") / "c.json")
+    )
     engine = LocalEngine(model="tiny", devices=["cpu", "cpu"])
     for w in engine.world.workers:
         w.eta.avg_ipm = 60.0
@@ -94,6 +106,34 @@ class TestControl:
             "/sdapi/v1/options", json={"sd_model_checkpoint": "nope"}
         )
         assert r.status_code == 404
+
+    def test_settings_round_trip(self, client):
+        import json as _json
+        import os
+
+        before = client.get("/sdwd/settings").json()
+        assert before["complement_production"] is True
+        r = client.post(
+            "/sdwd/settings",
+            json={"job_timeout": 7.5, "distribute_img2img": False},
+        )
+        assert r.status_code == 200
+        after = client.get("/sdwd/settings").json()
+        assert after["job_timeout"] == 7.5
+        assert after["distribute_img2img"] is False
+        # persisted to the config file
+        with open(os.environ["SDWD_CONFIG"]) as fh:
+            on_disk = _json.load(fh)
+        assert on_disk["settings"]["job_timeout"] == 7.5
+        # restore for other tests
+        client.post(
+            "/sdwd/settings",
+            json={"job_timeout": 3.0, "distribute_img2img": True},
+        )
+
+    def test_settings_rejects_unknown(self, client):
+        r = client.post("/sdwd/settings", json={"warp_factor": 9})
+        assert r.status_code == 422
 
     def test_status(self, client):
         r = client.get("/sdwd/status").json()
