@@ -1,16 +1,18 @@
-"""Deterministic prompt tokenizer.
+"""Deterministic prompt tokenizer, with optional real CLIP BPE.
 
 There is no network access for the CLIP BPE vocab, and the engine runs
-random-init weights, so tokenization only has to be
-deterministic and well-shaped: lower-cased word pieces are hashed into the
-CLIP vocab range, bracketed by BOS/EOS, padded to 77. A real BPE vocab can
-be dropped in later via ``from_files`` without touching callers.
+random-init weights, so by default lower-cased word pieces
+are hashed into the CLIP vocab range, bracketed by BOS/EOS, padded to 77 —
+deterministic and well-shaped. When a user has the real vocab on disk,
+``use_bpe(vocab.json, merges.txt)`` (or ``BPETokenizer.from_files``) swaps
+in OpenAI-CLIP byte-level BPE without touching callers.
 """
 from __future__ import annotations
 
 import hashlib
+import json
 import re
-from typing import List
+from typing import Dict, List, Optional, Tuple
 
 import torch
 
@@ -27,9 +29,120 @@ def _hash_token(word: str) -> int:
     return h % (VOCAB_SIZE - 2)  # keep clear of BOS/EOS
 
 
+def _bytes_to_unicode() -> Dict[int, str]:
+    """GPT-2/CLIP byte→printable-unicode map (reversible, no control chars)."""
+    bs = (
+        list(range(ord("!"), ord("~") + 1))
+        + list(range(ord("¡"), ord("¬") + 1))
+        + list(range(ord("®"), ord("ÿ") + 1))
+    )
+    cs = bs[:]
+    n = 0
+    for b in range(256):
+        if b not in bs:
+            bs.append(b)
+            cs.append(256 + n)
+            n += 1
+    return dict(zip(bs, [chr(c) for c in cs]))
+
+
+# CLIP's word-split pattern (contractions, letter runs, single digits,
+# punctuation runs) — \p{L}/\p{N} rewritten for the stdlib re module.
+_bpe_word_re = re.compile(
+    r"'s|'t|'re|'ve|'m|'ll|'d|[^\W\d_]+|\d|[^\s\w]+", re.IGNORECASE
+)
+
+
+class BPETokenizer:
+    """OpenAI-CLIP byte-level BPE (lowercase, ``</w>`` end-of-word marker).
+
+    Symbols missing from the vocab fall back to the hash tokenizer so a
+    truncated vocab degrades gracefully instead of raising.
+    """
+
+    def __init__(self, vocab: Dict[str, int], merges: List[Tuple[str, str]]):
+        self.vocab = vocab
+        self.ranks = {pair: i for i, pair in enumerate(merges)}
+        self.byte_enc = _bytes_to_unicode()
+        self._cache: Dict[str, List[str]] = {}
+
+    @classmethod
+    def from_files(cls, vocab_json: str, merges_txt: str) -> "BPETokenizer":
+        with open(vocab_json, "r", encoding="utf-8") as fh:
+            vocab = json.load(fh)
+        merges: List[Tuple[str, str]] = []
+        with open(merges_txt, "r", encoding="utf-8") as fh:
+            for line in fh:
+                line = line.strip()
+                if not line or line.startswith("#"):
+                    continue
+                a, _, b = line.partition(" ")
+                if b:
+                    merges.append((a, b))
+        return cls(vocab, merges)
+
+    def _bpe(self, word: str) -> List[str]:
+        if word in self._cache:
+            return self._cache[word]
+        symbols = list(word)
+        symbols[-1] = symbols[-1] + "</w>"
+        while len(symbols) > 1:
+            pairs = [(symbols[i], symbols[i + 1]) for i in range(len(symbols) - 1)]
+            best = min(pairs, key=lambda p: self.ranks.get(p, 1 << 30))
+            if best not in self.ranks:
+                break
+            merged: List[str] = []
+            i = 0
+            while i < len(symbols):
+                if (
+                    i < len(symbols) - 1
+                    and (symbols[i], symbols[i + 1]) == best
+                ):
+                    merged.append(symbols[i] + symbols[i + 1])
+                    i += 2
+                else:
+                    merged.append(symbols[i])
+                    i += 1
+            symbols = merged
+        self._cache[word] = symbols
+        return symbols
+
+    def encode_text(self, text: str) -> List[int]:
+        """Text -> token ids (no BOS/EOS/padding)."""
+        text = re.sub(r"\s+", " ", text).strip().lower()
+        ids: List[int] = []
+        for word in _bpe_word_re.findall(text):
+            encoded = "".join(self.byte_enc[b] for b in word.encode("utf-8"))
+            for sym in self._bpe(encoded):
+                tid = self.vocab.get(sym)
+                ids.append(tid if tid is not None else _hash_token(sym))
+        return ids
+
+
+_active: Optional[BPETokenizer] = None
+
+
+def use_bpe(vocab_json: str, merges_txt: str) -> BPETokenizer:
+    """Switch the module to real CLIP BPE loaded from disk."""
+    global _active
+    _active = BPETokenizer.from_files(vocab_json, merges_txt)
+    return _active
+
+
+def use_hash() -> None:
+    """Revert to the deterministic hash tokenizer (the default)."""
+    global _active
+    _active = None
+
+
+def _fragment_ids(text: str) -> List[int]:
+    if _active is not None:
+        return _active.encode_text(text)
+    return [_hash_token(w) for w in _word_re.findall(text.lower())]
+
+
 def encode(text: str, max_len: int = MAX_LEN) -> List[int]:
-    words = _word_re.findall(text.lower())
-    ids = [BOS] + [_hash_token(w) for w in words][: max_len - 2] + [EOS]
+    ids = [BOS] + _fragment_ids(text)[: max_len - 2] + [EOS]
     ids += [EOS] * (max_len - len(ids))
     return ids
 
@@ -103,9 +216,9 @@ def encode_weighted(text: str, max_len: int = MAX_LEN):
     ids = [BOS]
     weights = [1.0]
     for frag, w in parse_weighted(text):
-        for word in _word_re.findall(frag.lower()):
+        for tid in _fragment_ids(frag):
             if len(ids) < max_len - 1:
-                ids.append(_hash_token(word))
+                ids.append(tid)
                 weights.append(w)
     ids.append(EOS)
     weights.append(1.0)

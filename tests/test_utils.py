@@ -74,6 +74,77 @@ class TestTokenizer:
         assert all(0 <= i < tokenizer.VOCAB_SIZE for i in ids)
 
 
+class TestBpeTokenizer:
+    """Real CLIP byte-level BPE path (synthetic tiny vocab on disk)."""
+
+    @staticmethod
+    def _write_fixture(tmp_path):
+        import json
+
+        from sdwd_amd.models.tokenizer import _bytes_to_unicode
+
+        # base alphabet: every byte symbol and its </w> form, then the
+        # merged tokens the merges below can produce.
+        syms = list(_bytes_to_unicode().values())
+        vocab = {}
+        for s in syms:
+            vocab[s] = len(vocab)
+        for s in syms:
+            vocab[s + "</w>"] = len(vocab)
+        merges = [("l", "o"), ("lo", "w</w>"), ("c", "o"), ("co", "w</w>")]
+        for a, b in merges:
+            if a + b not in vocab:
+                vocab[a + b] = len(vocab)
+        vpath = tmp_path / "vocab.json"
+        mpath = tmp_path / "merges.txt"
+        vpath.write_text(json.dumps(vocab))
+        mpath.write_text(
+            "#version: test\n" + "\n".join(f"{a} {b}" for a, b in merges)
+        )
+        return str(vpath), str(mpath), vocab
+
+    def test_merges_apply_in_rank_order(self, tmp_path):
+        vpath, mpath, vocab = self._write_fixture(tmp_path)
+        try:
+            bpe = tokenizer.use_bpe(vpath, mpath)
+            # "low" -> l+o merge first (rank 0), then lo+w</w> (rank 1)
+            assert bpe.encode_text("low") == [vocab["low</w>"]]
+            assert bpe.encode_text("cow") == [vocab["cow</w>"]]
+            # unmergeable word falls apart into symbols ending in </w>
+            ids = bpe.encode_text("ab")
+            assert ids == [vocab["a"], vocab["b</w>"]]
+        finally:
+            tokenizer.use_hash()
+
+    def test_module_encode_uses_bpe_when_active(self, tmp_path):
+        vpath, mpath, vocab = self._write_fixture(tmp_path)
+        try:
+            tokenizer.use_bpe(vpath, mpath)
+            ids = tokenizer.encode("low cow")
+            assert ids[0] == tokenizer.BOS
+            assert ids[1] == vocab["low</w>"]
+            assert ids[2] == vocab["cow</w>"]
+            assert ids[3] == tokenizer.EOS
+            assert len(ids) == 77
+            # weighted path shares the BPE fragments
+            wids, wts = tokenizer.encode_weighted("low (cow:1.5)")
+            assert wids[1] == vocab["low</w>"]
+            assert wids[2] == vocab["cow</w>"]
+            assert wts[1] == 1.0 and wts[2] == 1.5
+        finally:
+            tokenizer.use_hash()
+        # reverted: hash path again, still deterministic
+        assert tokenizer.encode("low cow") == tokenizer.encode("low cow")
+
+    def test_whitespace_and_case_normalised(self, tmp_path):
+        vpath, mpath, _ = self._write_fixture(tmp_path)
+        try:
+            bpe = tokenizer.use_bpe(vpath, mpath)
+            assert bpe.encode_text("  LOW\n\tcow ") == bpe.encode_text("low cow")
+        finally:
+            tokenizer.use_hash()
+
+
 class TestEtaDefaults:
     def test_unknown_sampler_cost(self):
         from sdwd_amd.core import sampler_cost
