@@ -215,6 +215,18 @@ def load_checkpoint(path: str, device="cpu", dtype=None) -> ModelBundle:
         meta = f.metadata() or {}
         arch = meta.get("arch", "sd15")
         keys = list(f.keys())
+        if any(k.startswith("model.diffusion_model.") for k in keys):
+            # a real sdwui/ldm checkpoint, not our native format: pick the
+            # arch from the conv_in / cross-attention shapes
+            from .convert import load_ldm_state_dict
+
+            ch = f.get_tensor("model.diffusion_model.input_blocks.0.0.weight"
+                              ).shape[0]
+            arch = {320: "sd15", 32: "tiny"}.get(int(ch), "sd15")
+            bundle = _BUILDERS[arch](arch)
+            load_ldm_state_dict(bundle, {k: f.get_tensor(k) for k in keys})
+            bundle.eval().to(device, dtype)
+            return bundle
         bundle = _BUILDERS[arch.split("/")[0] if arch in _BUILDERS else "sd15"](arch)
         by_prefix: Dict[str, Dict[str, torch.Tensor]] = {}
         for k in keys:

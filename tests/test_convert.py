@@ -1,0 +1,176 @@
+"""ldm/sdwui checkpoint-format conversion tests.
+
+No real checkpoint is available offline, so correctness is pinned two ways:
+round-tripping (export -> load must be bit-exact, including the quant-conv
+folds and CLIP qkv split/fuse), and canary checks that the exported key set
+for the sd15 config is exactly the canonical SD1.5 layout (686 UNet / 248
+VAE / 196 CLIP tensors, spot-checked names and shapes).
+"""
+import pytest
+import torch
+
+from sdwd_amd.models.convert import (
+    CLIP_PREFIX,
+    UNET_PREFIX,
+    VAE_PREFIX,
+    load_ldm_state_dict,
+    to_ldm_state_dict,
+)
+from sdwd_amd.models.registry import load_model
+
+
+def _perturb(bundle):
+    with torch.no_grad():
+        for p in bundle.unet.parameters():
+            p.add_(torch.randn_like(p) * 0.1)
+        for p in bundle.vae.parameters():
+            p.add_(torch.randn_like(p) * 0.1)
+        for p in bundle.text_encoder.parameters():
+            p.add_(torch.randn_like(p) * 0.1)
+
+
+class TestRoundTrip:
+    def test_tiny_bit_exact(self):
+        a = load_model("tiny", device="cpu", cache=False)
+        exported = to_ldm_state_dict(a)
+        b = load_model("tiny", device="cpu", cache=False)
+        _perturb(b)  # make sure the load actually has to do something
+        report = load_ldm_state_dict(b, exported)
+        assert not report["missing"], report["missing"][:5]
+        assert not report["unexpected"], report["unexpected"][:5]
+        for part in ("unet", "vae", "text_encoder"):
+            sa = getattr(a, part).state_dict()
+            sb = getattr(b, part).state_dict()
+            for k in sa:
+                assert torch.equal(sa[k], sb[k]), f"{part}.{k}"
+
+    def test_quant_conv_fold_is_exact(self):
+        """A non-identity quant/post_quant conv folds into conv_out/conv_in
+        so that encode/decode outputs match applying the two convs."""
+        import torch.nn.functional as F
+
+        a = load_model("tiny", device="cpu", cache=False)
+        exported = to_ldm_state_dict(a)
+        lat = 2 * a.vae.cfg.latent_channels
+        g = torch.Generator().manual_seed(5)
+        qw = torch.randn(lat, lat, 1, 1, generator=g) * 0.2 + torch.eye(
+            lat
+        ).reshape(lat, lat, 1, 1)
+        qb = torch.randn(lat, generator=g) * 0.1
+        exported[VAE_PREFIX + "quant_conv.weight"] = qw
+        exported[VAE_PREFIX + "quant_conv.bias"] = qb
+
+        b = load_model("tiny", device="cpu", cache=False)
+        _perturb(b)
+        load_ldm_state_dict(b, exported)
+
+        x = torch.randn(1, 3, 32, 32, generator=g)
+        with torch.no_grad():
+            ref = F.conv2d(a.vae.encoder(x), qw, qb)
+            got = b.vae.encoder(x)
+        assert torch.allclose(ref, got, atol=1e-4), (ref - got).abs().max()
+
+    def test_forward_equivalent(self):
+        a = load_model("tiny", device="cpu", cache=False)
+        b = load_model("tiny", device="cpu", cache=False)
+        _perturb(b)
+        load_ldm_state_dict(b, to_ldm_state_dict(a))
+        x = torch.randn(1, 4, 16, 16)
+        t = torch.tensor([3.0])
+        ctx = torch.randn(1, 77, a.unet.cfg.context_dim)
+        with torch.no_grad():
+            assert torch.equal(a.unet(x, t, ctx), b.unet(x, t, ctx))
+
+
+class TestCheckpointFile:
+    def test_ldm_safetensors_auto_detected(self, tmp_path):
+        """An sdwui-format .safetensors file loads through the normal
+        load_checkpoint entry point."""
+        from safetensors.torch import save_file
+
+        from sdwd_amd.models.registry import load_checkpoint
+
+        a = load_model("tiny", device="cpu", cache=False)
+        exported = {
+            k: v.contiguous().clone() for k, v in to_ldm_state_dict(a).items()
+        }
+        path = str(tmp_path / "sdwui_style.safetensors")
+        save_file(exported, path)
+        b = load_checkpoint(path)
+        x = torch.randn(1, 4, 16, 16)
+        t = torch.tensor([3.0])
+        ctx = torch.randn(1, 77, a.unet.cfg.context_dim)
+        with torch.no_grad():
+            assert torch.equal(a.unet(x, t, ctx), b.unet(x, t, ctx))
+
+
+@pytest.fixture(scope="module")
+def sd15_keys():
+    bundle = load_model("sd15", device="cpu", cache=False)
+    exported = to_ldm_state_dict(bundle)
+    return {k: tuple(v.shape) for k, v in exported.items()}
+
+
+class TestSd15Layout:
+    """Canonical SD1.5 checkpoint layout canaries."""
+
+    def test_tensor_counts(self, sd15_keys):
+        unet = [k for k in sd15_keys if k.startswith(UNET_PREFIX)]
+        vae = [k for k in sd15_keys if k.startswith(VAE_PREFIX)]
+        clip = [k for k in sd15_keys if k.startswith(CLIP_PREFIX)]
+        assert len(unet) == 686
+        assert len(vae) == 248
+        assert len(clip) == 196
+
+    @pytest.mark.parametrize(
+        "key,shape",
+        [
+            ("model.diffusion_model.input_blocks.0.0.weight", (320, 4, 3, 3)),
+            ("model.diffusion_model.input_blocks.1.0.in_layers.2.weight",
+             (320, 320, 3, 3)),
+            ("model.diffusion_model.input_blocks.1.1.transformer_blocks.0."
+             "attn2.to_k.weight", (320, 768)),
+            ("model.diffusion_model.input_blocks.3.0.op.weight",
+             (320, 320, 3, 3)),
+            ("model.diffusion_model.input_blocks.4.0.skip_connection.weight",
+             (640, 320, 1, 1)),
+            ("model.diffusion_model.middle_block.1.proj_in.weight",
+             (1280, 1280, 1, 1)),
+            ("model.diffusion_model.output_blocks.2.1.conv.weight",
+             (1280, 1280, 3, 3)),
+            ("model.diffusion_model.output_blocks.5.2.conv.weight",
+             (1280, 1280, 3, 3)),
+            ("model.diffusion_model.output_blocks.11.0.in_layers.2.weight",
+             (320, 640, 3, 3)),
+            ("model.diffusion_model.out.2.weight", (4, 320, 3, 3)),
+            ("first_stage_model.encoder.down.0.block.0.norm1.weight", (128,)),
+            ("first_stage_model.encoder.down.1.block.0.nin_shortcut.weight",
+             (256, 128, 1, 1)),
+            ("first_stage_model.encoder.mid.attn_1.q.weight",
+             (512, 512, 1, 1)),
+            ("first_stage_model.decoder.up.1.upsample.conv.weight",
+             (256, 256, 3, 3)),
+            ("first_stage_model.quant_conv.weight", (8, 8, 1, 1)),
+            ("first_stage_model.post_quant_conv.weight", (4, 4, 1, 1)),
+            ("cond_stage_model.transformer.text_model.embeddings."
+             "token_embedding.weight", (49408, 768)),
+            ("cond_stage_model.transformer.text_model.encoder.layers.11."
+             "self_attn.q_proj.weight", (768, 768)),
+            ("cond_stage_model.transformer.text_model.encoder.layers.0."
+             "mlp.fc1.weight", (3072, 768)),
+            ("cond_stage_model.transformer.text_model.final_layer_norm.weight",
+             (768,)),
+        ],
+    )
+    def test_canonical_key_shapes(self, sd15_keys, key, shape):
+        assert key in sd15_keys, key
+        assert sd15_keys[key] == shape
+
+    def test_no_upsample_at_level0(self, sd15_keys):
+        assert ("first_stage_model.decoder.up.0.upsample.conv.weight"
+                not in sd15_keys)
+        assert ("model.diffusion_model.output_blocks.11.1.conv.weight"
+                not in sd15_keys)
+        # but level-0 output blocks do carry attention in SD1.5
+        assert ("model.diffusion_model.output_blocks.11.1.transformer_blocks"
+                ".0.attn1.to_q.weight" in sd15_keys)
