@@ -84,12 +84,22 @@ class CLIPTextEncoder(nn.Module):
         self.register_buffer("causal_bias", mask, persistent=False)
 
     def forward(
-        self, tokens: torch.Tensor, penultimate: bool = False
+        self,
+        tokens: torch.Tensor,
+        penultimate: bool = False,
+        clip_skip: int = 1,
     ) -> torch.Tensor:
-        """tokens: [B, 77] int64 -> [B, 77, d_model] conditioning."""
+        """tokens: [B, 77] int64 -> [B, 77, d_model] conditioning.
+
+        ``penultimate``: SDXL conditioning — skip the last block and do NOT
+        apply the final LayerNorm (sgm "penultimate" convention).
+        ``clip_skip``: sdwui CLIP_stop_at_last_layers — skip the last
+        ``clip_skip - 1`` blocks but DO apply the final LayerNorm afterwards
+        (sd_hijack_clip semantics)."""
         x = self.token_emb(tokens) + self.pos_emb
         bias = self.causal_bias.to(x.dtype)
-        n = len(self.blocks) - (1 if penultimate else 0)
+        skip = 1 if penultimate else max(0, clip_skip - 1)
+        n = len(self.blocks) - skip
         for blk in self.blocks[:n]:
             x = blk(x, bias)
         if not penultimate:

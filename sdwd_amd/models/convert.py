@@ -154,7 +154,7 @@ def vae_key_map(vae) -> Dict[str, str]:
         if lvl != levels - 1:
             for p in ("weight", "bias"):
                 out[f"encoder.down.{lvl}.downsample.conv.{p}"] = (
-                    f"encoder.blocks.{idx}.{p}"
+                    f"encoder.blocks.{idx}.conv.{p}"
                 )
             idx += 1
     res("encoder.mid.0", "encoder.mid.block_1", vae.encoder.mid[0])

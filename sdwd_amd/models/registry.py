@@ -59,13 +59,21 @@ class ModelBundle:
 
 
 def _seeded_init(module: nn.Module, seed: int) -> None:
-    """Deterministic, device-independent random init."""
+    """Deterministic, device-independent random init.
+
+    Every parameter must come from the seeded generator: PyTorch's default
+    Linear/Conv bias init draws from the GLOBAL rng, which made two
+    processes build different "identical" models (and made seeded
+    generations irreproducible across runs). Norm weights stay at their
+    deterministic default (ones)."""
     gen = torch.Generator().manual_seed(seed)
     with torch.no_grad():
-        for p in module.parameters():
+        for name, p in module.named_parameters():
             if p.dim() >= 2:
                 nn.init.normal_(p, mean=0.0, std=0.02, generator=gen)
-            # keep 1-d params at their module defaults (ones/zeros)
+            elif name.endswith(".bias") or name == "bias":
+                nn.init.normal_(p, mean=0.0, std=0.01, generator=gen)
+            # 1-d norm weights keep their default (ones)
 
 
 def _build_sd15(name: str) -> ModelBundle:

@@ -85,6 +85,20 @@ class VAEAttention(nn.Module):
         return x + o
 
 
+class VAEDownsample(nn.Module):
+    """ldm VAE downsample: pad (0,1,0,1) then a stride-2 conv with no
+    padding. The asymmetric pad shifts the sampling grid half a latent pixel
+    relative to a symmetric pad-1 conv, so real checkpoint weights only
+    reproduce the reference's latents with this exact form."""
+
+    def __init__(self, ch: int):
+        super().__init__()
+        self.conv = SDConv2d(ch, ch, 3, stride=2, padding=0)
+
+    def forward(self, x):
+        return self.conv(torch.nn.functional.pad(x, (0, 1, 0, 1)))
+
+
 class VAEEncoder(nn.Module):
     def __init__(self, cfg: VAEConfig):
         super().__init__()
@@ -98,7 +112,7 @@ class VAEEncoder(nn.Module):
                 blocks.append(VAEResBlock(cur, out_ch, cfg.groups))
                 cur = out_ch
             if lvl != len(cfg.channel_mult) - 1:
-                blocks.append(SDConv2d(cur, cur, 3, stride=2, padding=1))
+                blocks.append(VAEDownsample(cur))
         self.blocks = nn.ModuleList(blocks)
         self.mid = nn.ModuleList(
             [

@@ -146,7 +146,6 @@ class StableDiffusionPipeline:
             prompts + negatives, device=self.device
         )
         m = self.model
-        penult = clip_skip >= 2
         if m.is_sdxl:
             h1 = m.text_encoder(tokens, penultimate=True)
             h2 = m.text_encoder_2(tokens, penultimate=True)
@@ -155,7 +154,7 @@ class StableDiffusionPipeline:
                 tokens, m.text_encoder_2(tokens)
             )
         else:
-            ctx = m.text_encoder(tokens, penultimate=penult)
+            ctx = m.text_encoder(tokens, clip_skip=clip_skip)
             pooled = None
         # prompt-attention weights (sdwui semantics): scale the hidden
         # states of weighted tokens, then restore the original mean

@@ -173,13 +173,24 @@ class TestImg2Img:
 
 class TestRegistry:
     def test_deterministic_weights_across_loads(self):
-        from sdwd_amd.models import clear_cache
-
+        """EVERY parameter (including 1-d biases, which PyTorch's default
+        init draws from the global rng) must be identical between two
+        independent builds — this is what lets N ranks build the same model
+        without a broadcast, and what makes seeded runs reproducible
+        across processes."""
         m1 = load_model("tiny", cache=False)
+        torch.randn(1000)  # perturb the global rng between builds
         m2 = load_model("tiny", cache=False)
-        p1 = list(m1.unet.parameters())[0]
-        p2 = list(m2.unet.parameters())[0]
-        assert torch.equal(p1, p2)
+        for part in ("unet", "vae", "text_encoder"):
+            s1 = getattr(m1, part).state_dict()
+            s2 = getattr(m2, part).state_dict()
+            for k in s1:
+                assert torch.equal(s1[k], s2[k]), f"{part}.{k}"
+
+    def test_bias_init_is_seeded_not_default(self):
+        m = load_model("tiny", cache=False)
+        bias = dict(m.unet.named_parameters())["conv_in.bias"]
+        assert bias.abs().sum() > 0  # not zeros: came from the seeded normal
 
     def test_unknown_model_raises(self):
         with pytest.raises(KeyError):
