@@ -35,6 +35,9 @@ SAMPLER_EVALS_PER_STEP = {
     "Heun": 2.0,
     "DPM2": 2.0,
     "DPM2 a": 2.0,
+    "DPM++ 2S a": 2.0,
+    "DPM++ 2S a Karras": 2.0,
+    "UniPC": 1.0,  # corrector eval is reused as the next predictor eval
 }
 
 

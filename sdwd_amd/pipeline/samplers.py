@@ -215,6 +215,117 @@ class DPM2Ancestral(DPM2):
     eta = 1.0
 
 
+class DPMpp2SAncestral(Sampler):
+    """DPM++ 2S a (k-diffusion sample_dpmpp_2s_ancestral): single-step
+    second-order update to the ancestral down-sigma, then ancestral noise."""
+
+    order = 2
+
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
+        denoised = _eval(model_fn, x, sigma, t)
+        if sigma_next <= 0:
+            return denoised
+        sd, su = _ancestral_sigmas(sigma, sigma_next)
+        tt, tn = -math.log(sigma), -math.log(sd)
+        h = tn - tt
+        s = tt + 0.5 * h
+        sig_s = math.exp(-s)
+        x2 = ops.lincomb(
+            x, denoised, sig_s / sigma, -math.expm1(-(s - tt))
+        )
+        t_mid = self._t_for(sig_s, sigma, sigma_next, t, t_next)
+        denoised2 = _eval(model_fn, x2, sig_s, t_mid)
+        x = ops.lincomb(x, denoised2, sd / sigma, -math.expm1(-h))
+        if su > 0 and noise_fn is not None:
+            x = ops.add_noise(x, noise_fn(), 1.0, su)
+        return x
+
+
+class UniPC(Sampler):
+    """UniPC (bh2 variant, data prediction, order <= 3) in sigma space.
+
+    Predictor-corrector multistep: the corrector's model evaluation at the
+    new point is cached and reused as the next step's predictor evaluation,
+    so the amortized cost is one model eval per step — the same trick the
+    original uni_pc sampling loop uses.
+    """
+
+    max_order = 3
+
+    def reset(self):
+        self.m: List[torch.Tensor] = []  # x0 predictions, newest last
+        self.lam: List[float] = []
+        self._pending = None  # (x tensor, model eval) from the corrector
+
+    def _rhos(self, h: float, rks: List[float]):
+        """bh2 coefficients: solve R rhos = b (R[k][i] = rks[i]**k)."""
+        hh = -h
+        h_phi_1 = math.expm1(hh)
+        b_h = h_phi_1  # bh2: B(h) = expm1(hh)
+        b = []
+        h_phi_k = h_phi_1 / hh - 1
+        for k in range(1, len(rks) + 1):
+            b.append(h_phi_k * math.factorial(k) / b_h)
+            h_phi_k = h_phi_k / hh - 1 / math.factorial(k + 1)
+        n = len(rks)
+        if n == 1:
+            return [b[0]], h_phi_1, b_h
+        import numpy as np
+
+        R = np.array([[r ** k for r in rks] for k in range(n)])
+        rhos = np.linalg.solve(R, np.array(b))
+        return [float(v) for v in rhos], h_phi_1, b_h
+
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
+        if self._pending is not None and self._pending[0] is x:
+            m0 = self._pending[1]  # corrector eval from the previous step
+        else:
+            m0 = _eval(model_fn, x, sigma, t)
+        self._pending = None
+        lam0 = -math.log(sigma)
+        self.m.append(m0)
+        self.lam.append(lam0)
+        if len(self.m) > self.max_order:
+            self.m.pop(0)
+            self.lam.pop(0)
+        if sigma_next <= 0:
+            return m0
+        lam_t = -math.log(sigma_next)
+        h = lam_t - lam0
+        order = min(len(self.m), self.max_order)
+        # history ratios and differences (newest-first back through history)
+        rks = [
+            (self.lam[-1 - k] - lam0) / h for k in range(1, order)
+        ]
+        d1s = [
+            ops.lincomb(self.m[-1 - k], m0, 1.0 / rks[k - 1], -1.0 / rks[k - 1])
+            for k in range(1, order)
+        ]
+        # predictor
+        base = ops.lincomb(x, m0, sigma_next / sigma, -math.expm1(-h))
+        if rks:
+            rhos_p, _, b_h = self._rhos(h, rks)
+            x_t = base
+            for r, d in zip(rhos_p, d1s):
+                x_t = ops.lincomb(x_t, d, 1.0, -b_h * r)
+        else:
+            x_t = base
+        # corrector (skipped at sigma 0, handled above)
+        m_t = _eval(
+            model_fn, x_t, sigma_next, t_next if t_next is not None else t
+        )
+        rhos_c, _, b_h = self._rhos(h, rks + [1.0])
+        x_c = base
+        for r, d in zip(rhos_c[:-1], d1s):
+            x_c = ops.lincomb(x_c, d, 1.0, -b_h * r)
+        d1_t = ops.lincomb(m_t, m0, 1.0, -1.0)
+        x_c = ops.lincomb(x_c, d1_t, 1.0, -b_h * rhos_c[-1])
+        self._pending = (x_c, m_t)
+        return x_c
+
+
 class LMS(Sampler):
     """Linear multistep (order <= 4) with exactly integrated Adams
     coefficients over each sigma interval (k-diffusion sample_lms)."""
@@ -271,6 +382,9 @@ SAMPLERS: Dict[str, type] = {
     "DPM2": DPM2,
     "DPM2 a": DPM2Ancestral,
     "DDPM": EulerAncestral,
+    "DPM++ 2S a": DPMpp2SAncestral,
+    "DPM++ 2S a Karras": DPMpp2SAncestral,
+    "UniPC": UniPC,
 }
 
 
