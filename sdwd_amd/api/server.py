@@ -234,15 +234,8 @@ def create_app(engine: Optional[LocalEngine] = None,
             name = req.sd_model_checkpoint
             if name not in available_models():
                 raise HTTPException(404, f"unknown model {name}")
-            log.info("switching model to %s on all ranks", name)
-            from ..pipeline import StableDiffusionPipeline
-
-            for label, pipe in engine.pipes.items():
-                engine.pipes[label] = StableDiffusionPipeline(
-                    name, device=pipe.device, dtype=pipe.dtype
-                )
+            engine.set_model(name)
             state.current_model = name
-            engine.model_name = name
         return {}
 
     @app.get("/sdapi/v1/options")

@@ -56,6 +56,7 @@ class GenerationRequest:
     control_model: str = ""
     control_scale: float = 1.0
     clip_skip: int = 1
+    model: str = ""  # hot-swap to this checkpoint first ("" = keep current)
 
     def sched(self) -> GenRequest:
         return GenRequest(
@@ -198,6 +199,19 @@ class LocalEngine(_EngineBase):
             # remotes at init); CPU ranks have no memory probe to ping
             self.world.ping(indiscriminate=True)
 
+    def set_model(self, name: str) -> None:
+        """Hot-swap the checkpoint on every rank (ref C13: the reference
+        pushed the model name to all remotes over POST /options,
+        world.py:784-811)."""
+        if not name or name == self.model_name:
+            return
+        log.info("switching model to %s on all ranks", name)
+        for label, pipe in self.pipes.items():
+            self.pipes[label] = StableDiffusionPipeline(
+                name, device=pipe.device, dtype=pipe.dtype
+            )
+        self.model_name = name
+
     # benchmark runner wired into core.Worker.benchmark (ref C8)
     def _bench_runner(self, worker: Worker, payload) -> float:
         pipe = self.pipes[worker.label]
@@ -281,6 +295,7 @@ class LocalEngine(_EngineBase):
 
     def generate(self, gen: GenerationRequest) -> GalleryResult:
         t0 = time.perf_counter()
+        self.set_model(gen.model)
         self.world.clear_interrupt()
         jobs = self.world.make_jobs(gen.sched())
         self._live_jobs = jobs
@@ -368,17 +383,10 @@ class DistributedEngine(_EngineBase):
         else:
             self.device = torch.device("cpu")
         self.label = f"gpu{self.rank}"
+        self.model_name = model
+        self._dtype = dtype
         self.pipe = StableDiffusionPipeline(model, device=self.device, dtype=dtype)
-        # weight sync from rank 0 (C13): registry init is deterministic, the
-        # broadcast makes identity unconditional.
-        for m in (
-            self.pipe.model.text_encoder,
-            self.pipe.model.text_encoder_2,
-            self.pipe.model.unet,
-            self.pipe.model.vae,
-        ):
-            if m is not None:
-                pg.sync_weights(m)
+        self._sync_pipe_weights()
         self.world = World.from_devices(self.world_size)
         self._store = None
         if self.is_dist:
@@ -389,6 +397,31 @@ class DistributedEngine(_EngineBase):
             except Exception:  # pragma: no cover
                 self._store = None
         self._interrupt_epoch = 0
+
+    def _sync_pipe_weights(self) -> None:
+        # weight sync from rank 0 (C13): registry init is deterministic, the
+        # broadcast makes identity unconditional.
+        for m in (
+            self.pipe.model.text_encoder,
+            self.pipe.model.text_encoder_2,
+            self.pipe.model.unet,
+            self.pipe.model.vae,
+        ):
+            if m is not None:
+                pg.sync_weights(m)
+
+    def set_model(self, name: str) -> None:
+        """Collective model hot-swap (C13): every rank rebuilds its pipeline
+        and re-syncs weights from rank 0. Must be called on ALL ranks
+        (generate() routes it through the plan broadcast so this holds)."""
+        if not name or name == self.model_name:
+            return
+        log.info("rank %d switching model to %s", self.rank, name)
+        self.pipe = StableDiffusionPipeline(
+            name, device=self.device, dtype=self._dtype
+        )
+        self._sync_pipe_weights()
+        self.model_name = name
 
     # -- heartbeat (C11): ranks stamp the store; anyone can read liveness ----
     def heartbeat(self) -> None:
@@ -498,10 +531,11 @@ class DistributedEngine(_EngineBase):
         self.heartbeat()
         self._clear_interrupt()
         if self.rank == 0:
-            jobs = self.world.make_jobs(gen.sched())
+            plan = (gen.model, self.world.make_jobs(gen.sched()))
         else:
-            jobs = None
-        jobs = pg.broadcast_object(jobs)
+            plan = None
+        model_name, jobs = pg.broadcast_object(plan)
+        self.set_model(model_name)
 
         mine = next((j for j in jobs if j.worker_label == self.label), None)
         hf = gen.hr_scale if gen.enable_hr else 1.0

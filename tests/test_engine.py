@@ -135,6 +135,17 @@ def _dist_worker(rank, world_size, port, tmpdir, fail_rank):
         for i in range(4):
             assert res.images[i].float().std() > 0, f"image {i} empty"
         torch.save(res.images, os.path.join(tmpdir, "gallery.pt"))
+    if fail_rank == -1:
+        # model hot-swap rides the plan broadcast: every rank rebuilds
+        res2 = eng.generate(
+            GenerationRequest(
+                prompt="swap", batch_size=2, width=64, height=64, steps=1,
+                seed=7, model="tiny-xl",
+            )
+        )
+        assert eng.model_name == "tiny-xl"
+        if rank == 0:
+            assert res2.images.shape == (2, 64, 64, 3)
     from sdwd_amd.parallel import destroy_group
 
     destroy_group()
@@ -215,6 +226,22 @@ class TestHeartbeat:
 
 
 class TestEdgeCases:
+    def test_model_hot_swap_per_request(self):
+        eng = make_engine(1)
+        res = eng.generate(
+            GenerationRequest(prompt="m", batch_size=1, width=64, height=64,
+                              steps=1, seed=1, model="tiny-xl")
+        )
+        assert eng.model_name == "tiny-xl"
+        assert res.images.shape == (1, 64, 64, 3)
+        # same-model request is a no-op (pipes keep their identity)
+        p = eng.pipes["gpu0"]
+        eng.generate(
+            GenerationRequest(prompt="m", batch_size=1, width=64, height=64,
+                              steps=1, seed=1, model="tiny-xl")
+        )
+        assert eng.pipes["gpu0"] is p
+
     def test_batch_smaller_than_ranks(self):
         eng = make_engine(3)
         res = eng.generate(
