@@ -54,6 +54,9 @@ class Txt2ImgRequest(BaseModel):
     # execute the controlnet unit natively, other scripts are ignored
     # with a warning as the reference's compat filter did)
     alwayson_scripts: Dict[str, Any] = Field(default_factory=dict)
+    # sdwui per-request overrides (sd_model_checkpoint,
+    # CLIP_stop_at_last_layers are honored; the rest are ignored)
+    override_settings: Dict[str, Any] = Field(default_factory=dict)
 
 
 class Img2ImgRequest(Txt2ImgRequest):
@@ -130,6 +133,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             state.started_at = time.time()
             try:
                 result = engine.generate(gen)
+                state.current_model = engine.model_name
             finally:
                 state.busy = False
         images = [
@@ -164,11 +168,21 @@ def create_app(engine: Optional[LocalEngine] = None,
 
         return HTMLResponse(PAGE)
 
+    def _overrides(req: Txt2ImgRequest):
+        """Per-request override_settings (sdwui convention)."""
+        ov = req.override_settings or {}
+        model = str(ov.get("sd_model_checkpoint") or "")
+        if model and model not in available_models():
+            raise HTTPException(404, f"unknown model {model}")
+        clip_skip = int(ov.get("CLIP_stop_at_last_layers") or req.clip_skip)
+        return model, clip_skip
+
     @app.post("/sdapi/v1/txt2img")
     def txt2img(req: Txt2ImgRequest):
         control_image, control_model, control_scale = _parse_controlnet(
             req.alwayson_scripts
         )
+        model, clip_skip = _overrides(req)
         gen = GenerationRequest(
             prompt=req.prompt,
             negative_prompt=req.negative_prompt,
@@ -185,10 +199,11 @@ def create_app(engine: Optional[LocalEngine] = None,
             hr_scale=req.hr_scale,
             hr_steps=req.hr_second_pass_steps,
             denoising_strength=req.denoising_strength,
-            clip_skip=req.clip_skip,
+            clip_skip=clip_skip,
             control_image=control_image,
             control_model=control_model,
             control_scale=control_scale,
+            model=model,
         )
         return run_generation(gen)
 
@@ -196,6 +211,7 @@ def create_app(engine: Optional[LocalEngine] = None,
     def img2img(req: Img2ImgRequest):
         if not req.init_images:
             raise HTTPException(422, "init_images required")
+        model, clip_skip = _overrides(req)
         try:
             inits = torch.stack(
                 [_decode_b64_png(d) for d in req.init_images]
@@ -225,6 +241,8 @@ def create_app(engine: Optional[LocalEngine] = None,
             init_images=inits,
             denoising_strength=req.denoising_strength,
             mask_image=mask_image,
+            clip_skip=clip_skip,
+            model=model,
         )
         return run_generation(gen)
 
@@ -269,7 +287,9 @@ def create_app(engine: Optional[LocalEngine] = None,
 
     @app.post("/sdapi/v1/refresh-checkpoints")
     def refresh_checkpoints():
-        return {}
+        from ..models.registry import refresh_checkpoint_files
+
+        return {"found": refresh_checkpoint_files()}
 
     @app.post("/sdapi/v1/refresh-loras")
     def refresh_loras():

@@ -140,10 +140,37 @@ _BUILDERS: Dict[str, Callable[[str], ModelBundle]] = {
 
 _cache: Dict[str, ModelBundle] = {}
 
+# file-backed checkpoints (sdwui's checkpoint folder + refresh button):
+# name (file stem) -> path, populated by refresh_checkpoint_files()
+_FILE_MODELS: Dict[str, str] = {}
+
+
+def checkpoint_dir() -> str:
+    import os
+
+    return os.environ.get("SDWD_CHECKPOINT_DIR", "checkpoints")
+
+
+def refresh_checkpoint_files(dirpath: Optional[str] = None) -> list:
+    """Scan the checkpoint directory for *.safetensors; each file becomes a
+    loadable model name (its stem). Both formats load: native (saved by
+    save_checkpoint) and sdwui/ldm (auto-detected by key prefix)."""
+    import os
+
+    global _FILE_MODELS
+    d = dirpath or checkpoint_dir()
+    found: Dict[str, str] = {}
+    if os.path.isdir(d):
+        for fn in sorted(os.listdir(d)):
+            if fn.endswith(".safetensors"):
+                found[os.path.splitext(fn)[0]] = os.path.join(d, fn)
+    _FILE_MODELS = found
+    return sorted(found)
+
 
 def available_models() -> list:
     """ref worker.py:623-644 (GET /sd-models)."""
-    return sorted(_BUILDERS.keys())
+    return sorted(set(_BUILDERS) | set(_FILE_MODELS))
 
 
 def load_model(
@@ -153,6 +180,14 @@ def load_model(
     key = f"{name}"
     if cache and key in _cache:
         return _cache[key].to(device, dtype)
+    if name not in _BUILDERS and name not in _FILE_MODELS:
+        refresh_checkpoint_files()
+    if name in _FILE_MODELS and name not in _BUILDERS:
+        log.info("loading checkpoint file '%s'", _FILE_MODELS[name])
+        bundle = load_checkpoint(_FILE_MODELS[name], device=device, dtype=dtype)
+        if cache:
+            _cache[key] = bundle
+        return bundle
     if name not in _BUILDERS:
         raise KeyError(f"unknown model '{name}'; have {available_models()}")
     log.info("building model '%s' (random-init, deterministic)", name)

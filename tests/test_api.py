@@ -131,6 +131,37 @@ class TestControl:
             json={"job_timeout": 3.0, "distribute_img2img": True},
         )
 
+    def test_override_settings_model_and_clip_skip(self, client):
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={
+                "prompt": "o", "steps": 1, "width": 64, "height": 64,
+                "seed": 3,
+                "override_settings": {
+                    "sd_model_checkpoint": "tiny-xl",
+                    "CLIP_stop_at_last_layers": 2,
+                },
+            },
+        )
+        assert r.status_code == 200
+        # the swap persisted (sdwui override_settings default behavior)
+        opts = client.get("/sdapi/v1/options").json()
+        assert opts["sd_model_checkpoint"] == "tiny-xl"
+        # switch back for the other tests
+        assert client.post(
+            "/sdapi/v1/options", json={"sd_model_checkpoint": "tiny"}
+        ).status_code == 200
+
+    def test_override_settings_unknown_model_404(self, client):
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={
+                "prompt": "o", "steps": 1, "width": 64, "height": 64,
+                "override_settings": {"sd_model_checkpoint": "nope"},
+            },
+        )
+        assert r.status_code == 404
+
     def test_settings_rejects_unknown(self, client):
         r = client.post("/sdwd/settings", json={"warp_factor": 9})
         assert r.status_code == 422

@@ -104,6 +104,37 @@ class TestCheckpointFile:
             assert torch.equal(a.unet(x, t, ctx), b.unet(x, t, ctx))
 
 
+class TestCheckpointDir:
+    def test_file_backed_models(self, tmp_path, monkeypatch):
+        """*.safetensors in SDWD_CHECKPOINT_DIR become loadable model names
+        (sdwui checkpoint-folder semantics)."""
+        from sdwd_amd.models.registry import (
+            available_models,
+            load_model as lm,
+            refresh_checkpoint_files,
+            save_checkpoint,
+        )
+
+        a = lm("tiny", device="cpu", cache=False)
+        d = tmp_path / "ckpts"
+        d.mkdir()
+        save_checkpoint(a, str(d / "my-custom-model.safetensors"))
+        monkeypatch.setenv("SDWD_CHECKPOINT_DIR", str(d))
+        try:
+            names = refresh_checkpoint_files()
+            assert names == ["my-custom-model"]
+            assert "my-custom-model" in available_models()
+            b = lm("my-custom-model", cache=False)
+            x = torch.randn(1, 4, 16, 16)
+            t = torch.tensor([3.0])
+            ctx = torch.randn(1, 77, a.unet.cfg.context_dim)
+            with torch.no_grad():
+                assert torch.equal(a.unet(x, t, ctx), b.unet(x, t, ctx))
+        finally:
+            refresh_checkpoint_files(dirpath=str(tmp_path / "none"))
+        assert "my-custom-model" not in available_models()
+
+
 @pytest.fixture(scope="module")
 def sd15_keys():
     bundle = load_model("sd15", device="cpu", cache=False)
