@@ -48,6 +48,7 @@ class Txt2ImgRequest(BaseModel):
     # hires fix (sdwui fields)
     enable_hr: bool = False
     hr_scale: float = 2.0
+    hr_upscaler: str = "Latent"
     hr_second_pass_steps: int = 0
     denoising_strength: float = 0.75
     # alwayson scripts (ref C17/C18: the reference forwarded these; we
@@ -198,6 +199,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             enable_hr=req.enable_hr,
             hr_scale=req.hr_scale,
             hr_steps=req.hr_second_pass_steps,
+            hr_upscaler=req.hr_upscaler,
             denoising_strength=req.denoising_strength,
             clip_skip=clip_skip,
             control_image=control_image,

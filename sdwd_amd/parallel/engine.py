@@ -52,6 +52,7 @@ class GenerationRequest:
     enable_hr: bool = False
     hr_scale: float = 2.0
     hr_steps: int = 0
+    hr_upscaler: str = "nearest"
     control_image: Optional[torch.Tensor] = None
     control_model: str = ""
     control_scale: float = 1.0
@@ -105,6 +106,7 @@ def _job_pipeline_request(
         enable_hr=gen.enable_hr,
         hr_scale=gen.hr_scale,
         hr_steps=gen.hr_steps,
+        hr_upscaler=gen.hr_upscaler,
         control_image=gen.control_image,
         control_model=gen.control_model,
         control_scale=gen.control_scale,

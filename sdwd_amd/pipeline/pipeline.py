@@ -48,6 +48,9 @@ class PipelineRequest:
     enable_hr: bool = False
     hr_scale: float = 2.0
     hr_steps: int = 0  # 0 = same as steps
+    # latent upscaler (sdwui hr_upscaler "Latent..." family):
+    # nearest | bilinear | bicubic | bilinear-antialiased | bicubic-antialiased
+    hr_upscaler: str = "nearest"
     # controlnet (ref C17 executed natively)
     control_image: Optional[torch.Tensor] = None  # [B,H,W,3] uint8
     control_model: str = ""  # e.g. "controlnet-sd15"; "" = off
@@ -82,6 +85,30 @@ def _slerp(a: torch.Tensor, b: torch.Tensor, t: float) -> torch.Tensor:
             math.sin(t * omega) / so
         ) * bf
     return out.reshape(a.shape).to(a.dtype)
+
+
+_HR_MODES = {
+    "nearest": ("nearest", False),
+    "latent": ("nearest", False),  # sdwui "Latent" = nearest
+    "bilinear": ("bilinear", False),
+    "latent (bilinear)": ("bilinear", False),
+    "bicubic": ("bicubic", False),
+    "latent (bicubic)": ("bicubic", False),
+    "bilinear-antialiased": ("bilinear", True),
+    "latent (bilinear antialiased)": ("bilinear", True),
+    "bicubic-antialiased": ("bicubic", True),
+    "latent (bicubic antialiased)": ("bicubic", True),
+}
+
+
+def _upscale_latent(x: torch.Tensor, scale: float, upscaler: str) -> torch.Tensor:
+    """Latent-space upscale for the hires-fix first->second pass handoff
+    (sdwui's "Latent ..." hr_upscaler family; ref CHANGELOG hires support)."""
+    mode, aa = _HR_MODES.get((upscaler or "nearest").lower(), ("nearest", False))
+    kwargs = {"antialias": True} if aa else {}
+    return torch.nn.functional.interpolate(
+        x, scale_factor=scale, mode=mode, **kwargs
+    )
 
 
 def _image_noise(
@@ -373,8 +400,8 @@ class StableDiffusionPipeline:
         # (ref eta_hr, worker.py:205-228 predicts exactly this shape).
         if req.enable_hr and req.hr_scale > 1.0 and not was_interrupted:
             hr_steps = req.hr_steps or req.steps
-            x = torch.nn.functional.interpolate(
-                x.float(), scale_factor=req.hr_scale, mode="nearest"
+            x = _upscale_latent(
+                x.float(), req.hr_scale, req.hr_upscaler
             ).to(self.dtype)
             hsched = schedule_for(req.sampler_name, hr_steps)
             start = max(

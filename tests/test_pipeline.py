@@ -198,6 +198,30 @@ class TestRegistry:
 
 
 class TestHiresFix:
+    def test_latent_upscaler_modes(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+        from sdwd_amd.pipeline.pipeline import _upscale_latent
+
+        x = torch.randn(1, 4, 8, 8)
+        for name in ("Latent", "Latent (bilinear)", "Latent (bicubic)",
+                     "Latent (bicubic antialiased)", "bilinear-antialiased"):
+            up = _upscale_latent(x, 2.0, name)
+            assert up.shape == (1, 4, 16, 16), name
+        assert not torch.equal(
+            _upscale_latent(x, 2.0, "Latent"),
+            _upscale_latent(x, 2.0, "Latent (bilinear)"),
+        )
+        # end-to-end: a non-default upscaler changes the hires output
+        base = dict(prompt="hr", steps=2, width=64, height=64, seeds=[9],
+                    enable_hr=True, hr_scale=2.0, hr_steps=2,
+                    denoising_strength=0.6)
+        a = pipe.generate(PipelineRequest(**base)).images
+        b = pipe.generate(
+            PipelineRequest(**base, hr_upscaler="Latent (bilinear)")
+        ).images
+        assert a.shape == b.shape == (1, 128, 128, 3)
+        assert not torch.equal(a, b)
+
     def test_two_pass_upscale(self, pipe):
         from sdwd_amd.pipeline import PipelineRequest
 
