@@ -82,6 +82,12 @@ class CLIPTextEncoder(nn.Module):
         self.ln_final = nn.LayerNorm(d_model)
         mask = torch.full((max_len, max_len), float("-inf")).triu(1)
         self.register_buffer("causal_bias", mask, persistent=False)
+        # open_clip text encoders project the pooled EOT embedding
+        # (pooled = h[eot] @ text_projection); absent for SD1.5-style CLIP
+        self.text_proj: torch.nn.Parameter | None = None
+
+    def set_text_projection(self, weight: torch.Tensor) -> None:
+        self.text_proj = nn.Parameter(weight.clone())  # auto-registers
 
     def forward(
         self,
@@ -109,4 +115,7 @@ class CLIPTextEncoder(nn.Module):
     def pooled(self, tokens: torch.Tensor, hidden: torch.Tensor) -> torch.Tensor:
         """EOT-token pooled embedding (SDXL conditioning)."""
         eot = tokens.argmax(dim=-1)  # highest id = end-of-text
-        return hidden[torch.arange(hidden.shape[0]), eot]
+        h = hidden[torch.arange(hidden.shape[0]), eot]
+        if self.text_proj is not None:
+            h = h @ self.text_proj.to(h.dtype)
+        return h

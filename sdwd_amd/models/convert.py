@@ -34,6 +34,10 @@ log = get_logger("sdwd.convert")
 UNET_PREFIX = "model.diffusion_model."
 VAE_PREFIX = "first_stage_model."
 CLIP_PREFIX = "cond_stage_model.transformer.text_model."
+# SDXL checkpoints: two text encoders under the sgm conditioner —
+# embedders.0 is SD-style CLIP-L, embedders.1 is open_clip (bigG) layout
+XL_CLIP_L_PREFIX = "conditioner.embedders.0.transformer.text_model."
+XL_CLIP_G_PREFIX = "conditioner.embedders.1.model."
 
 _RES_INNER = {
     "norm1": "in_layers.0",
@@ -210,6 +214,32 @@ def clip_key_map(enc) -> Tuple[Dict[str, str], List[int]]:
     return out, fused
 
 
+def openclip_key_map(enc) -> Dict[str, str]:
+    """open_clip text-tower layout (SDXL's second encoder). qkv is stored
+    fused (attn.in_proj_*) which matches the native fused qkv directly."""
+    out = {
+        "token_embedding.weight": "token_emb.weight",
+        "positional_embedding": "pos_emb",
+        "ln_final.weight": "ln_final.weight",
+        "ln_final.bias": "ln_final.bias",
+    }
+    for i in range(len(enc.blocks)):
+        l = f"transformer.resblocks.{i}"
+        o = f"blocks.{i}"
+        for a, b in (("ln_1", "ln1"), ("ln_2", "ln2")):
+            out[f"{l}.{a}.weight"] = f"{o}.{b}.weight"
+            out[f"{l}.{a}.bias"] = f"{o}.{b}.bias"
+        out[f"{l}.attn.in_proj_weight"] = f"{o}.attn.qkv.weight"
+        out[f"{l}.attn.in_proj_bias"] = f"{o}.attn.qkv.bias"
+        out[f"{l}.attn.out_proj.weight"] = f"{o}.attn.out.weight"
+        out[f"{l}.attn.out_proj.bias"] = f"{o}.attn.out.bias"
+        out[f"{l}.mlp.c_fc.weight"] = f"{o}.mlp.0.weight"
+        out[f"{l}.mlp.c_fc.bias"] = f"{o}.mlp.0.bias"
+        out[f"{l}.mlp.c_proj.weight"] = f"{o}.mlp.2.weight"
+        out[f"{l}.mlp.c_proj.bias"] = f"{o}.mlp.2.bias"
+    return out
+
+
 def _fit(src: torch.Tensor, like: torch.Tensor) -> torch.Tensor:
     """Adapt a file tensor to the native parameter's shape: squeeze 1x1
     convs stored for linears and vice versa."""
@@ -258,6 +288,31 @@ def _load_part(mod, sub: Dict[str, torch.Tensor], key_map: Dict[str, str],
                 continue
             tgt.copy_(_fit(sub.pop(ldm_key), tgt).to(tgt.dtype))
             report["loaded"].append(ldm_key)
+
+
+def _load_sd_clip(enc, clip_sd: Dict[str, torch.Tensor],
+                  report: Dict[str, list]) -> None:
+    """SD-style CLIP (split q/k/v projections) -> native fused qkv."""
+    kmap, fused = clip_key_map(enc)
+    _load_part(enc, clip_sd, kmap, report)
+    with torch.no_grad():
+        for i in fused:
+            parts_w, parts_b = [], []
+            ok = True
+            for n in ("q_proj", "k_proj", "v_proj"):
+                kw = f"encoder.layers.{i}.self_attn.{n}.weight"
+                kb = f"encoder.layers.{i}.self_attn.{n}.bias"
+                if kw not in clip_sd or kb not in clip_sd:
+                    ok = False
+                    report["missing"].append(kw)
+                    continue
+                parts_w.append(clip_sd.pop(kw))
+                parts_b.append(clip_sd.pop(kb))
+            if ok:
+                qkv = enc.blocks[i].attn.qkv
+                qkv.weight.copy_(torch.cat(parts_w, 0).to(qkv.weight.dtype))
+                qkv.bias.copy_(torch.cat(parts_b, 0).to(qkv.bias.dtype))
+                report["loaded"].append(f"encoder.layers.{i}.self_attn.qkv")
 
 
 def load_ldm_state_dict(bundle, state: Dict[str, torch.Tensor]) -> Dict[str, list]:
@@ -309,34 +364,27 @@ def load_ldm_state_dict(bundle, state: Dict[str, torch.Tensor]) -> Dict[str, lis
                 dec["conv_in.bias"].copy_(b.to(dec["conv_in.bias"].dtype))
         report["unexpected"] += [VAE_PREFIX + k for k in vae_sd]
     if clip_sd and bundle.text_encoder is not None:
-        enc = bundle.text_encoder
-        kmap, fused = clip_key_map(enc)
-        _load_part(enc, clip_sd, kmap, report)
-        with torch.no_grad():
-            for i in fused:
-                parts_w, parts_b = [], []
-                ok = True
-                for n in ("q_proj", "k_proj", "v_proj"):
-                    kw = f"encoder.layers.{i}.self_attn.{n}.weight"
-                    kb = f"encoder.layers.{i}.self_attn.{n}.bias"
-                    if kw not in clip_sd or kb not in clip_sd:
-                        ok = False
-                        report["missing"].append(kw)
-                        continue
-                    parts_w.append(clip_sd.pop(kw))
-                    parts_b.append(clip_sd.pop(kb))
-                if ok:
-                    qkv = enc.blocks[i].attn.qkv
-                    qkv.weight.copy_(torch.cat(parts_w, 0).to(qkv.weight.dtype))
-                    qkv.bias.copy_(torch.cat(parts_b, 0).to(qkv.bias.dtype))
-                    report["loaded"].append(f"encoder.layers.{i}.self_attn.qkv")
+        _load_sd_clip(bundle.text_encoder, clip_sd, report)
         report["unexpected"] += [CLIP_PREFIX + k for k in clip_sd]
+    xl_l = take(XL_CLIP_L_PREFIX)
+    if xl_l and bundle.text_encoder is not None:
+        _load_sd_clip(bundle.text_encoder, xl_l, report)
+        report["unexpected"] += [XL_CLIP_L_PREFIX + k for k in xl_l]
+    xl_g = take(XL_CLIP_G_PREFIX)
+    if xl_g and getattr(bundle, "text_encoder_2", None) is not None:
+        enc2 = bundle.text_encoder_2
+        _load_part(enc2, xl_g, openclip_key_map(enc2), report)
+        if "text_projection" in xl_g:
+            enc2.set_text_projection(xl_g.pop("text_projection").float())
+            report["loaded"].append("text_projection")
+        xl_g.pop("logit_scale", None)
+        report["unexpected"] += [XL_CLIP_G_PREFIX + k for k in xl_g]
     # whatever is left in the file we never claimed to consume
     report["unexpected"] += [
         k for k in state
-        if not k.startswith(("cond_stage_model.", "model_ema.", "alphas", "betas",
-                             "sqrt_", "log_one_minus", "posterior_", "model.",
-                             "first_stage_model."))
+        if not k.startswith(("cond_stage_model.", "conditioner.", "model_ema.",
+                             "alphas", "betas", "sqrt_", "log_one_minus",
+                             "posterior_", "model.", "first_stage_model."))
     ]
     if report["missing"]:
         log.warning("ldm load: %d keys missing (first: %s)",
@@ -372,16 +420,26 @@ def to_ldm_state_dict(bundle) -> Dict[str, torch.Tensor]:
     eye = torch.eye(lat, dtype=eye.dtype)
     out[VAE_PREFIX + "post_quant_conv.weight"] = eye.reshape(lat, lat, 1, 1)
     out[VAE_PREFIX + "post_quant_conv.bias"] = torch.zeros(lat, dtype=eye.dtype)
+    clip_l_prefix = XL_CLIP_L_PREFIX if bundle.is_sdxl else CLIP_PREFIX
     if bundle.text_encoder is not None:
         enc = bundle.text_encoder
         clip_sd = enc.state_dict()
         kmap, fused = clip_key_map(enc)
         for ldm_key, our_key in kmap.items():
-            out[CLIP_PREFIX + ldm_key] = clip_sd[our_key]
+            out[clip_l_prefix + ldm_key] = clip_sd[our_key]
         for i in fused:
             w = clip_sd[f"blocks.{i}.attn.qkv.weight"].chunk(3, 0)
             b = clip_sd[f"blocks.{i}.attn.qkv.bias"].chunk(3, 0)
             for j, n in enumerate(("q_proj", "k_proj", "v_proj")):
-                out[f"{CLIP_PREFIX}encoder.layers.{i}.self_attn.{n}.weight"] = w[j]
-                out[f"{CLIP_PREFIX}encoder.layers.{i}.self_attn.{n}.bias"] = b[j]
+                out[f"{clip_l_prefix}encoder.layers.{i}.self_attn.{n}.weight"] = w[j]
+                out[f"{clip_l_prefix}encoder.layers.{i}.self_attn.{n}.bias"] = b[j]
+    if bundle.is_sdxl and getattr(bundle, "text_encoder_2", None) is not None:
+        enc2 = bundle.text_encoder_2
+        sd2 = enc2.state_dict()
+        for ldm_key, our_key in openclip_key_map(enc2).items():
+            out[XL_CLIP_G_PREFIX + ldm_key] = sd2[our_key]
+        tp = sd2.get("text_proj")
+        if tp is None:
+            tp = torch.eye(enc2.d_model)
+        out[XL_CLIP_G_PREFIX + "text_projection"] = tp
     return out

@@ -265,7 +265,11 @@ def load_checkpoint(path: str, device="cpu", dtype=None) -> ModelBundle:
 
             ch = f.get_tensor("model.diffusion_model.input_blocks.0.0.weight"
                               ).shape[0]
-            arch = {320: "sd15", 32: "tiny"}.get(int(ch), "sd15")
+            is_xl = any(k.startswith("conditioner.") for k in keys)
+            arch = (
+                ("sdxl" if is_xl else "sd15") if int(ch) == 320
+                else ("tiny-xl" if is_xl else "tiny")
+            )
             bundle = _BUILDERS[arch](arch)
             load_ldm_state_dict(bundle, {k: f.get_tensor(k) for k in keys})
             bundle.eval().to(device, dtype)

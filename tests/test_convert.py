@@ -82,6 +82,47 @@ class TestRoundTrip:
             assert torch.equal(a.unet(x, t, ctx), b.unet(x, t, ctx))
 
 
+class TestSdxlRoundTrip:
+    def test_tiny_xl_bit_exact(self):
+        """SDXL-layout export/load: dual encoders (CLIP-L split-qkv +
+        open_clip fused-qkv), label_emb, text_projection."""
+        a = load_model("tiny-xl", device="cpu", cache=False)
+        exported = to_ldm_state_dict(a)
+        assert any(k.startswith("conditioner.embedders.0.") for k in exported)
+        assert any(
+            ".attn.in_proj_weight" in k
+            for k in exported if k.startswith("conditioner.embedders.1.")
+        )
+        b = load_model("tiny-xl", device="cpu", cache=False)
+        _perturb(b)
+        with torch.no_grad():
+            for p in b.text_encoder_2.parameters():
+                p.add_(torch.randn_like(p) * 0.1)
+        report = load_ldm_state_dict(b, exported)
+        assert not report["missing"], report["missing"][:5]
+        assert not report["unexpected"], report["unexpected"][:5]
+        for part in ("unet", "vae", "text_encoder"):
+            sa = getattr(a, part).state_dict()
+            sb = getattr(b, part).state_dict()
+            for k in sa:
+                assert torch.equal(sa[k], sb[k]), f"{part}.{k}"
+        sa = a.text_encoder_2.state_dict()
+        sb = b.text_encoder_2.state_dict()
+        for k in sa:  # b additionally carries the identity text_projection
+            assert torch.equal(sa[k], sb[k]), f"text_encoder_2.{k}"
+        # pooled output unchanged by the identity projection
+        tokens = torch.randint(0, 100, (2, 77))
+        tokens[:, -1] = 49407
+        with torch.no_grad():
+            ha = a.text_encoder_2(tokens)
+            hb = b.text_encoder_2(tokens)
+            assert torch.equal(ha, hb)
+            assert torch.allclose(
+                a.text_encoder_2.pooled(tokens, ha),
+                b.text_encoder_2.pooled(tokens, hb), atol=1e-5,
+            )
+
+
 class TestCheckpointFile:
     def test_ldm_safetensors_auto_detected(self, tmp_path):
         """An sdwui-format .safetensors file loads through the normal
