@@ -38,6 +38,8 @@ CLIP_PREFIX = "cond_stage_model.transformer.text_model."
 # embedders.0 is SD-style CLIP-L, embedders.1 is open_clip (bigG) layout
 XL_CLIP_L_PREFIX = "conditioner.embedders.0.transformer.text_model."
 XL_CLIP_G_PREFIX = "conditioner.embedders.1.model."
+# SD2.x: single open_clip (ViT-H) text tower
+SD2_CLIP_PREFIX = "cond_stage_model.model."
 
 _RES_INNER = {
     "norm1": "in_layers.0",
@@ -366,6 +368,15 @@ def load_ldm_state_dict(bundle, state: Dict[str, torch.Tensor]) -> Dict[str, lis
     if clip_sd and bundle.text_encoder is not None:
         _load_sd_clip(bundle.text_encoder, clip_sd, report)
         report["unexpected"] += [CLIP_PREFIX + k for k in clip_sd]
+    sd2_clip = take(SD2_CLIP_PREFIX)
+    if sd2_clip and bundle.text_encoder is not None:
+        enc = bundle.text_encoder
+        _load_part(enc, sd2_clip, openclip_key_map(enc), report)
+        if "text_projection" in sd2_clip:
+            enc.set_text_projection(sd2_clip.pop("text_projection").float())
+            report["loaded"].append("text_projection")
+        sd2_clip.pop("logit_scale", None)
+        report["unexpected"] += [SD2_CLIP_PREFIX + k for k in sd2_clip]
     xl_l = take(XL_CLIP_L_PREFIX)
     if xl_l and bundle.text_encoder is not None:
         _load_sd_clip(bundle.text_encoder, xl_l, report)
@@ -421,6 +432,19 @@ def to_ldm_state_dict(bundle) -> Dict[str, torch.Tensor]:
     out[VAE_PREFIX + "post_quant_conv.weight"] = eye.reshape(lat, lat, 1, 1)
     out[VAE_PREFIX + "post_quant_conv.bias"] = torch.zeros(lat, dtype=eye.dtype)
     clip_l_prefix = XL_CLIP_L_PREFIX if bundle.is_sdxl else CLIP_PREFIX
+    # SD2.x lineage: single open_clip tower (d_model 1024 is unambiguous —
+    # SD1.x text encoders are always 768-wide HF-CLIP)
+    if (not bundle.is_sdxl and bundle.text_encoder is not None
+            and bundle.text_encoder.d_model == 1024):
+        enc = bundle.text_encoder
+        sd2 = enc.state_dict()
+        for ldm_key, our_key in openclip_key_map(enc).items():
+            out[SD2_CLIP_PREFIX + ldm_key] = sd2[our_key]
+        tp = sd2.get("text_proj")
+        out[SD2_CLIP_PREFIX + "text_projection"] = (
+            tp if tp is not None else torch.eye(enc.d_model)
+        )
+        return out
     if bundle.text_encoder is not None:
         enc = bundle.text_encoder
         clip_sd = enc.state_dict()

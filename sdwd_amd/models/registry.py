@@ -85,6 +85,20 @@ def _build_sd15(name: str) -> ModelBundle:
     return ModelBundle(name, te, None, unet, vae, context_dim=768)
 
 
+def _build_sd21(name: str) -> ModelBundle:
+    """SD2.1-base shape: open_clip ViT-H text tower (width 1024, 24 layers,
+    16 heads, fused qkv — the native layout), UNet with head_dim 64 and
+    context 1024, same VAE. eps-prediction (the 512-base lineage)."""
+    te = CLIPTextEncoder(d_model=1024, layers=24, heads=16)
+    unet = UNetModel(
+        UNetConfig(context_dim=1024, num_heads=0)  # num_heads 0 -> head_dim 64
+    )
+    vae = AutoencoderKL(VAEConfig.sd())
+    for seed_off, m in enumerate((te, unet, vae)):
+        _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
+    return ModelBundle(name, te, None, unet, vae, context_dim=1024)
+
+
 def _build_sdxl(name: str) -> ModelBundle:
     te = CLIPTextEncoder(d_model=768, layers=12, heads=12)
     te2 = CLIPTextEncoder(d_model=1280, layers=32, heads=20)
@@ -133,6 +147,7 @@ def _build_tiny(name: str) -> ModelBundle:
 
 _BUILDERS: Dict[str, Callable[[str], ModelBundle]] = {
     "sd15": _build_sd15,
+    "sd21": _build_sd21,
     "sdxl": _build_sdxl,
     "tiny": _build_tiny,
     "tiny-xl": _build_tiny_xl,
@@ -266,10 +281,11 @@ def load_checkpoint(path: str, device="cpu", dtype=None) -> ModelBundle:
             ch = f.get_tensor("model.diffusion_model.input_blocks.0.0.weight"
                               ).shape[0]
             is_xl = any(k.startswith("conditioner.") for k in keys)
-            arch = (
-                ("sdxl" if is_xl else "sd15") if int(ch) == 320
-                else ("tiny-xl" if is_xl else "tiny")
-            )
+            is_sd2 = any(k.startswith("cond_stage_model.model.") for k in keys)
+            if int(ch) == 320:
+                arch = "sdxl" if is_xl else ("sd21" if is_sd2 else "sd15")
+            else:
+                arch = "tiny-xl" if is_xl else "tiny"
             bundle = _BUILDERS[arch](arch)
             load_ldm_state_dict(bundle, {k: f.get_tensor(k) for k in keys})
             bundle.eval().to(device, dtype)

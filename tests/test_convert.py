@@ -123,6 +123,58 @@ class TestSdxlRoundTrip:
             )
 
 
+class TestSd2RoundTrip:
+    @staticmethod
+    def _small_sd2_bundle():
+        """Hand-built SD2-shaped bundle: 1024-wide open_clip-style text
+        tower (the export heuristic keys off d_model == 1024), tiny UNet/VAE."""
+        import zlib
+
+        from sdwd_amd.models.clip import CLIPTextEncoder
+        from sdwd_amd.models.registry import ModelBundle, _seeded_init
+        from sdwd_amd.models.unet import UNetConfig, UNetModel
+        from sdwd_amd.models.vae import AutoencoderKL, VAEConfig
+
+        te = CLIPTextEncoder(d_model=1024, layers=2, heads=4)
+        unet = UNetModel(
+            UNetConfig(model_channels=32, channel_mult=[1, 2],
+                       num_res_blocks=1, transformer_depth=[1, 1],
+                       context_dim=1024, num_heads=2, groups=8)
+        )
+        vae = AutoencoderKL(VAEConfig.tiny())
+        for off, m in enumerate((te, unet, vae)):
+            _seeded_init(m, zlib.crc32(b"sd2test") % (2 ** 31) + off)
+        return ModelBundle("sd2test", te, None, unet, vae, context_dim=1024)
+
+    def test_sd2_layout_round_trip(self):
+        a = self._small_sd2_bundle()
+        exported = to_ldm_state_dict(a)
+        # SD2 naming: open_clip tower under cond_stage_model.model.
+        assert any(
+            k.startswith("cond_stage_model.model.transformer.resblocks.")
+            for k in exported
+        )
+        assert "cond_stage_model.model.text_projection" in exported
+        assert not any(
+            k.startswith("cond_stage_model.transformer.") for k in exported
+        )
+        b = self._small_sd2_bundle()
+        _perturb(b)
+        report = load_ldm_state_dict(b, exported)
+        assert not report["missing"], report["missing"][:5]
+        assert not report["unexpected"], report["unexpected"][:5]
+        for part in ("unet", "vae", "text_encoder"):
+            sa = getattr(a, part).state_dict()
+            sb = getattr(b, part).state_dict()
+            for k in sa:
+                assert torch.equal(sa[k], sb[k]), f"{part}.{k}"
+
+    def test_sd21_builder_registered(self):
+        from sdwd_amd.models.registry import available_models
+
+        assert "sd21" in available_models()
+
+
 class TestCheckpointFile:
     def test_ldm_safetensors_auto_detected(self, tmp_path):
         """An sdwui-format .safetensors file loads through the normal
