@@ -55,6 +55,9 @@ class Txt2ImgRequest(BaseModel):
     # execute the controlnet unit natively, other scripts are ignored
     # with a warning as the reference's compat filter did)
     alwayson_scripts: Dict[str, Any] = Field(default_factory=dict)
+    # sdwui refiner fields
+    refiner_checkpoint: str = ""
+    refiner_switch_at: float = 0.8
     # sdwui per-request overrides (sd_model_checkpoint,
     # CLIP_stop_at_last_layers are honored; the rest are ignored)
     override_settings: Dict[str, Any] = Field(default_factory=dict)
@@ -176,6 +179,10 @@ def create_app(engine: Optional[LocalEngine] = None,
         if model and model not in available_models():
             raise HTTPException(404, f"unknown model {model}")
         clip_skip = int(ov.get("CLIP_stop_at_last_layers") or req.clip_skip)
+        if req.refiner_checkpoint and (
+            req.refiner_checkpoint not in available_models()
+        ):
+            raise HTTPException(404, f"unknown refiner {req.refiner_checkpoint}")
         return model, clip_skip
 
     @app.post("/sdapi/v1/txt2img")
@@ -206,6 +213,8 @@ def create_app(engine: Optional[LocalEngine] = None,
             control_model=control_model,
             control_scale=control_scale,
             model=model,
+            refiner_model=req.refiner_checkpoint,
+            refiner_switch_at=req.refiner_switch_at,
         )
         return run_generation(gen)
 

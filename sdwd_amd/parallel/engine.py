@@ -58,6 +58,8 @@ class GenerationRequest:
     control_scale: float = 1.0
     clip_skip: int = 1
     model: str = ""  # hot-swap to this checkpoint first ("" = keep current)
+    refiner_model: str = ""  # two-model refiner handoff (sdwui refiner)
+    refiner_switch_at: float = 0.8
 
     def sched(self) -> GenRequest:
         return GenRequest(
@@ -111,6 +113,8 @@ def _job_pipeline_request(
         control_model=gen.control_model,
         control_scale=gen.control_scale,
         clip_skip=gen.clip_skip,
+        refiner_model=gen.refiner_model,
+        refiner_switch_at=gen.refiner_switch_at,
     )
 
 

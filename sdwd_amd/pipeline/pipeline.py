@@ -51,6 +51,11 @@ class PipelineRequest:
     # latent upscaler (sdwui hr_upscaler "Latent..." family):
     # nearest | bilinear | bicubic | bilinear-antialiased | bicubic-antialiased
     hr_upscaler: str = "nearest"
+    # two-model refiner (sdwui refiner_checkpoint/refiner_switch_at): the
+    # base model denoises the first switch_at fraction of steps, the
+    # refiner model finishes (both share the latent space / VAE)
+    refiner_model: str = ""
+    refiner_switch_at: float = 0.8
     # controlnet (ref C17 executed natively)
     control_image: Optional[torch.Tensor] = None  # [B,H,W,3] uint8
     control_model: str = ""  # e.g. "controlnet-sd15"; "" = off
@@ -167,12 +172,13 @@ class StableDiffusionPipeline:
     # -- conditioning --------------------------------------------------------
     @torch.no_grad()
     def encode_prompts(
-        self, prompts: List[str], negatives: List[str], clip_skip: int = 1
+        self, prompts: List[str], negatives: List[str], clip_skip: int = 1,
+        bundle=None,
     ) -> tuple:
         tokens, weights = tokenizer.encode_batch_weighted(
             prompts + negatives, device=self.device
         )
-        m = self.model
+        m = bundle if bundle is not None else self.model
         if m.is_sdxl:
             h1 = m.text_encoder(tokens, penultimate=True)
             h2 = m.text_encoder_2(tokens, penultimate=True)
@@ -357,6 +363,51 @@ class StableDiffusionPipeline:
                 )
                 yc = y[: x_in.shape[0]] if y is not None else None
                 return denoiser(x_in, ts, cond, yc)
+
+        if req.refiner_model:
+            # two-model handoff: t descends through the schedule, so the
+            # refiner takes over once t falls to the switch timestep
+            base_fn = model_fn
+            refiner = load_model(
+                req.refiner_model, device=self.device, dtype=self.dtype
+            )
+            if self.device.type == "cuda":
+                refiner.unet.to(memory_format=torch.channels_last)
+            r_c1, r_u1, r_p1 = self.encode_prompts(
+                [req.prompt], [req.negative_prompt], req.clip_skip,
+                bundle=refiner,
+            )
+            r_ctx = torch.cat(
+                [r_c1.expand(b, -1, -1), r_u1.expand(b, -1, -1)], dim=0
+            )
+            r_y = None
+            if refiner.is_sdxl and r_p1 is not None:
+                r_y = torch.cat(
+                    [
+                        self._sdxl_vector(req, r_p1[0].expand(b, -1)),
+                        self._sdxl_vector(req, r_p1[1].expand(b, -1)),
+                    ]
+                )
+            ts_list = sched.timesteps.tolist()
+            si = max(0, int(round(len(ts_list) * req.refiner_switch_at)))
+            # switch_at is the fraction of steps the BASE model runs;
+            # 1.0 means the refiner never fires
+            t_switch = ts_list[si] if si < len(ts_list) else float("-inf")
+            r_unet = refiner.unet
+
+            def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:  # noqa: F811
+                if t > t_switch:
+                    return base_fn(x_in, t)
+                ts = torch.full(
+                    (x_in.shape[0] * 2,), t, device=self.device,
+                    dtype=torch.float32,
+                )
+                x2 = torch.cat([x_in, x_in], dim=0)
+                eps = r_unet(x2, ts, r_ctx, y=r_y)
+                eps_c, eps_u = eps.chunk(2, dim=0)
+                from .. import ops as _ops
+
+                return _ops.lincomb(eps_c, eps_u, cfg, 1.0 - cfg)
 
         was_interrupted = False
 

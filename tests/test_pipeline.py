@@ -467,6 +467,45 @@ class TestInpainting:
         assert torch.equal(a, b)
 
 
+class TestRefiner:
+    def test_same_model_refiner_is_identity(self, pipe):
+        """refiner == base model: the handoff must not perturb the
+        trajectory (same unet, same conditioning -> bitwise-equal output)."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="r", steps=4, width=64, height=64, seeds=[21])
+        a = pipe.generate(PipelineRequest(**base)).images
+        b = pipe.generate(
+            PipelineRequest(**base, refiner_model="tiny",
+                            refiner_switch_at=0.5)
+        ).images
+        assert torch.equal(a, b)
+
+    def test_different_refiner_changes_output(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="r", steps=4, width=64, height=64, seeds=[21])
+        a = pipe.generate(PipelineRequest(**base)).images
+        b = pipe.generate(
+            PipelineRequest(**base, refiner_model="tiny-xl",
+                            refiner_switch_at=0.5)
+        ).images
+        assert a.shape == b.shape
+        assert not torch.equal(a, b)
+        assert torch.isfinite(b.float()).all()
+
+    def test_switch_at_one_never_refines(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="r", steps=4, width=64, height=64, seeds=[21])
+        a = pipe.generate(PipelineRequest(**base)).images
+        b = pipe.generate(
+            PipelineRequest(**base, refiner_model="tiny-xl",
+                            refiner_switch_at=1.0)
+        ).images
+        assert torch.equal(a, b)
+
+
 class TestPromptWeighting:
     def test_parse_weighted(self):
         from sdwd_amd.models.tokenizer import parse_weighted
