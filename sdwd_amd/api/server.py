@@ -304,7 +304,40 @@ def create_app(engine: Optional[LocalEngine] = None,
 
     @app.post("/sdapi/v1/refresh-loras")
     def refresh_loras():
-        return {}
+        from ..models.lora import refresh_lora_files
+
+        return {"found": sorted(refresh_lora_files())}
+
+    @app.get("/sdapi/v1/loras")
+    def loras():
+        """Available LoRA adapters: files from SDWD_LORA_DIR plus any
+        registered in the engine's manager (sdwui /sdapi/v1/loras shape)."""
+        from ..models.lora import lora_files, refresh_lora_files
+
+        if not lora_files():
+            refresh_lora_files()
+        files = lora_files()
+        names = set(files)
+        for pipe in engine.pipes.values():
+            names.update(pipe.lora._registry)
+        return [
+            {"name": n, "alias": n, "path": files.get(n, "")}
+            for n in sorted(names)
+        ]
+
+    @app.post("/sdapi/v1/png-info")
+    def png_info(body: Dict[str, Any]):
+        """Read back the 'parameters' infotext from a generated PNG
+        (sdwui /sdapi/v1/png-info)."""
+        from ..utils.images import png_parameters
+
+        data = body.get("image", "")
+        try:
+            raw = base64.b64decode(data.split(",", 1)[-1])
+            info = png_parameters(raw)
+        except Exception as exc:
+            raise HTTPException(422, f"bad png: {exc}")
+        return {"info": info or "", "items": {}}
 
     @app.get("/sdapi/v1/progress")
     def progress():

@@ -90,6 +90,35 @@ def save_lora_file(lora: LoRA, path: str) -> str:
     return path
 
 
+_lora_files: Dict[str, str] = {}
+
+
+def lora_dir() -> str:
+    import os
+
+    return os.environ.get("SDWD_LORA_DIR", "loras")
+
+
+def refresh_lora_files(dirpath: Optional[str] = None) -> Dict[str, str]:
+    """Scan the lora directory (sdwui's Lora folder + refresh-loras route);
+    *.safetensors become addressable <lora:stem:scale> names."""
+    import os
+
+    global _lora_files
+    d = dirpath or lora_dir()
+    found: Dict[str, str] = {}
+    if os.path.isdir(d):
+        for fn in sorted(os.listdir(d)):
+            if fn.endswith(".safetensors"):
+                found[os.path.splitext(fn)[0]] = os.path.join(d, fn)
+    _lora_files = found
+    return dict(found)
+
+
+def lora_files() -> Dict[str, str]:
+    return _lora_files
+
+
 class LoraManager:
     """Tracks the adapter set merged into one UNet; swaps sets reversibly."""
 
@@ -106,8 +135,12 @@ class LoraManager:
 
     def get(self, name: str) -> LoRA:
         if name not in self._registry:
-            # deterministic random fallback (no files in this environment)
-            self._registry[name] = make_random_lora(name, self.unet)
+            path = lora_files().get(name)
+            if path is not None:
+                self._registry[name] = load_lora_file(path)
+            else:
+                # deterministic random fallback (no files in this environment)
+                self._registry[name] = make_random_lora(name, self.unet)
         return self._registry[name]
 
     @torch.no_grad()

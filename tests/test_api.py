@@ -162,6 +162,37 @@ class TestControl:
         )
         assert r.status_code == 404
 
+    def test_png_info_round_trip(self, client):
+        from sdwd_amd.utils.images import encode_png
+
+        img = torch.zeros(8, 8, 3, dtype=torch.uint8)
+        data = base64.b64encode(
+            encode_png(img, parameters="a cow\nSteps: 4, Seed: 9")
+        ).decode()
+        r = client.post("/sdapi/v1/png-info", json={"image": data})
+        assert r.status_code == 200
+        assert "Steps: 4" in r.json()["info"]
+
+    def test_loras_routes(self, client, tmp_path, monkeypatch_module):
+        import torch as _t
+
+        from sdwd_amd.models.lora import (
+            make_random_lora, refresh_lora_files, save_lora_file,
+        )
+
+        d = tmp_path / "loras"
+        d.mkdir()
+        pipe = next(iter(client.app.state.engine.pipes.values()))
+        save_lora_file(
+            make_random_lora("filelora", pipe.model.unet), str(d / "filelora.safetensors")
+        )
+        monkeypatch_module.setenv("SDWD_LORA_DIR", str(d))
+        r = client.post("/sdapi/v1/refresh-loras")
+        assert r.json()["found"] == ["filelora"]
+        names = [e["name"] for e in client.get("/sdapi/v1/loras").json()]
+        assert "filelora" in names
+        refresh_lora_files(dirpath=str(tmp_path / "none"))
+
     def test_settings_rejects_unknown(self, client):
         r = client.post("/sdwd/settings", json={"warp_factor": 9})
         assert r.status_code == 422

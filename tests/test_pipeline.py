@@ -426,6 +426,26 @@ class TestLoRA:
         k = next(iter(lora.tensors))
         assert torch.equal(back.tensors[k][0], lora.tensors[k][0])
 
+    def test_file_backed_lora_resolves_from_dir(self, pipe, tmp_path):
+        """A <lora:name:scale> tag resolves to SDWD_LORA_DIR/name.safetensors
+        when present (not the deterministic random fallback)."""
+        from sdwd_amd.models.lora import (
+            make_random_lora, refresh_lora_files, save_lora_file,
+        )
+
+        lora = make_random_lora("diskvariant", pipe.model.unet)
+        k = next(iter(lora.tensors))
+        with torch.no_grad():
+            lora.tensors[k][0].mul_(2.0)  # make the file differ from fallback
+        save_lora_file(lora, str(tmp_path / "diskvariant.safetensors"))
+        refresh_lora_files(dirpath=str(tmp_path))
+        try:
+            got = pipe.lora.get("diskvariant")
+            assert torch.equal(got.tensors[k][0], lora.tensors[k][0])
+        finally:
+            refresh_lora_files(dirpath=str(tmp_path / "none"))
+            pipe.lora._registry.pop("diskvariant", None)
+
 
 class TestInpainting:
     def test_mask_preserves_unmasked_region(self, pipe):
