@@ -517,11 +517,36 @@ class StableDiffusionPipeline:
             images = x.cpu()
 
         elapsed = time.perf_counter() - t0
+        # sdwui infotext format: optional fields appear only when active,
+        # so downstream "send to txt2img"-style parsers reproduce the run
+        extra = ""
+        if req.clip_skip > 1:
+            extra += f", Clip skip: {req.clip_skip}"
+        if req.init_latents is not None:
+            extra += f", Denoising strength: {req.denoising_strength}"
+        if req.subseed_strength > 0 and req.subseeds:
+            extra += f", Variation seed strength: {req.subseed_strength}"
+        if req.enable_hr and req.hr_scale > 1.0:
+            extra += (
+                f", Hires upscale: {req.hr_scale}"
+                f", Hires steps: {req.hr_steps or req.steps}"
+                f", Hires upscaler: {req.hr_upscaler}"
+            )
+        if req.refiner_model:
+            extra += (
+                f", Refiner: {req.refiner_model}"
+                f", Refiner switch at: {req.refiner_switch_at}"
+            )
         infotexts = [
             f"{req.prompt}\nNegative prompt: {req.negative_prompt}\n"
             f"Steps: {req.steps}, Sampler: {req.sampler_name}, "
             f"CFG scale: {req.cfg_scale}, Seed: {req.seeds[i]}, "
             f"Size: {req.width}x{req.height}, Model: {self.model.name}"
+            + (
+                f", Variation seed: {subseeds[i]}"
+                if req.subseed_strength > 0 and subseeds[i] != -1 else ""
+            )
+            + extra
             for i in range(b)
         ]
         return PipelineResult(

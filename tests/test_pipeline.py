@@ -487,6 +487,35 @@ class TestInpainting:
         assert torch.equal(a, b)
 
 
+class TestInfotext:
+    def test_optional_fields_present_when_active(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        req = PipelineRequest(
+            prompt="c", steps=2, width=64, height=64, seeds=[7],
+            subseeds=[9], subseed_strength=0.4, clip_skip=2,
+            enable_hr=True, hr_scale=2.0, hr_steps=1,
+            hr_upscaler="Latent (bilinear)", denoising_strength=0.6,
+        )
+        info = pipe.generate(req).infotexts[0]
+        for frag in ("Clip skip: 2", "Variation seed: 9",
+                     "Variation seed strength: 0.4", "Hires upscale: 2.0",
+                     "Hires upscaler: Latent (bilinear)", "Model: tiny"):
+            assert frag in info, (frag, info)
+
+    def test_optional_fields_absent_by_default(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        info = pipe.generate(
+            PipelineRequest(prompt="c", steps=1, width=64, height=64,
+                            seeds=[7])
+        ).infotexts[0]
+        assert "Clip skip" not in info
+        assert "Hires" not in info
+        assert "Variation" not in info
+        assert "Refiner" not in info
+
+
 class TestRefiner:
     def test_same_model_refiner_is_identity(self, pipe):
         """refiner == base model: the handoff must not perturb the
