@@ -35,6 +35,9 @@ class ModelBundle:
     vae: AutoencoderKL
     context_dim: int
     is_sdxl: bool = False
+    # "eps" (noise prediction) or "v" (velocity, SD2.x-768 lineage);
+    # the pipeline converts v -> eps algebraically before the sampler
+    prediction_type: str = "eps"
 
     @property
     def latent_channels(self) -> int:
@@ -99,6 +102,20 @@ def _build_sd21(name: str) -> ModelBundle:
     return ModelBundle(name, te, None, unet, vae, context_dim=1024)
 
 
+def _build_sd21v(name: str) -> ModelBundle:
+    """SD2.1-768 lineage: same architecture as sd21, v-prediction."""
+    b = _build_sd21(name)
+    b.prediction_type = "v"
+    return b
+
+
+def _build_tiny_v(name: str) -> ModelBundle:
+    """CPU-test v-prediction model (tiny arch, velocity output)."""
+    b = _build_tiny(name)
+    b.prediction_type = "v"
+    return b
+
+
 def _build_sdxl(name: str) -> ModelBundle:
     te = CLIPTextEncoder(d_model=768, layers=12, heads=12)
     te2 = CLIPTextEncoder(d_model=1280, layers=32, heads=20)
@@ -148,7 +165,9 @@ def _build_tiny(name: str) -> ModelBundle:
 _BUILDERS: Dict[str, Callable[[str], ModelBundle]] = {
     "sd15": _build_sd15,
     "sd21": _build_sd21,
+    "sd21v": _build_sd21v,
     "sdxl": _build_sdxl,
+    "tiny-v": _build_tiny_v,
     "tiny": _build_tiny,
     "tiny-xl": _build_tiny_xl,
 }

@@ -22,7 +22,7 @@ from ..models import tokenizer
 from ..utils import get_logger
 from .graphs import GraphedDenoiser
 from .samplers import build_sampler
-from .schedule import schedule_for
+from .schedule import schedule_for, sigma_for_t
 
 log = get_logger("pipeline")
 
@@ -337,6 +337,20 @@ class StableDiffusionPipeline:
             if hint.shape[0] == 1 and b > 1:
                 hint = hint.expand(b, -1, -1, -1)
 
+        def _to_eps(out, x_scaled, t, pred_type):
+            """v-prediction -> epsilon on the sampler's scaled input:
+            denoised = c_skip*x + c_out*v gives eps = c_in*(sigma*x_scaled + v)
+            (c_skip = 1/(s^2+1), c_out = -s/sqrt(s^2+1), c_in = 1/sqrt(s^2+1))."""
+            if pred_type != "v":
+                return out
+            from .. import ops as _ops
+
+            s = sigma_for_t(t)
+            c_in = 1.0 / math.sqrt(s * s + 1.0)
+            return _ops.lincomb(x_scaled, out, s * c_in, c_in)
+
+        pred_type = self.model.prediction_type
+
         def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:
             ts = torch.full(
                 (x_in.shape[0] * 2,), t, device=self.device,
@@ -349,6 +363,7 @@ class StableDiffusionPipeline:
                 eps = unet(x2, ts, ctx, y=y, control=ctrl)
             else:
                 eps = denoiser(x2, ts, ctx, y)
+            eps = _to_eps(eps, x2, t, pred_type)
             eps_c, eps_u = eps.chunk(2, dim=0)
             from .. import ops as _ops
 
@@ -362,7 +377,9 @@ class StableDiffusionPipeline:
                     dtype=torch.float32,
                 )
                 yc = y[: x_in.shape[0]] if y is not None else None
-                return denoiser(x_in, ts, cond, yc)
+                return _to_eps(
+                    denoiser(x_in, ts, cond, yc), x_in, t, pred_type
+                )
 
         if req.refiner_model:
             # two-model handoff: t descends through the schedule, so the
@@ -404,6 +421,7 @@ class StableDiffusionPipeline:
                 )
                 x2 = torch.cat([x_in, x_in], dim=0)
                 eps = r_unet(x2, ts, r_ctx, y=r_y)
+                eps = _to_eps(eps, x2, t, refiner.prediction_type)
                 eps_c, eps_u = eps.chunk(2, dim=0)
                 from .. import ops as _ops
 

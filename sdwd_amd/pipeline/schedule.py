@@ -27,6 +27,23 @@ def make_sigmas_full() -> torch.Tensor:
     return ((1 - alphas_cum) / alphas_cum).sqrt()
 
 
+_table_cache: torch.Tensor | None = None
+
+
+def sigma_for_t(t: float) -> float:
+    """Inverse of _timesteps_for: fractional train-timestep -> sigma
+    (log-sigma linear interpolation on the train table)."""
+    global _table_cache
+    if _table_cache is None:
+        _table_cache = make_sigmas_full().log()
+    logt = _table_cache
+    lo = int(t)
+    lo = max(0, min(lo, len(logt) - 1))
+    hi = min(lo + 1, len(logt) - 1)
+    frac = float(t) - lo
+    return float((logt[lo] * (1 - frac) + logt[hi] * frac).exp())
+
+
 @dataclass
 class Schedule:
     sigmas: torch.Tensor  # [steps+1] descending, last = 0

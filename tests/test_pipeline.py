@@ -487,6 +487,56 @@ class TestInpainting:
         assert torch.equal(a, b)
 
 
+class TestVPrediction:
+    def test_v_to_eps_algebra(self):
+        """The pipeline's v->eps rewrite must reproduce the canonical
+        denoised = c_skip*x + c_out*v parametrization exactly."""
+        import math
+
+        from sdwd_amd.pipeline.schedule import sigma_for_t
+
+        for t in (10.0, 500.0, 981.5):
+            s = sigma_for_t(t)
+            x = torch.randn(2, 4, 8, 8, dtype=torch.float64)
+            v = torch.randn_like(x)
+            c_in = 1.0 / math.sqrt(s * s + 1.0)
+            x_scaled = x * c_in
+            eps = (s * c_in) * x_scaled + c_in * v  # pipeline formula
+            denoised_via_eps = x - s * eps
+            c_skip = 1.0 / (s * s + 1.0)
+            c_out = -s / math.sqrt(s * s + 1.0)
+            denoised_direct = c_skip * x + c_out * v
+            assert torch.allclose(denoised_via_eps, denoised_direct,
+                                  atol=1e-10), t
+
+    def test_sigma_for_t_inverts_timesteps(self):
+        from sdwd_amd.pipeline.schedule import (
+            discrete_schedule, sigma_for_t,
+        )
+
+        sched = discrete_schedule(12)
+        for sig, t in zip(sched.sigmas.tolist(), sched.timesteps.tolist()):
+            assert abs(sigma_for_t(t) - sig) / sig < 1e-6
+
+    def test_tiny_v_generates(self):
+        from sdwd_amd.models import load_model
+        from sdwd_amd.pipeline import PipelineRequest
+
+        assert load_model("tiny-v", cache=False).prediction_type == "v"
+        # same weights, different parametrization -> different trajectory
+        mv = load_model("tiny", cache=False)
+        mv.prediction_type = "v"
+        pv = StableDiffusionPipeline(mv, device="cpu")
+        req = PipelineRequest(prompt="v", steps=3, width=64, height=64,
+                              seeds=[4])
+        out_v = pv.generate(req).images
+        assert torch.isfinite(out_v.float()).all()
+        pe = StableDiffusionPipeline(load_model("tiny", cache=False),
+                                     device="cpu")
+        out_e = pe.generate(req).images
+        assert not torch.equal(out_v, out_e)
+
+
 class TestInfotext:
     def test_optional_fields_present_when_active(self, pipe):
         from sdwd_amd.pipeline import PipelineRequest
