@@ -219,6 +219,7 @@ def load_model(
     if name in _FILE_MODELS and name not in _BUILDERS:
         log.info("loading checkpoint file '%s'", _FILE_MODELS[name])
         bundle = load_checkpoint(_FILE_MODELS[name], device=device, dtype=dtype)
+        bundle.name = name  # address by file stem, not the inferred arch
         if cache:
             _cache[key] = bundle
         return bundle

@@ -253,6 +253,15 @@ class LocalEngine(_EngineBase):
     ) -> None:
         worker = self.world.get_worker(job.worker_label)
         pipe = self.pipes[job.worker_label]
+        # per-worker checkpoint override (ref ui.py:161-171: a remote could
+        # pin its own model); rebuilt lazily when it changes
+        want = worker.model_override or self.model_name
+        if pipe.model.name != want:
+            log.info("%s: model override -> %s", job.worker_label, want)
+            pipe = StableDiffusionPipeline(
+                want, device=pipe.device, dtype=pipe.dtype
+            )
+            self.pipes[job.worker_label] = pipe
         try:
             if self._fail_injection.pop(job.worker_label, False):
                 raise RuntimeError("injected failure")
@@ -542,6 +551,18 @@ class DistributedEngine(_EngineBase):
             plan = None
         model_name, jobs = pg.broadcast_object(plan)
         self.set_model(model_name)
+        # per-worker checkpoint override (ref ui.py:161-171): purely local —
+        # no collective weight sync, the deterministic registry (or the
+        # checkpoint file) guarantees any rank builds identical weights
+        override = getattr(
+            self.world.get_worker(self.label), "model_override", None
+        )
+        want = override or self.model_name
+        if self.pipe.model.name != want:
+            log.info("rank %d: model override -> %s", self.rank, want)
+            self.pipe = StableDiffusionPipeline(
+                want, device=self.device, dtype=self._dtype
+            )
 
         mine = next((j for j in jobs if j.worker_label == self.label), None)
         hf = gen.hr_scale if gen.enable_hr else 1.0

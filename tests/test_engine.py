@@ -242,6 +242,26 @@ class TestEdgeCases:
         )
         assert eng.pipes["gpu0"] is p
 
+    def test_per_worker_model_override(self):
+        """ref ui.py:161-171: a worker can pin its own checkpoint; the
+        shard still joins the shared gallery."""
+        eng = make_engine(2)
+        eng.world.get_worker("gpu1").model_override = "tiny-xl"
+        res = eng.generate(
+            GenerationRequest(prompt="ov", batch_size=2, width=64, height=64,
+                              steps=2, seed=31)
+        )
+        assert res.images.shape == (2, 64, 64, 3)
+        assert eng.pipes["gpu0"].model.name == "tiny"
+        assert eng.pipes["gpu1"].model.name == "tiny-xl"
+        # removing the override reverts on the next run
+        eng.world.get_worker("gpu1").model_override = None
+        eng.generate(
+            GenerationRequest(prompt="ov", batch_size=2, width=64, height=64,
+                              steps=1, seed=32)
+        )
+        assert eng.pipes["gpu1"].model.name == "tiny"
+
     def test_batch_smaller_than_ranks(self):
         eng = make_engine(3)
         res = eng.generate(
