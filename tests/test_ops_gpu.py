@@ -426,3 +426,47 @@ class TestAllSamplersGPU:
                                 seeds=[1], sampler_name=name)
             )
             assert torch.isfinite(res.images.float()).all(), name
+
+
+class TestVaeDownsampleGPU:
+    def test_asym_pad_encode_matches_cpu_fp32(self, dev):
+        """VAE encode (ldm asymmetric-pad downsample) on GPU bf16 vs CPU
+        fp32: the sampling grid must agree (only precision noise differs)."""
+        from sdwd_amd.pipeline import StableDiffusionPipeline
+
+        gp = StableDiffusionPipeline("tiny", device=dev)
+        cp = StableDiffusionPipeline("tiny", device="cpu",
+                                     dtype=torch.float32)
+        img = torch.randint(0, 255, (2, 64, 64, 3), dtype=torch.uint8)
+        lg = gp.encode_image(img, seeds=[1, 2]).float().cpu()
+        lc = cp.encode_image(img, seeds=[1, 2]).float()
+        assert lg.shape == lc.shape == (2, 4, 32, 32)
+        rel = (lg - lc).abs().mean() / lc.abs().mean().clamp_min(1e-6)
+        assert rel < 0.15, f"latent grids diverge: rel={rel}"
+
+
+class TestRefinerGPU:
+    def test_same_model_refiner_identity_on_gpu(self, dev):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny", device=dev)
+        base = dict(prompt="r", steps=3, width=64, height=64, seeds=[7])
+        a = pipe.generate(PipelineRequest(**base)).images
+        b = pipe.generate(
+            PipelineRequest(**base, refiner_model="tiny",
+                            refiner_switch_at=0.5)
+        ).images
+        assert torch.equal(a, b)
+
+
+class TestVPredictionGPU:
+    def test_v_model_generates_on_gpu(self, dev):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny-v", device=dev)
+        res = pipe.generate(
+            PipelineRequest(prompt="v", steps=3, width=64, height=64,
+                            seeds=[3])
+        )
+        assert torch.isfinite(res.images.float()).all()
+        assert res.images.float().std() > 1.0
