@@ -44,6 +44,7 @@ class Txt2ImgRequest(BaseModel):
     n_iter: int = Field(default=1, ge=1, le=64)
     sampler_name: str = "Euler a"
     sampler_index: Optional[str] = None  # legacy alias
+    scheduler: str = "Automatic"
     clip_skip: int = 1
     # hires fix (sdwui fields)
     enable_hr: bool = False
@@ -200,6 +201,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             steps=req.steps,
             cfg_scale=req.cfg_scale,
             sampler_name=req.sampler_name or req.sampler_index or "Euler a",
+            scheduler=req.scheduler,
             seed=req.seed,
             subseed=req.subseed,
             subseed_strength=req.subseed_strength,
@@ -246,6 +248,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             steps=req.steps,
             cfg_scale=req.cfg_scale,
             sampler_name=req.sampler_name or "Euler a",
+            scheduler=req.scheduler,
             seed=req.seed,
             subseed=req.subseed,
             subseed_strength=req.subseed_strength,
@@ -281,6 +284,15 @@ def create_app(engine: Optional[LocalEngine] = None,
     @app.get("/sdapi/v1/samplers")
     def samplers():
         return [{"name": s, "aliases": [s]} for s in sampler_names()]
+
+    @app.get("/sdapi/v1/schedulers")
+    def schedulers():
+        from ..pipeline.schedule import scheduler_names
+
+        return [
+            {"name": n.lower().replace(" ", "_"), "label": n}
+            for n in scheduler_names()
+        ]
 
     @app.get("/sdapi/v1/memory")
     def memory():

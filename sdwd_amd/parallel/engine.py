@@ -43,6 +43,7 @@ class GenerationRequest:
     steps: int = 20
     cfg_scale: float = 7.0
     sampler_name: str = "Euler a"
+    scheduler: str = "Automatic"
     seed: int = -1
     subseed: int = -1
     subseed_strength: float = 0.0
@@ -99,6 +100,7 @@ def _job_pipeline_request(
         height=gen.height,
         cfg_scale=gen.cfg_scale,
         sampler_name=gen.sampler_name,
+        scheduler=gen.scheduler,
         seeds=list(job.seeds),
         subseeds=list(job.subseeds),
         subseed_strength=gen.subseed_strength,

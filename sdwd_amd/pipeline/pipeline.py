@@ -36,6 +36,7 @@ class PipelineRequest:
     height: int = 512
     cfg_scale: float = 7.0
     sampler_name: str = "Euler a"
+    scheduler: str = "Automatic"  # sdwui scheduler dropdown (sigma spacing)
     seeds: List[int] = field(default_factory=lambda: [0])
     subseeds: List[int] = field(default_factory=list)
     subseed_strength: float = 0.0
@@ -265,7 +266,7 @@ class StableDiffusionPipeline:
                 ]
             )
 
-        sched = schedule_for(req.sampler_name, req.steps)
+        sched = schedule_for(req.sampler_name, req.steps, req.scheduler)
         sampler = build_sampler(req.sampler_name, sched)
 
         noise = torch.stack(
@@ -472,7 +473,7 @@ class StableDiffusionPipeline:
             x = _upscale_latent(
                 x.float(), req.hr_scale, req.hr_upscaler
             ).to(self.dtype)
-            hsched = schedule_for(req.sampler_name, hr_steps)
+            hsched = schedule_for(req.sampler_name, hr_steps, req.scheduler)
             start = max(
                 0, hr_steps - max(1, int(hr_steps * req.denoising_strength))
             )

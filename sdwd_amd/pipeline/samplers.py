@@ -326,6 +326,18 @@ class UniPC(Sampler):
         return x_c
 
 
+class LCM(Sampler):
+    """Latent-consistency sampling: jump straight to the denoised estimate,
+    then re-noise to the next sigma (sdwui's LCM sampler shape)."""
+
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
+        denoised = _eval(model_fn, x, sigma, t)
+        if sigma_next <= 0 or noise_fn is None:
+            return denoised
+        return ops.add_noise(denoised, noise_fn(), 1.0, sigma_next)
+
+
 class LMS(Sampler):
     """Linear multistep (order <= 4) with exactly integrated Adams
     coefficients over each sigma interval (k-diffusion sample_lms)."""
@@ -385,6 +397,7 @@ SAMPLERS: Dict[str, type] = {
     "DPM++ 2S a": DPMpp2SAncestral,
     "DPM++ 2S a Karras": DPMpp2SAncestral,
     "UniPC": UniPC,
+    "LCM": LCM,
 }
 
 

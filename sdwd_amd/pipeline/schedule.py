@@ -6,6 +6,7 @@ epsilon-predictor taking the *scaled* input x/sqrt(sigma^2+1) and the
 """
 from __future__ import annotations
 
+import math
 from dataclasses import dataclass
 
 import torch
@@ -88,7 +89,56 @@ def karras_schedule(steps: int, rho: float = 7.0) -> Schedule:
     return Schedule(sigmas=sigmas.float(), timesteps=ts.float())
 
 
-def schedule_for(sampler_name: str, steps: int) -> Schedule:
+def exponential_schedule(steps: int) -> Schedule:
+    """Log-linear sigmas (k-diffusion get_sigmas_exponential)."""
+    table = make_sigmas_full()
+    smin, smax = float(table[0]), float(table[-1])
+    sig = torch.linspace(
+        math.log(smax), math.log(smin), steps, dtype=torch.float64
+    ).exp()
+    ts = _timesteps_for(sig, table)
+    sigmas = torch.cat([sig, torch.zeros(1, dtype=torch.float64)])
+    return Schedule(sigmas=sigmas.float(), timesteps=ts.float())
+
+
+def sgm_uniform_schedule(steps: int) -> Schedule:
+    """Uniform in train-timestep WITHOUT the final zero-sigma training step
+    (sgm convention: endpoints chosen so the last denoise lands on the
+    lowest-noise trained step)."""
+    table = make_sigmas_full()
+    idx = torch.linspace(
+        TRAIN_STEPS - 1, 0, steps + 1, dtype=torch.float64
+    )[:-1].round().long()
+    sig = table[idx]
+    sigmas = torch.cat([sig, torch.zeros(1, dtype=torch.float64)])
+    return Schedule(sigmas=sigmas.float(), timesteps=idx.double().float())
+
+
+SCHEDULERS = {
+    "automatic": None,  # resolved from the sampler name
+    "uniform": discrete_schedule,
+    "karras": karras_schedule,
+    "exponential": exponential_schedule,
+    "sgm uniform": sgm_uniform_schedule,
+    "sgm_uniform": sgm_uniform_schedule,
+}
+
+
+def scheduler_names() -> list:
+    return ["Automatic", "Uniform", "Karras", "Exponential", "SGM Uniform"]
+
+
+def schedule_for(
+    sampler_name: str, steps: int, scheduler: str = "Automatic"
+) -> Schedule:
+    """sdwui scheduler dropdown semantics: an explicit scheduler wins;
+    "Automatic" keeps the sampler-name convention ("... Karras")."""
+    key = (scheduler or "Automatic").strip().lower()
+    fn = SCHEDULERS.get(key)
+    if fn is not None:
+        return fn(steps)
+    if key not in SCHEDULERS:
+        raise KeyError(f"unknown scheduler '{scheduler}'")
     if "Karras" in sampler_name:
         return karras_schedule(steps)
     return discrete_schedule(steps)

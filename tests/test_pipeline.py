@@ -487,6 +487,71 @@ class TestInpainting:
         assert torch.equal(a, b)
 
 
+class TestSchedulers:
+    def test_exponential_and_sgm_monotone(self):
+        from sdwd_amd.pipeline.schedule import (
+            exponential_schedule, sgm_uniform_schedule,
+        )
+
+        for fn in (exponential_schedule, sgm_uniform_schedule):
+            s = fn(12)
+            assert len(s.sigmas) == 13
+            assert s.sigmas[-1] == 0
+            assert (s.sigmas[:-1].diff() < 0).all(), fn.__name__
+
+    def test_explicit_scheduler_overrides_name(self):
+        from sdwd_amd.pipeline.schedule import (
+            karras_schedule, schedule_for,
+        )
+
+        k = schedule_for("Euler", 10, "Karras")
+        assert torch.equal(k.sigmas, karras_schedule(10).sigmas)
+        # Automatic keeps the sampler-name convention
+        k2 = schedule_for("DPM++ 2M Karras", 10, "Automatic")
+        assert torch.equal(k2.sigmas, karras_schedule(10).sigmas)
+
+    def test_unknown_scheduler_raises(self):
+        from sdwd_amd.pipeline.schedule import schedule_for
+
+        with pytest.raises(KeyError):
+            schedule_for("Euler", 10, "Quantum")
+
+    def test_scheduler_changes_output(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="s", steps=4, width=64, height=64, seeds=[6])
+        a = pipe.generate(PipelineRequest(**base)).images
+        b = pipe.generate(
+            PipelineRequest(**base, scheduler="Exponential")
+        ).images
+        assert not torch.equal(a, b)
+
+    def test_lcm_point_mass(self):
+        """LCM with an exact-denoiser oracle lands on x0 at sigma 0."""
+        from sdwd_amd.pipeline.samplers import build_sampler
+        from sdwd_amd.pipeline.schedule import (
+            discrete_schedule, make_sigmas_full,
+        )
+
+        table = make_sigmas_full()
+        x0 = torch.full((1, 4, 8, 8), 0.4)
+        sched = discrete_schedule(6)
+        sampler = build_sampler("LCM", sched)
+
+        def model_fn(x_scaled, t):
+            sigma = float(table[int(round(t))])
+            c_in = 1.0 / (sigma * sigma + 1.0) ** 0.5
+            x = x_scaled.float() / c_in
+            return (x - x0) / sigma
+
+        g = torch.Generator().manual_seed(3)
+        x = torch.randn(x0.shape, generator=g) * float(sched.sigmas[0])
+        out = sampler.sample(
+            model_fn, x, noise_fn=lambda: torch.randn(x0.shape, generator=g)
+        )
+        assert (out - x0).abs().max() < 1e-5
+
+
 class TestVPrediction:
     def test_v_to_eps_algebra(self):
         """The pipeline's v->eps rewrite must reproduce the canonical
