@@ -285,6 +285,39 @@ def create_app(engine: Optional[LocalEngine] = None,
     def samplers():
         return [{"name": s, "aliases": [s]} for s in sampler_names()]
 
+    @app.post("/sdapi/v1/extra-single-image")
+    def extra_single_image(body: Dict[str, Any]):
+        """Pixel-space upscale (sdwui extras tab surface; the model-free
+        upscalers: Nearest / Bilinear / Bicubic / Lanczos-approx)."""
+        try:
+            img = _decode_b64_png(body.get("image", ""))
+        except Exception as exc:
+            raise HTTPException(422, f"bad image: {exc}")
+        scale = float(body.get("upscaling_resize", 2.0))
+        w2 = int(body.get("upscaling_resize_w", 0))
+        h2 = int(body.get("upscaling_resize_h", 0))
+        mode = str(body.get("upscaler_1", "Bilinear")).lower()
+        mode_map = {
+            "nearest": ("nearest", False),
+            "none": ("nearest", False),
+            "bilinear": ("bilinear", True),
+            "bicubic": ("bicubic", True),
+            "lanczos": ("bicubic", True),  # closest torch kernel
+        }
+        m, aa = mode_map.get(mode, ("bilinear", True))
+        x = img.permute(2, 0, 1)[None].float()
+        kwargs: Dict[str, Any] = {"mode": m}
+        if aa:
+            kwargs["antialias"] = True
+        if w2 and h2:
+            out = torch.nn.functional.interpolate(x, size=(h2, w2), **kwargs)
+        else:
+            out = torch.nn.functional.interpolate(
+                x, scale_factor=scale, **kwargs
+            )
+        out8 = out.clamp(0, 255).to(torch.uint8)[0].permute(1, 2, 0)
+        return {"image": _b64_png(out8), "html_info": ""}
+
     @app.get("/sdapi/v1/schedulers")
     def schedulers():
         from ..pipeline.schedule import scheduler_names

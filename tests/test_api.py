@@ -162,6 +162,34 @@ class TestControl:
         )
         assert r.status_code == 404
 
+    def test_extras_upscale(self, client):
+        from sdwd_amd.utils.images import decode_png, encode_png
+
+        img = torch.randint(0, 255, (16, 16, 3), dtype=torch.uint8)
+        data = base64.b64encode(encode_png(img)).decode()
+        r = client.post(
+            "/sdapi/v1/extra-single-image",
+            json={"image": data, "upscaling_resize": 2.0,
+                  "upscaler_1": "Nearest"},
+        )
+        assert r.status_code == 200
+        out = decode_png(base64.b64decode(r.json()["image"]))
+        assert out.shape == (32, 32, 3)
+        # nearest 2x: every 2x2 block equals the source pixel
+        assert torch.equal(out[::2, ::2], img)
+        # explicit target size
+        r2 = client.post(
+            "/sdapi/v1/extra-single-image",
+            json={"image": data, "upscaling_resize_w": 24,
+                  "upscaling_resize_h": 40, "upscaler_1": "Bicubic"},
+        )
+        out2 = decode_png(base64.b64decode(r2.json()["image"]))
+        assert out2.shape == (40, 24, 3)
+
+    def test_schedulers_route(self, client):
+        names = [s["label"] for s in client.get("/sdapi/v1/schedulers").json()]
+        assert "Karras" in names and "Exponential" in names
+
     def test_png_info_round_trip(self, client):
         from sdwd_amd.utils.images import encode_png
 
