@@ -9,7 +9,7 @@ from __future__ import annotations
 
 import enum
 import threading
-from typing import Callable, Dict, List, Optional, Set
+from typing import Callable, Dict, List, Set
 
 
 class State(enum.Enum):

@@ -11,7 +11,6 @@ instead of POST /options.
 from __future__ import annotations
 
 import threading
-import time
 from typing import Callable, Optional, Tuple
 
 from ..config import (

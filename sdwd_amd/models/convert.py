@@ -132,8 +132,6 @@ def unet_key_map(unet) -> Dict[str, str]:
 
 def vae_key_map(vae) -> Dict[str, str]:
     """-> {ldm_key (no prefix): native_key}. quant convs handled separately."""
-    from .vae import VAEResBlock
-
     cfg = vae.cfg
     out: Dict[str, str] = {}
     levels = len(cfg.channel_mult)

@@ -7,7 +7,7 @@ from __future__ import annotations
 import math
 import struct
 import zlib
-from typing import List, Optional
+from typing import Optional
 
 import torch
 
