@@ -38,6 +38,16 @@ PAGE = """<!DOCTYPE html>
 <h2>utils</h2>
 <button onclick="fetch('/sdwd/benchmark',{method:'POST'})">re-benchmark</button>
 <button onclick="fetch('/sdwd/sync-script',{method:'POST'})">run sync script</button>
+<h2>settings</h2>
+<form onsubmit="saveSettings(event)" id="settingsForm">
+ job timeout <input id="s_job_timeout" size="4"/>
+ <label><input type="checkbox" id="s_complement_production"/>complementary</label>
+ <label><input type="checkbox" id="s_step_scaling"/>step scaling</label>
+ <label><input type="checkbox" id="s_thin_client"/>thin client</label>
+ <label><input type="checkbox" id="s_distribute_txt2img"/>dist txt2img</label>
+ <label><input type="checkbox" id="s_distribute_img2img"/>dist img2img</label>
+ <button>save</button> <span id="s_saved"></span>
+</form>
 <h2>log</h2><div id="log"></div>
 <script>
 async function refresh(){
@@ -77,7 +87,30 @@ async function gen(ev){
   document.getElementById('gallery').innerHTML =
     out.images.map(b => `<img src="data:image/png;base64,${b}"/>`).join('');
 }
+const BOOL_SETTINGS = ['complement_production','step_scaling','thin_client',
+                       'distribute_txt2img','distribute_img2img'];
+async function loadSettings(){
+  try{
+    const s = await (await fetch('/sdwd/settings')).json();
+    document.getElementById('s_job_timeout').value = s.job_timeout;
+    for(const k of BOOL_SETTINGS)
+      document.getElementById('s_'+k).checked = !!s[k];
+  }catch(e){}
+}
+async function saveSettings(ev){
+  ev.preventDefault();
+  const body = {job_timeout: parseFloat(
+    document.getElementById('s_job_timeout').value)};
+  for(const k of BOOL_SETTINGS)
+    body[k] = document.getElementById('s_'+k).checked;
+  const r = await fetch('/sdwd/settings',{method:'POST',
+    headers:{'Content-Type':'application/json'}, body: JSON.stringify(body)});
+  document.getElementById('s_saved').textContent =
+    r.ok ? 'saved' : 'error';
+  setTimeout(()=>{document.getElementById('s_saved').textContent='';}, 2000);
+}
 setInterval(refresh, 1500);  // ref distributed.js:7-23 auto-refresh cadence
 refresh();
+loadSettings();
 </script>
 </body></html>"""

@@ -285,6 +285,7 @@ class TestWebUI:
         assert r.status_code == 200
         assert "sdwd_amd" in r.text
         assert "/sdwd/status" in r.text
+        assert "/sdwd/settings" in r.text  # settings tab (ref Settings tab)
 
 
 class TestRestart:
