@@ -245,9 +245,16 @@ class StableDiffusionPipeline:
         lat_c = self.model.latent_channels
         subseeds = req.subseeds or [-1] * b
 
+        # prompt editing ([from:to:when] / [a|b]) -> per-step text segments;
+        # the common single-segment case stays on the encode-once path
+        from .prompt_schedule import prompt_schedule
+
+        p_segs = prompt_schedule(req.prompt, req.steps)
+        n_segs = prompt_schedule(req.negative_prompt, req.steps)
+
         # one prompt per request: encode once, broadcast across the batch
         cond1, uncond1, pooled1 = self.encode_prompts(
-            [req.prompt], [req.negative_prompt], req.clip_skip
+            [p_segs[0][1]], [n_segs[0][1]], req.clip_skip
         )
         cond = cond1.expand(b, -1, -1)
         uncond = uncond1.expand(b, -1, -1)
@@ -320,6 +327,48 @@ class StableDiffusionPipeline:
         cfg = float(req.cfg_scale)
         unet = self.model.unet
 
+        # prompt-editing segments: (t_threshold, ctx, y) per conditioning
+        # change, selected by the sampler's current timestep in model_fn
+        seg_tensors: List[tuple] = []
+        if len(p_segs) > 1 or len(n_segs) > 1:
+            ts_all = sched.timesteps.tolist()
+            boundaries = sorted(
+                {s for s, _ in p_segs} | {s for s, _ in n_segs}
+            )
+            enc_cache: dict = {}
+            for s in boundaries:
+                if s >= len(ts_all):
+                    continue
+                p = [txt for st, txt in p_segs if st <= s][-1]
+                n = [txt for st, txt in n_segs if st <= s][-1]
+                if (p, n) not in enc_cache:
+                    c1, u1, pl1 = self.encode_prompts(
+                        [p], [n], req.clip_skip
+                    )
+                    ctx_b = torch.cat(
+                        [c1.expand(b, -1, -1), u1.expand(b, -1, -1)], dim=0
+                    )
+                    y_b = None
+                    if self.model.is_sdxl and pl1 is not None:
+                        y_b = torch.cat(
+                            [
+                                self._sdxl_vector(req, pl1[0].expand(b, -1)),
+                                self._sdxl_vector(req, pl1[1].expand(b, -1)),
+                            ]
+                        )
+                    enc_cache[(p, n)] = (ctx_b, y_b)
+                ctx_b, y_b = enc_cache[(p, n)]
+                seg_tensors.append((ts_all[s], ctx_b, y_b))
+
+        def _ctx_y_for(t: float):
+            if not seg_tensors:
+                return ctx, y
+            sel = seg_tensors[0]
+            for s in seg_tensors:
+                if t <= s[0] + 1e-6:
+                    sel = s
+            return sel[1], sel[2]
+
         denoiser = self._denoiser
 
         controlnet = None
@@ -353,6 +402,7 @@ class StableDiffusionPipeline:
         pred_type = self.model.prediction_type
 
         def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:
+            c_ctx, c_y = _ctx_y_for(t)
             ts = torch.full(
                 (x_in.shape[0] * 2,), t, device=self.device,
                 dtype=torch.float32,
@@ -360,10 +410,10 @@ class StableDiffusionPipeline:
             x2 = torch.cat([x_in, x_in], dim=0)
             if controlnet is not None:
                 h2 = torch.cat([hint, hint], dim=0)
-                ctrl = controlnet(x2, h2, ts, ctx, req.control_scale)
-                eps = unet(x2, ts, ctx, y=y, control=ctrl)
+                ctrl = controlnet(x2, h2, ts, c_ctx, req.control_scale)
+                eps = unet(x2, ts, c_ctx, y=c_y, control=ctrl)
             else:
-                eps = denoiser(x2, ts, ctx, y)
+                eps = denoiser(x2, ts, c_ctx, c_y)
             eps = _to_eps(eps, x2, t, pred_type)
             eps_c, eps_u = eps.chunk(2, dim=0)
             from .. import ops as _ops
@@ -373,13 +423,14 @@ class StableDiffusionPipeline:
         if cfg == 1.0:
 
             def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:  # noqa: F811
+                c_ctx, c_y = _ctx_y_for(t)
+                nb = x_in.shape[0]
                 ts = torch.full(
-                    (x_in.shape[0],), t, device=self.device,
-                    dtype=torch.float32,
+                    (nb,), t, device=self.device, dtype=torch.float32,
                 )
-                yc = y[: x_in.shape[0]] if y is not None else None
+                yc = c_y[:nb] if c_y is not None else None
                 return _to_eps(
-                    denoiser(x_in, ts, cond, yc), x_in, t, pred_type
+                    denoiser(x_in, ts, c_ctx[:nb], yc), x_in, t, pred_type
                 )
 
         if req.refiner_model:

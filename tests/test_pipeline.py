@@ -487,6 +487,65 @@ class TestInpainting:
         assert torch.equal(a, b)
 
 
+class TestPromptEditing:
+    def test_schedule_parsing(self):
+        from sdwd_amd.pipeline.prompt_schedule import (
+            prompt_at_step, prompt_schedule,
+        )
+
+        assert prompt_schedule("a [cat:dog:0.5] x", 10) == [
+            (0, "a cat x"), (5, "a dog x"),
+        ]
+        # absolute-step threshold
+        assert prompt_at_step("[a::3] b", 2, 10) == "a b"
+        assert prompt_at_step("[a::3] b", 3, 10) == " b"
+        # late addition [to:when]
+        assert prompt_at_step("[x:0.5]", 2, 10) == ""
+        assert prompt_at_step("[x:0.5]", 7, 10) == "x"
+        # alternation
+        assert prompt_at_step("[a|b]", 0, 4) == "a"
+        assert prompt_at_step("[a|b]", 1, 4) == "b"
+        assert prompt_at_step("[a|b]", 2, 4) == "a"
+        # plain attention brackets untouched
+        assert prompt_at_step("a [word] b", 1, 4) == "a [word] b"
+
+    def test_single_segment_fast_path(self):
+        from sdwd_amd.pipeline.prompt_schedule import prompt_schedule
+
+        assert prompt_schedule("plain prompt", 20) == [(0, "plain prompt")]
+
+    def test_editing_changes_output(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(steps=4, width=64, height=64, seeds=[12])
+        edited = pipe.generate(
+            PipelineRequest(prompt="a [cat:dog:0.5] x", **base)
+        ).images
+        all_cat = pipe.generate(
+            PipelineRequest(prompt="a cat x", **base)
+        ).images
+        all_dog = pipe.generate(
+            PipelineRequest(prompt="a dog x", **base)
+        ).images
+        assert not torch.equal(edited, all_cat)
+        assert not torch.equal(edited, all_dog)
+        # deterministic
+        again = pipe.generate(
+            PipelineRequest(prompt="a [cat:dog:0.5] x", **base)
+        ).images
+        assert torch.equal(edited, again)
+
+    def test_editing_at_one_is_first_prompt(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(steps=3, width=64, height=64, seeds=[2])
+        a = pipe.generate(
+            PipelineRequest(prompt="a [cat:dog:1.0] x", **base)
+        ).images
+        c = pipe.generate(PipelineRequest(prompt="a cat x", **base)).images
+        assert torch.equal(a, c)
+
+
 class TestSchedulers:
     def test_exponential_and_sgm_monotone(self):
         from sdwd_amd.pipeline.schedule import (
