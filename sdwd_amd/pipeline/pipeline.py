@@ -221,6 +221,33 @@ class StableDiffusionPipeline:
             [pooled.float(), emb[None].expand(b, -1)], dim=-1
         ).to(self.dtype)
 
+    def _build_ctx(self, p_text: str, n_text: str, req: "PipelineRequest",
+                   b: int):
+        """Conditioning for one prompt segment: handles AND composition.
+
+        Returns (ctx, y, weights): ctx rows are [cond_1..cond_k, uncond],
+        each broadcast to the batch; weights are the AND sub-prompt weights
+        (k == 1, weight 1.0 for a plain prompt)."""
+        from .prompt_schedule import split_and
+
+        parts = split_and(p_text)
+        prompts = [t for t, _ in parts]
+        ws = [w for _, w in parts]
+        c, u, pl = self.encode_prompts(prompts, [n_text], req.clip_skip)
+        rows = [c[i : i + 1].expand(b, -1, -1) for i in range(len(prompts))]
+        rows.append(u.expand(b, -1, -1))
+        ctx = torch.cat(rows, dim=0)
+        y = None
+        if self.model.is_sdxl and pl is not None:
+            pc, pu = pl
+            yrows = [
+                self._sdxl_vector(req, pc[i : i + 1].expand(b, -1))
+                for i in range(len(prompts))
+            ]
+            yrows.append(self._sdxl_vector(req, pu.expand(b, -1)))
+            y = torch.cat(yrows)
+        return ctx, y, ws
+
     # -- the denoise loop ----------------------------------------------------
     @torch.no_grad()
     def generate(
@@ -252,26 +279,11 @@ class StableDiffusionPipeline:
         p_segs = prompt_schedule(req.prompt, req.steps)
         n_segs = prompt_schedule(req.negative_prompt, req.steps)
 
-        # one prompt per request: encode once, broadcast across the batch
-        cond1, uncond1, pooled1 = self.encode_prompts(
-            [p_segs[0][1]], [n_segs[0][1]], req.clip_skip
+        # one conditioning set per request (encode once, broadcast across
+        # the batch); AND composition yields k cond rows + 1 uncond row
+        ctx, y, and_ws = self._build_ctx(
+            p_segs[0][1], n_segs[0][1], req, b
         )
-        cond = cond1.expand(b, -1, -1)
-        uncond = uncond1.expand(b, -1, -1)
-        pooled_cu = None
-        if pooled1 is not None:
-            pooled_cu = (
-                pooled1[0].expand(b, -1),
-                pooled1[1].expand(b, -1),
-            )
-        y = None
-        if self.model.is_sdxl and pooled_cu is not None:
-            y = torch.cat(
-                [
-                    self._sdxl_vector(req, pooled_cu[0]),
-                    self._sdxl_vector(req, pooled_cu[1]),
-                ]
-            )
 
         sched = schedule_for(req.sampler_name, req.steps, req.scheduler)
         sampler = build_sampler(req.sampler_name, sched)
@@ -323,12 +335,11 @@ class StableDiffusionPipeline:
             )
             return n.to(self.device, self.dtype)
 
-        ctx = torch.cat([cond, uncond], dim=0)
         cfg = float(req.cfg_scale)
         unet = self.model.unet
 
-        # prompt-editing segments: (t_threshold, ctx, y) per conditioning
-        # change, selected by the sampler's current timestep in model_fn
+        # prompt-editing segments: (t_threshold, ctx, y, weights) per
+        # conditioning change, selected by the current timestep in model_fn
         seg_tensors: List[tuple] = []
         if len(p_segs) > 1 or len(n_segs) > 1:
             ts_all = sched.timesteps.tolist()
@@ -342,32 +353,18 @@ class StableDiffusionPipeline:
                 p = [txt for st, txt in p_segs if st <= s][-1]
                 n = [txt for st, txt in n_segs if st <= s][-1]
                 if (p, n) not in enc_cache:
-                    c1, u1, pl1 = self.encode_prompts(
-                        [p], [n], req.clip_skip
-                    )
-                    ctx_b = torch.cat(
-                        [c1.expand(b, -1, -1), u1.expand(b, -1, -1)], dim=0
-                    )
-                    y_b = None
-                    if self.model.is_sdxl and pl1 is not None:
-                        y_b = torch.cat(
-                            [
-                                self._sdxl_vector(req, pl1[0].expand(b, -1)),
-                                self._sdxl_vector(req, pl1[1].expand(b, -1)),
-                            ]
-                        )
-                    enc_cache[(p, n)] = (ctx_b, y_b)
-                ctx_b, y_b = enc_cache[(p, n)]
-                seg_tensors.append((ts_all[s], ctx_b, y_b))
+                    enc_cache[(p, n)] = self._build_ctx(p, n, req, b)
+                ctx_b, y_b, ws_b = enc_cache[(p, n)]
+                seg_tensors.append((ts_all[s], ctx_b, y_b, ws_b))
 
         def _ctx_y_for(t: float):
             if not seg_tensors:
-                return ctx, y
+                return ctx, y, and_ws
             sel = seg_tensors[0]
             for s in seg_tensors:
                 if t <= s[0] + 1e-6:
                     sel = s
-            return sel[1], sel[2]
+            return sel[1], sel[2], sel[3]
 
         denoiser = self._denoiser
 
@@ -402,35 +399,43 @@ class StableDiffusionPipeline:
         pred_type = self.model.prediction_type
 
         def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:
-            c_ctx, c_y = _ctx_y_for(t)
+            c_ctx, c_y, ws = _ctx_y_for(t)
+            k1 = len(ws) + 1  # k AND-conds + 1 uncond
+            nb = x_in.shape[0]
             ts = torch.full(
-                (x_in.shape[0] * 2,), t, device=self.device,
-                dtype=torch.float32,
+                (nb * k1,), t, device=self.device, dtype=torch.float32,
             )
-            x2 = torch.cat([x_in, x_in], dim=0)
+            xk = torch.cat([x_in] * k1, dim=0)
             if controlnet is not None:
-                h2 = torch.cat([hint, hint], dim=0)
-                ctrl = controlnet(x2, h2, ts, c_ctx, req.control_scale)
-                eps = unet(x2, ts, c_ctx, y=c_y, control=ctrl)
+                hk = torch.cat([hint] * k1, dim=0)
+                ctrl = controlnet(xk, hk, ts, c_ctx, req.control_scale)
+                eps = unet(xk, ts, c_ctx, y=c_y, control=ctrl)
             else:
-                eps = denoiser(x2, ts, c_ctx, c_y)
-            eps = _to_eps(eps, x2, t, pred_type)
-            eps_c, eps_u = eps.chunk(2, dim=0)
+                eps = denoiser(xk, ts, c_ctx, c_y)
+            eps = _to_eps(eps, xk, t, pred_type)
+            parts = eps.chunk(k1, dim=0)
             from .. import ops as _ops
 
-            return _ops.lincomb(eps_c, eps_u, cfg, 1.0 - cfg)
+            if k1 == 2:
+                w = cfg * ws[0]
+                return _ops.lincomb(parts[0], parts[1], w, 1.0 - w)
+            # composable diffusion: u + cfg * sum_i w_i (c_i - u)
+            out = parts[-1]
+            for w, ec in zip(ws, parts[:-1]):
+                d = _ops.lincomb(ec, parts[-1], 1.0, -1.0)
+                out = _ops.lincomb(out, d, 1.0, cfg * w)
+            return out
 
-        if cfg == 1.0:
+        if cfg == 1.0 and not seg_tensors and and_ws == [1.0]:
 
             def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:  # noqa: F811
-                c_ctx, c_y = _ctx_y_for(t)
                 nb = x_in.shape[0]
                 ts = torch.full(
                     (nb,), t, device=self.device, dtype=torch.float32,
                 )
-                yc = c_y[:nb] if c_y is not None else None
+                yc = y[:nb] if y is not None else None
                 return _to_eps(
-                    denoiser(x_in, ts, c_ctx[:nb], yc), x_in, t, pred_type
+                    denoiser(x_in, ts, ctx[:nb], yc), x_in, t, pred_type
                 )
 
         if req.refiner_model:

@@ -14,6 +14,23 @@ import re
 from typing import List, Tuple
 
 _inner = re.compile(r"\[([^\[\]]*)\]")
+
+
+def split_and(text: str) -> List[Tuple[str, float]]:
+    """sdwui composable diffusion: "a AND b:0.6" -> [("a",1.0),("b",0.6)].
+    The optional trailing ``:number`` on each sub-prompt is its weight."""
+    parts = []
+    for sub in text.split(" AND "):
+        head, sep, tail = sub.rpartition(":")
+        w = 1.0
+        if sep and head.strip():
+            try:
+                w = float(tail.strip())
+                sub = head
+            except ValueError:
+                pass
+        parts.append((sub.strip(), w))
+    return parts or [(text, 1.0)]
 # sentinels for brackets we must preserve (attention syntax)
 _L, _R = "\x01", "\x02"
 

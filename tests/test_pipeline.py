@@ -546,6 +546,49 @@ class TestPromptEditing:
         assert torch.equal(a, c)
 
 
+class TestComposableAnd:
+    def test_split_and(self):
+        from sdwd_amd.pipeline.prompt_schedule import split_and
+
+        assert split_and("a cat AND a dog") == [("a cat", 1.0), ("a dog", 1.0)]
+        assert split_and("a cat AND a dog:0.4") == [
+            ("a cat", 1.0), ("a dog", 0.4),
+        ]
+        assert split_and("plain") == [("plain", 1.0)]
+        # attention/editing colons are not weights
+        assert split_and("a (cat:1.3)") == [("a (cat:1.3)", 1.0)]
+        assert split_and("[a:b:0.5]") == [("[a:b:0.5]", 1.0)]
+
+    def test_and_changes_output(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(steps=3, width=64, height=64, seeds=[8])
+        ab = pipe.generate(
+            PipelineRequest(prompt="a cat AND a dog", **base)
+        ).images
+        a = pipe.generate(PipelineRequest(prompt="a cat", **base)).images
+        assert not torch.equal(ab, a)
+        assert torch.isfinite(ab.float()).all()
+        again = pipe.generate(
+            PipelineRequest(prompt="a cat AND a dog", **base)
+        ).images
+        assert torch.equal(ab, again)
+
+    def test_zero_weight_equals_single(self, pipe):
+        """AND with weight 0 on the second prompt reduces to plain CFG of
+        the first (up to batch-blocking last-bit quantization)."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(steps=3, width=64, height=64, seeds=[8])
+        ab = pipe.generate(
+            PipelineRequest(prompt="a cat AND a dog:0.0", **base)
+        ).images
+        a = pipe.generate(PipelineRequest(prompt="a cat", **base)).images
+        diff = (ab.float() - a.float()).abs()
+        assert diff.max() <= 1.0
+        assert (diff > 0).float().mean() < 0.02
+
+
 class TestSchedulers:
     def test_exponential_and_sgm_monotone(self):
         from sdwd_amd.pipeline.schedule import (
