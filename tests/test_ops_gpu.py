@@ -470,3 +470,21 @@ class TestVPredictionGPU:
         )
         assert torch.isfinite(res.images.float()).all()
         assert res.images.float().std() > 1.0
+
+
+class TestPromptFeaturesGPU:
+    def test_editing_and_composition_on_gpu(self, dev):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny", device=dev)
+        base = dict(steps=3, width=64, height=64, seeds=[5])
+        edited = pipe.generate(
+            PipelineRequest(prompt="a [cat:dog:0.5] x", **base)
+        ).images
+        comp = pipe.generate(
+            PipelineRequest(prompt="a cat AND a dog:0.5", **base)
+        ).images
+        plain = pipe.generate(PipelineRequest(prompt="a cat", **base)).images
+        for out in (edited, comp):
+            assert torch.isfinite(out.float()).all()
+            assert not torch.equal(out, plain)
