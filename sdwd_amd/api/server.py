@@ -45,6 +45,7 @@ class Txt2ImgRequest(BaseModel):
     sampler_name: str = "Euler a"
     sampler_index: Optional[str] = None  # legacy alias
     scheduler: str = "Automatic"
+    styles: List[str] = Field(default_factory=list)
     clip_skip: int = 1
     # hires fix (sdwui fields)
     enable_hr: bool = False
@@ -180,6 +181,14 @@ def create_app(engine: Optional[LocalEngine] = None,
         if model and model not in available_models():
             raise HTTPException(404, f"unknown model {model}")
         clip_skip = int(ov.get("CLIP_stop_at_last_layers") or req.clip_skip)
+        if req.styles:
+            from ..pipeline.styles import all_styles, apply_styles, refresh_styles
+
+            if not all_styles():
+                refresh_styles()
+            req.prompt, req.negative_prompt = apply_styles(
+                req.prompt, req.negative_prompt, req.styles
+            )
         if req.refiner_checkpoint and (
             req.refiner_checkpoint not in available_models()
         ):
@@ -317,6 +326,17 @@ def create_app(engine: Optional[LocalEngine] = None,
             )
         out8 = out.clamp(0, 255).to(torch.uint8)[0].permute(1, 2, 0)
         return {"image": _b64_png(out8), "html_info": ""}
+
+    @app.get("/sdapi/v1/prompt-styles")
+    def prompt_styles():
+        from ..pipeline.styles import all_styles, refresh_styles
+
+        if not all_styles():
+            refresh_styles()
+        return [
+            {"name": n, "prompt": p, "negative_prompt": np}
+            for n, (p, np) in sorted(all_styles().items())
+        ]
 
     @app.get("/sdapi/v1/schedulers")
     def schedulers():

@@ -162,6 +162,39 @@ class TestControl:
         )
         assert r.status_code == 404
 
+    def test_styles(self, client, tmp_path, monkeypatch_module):
+        from sdwd_amd.pipeline.styles import apply_styles, refresh_styles
+
+        p = tmp_path / "styles.csv"
+        p.write_text(
+            'name,prompt,negative_prompt\n'
+            'cinematic,"{prompt}, film grain, 35mm",blurry\n'
+            'plainsuffix,"golden hour",\n'
+        )
+        monkeypatch_module.setenv("SDWD_STYLES_FILE", str(p))
+        try:
+            names = refresh_styles()
+            assert names == ["cinematic", "plainsuffix"]
+            pr, ng = apply_styles("a cow", "bad", ["cinematic"])
+            assert pr == "a cow, film grain, 35mm"
+            assert ng == "bad, blurry"
+            pr2, _ = apply_styles("a cow", "", ["plainsuffix"])
+            assert pr2 == "a cow, golden hour"
+            r = client.get("/sdapi/v1/prompt-styles")
+            assert any(e["name"] == "cinematic" for e in r.json())
+            # end-to-end through txt2img (styled prompt lands in infotext)
+            import json as _json
+
+            resp = client.post(
+                "/sdapi/v1/txt2img",
+                json={"prompt": "a cow", "steps": 1, "width": 64,
+                      "height": 64, "seed": 5, "styles": ["cinematic"]},
+            )
+            info = _json.loads(resp.json()["info"])
+            assert "film grain" in info["infotexts"][0]
+        finally:
+            refresh_styles(path=str(tmp_path / "none.csv"))
+
     def test_extras_upscale(self, client):
         from sdwd_amd.utils.images import decode_png, encode_png
 
