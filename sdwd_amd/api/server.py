@@ -11,6 +11,7 @@ from __future__ import annotations
 
 import base64
 import json
+import os
 import threading
 import time
 from typing import Any, Dict, List, Optional
@@ -130,6 +131,36 @@ def create_app(engine: Optional[LocalEngine] = None,
     state = ServerState(engine)
     app = FastAPI(title="sdwd_amd", version="0.1.0")
     app.state.engine = engine
+
+    # HTTP basic auth, sdwui --api-auth parity (the reference's worker
+    # records carried user:pass credentials, pmodels.py:12-34):
+    # SDWD_API_AUTH="user:pass" protects every route.
+    auth_cfg = os.environ.get("SDWD_API_AUTH", "")
+    if auth_cfg and ":" in auth_cfg:
+        import hmac
+        import base64 as _b64
+
+        expected = auth_cfg.encode()
+
+        @app.middleware("http")
+        async def _basic_auth(request, call_next):
+            from fastapi.responses import JSONResponse
+
+            hdr = request.headers.get("authorization", "")
+            ok = False
+            if hdr.lower().startswith("basic "):
+                try:
+                    got = _b64.b64decode(hdr[6:])
+                    ok = hmac.compare_digest(got, expected)
+                except Exception:
+                    ok = False
+            if not ok:
+                return JSONResponse(
+                    {"detail": "unauthorized"},
+                    status_code=401,
+                    headers={"WWW-Authenticate": "Basic"},
+                )
+            return await call_next(request)
 
     def run_generation(gen: GenerationRequest) -> Dict[str, Any]:
         # one generation at a time (the reference serialized on the host's

@@ -88,7 +88,7 @@ def _transformer_map(st, ours: str, ldm: str, out: Dict[str, str]) -> None:
         out[f"{l}.ff.net.2.bias"] = f"{o}.ff.1.bias"
 
 
-def _seq_map(seq, ours: str, ldm: str, upsample: bool,
+def _seq_map(seq, ours: str, ldm: str,
              out: Dict[str, str]) -> None:
     from .unet import Downsample, ResBlock, SpatialTransformer, Upsample
 
@@ -123,10 +123,10 @@ def unet_key_map(unet) -> Dict[str, str]:
         out[f"out.0.{p}"] = f"norm_out.{p}"
         out[f"out.2.{p}"] = f"conv_out.{p}"
     for n, seq in enumerate(unet.down):
-        _seq_map(seq, f"down.{n}", f"input_blocks.{n + 1}", False, out)
-    _seq_map(unet.mid, "mid", "middle_block", False, out)
+        _seq_map(seq, f"down.{n}", f"input_blocks.{n + 1}", out)
+    _seq_map(unet.mid, "mid", "middle_block", out)
     for n, seq in enumerate(unet.up):
-        _seq_map(seq, f"up.{n}", f"output_blocks.{n}", True, out)
+        _seq_map(seq, f"up.{n}", f"output_blocks.{n}", out)
     return out
 
 

@@ -162,6 +162,25 @@ class TestControl:
         )
         assert r.status_code == 404
 
+    def test_basic_auth(self, monkeypatch):
+        import base64 as b64
+
+        from fastapi.testclient import TestClient
+
+        from sdwd_amd.api import create_app
+        from sdwd_amd.parallel import LocalEngine
+
+        monkeypatch.setenv("SDWD_API_AUTH", "user:secret")
+        eng = LocalEngine(model="tiny", devices=["cpu"])
+        c = TestClient(create_app(engine=eng))
+        assert c.get("/sdapi/v1/sd-models").status_code == 401
+        hdr = {"Authorization": "Basic " + b64.b64encode(
+            b"user:secret").decode()}
+        assert c.get("/sdapi/v1/sd-models", headers=hdr).status_code == 200
+        bad = {"Authorization": "Basic " + b64.b64encode(
+            b"user:wrong").decode()}
+        assert c.get("/sdapi/v1/sd-models", headers=bad).status_code == 401
+
     def test_styles(self, client, tmp_path, monkeypatch_module):
         from sdwd_amd.pipeline.styles import apply_styles, refresh_styles
 

@@ -29,6 +29,12 @@ def main() -> int:
     ap.add_argument("--height", type=int, default=512)
     ap.add_argument("--cfg", type=float, default=7.0)
     ap.add_argument("--sampler", default="Euler a")
+    ap.add_argument("--scheduler", default="Automatic",
+                    help="Uniform | Karras | Exponential | SGM Uniform")
+    ap.add_argument("--clip-skip", type=int, default=1)
+    ap.add_argument("--hr-upscaler", default="Latent")
+    ap.add_argument("--refiner", default="", help="refiner model name")
+    ap.add_argument("--refiner-switch-at", type=float, default=0.8)
     ap.add_argument("--seed", type=int, default=-1)
     ap.add_argument("--model", default="sd15")
     ap.add_argument("--init", default="", help="PNG for img2img")
@@ -62,11 +68,16 @@ def main() -> int:
             steps=args.steps,
             cfg_scale=args.cfg,
             sampler_name=args.sampler,
+            scheduler=args.scheduler,
+            clip_skip=args.clip_skip,
             seed=args.seed,
             init_images=init_images,
             denoising_strength=args.strength,
             enable_hr=args.hires,
             hr_scale=args.hr_scale,
+            hr_upscaler=args.hr_upscaler,
+            refiner_model=args.refiner,
+            refiner_switch_at=args.refiner_switch_at,
         )
     )
     os.makedirs(args.out, exist_ok=True)
