@@ -70,6 +70,8 @@ class Img2ImgRequest(Txt2ImgRequest):
     init_images: List[str] = Field(default_factory=list)  # base64 PNG
     denoising_strength: float = 0.75
     mask: Optional[str] = None  # base64 PNG, white = repaint
+    inpaint_full_res: bool = False  # sdwui "Inpaint area: Only masked"
+    inpaint_full_res_padding: int = 32
 
 
 class OptionsRequest(BaseModel):
@@ -297,6 +299,8 @@ def create_app(engine: Optional[LocalEngine] = None,
             mask_image=mask_image,
             clip_skip=clip_skip,
             model=model,
+            inpaint_full_res=req.inpaint_full_res,
+            inpaint_full_res_padding=req.inpaint_full_res_padding,
         )
         return run_generation(gen)
 

@@ -17,7 +17,7 @@ from __future__ import annotations
 
 import threading
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass, field, replace as replace_dc
 from typing import Dict, List, Optional
 
 import torch
@@ -61,6 +61,10 @@ class GenerationRequest:
     model: str = ""  # hot-swap to this checkpoint first ("" = keep current)
     refiner_model: str = ""  # two-model refiner handoff (sdwui refiner)
     refiner_switch_at: float = 0.8
+    # sdwui "Inpaint area: Only masked": crop the padded mask bbox, run the
+    # generation on the crop at full W x H, paste the result back
+    inpaint_full_res: bool = False
+    inpaint_full_res_padding: int = 32
 
     def sched(self) -> GenRequest:
         return GenRequest(
@@ -118,6 +122,92 @@ def _job_pipeline_request(
         refiner_model=gen.refiner_model,
         refiner_switch_at=gen.refiner_switch_at,
     )
+
+
+def _crop_for_inpaint_full_res(gen: GenerationRequest):
+    """-> (cropped_gen, paste_ctx) or (gen, None).
+
+    Crops the padded mask bounding box (expanded toward the W:H aspect),
+    resizes init+mask crops to the full generation size, and returns what
+    _paste_inpaint_full_res needs to put results back (sdwui
+    "Only masked" semantics)."""
+    import torch.nn.functional as F
+
+    if not (
+        gen.inpaint_full_res
+        and gen.init_images is not None
+        and gen.mask_image is not None
+    ):
+        return gen, None
+    mask = gen.mask_image
+    if mask.dim() == 3:
+        mask = mask[0]
+    h, w = mask.shape
+    ys, xs = (mask > 127).nonzero(as_tuple=True)
+    if len(ys) == 0:
+        return gen, None
+    pad = max(0, int(gen.inpaint_full_res_padding))
+    y0, y1 = max(0, int(ys.min()) - pad), min(h, int(ys.max()) + 1 + pad)
+    x0, x1 = max(0, int(xs.min()) - pad), min(w, int(xs.max()) + 1 + pad)
+    # expand the short side toward the generation aspect ratio
+    target_ar = gen.width / gen.height
+    ch, cw = y1 - y0, x1 - x0
+    if cw / ch < target_ar:  # too narrow -> widen
+        want = min(w, int(round(ch * target_ar)))
+        grow = want - cw
+        x0 = max(0, x0 - grow // 2)
+        x1 = min(w, x0 + want)
+        x0 = max(0, x1 - want)
+    elif cw / ch > target_ar:  # too short -> heighten
+        want = min(h, int(round(cw / target_ar)))
+        grow = want - ch
+        y0 = max(0, y0 - grow // 2)
+        y1 = min(h, y0 + want)
+        y0 = max(0, y1 - want)
+
+    def _resize(img_f32, size):
+        return F.interpolate(
+            img_f32, size=size, mode="bilinear", antialias=True
+        )
+
+    inits = gen.init_images.float().permute(0, 3, 1, 2)  # [B,3,H,W]
+    crop = _resize(
+        inits[:, :, y0:y1, x0:x1], (gen.height, gen.width)
+    ).clamp(0, 255).to(torch.uint8).permute(0, 2, 3, 1)
+    mcrop = _resize(
+        mask[None, None, y0:y1, x0:x1].float(), (gen.height, gen.width)
+    )[0, 0].clamp(0, 255).to(torch.uint8)
+    new_gen = replace_dc(gen, init_images=crop, mask_image=mcrop)
+    paste_ctx = {
+        "orig_inits": gen.init_images,
+        "box": (y0, y1, x0, x1),
+        "mask": (mask[y0:y1, x0:x1].float() / 255.0),
+    }
+    return new_gen, paste_ctx
+
+
+def _paste_inpaint_full_res(result: "GalleryResult", paste_ctx) -> None:
+    """Paste generated crops back into the original images (in place)."""
+    import torch.nn.functional as F
+
+    if paste_ctx is None:
+        return
+    y0, y1, x0, x1 = paste_ctx["box"]
+    m = paste_ctx["mask"][..., None]  # [ch,cw,1] in 0..1
+    orig = paste_ctx["orig_inits"]
+    n_orig = orig.shape[0]
+    out = []
+    for i in range(result.images.shape[0]):
+        base = orig[i % n_orig].float().clone()
+        region = F.interpolate(
+            result.images[i].float().permute(2, 0, 1)[None],
+            size=(y1 - y0, x1 - x0), mode="bilinear", antialias=True,
+        )[0].permute(1, 2, 0)
+        patch = base[y0:y1, x0:x1]
+        base[y0:y1, x0:x1] = m * region + (1.0 - m) * patch
+        out.append(base.clamp(0, 255).to(torch.uint8))
+    result.images = torch.stack(out)
+    result.grid = make_grid(result.images) if len(out) > 1 else None
 
 
 class _EngineBase:
@@ -313,6 +403,7 @@ class LocalEngine(_EngineBase):
     def generate(self, gen: GenerationRequest) -> GalleryResult:
         t0 = time.perf_counter()
         self.set_model(gen.model)
+        gen, paste_ctx = _crop_for_inpaint_full_res(gen)
         self.world.clear_interrupt()
         jobs = self.world.make_jobs(gen.sched())
         self._live_jobs = jobs
@@ -358,6 +449,7 @@ class LocalEngine(_EngineBase):
         result = self._assemble(
             gen, jobs, shards, infos, time.perf_counter() - t0, interrupted
         )
+        _paste_inpaint_full_res(result, paste_ctx)
         self._autosave()
         return result
 
@@ -565,6 +657,7 @@ class DistributedEngine(_EngineBase):
             self.pipe = StableDiffusionPipeline(
                 want, device=self.device, dtype=self._dtype
             )
+        gen, paste_ctx = _crop_for_inpaint_full_res(gen)
 
         mine = next((j for j in jobs if j.worker_label == self.label), None)
         hf = gen.hr_scale if gen.enable_hr else 1.0
@@ -663,6 +756,8 @@ class DistributedEngine(_EngineBase):
                     res = self.pipe.generate(_job_pipeline_request(gen, job))
                     shards[job.worker_label] = res.images
 
-        return self._assemble(
+        result = self._assemble(
             gen, jobs, shards, infos, time.perf_counter() - t0, interrupted
         )
+        _paste_inpaint_full_res(result, paste_ctx)
+        return result

@@ -242,6 +242,43 @@ class TestEdgeCases:
         )
         assert eng.pipes["gpu0"] is p
 
+    def test_inpaint_full_res_preserves_outside(self):
+        """sdwui "Only masked": pixels outside the padded crop box equal
+        the original init EXACTLY (paste, not regeneration)."""
+        eng = make_engine(1)
+        init = torch.zeros(1, 64, 64, 3, dtype=torch.uint8)
+        init[:, :, :32] = 50
+        init[:, :, 32:] = 200
+        mask = torch.zeros(64, 64, dtype=torch.uint8)
+        mask[24:40, 24:40] = 255
+        res = eng.generate(
+            GenerationRequest(
+                prompt="patch", batch_size=1, width=64, height=64, steps=2,
+                seed=3, init_images=init, mask_image=mask,
+                denoising_strength=1.0, inpaint_full_res=True,
+                inpaint_full_res_padding=8,
+            )
+        )
+        assert res.images.shape == (1, 64, 64, 3)
+        # crop box is [16:48, 16:48]; everything outside is untouched
+        out = res.images[0]
+        assert torch.equal(out[:16], init[0, :16])
+        assert torch.equal(out[48:], init[0, 48:])
+        assert torch.equal(out[:, :16], init[0][:, :16])
+        assert torch.equal(out[:, 48:], init[0][:, 48:])
+        # inside the mask something was generated
+        assert not torch.equal(out[24:40, 24:40], init[0, 24:40, 24:40])
+        # deterministic
+        res2 = eng.generate(
+            GenerationRequest(
+                prompt="patch", batch_size=1, width=64, height=64, steps=2,
+                seed=3, init_images=init, mask_image=mask,
+                denoising_strength=1.0, inpaint_full_res=True,
+                inpaint_full_res_padding=8,
+            )
+        )
+        assert torch.equal(res.images, res2.images)
+
     def test_per_worker_model_override(self):
         """ref ui.py:161-171: a worker can pin its own checkpoint; the
         shard still joins the shared gallery."""
