@@ -47,6 +47,7 @@ class Txt2ImgRequest(BaseModel):
     sampler_index: Optional[str] = None  # legacy alias
     scheduler: str = "Automatic"
     styles: List[str] = Field(default_factory=list)
+    tiling: bool = False
     clip_skip: int = 1
     # hires fix (sdwui fields)
     enable_hr: bool = False
@@ -72,6 +73,9 @@ class Img2ImgRequest(Txt2ImgRequest):
     mask: Optional[str] = None  # base64 PNG, white = repaint
     inpaint_full_res: bool = False  # sdwui "Inpaint area: Only masked"
     inpaint_full_res_padding: int = 32
+    mask_blur: int = 4
+    # sdwui resize_mode: 0 just resize, 1 crop and resize, 2 resize and fill
+    resize_mode: int = 0
 
 
 class OptionsRequest(BaseModel):
@@ -99,6 +103,40 @@ def _decode_b64_png(data: str) -> torch.Tensor:
     if "," in data[:64]:  # data URI prefix
         data = data.split(",", 1)[1]
     return decode_png(base64.b64decode(data))
+
+
+def _resize_init(img: torch.Tensor, w: int, h: int, mode: int) -> torch.Tensor:
+    """sdwui img2img resize modes for an init image of any size:
+    0 = just resize (stretch), 1 = crop and resize (cover, center-crop),
+    2 = resize and fill (contain, edges replicated)."""
+    import torch.nn.functional as F
+
+    ih, iw = img.shape[0], img.shape[1]
+    if (ih, iw) == (h, w):
+        return img
+    x = img.float().permute(2, 0, 1)[None]
+
+    def interp(t, size):
+        return F.interpolate(t, size=size, mode="bilinear", antialias=True)
+
+    if mode == 1:  # cover then center-crop
+        scale = max(w / iw, h / ih)
+        rh, rw = int(round(ih * scale)), int(round(iw * scale))
+        x = interp(x, (rh, rw))
+        top, left = (rh - h) // 2, (rw - w) // 2
+        x = x[:, :, top : top + h, left : left + w]
+    elif mode == 2:  # contain then replicate-pad
+        scale = min(w / iw, h / ih)
+        rh, rw = int(round(ih * scale)), int(round(iw * scale))
+        x = interp(x, (rh, rw))
+        pt = (h - rh) // 2
+        pl = (w - rw) // 2
+        x = F.pad(
+            x, (pl, w - rw - pl, pt, h - rh - pt), mode="replicate"
+        )
+    else:  # just resize
+        x = interp(x, (h, w))
+    return x[0].permute(1, 2, 0).clamp(0, 255).to(torch.uint8)
 
 
 def _parse_controlnet(alwayson: Dict[str, Any]):
@@ -247,6 +285,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             seed=req.seed,
             subseed=req.subseed,
             subseed_strength=req.subseed_strength,
+            tiling=req.tiling,
             enable_hr=req.enable_hr,
             hr_scale=req.hr_scale,
             hr_steps=req.hr_second_pass_steps,
@@ -269,16 +308,23 @@ def create_app(engine: Optional[LocalEngine] = None,
         model, clip_skip = _overrides(req)
         try:
             inits = torch.stack(
-                [_decode_b64_png(d) for d in req.init_images]
+                [
+                    _resize_init(
+                        _decode_b64_png(d), req.width, req.height,
+                        req.resize_mode,
+                    )
+                    for d in req.init_images
+                ]
             )
         except Exception as exc:
             raise HTTPException(422, f"bad init image: {exc}")
         mask_image = None
         if req.mask:
             try:
-                mask_image = _decode_b64_png(req.mask).float().mean(-1).to(
-                    torch.uint8
-                )
+                m = _decode_b64_png(req.mask)
+                mask_image = _resize_init(
+                    m, req.width, req.height, 0
+                ).float().mean(-1).to(torch.uint8)
             except Exception as exc:
                 raise HTTPException(422, f"bad mask image: {exc}")
         gen = GenerationRequest(
@@ -294,6 +340,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             seed=req.seed,
             subseed=req.subseed,
             subseed_strength=req.subseed_strength,
+            tiling=req.tiling,
             init_images=inits,
             denoising_strength=req.denoising_strength,
             mask_image=mask_image,
@@ -301,6 +348,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             model=model,
             inpaint_full_res=req.inpaint_full_res,
             inpaint_full_res_padding=req.inpaint_full_res_padding,
+            mask_blur=req.mask_blur,
         )
         return run_generation(gen)
 

@@ -83,6 +83,20 @@ class SDConv2d(nn.Conv2d):
         from .. import ops
 
         k = self.kernel_size
+        if getattr(self, "circular", False) and self.padding != (0, 0):
+            # seamless-tiling mode (sdwui "Tiling"): wrap-around padding;
+            # runs the library conv (the HIP kernel is zero-pad only)
+            ph, pw = self.padding
+            xp = F.pad(x, (pw, pw, ph, ph), mode="circular")
+            out = F.conv2d(
+                xp, self.weight, self.bias, self.stride, 0,
+                self.dilation, self.groups,
+            )
+            if chan_bias is not None:
+                out = out + chan_bias.to(out.dtype)[:, :, None, None]
+            if residual is not None:
+                out = out + residual
+            return out
         if x.is_cuda and x.dtype == torch.bfloat16:
             if (
                 k == (3, 3)

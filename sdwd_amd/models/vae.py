@@ -96,7 +96,10 @@ class VAEDownsample(nn.Module):
         self.conv = SDConv2d(ch, ch, 3, stride=2, padding=0)
 
     def forward(self, x):
-        return self.conv(torch.nn.functional.pad(x, (0, 1, 0, 1)))
+        mode = (
+            "circular" if getattr(self.conv, "circular", False) else "constant"
+        )
+        return self.conv(torch.nn.functional.pad(x, (0, 1, 0, 1), mode=mode))
 
 
 class VAEEncoder(nn.Module):

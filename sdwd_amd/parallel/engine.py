@@ -65,6 +65,8 @@ class GenerationRequest:
     # generation on the crop at full W x H, paste the result back
     inpaint_full_res: bool = False
     inpaint_full_res_padding: int = 32
+    mask_blur: int = 0  # gaussian blur radius on the mask (sdwui mask_blur)
+    tiling: bool = False
 
     def sched(self) -> GenRequest:
         return GenRequest(
@@ -121,7 +123,38 @@ def _job_pipeline_request(
         clip_skip=gen.clip_skip,
         refiner_model=gen.refiner_model,
         refiner_switch_at=gen.refiner_switch_at,
+        tiling=gen.tiling,
     )
+
+
+def _blur_mask(mask: torch.Tensor, radius: int) -> torch.Tensor:
+    """Separable gaussian blur on a [H,W] uint8 mask (sdwui mask_blur:
+    softens the inpaint boundary in both the latent mask and the paste)."""
+    import torch.nn.functional as F
+
+    if radius <= 0:
+        return mask
+    sigma = max(0.5, radius / 2.0)
+    k = int(2 * round(3 * sigma) + 1)
+    xs = torch.arange(k, dtype=torch.float32) - (k - 1) / 2
+    g = torch.exp(-0.5 * (xs / sigma) ** 2)
+    g = (g / g.sum()).reshape(1, 1, 1, k)
+    m = mask.float()[None, None]
+    m = F.conv2d(F.pad(m, (k // 2, k // 2, 0, 0), mode="replicate"), g)
+    m = F.conv2d(
+        F.pad(m, (0, 0, k // 2, k // 2), mode="replicate"),
+        g.reshape(1, 1, k, 1),
+    )
+    return m[0, 0].clamp(0, 255).to(torch.uint8)
+
+
+def _preprocess_mask(gen: GenerationRequest) -> GenerationRequest:
+    if gen.mask_image is not None and gen.mask_blur > 0:
+        m = gen.mask_image
+        if m.dim() == 3:
+            m = m[0]
+        return replace_dc(gen, mask_image=_blur_mask(m, gen.mask_blur))
+    return gen
 
 
 def _crop_for_inpaint_full_res(gen: GenerationRequest):
@@ -403,6 +436,7 @@ class LocalEngine(_EngineBase):
     def generate(self, gen: GenerationRequest) -> GalleryResult:
         t0 = time.perf_counter()
         self.set_model(gen.model)
+        gen = _preprocess_mask(gen)
         gen, paste_ctx = _crop_for_inpaint_full_res(gen)
         self.world.clear_interrupt()
         jobs = self.world.make_jobs(gen.sched())
@@ -657,6 +691,7 @@ class DistributedEngine(_EngineBase):
             self.pipe = StableDiffusionPipeline(
                 want, device=self.device, dtype=self._dtype
             )
+        gen = _preprocess_mask(gen)
         gen, paste_ctx = _crop_for_inpaint_full_res(gen)
 
         mine = next((j for j in jobs if j.worker_label == self.label), None)

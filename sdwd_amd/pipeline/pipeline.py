@@ -57,6 +57,7 @@ class PipelineRequest:
     # refiner model finishes (both share the latent space / VAE)
     refiner_model: str = ""
     refiner_switch_at: float = 0.8
+    tiling: bool = False  # seamless tiles: wrap-around conv padding
     # controlnet (ref C17 executed natively)
     control_image: Optional[torch.Tensor] = None  # [B,H,W,3] uint8
     control_model: str = ""  # e.g. "controlnet-sd15"; "" = off
@@ -165,6 +166,7 @@ class StableDiffusionPipeline:
         from ..models.lora import LoraManager
 
         self.lora = LoraManager(self.model.unet)
+        self._tiling = False
         self._denoiser = GraphedDenoiser(
             lambda x, ts, ctx, y: self.model.unet(x, ts, ctx, y=y),
             self.device,
@@ -266,6 +268,17 @@ class StableDiffusionPipeline:
         if loras or self.lora.active:
             req = replace(req, prompt=prompt)
             self.lora.set_active(loras)
+        if req.tiling != self._tiling:
+            from ..models.layers import SDConv2d
+
+            for m in list(self.model.unet.modules()) + list(
+                self.model.vae.modules()
+            ):
+                if isinstance(m, SDConv2d):
+                    m.circular = req.tiling
+            # captured graphs baked the old padding path
+            self._denoiser.cache.clear()
+            self._tiling = req.tiling
         b = req.batch_size
         f = self.model.vae.cfg.downsample_factor
         lat_h, lat_w = req.height // f, req.width // f
@@ -612,6 +625,8 @@ class StableDiffusionPipeline:
                 f", Refiner: {req.refiner_model}"
                 f", Refiner switch at: {req.refiner_switch_at}"
             )
+        if req.tiling:
+            extra += ", Tiling: True"
         infotexts = [
             f"{req.prompt}\nNegative prompt: {req.negative_prompt}\n"
             f"Steps: {req.steps}, Sampler: {req.sampler_name}, "

@@ -546,6 +546,50 @@ class TestPromptEditing:
         assert torch.equal(a, c)
 
 
+class TestTiling:
+    def test_circular_conv_matches_reference(self):
+        import torch.nn as nn
+
+        from sdwd_amd.models.layers import SDConv2d
+
+        conv = SDConv2d(8, 8, 3, padding=1)
+        conv.circular = True
+        ref = nn.Conv2d(8, 8, 3, padding=1, padding_mode="circular")
+        with torch.no_grad():
+            ref.weight.copy_(conv.weight)
+            ref.bias.copy_(conv.bias)
+        x = torch.randn(2, 8, 16, 16)
+        assert torch.allclose(conv(x), ref(x), atol=1e-6)
+
+    def test_tiling_changes_output_and_reverts(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="t", steps=2, width=64, height=64, seeds=[4])
+        plain = pipe.generate(PipelineRequest(**base)).images
+        tiled = pipe.generate(PipelineRequest(**base, tiling=True)).images
+        assert not torch.equal(plain, tiled)
+        # flag reverts cleanly on the next request
+        plain2 = pipe.generate(PipelineRequest(**base)).images
+        assert torch.equal(plain, plain2)
+        info = pipe.generate(
+            PipelineRequest(**base, tiling=True)
+        ).infotexts[0]
+        assert "Tiling: True" in info
+
+    def test_mask_blur_softens_mask(self):
+        from sdwd_amd.parallel.engine import _blur_mask
+
+        m = torch.zeros(32, 32, dtype=torch.uint8)
+        m[8:24, 8:24] = 255
+        blurred = _blur_mask(m, 4)
+        assert blurred.shape == m.shape
+        # edge is now soft: intermediate values exist
+        inter = ((blurred > 10) & (blurred < 245)).sum()
+        assert inter > 0
+        # interior remains fully masked
+        assert blurred[16, 16] > 250
+
+
 class TestComposableAnd:
     def test_split_and(self):
         from sdwd_amd.pipeline.prompt_schedule import split_and
