@@ -74,6 +74,7 @@ class Img2ImgRequest(Txt2ImgRequest):
     inpaint_full_res: bool = False  # sdwui "Inpaint area: Only masked"
     inpaint_full_res_padding: int = 32
     mask_blur: int = 4
+    inpainting_fill: int = 1  # 0 fill, 1 original, 2 latent noise, 3 nothing
     # sdwui resize_mode: 0 just resize, 1 crop and resize, 2 resize and fill
     resize_mode: int = 0
 
@@ -349,6 +350,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             inpaint_full_res=req.inpaint_full_res,
             inpaint_full_res_padding=req.inpaint_full_res_padding,
             mask_blur=req.mask_blur,
+            inpainting_fill=req.inpainting_fill,
         )
         return run_generation(gen)
 

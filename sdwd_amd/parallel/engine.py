@@ -67,6 +67,9 @@ class GenerationRequest:
     inpaint_full_res_padding: int = 32
     mask_blur: int = 0  # gaussian blur radius on the mask (sdwui mask_blur)
     tiling: bool = False
+    # sdwui "Masked content": 0 fill (mean color), 1 original,
+    # 2 latent noise, 3 latent nothing
+    inpainting_fill: int = 1
 
     def sched(self) -> GenRequest:
         return GenRequest(
@@ -113,6 +116,7 @@ def _job_pipeline_request(
         init_latents=init_latents,
         denoising_strength=gen.denoising_strength,
         mask_image=gen.mask_image,
+        inpainting_fill=gen.inpainting_fill,
         enable_hr=gen.enable_hr,
         hr_scale=gen.hr_scale,
         hr_steps=gen.hr_steps,
@@ -149,11 +153,26 @@ def _blur_mask(mask: torch.Tensor, radius: int) -> torch.Tensor:
 
 
 def _preprocess_mask(gen: GenerationRequest) -> GenerationRequest:
-    if gen.mask_image is not None and gen.mask_blur > 0:
+    if gen.mask_image is None:
+        return gen
+    m = gen.mask_image
+    if m.dim() == 3:
+        m = m[0]
+    if gen.mask_blur > 0:
+        gen = replace_dc(gen, mask_image=_blur_mask(m, gen.mask_blur))
         m = gen.mask_image
-        if m.dim() == 3:
-            m = m[0]
-        return replace_dc(gen, mask_image=_blur_mask(m, gen.mask_blur))
+    if gen.inpainting_fill == 0 and gen.init_images is not None:
+        # "fill": seed the masked region with the image's unmasked mean
+        # color before encoding (approximates sdwui's blurred fill)
+        sel = (m > 127)
+        if sel.any() and (~sel).any():
+            imgs = gen.init_images.float().clone()
+            for i in range(imgs.shape[0]):
+                mean = imgs[i][~sel].reshape(-1, 3).mean(dim=0)
+                imgs[i][sel] = mean
+            gen = replace_dc(
+                gen, init_images=imgs.clamp(0, 255).to(torch.uint8)
+            )
     return gen
 
 

@@ -45,6 +45,9 @@ class PipelineRequest:
     denoising_strength: float = 0.75
     # inpainting: uint8 mask, 255 = repaint region (sdwui convention)
     mask_image: Optional[torch.Tensor] = None    # [H,W] or [B,H,W]
+    # sdwui "Masked content": 1 original (default), 2 latent noise,
+    # 3 latent nothing ("fill"=0 is approximated upstream in the engine)
+    inpainting_fill: int = 1
     # hires fix (sdwui two-pass: base gen -> latent upscale -> img2img pass)
     enable_hr: bool = False
     hr_scale: float = 2.0
@@ -250,6 +253,19 @@ class StableDiffusionPipeline:
             y = torch.cat(yrows)
         return ctx, y, ws
 
+    def _latent_mask(self, mask_image, b, lat_h, lat_w):
+        """[H,W]/[B,H,W] uint8 mask -> [B,1,lat_h,lat_w] float in 0..1."""
+        mk = mask_image
+        if mk.dim() == 2:
+            mk = mk[None]
+        lat_mask = torch.nn.functional.interpolate(
+            (mk.float() / 255.0)[:, None], size=(lat_h, lat_w),
+            mode="area",
+        ).clamp(0, 1).to(self.device)
+        if lat_mask.shape[0] == 1 and b > 1:
+            lat_mask = lat_mask.expand(b, -1, -1, -1)
+        return lat_mask
+
     # -- the denoise loop ----------------------------------------------------
     @torch.no_grad()
     def generate(
@@ -323,10 +339,27 @@ class StableDiffusionPipeline:
             )
             sampler = build_sampler(req.sampler_name, sched)
             s0 = float(sched.sigmas[0])
-            x = (
-                req.init_latents.to(self.device, self.dtype).float()
-                + noise.float() * s0
-            ).to(self.dtype)
+            init_lat0 = req.init_latents.to(self.device).float()
+            if req.mask_image is not None and req.inpainting_fill in (2, 3):
+                # masked content: replace the region's starting latents
+                lm = self._latent_mask(
+                    req.mask_image, b, lat_h, lat_w
+                )
+                if req.inpainting_fill == 2:  # latent noise
+                    content = torch.stack([
+                        torch.randn(
+                            (lat_c, lat_h, lat_w),
+                            generator=torch.Generator("cpu").manual_seed(
+                                (int(sd) ^ 0xF111) & 0xFFFFFFFF
+                            ),
+                            dtype=torch.float32,
+                        )
+                        for sd in req.seeds
+                    ]).to(self.device)
+                else:  # latent nothing
+                    content = torch.zeros_like(init_lat0)
+                init_lat0 = (1.0 - lm) * init_lat0 + lm * content
+            x = (init_lat0 + noise.float() * s0).to(self.dtype)
         else:
             x = (noise.float() * float(sig[0])).to(self.dtype)
 
@@ -511,15 +544,7 @@ class StableDiffusionPipeline:
             # latent-space inpainting: outside the mask the trajectory is
             # pinned to the init re-noised at the current sigma (sdwui
             # masked-img2img semantics; deterministic via the init noise)
-            mk = req.mask_image
-            if mk.dim() == 2:
-                mk = mk[None]
-            lat_mask = torch.nn.functional.interpolate(
-                (mk.float() / 255.0)[:, None], size=(lat_h, lat_w),
-                mode="area",
-            ).clamp(0, 1).to(self.device)
-            if lat_mask.shape[0] == 1 and b > 1:
-                lat_mask = lat_mask.expand(b, -1, -1, -1)
+            lat_mask = self._latent_mask(req.mask_image, b, lat_h, lat_w)
             init_lat = req.init_latents.to(self.device).float()
             noise_f32 = noise.float()
 

@@ -546,6 +546,58 @@ class TestPromptEditing:
         assert torch.equal(a, c)
 
 
+class TestInpaintingFill:
+    def _run(self, pipe, fill):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        init = torch.full((1, 64, 64, 3), 180, dtype=torch.uint8)
+        lat = pipe.encode_image(init, seeds=[6])
+        mask = torch.zeros(64, 64, dtype=torch.uint8)
+        mask[16:48, 16:48] = 255
+        return pipe.generate(
+            PipelineRequest(
+                prompt="fill", steps=2, width=64, height=64, seeds=[6],
+                init_latents=lat, mask_image=mask, denoising_strength=0.5,
+                inpainting_fill=fill,
+            )
+        ).images
+
+    def test_fill_modes_differ_inside_mask(self, pipe):
+        orig = self._run(pipe, 1)
+        noise = self._run(pipe, 2)
+        nothing = self._run(pipe, 3)
+        # inside the mask the starting content differs per mode
+        assert not torch.equal(orig[:, 20:44, 20:44], noise[:, 20:44, 20:44])
+        assert not torch.equal(orig[:, 20:44, 20:44],
+                               nothing[:, 20:44, 20:44])
+        for out in (orig, noise, nothing):
+            assert torch.isfinite(out.float()).all()
+
+    def test_latent_modes_deterministic(self, pipe):
+        a = self._run(pipe, 2)
+        b = self._run(pipe, 2)
+        assert torch.equal(a, b)
+
+    def test_pixel_fill_mode_in_engine(self):
+        from sdwd_amd.parallel import GenerationRequest
+        from sdwd_amd.parallel.engine import _preprocess_mask
+
+        init = torch.zeros(1, 32, 32, 3, dtype=torch.uint8)
+        init[:, :, :16] = 60
+        init[:, :, 16:] = 200  # mean of unmasked region will mix both
+        mask = torch.zeros(32, 32, dtype=torch.uint8)
+        mask[12:20, 12:20] = 255
+        gen = GenerationRequest(
+            prompt="f", init_images=init, mask_image=mask,
+            inpainting_fill=0, width=32, height=32,
+        )
+        out = _preprocess_mask(gen)
+        filled = out.init_images[0, 14:18, 14:18].float()
+        # region replaced by one flat color (the unmasked mean)
+        assert filled.std() < 1.0
+        assert 60 < filled.mean() < 200
+
+
 class TestTiling:
     def test_circular_conv_matches_reference(self):
         import torch.nn as nn
