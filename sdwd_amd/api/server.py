@@ -49,6 +49,11 @@ class Txt2ImgRequest(BaseModel):
     styles: List[str] = Field(default_factory=list)
     tiling: bool = False
     clip_skip: int = 1
+    s_churn: float = 0.0
+    s_tmin: float = 0.0
+    s_tmax: float = 0.0
+    s_noise: float = 1.0
+    s_min_uncond: float = 0.0
     # hires fix (sdwui fields)
     enable_hr: bool = False
     hr_scale: float = 2.0
@@ -287,6 +292,11 @@ def create_app(engine: Optional[LocalEngine] = None,
             subseed=req.subseed,
             subseed_strength=req.subseed_strength,
             tiling=req.tiling,
+            s_churn=req.s_churn,
+            s_tmin=req.s_tmin,
+            s_tmax=req.s_tmax,
+            s_noise=req.s_noise,
+            s_min_uncond=req.s_min_uncond,
             enable_hr=req.enable_hr,
             hr_scale=req.hr_scale,
             hr_steps=req.hr_second_pass_steps,
@@ -342,6 +352,11 @@ def create_app(engine: Optional[LocalEngine] = None,
             subseed=req.subseed,
             subseed_strength=req.subseed_strength,
             tiling=req.tiling,
+            s_churn=req.s_churn,
+            s_tmin=req.s_tmin,
+            s_tmax=req.s_tmax,
+            s_noise=req.s_noise,
+            s_min_uncond=req.s_min_uncond,
             init_images=inits,
             denoising_strength=req.denoising_strength,
             mask_image=mask_image,

@@ -70,6 +70,12 @@ class GenerationRequest:
     # sdwui "Masked content": 0 fill (mean color), 1 original,
     # 2 latent noise, 3 latent nothing
     inpainting_fill: int = 1
+    # sampler stochasticity + uncond-skip perf knob (sdwui Sampler params)
+    s_churn: float = 0.0
+    s_tmin: float = 0.0
+    s_tmax: float = 0.0
+    s_noise: float = 1.0
+    s_min_uncond: float = 0.0
 
     def sched(self) -> GenRequest:
         return GenRequest(
@@ -128,6 +134,11 @@ def _job_pipeline_request(
         refiner_model=gen.refiner_model,
         refiner_switch_at=gen.refiner_switch_at,
         tiling=gen.tiling,
+        s_churn=gen.s_churn,
+        s_tmin=gen.s_tmin,
+        s_tmax=gen.s_tmax,
+        s_noise=gen.s_noise,
+        s_min_uncond=gen.s_min_uncond,
     )
 
 

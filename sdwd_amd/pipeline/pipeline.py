@@ -61,6 +61,13 @@ class PipelineRequest:
     refiner_model: str = ""
     refiner_switch_at: float = 0.8
     tiling: bool = False  # seamless tiles: wrap-around conv padding
+    # k-diffusion stochasticity (sdwui Sampler parameters section)
+    s_churn: float = 0.0
+    s_tmin: float = 0.0
+    s_tmax: float = 0.0   # 0 = unlimited (sdwui convention)
+    s_noise: float = 1.0
+    # skip the uncond eval below this sigma (sdwui s_min_uncond perf knob)
+    s_min_uncond: float = 0.0
     # controlnet (ref C17 executed natively)
     control_image: Optional[torch.Tensor] = None  # [B,H,W,3] uint8
     control_model: str = ""  # e.g. "controlnet-sd15"; "" = off
@@ -119,6 +126,13 @@ def _upscale_latent(x: torch.Tensor, scale: float, upscaler: str) -> torch.Tenso
     return torch.nn.functional.interpolate(
         x, scale_factor=scale, mode=mode, **kwargs
     )
+
+
+def _apply_sampler_params(sampler, req: "PipelineRequest") -> None:
+    sampler.s_churn = float(req.s_churn)
+    sampler.s_tmin = float(req.s_tmin)
+    sampler.s_tmax = float(req.s_tmax) if req.s_tmax > 0 else float("inf")
+    sampler.s_noise = float(req.s_noise)
 
 
 def _image_noise(
@@ -316,6 +330,7 @@ class StableDiffusionPipeline:
 
         sched = schedule_for(req.sampler_name, req.steps, req.scheduler)
         sampler = build_sampler(req.sampler_name, sched)
+        _apply_sampler_params(sampler, req)
 
         noise = torch.stack(
             [
@@ -338,6 +353,7 @@ class StableDiffusionPipeline:
                 sigmas=sched.sigmas[start:], timesteps=sched.timesteps[start:]
             )
             sampler = build_sampler(req.sampler_name, sched)
+            _apply_sampler_params(sampler, req)
             s0 = float(sched.sigmas[0])
             init_lat0 = req.init_latents.to(self.device).float()
             if req.mask_image is not None and req.inpainting_fill in (2, 3):
@@ -444,10 +460,26 @@ class StableDiffusionPipeline:
 
         pred_type = self.model.prediction_type
 
+        s_min_uncond = float(req.s_min_uncond)
+
         def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:
             c_ctx, c_y, ws = _ctx_y_for(t)
-            k1 = len(ws) + 1  # k AND-conds + 1 uncond
             nb = x_in.shape[0]
+            if (
+                s_min_uncond > 0
+                and len(ws) == 1
+                and sigma_for_t(t) < s_min_uncond
+            ):
+                # sdwui s_min_uncond: at low noise the uncond eval barely
+                # changes the output — skip it (halves the model cost)
+                ts1 = torch.full(
+                    (nb,), t, device=self.device, dtype=torch.float32,
+                )
+                yc = c_y[:nb] if c_y is not None else None
+                return _to_eps(
+                    denoiser(x_in, ts1, c_ctx[:nb], yc), x_in, t, pred_type
+                )
+            k1 = len(ws) + 1  # k AND-conds + 1 uncond
             ts = torch.full(
                 (nb * k1,), t, device=self.device, dtype=torch.float32,
             )
@@ -575,6 +607,7 @@ class StableDiffusionPipeline:
                 sigmas=hsched.sigmas[start:], timesteps=hsched.timesteps[start:]
             )
             hsampler = build_sampler(req.sampler_name, hsched)
+            _apply_sampler_params(hsampler, req)
             hh, hw = x.shape[2], x.shape[3]
             hr_noise = torch.stack(
                 [

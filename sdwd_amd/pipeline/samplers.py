@@ -56,6 +56,26 @@ class Sampler:
 
     def __init__(self, schedule: Schedule):
         self.schedule = schedule
+        # k-diffusion stochasticity knobs (sdwui s_churn/s_tmin/s_tmax/
+        # s_noise); honored by the Euler/Heun family via _churned
+        self.s_churn = 0.0
+        self.s_tmin = 0.0
+        self.s_tmax = float("inf")
+        self.s_noise = 1.0
+
+    def _churned(self, x, sigma, noise_fn):
+        """Temporarily raise sigma by gamma (k-diffusion churn); returns
+        (x_hat, sigma_hat)."""
+        if self.s_churn <= 0 or not (self.s_tmin <= sigma <= self.s_tmax):
+            return x, sigma
+        gamma = min(
+            self.s_churn / max(1, self.schedule.steps), 2 ** 0.5 - 1
+        )
+        sigma_hat = sigma * (1 + gamma)
+        if noise_fn is not None and sigma_hat > sigma:
+            extra = (sigma_hat ** 2 - sigma ** 2) ** 0.5 * self.s_noise
+            x = ops.add_noise(x, noise_fn(), 1.0, extra)
+        return x, sigma_hat
 
     def sample(
         self,
@@ -106,6 +126,7 @@ class Sampler:
 class Euler(Sampler):
     def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
              t_next=None):
+        x, sigma = self._churned(x, sigma, noise_fn)
         denoised = _eval(model_fn, x, sigma, t)
         return ops.euler_step(x, denoised, sigma, sigma_next)
 
@@ -126,6 +147,7 @@ class Heun(Sampler):
 
     def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
              t_next=None):
+        x, sigma = self._churned(x, sigma, noise_fn)
         denoised = _eval(model_fn, x, sigma, t)
         if sigma_next <= 0:
             return ops.euler_step(x, denoised, sigma, sigma_next)

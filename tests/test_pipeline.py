@@ -546,6 +546,43 @@ class TestPromptEditing:
         assert torch.equal(a, c)
 
 
+class TestSamplerParams:
+    def test_churn_changes_euler_output(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="c", steps=4, width=64, height=64, seeds=[9],
+                    sampler_name="Euler")
+        a = pipe.generate(PipelineRequest(**base)).images
+        b = pipe.generate(PipelineRequest(**base, s_churn=1.0)).images
+        assert not torch.equal(a, b)
+        # deterministic (churn noise comes from the seeded noise_fn)
+        b2 = pipe.generate(PipelineRequest(**base, s_churn=1.0)).images
+        assert torch.equal(b, b2)
+
+    def test_s_min_uncond_skips_low_sigma_uncond(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="u", steps=4, width=64, height=64, seeds=[9])
+        a = pipe.generate(PipelineRequest(**base)).images
+        # huge threshold -> EVERY step skips uncond -> equivalent to cfg=1
+        b = pipe.generate(
+            PipelineRequest(**base, s_min_uncond=1e9)
+        ).images
+        c = pipe.generate(PipelineRequest(**base, cfg_scale=1.0)).images
+        assert not torch.equal(a, b)
+        diff = (b.float() - c.float()).abs()
+        assert diff.max() <= 1.0  # same compute, batch-blocking LSB only
+
+    def test_churn_zero_is_default(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="c", steps=3, width=64, height=64, seeds=[9],
+                    sampler_name="Heun")
+        a = pipe.generate(PipelineRequest(**base)).images
+        b = pipe.generate(PipelineRequest(**base, s_churn=0.0)).images
+        assert torch.equal(a, b)
+
+
 class TestInpaintingFill:
     def _run(self, pipe, fill):
         from sdwd_amd.pipeline import PipelineRequest
