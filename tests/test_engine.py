@@ -115,8 +115,9 @@ def _dist_worker(rank, world_size, port, tmpdir, fail_rank):
     from sdwd_amd.parallel import DistributedEngine, GenerationRequest
 
     eng = DistributedEngine(model="tiny", backend="gloo")
-    for w in eng.world.workers:
-        w.eta.avg_ipm = 60.0
+    # uneven speeds: the plan (made on rank 0) must shard 1:3
+    eng.world.get_worker("gpu0").eta.avg_ipm = 30.0
+    eng.world.get_worker("gpu1").eta.avg_ipm = 90.0
     if fail_rank == rank:
         # simulate a failing shard on this rank
         def boom(*a, **k):
@@ -134,6 +135,21 @@ def _dist_worker(rank, world_size, port, tmpdir, fail_rank):
         assert res.seeds == [900, 901, 902, 903]
         for i in range(4):
             assert res.images[i].float().std() > 0, f"image {i} empty"
+        if fail_rank == -1:
+            # benchmark-weighted split honored across processes: the 3x
+            # faster rank got the bigger (non-complementary) shard
+            import re as _re
+
+            sizes = {}
+            for line in res.job_summary:
+                if "(complementary)" in line:
+                    continue
+                m = _re.match(r"(gpu\d): (\d+) image", line)
+                if m:
+                    sizes[m.group(1)] = sizes.get(m.group(1), 0) + int(
+                        m.group(2)
+                    )
+            assert sizes.get("gpu1", 0) > sizes.get("gpu0", 0), res.job_summary
         torch.save(res.images, os.path.join(tmpdir, "gallery.pt"))
     if fail_rank == -1:
         # model hot-swap rides the plan broadcast: every rank rebuilds
