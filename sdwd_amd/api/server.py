@@ -80,6 +80,7 @@ class Img2ImgRequest(Txt2ImgRequest):
     inpaint_full_res_padding: int = 32
     mask_blur: int = 4
     inpainting_fill: int = 1  # 0 fill, 1 original, 2 latent noise, 3 nothing
+    color_correction: bool = False
     # sdwui resize_mode: 0 just resize, 1 crop and resize, 2 resize and fill
     resize_mode: int = 0
 
@@ -366,6 +367,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             inpaint_full_res_padding=req.inpaint_full_res_padding,
             mask_blur=req.mask_blur,
             inpainting_fill=req.inpainting_fill,
+            color_correction=req.color_correction,
         )
         return run_generation(gen)
 
@@ -502,7 +504,10 @@ def create_app(engine: Optional[LocalEngine] = None,
             info = png_parameters(raw)
         except Exception as exc:
             raise HTTPException(422, f"bad png: {exc}")
-        return {"info": info or "", "items": {}}
+        from ..utils.images import parse_infotext
+
+        parsed = parse_infotext(info) if info else {}
+        return {"info": info or "", "items": {}, "parameters": parsed}
 
     @app.get("/sdapi/v1/progress")
     def progress():

@@ -76,6 +76,8 @@ class GenerationRequest:
     s_tmax: float = 0.0
     s_noise: float = 1.0
     s_min_uncond: float = 0.0
+    # sdwui img2img color correction: match output statistics to the init
+    color_correction: bool = False
 
     def sched(self) -> GenRequest:
         return GenRequest(
@@ -247,6 +249,23 @@ def _crop_for_inpaint_full_res(gen: GenerationRequest):
         "mask": (mask[y0:y1, x0:x1].float() / 255.0),
     }
     return new_gen, paste_ctx
+
+
+def _apply_color_correction(result: "GalleryResult",
+                            gen: GenerationRequest) -> None:
+    """sdwui img2img color correction: match each output's channel
+    statistics to its init image (in place, after assembly/paste)."""
+    from ..utils.images import color_correct
+
+    if not (gen.color_correction and gen.init_images is not None):
+        return
+    n_init = gen.init_images.shape[0]
+    imgs = []
+    for i in range(result.images.shape[0]):
+        ref = gen.init_images[i % n_init]
+        imgs.append(color_correct(result.images[i], ref))
+    result.images = torch.stack(imgs)
+    result.grid = make_grid(result.images) if len(imgs) > 1 else None
 
 
 def _paste_inpaint_full_res(result: "GalleryResult", paste_ctx) -> None:
@@ -514,6 +533,7 @@ class LocalEngine(_EngineBase):
             gen, jobs, shards, infos, time.perf_counter() - t0, interrupted
         )
         _paste_inpaint_full_res(result, paste_ctx)
+        _apply_color_correction(result, gen)
         self._autosave()
         return result
 
@@ -825,4 +845,5 @@ class DistributedEngine(_EngineBase):
             gen, jobs, shards, infos, time.perf_counter() - t0, interrupted
         )
         _paste_inpaint_full_res(result, paste_ctx)
+        _apply_color_correction(result, gen)
         return result

@@ -80,6 +80,59 @@ def png_parameters(data: bytes) -> Optional[str]:
     return None
 
 
+def color_correct(out: torch.Tensor, ref: torch.Tensor) -> torch.Tensor:
+    """Match per-channel mean/std of ``out`` to ``ref`` ([H,W,3] uint8 each;
+    sdwui's img2img color correction, approximated in RGB)."""
+    o = out.float()
+    r = ref.float()
+    om = o.mean(dim=(0, 1))
+    osd = o.std(dim=(0, 1)).clamp_min(1e-5)
+    rm = r.mean(dim=(0, 1))
+    rsd = r.std(dim=(0, 1))
+    corrected = (o - om) / osd * rsd + rm
+    return corrected.clamp(0, 255).to(torch.uint8)
+
+
+def parse_infotext(text: str) -> dict:
+    """Parse an sdwui 'parameters' infotext back into request fields
+    (inverse of the generator in pipeline.py; sdwui
+    parse_generation_parameters shape)."""
+    import re
+
+    out: dict = {}
+    lines = text.split("\n")
+    kv_line = ""
+    prompt_lines = []
+    for ln in lines:
+        if re.match(r"^Steps: \d+", ln.strip()):
+            kv_line = ln
+            break
+        prompt_lines.append(ln)
+    neg = ""
+    prompts = []
+    for ln in prompt_lines:
+        if ln.startswith("Negative prompt: "):
+            neg = ln[len("Negative prompt: "):]
+        else:
+            prompts.append(ln)
+    out["prompt"] = "\n".join(prompts).strip()
+    out["negative_prompt"] = neg
+    for m in re.finditer(r"([A-Za-z ]+): ([^,]+)(?:, |$)", kv_line):
+        key = m.group(1).strip().lower().replace(" ", "_")
+        val = m.group(2).strip()
+        out[key] = val
+    if "size" in out and "x" in out["size"]:
+        w, h = out.pop("size").split("x")
+        out["width"], out["height"] = int(w), int(h)
+    for k in ("steps", "seed", "clip_skip"):
+        if k in out:
+            out[k] = int(out[k])
+    for k in ("cfg_scale", "denoising_strength"):
+        if k in out:
+            out[k] = float(out[k])
+    return out
+
+
 def save_png(
     image: torch.Tensor, path: str, parameters: Optional[str] = None
 ) -> str:

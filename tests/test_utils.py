@@ -153,6 +153,47 @@ class TestEtaDefaults:
         assert sampler_cost("Heun") == 2.0
 
 
+class TestColorCorrect:
+    def test_matches_reference_stats(self):
+        from sdwd_amd.utils.images import color_correct
+
+        g = torch.Generator().manual_seed(1)
+        out = (torch.randn(32, 32, 3, generator=g) * 20 + 60).clamp(
+            0, 255
+        ).to(torch.uint8)
+        ref = (torch.randn(32, 32, 3, generator=g) * 40 + 180).clamp(
+            0, 255
+        ).to(torch.uint8)
+        corrected = color_correct(out, ref)
+        cm = corrected.float().mean(dim=(0, 1))
+        rm = ref.float().mean(dim=(0, 1))
+        assert (cm - rm).abs().max() < 8.0  # uint8 rounding/clamping slack
+
+
+class TestParseInfotext:
+    def test_round_trip_fields(self):
+        from sdwd_amd.utils.images import parse_infotext
+
+        text = ("a cow, detailed\nNegative prompt: blurry, bad\n"
+                "Steps: 20, Sampler: Euler a, CFG scale: 7.0, Seed: 42, "
+                "Size: 512x768, Model: sd15, Clip skip: 2")
+        p = parse_infotext(text)
+        assert p["prompt"] == "a cow, detailed"
+        assert p["negative_prompt"] == "blurry, bad"
+        assert p["steps"] == 20 and p["seed"] == 42
+        assert p["cfg_scale"] == 7.0
+        assert p["width"] == 512 and p["height"] == 768
+        assert p["sampler"] == "Euler a"
+        assert p["clip_skip"] == 2
+
+    def test_no_negative(self):
+        from sdwd_amd.utils.images import parse_infotext
+
+        p = parse_infotext("just a prompt\nSteps: 4, Seed: 1")
+        assert p["prompt"] == "just a prompt"
+        assert p["negative_prompt"] == ""
+
+
 class TestPngMetadata:
     def test_parameters_round_trip(self):
         from sdwd_amd.utils.images import encode_png, png_parameters
