@@ -427,8 +427,17 @@ def sampler_names() -> List[str]:
     return sorted(SAMPLERS.keys())
 
 
-def build_sampler(name: str, schedule: Schedule) -> Sampler:
+def build_sampler(name: str, schedule: Schedule, strict: bool = False) -> Sampler:
+    """``strict=False`` falls back to Euler a on an unknown name, matching
+    the reference's sampler-not-found retry (ref worker.py:456-467)."""
     cls = SAMPLERS.get(name)
     if cls is None:
-        raise KeyError(f"unknown sampler '{name}'")
+        if strict:
+            raise KeyError(f"unknown sampler '{name}'")
+        from ..utils import get_logger
+
+        get_logger("samplers").warning(
+            "unknown sampler '%s'; falling back to Euler a", name
+        )
+        cls = EulerAncestral
     return cls(schedule)

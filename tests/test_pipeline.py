@@ -751,6 +751,24 @@ class TestSchedulers:
         with pytest.raises(KeyError):
             schedule_for("Euler", 10, "Quantum")
 
+    def test_unknown_sampler_falls_back(self, pipe):
+        """ref worker.py:456-467: sampler-not-found retries as Euler a."""
+        from sdwd_amd.pipeline import PipelineRequest
+        from sdwd_amd.pipeline.samplers import build_sampler
+        from sdwd_amd.pipeline.schedule import discrete_schedule
+
+        with pytest.raises(KeyError):
+            build_sampler("Future Sampler 9000", discrete_schedule(4),
+                          strict=True)
+        base = dict(prompt="f", steps=2, width=64, height=64, seeds=[3])
+        a = pipe.generate(
+            PipelineRequest(**base, sampler_name="Future Sampler 9000")
+        ).images
+        b = pipe.generate(
+            PipelineRequest(**base, sampler_name="Euler a")
+        ).images
+        assert torch.equal(a, b)
+
     def test_scheduler_changes_output(self, pipe):
         from sdwd_amd.pipeline import PipelineRequest
 
