@@ -162,6 +162,27 @@ class TestControl:
         )
         assert r.status_code == 404
 
+    def test_all_knobs_accepted(self, client):
+        """Every sdwui-compat request field plumbs through end to end."""
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={
+                "prompt": "a [cat:dog:0.5] AND a cow:0.7 <lora:style-z:0.5>",
+                "negative_prompt": "blurry",
+                "steps": 2, "width": 64, "height": 64, "seed": 11,
+                "sampler_name": "Heun", "scheduler": "Exponential",
+                "cfg_scale": 4.0, "clip_skip": 2, "tiling": True,
+                "s_churn": 0.5, "s_noise": 1.1, "s_min_uncond": 0.1,
+                "enable_hr": True, "hr_scale": 2.0,
+                "hr_upscaler": "Latent (bilinear)",
+                "hr_second_pass_steps": 1,
+                "refiner_checkpoint": "tiny-xl", "refiner_switch_at": 0.5,
+            },
+        )
+        assert r.status_code == 200
+        body = r.json()
+        assert len(body["images"]) >= 1
+
     def test_basic_auth(self, monkeypatch):
         import base64 as b64
 
