@@ -321,11 +321,23 @@ class StableDiffusionPipeline:
 
         p_segs = prompt_schedule(req.prompt, req.steps)
         n_segs = prompt_schedule(req.negative_prompt, req.steps)
+        # img2img runs only the schedule tail; editing thresholds are
+        # fractions of req.steps, so segments before the tail's first step
+        # collapse into the base conditioning
+        start_off = 0
+        if req.init_latents is not None:
+            start_off = max(
+                0, req.steps - max(1, int(req.steps * req.denoising_strength))
+            )
+
+        def _seg_text_at(segs, step):
+            return [txt for st, txt in segs if st <= step][-1]
 
         # one conditioning set per request (encode once, broadcast across
         # the batch); AND composition yields k cond rows + 1 uncond row
         ctx, y, and_ws = self._build_ctx(
-            p_segs[0][1], n_segs[0][1], req, b
+            _seg_text_at(p_segs, start_off), _seg_text_at(n_segs, start_off),
+            req, b,
         )
 
         sched = schedule_for(req.sampler_name, req.steps, req.scheduler)
@@ -346,9 +358,7 @@ class StableDiffusionPipeline:
         if req.init_latents is not None:
             # img2img: noise the init latents to the strength point, run the
             # tail of the schedule from there.
-            start = max(
-                0, req.steps - max(1, int(req.steps * req.denoising_strength))
-            )
+            start = start_off
             sched = type(sched)(
                 sigmas=sched.sigmas[start:], timesteps=sched.timesteps[start:]
             )
@@ -410,22 +420,23 @@ class StableDiffusionPipeline:
             )
             enc_cache: dict = {}
             for s in boundaries:
-                if s >= len(ts_all):
-                    continue
+                eff = s - start_off
+                if eff <= 0 or eff >= len(ts_all):
+                    continue  # the base conditioning already covers it
                 p = [txt for st, txt in p_segs if st <= s][-1]
                 n = [txt for st, txt in n_segs if st <= s][-1]
                 if (p, n) not in enc_cache:
                     enc_cache[(p, n)] = self._build_ctx(p, n, req, b)
                 ctx_b, y_b, ws_b = enc_cache[(p, n)]
-                seg_tensors.append((ts_all[s], ctx_b, y_b, ws_b))
+                seg_tensors.append((ts_all[eff], ctx_b, y_b, ws_b))
 
         def _ctx_y_for(t: float):
-            if not seg_tensors:
-                return ctx, y, and_ws
-            sel = seg_tensors[0]
+            sel = None
             for s in seg_tensors:
                 if t <= s[0] + 1e-6:
                     sel = s
+            if sel is None:  # before the first switch: base conditioning
+                return ctx, y, and_ws
             return sel[1], sel[2], sel[3]
 
         denoiser = self._denoiser

@@ -535,6 +535,25 @@ class TestPromptEditing:
         ).images
         assert torch.equal(edited, again)
 
+    def test_editing_in_img2img_tail(self, pipe):
+        """img2img runs only the schedule tail; a switch scheduled BEFORE
+        the tail collapses into the base conditioning (the 'to' text)."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        img = torch.full((1, 64, 64, 3), 140, dtype=torch.uint8)
+        lat = pipe.encode_image(img, seeds=[4])
+        base = dict(steps=10, width=64, height=64, seeds=[4],
+                    init_latents=lat, denoising_strength=0.3)
+        # switch at 10% of 10 steps = step 1; the tail starts at step 7,
+        # so the whole run must use "dog"
+        edited = pipe.generate(
+            PipelineRequest(prompt="a [cat:dog:0.1] x", **base)
+        ).images
+        all_dog = pipe.generate(
+            PipelineRequest(prompt="a dog x", **base)
+        ).images
+        assert torch.equal(edited, all_dog)
+
     def test_editing_at_one_is_first_prompt(self, pipe):
         from sdwd_amd.pipeline import PipelineRequest
 
