@@ -41,7 +41,9 @@ class ModelBundle:
 
     @property
     def latent_channels(self) -> int:
-        return self.unet.cfg.in_channels
+        # from the VAE, not unet.in_channels: inpainting UNets take
+        # 2*latent+1 input channels but the latent state is still 4-wide
+        return self.vae.cfg.latent_channels
 
     def to(self, device, dtype=None) -> "ModelBundle":
         for m in (self.text_encoder, self.text_encoder_2, self.unet, self.vae):
@@ -100,6 +102,28 @@ def _build_sd21(name: str) -> ModelBundle:
     for seed_off, m in enumerate((te, unet, vae)):
         _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
     return ModelBundle(name, te, None, unet, vae, context_dim=1024)
+
+
+def _build_sd15_inpaint(name: str) -> ModelBundle:
+    """sd15-inpainting lineage: 9-channel UNet input (4 latent + 1 mask +
+    4 masked-image latent, the runwayml inpainting conditioning)."""
+    te = CLIPTextEncoder()
+    unet = UNetModel(UNetConfig(in_channels=9))
+    vae = AutoencoderKL(VAEConfig.sd())
+    for seed_off, m in enumerate((te, unet, vae)):
+        _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
+    return ModelBundle(name, te, None, unet, vae, context_dim=768)
+
+
+def _build_tiny_inpaint(name: str) -> ModelBundle:
+    te = CLIPTextEncoder(d_model=64, layers=2, heads=2, max_len=77)
+    cfg = UNetConfig.tiny()
+    cfg.in_channels = 9
+    unet = UNetModel(cfg)
+    vae = AutoencoderKL(VAEConfig.tiny())
+    for seed_off, m in enumerate((te, unet, vae)):
+        _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
+    return ModelBundle(name, te, None, unet, vae, context_dim=64)
 
 
 def _build_sd21v(name: str) -> ModelBundle:
@@ -166,8 +190,10 @@ _BUILDERS: Dict[str, Callable[[str], ModelBundle]] = {
     "sd15": _build_sd15,
     "sd21": _build_sd21,
     "sd21v": _build_sd21v,
+    "sd15-inpaint": _build_sd15_inpaint,
     "sdxl": _build_sdxl,
     "tiny-v": _build_tiny_v,
+    "tiny-inpaint": _build_tiny_inpaint,
     "tiny": _build_tiny,
     "tiny-xl": _build_tiny_xl,
 }

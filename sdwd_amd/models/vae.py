@@ -86,20 +86,23 @@ class VAEAttention(nn.Module):
 
 
 class VAEDownsample(nn.Module):
-    """ldm VAE downsample: pad (0,1,0,1) then a stride-2 conv with no
-    padding. The asymmetric pad shifts the sampling grid half a latent pixel
-    relative to a symmetric pad-1 conv, so real checkpoint weights only
-    reproduce the reference's latents with this exact form."""
+    """ldm VAE downsample: zero pad (0,1,0,1) then a stride-2 valid conv.
+    The asymmetric pad shifts the sampling grid half a latent pixel vs a
+    symmetric pad-1 conv, so real checkpoint weights only reproduce the
+    reference's latents with this exact arithmetic.
+
+    Implemented as a stride-1 pad-1 conv subsampled at odd indices —
+    output i then reads rows 2i..2i+2 with a zero row/col past the edge,
+    which is the identical computation, and it keeps the conv on the
+    pad-1 paths (the HIP kernel is pad-1-only, and MIOpen's pad-0
+    stride-2 NHWC bf16 fallback raises on small channel counts)."""
 
     def __init__(self, ch: int):
         super().__init__()
-        self.conv = SDConv2d(ch, ch, 3, stride=2, padding=0)
+        self.conv = SDConv2d(ch, ch, 3, stride=1, padding=1)
 
     def forward(self, x):
-        mode = (
-            "circular" if getattr(self.conv, "circular", False) else "constant"
-        )
-        return self.conv(torch.nn.functional.pad(x, (0, 1, 0, 1), mode=mode))
+        return self.conv(x)[:, :, 1::2, 1::2]
 
 
 class VAEEncoder(nn.Module):

@@ -473,6 +473,38 @@ class StableDiffusionPipeline:
 
         s_min_uncond = float(req.s_min_uncond)
 
+        # 9-channel inpainting models (sd15-inpaint): the UNet input is
+        # [z_t, mask(1ch), masked-init latents]; with no mask the
+        # convention is mask=1 everywhere + zero masked-image (runwayml)
+        is_inpaint_model = unet.cfg.in_channels == 2 * lat_c + 1
+        _inpaint_cache: dict = {}
+
+        def _inpaint_extra(h: int, w: int) -> torch.Tensor:
+            key = (h, w)
+            if key not in _inpaint_cache:
+                if (req.init_latents is not None
+                        and req.mask_image is not None):
+                    lm = self._latent_mask(req.mask_image, b, lat_h, lat_w)
+                    init_l = req.init_latents.to(self.device).float()
+                    masked_z = init_l * (1.0 - (lm > 0.5).float())
+                    ex = torch.cat([lm, masked_z], dim=1)
+                    if (h, w) != (lat_h, lat_w):
+                        ex = torch.nn.functional.interpolate(
+                            ex, size=(h, w), mode="nearest"
+                        )
+                else:
+                    ex = torch.cat(
+                        [
+                            torch.ones(b, 1, h, w, device=self.device),
+                            torch.zeros(
+                                b, lat_c, h, w, device=self.device
+                            ),
+                        ],
+                        dim=1,
+                    )
+                _inpaint_cache[key] = ex.to(self.dtype).contiguous()
+            return _inpaint_cache[key]
+
         def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:
             c_ctx, c_y, ws = _ctx_y_for(t)
             nb = x_in.shape[0]
@@ -487,14 +519,23 @@ class StableDiffusionPipeline:
                     (nb,), t, device=self.device, dtype=torch.float32,
                 )
                 yc = c_y[:nb] if c_y is not None else None
+                x1 = x_in
+                if is_inpaint_model:
+                    x1 = torch.cat(
+                        [x_in, _inpaint_extra(x_in.shape[2], x_in.shape[3])],
+                        dim=1,
+                    )
                 return _to_eps(
-                    denoiser(x_in, ts1, c_ctx[:nb], yc), x_in, t, pred_type
+                    denoiser(x1, ts1, c_ctx[:nb], yc), x_in, t, pred_type
                 )
             k1 = len(ws) + 1  # k AND-conds + 1 uncond
             ts = torch.full(
                 (nb * k1,), t, device=self.device, dtype=torch.float32,
             )
             xk = torch.cat([x_in] * k1, dim=0)
+            if is_inpaint_model:
+                ex = _inpaint_extra(x_in.shape[2], x_in.shape[3])
+                xk = torch.cat([xk, torch.cat([ex] * k1, dim=0)], dim=1)
             if controlnet is not None:
                 hk = torch.cat([hint] * k1, dim=0)
                 ctrl = controlnet(xk, hk, ts, c_ctx, req.control_scale)
@@ -523,8 +564,14 @@ class StableDiffusionPipeline:
                     (nb,), t, device=self.device, dtype=torch.float32,
                 )
                 yc = y[:nb] if y is not None else None
+                x1 = x_in
+                if is_inpaint_model:
+                    x1 = torch.cat(
+                        [x_in, _inpaint_extra(x_in.shape[2], x_in.shape[3])],
+                        dim=1,
+                    )
                 return _to_eps(
-                    denoiser(x_in, ts, ctx[:nb], yc), x_in, t, pred_type
+                    denoiser(x1, ts, ctx[:nb], yc), x_in, t, pred_type
                 )
 
         if req.refiner_model:

@@ -602,6 +602,101 @@ class TestSamplerParams:
         assert torch.equal(a, b)
 
 
+class TestInpaintModel9ch:
+    def test_txt2img_with_inpaint_model(self):
+        """No mask: the runwayml convention feeds mask=1 + zero latents."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        pipe = StableDiffusionPipeline("tiny-inpaint", device="cpu")
+        assert pipe.model.unet.cfg.in_channels == 9
+        assert pipe.model.latent_channels == 4
+        res = pipe.generate(
+            PipelineRequest(prompt="p", steps=2, width=64, height=64,
+                            seeds=[1])
+        )
+        assert res.images.shape == (1, 64, 64, 3)
+        assert torch.isfinite(res.images.float()).all()
+
+    def test_masked_inpaint_model(self):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        pipe = StableDiffusionPipeline("tiny-inpaint", device="cpu")
+        init = torch.full((1, 64, 64, 3), 120, dtype=torch.uint8)
+        lat = pipe.encode_image(init, seeds=[2])
+        mask = torch.zeros(64, 64, dtype=torch.uint8)
+        mask[20:40, 20:40] = 255
+        req = PipelineRequest(
+            prompt="m", steps=3, width=64, height=64, seeds=[2],
+            init_latents=lat, mask_image=mask, denoising_strength=0.8,
+        )
+        a = pipe.generate(req).images
+        b = pipe.generate(req).images
+        assert torch.equal(a, b)
+        assert torch.isfinite(a.float()).all()
+
+    def test_hires_with_inpaint_model(self):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        pipe = StableDiffusionPipeline("tiny-inpaint", device="cpu")
+        res = pipe.generate(
+            PipelineRequest(prompt="h", steps=2, width=64, height=64,
+                            seeds=[3], enable_hr=True, hr_scale=2.0,
+                            hr_steps=1, denoising_strength=0.5)
+        )
+        assert res.images.shape == (1, 128, 128, 3)
+
+    def test_ldm_arch_inference_9ch(self, tmp_path):
+        """A 9-channel ldm checkpoint round-trips through the converter."""
+        from sdwd_amd.models.convert import (
+            load_ldm_state_dict, to_ldm_state_dict,
+        )
+
+        a = load_model("tiny-inpaint", device="cpu", cache=False)
+        exported = to_ldm_state_dict(a)
+        assert exported[
+            "model.diffusion_model.input_blocks.0.0.weight"
+        ].shape[1] == 9
+        b = load_model("tiny-inpaint", device="cpu", cache=False)
+        _perturb_bundle(b)
+        report = load_ldm_state_dict(b, exported)
+        assert not report["missing"]
+        sa = a.unet.state_dict()
+        sb = b.unet.state_dict()
+        for k in sa:
+            assert torch.equal(sa[k], sb[k]), k
+
+
+def _perturb_bundle(bundle):
+    with torch.no_grad():
+        for p in bundle.unet.parameters():
+            p.add_(torch.randn_like(p) * 0.1)
+
+
+class TestVaeDownsample:
+    def test_subsample_equals_asym_pad_stride2(self):
+        """The stride-1 + odd-subsample form must equal the canonical ldm
+        pad-(0,1,0,1) stride-2 valid conv exactly."""
+        import torch.nn.functional as F
+
+        from sdwd_amd.models.vae import VAEDownsample
+
+        ds = VAEDownsample(8)
+        x = torch.randn(2, 8, 16, 16)
+        got = ds(x)
+        ref = F.conv2d(
+            F.pad(x, (0, 1, 0, 1)), ds.conv.weight, ds.conv.bias, stride=2
+        )
+        assert got.shape == ref.shape == (2, 8, 8, 8)
+        assert torch.allclose(got, ref, atol=1e-6)
+
+    def test_odd_input(self):
+        from sdwd_amd.models.vae import VAEDownsample
+
+        ds = VAEDownsample(8)
+        out = ds(torch.randn(1, 8, 15, 15))
+        assert out.shape == (1, 8, 7, 7)
+
+
 class TestInpaintingFill:
     def _run(self, pipe, fill):
         from sdwd_amd.pipeline import PipelineRequest
