@@ -488,3 +488,15 @@ class TestPromptFeaturesGPU:
         for out in (edited, comp):
             assert torch.isfinite(out.float()).all()
             assert not torch.equal(out, plain)
+
+
+class TestInpaintModelGPU:
+    def test_9ch_model_generates(self, dev):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny-inpaint", device=dev)
+        res = pipe.generate(
+            PipelineRequest(prompt="i", steps=2, width=64, height=64,
+                            seeds=[1])
+        )
+        assert torch.isfinite(res.images.float()).all()
