@@ -324,14 +324,20 @@ def load_checkpoint(path: str, device="cpu", dtype=None) -> ModelBundle:
             # arch from the conv_in / cross-attention shapes
             from .convert import load_ldm_state_dict
 
-            ch = f.get_tensor("model.diffusion_model.input_blocks.0.0.weight"
-                              ).shape[0]
+            w_in = f.get_tensor(
+                "model.diffusion_model.input_blocks.0.0.weight"
+            )
+            ch, in_ch = int(w_in.shape[0]), int(w_in.shape[1])
             is_xl = any(k.startswith("conditioner.") for k in keys)
             is_sd2 = any(k.startswith("cond_stage_model.model.") for k in keys)
-            if int(ch) == 320:
+            if ch == 320:
                 arch = "sdxl" if is_xl else ("sd21" if is_sd2 else "sd15")
+                if in_ch == 9 and arch == "sd15":
+                    arch = "sd15-inpaint"
             else:
-                arch = "tiny-xl" if is_xl else "tiny"
+                arch = "tiny-xl" if is_xl else (
+                    "tiny-inpaint" if in_ch == 9 else "tiny"
+                )
             bundle = _BUILDERS[arch](arch)
             load_ldm_state_dict(bundle, {k: f.get_tensor(k) for k in keys})
             bundle.eval().to(device, dtype)

@@ -197,6 +197,23 @@ class TestCheckpointFile:
             assert torch.equal(a.unet(x, t, ctx), b.unet(x, t, ctx))
 
 
+class TestInpaintArchInference:
+    def test_9ch_file_loads_as_inpaint_arch(self, tmp_path):
+        from safetensors.torch import save_file
+
+        from sdwd_amd.models.registry import load_checkpoint
+
+        a = load_model("tiny-inpaint", device="cpu", cache=False)
+        exported = {
+            k: v.contiguous().clone() for k, v in to_ldm_state_dict(a).items()
+        }
+        path = str(tmp_path / "inpaint.safetensors")
+        save_file(exported, path)
+        b = load_checkpoint(path)
+        assert b.unet.cfg.in_channels == 9
+        assert b.latent_channels == 4
+
+
 class TestCheckpointDir:
     def test_file_backed_models(self, tmp_path, monkeypatch):
         """*.safetensors in SDWD_CHECKPOINT_DIR become loadable model names
