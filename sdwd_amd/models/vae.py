@@ -102,7 +102,12 @@ class VAEDownsample(nn.Module):
         self.conv = SDConv2d(ch, ch, 3, stride=1, padding=1)
 
     def forward(self, x):
-        return self.conv(x)[:, :, 1::2, 1::2]
+        out = self.conv(x)[:, :, 1::2, 1::2]
+        # the strided view is non-contiguous; the fused GN kernel (and the
+        # NHWC conv path) need a packed layout
+        if out.is_cuda:
+            return out.contiguous(memory_format=torch.channels_last)
+        return out.contiguous()
 
 
 class VAEEncoder(nn.Module):
