@@ -432,11 +432,17 @@ class TestVaeDownsampleGPU:
     def test_asym_pad_encode_matches_cpu_fp32(self, dev):
         """VAE encode (ldm asymmetric-pad downsample) on GPU bf16 vs CPU
         fp32: the sampling grid must agree (only precision noise differs)."""
+        from sdwd_amd.models import load_model
         from sdwd_amd.pipeline import StableDiffusionPipeline
 
         gp = StableDiffusionPipeline("tiny", device=dev)
-        cp = StableDiffusionPipeline("tiny", device="cpu",
-                                     dtype=torch.float32)
+        # cache=False: sharing the cached bundle would move gp's weights
+        # to CPU/fp32 under its feet
+        cp = StableDiffusionPipeline(
+            load_model("tiny", device="cpu", dtype=torch.float32,
+                       cache=False),
+            device="cpu", dtype=torch.float32,
+        )
         img = torch.randint(0, 255, (2, 64, 64, 3), dtype=torch.uint8)
         lg = gp.encode_image(img, seeds=[1, 2]).float().cpu()
         lc = cp.encode_image(img, seeds=[1, 2]).float()
