@@ -198,7 +198,7 @@ _BUILDERS: Dict[str, Callable[[str], ModelBundle]] = {
     "tiny-xl": _build_tiny_xl,
 }
 
-_cache: Dict[str, ModelBundle] = {}
+_cache: Dict[tuple, ModelBundle] = {}
 
 # file-backed checkpoints (sdwui's checkpoint folder + refresh button):
 # name (file stem) -> path, populated by refresh_checkpoint_files()
@@ -237,9 +237,12 @@ def load_model(
     name: str, device="cpu", dtype: Optional[torch.dtype] = None,
     cache: bool = True,
 ) -> ModelBundle:
-    key = f"{name}"
+    # cache per (name, device, dtype): a shared single instance would be
+    # silently .to()-moved under the feet of every other holder (e.g. a
+    # LocalEngine with one pipeline per GPU in one process)
+    key = (name, str(device), str(dtype))
     if cache and key in _cache:
-        return _cache[key].to(device, dtype)
+        return _cache[key]
     if name not in _BUILDERS and name not in _FILE_MODELS:
         refresh_checkpoint_files()
     if name in _FILE_MODELS and name not in _BUILDERS:

@@ -315,3 +315,19 @@ class TestSd15Layout:
         # but level-0 output blocks do carry attention in SD1.5
         assert ("model.diffusion_model.output_blocks.11.1.transformer_blocks"
                 ".0.attn1.to_q.weight" in sd15_keys)
+
+
+class TestModelCacheIsolation:
+    def test_cache_keys_by_device_and_dtype(self):
+        """Two holders with different dtype targets must not alias — a
+        shared instance would be .to()-moved under the first holder."""
+        import torch as _t
+
+        a = load_model("tiny", device="cpu", dtype=_t.float32)
+        b = load_model("tiny", device="cpu", dtype=_t.float64)
+        assert a.unet is not b.unet
+        assert next(a.unet.parameters()).dtype == _t.float32
+        assert next(b.unet.parameters()).dtype == _t.float64
+        # same key -> same instance (the cache still caches)
+        c = load_model("tiny", device="cpu", dtype=_t.float32)
+        assert c.unet is a.unet
