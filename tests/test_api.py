@@ -162,6 +162,31 @@ class TestControl:
         )
         assert r.status_code == 404
 
+    def test_img2img_inpaint_full_res_end_to_end(self, client):
+        from sdwd_amd.utils.images import decode_png, encode_png
+
+        init = torch.full((64, 64, 3), 90, dtype=torch.uint8)
+        mask = torch.zeros(64, 64, 3, dtype=torch.uint8)
+        mask[24:40, 24:40] = 255
+        r = client.post(
+            "/sdapi/v1/img2img",
+            json={
+                "prompt": "patch", "steps": 2, "width": 64, "height": 64,
+                "seed": 5, "denoising_strength": 1.0,
+                "init_images": [base64.b64encode(encode_png(init)).decode()],
+                "mask": base64.b64encode(encode_png(mask)).decode(),
+                "inpaint_full_res": True, "inpaint_full_res_padding": 8,
+                "mask_blur": 2, "inpainting_fill": 2,
+                "resize_mode": 0,
+            },
+        )
+        assert r.status_code == 200
+        imgs = r.json()["images"]
+        out = decode_png(base64.b64decode(imgs[-1]))
+        assert out.shape == (64, 64, 3)
+        # far corner is pasted from the original init exactly
+        assert torch.equal(out[:8, :8], init[:8, :8])
+
     def test_all_knobs_accepted(self, client):
         """Every sdwui-compat request field plumbs through end to end."""
         r = client.post(
