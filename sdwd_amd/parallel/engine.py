@@ -36,6 +36,9 @@ class GenerationRequest:
     """User-facing request = scheduling fields + pipeline fields."""
 
     prompt: str = ""
+    # optional distinct prompt per gallery image (len == batch_size);
+    # sharded with the seeds so the N-GPU gallery stays 1-GPU-identical
+    prompts: Optional[List[str]] = None
     negative_prompt: str = ""
     batch_size: int = 1
     width: int = 512
@@ -109,8 +112,16 @@ class GalleryResult:
 def _job_pipeline_request(
     gen: GenerationRequest, job: Job, init_latents=None
 ) -> PipelineRequest:
+    job_prompts = None
+    if gen.prompts:
+        n = len(gen.prompts)
+        job_prompts = [
+            gen.prompts[(job.gallery_offset + i) % n]
+            for i in range(job.batch_size)
+        ]
     return PipelineRequest(
         prompt=gen.prompt,
+        prompts=job_prompts,
         negative_prompt=gen.negative_prompt,
         steps=job.step_override or gen.steps,
         width=gen.width,

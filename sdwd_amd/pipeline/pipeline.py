@@ -30,6 +30,9 @@ log = get_logger("pipeline")
 @dataclass
 class PipelineRequest:
     prompt: str = ""
+    # optional per-image prompts (len == batch); overrides `prompt` and
+    # disables editing/AND composition for the request (plain encode)
+    prompts: Optional[List[str]] = None
     negative_prompt: str = ""
     steps: int = 20
     width: int = 512
@@ -319,8 +322,13 @@ class StableDiffusionPipeline:
         # the common single-segment case stays on the encode-once path
         from .prompt_schedule import prompt_schedule
 
-        p_segs = prompt_schedule(req.prompt, req.steps)
-        n_segs = prompt_schedule(req.negative_prompt, req.steps)
+        per_image = bool(req.prompts) and len(req.prompts) == b
+        if per_image:
+            p_segs = [(0, req.prompts[0])]
+            n_segs = [(0, req.negative_prompt)]
+        else:
+            p_segs = prompt_schedule(req.prompt, req.steps)
+            n_segs = prompt_schedule(req.negative_prompt, req.steps)
         # img2img runs only the schedule tail; editing thresholds are
         # fractions of req.steps, so segments before the tail's first step
         # collapse into the base conditioning
@@ -334,11 +342,28 @@ class StableDiffusionPipeline:
             return [txt for st, txt in segs if st <= step][-1]
 
         # one conditioning set per request (encode once, broadcast across
-        # the batch); AND composition yields k cond rows + 1 uncond row
-        ctx, y, and_ws = self._build_ctx(
-            _seg_text_at(p_segs, start_off), _seg_text_at(n_segs, start_off),
-            req, b,
-        )
+        # the batch); AND composition yields k cond rows + 1 uncond row.
+        # Per-image prompts encode a cond row per image instead.
+        if per_image:
+            c_rows, u_rows, pl = self.encode_prompts(
+                list(req.prompts), [req.negative_prompt] * b, req.clip_skip
+            )
+            ctx = torch.cat([c_rows, u_rows], dim=0)
+            and_ws = [1.0]
+            y = None
+            if self.model.is_sdxl and pl is not None:
+                y = torch.cat(
+                    [
+                        self._sdxl_vector(req, pl[0]),
+                        self._sdxl_vector(req, pl[1]),
+                    ]
+                )
+        else:
+            ctx, y, and_ws = self._build_ctx(
+                _seg_text_at(p_segs, start_off),
+                _seg_text_at(n_segs, start_off),
+                req, b,
+            )
 
         sched = schedule_for(req.sampler_name, req.steps, req.scheduler)
         sampler = build_sampler(req.sampler_name, sched)
@@ -413,7 +438,7 @@ class StableDiffusionPipeline:
         # prompt-editing segments: (t_threshold, ctx, y, weights) per
         # conditioning change, selected by the current timestep in model_fn
         seg_tensors: List[tuple] = []
-        if len(p_segs) > 1 or len(n_segs) > 1:
+        if (len(p_segs) > 1 or len(n_segs) > 1) and not per_image:
             ts_all = sched.timesteps.tolist()
             boundaries = sorted(
                 {s for s, _ in p_segs} | {s for s, _ in n_segs}
@@ -743,8 +768,11 @@ class StableDiffusionPipeline:
             )
         if req.tiling:
             extra += ", Tiling: True"
+        prompt_of = (
+            (lambda i: req.prompts[i]) if per_image else (lambda i: req.prompt)
+        )
         infotexts = [
-            f"{req.prompt}\nNegative prompt: {req.negative_prompt}\n"
+            f"{prompt_of(i)}\nNegative prompt: {req.negative_prompt}\n"
             f"Steps: {req.steps}, Sampler: {req.sampler_name}, "
             f"CFG scale: {req.cfg_scale}, Seed: {req.seeds[i]}, "
             f"Size: {req.width}x{req.height}, Model: {self.model.name}"

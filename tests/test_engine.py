@@ -258,6 +258,28 @@ class TestEdgeCases:
         )
         assert eng.pipes["gpu0"] is p
 
+    def test_per_image_prompts_shard_deterministically(self):
+        """Distinct prompt per gallery row: the 2-rank gallery must equal
+        the 1-rank gallery image-for-image (prompt follows the seed)."""
+        prompts = ["a cow", "a dog", "a cat", "a bird"]
+        base = dict(batch_size=4, width=64, height=64, steps=2, seed=70,
+                    prompts=prompts)
+        one = make_engine(1).generate(GenerationRequest(**base))
+        two = make_engine(2).generate(GenerationRequest(**base))
+        diff = (one.images.float() - two.images.float()).abs()
+        assert diff.max() <= 1.0
+        # each infotext carries its own prompt
+        for i, p in enumerate(prompts):
+            assert two.infotexts[i].startswith(p), two.infotexts[i]
+        # and differs from a single-prompt run
+        single = make_engine(1).generate(
+            GenerationRequest(**{**base, "prompts": None, "prompt": "a cow"})
+        )
+        assert not torch.equal(one.images[1], single.images[1])
+        assert torch.allclose(
+            one.images[0].float(), single.images[0].float(), atol=1.0
+        )
+
     def test_inpaint_full_res_preserves_outside(self):
         """sdwui "Only masked": pixels outside the padded crop box equal
         the original init EXACTLY (paste, not regeneration)."""
