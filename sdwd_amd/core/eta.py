@@ -30,6 +30,9 @@ SAMPLER_EVALS_PER_STEP = {
     "LMS": 1.0,
     "DPM++ 2M": 1.0,
     "DPM++ 2M Karras": 1.0,
+    "DPM++ 2M SDE": 1.0,
+    "DPM++ 2M SDE Karras": 1.0,
+    "LCM": 1.0,
     "DPM++ SDE": 2.0,
     "DPM++ SDE Karras": 2.0,
     "Heun": 2.0,

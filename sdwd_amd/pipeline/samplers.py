@@ -208,6 +208,40 @@ class DPMppSDE(Sampler):
         return x
 
 
+class DPMpp2MSDE(Sampler):
+    """DPM++ 2M SDE (k-diffusion sample_dpmpp_2m_sde, midpoint solver,
+    eta=1): the 2M multistep update with an SDE noise injection."""
+
+    eta = 1.0
+
+    def reset(self):
+        self.old_denoised = None
+        self.h_last = None
+
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
+        denoised = _eval(model_fn, x, sigma, t)
+        if sigma_next <= 0:
+            self.old_denoised = denoised
+            return denoised
+        tt, tn = -math.log(sigma), -math.log(sigma_next)
+        h = tn - tt
+        eta_h = self.eta * h
+        decay = math.exp(-eta_h)
+        phi = -math.expm1(-h - eta_h)
+        x = ops.lincomb(x, denoised, (sigma_next / sigma) * decay, phi)
+        if self.old_denoised is not None and self.h_last is not None:
+            r = self.h_last / h
+            d = ops.lincomb(denoised, self.old_denoised, 1.0, -1.0)
+            x = ops.lincomb(x, d, 1.0, 0.5 * phi / r)
+        if self.eta > 0 and noise_fn is not None:
+            amp = sigma_next * math.sqrt(max(0.0, -math.expm1(-2 * eta_h)))
+            x = ops.add_noise(x, noise_fn(), 1.0, amp)
+        self.old_denoised = denoised
+        self.h_last = h
+        return x
+
+
 class DPM2(Sampler):
     """k-diffusion sample_dpm_2: explicit midpoint in log-sigma."""
 
@@ -411,6 +445,8 @@ SAMPLERS: Dict[str, type] = {
     "DPM++ 2M Karras": DPMpp2M,
     "DPM++ SDE": DPMppSDE,
     "DPM++ SDE Karras": DPMppSDE,
+    "DPM++ 2M SDE": DPMpp2MSDE,
+    "DPM++ 2M SDE Karras": DPMpp2MSDE,
     "LMS": LMS,
     "LMS Karras": LMS,
     "DPM2": DPM2,

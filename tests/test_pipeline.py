@@ -311,7 +311,8 @@ class TestSamplerOracle:
     @pytest.mark.parametrize("name", ["Euler", "Euler a", "DDIM", "Heun",
                                       "DPM++ 2M", "DPM++ SDE", "LMS",
                                       "DPM2", "DPM2 a", "DDPM",
-                                      "DPM++ 2S a", "UniPC"])
+                                      "DPM++ 2S a", "UniPC",
+                                      "DPM++ 2M SDE"])
     def test_converges_to_point_mass(self, name):
         from sdwd_amd.pipeline.samplers import build_sampler
         from sdwd_amd.pipeline.schedule import discrete_schedule, make_sigmas_full
@@ -342,7 +343,8 @@ class TestSamplerOracle:
 
 class TestKarrasOracle:
     @pytest.mark.parametrize("name", ["DPM++ 2M Karras", "DPM++ SDE Karras",
-                                      "DPM++ 2S a Karras"])
+                                      "DPM++ 2S a Karras",
+                                      "DPM++ 2M SDE Karras"])
     def test_karras_also_converges(self, name):
         """Same point-mass oracle through the Karras sigma schedule (uses
         fractional timesteps -> sigma via exp-interp of the log table)."""
