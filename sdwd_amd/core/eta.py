@@ -32,6 +32,8 @@ SAMPLER_EVALS_PER_STEP = {
     "DPM++ 2M Karras": 1.0,
     "DPM++ 2M SDE": 1.0,
     "DPM++ 2M SDE Karras": 1.0,
+    "DPM++ 3M SDE": 1.0,
+    "DPM++ 3M SDE Karras": 1.0,
     "LCM": 1.0,
     "DPM++ SDE": 2.0,
     "DPM++ SDE Karras": 2.0,

@@ -242,6 +242,62 @@ class DPMpp2MSDE(Sampler):
         return x
 
 
+class DPMpp3MSDE(Sampler):
+    """DPM++ 3M SDE (k-diffusion sample_dpmpp_3m_sde, eta=1): third-order
+    multistep with SDE noise."""
+
+    eta = 1.0
+
+    def reset(self):
+        self.den_1 = None
+        self.den_2 = None
+        self.h_1 = None
+        self.h_2 = None
+
+    def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
+             t_next=None):
+        denoised = _eval(model_fn, x, sigma, t)
+        if sigma_next <= 0:
+            return denoised
+        tt, tn = -math.log(sigma), -math.log(sigma_next)
+        h = tn - tt
+        h_eta = h * (self.eta + 1.0)
+        phi_1 = -math.expm1(-h_eta)
+        x = ops.lincomb(x, denoised, math.exp(-h_eta), phi_1)
+        if self.den_1 is not None and self.h_1 is not None:
+            phi_2 = phi_1 / h_eta - 1.0  # note: negative of k-diff's phi_2
+            if self.den_2 is not None and self.h_2 is not None:
+                r0, r1 = self.h_1 / h, self.h_2 / h
+                d1_0 = ops.lincomb(denoised, self.den_1, 1 / r0, -1 / r0)
+                d1_1 = ops.lincomb(self.den_1, self.den_2, 1 / r1, -1 / r1)
+                d1 = ops.lincomb(
+                    d1_0, ops.lincomb(d1_0, d1_1, 1.0, -1.0),
+                    1.0, r0 / (r0 + r1),
+                )
+                d2 = ops.lincomb(
+                    d1_0, d1_1, 1 / (r0 + r1), -1 / (r0 + r1)
+                )
+                # k-diff: x += phi2_kd*d1 - phi3_kd*d2 with
+                # phi2_kd = -phi_2, phi3_kd = phi2_kd/h_eta - 0.5
+                phi_3 = phi_2 / h_eta + 0.5  # = -phi3_kd
+                x = ops.lincomb(x, d1, 1.0, -phi_2)
+                x = ops.lincomb(x, d2, 1.0, phi_3)
+            else:
+                r = self.h_1 / h
+                d = ops.lincomb(denoised, self.den_1, 1 / r, -1 / r)
+                x = ops.lincomb(x, d, 1.0, -phi_2)
+        if self.eta > 0 and noise_fn is not None:
+            amp = sigma_next * math.sqrt(
+                max(0.0, -math.expm1(-2 * h * self.eta))
+            )
+            x = ops.add_noise(x, noise_fn(), 1.0, amp)
+        self.den_2 = self.den_1
+        self.den_1 = denoised
+        self.h_2 = self.h_1
+        self.h_1 = h
+        return x
+
+
 class DPM2(Sampler):
     """k-diffusion sample_dpm_2: explicit midpoint in log-sigma."""
 
@@ -447,6 +503,8 @@ SAMPLERS: Dict[str, type] = {
     "DPM++ SDE Karras": DPMppSDE,
     "DPM++ 2M SDE": DPMpp2MSDE,
     "DPM++ 2M SDE Karras": DPMpp2MSDE,
+    "DPM++ 3M SDE": DPMpp3MSDE,
+    "DPM++ 3M SDE Karras": DPMpp3MSDE,
     "LMS": LMS,
     "LMS Karras": LMS,
     "DPM2": DPM2,

@@ -312,7 +312,7 @@ class TestSamplerOracle:
                                       "DPM++ 2M", "DPM++ SDE", "LMS",
                                       "DPM2", "DPM2 a", "DDPM",
                                       "DPM++ 2S a", "UniPC",
-                                      "DPM++ 2M SDE"])
+                                      "DPM++ 2M SDE", "DPM++ 3M SDE"])
     def test_converges_to_point_mass(self, name):
         from sdwd_amd.pipeline.samplers import build_sampler
         from sdwd_amd.pipeline.schedule import discrete_schedule, make_sigmas_full
@@ -344,7 +344,8 @@ class TestSamplerOracle:
 class TestKarrasOracle:
     @pytest.mark.parametrize("name", ["DPM++ 2M Karras", "DPM++ SDE Karras",
                                       "DPM++ 2S a Karras",
-                                      "DPM++ 2M SDE Karras"])
+                                      "DPM++ 2M SDE Karras",
+                                      "DPM++ 3M SDE Karras"])
     def test_karras_also_converges(self, name):
         """Same point-mass oracle through the Karras sigma schedule (uses
         fractional timesteps -> sigma via exp-interp of the log table)."""
