@@ -148,27 +148,26 @@ def _resize_init(img: torch.Tensor, w: int, h: int, mode: int) -> torch.Tensor:
 
 def _parse_controlnet(alwayson: Dict[str, Any]):
     """sdwui controlnet payload: {"controlnet": {"args": [unit, ...]}}
-    (ref control_net.py:20-79 packed this; we consume unit 0 natively).
-    Unknown scripts are logged and skipped (ref C18 compat filter)."""
-    control_image = None
-    control_model = ""
-    control_scale = 1.0
+    (ref control_net.py:20-79 packed this; every unit is executed natively
+    with its weight and guidance window). Unknown scripts are logged and
+    skipped (ref C18 compat filter)."""
+    units = []
     for name, body in (alwayson or {}).items():
         if name.lower() != "controlnet":
             log.warning("ignoring unsupported alwayson script '%s'", name)
             continue
-        args = (body or {}).get("args", [])
-        if not args:
-            continue
-        unit = args[0]
-        img_b64 = unit.get("input_image") or unit.get("image")
-        if img_b64:
-            control_image = _decode_b64_png(img_b64)[None]
-        control_model = unit.get("model", "controlnet-sd15")
-        control_scale = float(unit.get("weight", 1.0))
-        if len(args) > 1:
-            log.warning("only the first controlnet unit is applied")
-    return control_image, control_model, control_scale
+        for unit in (body or {}).get("args", []):
+            img_b64 = unit.get("input_image") or unit.get("image")
+            if not img_b64:
+                continue
+            units.append({
+                "image": _decode_b64_png(img_b64)[None],
+                "model": unit.get("model", "controlnet-sd15"),
+                "scale": float(unit.get("weight", 1.0)),
+                "guidance_start": float(unit.get("guidance_start", 0.0)),
+                "guidance_end": float(unit.get("guidance_end", 1.0)),
+            })
+    return units
 
 
 def create_app(engine: Optional[LocalEngine] = None,
@@ -275,9 +274,7 @@ def create_app(engine: Optional[LocalEngine] = None,
 
     @app.post("/sdapi/v1/txt2img")
     def txt2img(req: Txt2ImgRequest):
-        control_image, control_model, control_scale = _parse_controlnet(
-            req.alwayson_scripts
-        )
+        control_units = _parse_controlnet(req.alwayson_scripts)
         model, clip_skip = _overrides(req)
         gen = GenerationRequest(
             prompt=req.prompt,
@@ -304,9 +301,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             hr_upscaler=req.hr_upscaler,
             denoising_strength=req.denoising_strength,
             clip_skip=clip_skip,
-            control_image=control_image,
-            control_model=control_model,
-            control_scale=control_scale,
+            control_units=control_units,
             model=model,
             refiner_model=req.refiner_checkpoint,
             refiner_switch_at=req.refiner_switch_at,

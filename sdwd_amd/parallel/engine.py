@@ -60,6 +60,7 @@ class GenerationRequest:
     control_image: Optional[torch.Tensor] = None
     control_model: str = ""
     control_scale: float = 1.0
+    control_units: Optional[List[dict]] = None  # multi-unit controlnet
     clip_skip: int = 1
     model: str = ""  # hot-swap to this checkpoint first ("" = keep current)
     refiner_model: str = ""  # two-model refiner handoff (sdwui refiner)
@@ -143,6 +144,7 @@ def _job_pipeline_request(
         control_image=gen.control_image,
         control_model=gen.control_model,
         control_scale=gen.control_scale,
+        control_units=gen.control_units,
         clip_skip=gen.clip_skip,
         refiner_model=gen.refiner_model,
         refiner_switch_at=gen.refiner_switch_at,

@@ -71,10 +71,13 @@ class PipelineRequest:
     s_noise: float = 1.0
     # skip the uncond eval below this sigma (sdwui s_min_uncond perf knob)
     s_min_uncond: float = 0.0
-    # controlnet (ref C17 executed natively)
+    # controlnet (ref C17 executed natively); either the single-unit
+    # legacy fields or a list of unit dicts with
+    # {image, model, scale, guidance_start, guidance_end}
     control_image: Optional[torch.Tensor] = None  # [B,H,W,3] uint8
     control_model: str = ""  # e.g. "controlnet-sd15"; "" = off
     control_scale: float = 1.0
+    control_units: Optional[List[dict]] = None
     clip_skip: int = 1  # 1 = final layer; 2 = penultimate (sdwui setting)
 
     @property
@@ -466,21 +469,75 @@ class StableDiffusionPipeline:
 
         denoiser = self._denoiser
 
-        controlnet = None
-        hint = None
+        # controlnet units: [(module, hint, scale, t_begin, t_end)] with a
+        # per-unit guidance window in t-space (sdwui guidance_start/end)
+        units = list(req.control_units or [])
         if req.control_model and req.control_image is not None:
+            units.insert(0, {
+                "image": req.control_image, "model": req.control_model,
+                "scale": req.control_scale,
+            })
+        cn_units = []
+        if units:
             from ..models.registry import load_controlnet
 
-            controlnet = load_controlnet(
-                req.control_model, device=self.device, dtype=self.dtype
-            )
-            if self.device.type == "cuda":
-                controlnet.to(memory_format=torch.channels_last)
-            hint = (
-                req.control_image.permute(0, 3, 1, 2).float() / 255.0
-            ).to(self.device, self.dtype)
-            if hint.shape[0] == 1 and b > 1:
-                hint = hint.expand(b, -1, -1, -1)
+            ts_full = sched.timesteps.tolist()
+            for u in units:
+                img = u.get("image")
+                name = u.get("model") or "controlnet-sd15"
+                if img is None:
+                    continue
+                mod = load_controlnet(
+                    name, device=self.device, dtype=self.dtype
+                )
+                if self.device.type == "cuda":
+                    mod.to(memory_format=torch.channels_last)
+                hint_u = (
+                    img.permute(0, 3, 1, 2).float() / 255.0
+                ).to(self.device, self.dtype)
+                if hint_u.shape[0] == 1 and b > 1:
+                    hint_u = hint_u.expand(b, -1, -1, -1)
+                g0 = u.get("guidance_start", 0.0)
+                g0 = 0.0 if g0 is None else float(g0)
+                g1 = u.get("guidance_end", 1.0)
+                g1 = 1.0 if g1 is None else float(g1)
+                i0 = max(0, min(len(ts_full) - 1, int(g0 * len(ts_full))))
+                i1 = int(g1 * len(ts_full))
+                t_begin = ts_full[i0] if g0 > 0 else float("inf")
+                t_end = (
+                    ts_full[i1] if i1 < len(ts_full) else float("-inf")
+                )
+                cn_units.append((
+                    mod, hint_u, float(u.get("scale", 1.0)), t_begin, t_end,
+                ))
+        controlnet = bool(cn_units)
+
+        def _control_residuals(xk, ts, c_ctx, t):
+            """Sum the active units' residuals at timestep t."""
+            total = None
+            for mod, hint_u, scale, t_begin, t_end in cn_units:
+                if not (t <= t_begin + 1e-6 and t > t_end):
+                    continue
+                k1 = xk.shape[0] // hint_u.shape[0]
+                hk = torch.cat([hint_u] * k1, dim=0)
+                xc = xk
+                if xk.shape[1] != mod.cfg.in_channels:
+                    # inpaint models carry extra conditioning channels the
+                    # controlnet (a 4ch base-UNet copy) doesn't take
+                    xc = xk[:, : mod.cfg.in_channels]
+                ctrl = mod(xc, hk, ts, c_ctx, scale)
+                if total is None:
+                    total = ctrl
+                else:
+                    total = {
+                        "down": [
+                            a + bb for a, bb in zip(
+                                total["down"], ctrl["down"]
+                            )
+                        ],
+                        "mid": total["mid"] + ctrl["mid"],
+                    }
+            return total
 
         def _to_eps(out, x_scaled, t, pred_type):
             """v-prediction -> epsilon on the sampler's scaled input:
@@ -561,9 +618,10 @@ class StableDiffusionPipeline:
             if is_inpaint_model:
                 ex = _inpaint_extra(x_in.shape[2], x_in.shape[3])
                 xk = torch.cat([xk, torch.cat([ex] * k1, dim=0)], dim=1)
-            if controlnet is not None:
-                hk = torch.cat([hint] * k1, dim=0)
-                ctrl = controlnet(xk, hk, ts, c_ctx, req.control_scale)
+            ctrl = (
+                _control_residuals(xk, ts, c_ctx, t) if controlnet else None
+            )
+            if ctrl is not None:
                 eps = unet(xk, ts, c_ctx, y=c_y, control=ctrl)
             else:
                 eps = denoiser(xk, ts, c_ctx, c_y)

@@ -568,6 +568,78 @@ class TestPromptEditing:
         assert torch.equal(a, c)
 
 
+class TestControlNetUnits:
+    def _hint(self, seed=0):
+        g = torch.Generator().manual_seed(seed)
+        return torch.randint(0, 255, (1, 64, 64, 3), generator=g,
+                             dtype=torch.int64).to(torch.uint8)
+
+    def test_unit_list_equals_legacy_single(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="c", steps=2, width=64, height=64, seeds=[4])
+        hint = self._hint(1)
+        legacy = pipe.generate(PipelineRequest(
+            **base, control_image=hint, control_model="controlnet-tiny",
+            control_scale=0.8,
+        )).images
+        unit = pipe.generate(PipelineRequest(
+            **base, control_units=[{
+                "image": hint, "model": "controlnet-tiny", "scale": 0.8,
+            }],
+        )).images
+        assert torch.equal(legacy, unit)
+
+    def test_two_units_differ_from_one(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="c", steps=2, width=64, height=64, seeds=[4])
+        u1 = {"image": self._hint(1), "model": "controlnet-tiny",
+              "scale": 0.5}
+        u2 = {"image": self._hint(2), "model": "controlnet-tiny",
+              "scale": 0.5}
+        one = pipe.generate(
+            PipelineRequest(**base, control_units=[u1])
+        ).images
+        two = pipe.generate(
+            PipelineRequest(**base, control_units=[u1, u2])
+        ).images
+        assert not torch.equal(one, two)
+
+    def test_zero_guidance_window_is_no_op(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="c", steps=3, width=64, height=64, seeds=[4])
+        plain = pipe.generate(PipelineRequest(**base)).images
+        gated = pipe.generate(PipelineRequest(
+            **base, control_units=[{
+                "image": self._hint(1), "model": "controlnet-tiny",
+                "scale": 1.0, "guidance_start": 0.0, "guidance_end": 0.0,
+            }],
+        )).images
+        assert torch.equal(plain, gated)
+
+    def test_late_window_changes_only_late_steps(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="c", steps=4, width=64, height=64, seeds=[4])
+        full = pipe.generate(PipelineRequest(
+            **base, control_units=[{
+                "image": self._hint(1), "model": "controlnet-tiny",
+                "scale": 1.0,
+            }],
+        )).images
+        late = pipe.generate(PipelineRequest(
+            **base, control_units=[{
+                "image": self._hint(1), "model": "controlnet-tiny",
+                "scale": 1.0, "guidance_start": 0.5,
+            }],
+        )).images
+        plain = pipe.generate(PipelineRequest(**base)).images
+        assert not torch.equal(late, plain)
+        assert not torch.equal(late, full)
+
+
 class TestSamplerParams:
     def test_churn_changes_euler_output(self, pipe):
         from sdwd_amd.pipeline import PipelineRequest
