@@ -435,6 +435,43 @@ def create_app(engine: Optional[LocalEngine] = None,
             for n, (p, np) in sorted(all_styles().items())
         ]
 
+    # -- static enumerations sdwui GUIs query at startup ---------------------
+    @app.get("/sdapi/v1/upscalers")
+    def upscalers():
+        return [
+            {"name": n, "model_name": None, "model_path": None, "scale": 4}
+            for n in ("None", "Nearest", "Bilinear", "Bicubic", "Lanczos")
+        ]
+
+    @app.get("/sdapi/v1/latent-upscale-modes")
+    def latent_upscale_modes():
+        return [
+            {"name": n}
+            for n in ("Latent", "Latent (bilinear)", "Latent (bicubic)",
+                      "Latent (bilinear antialiased)",
+                      "Latent (bicubic antialiased)")
+        ]
+
+    @app.get("/sdapi/v1/face-restorers")
+    def face_restorers():
+        return [{"name": "None", "cmd_dir": None}]
+
+    @app.get("/sdapi/v1/embeddings")
+    def embeddings():
+        return {"loaded": {}, "skipped": {}}
+
+    @app.get("/sdapi/v1/hypernetworks")
+    def hypernetworks():
+        return []
+
+    @app.get("/sdapi/v1/scripts")
+    def scripts():
+        return {"txt2img": [], "img2img": []}
+
+    @app.get("/sdapi/v1/sd-vae")
+    def sd_vae():
+        return [{"model_name": "auto", "filename": ""}]
+
     @app.get("/sdapi/v1/schedulers")
     def schedulers():
         from ..pipeline.schedule import scheduler_names

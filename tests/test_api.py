@@ -284,6 +284,15 @@ class TestControl:
         out2 = decode_png(base64.b64decode(r2.json()["image"]))
         assert out2.shape == (40, 24, 3)
 
+    def test_startup_enumeration_routes(self, client):
+        """The static lists sdwui GUIs fetch at startup all answer."""
+        for route in ("upscalers", "latent-upscale-modes", "face-restorers",
+                      "hypernetworks", "scripts", "embeddings", "sd-vae"):
+            r = client.get(f"/sdapi/v1/{route}")
+            assert r.status_code == 200, route
+        ups = [u["name"] for u in client.get("/sdapi/v1/upscalers").json()]
+        assert "Lanczos" in ups
+
     def test_schedulers_route(self, client):
         names = [s["label"] for s in client.get("/sdapi/v1/schedulers").json()]
         assert "Karras" in names and "Exponential" in names
