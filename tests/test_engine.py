@@ -198,6 +198,51 @@ class TestDistributedEngine:
         assert all(gallery[i].float().std() > 0 for i in range(4))
 
 
+def _thin_worker(rank, world_size, port, tmpdir):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from sdwd_amd.parallel import DistributedEngine, GenerationRequest, destroy_group
+
+    eng = DistributedEngine(model="tiny", backend="gloo")
+    eng.world.settings.thin_client = True
+    for w in eng.world.workers:
+        w.eta.avg_ipm = 60.0
+    res = eng.generate(
+        GenerationRequest(prompt="thin", batch_size=3, width=64, height=64,
+                          steps=1, seed=40)
+    )
+    if rank == 0:
+        assert res.images.shape == (3, 64, 64, 3)
+        # rank 0 orchestrates only: every image came from gpu1
+        for line in res.job_summary:
+            if line.startswith("gpu0:") and "(complementary)" not in line:
+                n = int(line.split(":")[1].strip().split(" ")[0])
+                assert n == 0, res.job_summary
+        with open(os.path.join(tmpdir, "thin_ok"), "w") as fh:
+            fh.write("ok")
+    destroy_group()
+
+
+@pytest.mark.timeout(300)
+class TestThinClientDistributed:
+    def test_master_takes_no_shard(self, tmp_path):
+        import socket
+
+        import torch.multiprocessing as mp
+
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        mp.start_processes(
+            _thin_worker, args=(2, port, str(tmp_path)), nprocs=2,
+            start_method="spawn", join=True,
+        )
+        assert (tmp_path / "thin_ok").exists()
+
+
 def _hb_worker(rank, world_size, port, tmpdir):
     os.environ["RANK"] = str(rank)
     os.environ["LOCAL_RANK"] = str(rank)
