@@ -544,6 +544,16 @@ def create_app(engine: Optional[LocalEngine] = None,
     @app.get("/sdapi/v1/progress")
     def progress():
         frac = engine.progress() if state.busy else 1.0
+        current = None
+        if state.busy:
+            # cheap live preview (sdwui current_image): approximate
+            # latent->RGB of rank 0's in-flight latents, no VAE decode
+            try:
+                pv = engine.pipes["gpu0"].preview_image()
+                if pv is not None:
+                    current = _b64_png(pv)
+            except Exception:  # preview must never break progress polling
+                current = None
         return {
             "progress": frac,
             "eta_relative": 0.0,
@@ -551,7 +561,7 @@ def create_app(engine: Optional[LocalEngine] = None,
                 "job": "generate" if state.busy else "",
                 "interrupted": engine.world.interrupted.is_set(),
             },
-            "current_image": None,
+            "current_image": current,
         }
 
     @app.post("/sdapi/v1/server-restart")

@@ -190,6 +190,7 @@ class StableDiffusionPipeline:
 
         self.lora = LoraManager(self.model.unet)
         self._tiling = False
+        self._last_preview = None  # latest denoise-loop latents (preview)
         self._denoiser = GraphedDenoiser(
             lambda x, ts, ctx, y: self.model.unet(x, ts, ctx, y=y),
             self.device,
@@ -727,6 +728,14 @@ class StableDiffusionPipeline:
                     lat_mask * xc.float() + (1.0 - lat_mask) * keep
                 ).to(xc.dtype)
 
+        orig_post = post_step
+
+        def post_step(xc, sigma_next):  # noqa: F811 - compose preview store
+            if orig_post is not None:
+                xc = orig_post(xc, sigma_next)
+            self._last_preview = xc
+            return xc
+
         x = sampler.sample(
             model_fn, x, noise_fn=noise_fn, callback=step_callback,
             interrupt=_interrupt, post_step=post_step,
@@ -849,6 +858,27 @@ class StableDiffusionPipeline:
             elapsed=elapsed,
             interrupted=was_interrupted,
         )
+
+    # approximate latent->RGB projection (the well-known SD 4ch linear
+    # map; sdwui's "cheap" live-preview mode)
+    _L2RGB = torch.tensor([
+        [0.298, 0.207, 0.208],
+        [0.187, 0.286, 0.173],
+        [-0.158, 0.189, 0.264],
+        [-0.184, -0.271, -0.473],
+    ])
+
+    @torch.no_grad()
+    def preview_image(self) -> Optional[torch.Tensor]:
+        """Cheap preview of the generation in flight: [h,w,3] uint8 at
+        latent resolution (no VAE decode), or None."""
+        lat = self._last_preview
+        if lat is None:
+            return None
+        l4 = lat[0, :4].float().cpu()  # first image, first 4 channels
+        rgb = torch.einsum("chw,cr->rhw", l4, self._L2RGB)
+        rgb = ((rgb / 3.0 + 0.5) * 255.0).clamp(0, 255)
+        return rgb.permute(1, 2, 0).to(torch.uint8)
 
     @torch.no_grad()
     def encode_image(

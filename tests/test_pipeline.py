@@ -568,6 +568,20 @@ class TestPromptEditing:
         assert torch.equal(a, c)
 
 
+class TestPreview:
+    def test_preview_available_after_generation(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        pipe.generate(
+            PipelineRequest(prompt="p", steps=2, width=64, height=64,
+                            seeds=[1])
+        )
+        pv = pipe.preview_image()
+        assert pv is not None
+        assert pv.shape == (32, 32, 3)  # latent resolution (f=2 tiny VAE)
+        assert pv.dtype == torch.uint8
+
+
 class TestControlNetUnits:
     def _hint(self, seed=0):
         g = torch.Generator().manual_seed(seed)
