@@ -608,9 +608,15 @@ class StableDiffusionPipeline:
                         [x_in, _inpaint_extra(x_in.shape[2], x_in.shape[3])],
                         dim=1,
                     )
-                return _to_eps(
-                    denoiser(x1, ts1, c_ctx[:nb], yc), x_in, t, pred_type
+                ctrl1 = (
+                    _control_residuals(x1, ts1, c_ctx[:nb], t)
+                    if controlnet else None
                 )
+                if ctrl1 is not None:
+                    out1 = unet(x1, ts1, c_ctx[:nb], y=yc, control=ctrl1)
+                else:
+                    out1 = denoiser(x1, ts1, c_ctx[:nb], yc)
+                return _to_eps(out1, x_in, t, pred_type)
             k1 = len(ws) + 1  # k AND-conds + 1 uncond
             ts = torch.full(
                 (nb * k1,), t, device=self.device, dtype=torch.float32,
@@ -640,7 +646,7 @@ class StableDiffusionPipeline:
                 out = _ops.lincomb(out, d, 1.0, cfg * w)
             return out
 
-        if cfg == 1.0 and not seg_tensors and and_ws == [1.0]:
+        if cfg == 1.0 and not seg_tensors and and_ws == [1.0] and not cn_units:
 
             def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:  # noqa: F811
                 nb = x_in.shape[0]

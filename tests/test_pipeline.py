@@ -633,6 +633,36 @@ class TestControlNetUnits:
         )).images
         assert torch.equal(plain, gated)
 
+    def test_controlnet_applies_at_cfg_one(self, pipe):
+        """Regression: the cfg==1 fast path must not bypass control."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="c", steps=2, width=64, height=64, seeds=[4],
+                    cfg_scale=1.0)
+        plain = pipe.generate(PipelineRequest(**base)).images
+        controlled = pipe.generate(PipelineRequest(
+            **base, control_units=[{
+                "image": self._hint(1), "model": "controlnet-tiny",
+                "scale": 1.0,
+            }],
+        )).images
+        assert not torch.equal(plain, controlled)
+
+    def test_controlnet_applies_with_s_min_uncond(self, pipe):
+        """Regression: the uncond-skip path must still apply control."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="c", steps=2, width=64, height=64, seeds=[4],
+                    s_min_uncond=1e9)
+        plain = pipe.generate(PipelineRequest(**base)).images
+        controlled = pipe.generate(PipelineRequest(
+            **base, control_units=[{
+                "image": self._hint(1), "model": "controlnet-tiny",
+                "scale": 1.0,
+            }],
+        )).images
+        assert not torch.equal(plain, controlled)
+
     def test_late_window_changes_only_late_steps(self, pipe):
         from sdwd_amd.pipeline import PipelineRequest
 
