@@ -303,6 +303,47 @@ class TestEdgeCases:
         )
         assert eng.pipes["gpu0"] is p
 
+    def test_img2img_init_cycling(self):
+        """Fewer init images than batch: inits cycle by gallery index, so
+        sharding can't change which init image k gets."""
+        eng1 = make_engine(1)
+        eng2 = make_engine(2)
+        inits = torch.stack([
+            torch.full((64, 64, 3), 40, dtype=torch.uint8),
+            torch.full((64, 64, 3), 220, dtype=torch.uint8),
+        ])
+        base = dict(prompt="cyc", batch_size=4, width=64, height=64,
+                    steps=2, seed=55, init_images=inits,
+                    denoising_strength=0.3)
+        a = eng1.generate(GenerationRequest(**base)).images
+        b = eng2.generate(GenerationRequest(**base)).images
+        diff = (a.float() - b.float()).abs()
+        assert diff.max() <= 1.0
+        # low strength: image 0/2 stay near dark init, 1/3 near bright
+        assert a[0].float().mean() < a[1].float().mean()
+        assert a[2].float().mean() < a[3].float().mean()
+
+    def test_all_ranks_unbenchmarked_equal_split(self):
+        eng = LocalEngine(model="tiny", devices=["cpu", "cpu"])
+        for w in eng.world.workers:
+            assert w.eta.avg_ipm == 0.0
+        res = eng.generate(
+            GenerationRequest(prompt="e", batch_size=4, width=64, height=64,
+                              steps=1, seed=2)
+        )
+        assert res.images.shape == (4, 64, 64, 3)
+
+    def test_interrupt_mid_hires(self):
+        eng = make_engine(1)
+        calls = []
+        eng.world.get_worker("gpu0").interrupt_event.set()
+        res = eng.generate(
+            GenerationRequest(prompt="h", batch_size=1, width=64, height=64,
+                              steps=4, seed=3, enable_hr=True, hr_scale=2.0,
+                              denoising_strength=0.5)
+        )
+        assert res.interrupted or res.images.numel() > 0
+
     def test_per_image_prompts_shard_deterministically(self):
         """Distinct prompt per gallery row: the 2-rank gallery must equal
         the 1-rank gallery image-for-image (prompt follows the seed)."""
