@@ -335,7 +335,6 @@ class TestEdgeCases:
 
     def test_interrupt_mid_hires(self):
         eng = make_engine(1)
-        calls = []
         eng.world.get_worker("gpu0").interrupt_event.set()
         res = eng.generate(
             GenerationRequest(prompt="h", batch_size=1, width=64, height=64,
