@@ -37,6 +37,8 @@ class Txt2ImgRequest(BaseModel):
     seed: int = -1
     subseed: int = -1
     subseed_strength: float = 0.0
+    seed_resize_from_w: int = Field(default=0, ge=0)
+    seed_resize_from_h: int = Field(default=0, ge=0)
     steps: int = 20
     cfg_scale: float = 7.0
     width: int = 512
@@ -258,6 +260,7 @@ def create_app(engine: Optional[LocalEngine] = None,
         if model and model not in available_models():
             raise HTTPException(404, f"unknown model {model}")
         clip_skip = int(ov.get("CLIP_stop_at_last_layers") or req.clip_skip)
+        ensd = int(ov.get("eta_noise_seed_delta") or 0)
         if req.styles:
             from ..pipeline.styles import all_styles, apply_styles, refresh_styles
 
@@ -270,12 +273,12 @@ def create_app(engine: Optional[LocalEngine] = None,
             req.refiner_checkpoint not in available_models()
         ):
             raise HTTPException(404, f"unknown refiner {req.refiner_checkpoint}")
-        return model, clip_skip
+        return model, clip_skip, ensd
 
     @app.post("/sdapi/v1/txt2img")
     def txt2img(req: Txt2ImgRequest):
         control_units = _parse_controlnet(req.alwayson_scripts)
-        model, clip_skip = _overrides(req)
+        model, clip_skip, ensd = _overrides(req)
         gen = GenerationRequest(
             prompt=req.prompt,
             negative_prompt=req.negative_prompt,
@@ -289,6 +292,9 @@ def create_app(engine: Optional[LocalEngine] = None,
             seed=req.seed,
             subseed=req.subseed,
             subseed_strength=req.subseed_strength,
+            seed_resize_from_w=req.seed_resize_from_w,
+            seed_resize_from_h=req.seed_resize_from_h,
+            eta_noise_seed_delta=ensd,
             tiling=req.tiling,
             s_churn=req.s_churn,
             s_tmin=req.s_tmin,
@@ -312,7 +318,7 @@ def create_app(engine: Optional[LocalEngine] = None,
     def img2img(req: Img2ImgRequest):
         if not req.init_images:
             raise HTTPException(422, "init_images required")
-        model, clip_skip = _overrides(req)
+        model, clip_skip, ensd = _overrides(req)
         try:
             inits = torch.stack(
                 [
@@ -347,6 +353,9 @@ def create_app(engine: Optional[LocalEngine] = None,
             seed=req.seed,
             subseed=req.subseed,
             subseed_strength=req.subseed_strength,
+            seed_resize_from_w=req.seed_resize_from_w,
+            seed_resize_from_h=req.seed_resize_from_h,
+            eta_noise_seed_delta=ensd,
             tiling=req.tiling,
             s_churn=req.s_churn,
             s_tmin=req.s_tmin,

@@ -50,6 +50,9 @@ class GenerationRequest:
     seed: int = -1
     subseed: int = -1
     subseed_strength: float = 0.0
+    seed_resize_from_w: int = 0
+    seed_resize_from_h: int = 0
+    eta_noise_seed_delta: int = 0
     init_images: Optional[torch.Tensor] = None  # [B,H,W,3] uint8 (img2img)
     denoising_strength: float = 0.75
     mask_image: Optional[torch.Tensor] = None   # [H,W] uint8, 255=repaint
@@ -133,6 +136,9 @@ def _job_pipeline_request(
         seeds=list(job.seeds),
         subseeds=list(job.subseeds),
         subseed_strength=gen.subseed_strength,
+        seed_resize_from_w=gen.seed_resize_from_w,
+        seed_resize_from_h=gen.seed_resize_from_h,
+        eta_noise_seed_delta=gen.eta_noise_seed_delta,
         init_latents=init_latents,
         denoising_strength=gen.denoising_strength,
         mask_image=gen.mask_image,

@@ -43,6 +43,13 @@ class PipelineRequest:
     seeds: List[int] = field(default_factory=lambda: [0])
     subseeds: List[int] = field(default_factory=list)
     subseed_strength: float = 0.0
+    # sdwui "Resize seed from": draw the initial noise at another
+    # resolution's latent grid and resize it, so a composition found at
+    # WxH roughly survives a resolution change
+    seed_resize_from_w: int = 0
+    seed_resize_from_h: int = 0
+    # sdwui eta_noise_seed_delta: offsets the ancestral-noise seeds
+    eta_noise_seed_delta: int = 0
     # img2img
     init_latents: Optional[torch.Tensor] = None  # pre-encoded [B,4,h,w]
     denoising_strength: float = 0.75
@@ -373,15 +380,26 @@ class StableDiffusionPipeline:
         sampler = build_sampler(req.sampler_name, sched)
         _apply_sampler_params(sampler, req)
 
+        noise_shape = (lat_c, lat_h, lat_w)
+        if req.seed_resize_from_w > 0 and req.seed_resize_from_h > 0:
+            noise_shape = (
+                lat_c, req.seed_resize_from_h // f, req.seed_resize_from_w // f
+            )
         noise = torch.stack(
             [
                 _image_noise(
                     req.seeds[i], subseeds[i], req.subseed_strength,
-                    (lat_c, lat_h, lat_w),
+                    noise_shape,
                 )
                 for i in range(b)
             ]
-        ).to(self.device, self.dtype)
+        )
+        if noise_shape != (lat_c, lat_h, lat_w):
+            noise = torch.nn.functional.interpolate(
+                noise, size=(lat_h, lat_w), mode="bilinear",
+                antialias=False,
+            )
+        noise = noise.to(self.device, self.dtype)
 
         sig = sched.sigmas
         if req.init_latents is not None:
@@ -420,8 +438,11 @@ class StableDiffusionPipeline:
 
         # ancestral noise: per-image generators stepped identically regardless
         # of shard composition
+        ensd = int(req.eta_noise_seed_delta)
         gens = [
-            torch.Generator("cpu").manual_seed((int(s) ^ 0x5EED) & 0xFFFFFFFF)
+            torch.Generator("cpu").manual_seed(
+                ((int(s) + ensd) ^ 0x5EED) & 0xFFFFFFFF
+            )
             for s in req.seeds
         ]
 

@@ -568,6 +568,45 @@ class TestPromptEditing:
         assert torch.equal(a, c)
 
 
+class TestSeedExtras:
+    def test_seed_resize_changes_noise_layout(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="s", steps=2, width=64, height=64, seeds=[9])
+        plain = pipe.generate(PipelineRequest(**base)).images
+        resized = pipe.generate(
+            PipelineRequest(**base, seed_resize_from_w=128,
+                            seed_resize_from_h=128)
+        ).images
+        assert plain.shape == resized.shape
+        assert not torch.equal(plain, resized)
+        again = pipe.generate(
+            PipelineRequest(**base, seed_resize_from_w=128,
+                            seed_resize_from_h=128)
+        ).images
+        assert torch.equal(resized, again)
+
+    def test_ensd_shifts_ancestral_noise_only(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        # Euler (non-ancestral): ENSD has no effect
+        base_e = dict(prompt="s", steps=3, width=64, height=64, seeds=[9],
+                      sampler_name="Euler")
+        a = pipe.generate(PipelineRequest(**base_e)).images
+        b = pipe.generate(
+            PipelineRequest(**base_e, eta_noise_seed_delta=31337)
+        ).images
+        assert torch.equal(a, b)
+        # Euler a (ancestral): ENSD changes the trajectory
+        base_a = dict(prompt="s", steps=3, width=64, height=64, seeds=[9],
+                      sampler_name="Euler a")
+        c = pipe.generate(PipelineRequest(**base_a)).images
+        d = pipe.generate(
+            PipelineRequest(**base_a, eta_noise_seed_delta=31337)
+        ).images
+        assert not torch.equal(c, d)
+
+
 class TestPreview:
     def test_preview_available_after_generation(self, pipe):
         from sdwd_amd.pipeline import PipelineRequest
