@@ -72,6 +72,8 @@ class Txt2ImgRequest(BaseModel):
     # sdwui per-request overrides (sd_model_checkpoint,
     # CLIP_stop_at_last_layers are honored; the rest are ignored)
     override_settings: Dict[str, Any] = Field(default_factory=dict)
+    send_images: bool = True   # omit base64 images from the response
+    save_images: bool = False  # persist PNGs server-side (SDWD_OUTPUT_DIR)
 
 
 class Img2ImgRequest(Txt2ImgRequest):
@@ -210,7 +212,8 @@ def create_app(engine: Optional[LocalEngine] = None,
                 )
             return await call_next(request)
 
-    def run_generation(gen: GenerationRequest) -> Dict[str, Any]:
+    def run_generation(gen: GenerationRequest, send_images: bool = True,
+                       save_images: bool = False) -> Dict[str, Any]:
         # one generation at a time (the reference serialized on the host's
         # queue_lock, world.py:244,273); concurrent requests queue here
         with state.lock:
@@ -221,14 +224,30 @@ def create_app(engine: Optional[LocalEngine] = None,
                 state.current_model = engine.model_name
             finally:
                 state.busy = False
-        images = [
-            base64.b64encode(
-                encode_png(result.images[i], result.infotexts[i])
-            ).decode()
-            for i in range(result.images.shape[0])
-        ]
-        if result.grid is not None:
-            images.insert(0, _b64_png(result.grid))
+        if save_images:
+            from ..utils.images import save_png
+
+            outdir = os.environ.get("SDWD_OUTPUT_DIR", "outputs")
+            os.makedirs(outdir, exist_ok=True)
+            stamp = time.strftime("%Y%m%d-%H%M%S")
+            for i in range(result.images.shape[0]):
+                save_png(
+                    result.images[i],
+                    os.path.join(
+                        outdir, f"{stamp}-{result.seeds[i]}-{i:03d}.png"
+                    ),
+                    parameters=result.infotexts[i],
+                )
+        images = []
+        if send_images:
+            images = [
+                base64.b64encode(
+                    encode_png(result.images[i], result.infotexts[i])
+                ).decode()
+                for i in range(result.images.shape[0])
+            ]
+            if result.grid is not None:
+                images.insert(0, _b64_png(result.grid))
         info = {
             "all_seeds": result.seeds,
             "all_subseeds": [-1] * len(result.seeds),
@@ -312,7 +331,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             refiner_model=req.refiner_checkpoint,
             refiner_switch_at=req.refiner_switch_at,
         )
-        return run_generation(gen)
+        return run_generation(gen, req.send_images, req.save_images)
 
     @app.post("/sdapi/v1/img2img")
     def img2img(req: Img2ImgRequest):
@@ -373,7 +392,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             inpainting_fill=req.inpainting_fill,
             color_correction=req.color_correction,
         )
-        return run_generation(gen)
+        return run_generation(gen, req.send_images, req.save_images)
 
     @app.post("/sdapi/v1/options")
     def set_options(req: OptionsRequest):

@@ -187,6 +187,32 @@ class TestControl:
         # far corner is pasted from the original init exactly
         assert torch.equal(out[:8, :8], init[:8, :8])
 
+    def test_save_and_send_images_flags(self, client, tmp_path,
+                                        monkeypatch_module):
+        import os
+
+        monkeypatch_module.setenv("SDWD_OUTPUT_DIR", str(tmp_path / "out"))
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={"prompt": "s", "steps": 1, "width": 64, "height": 64,
+                  "seed": 8, "send_images": False, "save_images": True},
+        )
+        assert r.status_code == 200
+        assert r.json()["images"] == []
+        saved = os.listdir(tmp_path / "out")
+        assert len(saved) == 1 and saved[0].endswith(".png")
+
+    def test_unknown_fields_ignored(self, client):
+        """sdwui clients send fields we don't implement (restore_faces,
+        script_args, ...); they must not 422."""
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={"prompt": "x", "steps": 1, "width": 64, "height": 64,
+                  "seed": 1, "restore_faces": False, "script_args": [],
+                  "script_name": None, "comments": {}},
+        )
+        assert r.status_code == 200
+
     def test_all_knobs_accepted(self, client):
         """Every sdwui-compat request field plumbs through end to end."""
         r = client.post(
