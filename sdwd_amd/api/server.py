@@ -667,6 +667,30 @@ def create_app(engine: Optional[LocalEngine] = None,
         engine.world.save()
         return engine.world.settings.model_dump()
 
+    @app.get("/sdwd/benchmark-payload")
+    def get_benchmark_payload():
+        """The canonical benchmark payload (ref shared.py:63-77 constants,
+        editable via config like the reference's Benchmark_Payload)."""
+        return engine.world.benchmark_payload.model_dump()
+
+    @app.post("/sdwd/benchmark-payload")
+    def set_benchmark_payload(body: Dict[str, Any]):
+        from ..config.models import BenchmarkPayload
+
+        current = engine.world.benchmark_payload.model_dump()
+        unknown = set(body) - set(current)
+        if unknown:
+            raise HTTPException(422, f"unknown fields: {sorted(unknown)}")
+        current.update(body)
+        try:
+            engine.world.benchmark_payload = BenchmarkPayload.model_validate(
+                current
+            )
+        except Exception as exc:
+            raise HTTPException(422, str(exc))
+        engine.world.save()
+        return engine.world.benchmark_payload.model_dump()
+
     @app.post("/sdwd/worker/{label}/enable")
     def enable_worker(label: str):
         w = engine.world.get_worker(label)

@@ -354,6 +354,21 @@ class TestControl:
         assert "filelora" in names
         refresh_lora_files(dirpath=str(tmp_path / "none"))
 
+    def test_benchmark_payload_round_trip(self, client):
+        before = client.get("/sdwd/benchmark-payload").json()
+        assert before["steps"] == 20
+        r = client.post(
+            "/sdwd/benchmark-payload", json={"steps": 8, "batch_size": 2}
+        )
+        assert r.status_code == 200
+        after = client.get("/sdwd/benchmark-payload").json()
+        assert after["steps"] == 8 and after["batch_size"] == 2
+        assert client.post(
+            "/sdwd/benchmark-payload", json={"warp": 1}
+        ).status_code == 422
+        client.post("/sdwd/benchmark-payload",
+                    json={"steps": 20, "batch_size": 1})
+
     def test_settings_rejects_unknown(self, client):
         r = client.post("/sdwd/settings", json={"warp_factor": 9})
         assert r.status_code == 422
