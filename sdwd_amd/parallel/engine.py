@@ -163,6 +163,26 @@ def _job_pipeline_request(
     )
 
 
+def _learn_pixel_cap(worker, gen: GenerationRequest, batch: int) -> bool:
+    """After an out-of-memory failure, cap the rank's future jobs below the
+    attempted size instead of sidelining a healthy GPU (an improvement on
+    the reference, whose pixel caps were manual-only, ui.py:313-319).
+    Returns True if a cap was applied."""
+    attempted = batch * gen.width * gen.height
+    if attempted <= 0:
+        return False
+    cap = int(attempted * 0.8)
+    if worker.pixel_cap and worker.pixel_cap <= cap:
+        cap = int(worker.pixel_cap * 0.8)  # cap again, smaller
+    if cap < gen.width * gen.height:
+        return False  # can't even fit one image; leave UNAVAILABLE
+    worker.pixel_cap = cap
+    log.warning(
+        "%s: OOM at %d px; learned pixel cap %d", worker.label, attempted, cap
+    )
+    return True
+
+
 def _blur_mask(mask: torch.Tensor, radius: int) -> torch.Tensor:
     """Separable gaussian blur on a [H,W] uint8 mask (sdwui mask_blur:
     softens the inpaint boundary in both the latent mask and the paste)."""
@@ -392,7 +412,8 @@ class LocalEngine(_EngineBase):
             self.pipes[f"gpu{i}"] = StableDiffusionPipeline(
                 model, device=dev, dtype=dtype
             )
-        self._fail_injection: Dict[str, bool] = {}
+        self._fail_injection: Dict[str, object] = {}
+        self._oom_capped: set = set()
         if torch.cuda.is_available():
             # startup liveness sweep (ref distributed.py:48-52 pinged all
             # remotes at init); CPU ranks have no memory probe to ping
@@ -432,9 +453,10 @@ class LocalEngine(_EngineBase):
     def benchmark(self, rebenchmark: bool = False) -> Dict[str, float]:
         return self.world.benchmark(self._bench_runner, rebenchmark=rebenchmark)
 
-    def inject_failure(self, label: str) -> None:
-        """Test hook: make the next shard on this rank raise."""
-        self._fail_injection[label] = True
+    def inject_failure(self, label: str, oom: bool = False) -> None:
+        """Test hook: make the next shard on this rank raise (optionally
+        as an out-of-memory error to exercise pixel-cap learning)."""
+        self._fail_injection[label] = "oom" if oom else True
 
     def _run_job(
         self,
@@ -456,7 +478,10 @@ class LocalEngine(_EngineBase):
             )
             self.pipes[job.worker_label] = pipe
         try:
-            if self._fail_injection.pop(job.worker_label, False):
+            kind = self._fail_injection.pop(job.worker_label, False)
+            if kind == "oom":
+                raise torch.cuda.OutOfMemoryError("injected OOM")
+            if kind:
                 raise RuntimeError("injected failure")
             worker.set_state(State.WORKING)
             init_latents = None
@@ -489,6 +514,12 @@ class LocalEngine(_EngineBase):
             log.warning("rank %s failed: %s", job.worker_label, exc)
             errors[job.worker_label] = exc
             worker.set_state(State.UNAVAILABLE)
+            if isinstance(exc, torch.cuda.OutOfMemoryError) and (
+                _learn_pixel_cap(worker, gen, job.batch_size)
+            ):
+                # healthy GPU, job just too big: re-admit it AFTER this
+                # generation's requeue (which excludes UNAVAILABLE ranks)
+                self._oom_capped.add(job.worker_label)
 
     def progress(self) -> float:
         """0..1 across the jobs of the generation in flight (ref /progress)."""
@@ -546,6 +577,13 @@ class LocalEngine(_EngineBase):
                 t.start()
             for t in rts:
                 t.join()
+
+        # OOM-capped ranks come back for the next plan (now capped)
+        for label in self._oom_capped:
+            w = self.world.get_worker(label)
+            if w is not None and w.state is State.UNAVAILABLE:
+                w.set_state(State.IDLE, strict=False)
+        self._oom_capped.clear()
 
         interrupted = self.world.interrupted.is_set()
         result = self._assemble(

@@ -303,6 +303,30 @@ class TestEdgeCases:
         )
         assert eng.pipes["gpu0"] is p
 
+    def test_oom_learns_pixel_cap(self):
+        """An OOM shard caps the rank below the attempted size and keeps it
+        schedulable; the gallery still completes via requeue."""
+        eng = make_engine(2)
+        eng.inject_failure("gpu1", oom=True)
+        res = eng.generate(
+            GenerationRequest(prompt="o", batch_size=4, width=64, height=64,
+                              steps=1, seed=21)
+        )
+        assert res.images.shape == (4, 64, 64, 3)
+        w1 = eng.world.get_worker("gpu1")
+        assert 0 < w1.pixel_cap < 2 * 64 * 64  # below the 2-image attempt
+        assert w1.state is not State.UNAVAILABLE
+        # next run plans within the cap (no failure injected now)
+        res2 = eng.generate(
+            GenerationRequest(prompt="o", batch_size=4, width=64, height=64,
+                              steps=1, seed=22)
+        )
+        assert res2.images.shape == (4, 64, 64, 3)
+        for line in res2.job_summary:
+            if line.startswith("gpu1:") and "(complementary)" not in line:
+                n = int(line.split(":")[1].strip().split(" ")[0])
+                assert n <= w1.pixel_cap // (64 * 64)
+
     def test_img2img_init_cycling(self):
         """Fewer init images than batch: inits cycle by gallery index, so
         sharding can't change which init image k gets."""
