@@ -805,7 +805,7 @@ class DistributedEngine(_EngineBase):
         hf = gen.hr_scale if gen.enable_hr else 1.0
         h, w = int(gen.height * hf), int(gen.width * hf)
         my_images = torch.zeros(0, h, w, 3, dtype=torch.uint8)
-        my_ok, my_elapsed = 1.0, 0.0
+        my_ok, my_elapsed, my_cap = 1.0, 0.0, 0.0
         my_infos: List[str] = []
         interrupted = False
         if mine is not None and mine.batch_size > 0:
@@ -851,6 +851,10 @@ class DistributedEngine(_EngineBase):
             except Exception as exc:  # noqa: BLE001
                 log.warning("rank %d shard failed: %s", self.rank, exc)
                 my_ok = 0.0
+                if isinstance(exc, torch.cuda.OutOfMemoryError):
+                    me = self.world.get_worker(self.label)
+                    if _learn_pixel_cap(me, gen, mine.batch_size):
+                        my_cap = float(me.pixel_cap)
 
         # status + images to everyone (rank 0 consumes)
         sizes = [
@@ -864,7 +868,13 @@ class DistributedEngine(_EngineBase):
             )
             for r in range(self.world_size)
         ]
-        status = pg.allgather_floats([my_ok, my_elapsed], self.device)
+        status = pg.allgather_floats(
+            [my_ok, my_elapsed, my_cap], self.device
+        )
+        # every rank applies learned pixel caps so future plans agree
+        for r in range(self.world_size):
+            if len(status[r]) > 2 and status[r][2] > 0:
+                self.world.get_worker(f"gpu{r}").pixel_cap = int(status[r][2])
         shard_dev = my_images.to(self.device)
         gathered = pg.gather_images(shard_dev, sizes, self.device)
         all_images = gathered.cpu() if gathered is not None else my_images
