@@ -667,6 +667,15 @@ def create_app(engine: Optional[LocalEngine] = None,
         engine.world.save()
         return engine.world.settings.model_dump()
 
+    @app.post("/sdwd/release-lock")
+    def release_lock():
+        """Debug escape hatch (ref ui.py:69-70 force-released the host's
+        queue lock): clears a stuck busy flag and replaces the lock."""
+        state.busy = False
+        state.lock = threading.Lock()
+        log.warning("generation lock force-released")
+        return {}
+
     @app.get("/sdwd/benchmark-payload")
     def get_benchmark_payload():
         """The canonical benchmark payload (ref shared.py:63-77 constants,

@@ -354,6 +354,16 @@ class TestControl:
         assert "filelora" in names
         refresh_lora_files(dirpath=str(tmp_path / "none"))
 
+    def test_release_lock(self, client):
+        assert client.post("/sdwd/release-lock").status_code == 200
+        # engine still serves afterwards
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={"prompt": "l", "steps": 1, "width": 64, "height": 64,
+                  "seed": 2},
+        )
+        assert r.status_code == 200
+
     def test_benchmark_payload_round_trip(self, client):
         before = client.get("/sdwd/benchmark-payload").json()
         assert before["steps"] == 20
