@@ -43,6 +43,7 @@ SAMPLER_EVALS_PER_STEP = {
     "DPM++ 2S a": 2.0,
     "DPM++ 2S a Karras": 2.0,
     "UniPC": 1.0,  # corrector eval is reused as the next predictor eval
+    "Restart": 2.6,  # Heun cost + ~30% restart overhead at 20 steps
 }
 
 

@@ -312,7 +312,8 @@ class TestSamplerOracle:
                                       "DPM++ 2M", "DPM++ SDE", "LMS",
                                       "DPM2", "DPM2 a", "DDPM",
                                       "DPM++ 2S a", "UniPC",
-                                      "DPM++ 2M SDE", "DPM++ 3M SDE"])
+                                      "DPM++ 2M SDE", "DPM++ 3M SDE",
+                                      "Restart"])
     def test_converges_to_point_mass(self, name):
         from sdwd_amd.pipeline.samplers import build_sampler
         from sdwd_amd.pipeline.schedule import discrete_schedule, make_sigmas_full
