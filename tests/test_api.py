@@ -233,6 +233,13 @@ class TestControl:
         assert r.status_code == 200
         body = r.json()
         assert len(body["images"]) >= 1
+        # unmerge the <lora:...> tag so the shared engine weights are
+        # pristine for whatever test runs next
+        client.post(
+            "/sdapi/v1/txt2img",
+            json={"prompt": "plain", "steps": 1, "width": 64, "height": 64,
+                  "seed": 1},
+        )
 
     def test_basic_auth(self, monkeypatch):
         import base64 as b64

@@ -16,7 +16,13 @@ from sdwd_amd.pipeline import (
 
 @pytest.fixture(scope="module")
 def pipe():
-    return StableDiffusionPipeline("tiny", device="cpu")
+    # an uncached bundle: other test modules share the registry cache and
+    # could otherwise mutate this fixture's weights (LoRA merges, tiling
+    # flags) from another xdist-scheduled test
+    return StableDiffusionPipeline(
+        load_model("tiny", device="cpu", dtype=torch.float32, cache=False),
+        device="cpu", dtype=torch.float32,
+    )
 
 
 class TestSchedule:
