@@ -1,11 +1,13 @@
 """Samplers (k-diffusion style, sigma space).
 
 The reference's users pick these by name (its sampler-speed table
-worker.py:75-94 lists what people run); implemented natively here: Euler,
-Euler a, DDIM (= Euler on the discrete schedule), Heun, DPM++ 2M
-(+ Karras), DPM++ SDE. The per-step state update is a fused elementwise
-HIP kernel on GPU (ops.euler_step / ops.add_noise) so the denoise loop is
-hipGraph-capturable.
+worker.py:75-94 lists what people run); implemented natively: Euler,
+Euler a, DDIM (= Euler on the discrete schedule), DDPM, Heun, LMS,
+DPM2/DPM2 a, DPM++ 2M / SDE / 2S a / 2M SDE / 3M SDE, UniPC, LCM and
+Restart, each verified against a point-mass oracle (the update algebra
+must land exactly on x0 with an exact denoiser). Per-step state updates
+are fused elementwise HIP kernels on GPU (ops.euler_step /
+ops.add_noise / ops.lincomb) so the denoise loop is hipGraph-capturable.
 
 model_fn(x_scaled, t) -> eps; the driver loop in pipeline.py owns CFG,
 interrupts and callbacks.
@@ -489,8 +491,7 @@ class Restart(Sampler):
 
         x, stop = descend(x, 0, bi)
         if not stop and restarts and noise_fn is not None:
-            s_hi, s_lo = sig[a], sig[max(a, bi - 1) if bi <= n else a]
-            s_lo = sig[bi] if bi < len(sig) else 0.0
+            s_hi, s_lo = sig[a], sig[bi]
             for _ in range(restarts):
                 bump = math.sqrt(max(0.0, s_hi * s_hi - s_lo * s_lo))
                 x = ops.add_noise(x, noise_fn(), 1.0, bump)
