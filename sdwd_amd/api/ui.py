@@ -35,6 +35,7 @@ PAGE = """<!DOCTYPE html>
    interrupt</button>
 </form>
 <div id="gallery"></div>
+<div>live: <img id="preview" style="max-width:128px;display:none"/></div>
 <h2>utils</h2>
 <button onclick="fetch('/sdwd/benchmark',{method:'POST'})">re-benchmark</button>
 <button onclick="fetch('/sdwd/sync-script',{method:'POST'})">run sync script</button>
@@ -63,6 +64,16 @@ async function refresh(){
     }
     h += `</table><p>model: ${s.model} — busy: ${s.busy}</p>`;
     document.getElementById('status').innerHTML = h;
+    if (s.busy) {
+      const pr = await (await fetch('/sdapi/v1/progress')).json();
+      const img = document.getElementById('preview');
+      if (pr.current_image) {
+        img.src = 'data:image/png;base64,' + pr.current_image;
+        img.style.display = 'inline';
+      }
+    } else {
+      document.getElementById('preview').style.display = 'none';
+    }
     document.getElementById('log').textContent = s.log.slice(-16).join('\\n');
   }catch(e){}
 }
