@@ -667,6 +667,15 @@ def create_app(engine: Optional[LocalEngine] = None,
         engine.world.save()
         return engine.world.settings.model_dump()
 
+    @app.post("/sdwd/reset-mpe")
+    def reset_mpe():
+        """Clear every rank's ETA error-correction history (ref 2.0.0
+        debug utility)."""
+        for w in engine.world.workers:
+            w.eta.reset_errors()
+        engine.world.save()
+        return {}
+
     @app.post("/sdwd/release-lock")
     def release_lock():
         """Debug escape hatch (ref ui.py:69-70 force-released the host's

@@ -108,6 +108,11 @@ class EtaPredictor:
             return
         self.percent_errors.append(pct)
 
+    def reset_errors(self) -> None:
+        """Drop the error-correction history (ref 2.0.0 'debug option for
+        resetting error correction at runtime')."""
+        self.percent_errors.clear()
+
     def mpe(self) -> float:
         if not self.percent_errors:
             return 0.0

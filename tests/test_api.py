@@ -361,6 +361,13 @@ class TestControl:
         assert "filelora" in names
         refresh_lora_files(dirpath=str(tmp_path / "none"))
 
+    def test_reset_mpe(self, client):
+        eng = client.app.state.engine
+        eng.world.get_worker("gpu0").eta.record_outcome(1.0, 2.0)
+        assert eng.world.get_worker("gpu0").eta.mpe() != 0.0
+        assert client.post("/sdwd/reset-mpe").status_code == 200
+        assert eng.world.get_worker("gpu0").eta.mpe() == 0.0
+
     def test_release_lock(self, client):
         assert client.post("/sdwd/release-lock").status_code == 200
         # engine still serves afterwards
