@@ -3,7 +3,7 @@ remote webui's CLIP via POST /txt2img — SURVEY.md §2.5).
 
 A ViT-L/14-text-shaped transformer: vocab 49408, 77 positions, causal mask,
 quick-GELU MLPs, final LayerNorm. SDXL adds a second, bigger encoder
-(penultimate-layer output + pooled embedding) — see sdxl.py.
+(penultimate-layer output + pooled embedding) — see registry._build_sdxl.
 
 Weights are random-init (no network in this environment); the architecture
 and shapes match SD1.5 so the benchmark works the same compute.
