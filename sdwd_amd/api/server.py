@@ -486,7 +486,22 @@ def create_app(engine: Optional[LocalEngine] = None,
 
     @app.get("/sdapi/v1/embeddings")
     def embeddings():
-        return {"loaded": {}, "skipped": {}}
+        from ..models.embeddings import loaded
+
+        return {
+            "loaded": {
+                n: {"step": None, "sd_checkpoint": None,
+                    "sd_checkpoint_name": None, "shape": None, "vectors": k}
+                for n, k in loaded().items()
+            },
+            "skipped": {},
+        }
+
+    @app.post("/sdapi/v1/refresh-embeddings")
+    def refresh_embeddings():
+        from ..models.embeddings import refresh_embedding_files
+
+        return {"found": refresh_embedding_files()}
 
     @app.get("/sdapi/v1/hypernetworks")
     def hypernetworks():

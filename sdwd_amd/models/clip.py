@@ -102,7 +102,12 @@ class CLIPTextEncoder(nn.Module):
         ``clip_skip``: sdwui CLIP_stop_at_last_layers — skip the last
         ``clip_skip - 1`` blocks but DO apply the final LayerNorm afterwards
         (sd_hijack_clip semantics)."""
-        x = self.token_emb(tokens) + self.pos_emb
+        safe = tokens.clamp_max(self.token_emb.num_embeddings - 1)
+        x = self.token_emb(safe) + self.pos_emb
+        # textual-inversion placeholders (ids >= vocab) -> trained vectors
+        from .embeddings import apply_to_hidden
+
+        x = apply_to_hidden(tokens, x)
         bias = self.causal_bias.to(x.dtype)
         skip = 1 if penultimate else max(0, clip_skip - 1)
         n = len(self.blocks) - skip
@@ -114,7 +119,11 @@ class CLIPTextEncoder(nn.Module):
 
     def pooled(self, tokens: torch.Tensor, hidden: torch.Tensor) -> torch.Tensor:
         """EOT-token pooled embedding (SDXL conditioning)."""
-        eot = tokens.argmax(dim=-1)  # highest id = end-of-text
+        from .embeddings import PLACEHOLDER_BASE
+
+        tok = tokens.clone()
+        tok[tok >= PLACEHOLDER_BASE] = 0  # TI placeholders are not EOT
+        eot = tok.argmax(dim=-1)  # highest id = end-of-text
         h = hidden[torch.arange(hidden.shape[0]), eot]
         if self.text_proj is not None:
             h = h @ self.text_proj.to(h.dtype)

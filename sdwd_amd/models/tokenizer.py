@@ -136,9 +136,22 @@ def use_hash() -> None:
 
 
 def _fragment_ids(text: str) -> List[int]:
+    from .embeddings import trigger_ids
+
+    out: List[int] = []
     if _active is not None:
-        return _active.encode_text(text)
-    return [_hash_token(w) for w in _word_re.findall(text.lower())]
+        # BPE path: check whole words for textual-inversion triggers first
+        for word in text.lower().split():
+            tids = trigger_ids(word)
+            if tids is not None:
+                out.extend(tids)
+            else:
+                out.extend(_active.encode_text(word))
+        return out
+    for w in _word_re.findall(text.lower()):
+        tids = trigger_ids(w)
+        out.extend(tids if tids is not None else [_hash_token(w)])
+    return out
 
 
 def encode(text: str, max_len: int = MAX_LEN) -> List[int]:

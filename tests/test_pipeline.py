@@ -1239,3 +1239,65 @@ class TestPromptWeighting:
         a = pipe.generate(PipelineRequest(**base, clip_skip=1)).images
         b = pipe.generate(PipelineRequest(**base, clip_skip=2)).images
         assert not torch.equal(a, b)
+
+
+class TestTextualInversion:
+    def test_trigger_changes_output_deterministically(self, pipe):
+        from sdwd_amd.models import embeddings
+        from sdwd_amd.pipeline import PipelineRequest
+
+        g = torch.Generator().manual_seed(7)
+        vec = torch.randn(2, 64, generator=g)  # tiny clip d_model = 64
+        base = dict(steps=2, width=64, height=64, seeds=[5])
+        plain = pipe.generate(
+            PipelineRequest(prompt="a mystyle cow", **base)
+        ).images
+        embeddings.register("mystyle", vec)
+        try:
+            ti = pipe.generate(
+                PipelineRequest(prompt="a mystyle cow", **base)
+            ).images
+            ti2 = pipe.generate(
+                PipelineRequest(prompt="a mystyle cow", **base)
+            ).images
+            assert not torch.equal(plain, ti)
+            assert torch.equal(ti, ti2)
+        finally:
+            embeddings.clear()
+        back = pipe.generate(
+            PipelineRequest(prompt="a mystyle cow", **base)
+        ).images
+        assert torch.equal(plain, back)
+
+    def test_file_loading(self, tmp_path):
+        from safetensors.torch import save_file
+
+        from sdwd_amd.models import embeddings
+
+        save_file({"emb_params": torch.randn(3, 64)},
+                  str(tmp_path / "trigword.safetensors"))
+        try:
+            names = embeddings.refresh_embedding_files(str(tmp_path))
+            assert names == ["trigword"]
+            assert embeddings.loaded() == {"trigword": 3}
+            from sdwd_amd.models import tokenizer
+
+            ids = tokenizer.encode("a trigword")
+            assert sum(1 for i in ids
+                       if i >= embeddings.PLACEHOLDER_BASE) == 3
+        finally:
+            embeddings.clear()
+
+    def test_mismatched_width_skipped(self, pipe):
+        from sdwd_amd.models import embeddings
+        from sdwd_amd.pipeline import PipelineRequest
+
+        embeddings.register("wrongwidth", torch.randn(1, 999))
+        try:
+            res = pipe.generate(
+                PipelineRequest(prompt="a wrongwidth cow", steps=1,
+                                width=64, height=64, seeds=[5])
+            )
+            assert torch.isfinite(res.images.float()).all()
+        finally:
+            embeddings.clear()
