@@ -61,6 +61,8 @@ class Txt2ImgRequest(BaseModel):
     hr_scale: float = 2.0
     hr_upscaler: str = "Latent"
     hr_second_pass_steps: int = 0
+    hr_prompt: str = ""
+    hr_negative_prompt: str = ""
     denoising_strength: float = 0.75
     # alwayson scripts (ref C17/C18: the reference forwarded these; we
     # execute the controlnet unit natively, other scripts are ignored
@@ -324,6 +326,8 @@ def create_app(engine: Optional[LocalEngine] = None,
             hr_scale=req.hr_scale,
             hr_steps=req.hr_second_pass_steps,
             hr_upscaler=req.hr_upscaler,
+            hr_prompt=req.hr_prompt,
+            hr_negative_prompt=req.hr_negative_prompt,
             denoising_strength=req.denoising_strength,
             clip_skip=clip_skip,
             control_units=control_units,

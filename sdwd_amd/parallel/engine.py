@@ -60,6 +60,8 @@ class GenerationRequest:
     hr_scale: float = 2.0
     hr_steps: int = 0
     hr_upscaler: str = "nearest"
+    hr_prompt: str = ""
+    hr_negative_prompt: str = ""
     control_image: Optional[torch.Tensor] = None
     control_model: str = ""
     control_scale: float = 1.0
@@ -147,6 +149,8 @@ def _job_pipeline_request(
         hr_scale=gen.hr_scale,
         hr_steps=gen.hr_steps,
         hr_upscaler=gen.hr_upscaler,
+        hr_prompt=gen.hr_prompt,
+        hr_negative_prompt=gen.hr_negative_prompt,
         control_image=gen.control_image,
         control_model=gen.control_model,
         control_scale=gen.control_scale,

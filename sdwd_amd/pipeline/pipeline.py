@@ -65,6 +65,10 @@ class PipelineRequest:
     # latent upscaler (sdwui hr_upscaler "Latent..." family):
     # nearest | bilinear | bicubic | bilinear-antialiased | bicubic-antialiased
     hr_upscaler: str = "nearest"
+    # optional different conditioning for the hires second pass
+    # (sdwui hr_prompt / hr_negative_prompt; "" = reuse the base prompt)
+    hr_prompt: str = ""
+    hr_negative_prompt: str = ""
     # two-model refiner (sdwui refiner_checkpoint/refiner_switch_at): the
     # base model denoises the first switch_at fraction of steps, the
     # refiner model finishes (both share the latent space / VAE)
@@ -773,6 +777,16 @@ class StableDiffusionPipeline:
         # (ref eta_hr, worker.py:205-228 predicts exactly this shape).
         if req.enable_hr and req.hr_scale > 1.0 and not was_interrupted:
             hr_steps = req.hr_steps or req.steps
+            if req.hr_prompt or req.hr_negative_prompt:
+                # the hires pass denoises under its own conditioning;
+                # model_fn reads ctx/y/and_ws through the closure
+                from ..models.lora import parse_prompt_loras
+
+                hp = req.hr_prompt or req.prompt
+                hp, _ = parse_prompt_loras(hp)  # tags stripped, set unchanged
+                hn = req.hr_negative_prompt or req.negative_prompt
+                ctx, y, and_ws = self._build_ctx(hp, hn, req, b)
+                seg_tensors.clear()
             x = _upscale_latent(
                 x.float(), req.hr_scale, req.hr_upscaler
             ).to(self.dtype)

@@ -1301,3 +1301,33 @@ class TestTextualInversion:
             assert torch.isfinite(res.images.float()).all()
         finally:
             embeddings.clear()
+
+
+class TestHrPrompt:
+    def test_hr_prompt_changes_second_pass(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(steps=3, width=64, height=64, seeds=[9],
+                    enable_hr=True, hr_scale=2.0, hr_steps=2,
+                    denoising_strength=0.6, prompt="a cat")
+        plain = pipe.generate(PipelineRequest(**base)).images
+        changed = pipe.generate(
+            PipelineRequest(**base, hr_prompt="a dog")
+        ).images
+        same = pipe.generate(
+            PipelineRequest(**base, hr_prompt="a cat")
+        ).images
+        assert not torch.equal(plain, changed)
+        assert torch.equal(plain, same)
+
+    def test_hr_negative_only(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(steps=2, width=64, height=64, seeds=[9],
+                    enable_hr=True, hr_scale=2.0, hr_steps=1,
+                    denoising_strength=0.6, prompt="a cat")
+        a = pipe.generate(PipelineRequest(**base)).images
+        b = pipe.generate(
+            PipelineRequest(**base, hr_negative_prompt="blurry")
+        ).images
+        assert not torch.equal(a, b)
