@@ -63,6 +63,8 @@ class Txt2ImgRequest(BaseModel):
     hr_second_pass_steps: int = 0
     hr_prompt: str = ""
     hr_negative_prompt: str = ""
+    hr_resize_x: int = Field(default=0, ge=0)
+    hr_resize_y: int = Field(default=0, ge=0)
     denoising_strength: float = 0.75
     # alwayson scripts (ref C17/C18: the reference forwarded these; we
     # execute the controlnet unit natively, other scripts are ignored
@@ -328,6 +330,8 @@ def create_app(engine: Optional[LocalEngine] = None,
             hr_upscaler=req.hr_upscaler,
             hr_prompt=req.hr_prompt,
             hr_negative_prompt=req.hr_negative_prompt,
+            hr_resize_x=req.hr_resize_x,
+            hr_resize_y=req.hr_resize_y,
             denoising_strength=req.denoising_strength,
             clip_skip=clip_skip,
             control_units=control_units,

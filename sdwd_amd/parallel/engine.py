@@ -62,6 +62,8 @@ class GenerationRequest:
     hr_upscaler: str = "nearest"
     hr_prompt: str = ""
     hr_negative_prompt: str = ""
+    hr_resize_x: int = 0
+    hr_resize_y: int = 0
     control_image: Optional[torch.Tensor] = None
     control_model: str = ""
     control_scale: float = 1.0
@@ -151,6 +153,8 @@ def _job_pipeline_request(
         hr_upscaler=gen.hr_upscaler,
         hr_prompt=gen.hr_prompt,
         hr_negative_prompt=gen.hr_negative_prompt,
+        hr_resize_x=gen.hr_resize_x,
+        hr_resize_y=gen.hr_resize_y,
         control_image=gen.control_image,
         control_model=gen.control_model,
         control_scale=gen.control_scale,

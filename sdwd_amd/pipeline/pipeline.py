@@ -69,6 +69,9 @@ class PipelineRequest:
     # (sdwui hr_prompt / hr_negative_prompt; "" = reuse the base prompt)
     hr_prompt: str = ""
     hr_negative_prompt: str = ""
+    # explicit hires target size (sdwui hr_resize_x/y; 0 = use hr_scale)
+    hr_resize_x: int = 0
+    hr_resize_y: int = 0
     # two-model refiner (sdwui refiner_checkpoint/refiner_switch_at): the
     # base model denoises the first switch_at fraction of steps, the
     # refiner model finishes (both share the latent space / VAE)
@@ -135,11 +138,17 @@ _HR_MODES = {
 }
 
 
-def _upscale_latent(x: torch.Tensor, scale: float, upscaler: str) -> torch.Tensor:
+def _upscale_latent(x: torch.Tensor, scale: float, upscaler: str,
+                    size=None) -> torch.Tensor:
     """Latent-space upscale for the hires-fix first->second pass handoff
-    (sdwui's "Latent ..." hr_upscaler family; ref CHANGELOG hires support)."""
+    (sdwui's "Latent ..." hr_upscaler family; ref CHANGELOG hires support).
+    ``size`` (lat_h, lat_w) overrides ``scale`` (sdwui hr_resize_x/y)."""
     mode, aa = _HR_MODES.get((upscaler or "nearest").lower(), ("nearest", False))
     kwargs = {"antialias": True} if aa else {}
+    if size is not None:
+        return torch.nn.functional.interpolate(
+            x, size=size, mode=mode, **kwargs
+        )
     return torch.nn.functional.interpolate(
         x, scale_factor=scale, mode=mode, **kwargs
     )
@@ -787,8 +796,11 @@ class StableDiffusionPipeline:
                 hn = req.hr_negative_prompt or req.negative_prompt
                 ctx, y, and_ws = self._build_ctx(hp, hn, req, b)
                 seg_tensors.clear()
+            hr_size = None
+            if req.hr_resize_x > 0 and req.hr_resize_y > 0:
+                hr_size = (req.hr_resize_y // f, req.hr_resize_x // f)
             x = _upscale_latent(
-                x.float(), req.hr_scale, req.hr_upscaler
+                x.float(), req.hr_scale, req.hr_upscaler, size=hr_size
             ).to(self.dtype)
             hsched = schedule_for(req.sampler_name, hr_steps, req.scheduler)
             start = max(

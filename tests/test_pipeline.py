@@ -1331,3 +1331,16 @@ class TestHrPrompt:
             PipelineRequest(**base, hr_negative_prompt="blurry")
         ).images
         assert not torch.equal(a, b)
+
+
+class TestHrResize:
+    def test_explicit_target_size(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        res = pipe.generate(
+            PipelineRequest(prompt="r", steps=2, width=64, height=64,
+                            seeds=[3], enable_hr=True, hr_scale=2.0,
+                            hr_steps=1, denoising_strength=0.5,
+                            hr_resize_x=96, hr_resize_y=128)
+        )
+        assert res.images.shape == (1, 128, 96, 3)
