@@ -65,6 +65,7 @@ class Txt2ImgRequest(BaseModel):
     hr_negative_prompt: str = ""
     hr_resize_x: int = Field(default=0, ge=0)
     hr_resize_y: int = Field(default=0, ge=0)
+    hr_sampler_name: str = ""
     denoising_strength: float = 0.75
     # alwayson scripts (ref C17/C18: the reference forwarded these; we
     # execute the controlnet unit natively, other scripts are ignored
@@ -332,6 +333,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             hr_negative_prompt=req.hr_negative_prompt,
             hr_resize_x=req.hr_resize_x,
             hr_resize_y=req.hr_resize_y,
+            hr_sampler_name=req.hr_sampler_name,
             denoising_strength=req.denoising_strength,
             clip_skip=clip_skip,
             control_units=control_units,

@@ -64,6 +64,7 @@ class GenerationRequest:
     hr_negative_prompt: str = ""
     hr_resize_x: int = 0
     hr_resize_y: int = 0
+    hr_sampler_name: str = ""
     control_image: Optional[torch.Tensor] = None
     control_model: str = ""
     control_scale: float = 1.0
@@ -155,6 +156,7 @@ def _job_pipeline_request(
         hr_negative_prompt=gen.hr_negative_prompt,
         hr_resize_x=gen.hr_resize_x,
         hr_resize_y=gen.hr_resize_y,
+        hr_sampler_name=gen.hr_sampler_name,
         control_image=gen.control_image,
         control_model=gen.control_model,
         control_scale=gen.control_scale,

@@ -72,6 +72,7 @@ class PipelineRequest:
     # explicit hires target size (sdwui hr_resize_x/y; 0 = use hr_scale)
     hr_resize_x: int = 0
     hr_resize_y: int = 0
+    hr_sampler_name: str = ""  # "" = same sampler as the first pass
     # two-model refiner (sdwui refiner_checkpoint/refiner_switch_at): the
     # base model denoises the first switch_at fraction of steps, the
     # refiner model finishes (both share the latent space / VAE)
@@ -802,14 +803,15 @@ class StableDiffusionPipeline:
             x = _upscale_latent(
                 x.float(), req.hr_scale, req.hr_upscaler, size=hr_size
             ).to(self.dtype)
-            hsched = schedule_for(req.sampler_name, hr_steps, req.scheduler)
+            hr_sampler = req.hr_sampler_name or req.sampler_name
+            hsched = schedule_for(hr_sampler, hr_steps, req.scheduler)
             start = max(
                 0, hr_steps - max(1, int(hr_steps * req.denoising_strength))
             )
             hsched = type(hsched)(
                 sigmas=hsched.sigmas[start:], timesteps=hsched.timesteps[start:]
             )
-            hsampler = build_sampler(req.sampler_name, hsched)
+            hsampler = build_sampler(hr_sampler, hsched)
             _apply_sampler_params(hsampler, req)
             hh, hw = x.shape[2], x.shape[3]
             hr_noise = torch.stack(

@@ -1344,3 +1344,21 @@ class TestHrResize:
                             hr_resize_x=96, hr_resize_y=128)
         )
         assert res.images.shape == (1, 128, 96, 3)
+
+
+class TestHrSampler:
+    def test_different_hires_sampler(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="h", steps=3, width=64, height=64, seeds=[6],
+                    enable_hr=True, hr_scale=2.0, hr_steps=4,
+                    denoising_strength=0.9)
+        a = pipe.generate(PipelineRequest(**base)).images
+        b = pipe.generate(
+            PipelineRequest(**base, hr_sampler_name="Heun")
+        ).images
+        assert not torch.equal(a, b)
+        same = pipe.generate(
+            PipelineRequest(**base, hr_sampler_name="Euler a")
+        ).images
+        assert torch.equal(a, same)
