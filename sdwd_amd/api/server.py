@@ -109,6 +109,9 @@ class ServerState:
         self.progress = 0.0
         self.started_at = 0.0
         self.lock = threading.Lock()
+        # global defaults settable via POST /sdapi/v1/options
+        self.default_clip_skip = 1
+        self.ensd = 0
 
 
 def _b64_png(img: torch.Tensor) -> str:
@@ -283,8 +286,13 @@ def create_app(engine: Optional[LocalEngine] = None,
         model = str(ov.get("sd_model_checkpoint") or "")
         if model and model not in available_models():
             raise HTTPException(404, f"unknown model {model}")
-        clip_skip = int(ov.get("CLIP_stop_at_last_layers") or req.clip_skip)
-        ensd = int(ov.get("eta_noise_seed_delta") or 0)
+        # per-request override > request field > global option default
+        clip_skip = int(
+            ov.get("CLIP_stop_at_last_layers")
+            or (req.clip_skip if req.clip_skip > 1 else 0)
+            or state.default_clip_skip
+        )
+        ensd = int(ov.get("eta_noise_seed_delta") or state.ensd)
         if req.styles:
             from ..pipeline.styles import all_styles, apply_styles, refresh_styles
 
@@ -406,6 +414,13 @@ def create_app(engine: Optional[LocalEngine] = None,
 
     @app.post("/sdapi/v1/options")
     def set_options(req: OptionsRequest):
+        extras = req.model_extra or {}
+        if "CLIP_stop_at_last_layers" in extras:
+            state.default_clip_skip = max(
+                1, int(extras["CLIP_stop_at_last_layers"] or 1)
+            )
+        if "eta_noise_seed_delta" in extras:
+            state.ensd = int(extras["eta_noise_seed_delta"] or 0)
         if req.sd_model_checkpoint and req.sd_model_checkpoint != state.current_model:
             name = req.sd_model_checkpoint
             if name not in available_models():
@@ -416,7 +431,12 @@ def create_app(engine: Optional[LocalEngine] = None,
 
     @app.get("/sdapi/v1/options")
     def get_options():
-        return {"sd_model_checkpoint": state.current_model, "sd_vae": "auto"}
+        return {
+            "sd_model_checkpoint": state.current_model,
+            "sd_vae": "auto",
+            "CLIP_stop_at_last_layers": state.default_clip_skip,
+            "eta_noise_seed_delta": state.ensd,
+        }
 
     @app.get("/sdapi/v1/sd-models")
     def sd_models():

@@ -481,3 +481,28 @@ class TestBadInput:
                   "steps": 1, "width": 64, "height": 64},
         )
         assert r.status_code == 422
+
+
+class TestGlobalOptions:
+    def test_clip_skip_and_ensd_via_options(self, client):
+        r = client.post(
+            "/sdapi/v1/options",
+            json={"CLIP_stop_at_last_layers": 2, "eta_noise_seed_delta": 31337},
+        )
+        assert r.status_code == 200
+        opts = client.get("/sdapi/v1/options").json()
+        assert opts["CLIP_stop_at_last_layers"] == 2
+        assert opts["eta_noise_seed_delta"] == 31337
+        # per-request override still wins
+        r2 = client.post(
+            "/sdapi/v1/txt2img",
+            json={"prompt": "g", "steps": 1, "width": 64, "height": 64,
+                  "seed": 4,
+                  "override_settings": {"CLIP_stop_at_last_layers": 1}},
+        )
+        assert r2.status_code == 200
+        # reset
+        client.post(
+            "/sdapi/v1/options",
+            json={"CLIP_stop_at_last_layers": 1, "eta_noise_seed_delta": 0},
+        )
