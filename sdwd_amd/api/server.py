@@ -385,7 +385,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             height=req.height,
             steps=req.steps,
             cfg_scale=req.cfg_scale,
-            sampler_name=req.sampler_name or "Euler a",
+            sampler_name=req.sampler_name or req.sampler_index or "Euler a",
             scheduler=req.scheduler,
             seed=req.seed,
             subseed=req.subseed,

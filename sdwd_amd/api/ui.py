@@ -26,6 +26,8 @@ PAGE = """<!DOCTYPE html>
 <h2>generate</h2>
 <form onsubmit="gen(event)">
  prompt <input id="prompt" size="50" value="a herd of cows"/>
+ negative <input id="negative" size="20" value=""/>
+ sampler <select id="sampler"></select>
  batch <input id="batch" size="3" value="4"/>
  steps <input id="steps" size="3" value="20"/>
  size <input id="size" size="4" value="512"/>
@@ -86,6 +88,8 @@ async function gen(ev){
   ev.preventDefault();
   const body = {
     prompt: document.getElementById('prompt').value,
+    negative_prompt: document.getElementById('negative').value,
+    sampler_name: document.getElementById('sampler').value || 'Euler a',
     batch_size: parseInt(document.getElementById('batch').value),
     steps: parseInt(document.getElementById('steps').value),
     width: parseInt(document.getElementById('size').value),
@@ -120,8 +124,18 @@ async function saveSettings(ev){
     r.ok ? 'saved' : 'error';
   setTimeout(()=>{document.getElementById('s_saved').textContent='';}, 2000);
 }
+async function loadSamplers(){
+  try{
+    const ss = await (await fetch('/sdapi/v1/samplers')).json();
+    const sel = document.getElementById('sampler');
+    sel.innerHTML = ss.map(s =>
+      `<option ${s.name==='Euler a'?'selected':''}>${s.name}</option>`
+    ).join('');
+  }catch(e){}
+}
 setInterval(refresh, 1500);  // ref distributed.js:7-23 auto-refresh cadence
 refresh();
 loadSettings();
+loadSamplers();
 </script>
 </body></html>"""
