@@ -265,9 +265,19 @@ def create_app(engine: Optional[LocalEngine] = None,
             "job_summary": result.job_summary,
             "elapsed": result.elapsed,
         }
+        params = {
+            k: (None if isinstance(v, torch.Tensor) else v)
+            for k, v in gen.__dict__.items()
+        }
+        if params.get("control_units"):
+            params["control_units"] = [
+                {kk: (None if isinstance(vv, torch.Tensor) else vv)
+                 for kk, vv in u.items()}
+                for u in params["control_units"]
+            ]
         return {
             "images": images,
-            "parameters": gen.__dict__ | {"init_images": None},
+            "parameters": params,
             "info": json.dumps(info),
         }
 

@@ -506,3 +506,26 @@ class TestGlobalOptions:
             "/sdapi/v1/options",
             json={"CLIP_stop_at_last_layers": 1, "eta_noise_seed_delta": 0},
         )
+
+
+class TestResponseHygiene:
+    def test_parameters_never_echo_tensors(self, client):
+        init = torch.full((64, 64, 3), 90, dtype=torch.uint8)
+        mask = torch.zeros(64, 64, 3, dtype=torch.uint8)
+        mask[8:16, 8:16] = 255
+        r = client.post(
+            "/sdapi/v1/img2img",
+            json={
+                "prompt": "t", "steps": 1, "width": 64, "height": 64,
+                "seed": 2, "init_images": [
+                    base64.b64encode(encode_png(init)).decode()
+                ],
+                "mask": base64.b64encode(encode_png(mask)).decode(),
+            },
+        )
+        assert r.status_code == 200
+        params = r.json()["parameters"]
+        assert params["init_images"] is None
+        assert params["mask_image"] is None
+        # the whole response stays small (no tensor-as-list blowup)
+        assert len(r.content) < 200_000
