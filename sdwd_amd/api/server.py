@@ -578,6 +578,30 @@ def create_app(engine: Optional[LocalEngine] = None,
         engine.interrupt()
         return {}
 
+    @app.post("/sdapi/v1/unload-checkpoint")
+    def unload_checkpoint():
+        """Move every rank's weights to host memory and release VRAM
+        (sdwui parity; reload-checkpoint restores)."""
+        for pipe in engine.pipes.values():
+            pipe.model.to("cpu")
+            pipe._denoiser.cache.clear()
+        if torch.cuda.is_available():
+            torch.cuda.empty_cache()
+        return {}
+
+    @app.post("/sdapi/v1/reload-checkpoint")
+    def reload_checkpoint():
+        for pipe in engine.pipes.values():
+            pipe.model.to(pipe.device, pipe.dtype)
+        return {}
+
+    @app.post("/sdapi/v1/interrogate")
+    def interrogate():
+        # no caption model ships in this offline environment
+        raise HTTPException(
+            501, "interrogate requires a caption model (not available)"
+        )
+
     @app.post("/sdapi/v1/refresh-checkpoints")
     def refresh_checkpoints():
         from ..models.registry import refresh_checkpoint_files

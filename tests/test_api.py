@@ -529,3 +529,18 @@ class TestResponseHygiene:
         assert params["mask_image"] is None
         # the whole response stays small (no tensor-as-list blowup)
         assert len(r.content) < 200_000
+
+
+class TestUnloadReload:
+    def test_unload_then_reload_then_generate(self, client):
+        assert client.post("/sdapi/v1/unload-checkpoint").status_code == 200
+        assert client.post("/sdapi/v1/reload-checkpoint").status_code == 200
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={"prompt": "u", "steps": 1, "width": 64, "height": 64,
+                  "seed": 3},
+        )
+        assert r.status_code == 200
+
+    def test_interrogate_graceful(self, client):
+        assert client.post("/sdapi/v1/interrogate").status_code == 501
