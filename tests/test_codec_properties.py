@@ -98,3 +98,32 @@ def test_color_correct_is_stat_projection(seed, h, w):
     # idempotent up to rounding: correcting twice changes little
     c2 = color_correct(c, ref)
     assert (c2.float() - c.float()).abs().mean() < 3.0
+
+
+@hsettings(max_examples=40, deadline=None)
+@given(
+    steps=st.integers(1, 150),
+    seed=st.integers(0, 2**31 - 1),
+    cfg=st.floats(1.0, 30.0),
+    w=st.sampled_from([64, 512, 768, 1024]),
+    h=st.sampled_from([64, 512, 768, 1024]),
+    sampler=st.sampled_from(["Euler a", "DPM++ 2M Karras", "UniPC"]),
+    prompt=st.text(alphabet=st.sampled_from("abc xyz,"), min_size=1,
+                   max_size=40),
+)
+def test_infotext_emit_parse_round_trip(steps, seed, cfg, w, h, sampler,
+                                        prompt):
+    """The infotext our pipeline emits must be recoverable by our own
+    parser (the sdwui 'send to txt2img' loop)."""
+    from sdwd_amd.utils.images import parse_infotext
+
+    prompt = prompt.replace("\n", " ").strip() or "x"
+    text = (f"{prompt}\nNegative prompt: neg\n"
+            f"Steps: {steps}, Sampler: {sampler}, CFG scale: {cfg}, "
+            f"Seed: {seed}, Size: {w}x{h}, Model: sd15")
+    p = parse_infotext(text)
+    assert p["prompt"] == prompt
+    assert p["steps"] == steps
+    assert p["seed"] == seed
+    assert abs(p["cfg_scale"] - cfg) < 1e-6 or str(cfg) in str(p["cfg_scale"])
+    assert p["width"] == w and p["height"] == h
