@@ -839,9 +839,15 @@ class StableDiffusionPipeline:
                 )
                 return n.to(self.device, self.dtype)
 
+            def hr_post(xc, sigma_next):
+                # live preview only (the inpainting re-imposition, if any,
+                # is keyed to the base-resolution mask)
+                self._last_preview = xc
+                return xc
+
             x = hsampler.sample(
                 model_fn, x, noise_fn=hr_noise_fn, callback=step_callback,
-                interrupt=_interrupt,
+                interrupt=_interrupt, post_step=hr_post,
             )
 
         if decode:
