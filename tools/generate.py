@@ -21,7 +21,9 @@ from sdwd_amd.utils.images import decode_png, save_png  # noqa: E402
 
 def main() -> int:
     ap = argparse.ArgumentParser(description=__doc__)
-    ap.add_argument("--prompt", required=True)
+    ap.add_argument("--prompt", default="")
+    ap.add_argument("--prompts-file", default="",
+                    help="one prompt per line -> one image per prompt")
     ap.add_argument("--negative", default="")
     ap.add_argument("--batch", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
@@ -47,6 +49,15 @@ def main() -> int:
     add_flags(ap)
     args = ap.parse_args()
 
+    if not args.prompt and not args.prompts_file:
+        ap.error("--prompt or --prompts-file required")
+    prompts = None
+    if args.prompts_file:
+        with open(args.prompts_file) as fh:
+            prompts = [ln.strip() for ln in fh if ln.strip()]
+        args.batch = len(prompts)
+        if not args.prompt:
+            args.prompt = prompts[0]
     engine = LocalEngine(model=args.model)
     print(f"{len(engine.devices)} rank(s): {engine.devices}")
     if args.benchmark:
@@ -61,6 +72,7 @@ def main() -> int:
     res = engine.generate(
         GenerationRequest(
             prompt=args.prompt,
+            prompts=prompts,
             negative_prompt=args.negative,
             batch_size=args.batch,
             width=args.width,
