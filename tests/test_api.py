@@ -544,3 +544,31 @@ class TestUnloadReload:
 
     def test_interrogate_graceful(self, client):
         assert client.post("/sdapi/v1/interrogate").status_code == 501
+
+
+class TestEmbeddingsApi:
+    def test_loaded_listing_and_refresh(self, client, tmp_path,
+                                        monkeypatch_module):
+        from safetensors.torch import save_file
+
+        from sdwd_amd.models import embeddings
+
+        d = tmp_path / "embs"
+        d.mkdir()
+        save_file({"emb_params": torch.randn(2, 64)},
+                  str(d / "apitrigger.safetensors"))
+        monkeypatch_module.setenv("SDWD_EMBEDDINGS_DIR", str(d))
+        try:
+            r = client.post("/sdapi/v1/refresh-embeddings")
+            assert r.json()["found"] == ["apitrigger"]
+            listed = client.get("/sdapi/v1/embeddings").json()["loaded"]
+            assert listed["apitrigger"]["vectors"] == 2
+            # generation with the trigger goes through
+            g = client.post(
+                "/sdapi/v1/txt2img",
+                json={"prompt": "an apitrigger cow", "steps": 1,
+                      "width": 64, "height": 64, "seed": 9},
+            )
+            assert g.status_code == 200
+        finally:
+            embeddings.clear()
