@@ -225,25 +225,46 @@ def parse_weighted(text: str):
 
 
 def encode_weighted(text: str, max_len: int = MAX_LEN):
-    """-> (ids [77], weights [77] float). BOS/EOS/pad carry weight 1."""
-    ids = [BOS]
-    weights = [1.0]
+    """-> (ids [k*77], weights [k*77] float): sdwui "unlimited prompt
+    length" — tokens beyond 75 spill into additional BOS/EOS-bracketed
+    77-token chunks, each later encoded by CLIP separately. BOS/EOS/pad
+    carry weight 1."""
+    flat: List[tuple] = []
     for frag, w in parse_weighted(text):
         for tid in _fragment_ids(frag):
-            if len(ids) < max_len - 1:
-                ids.append(tid)
-                weights.append(w)
-    ids.append(EOS)
-    weights.append(1.0)
-    while len(ids) < max_len:
+            flat.append((tid, w))
+    body = max_len - 2
+    chunks = max(1, -(-max(1, len(flat)) // body))
+    ids: List[int] = []
+    weights: List[float] = []
+    for c in range(chunks):
+        part = flat[c * body : (c + 1) * body]
+        ids.append(BOS)
+        weights.append(1.0)
+        for tid, w in part:
+            ids.append(tid)
+            weights.append(w)
         ids.append(EOS)
         weights.append(1.0)
+        while len(ids) % max_len:
+            ids.append(EOS)
+            weights.append(1.0)
     return ids, weights
 
 
 def encode_batch_weighted(texts: List[str], device="cpu"):
+    """-> (ids [B, K, 77], weights [B, K*77]); K = max chunk count over
+    the batch (short prompts pad with empty BOS/EOS chunks)."""
     pairs = [encode_weighted(t) for t in texts]
-    ids = torch.tensor([p[0] for p in pairs], dtype=torch.long, device=device)
-    wts = torch.tensor([p[1] for p in pairs], dtype=torch.float32,
-                       device=device)
-    return ids, wts
+    k = max(len(p[0]) // MAX_LEN for p in pairs)
+    pad_ids = [BOS] + [EOS] * (MAX_LEN - 1)
+    ids_rows, wt_rows = [], []
+    for pids, pwts in pairs:
+        while len(pids) < k * MAX_LEN:
+            pids = pids + pad_ids
+            pwts = pwts + [1.0] * MAX_LEN
+        ids_rows.append(pids)
+        wt_rows.append(pwts)
+    ids = torch.tensor(ids_rows, dtype=torch.long, device=device)
+    wts = torch.tensor(wt_rows, dtype=torch.float32, device=device)
+    return ids.reshape(len(texts), k, MAX_LEN), wts

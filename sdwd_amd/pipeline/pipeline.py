@@ -226,16 +226,26 @@ class StableDiffusionPipeline:
         tokens, weights = tokenizer.encode_batch_weighted(
             prompts + negatives, device=self.device
         )
+        # [N, K, 77]: long prompts spill into K chunks, each run through
+        # CLIP separately and concatenated (sdwui unlimited prompt length)
+        n, k, L = tokens.shape
+        flat = tokens.reshape(n * k, L)
         m = bundle if bundle is not None else self.model
+
+        def _cat_chunks(h):
+            return h.reshape(n, k * L, h.shape[-1])
+
         if m.is_sdxl:
-            h1 = m.text_encoder(tokens, penultimate=True)
-            h2 = m.text_encoder_2(tokens, penultimate=True)
+            h1 = _cat_chunks(m.text_encoder(flat, penultimate=True))
+            h2 = _cat_chunks(m.text_encoder_2(flat, penultimate=True))
             ctx = torch.cat([h1, h2], dim=-1)
+            # pooled conditioning comes from the FIRST chunk's EOT
+            first = tokens[:, 0]
             pooled = m.text_encoder_2.pooled(
-                tokens, m.text_encoder_2(tokens)
+                first, m.text_encoder_2(first)
             )
         else:
-            ctx = m.text_encoder(tokens, clip_skip=clip_skip)
+            ctx = _cat_chunks(m.text_encoder(flat, clip_skip=clip_skip))
             pooled = None
         # prompt-attention weights (sdwui semantics): scale the hidden
         # states of weighted tokens, then restore the original mean

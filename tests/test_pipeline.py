@@ -1362,3 +1362,42 @@ class TestHrSampler:
             PipelineRequest(**base, hr_sampler_name="Euler a")
         ).images
         assert torch.equal(a, same)
+
+
+class TestLongPrompts:
+    def test_chunked_encoding_shapes(self):
+        from sdwd_amd.models import tokenizer as tk
+
+        long_prompt = " ".join(f"word{i}" for i in range(160))
+        ids, wts = tk.encode_batch_weighted([long_prompt])
+        assert ids.shape == (1, 3, 77)  # 160 tokens -> 3 chunks of 75
+        assert wts.shape == (1, 231)
+        # every chunk is BOS-bracketed
+        assert (ids[0, :, 0] == tk.BOS).all()
+
+    def test_long_prompt_generates_and_uses_tail(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        words = [f"word{i}" for i in range(100)]
+        base = dict(steps=2, width=64, height=64, seeds=[3])
+        a = pipe.generate(
+            PipelineRequest(prompt=" ".join(words), **base)
+        ).images
+        # change ONLY a token beyond position 75: output must change
+        words[90] = "different"
+        b = pipe.generate(
+            PipelineRequest(prompt=" ".join(words), **base)
+        ).images
+        assert not torch.equal(a, b)
+        assert torch.isfinite(a.float()).all()
+
+    def test_short_prompts_unchanged(self, pipe):
+        """K=1 path must stay bitwise identical to the pre-chunking
+        behavior (deterministic regression anchor)."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        req = PipelineRequest(prompt="a cow", steps=2, width=64, height=64,
+                              seeds=[7])
+        a = pipe.generate(req).images
+        b = pipe.generate(req).images
+        assert torch.equal(a, b)
