@@ -506,3 +506,35 @@ class TestInpaintModelGPU:
                             seeds=[1])
         )
         assert torch.isfinite(res.images.float()).all()
+
+
+class TestLongPromptGPU:
+    def test_cross_attention_154_keys(self, dev):
+        """Two-chunk prompts give Sk=154 cross-attention (2*64 + 26 tail);
+        numerics vs fp32 reference at the exact production layout."""
+        from sdwd_amd import ops
+
+        torch.manual_seed(5)
+        b, sq, sk, h, d = 2, 1024, 154, 8, 40
+        q = torch.randn(b, sq, h, d, device=dev, dtype=torch.bfloat16)
+        k = torch.randn(b, sk, h, d, device=dev, dtype=torch.bfloat16)
+        v = torch.randn(b, sk, h, d, device=dev, dtype=torch.bfloat16)
+        out = ops.attention_bshd(q, k, v)
+        qf = q.float().permute(0, 2, 1, 3)
+        kf = k.float().permute(0, 2, 1, 3)
+        vf = v.float().permute(0, 2, 1, 3)
+        s = torch.matmul(qf, kf.transpose(-1, -2)) * (d ** -0.5)
+        ref = torch.matmul(s.softmax(-1), vf).permute(0, 2, 1, 3)
+        err = (out.float() - ref).abs().max().item()
+        assert err < 0.06, err
+
+    def test_long_prompt_pipeline_gpu(self, dev):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny", device=dev)
+        prompt = " ".join(f"w{i}" for i in range(100))
+        res = pipe.generate(
+            PipelineRequest(prompt=prompt, steps=2, width=64, height=64,
+                            seeds=[4])
+        )
+        assert torch.isfinite(res.images.float()).all()
