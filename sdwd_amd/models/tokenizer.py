@@ -227,18 +227,24 @@ def parse_weighted(text: str):
 def encode_weighted(text: str, max_len: int = MAX_LEN):
     """-> (ids [k*77], weights [k*77] float): sdwui "unlimited prompt
     length" — tokens beyond 75 spill into additional BOS/EOS-bracketed
-    77-token chunks, each later encoded by CLIP separately. BOS/EOS/pad
-    carry weight 1."""
-    flat: List[tuple] = []
-    for frag, w in parse_weighted(text):
-        for tid in _fragment_ids(frag):
-            flat.append((tid, w))
+    77-token chunks, each later encoded by CLIP separately; the literal
+    keyword ``BREAK`` forces a chunk boundary. BOS/EOS/pad carry
+    weight 1."""
+    import re as _re
+
     body = max_len - 2
-    chunks = max(1, -(-max(1, len(flat)) // body))
+    chunk_lists: List[List[tuple]] = []
+    for section in _re.split(r"\bBREAK\b", text):
+        flat: List[tuple] = []
+        for frag, w in parse_weighted(section):
+            for tid in _fragment_ids(frag):
+                flat.append((tid, w))
+        n_chunks = max(1, -(-max(1, len(flat)) // body))
+        for c in range(n_chunks):
+            chunk_lists.append(flat[c * body : (c + 1) * body])
     ids: List[int] = []
     weights: List[float] = []
-    for c in range(chunks):
-        part = flat[c * body : (c + 1) * body]
+    for part in chunk_lists:
         ids.append(BOS)
         weights.append(1.0)
         for tid, w in part:

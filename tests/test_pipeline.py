@@ -1401,3 +1401,27 @@ class TestLongPrompts:
         a = pipe.generate(req).images
         b = pipe.generate(req).images
         assert torch.equal(a, b)
+
+
+class TestBreakKeyword:
+    def test_break_forces_chunk_boundary(self):
+        from sdwd_amd.models import tokenizer as tk
+
+        ids, _ = tk.encode_weighted("a cow BREAK a dog")
+        assert len(ids) == 2 * 77
+        assert ids[0] == tk.BOS and ids[77] == tk.BOS
+        # lowercase 'break' is just a word, not a separator
+        ids2, _ = tk.encode_weighted("a cow break a dog")
+        assert len(ids2) == 77
+
+    def test_break_changes_output(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(steps=2, width=64, height=64, seeds=[8])
+        a = pipe.generate(
+            PipelineRequest(prompt="a cow BREAK a dog", **base)
+        ).images
+        b = pipe.generate(
+            PipelineRequest(prompt="a cow a dog", **base)
+        ).images
+        assert not torch.equal(a, b)
