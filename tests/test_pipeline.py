@@ -1425,3 +1425,25 @@ class TestBreakKeyword:
             PipelineRequest(prompt="a cow a dog", **base)
         ).images
         assert not torch.equal(a, b)
+
+
+class TestEditingLongInteraction:
+    def test_segment_with_different_chunk_count(self, pipe):
+        """A prompt edit that switches to a >75-token text mid-run: the
+        conditioning length changes between segments."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        long_tail = " ".join(f"w{i}" for i in range(90))
+        req = PipelineRequest(
+            prompt=f"[a cow:{long_tail}:0.5]", steps=4, width=64, height=64,
+            seeds=[2],
+        )
+        a = pipe.generate(req).images
+        b = pipe.generate(req).images
+        assert torch.equal(a, b)
+        assert torch.isfinite(a.float()).all()
+        short = pipe.generate(
+            PipelineRequest(prompt="a cow", steps=4, width=64, height=64,
+                            seeds=[2])
+        ).images
+        assert not torch.equal(a, short)
