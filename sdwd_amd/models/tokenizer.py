@@ -164,12 +164,17 @@ def encode_batch(texts: List[str], device="cpu") -> torch.Tensor:
     return torch.tensor([encode(t) for t in texts], dtype=torch.long, device=device)
 
 
+_ESC = {"\\(": "\x03", "\\)": "\x04", "\\[": "\x05", "\\]": "\x06"}
+
+
 def parse_weighted(text: str):
     """sdwui prompt-attention syntax -> [(fragment, weight)].
 
     ``(x)`` -> 1.1x, ``((x))`` -> 1.21x, ``[x]`` -> /1.1, ``(x:1.3)`` ->
-    exactly 1.3; nesting multiplies.
+    exactly 1.3; nesting multiplies. Escaped ``\\(`` etc. are literal.
     """
+    for esc, sent in _ESC.items():
+        text = text.replace(esc, sent)
     out = []
     stack = [1.0]
     buf = ""
@@ -221,7 +226,14 @@ def parse_weighted(text: str):
             buf += ch
         i += 1
     flush()
-    return [(f, w) for f, w in out if f.strip()]
+    rev = {v: k[-1] for k, v in _ESC.items()}
+
+    def unesc(f):
+        for sent, ch in rev.items():
+            f = f.replace(sent, ch)
+        return f
+
+    return [(unesc(f), w) for f, w in out if f.strip()]
 
 
 def encode_weighted(text: str, max_len: int = MAX_LEN):

@@ -1447,3 +1447,16 @@ class TestEditingLongInteraction:
                             seeds=[2])
         ).images
         assert not torch.equal(a, short)
+
+
+class TestEscapedParens:
+    def test_literal_brackets(self):
+        from sdwd_amd.models.tokenizer import parse_weighted
+
+        assert parse_weighted(r"a \(literal\) cow") == [
+            (r"a (literal) cow", 1.0)
+        ]
+        assert parse_weighted(r"\[not down\]") == [("[not down]", 1.0)]
+        # unescaped still weights
+        frags = parse_weighted(r"mix (up) and \(flat\)")
+        assert ("up", 1.1) in [(f.strip(), w) for f, w in frags]
