@@ -114,6 +114,32 @@ def sgm_uniform_schedule(steps: int) -> Schedule:
     return Schedule(sigmas=sigmas.float(), timesteps=idx.double().float())
 
 
+def polyexponential_schedule(steps: int, rho: float = 1.0) -> Schedule:
+    """Log-linear ramp raised to rho (k-diffusion
+    get_sigmas_polyexponential; rho=1 equals Exponential)."""
+    table = make_sigmas_full()
+    smin, smax = float(table[0]), float(table[-1])
+    ramp = torch.linspace(1, 0, steps, dtype=torch.float64) ** rho
+    sig = (ramp * (math.log(smax) - math.log(smin)) + math.log(smin)).exp()
+    ts = _timesteps_for(sig, table)
+    sigmas = torch.cat([sig, torch.zeros(1, dtype=torch.float64)])
+    return Schedule(sigmas=sigmas.float(), timesteps=ts.float())
+
+
+def kl_optimal_schedule(steps: int) -> Schedule:
+    """KL-optimal spacing (arctan-uniform; sdwui "KL Optimal",
+    from "Align Your Steps"' analysis)."""
+    table = make_sigmas_full()
+    smin, smax = float(table[0]), float(table[-1])
+    t = torch.linspace(
+        math.atan(smax), math.atan(smin), steps, dtype=torch.float64
+    )
+    sig = torch.tan(t)
+    ts = _timesteps_for(sig, table)
+    sigmas = torch.cat([sig, torch.zeros(1, dtype=torch.float64)])
+    return Schedule(sigmas=sigmas.float(), timesteps=ts.float())
+
+
 SCHEDULERS = {
     "automatic": None,  # resolved from the sampler name
     "uniform": discrete_schedule,
@@ -121,11 +147,15 @@ SCHEDULERS = {
     "exponential": exponential_schedule,
     "sgm uniform": sgm_uniform_schedule,
     "sgm_uniform": sgm_uniform_schedule,
+    "polyexponential": polyexponential_schedule,
+    "kl optimal": kl_optimal_schedule,
+    "kl_optimal": kl_optimal_schedule,
 }
 
 
 def scheduler_names() -> list:
-    return ["Automatic", "Uniform", "Karras", "Exponential", "SGM Uniform"]
+    return ["Automatic", "Uniform", "Karras", "Exponential", "SGM Uniform",
+            "Polyexponential", "KL Optimal"]
 
 
 def schedule_for(

@@ -1460,3 +1460,32 @@ class TestEscapedParens:
         # unescaped still weights
         frags = parse_weighted(r"mix (up) and \(flat\)")
         assert ("up", 1.1) in [(f.strip(), w) for f, w in frags]
+
+
+class TestMoreSchedulers:
+    def test_poly_and_kl_monotone(self):
+        from sdwd_amd.pipeline.schedule import (
+            exponential_schedule, kl_optimal_schedule,
+            polyexponential_schedule,
+        )
+
+        for fn in (polyexponential_schedule, kl_optimal_schedule):
+            sch = fn(14)
+            assert len(sch.sigmas) == 15 and sch.sigmas[-1] == 0
+            assert (sch.sigmas[:-1].diff() < 0).all(), fn.__name__
+        # rho=1 polyexponential == exponential
+        a = polyexponential_schedule(10).sigmas
+        b = exponential_schedule(10).sigmas
+        assert torch.allclose(a, b, rtol=1e-5)
+
+    def test_selectable_and_distinct(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="s", steps=4, width=64, height=64, seeds=[6])
+        outs = {}
+        for sched in ("Karras", "KL Optimal", "Polyexponential"):
+            outs[sched] = pipe.generate(
+                PipelineRequest(**base, scheduler=sched)
+            ).images
+        assert not torch.equal(outs["Karras"], outs["KL Optimal"])
+        assert not torch.equal(outs["KL Optimal"], outs["Polyexponential"])
