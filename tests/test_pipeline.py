@@ -1489,3 +1489,36 @@ class TestMoreSchedulers:
             ).images
         assert not torch.equal(outs["Karras"], outs["KL Optimal"])
         assert not torch.equal(outs["KL Optimal"], outs["Polyexponential"])
+
+
+class TestSurfaceConsistency:
+    def test_every_sampler_has_sane_eta_cost(self):
+        from sdwd_amd.core.eta import SAMPLER_EVALS_PER_STEP, sampler_cost
+        from sdwd_amd.pipeline.samplers import SAMPLERS
+
+        for name in SAMPLERS:
+            assert 0.5 <= sampler_cost(name) <= 4.0, name
+        # no stale entries for samplers that no longer exist
+        for name in SAMPLER_EVALS_PER_STEP:
+            assert name in SAMPLERS, f"stale eta entry {name}"
+
+    def test_scheduler_names_all_resolve(self):
+        from sdwd_amd.pipeline.schedule import (
+            SCHEDULERS, schedule_for, scheduler_names,
+        )
+
+        for label in scheduler_names():
+            sch = schedule_for("Euler", 8, label)
+            assert sch.steps == 8, label
+        for key, fn in SCHEDULERS.items():
+            if fn is not None:
+                assert (fn(6).sigmas[:-1].diff() < 0).all(), key
+
+    def test_karras_suffix_samplers_share_rule(self):
+        """'X Karras' must map to the same update rule as 'X'."""
+        from sdwd_amd.pipeline.samplers import SAMPLERS
+
+        for name, cls in SAMPLERS.items():
+            if name.endswith(" Karras"):
+                base = name[: -len(" Karras")]
+                assert SAMPLERS.get(base) is cls, name
