@@ -89,6 +89,7 @@ class Img2ImgRequest(Txt2ImgRequest):
     inpaint_full_res_padding: int = 32
     mask_blur: int = 4
     inpainting_fill: int = 1  # 0 fill, 1 original, 2 latent noise, 3 nothing
+    inpainting_mask_invert: int = 0  # 1 = inpaint NOT-masked region
     color_correction: bool = False
     # sdwui resize_mode: 0 just resize, 1 crop and resize, 2 resize and fill
     resize_mode: int = 0
@@ -418,6 +419,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             inpaint_full_res_padding=req.inpaint_full_res_padding,
             mask_blur=req.mask_blur,
             inpainting_fill=req.inpainting_fill,
+            inpainting_mask_invert=req.inpainting_mask_invert,
             color_correction=req.color_correction,
         )
         return run_generation(gen, req.send_images, req.save_images)

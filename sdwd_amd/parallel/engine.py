@@ -82,6 +82,8 @@ class GenerationRequest:
     # sdwui "Masked content": 0 fill (mean color), 1 original,
     # 2 latent noise, 3 latent nothing
     inpainting_fill: int = 1
+    # sdwui "Inpaint masked / not masked" (1 inverts the mask)
+    inpainting_mask_invert: int = 0
     # sampler stochasticity + uncond-skip perf knob (sdwui Sampler params)
     s_churn: float = 0.0
     s_tmin: float = 0.0
@@ -220,6 +222,9 @@ def _preprocess_mask(gen: GenerationRequest) -> GenerationRequest:
     m = gen.mask_image
     if m.dim() == 3:
         m = m[0]
+    if gen.inpainting_mask_invert:
+        m = 255 - m
+        gen = replace_dc(gen, mask_image=m)
     if gen.mask_blur > 0:
         gen = replace_dc(gen, mask_image=_blur_mask(m, gen.mask_blur))
         m = gen.mask_image
