@@ -494,6 +494,18 @@ def create_app(engine: Optional[LocalEngine] = None,
         out8 = out.clamp(0, 255).to(torch.uint8)[0].permute(1, 2, 0)
         return {"image": _b64_png(out8), "html_info": ""}
 
+    @app.post("/sdapi/v1/extra-batch-images")
+    def extra_batch_images(body: Dict[str, Any]):
+        """Batch version of extra-single-image (same pixel upscalers)."""
+        images = body.get("imageList") or body.get("image_list") or []
+        out = []
+        for entry in images:
+            data = entry.get("data") if isinstance(entry, dict) else entry
+            sub = dict(body)
+            sub["image"] = data
+            out.append(extra_single_image(sub)["image"])
+        return {"images": out, "html_info": ""}
+
     @app.get("/sdapi/v1/prompt-styles")
     def prompt_styles():
         from ..pipeline.styles import all_styles, refresh_styles

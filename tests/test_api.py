@@ -572,3 +572,27 @@ class TestEmbeddingsApi:
             assert g.status_code == 200
         finally:
             embeddings.clear()
+
+
+class TestExtrasBatch:
+    def test_batch_upscale(self, client):
+        from sdwd_amd.utils.images import decode_png, encode_png
+
+        imgs = [torch.randint(0, 255, (8, 8, 3), dtype=torch.uint8)
+                for _ in range(3)]
+        payload = {
+            "upscaling_resize": 2.0, "upscaler_1": "Nearest",
+            "imageList": [
+                {"data": base64.b64encode(encode_png(i)).decode(),
+                 "name": f"i{n}.png"}
+                for n, i in enumerate(imgs)
+            ],
+        }
+        r = client.post("/sdapi/v1/extra-batch-images", json=payload)
+        assert r.status_code == 200
+        outs = r.json()["images"]
+        assert len(outs) == 3
+        for src, b in zip(imgs, outs):
+            dec = decode_png(base64.b64decode(b))
+            assert dec.shape == (16, 16, 3)
+            assert torch.equal(dec[::2, ::2], src)
