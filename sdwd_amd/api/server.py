@@ -565,6 +565,22 @@ def create_app(engine: Optional[LocalEngine] = None,
     def scripts():
         return {"txt2img": [], "img2img": []}
 
+    @app.get("/sdapi/v1/script-info")
+    def script_info():
+        return []
+
+    @app.get("/sdapi/v1/cmd-flags")
+    def cmd_flags():
+        return {
+            "api": True,
+            "listen": None,
+            "port": None,
+            "device_id": None,
+            "medvram": False,
+            "lowvram": False,
+            "xformers": False,
+        }
+
     @app.get("/sdapi/v1/sd-vae")
     def sd_vae():
         return [{"model_name": "auto", "filename": ""}]

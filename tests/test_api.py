@@ -320,7 +320,8 @@ class TestControl:
     def test_startup_enumeration_routes(self, client):
         """The static lists sdwui GUIs fetch at startup all answer."""
         for route in ("upscalers", "latent-upscale-modes", "face-restorers",
-                      "hypernetworks", "scripts", "embeddings", "sd-vae"):
+                      "hypernetworks", "scripts", "embeddings", "sd-vae",
+                      "script-info", "cmd-flags"):
             r = client.get(f"/sdapi/v1/{route}")
             assert r.status_code == 200, route
         ups = [u["name"] for u in client.get("/sdapi/v1/upscalers").json()]
