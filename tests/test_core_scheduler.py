@@ -447,3 +447,18 @@ class TestPerTaskToggles:
         i2i = world.make_jobs(GenRequest(batch_size=4, task="img2img"))
         assert len(t2i) == 1
         assert len(i2i) == 2
+
+
+class TestCapPersistence:
+    def test_learned_caps_survive_save_load(self, tmp_path):
+        from sdwd_amd.core import World
+
+        path = str(tmp_path / "cfg.json")
+        w = World.from_devices(2, config_path=path)
+        w.get_worker("gpu1").pixel_cap = 123456
+        w.get_worker("gpu0").eta.avg_ipm = 44.0
+        w.save()
+        w2 = World.from_devices(2, config_path=path)
+        w2.load()
+        assert w2.get_worker("gpu1").pixel_cap == 123456
+        assert w2.get_worker("gpu0").eta.avg_ipm == 44.0
