@@ -114,7 +114,9 @@ class World:
         self._assign_seeds(jobs, request)
         return jobs
 
-    def optimize_jobs(self, request: GenRequest) -> List[Job]:
+    def optimize_jobs(
+        self, request: GenRequest, allow_complementary: bool = True
+    ) -> List[Job]:
         workers = self.active_workers()
         if not workers:
             raise RuntimeError("no available ranks")
@@ -142,7 +144,7 @@ class World:
         self._apply_pixel_caps(plan, workers, request)
 
         # complementary production for deferred ranks
-        if self.settings.complement_production:
+        if self.settings.complement_production and allow_complementary:
             self._add_complementary(plan, workers, request)
 
         jobs = [j for j in plan.values() if not j.empty]
@@ -342,7 +344,10 @@ class World:
         if not survivors:
             raise RuntimeError("no surviving ranks to requeue onto")
         sub = GenRequest(**{**request.__dict__, "batch_size": failed_job.batch_size})
-        jobs = self.optimize_jobs(sub)
+        # no complementary/bonus jobs on a requeue: the retry plan must total
+        # EXACTLY the failed shard's batch so the seed slices below align
+        # (a bonus job would read past failed_job.seeds and fall back to 0)
+        jobs = self.optimize_jobs(sub, allow_complementary=False)
         # keep the failed shard's slots and seeds
         pos = 0
         for job in jobs:

@@ -140,6 +140,14 @@ def _build_tiny_v(name: str) -> ModelBundle:
     return b
 
 
+def _build_tiny_inpaint_v(name: str) -> ModelBundle:
+    """CPU-test v-prediction 9-channel inpainting model (exercises the
+    _to_eps-on-pre-concat-latents path in pipeline.py model_fn)."""
+    b = _build_tiny_inpaint(name)
+    b.prediction_type = "v"
+    return b
+
+
 def _build_sdxl(name: str) -> ModelBundle:
     te = CLIPTextEncoder(d_model=768, layers=12, heads=12)
     te2 = CLIPTextEncoder(d_model=1280, layers=32, heads=20)
@@ -194,6 +202,7 @@ _BUILDERS: Dict[str, Callable[[str], ModelBundle]] = {
     "sdxl": _build_sdxl,
     "tiny-v": _build_tiny_v,
     "tiny-inpaint": _build_tiny_inpaint,
+    "tiny-inpaint-v": _build_tiny_inpaint_v,
     "tiny": _build_tiny,
     "tiny-xl": _build_tiny_xl,
 }

@@ -300,12 +300,13 @@ class UNetModel(nn.Module):
         for blk in self.down:
             h = blk(h, emb, context)
             skips.append(h)
-        if control is not None:
-            skips = [s + c for s, c in zip(skips, control["down"])]
-            h = skips[-1]
+        # cldm semantics: the mid block sees the UNMODIFIED last down
+        # activation; down residuals apply only at the up-path skip-concat
+        # and the mid residual after the mid block.
         h = self.mid(h, emb, context)
         if control is not None:
             h = h + control["mid"]
+            skips = [s + c for s, c in zip(skips, control["down"])]
         for blk in self.up:
             h = torch.cat([h, skips.pop()], dim=1)
             h = blk(h, emb, context)

@@ -791,6 +791,93 @@ class DistributedEngine(_EngineBase):
             self.world.get_worker(f"gpu{r}").eta.avg_ipm = ipm
         return {w.label: w.eta.avg_ipm for w in self.world.workers}
 
+    def _weight_digest(self) -> float:
+        """Cheap per-pipe weight digest (sum of |param|) for the cross-rank
+        identity check; cached until the pipe object is rebuilt."""
+        cached = getattr(self.pipe, "_sdwd_digest", None)
+        if cached is None:
+            total = 0.0
+            for m in (
+                self.pipe.model.text_encoder,
+                self.pipe.model.text_encoder_2,
+                self.pipe.model.unet,
+                self.pipe.model.vae,
+            ):
+                if m is None:
+                    continue
+                acc = None
+                for p in m.parameters():
+                    s = p.detach().float().abs().sum()
+                    acc = s if acc is None else acc + s
+                if acc is not None:
+                    total += float(acc.item())
+            self.pipe._sdwd_digest = cached = total
+        return cached
+
+    def _exec_shard(self, gen: GenerationRequest, job) -> dict:
+        """Run one shard on this rank. Returns images/infos/status; never
+        raises (failures are reported through ok=0 so the collectives that
+        follow stay symmetric across ranks)."""
+        hf = gen.hr_scale if gen.enable_hr else 1.0
+        h, w = int(gen.height * hf), int(gen.width * hf)
+        out = {
+            "images": torch.zeros(0, h, w, 3, dtype=torch.uint8),
+            "infos": [],
+            "ok": 1.0,
+            "elapsed": 0.0,
+            "cap": 0.0,
+            "interrupted": False,
+        }
+        if job is None or job.batch_size <= 0:
+            return out
+        try:
+            init_latents = None
+            if gen.init_images is not None:
+                src = gen.init_images
+                idx = [
+                    (job.gallery_offset + i) % src.shape[0]
+                    for i in range(job.batch_size)
+                ]
+                init_latents = self.pipe.encode_image(
+                    src[idx], seeds=job.seeds
+                )
+            ts = time.perf_counter()
+
+            last_pub = [0.0]
+
+            def on_step(i, n):
+                # publish rank progress out-of-band, throttled to the
+                # interrupt-poll cadence (C19 status for torchrun mode)
+                now = time.monotonic()
+                if self._store is not None and (
+                    now - last_pub[0] > 0.5 or i == n
+                ):
+                    last_pub[0] = now
+                    try:
+                        self._store.set(
+                            f"sdwd_prog_{self.rank}", f"{i}/{n}"
+                        )
+                    except Exception:
+                        pass
+
+            res = self.pipe.generate(
+                _job_pipeline_request(gen, job, init_latents),
+                interrupt=self._interrupted,
+                step_callback=on_step,
+            )
+            out["elapsed"] = time.perf_counter() - ts
+            out["images"] = res.images
+            out["infos"] = res.infotexts
+            out["interrupted"] = res.interrupted
+        except Exception as exc:  # noqa: BLE001
+            log.warning("rank %d shard failed: %s", self.rank, exc)
+            out["ok"] = 0.0
+            if isinstance(exc, torch.cuda.OutOfMemoryError):
+                me = self.world.get_worker(self.label)
+                if _learn_pixel_cap(me, gen, job.batch_size):
+                    out["cap"] = float(me.pixel_cap)
+        return out
+
     def generate(self, gen: GenerationRequest) -> GalleryResult:
         t0 = time.perf_counter()
         self.heartbeat()
@@ -813,65 +900,34 @@ class DistributedEngine(_EngineBase):
             self.pipe = StableDiffusionPipeline(
                 want, device=self.device, dtype=self._dtype
             )
+        # cross-rank weight-identity check: ranks that claim the same model
+        # name must hold identical weights (registry init is deterministic;
+        # file-backed checkpoints could silently diverge, e.g. a stale file
+        # on one node path). Cheap: the digest is cached per pipe build.
+        if self.is_dist:
+            pairs = pg.allgather_object((want, self._weight_digest()))
+            by_name: Dict[str, float] = {}
+            for r, (name, d) in enumerate(pairs):
+                if name in by_name:
+                    ref = by_name[name]
+                    if abs(ref - d) > 1e-3 * max(1.0, abs(ref)):
+                        raise RuntimeError(
+                            f"weight divergence for model '{name}' across "
+                            f"ranks ({ref:.6g} vs {d:.6g} on rank {r}); "
+                            "resync the checkpoint files"
+                        )
+                else:
+                    by_name[name] = d
         gen = _preprocess_mask(gen)
         gen, paste_ctx = _crop_for_inpaint_full_res(gen)
 
         mine = next((j for j in jobs if j.worker_label == self.label), None)
-        hf = gen.hr_scale if gen.enable_hr else 1.0
-        h, w = int(gen.height * hf), int(gen.width * hf)
-        my_images = torch.zeros(0, h, w, 3, dtype=torch.uint8)
-        my_ok, my_elapsed, my_cap = 1.0, 0.0, 0.0
-        my_infos: List[str] = []
-        interrupted = False
-        if mine is not None and mine.batch_size > 0:
-            try:
-                init_latents = None
-                if gen.init_images is not None:
-                    src = gen.init_images
-                    idx = [
-                        (mine.gallery_offset + i) % src.shape[0]
-                        for i in range(mine.batch_size)
-                    ]
-                    init_latents = self.pipe.encode_image(
-                        src[idx], seeds=mine.seeds
-                    )
-                ts = time.perf_counter()
+        r1 = self._exec_shard(gen, mine)
+        interrupted = r1["interrupted"]
 
-                last_pub = [0.0]
-
-                def on_step(i, n):
-                    # publish rank progress out-of-band, throttled to the
-                    # interrupt-poll cadence (C19 status for torchrun mode)
-                    now = time.monotonic()
-                    if self._store is not None and (
-                        now - last_pub[0] > 0.5 or i == n
-                    ):
-                        last_pub[0] = now
-                        try:
-                            self._store.set(
-                                f"sdwd_prog_{self.rank}", f"{i}/{n}"
-                            )
-                        except Exception:
-                            pass
-
-                res = self.pipe.generate(
-                    _job_pipeline_request(gen, mine, init_latents),
-                    interrupt=self._interrupted,
-                    step_callback=on_step,
-                )
-                my_elapsed = time.perf_counter() - ts
-                my_images = res.images
-                my_infos = res.infotexts
-                interrupted = res.interrupted
-            except Exception as exc:  # noqa: BLE001
-                log.warning("rank %d shard failed: %s", self.rank, exc)
-                my_ok = 0.0
-                if isinstance(exc, torch.cuda.OutOfMemoryError):
-                    me = self.world.get_worker(self.label)
-                    if _learn_pixel_cap(me, gen, mine.batch_size):
-                        my_cap = float(me.pixel_cap)
-
-        # status + images to everyone (rank 0 consumes)
+        # status + images + infotexts to everyone (rank 0 consumes; the
+        # infotext gather keeps ref parity: distributed.py:343-349 preserved
+        # per-image infotexts across all workers)
         sizes = [
             next(
                 (
@@ -884,15 +940,17 @@ class DistributedEngine(_EngineBase):
             for r in range(self.world_size)
         ]
         status = pg.allgather_floats(
-            [my_ok, my_elapsed, my_cap], self.device
+            [r1["ok"], r1["elapsed"], r1["cap"]], self.device
         )
         # every rank applies learned pixel caps so future plans agree
         for r in range(self.world_size):
             if len(status[r]) > 2 and status[r][2] > 0:
                 self.world.get_worker(f"gpu{r}").pixel_cap = int(status[r][2])
-        shard_dev = my_images.to(self.device)
-        gathered = pg.gather_images(shard_dev, sizes, self.device)
-        all_images = gathered.cpu() if gathered is not None else my_images
+        gathered = pg.gather_images(
+            r1["images"].to(self.device), sizes, self.device
+        )
+        all_images = gathered.cpu() if gathered is not None else r1["images"]
+        all_infos = pg.allgather_object(r1["infos"])
 
         # slice the gathered stack back into per-rank shards
         shards: Dict[str, torch.Tensor] = {}
@@ -901,27 +959,103 @@ class DistributedEngine(_EngineBase):
         for r in range(self.world_size):
             n = sizes[r]
             shards[f"gpu{r}"] = all_images[pos : pos + n]
+            infos[f"gpu{r}"] = list(all_infos[r]) if r < len(all_infos) else []
             pos += n
-        infos[self.label] = my_infos
 
-        # recovery: rank 0 re-runs failed shards locally
+        # eta bookkeeping + failure detection on EVERY rank (identical
+        # allgathered inputs keep the replicated World state in agreement)
+        failed: List[Job] = []
         for r in range(self.world_size):
             ok, elapsed = status[r][0], status[r][1]
             job = next(
                 (j for j in jobs if j.worker_label == f"gpu{r}"), None
             )
-            if job is not None:
-                job.elapsed = elapsed
-                if job.predicted_eta > 0 and elapsed > 0:
-                    self._record_eta(self.world, job)
-            if ok < 0.5 and job is not None:
+            if job is None:
+                continue
+            job.elapsed = elapsed
+            if job.predicted_eta > 0 and elapsed > 0:
+                self._record_eta(self.world, job)
+            if ok < 0.5:
                 self.world.get_worker(f"gpu{r}").set_state(State.UNAVAILABLE)
-                if self.rank == 0 and not job.complementary:
-                    log.warning(
-                        "re-running failed shard of %s on rank 0", job.worker_label
+                if not job.complementary:
+                    failed.append(job)
+                elif self.rank == 0:
+                    jobs.remove(job)  # bonus images are simply dropped
+
+        # recovery: reshard failed shards over the survivors — the same
+        # semantics LocalEngine has (seeds preserved, benchmark-weighted),
+        # instead of a serial re-run on rank 0. Rank 0 plans; the plan is
+        # broadcast so every rank runs its part of the retry collectives.
+        retries: List[Job] = []
+        if self.rank == 0:
+            for job in failed:
+                try:
+                    rjobs = self.world.requeue_failed(job, gen.sched())
+                except RuntimeError:
+                    log.error(
+                        "no survivors; shard of %s lost", job.worker_label
                     )
-                    res = self.pipe.generate(_job_pipeline_request(gen, job))
-                    shards[job.worker_label] = res.images
+                    continue
+                log.warning(
+                    "resharding failed shard of %s over %s",
+                    job.worker_label,
+                    [j.worker_label for j in rjobs],
+                )
+                jobs.remove(job)
+                jobs.extend(rjobs)
+                retries.extend(rjobs)
+        if self.is_dist and failed:
+            retries = pg.broadcast_object(retries if self.rank == 0 else None)
+
+        if retries:
+            my_retries = [
+                j for j in retries if j.worker_label == self.label
+            ]
+            imgs2, infos2 = [], []
+            ok2 = 1.0
+            for job in my_retries:
+                rr = self._exec_shard(gen, job)
+                imgs2.append(rr["images"])
+                infos2.extend(rr["infos"])
+                ok2 = min(ok2, rr["ok"])
+                interrupted = interrupted or rr["interrupted"]
+            my_img2 = (
+                torch.cat(imgs2, dim=0)
+                if imgs2
+                else torch.zeros(0, 1, 1, 3, dtype=torch.uint8)
+            )
+            sizes2 = [
+                sum(
+                    j.batch_size
+                    for j in retries
+                    if j.worker_label == f"gpu{r}"
+                )
+                for r in range(self.world_size)
+            ]
+            status2 = pg.allgather_floats([ok2], self.device)
+            g2 = pg.gather_images(
+                my_img2.to(self.device), sizes2, self.device
+            )
+            ai2 = g2.cpu() if g2 is not None else my_img2
+            inf2 = pg.allgather_object(infos2)
+            for r in range(self.world_size):
+                if status2[r][0] < 0.5:
+                    self.world.get_worker(f"gpu{r}").set_state(
+                        State.UNAVAILABLE, strict=False
+                    )
+            if self.rank == 0:
+                pos = 0
+                for r in range(self.world_size):
+                    ipos = 0
+                    for job in retries:
+                        if job.worker_label != f"gpu{r}":
+                            continue
+                        shards[id(job)] = ai2[pos : pos + job.batch_size]
+                        infos[id(job)] = inf2[r][
+                            ipos : ipos + job.batch_size
+                        ]
+                        pos += job.batch_size
+                        ipos += job.batch_size
 
         result = self._assemble(
             gen, jobs, shards, infos, time.perf_counter() - t0, interrupted

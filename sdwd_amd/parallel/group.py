@@ -79,6 +79,16 @@ def broadcast_object(obj, src: int = 0):
     return box[0]
 
 
+def allgather_object(obj) -> List:
+    """All-gather one small picklable object per rank (infotext lists,
+    (model, digest) pairs). Returns a world_size-long list in rank order."""
+    if not dist.is_initialized() or dist.get_world_size() == 1:
+        return [obj]
+    out: List = [None] * dist.get_world_size()
+    dist.all_gather_object(out, obj)
+    return out
+
+
 def allgather_floats(values: List[float], device) -> List[List[float]]:
     """All-gather a small per-rank float vector (ipm, elapsed, flags)."""
     if not dist.is_initialized() or dist.get_world_size() == 1:
@@ -121,14 +131,31 @@ def gather_images(
     """
     if not dist.is_initialized() or dist.get_world_size() == 1:
         return shard
-    max_n = max(shard_sizes)
-    tail_shape = shard.shape[1:]
-    padded = torch.zeros(
-        (max_n, *tail_shape), dtype=shard.dtype, device=device
+    world = dist.get_world_size()
+    # Agree on the padded H/W first: a rank with an EMPTY shard (idle or
+    # failed) sized its placeholder from request math, while generating
+    # ranks' output size is latent-derived (H//f*f, floored hires scale) —
+    # with fractional hr_scale the two can differ and all_gather would
+    # error/hang on mismatched shapes. The consensus shape comes from any
+    # rank that actually produced images.
+    local = list(shard.shape) + [0] * (4 - shard.dim())
+    shp = torch.tensor(local[:4], dtype=torch.int64, device=device)
+    shapes = [torch.empty_like(shp) for _ in range(world)]
+    dist.all_gather(shapes, shp)
+    rows = [s.cpu().tolist() for s in shapes]
+    tail = next(
+        (tuple(r[1:]) for r in rows if r[0] > 0), tuple(shard.shape[1:])
     )
+    if shard.shape[0] > 0 and tuple(shard.shape[1:]) != tail:
+        raise RuntimeError(
+            f"image shard shape mismatch across ranks: {tuple(shard.shape[1:])}"
+            f" vs consensus {tail}"
+        )
+    max_n = max(max(shard_sizes), 1)
+    padded = torch.zeros((max_n, *tail), dtype=shard.dtype, device=device)
     if shard.shape[0] > 0:
         padded[: shard.shape[0]] = shard.to(device)
-    out = [torch.empty_like(padded) for _ in range(dist.get_world_size())]
+    out = [torch.empty_like(padded) for _ in range(world)]
     dist.all_gather(out, padded)
     parts = [o[: shard_sizes[r]] for r, o in enumerate(out)]
     return torch.cat(parts, dim=0)

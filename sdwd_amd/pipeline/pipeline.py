@@ -633,12 +633,19 @@ class StableDiffusionPipeline:
                 _inpaint_cache[key] = ex.to(self.dtype).contiguous()
             return _inpaint_cache[key]
 
+        # sdwui's s_min_uncond skips the uncond eval only on every OTHER
+        # model eval (CFGDenoiser.forward: `self.step % 2`); the counter
+        # ticks once per model_fn call like sdwui's self.step
+        smu_step = [0]
+
         def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:
             c_ctx, c_y, ws = _ctx_y_for(t)
             nb = x_in.shape[0]
+            smu_step[0] += 1
             if (
                 s_min_uncond > 0
                 and len(ws) == 1
+                and (smu_step[0] - 1) % 2
                 and sigma_for_t(t) < s_min_uncond
             ):
                 # sdwui s_min_uncond: at low noise the uncond eval barely
@@ -667,6 +674,7 @@ class StableDiffusionPipeline:
                 (nb * k1,), t, device=self.device, dtype=torch.float32,
             )
             xk = torch.cat([x_in] * k1, dim=0)
+            x_rep = xk  # pre-concat latents: _to_eps needs lat_c channels
             if is_inpaint_model:
                 ex = _inpaint_extra(x_in.shape[2], x_in.shape[3])
                 xk = torch.cat([xk, torch.cat([ex] * k1, dim=0)], dim=1)
@@ -677,7 +685,7 @@ class StableDiffusionPipeline:
                 eps = unet(xk, ts, c_ctx, y=c_y, control=ctrl)
             else:
                 eps = denoiser(xk, ts, c_ctx, c_y)
-            eps = _to_eps(eps, xk, t, pred_type)
+            eps = _to_eps(eps, x_rep, t, pred_type)
             parts = eps.chunk(k1, dim=0)
             from .. import ops as _ops
 

@@ -224,6 +224,26 @@ class TestRequeue:
             seeds.extend(j.seeds)
         assert seeds == victim.seeds
 
+    def test_requeue_seed_alignment_with_complementary(self):
+        """A requeue plan must total EXACTLY the failed batch even when
+        complement production is on and the survivors are heterogeneous —
+        a bonus job would read past failed_job.seeds (blank gallery rows)."""
+        world = make_world(
+            [60.0, 2.0, 60.0], complement_production=True
+        )
+        req = GenRequest(batch_size=12, seed=4242)
+        jobs = world.make_jobs(req)
+        victim = next(j for j in jobs if j.worker_label == "gpu0")
+        replacement = world.requeue_failed(victim, req)
+        assert sum(j.batch_size for j in replacement) == victim.batch_size
+        assert not any(j.complementary for j in replacement)
+        seeds = []
+        for j in sorted(replacement, key=lambda j: j.gallery_offset):
+            # every slot has a real seed (no short slices)
+            assert len(j.seeds) == j.batch_size
+            seeds.extend(j.seeds)
+        assert seeds == victim.seeds
+
     def test_recovery_flips_idle(self):
         world = make_world([30.0, 30.0])
         w = world.get_worker("gpu1")

@@ -151,6 +151,8 @@ def _dist_worker(rank, world_size, port, tmpdir, fail_rank):
                     )
             assert sizes.get("gpu1", 0) > sizes.get("gpu0", 0), res.job_summary
         torch.save(res.images, os.path.join(tmpdir, "gallery.pt"))
+        with open(os.path.join(tmpdir, "infotexts.txt"), "w") as fh:
+            fh.write("\x1e".join(res.infotexts))
     if fail_rank == -1:
         # model hot-swap rides the plan broadcast: every rank rebuilds
         res2 = eng.generate(
@@ -183,19 +185,56 @@ def _spawn(world_size, tmpdir, fail_rank=-1):
     )
 
 
+def _single_rank_reference(batch=4, seed=900, prompt="dist"):
+    """1-process LocalEngine gallery for image-for-image comparison."""
+    from sdwd_amd.parallel import LocalEngine, GenerationRequest
+
+    eng = LocalEngine(model="tiny", devices=["cpu"])
+    return eng.generate(
+        GenerationRequest(
+            prompt=prompt, batch_size=batch, width=64, height=64, steps=2,
+            seed=seed,
+        )
+    )
+
+
 @pytest.mark.timeout(300)
 class TestDistributedEngine:
     def test_two_rank_gloo(self, tmp_path):
         _spawn(2, str(tmp_path))
         gallery = torch.load(tmp_path / "gallery.pt")
         assert gallery.shape == (4, 64, 64, 3)
+        # per-image infotexts preserved across ranks (ref
+        # distributed.py:343-349): every slot carries a non-empty base
+        # infotext ahead of its worker label, including remote shards
+        infos = (tmp_path / "infotexts.txt").read_text().split("\x1e")
+        assert len(infos) == 4
+        for t in infos:
+            base, _, worker = t.rpartition("\nWorker: ")
+            assert worker.startswith("gpu"), t
+            assert "dist" in base, f"lost base infotext: {t!r}"
+        # the 2-rank gallery matches a 1-rank run image-for-image
+        ref = _single_rank_reference()
+        diff = (ref.images.float() - gallery.float()).abs()
+        assert diff.max() <= 1.0
 
     def test_two_rank_failure_recovery(self, tmp_path):
-        """Rank 1's shard fails; rank 0 re-runs it so the gallery is full."""
+        """Rank 1's shard fails; it is resharded over the survivors and the
+        recovered gallery still matches a 1-rank run image-for-image."""
         _spawn(2, str(tmp_path), fail_rank=1)
         gallery = torch.load(tmp_path / "gallery.pt")
         assert gallery.shape == (4, 64, 64, 3)
         assert all(gallery[i].float().std() > 0 for i in range(4))
+        ref = _single_rank_reference()
+        diff = (ref.images.float() - gallery.float()).abs()
+        assert diff.max() <= 1.0
+        # recovered slots keep their infotexts too
+        infos = (tmp_path / "infotexts.txt").read_text().split("\x1e")
+        assert len(infos) == 4
+        for t in infos:
+            base, _, worker = t.rpartition("\nWorker: ")
+            assert worker.startswith("gpu"), t
+            assert "dist" in base, f"lost infotext after recovery: {t!r}"
 
 
 def _thin_worker(rank, world_size, port, tmpdir):

@@ -743,19 +743,30 @@ class TestSamplerParams:
         b2 = pipe.generate(PipelineRequest(**base, s_churn=1.0)).images
         assert torch.equal(b, b2)
 
-    def test_s_min_uncond_skips_low_sigma_uncond(self, pipe):
+    def test_s_min_uncond_skips_alternating(self, pipe):
+        """sdwui skips the uncond eval only on every OTHER model eval
+        (CFGDenoiser.forward `self.step % 2`), not on every qualifying
+        step — the unet batch trace must read 2,1,2,1."""
         from sdwd_amd.pipeline import PipelineRequest
 
         base = dict(prompt="u", steps=4, width=64, height=64, seeds=[9])
-        a = pipe.generate(PipelineRequest(**base)).images
-        # huge threshold -> EVERY step skips uncond -> equivalent to cfg=1
-        b = pipe.generate(
-            PipelineRequest(**base, s_min_uncond=1e9)
-        ).images
-        c = pipe.generate(PipelineRequest(**base, cfg_scale=1.0)).images
+        batches = []
+        h = pipe.model.unet.register_forward_pre_hook(
+            lambda m, args: batches.append(args[0].shape[0])
+        )
+        try:
+            a = pipe.generate(PipelineRequest(**base)).images
+            normal = list(batches)
+            batches.clear()
+            b = pipe.generate(
+                PipelineRequest(**base, s_min_uncond=1e9)
+            ).images
+            skipped = list(batches)
+        finally:
+            h.remove()
+        assert normal == [2, 2, 2, 2]
+        assert skipped == [2, 1, 2, 1]
         assert not torch.equal(a, b)
-        diff = (b.float() - c.float()).abs()
-        assert diff.max() <= 1.0  # same compute, batch-blocking LSB only
 
     def test_churn_zero_is_default(self, pipe):
         from sdwd_amd.pipeline import PipelineRequest
@@ -798,6 +809,33 @@ class TestInpaintModel9ch:
         b = pipe.generate(req).images
         assert torch.equal(a, b)
         assert torch.isfinite(a.float()).all()
+
+    def test_vpred_inpaint_model(self):
+        """v-prediction + 9-channel inpaint model: _to_eps must combine the
+        PRE-concat 4-channel latents with the UNet output (regression: the
+        9-channel tensor crashed lincomb under CFG)."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        pipe = StableDiffusionPipeline("tiny-inpaint-v", device="cpu")
+        assert pipe.model.prediction_type == "v"
+        assert pipe.model.unet.cfg.in_channels == 9
+        res = pipe.generate(
+            PipelineRequest(prompt="v", steps=2, width=64, height=64,
+                            seeds=[5], cfg_scale=7.0)
+        )
+        assert res.images.shape == (1, 64, 64, 3)
+        assert torch.isfinite(res.images.float()).all()
+        # and with a mask + the uncond-skip knob active
+        init = torch.full((1, 64, 64, 3), 90, dtype=torch.uint8)
+        lat = pipe.encode_image(init, seeds=[6])
+        mask = torch.zeros(64, 64, dtype=torch.uint8)
+        mask[10:30, 10:30] = 255
+        res2 = pipe.generate(
+            PipelineRequest(prompt="vm", steps=3, width=64, height=64,
+                            seeds=[6], init_latents=lat, mask_image=mask,
+                            denoising_strength=0.7, s_min_uncond=1e9)
+        )
+        assert torch.isfinite(res2.images.float()).all()
 
     def test_hires_with_inpaint_model(self):
         from sdwd_amd.pipeline import PipelineRequest
