@@ -219,13 +219,316 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// v3: same wave/tile geometry as v2 (4 waves x 64 q-rows, KVB=64 staged in
+// LDS, swapped QK^T, in-register softmax, permlane P->A relayout) with the
+// guide's attention-ladder levers applied:
+//  * async-STAGE split (T14): next tile's K/V issued to registers BEFORE the
+//    compute phase; only the LDS-write latency sits at the tile seam.
+//  * scale folded into Q in log2 units: the hot loop's scores are already
+//    scale*log2(e)*(q.k) so softmax is bare v_exp2 with no per-score mul.
+//  * full-tile fast path: the (kv+kk<Sk) mask VALU only runs on the ragged
+//    last tile; staged loads clamp rows/cols instead of branching per lane.
+//  * defer-max (T13): the O/l rescale (16 shfl + 32 mul per q-subtile) is
+//    skipped while the running max grows < 2^THR; P is bounded by 2^THR
+//    which fp32 accumulation absorbs (THR=8 -> P<=256).
+//  * s_setprio(1) brackets around the MFMA clusters (T5).
+// ---------------------------------------------------------------------------
+template <int DPAD, int QS = (DPAD <= 64 ? 2 : 1)>
+__launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
+    const __hip_bfloat16 *__restrict__ Q, const __hip_bfloat16 *__restrict__ K,
+    const __hip_bfloat16 *__restrict__ V, __hip_bfloat16 *__restrict__ O,
+    int H, long Sq, long Sk, int D, float scale, AttnStrides st) {
+  constexpr int KVB = 64;
+  constexpr int PADK = 8;
+  constexpr int NC = DPAD / 16;
+  constexpr int DV = (DPAD + 31) / 32 * 32;
+  constexpr int ND = DV / 32;
+  constexpr int NG = KVB * (DPAD / 8);        // staged bf16x8 groups per tile
+  constexpr int NST = (NG + 255) / 256;       // groups per thread
+  constexpr float THR = 8.0f;                 // defer-max threshold (log2)
+  // async register staging costs NST*16 VGPRs; at DPAD>=144 that spills,
+  // so big head dims stage synchronously instead
+  constexpr bool ASYNC = (DPAD <= 128);
+
+  __shared__ __align__(16) __bf16 kt[KVB][DPAD + PADK];
+  __shared__ __align__(16) __bf16 vt[DV][KVB + PADK];
+
+  const int bh = blockIdx.y;
+  const int qblk = blockIdx.x;
+  const int bb = bh / H, hh = bh % H;
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int lq = lane % 32;
+  const int half = lane / 32;
+  const long q0 = (long)qblk * (QS * 128) + wid * (QS * 32);
+
+  const __hip_bfloat16 *Qb = Q + bb * st.qb + hh * st.qh;
+  const __hip_bfloat16 *Kb = K + bb * st.kb + hh * st.kh;
+  const __hip_bfloat16 *Vb = V + bb * st.vb + hh * st.vh;
+  __hip_bfloat16 *Ob = O + bb * st.ob + hh * st.oh;
+
+  // staged-load coordinates, fixed per thread (row/col-group of each piece)
+  int srow[NST], scol[NST];
+#pragma unroll
+  for (int s = 0; s < NST; ++s) {
+    const int idx = threadIdx.x + s * 256;
+    srow[s] = idx / (DPAD / 8);
+    scol[s] = idx % (DPAD / 8);
+  }
+  // clamp the d-group so the ragged head dim never reads past a row
+  const int dmax = D / 8 - 1;  // last full 8-group (D%8==0 guaranteed)
+
+  // Q fragments with scale*log2(e) folded in (one extra bf16 rounding)
+  const float l2scale = scale * 1.4426950408889634f;
+  bf16x8 qf[QS][NC];
+#pragma unroll
+  for (int qs = 0; qs < QS; ++qs) {
+    const long qq = q0 + qs * 32 + lq;
+    const long qrow = (qq < Sq) ? qq : (Sq - 1);
+#pragma unroll
+    for (int c = 0; c < NC; ++c) {
+      const int d0 = c * 16 + 8 * half;
+      bf16x8 qv = (bf16x8){};
+      if (d0 + 8 <= D) {
+        qv = *(const bf16x8 *)(Qb + qrow * st.qr + d0);
+#pragma unroll
+        for (int i = 0; i < 8; ++i)
+          qv[i] = (__bf16)((float)qv[i] * l2scale);
+      }
+      qf[qs][c] = qv;
+    }
+  }
+
+  f32x16 o[QS][ND];
+#pragma unroll
+  for (int qs = 0; qs < QS; ++qs)
+#pragma unroll
+    for (int d = 0; d < ND; ++d) o[qs][d] = (f32x16){};
+  float m[QS], l[QS];
+#pragma unroll
+  for (int qs = 0; qs < QS; ++qs) { m[qs] = -1e30f; l[qs] = 0.f; }
+
+  // zero vt's pad rows once (D..DV); never re-staged
+  for (int idx = threadIdx.x + (D / 8) * 8 * (KVB + PADK);
+       idx < DV * (KVB + PADK); idx += 256)
+    ((__bf16 *)vt)[idx] = (__bf16)0.0f;
+
+  // per-thread running K/V pointers: tile 0 clamped; full-tile advances are
+  // one uniform-delta 64-bit add instead of per-tile strided re-derivation
+  const __hip_bfloat16 *kp[NST], *vp[NST];
+#pragma unroll
+  for (int s = 0; s < NST; ++s) {
+    const long r = (srow[s] < Sk) ? srow[s] : (Sk - 1);
+    const int cg = (scol[s] <= dmax) ? scol[s] : dmax;
+    kp[s] = Kb + r * st.kr + cg * 8;
+    vp[s] = Vb + r * st.vr + cg * 8;
+  }
+  const long kdelta = KVB * st.kr, vdelta = KVB * st.vr;
+
+  // prologue: issue tile 0's loads (clamped; OOB keys masked at score time)
+  bf16x8 kst[ASYNC ? NST : 1], vst[ASYNC ? NST : 1];
+  if constexpr (ASYNC) {
+#pragma unroll
+    for (int s = 0; s < NST; ++s) {
+      if (threadIdx.x + s * 256 < NG) {
+        kst[s] = *(const bf16x8 *)kp[s];
+        vst[s] = *(const bf16x8 *)vp[s];
+      }
+    }
+  }
+
+  for (long kv = 0; kv < Sk; kv += KVB) {
+    __syncthreads();  // previous tile's LDS reads complete
+    if constexpr (ASYNC) {
+#pragma unroll
+      for (int s = 0; s < NST; ++s) {
+        if (threadIdx.x + s * 256 < NG) {
+          *(bf16x8 *)&kt[srow[s]][scol[s] * 8] = kst[s];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            vt[scol[s] * 8 + j][srow[s]] = vst[s][j];
+        }
+      }
+    } else {
+      // synchronous cooperative stage (VGPR-tight big head dims)
+#pragma unroll
+      for (int s = 0; s < NST; ++s) {
+        if (threadIdx.x + s * 256 < NG) {
+          const bf16x8 kvec = *(const bf16x8 *)kp[s];
+          const bf16x8 vvec = *(const bf16x8 *)vp[s];
+          *(bf16x8 *)&kt[srow[s]][scol[s] * 8] = kvec;
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            vt[scol[s] * 8 + j][srow[s]] = vvec[j];
+        }
+      }
+      // advance for the next tile (clamped re-derivation on the ragged one)
+      if (kv + 2 * KVB <= Sk) {
+#pragma unroll
+        for (int s = 0; s < NST; ++s) { kp[s] += kdelta; vp[s] += vdelta; }
+      } else if (kv + KVB < Sk) {
+#pragma unroll
+        for (int s = 0; s < NST; ++s) {
+          const long rr = kv + KVB + srow[s];
+          const long r = (rr < Sk) ? rr : (Sk - 1);
+          const int cg = (scol[s] <= dmax) ? scol[s] : dmax;
+          kp[s] = Kb + r * st.kr + cg * 8;
+          vp[s] = Vb + r * st.vr + cg * 8;
+        }
+      }
+    }
+    __syncthreads();
+    // async-STAGE: issue the NEXT tile's loads now; they complete under
+    // this tile's compute (the compiler places the vmcnt at first reuse)
+    if constexpr (ASYNC) {
+      if (kv + KVB < Sk) {
+        if (kv + 2 * KVB <= Sk) {
+#pragma unroll
+          for (int s = 0; s < NST; ++s) { kp[s] += kdelta; vp[s] += vdelta; }
+        } else {
+#pragma unroll
+          for (int s = 0; s < NST; ++s) {
+            const long rr = kv + KVB + srow[s];
+            const long r = (rr < Sk) ? rr : (Sk - 1);
+            const int cg = (scol[s] <= dmax) ? scol[s] : dmax;
+            kp[s] = Kb + r * st.kr + cg * 8;
+            vp[s] = Vb + r * st.vr + cg * 8;
+          }
+        }
+#pragma unroll
+        for (int s = 0; s < NST; ++s) {
+          if (threadIdx.x + s * 256 < NG) {
+            kst[s] = *(const bf16x8 *)kp[s];
+            vst[s] = *(const bf16x8 *)vp[s];
+          }
+        }
+      }
+    }
+
+    const bool fulltile = (kv + KVB <= Sk);
+#pragma unroll
+    for (int qs = 0; qs < QS; ++qs) {
+      f32x16 stile[2] = {(f32x16){}, (f32x16){}};
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+        for (int c = 0; c < NC; ++c) {
+          bf16x8 kf =
+              *(const bf16x8 *)&kt[sub * 32 + lq][c * 16 + 8 * half];
+          stile[sub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              kf, qf[qs][c], stile[sub], 0, 0, 0);
+        }
+      __builtin_amdgcn_s_setprio(0);
+
+      // softmax IN the accumulator registers (no p[] copy: VGPR budget)
+      float pmax = -1e30f;
+      if (fulltile) {
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+          for (int r = 0; r < 16; ++r)
+            pmax = fmaxf(pmax, stile[sub][r]);
+      } else {
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int kk = sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+            if (kv + kk >= Sk) stile[sub][r] = -1e30f;
+            pmax = fmaxf(pmax, stile[sub][r]);
+          }
+      }
+      pmax = fmaxf(pmax, __shfl_xor(pmax, 32, WAVE));
+
+      // defer-max: rescale only when some row's max grew past THR
+      if (!__all(pmax <= m[qs] + THR)) {
+        const float mnew = fmaxf(m[qs], pmax);
+        const float alpha = __builtin_amdgcn_exp2f(m[qs] - mnew);
+        l[qs] *= alpha;
+        m[qs] = mnew;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
+          const float a = __shfl(alpha, qrow, WAVE);
+#pragma unroll
+          for (int d = 0; d < ND; ++d) o[qs][d][r] *= a;
+        }
+      }
+      float rowsum = 0.f;
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub)
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          stile[sub][r] = __builtin_amdgcn_exp2f(stile[sub][r] - m[qs]);
+          rowsum += stile[sub][r];
+        }
+      rowsum += __shfl_xor(rowsum, 32, WAVE);
+      l[qs] += rowsum;
+
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+        unsigned pk[8];
+#pragma unroll
+        for (int t = 0; t < 8; ++t)
+          pk[t] = pack_bf16(stile[sub][2 * t], stile[sub][2 * t + 1]);
+        bf16x8 pa0, pa1;
+        {
+          auto r0 =
+              __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
+          auto r1 =
+              __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
+          unsigned frag[4] = {r0[0], r1[0], r0[1], r1[1]};
+          pa0 = *(bf16x8 *)frag;
+          auto r2 =
+              __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
+          auto r3 =
+              __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
+          unsigned frag1[4] = {r2[0], r3[0], r2[1], r3[1]};
+          pa1 = *(bf16x8 *)frag1;
+        }
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int d = 0; d < ND; ++d) {
+          bf16x8 v0 =
+              *(const bf16x8 *)&vt[d * 32 + lq][sub * 32 + 8 * half];
+          o[qs][d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              pa0, v0, o[qs][d], 0, 0, 0);
+          bf16x8 v1 =
+              *(const bf16x8 *)&vt[d * 32 + lq][sub * 32 + 16 + 8 * half];
+          o[qs][d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              pa1, v1, o[qs][d], 0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int qs = 0; qs < QS; ++qs) {
+    const float linv = 1.0f / fmaxf(l[qs], 1e-30f);
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
+      const float inv = __shfl(linv, qrow, WAVE);
+      const long qq = q0 + qs * 32 + qrow;
+      if (qq >= Sq) continue;
+#pragma unroll
+      for (int d = 0; d < ND; ++d)
+        if (d * 32 + lq < D)
+          Ob[qq * st.or_ + d * 32 + lq] = f2bf(o[qs][d][r] * inv);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // host-side dispatch (torch API lives in ext.hip which includes this file)
 // ---------------------------------------------------------------------------
 bool flash_supported(long d_head) {
   return d_head > 16 && d_head <= 192 && (d_head % 8) == 0;
 }
 
-#ifdef __HIP_PLATFORM_AMD__
+#if defined(__HIP_PLATFORM_AMD__) && !defined(SDWD_NO_TORCH)
 static inline long round16(long d) { return (d + 15) / 16 * 16; }
 
 // qkv layout: "bhsd" (q.size = [B,H,S,D]) or "bshd" ([B,S,H,D]); tensors
@@ -255,17 +558,34 @@ static torch::Tensor flash_attention_raw(torch::Tensor q, torch::Tensor k,
     st.ob = out.stride(0); st.oh = out.stride(1); st.or_ = out.stride(2);
   }
   const long DP = round16(D);
-  dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)(B * H));
   dim3 block(256);
   auto stream = cur_stream();
+  static const bool use_v2 = [] {
+    const char *e = getenv("SDWD_ATTN");
+    return e && strcmp(e, "v2") == 0;
+  }();
 
 #define LAUNCH_FLASH(DP_)                                                   \
-  hipLaunchKernelGGL(flash_fwd_bf16_kernel<DP_>, grid, block, 0, stream,    \
-                     (const __hip_bfloat16 *)q.data_ptr(),                  \
-                     (const __hip_bfloat16 *)k.data_ptr(),                  \
-                     (const __hip_bfloat16 *)v.data_ptr(),                  \
-                     (__hip_bfloat16 *)out.data_ptr(), (int)H, Sq, Sk,      \
-                     (int)D, (float)scale, st)
+  do {                                                                      \
+    if (use_v2) {                                                           \
+      dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)(B * H));           \
+      hipLaunchKernelGGL(flash_fwd_bf16_kernel<DP_>, grid, block, 0,        \
+                         stream, (const __hip_bfloat16 *)q.data_ptr(),      \
+                         (const __hip_bfloat16 *)k.data_ptr(),              \
+                         (const __hip_bfloat16 *)v.data_ptr(),              \
+                         (__hip_bfloat16 *)out.data_ptr(), (int)H, Sq, Sk,  \
+                         (int)D, (float)scale, st);                         \
+    } else {                                                                \
+      constexpr long ROWS_ = (DP_ <= 64 ? 256 : 128); /* QS*128 */          \
+      dim3 grid((unsigned)((Sq + ROWS_ - 1) / ROWS_), (unsigned)(B * H));   \
+      hipLaunchKernelGGL(flash_fwd_bf16_v3<DP_>, grid, block, 0, stream,    \
+                         (const __hip_bfloat16 *)q.data_ptr(),              \
+                         (const __hip_bfloat16 *)k.data_ptr(),              \
+                         (const __hip_bfloat16 *)v.data_ptr(),              \
+                         (__hip_bfloat16 *)out.data_ptr(), (int)H, Sq, Sk,  \
+                         (int)D, (float)scale, st);                         \
+    }                                                                       \
+  } while (0)
 
   switch (DP) {
     case 32: LAUNCH_FLASH(32); break;

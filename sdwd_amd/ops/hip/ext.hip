@@ -306,6 +306,15 @@ torch::Tensor probe_mfma16(torch::Tensor A, torch::Tensor B) {
   return C;
 }
 
+torch::Tensor probe_tr16(long stride_bytes, long base_bytes) {
+  auto out = torch::zeros(
+      {64, 4}, torch::dtype(torch::kFloat32).device(torch::kCUDA));
+  hipLaunchKernelGGL(probe_tr_b16, dim3(1), dim3(64), 0, cur_stream(),
+                     out.data_ptr<float>(), (int)stride_bytes,
+                     (int)base_bytes);
+  return out;
+}
+
 torch::Tensor attention_fwd_bshd(torch::Tensor q, torch::Tensor k,
                                  torch::Tensor v, double scale) {
   // [B,S,H,D] (the natural projection layout; avoids transposes entirely)
@@ -328,6 +337,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_fwd_bshd", &attention_fwd_bshd);
   m.def("probe_mfma32", &probe_mfma32);
   m.def("probe_mfma16", &probe_mfma16);
+  m.def("probe_tr16", &probe_tr16);
   m.def("flash_supported", &flash_supported);
   m.def("conv3x3_nhwc", &conv3x3_nhwc);
   m.def("conv3x3_supported", &conv3x3_supported);

@@ -27,6 +27,39 @@ __global__ void probe_mfma_32x32x16(const float *__restrict__ A,
   }
 }
 
+// ds_read_b64_tr_b16 semantics probe: LDS holds bf16 element e = e (its own
+// index); each lane issues one tr read at addr = base + lane*stride and the
+// 4 received bf16 are dumped per lane. The host decodes which LDS element
+// each (lane, j) slot received, settling the exact transpose mapping for
+// the attention V path (guide T10 has no worked example on disk).
+__global__ void probe_tr_b16(float *__restrict__ out, int stride_bytes,
+                             int base_bytes) {
+  __shared__ __align__(16) __bf16 buf[2048];
+  const int l = threadIdx.x;
+  for (int i = l; i < 2048; i += 64) buf[i] = (__bf16)(float)i;
+  __syncthreads();
+  unsigned addr =
+      (unsigned)(unsigned long long)(const char *)buf + base_bytes +
+      (unsigned)l * (unsigned)stride_bytes;
+  unsigned long long d = 0;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+      : "=v"(d)
+      : "v"(addr)
+      : "memory");
+  __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const unsigned short bits = (unsigned short)(d >> (16 * j));
+    union {
+      unsigned u;
+      float f;
+    } cv;
+    cv.u = ((unsigned)bits) << 16;  // bf16 -> f32
+    out[l * 4 + j] = cv.f;
+  }
+}
+
 __global__ void probe_mfma_16x16x32(const float *__restrict__ A,
                                     const float *__restrict__ B,
                                     float *__restrict__ C) {
