@@ -233,7 +233,7 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_kernel(
 //    which fp32 accumulation absorbs (THR=8 -> P<=256).
 //  * s_setprio(1) brackets around the MFMA clusters (T5).
 // ---------------------------------------------------------------------------
-template <int DPAD, int QS = (DPAD <= 64 ? 2 : 1)>
+template <int DPAD, int QS = (DPAD <= 96 ? 2 : 1)>
 __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
     const __hip_bfloat16 *__restrict__ Q, const __hip_bfloat16 *__restrict__ K,
     const __hip_bfloat16 *__restrict__ V, __hip_bfloat16 *__restrict__ O,
@@ -248,7 +248,7 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
   constexpr float THR = 8.0f;                 // defer-max threshold (log2)
   // async register staging costs NST*16 VGPRs; at DPAD>=144 that spills,
   // so big head dims stage synchronously instead
-  constexpr bool ASYNC = (DPAD <= 128);
+  constexpr bool ASYNC = (DPAD <= 64);
 
   __shared__ __align__(16) __bf16 kt[KVB][DPAD + PADK];
   __shared__ __align__(16) __bf16 vt[DV][KVB + PADK];
