@@ -243,15 +243,25 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
   constexpr int NC = DPAD / 16;
   constexpr int DV = (DPAD + 31) / 32 * 32;
   constexpr int ND = DV / 32;
-  constexpr int NG = KVB * (DPAD / 8);        // staged bf16x8 groups per tile
-  constexpr int NST = (NG + 255) / 256;       // groups per thread
+  constexpr int PADV = (DV == 128) ? 24 : 8;  // tr-read conflict-free pad
+  constexpr int RSV = DV + PADV;              // vt2 row stride (elements)
+  constexpr int NG = KVB * (DPAD / 8);        // bf16x8 pieces per operand
+  constexpr int NST = 2 * NG / 256;           // pieces/thread (exact: 16*DPAD%256==0)
   constexpr float THR = 8.0f;                 // defer-max threshold (log2)
-  // async register staging costs NST*16 VGPRs; at DPAD>=144 that spills,
+  // async register staging costs NST*8 VGPRs; at DPAD>=80 that spills,
   // so big head dims stage synchronously instead
   constexpr bool ASYNC = (DPAD <= 64);
 
+  // BOTH K and V are staged row-major with pure vectorized writes: the PV
+  // B-fragments come from ds_read_b64_tr_b16 hardware transpose reads
+  // (mapping decoded by tools/tr16_probe.py: with addr(l) =
+  // &vt2[k0 + ((l&15)>>2)][d0 + 4*(l&3)], lane l receives V[k0+j][d0+(l&15)]
+  // for j=0..3). This removes the 8-scalar-LDS-write V transpose AND makes
+  // the staging exactly NST pieces for EVERY thread (wave-balanced barriers;
+  // the old 1.5-piece split parked half the waves at each barrier - the
+  // round-1 PMC showed 52.6% SQ_WAIT_ANY at D=48).
   __shared__ __align__(16) __bf16 kt[KVB][DPAD + PADK];
-  __shared__ __align__(16) __bf16 vt[DV][KVB + PADK];
+  __shared__ __align__(16) __bf16 vt2[KVB][RSV];
 
   const int bh = blockIdx.y;
   const int qblk = blockIdx.x;
@@ -267,13 +277,22 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
   const __hip_bfloat16 *Vb = V + bb * st.vb + hh * st.vh;
   __hip_bfloat16 *Ob = O + bb * st.ob + hh * st.oh;
 
-  // staged-load coordinates, fixed per thread (row/col-group of each piece)
-  int srow[NST], scol[NST];
+  // staged-piece coordinates: pieces [0,NG) are K rows, [NG,2NG) V rows;
+  // NG is a multiple of 64 so the K/V split is wave-uniform and every
+  // piece costs the same (one b128 write). Coordinates are recomputed on
+  // the rare clamped paths instead of held live (VGPR budget).
+  const auto piece = [&](int s, int &r, int &c, bool &v) {
+    const int pc = threadIdx.x + s * 256;
+    const int q = (pc < NG) ? pc : pc - NG;
+    v = pc >= NG;
+    r = q / (DPAD / 8);
+    c = q % (DPAD / 8);
+  };
+  int eoff[NST];
 #pragma unroll
   for (int s = 0; s < NST; ++s) {
-    const int idx = threadIdx.x + s * 256;
-    srow[s] = idx / (DPAD / 8);
-    scol[s] = idx % (DPAD / 8);
+    int r, c; bool v; piece(s, r, c, v);
+    eoff[s] = v ? (r * RSV + c * 8) : (r * (DPAD + PADK) + c * 8);
   }
   // clamp the d-group so the ragged head dim never reads past a row
   const int dmax = D / 8 - 1;  // last full 8-group (D%8==0 guaranteed)
@@ -299,6 +318,14 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
     }
   }
 
+  // per-lane tr-read base LDS byte address (probe-decoded mapping): the
+  // per-read row/col deltas are added as small unsigned offsets
+  const unsigned vbase =
+      (unsigned)(unsigned long long)(const char *)&vt2[0][0] +
+      ((((lane & 15) >> 2) + 8 * half) * RSV + 4 * (lane & 3) +
+       ((lane >> 4) & 1) * 16) *
+          2;
+
   f32x16 o[QS][ND];
 #pragma unroll
   for (int qs = 0; qs < QS; ++qs)
@@ -308,33 +335,23 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
 #pragma unroll
   for (int qs = 0; qs < QS; ++qs) { m[qs] = -1e30f; l[qs] = 0.f; }
 
-  // zero vt's pad rows once (D..DV); never re-staged
-  for (int idx = threadIdx.x + (D / 8) * 8 * (KVB + PADK);
-       idx < DV * (KVB + PADK); idx += 256)
-    ((__bf16 *)vt)[idx] = (__bf16)0.0f;
-
-  // per-thread running K/V pointers: tile 0 clamped; full-tile advances are
-  // one uniform-delta 64-bit add instead of per-tile strided re-derivation
-  const __hip_bfloat16 *kp[NST], *vp[NST];
+  // per-thread running piece pointers: tile 0 clamped; full-tile advances
+  // are one uniform-delta 64-bit add instead of strided re-derivation
+  const __hip_bfloat16 *pp[NST];
 #pragma unroll
   for (int s = 0; s < NST; ++s) {
-    const long r = (srow[s] < Sk) ? srow[s] : (Sk - 1);
-    const int cg = (scol[s] <= dmax) ? scol[s] : dmax;
-    kp[s] = Kb + r * st.kr + cg * 8;
-    vp[s] = Vb + r * st.vr + cg * 8;
+    int sr, sc; bool v; piece(s, sr, sc, v);
+    const long r = (sr < Sk) ? sr : (Sk - 1);
+    const int cg = (sc <= dmax) ? sc : dmax;
+    pp[s] = (v ? Vb + r * st.vr : Kb + r * st.kr) + cg * 8;
   }
   const long kdelta = KVB * st.kr, vdelta = KVB * st.vr;
 
   // prologue: issue tile 0's loads (clamped; OOB keys masked at score time)
-  bf16x8 kst[ASYNC ? NST : 1], vst[ASYNC ? NST : 1];
+  bf16x8 stg[ASYNC ? NST : 1];
   if constexpr (ASYNC) {
 #pragma unroll
-    for (int s = 0; s < NST; ++s) {
-      if (threadIdx.x + s * 256 < NG) {
-        kst[s] = *(const bf16x8 *)kp[s];
-        vst[s] = *(const bf16x8 *)vp[s];
-      }
-    }
+    for (int s = 0; s < NST; ++s) stg[s] = *(const bf16x8 *)pp[s];
   }
 
   for (long kv = 0; kv < Sk; kv += KVB) {
@@ -342,38 +359,29 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
     if constexpr (ASYNC) {
 #pragma unroll
       for (int s = 0; s < NST; ++s) {
-        if (threadIdx.x + s * 256 < NG) {
-          *(bf16x8 *)&kt[srow[s]][scol[s] * 8] = kst[s];
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            vt[scol[s] * 8 + j][srow[s]] = vst[s][j];
-        }
+        const bool v = threadIdx.x + s * 256 >= NG;
+        *(bf16x8 *)((v ? &vt2[0][0] : &kt[0][0]) + eoff[s]) = stg[s];
       }
     } else {
       // synchronous cooperative stage (VGPR-tight big head dims)
 #pragma unroll
       for (int s = 0; s < NST; ++s) {
-        if (threadIdx.x + s * 256 < NG) {
-          const bf16x8 kvec = *(const bf16x8 *)kp[s];
-          const bf16x8 vvec = *(const bf16x8 *)vp[s];
-          *(bf16x8 *)&kt[srow[s]][scol[s] * 8] = kvec;
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            vt[scol[s] * 8 + j][srow[s]] = vvec[j];
-        }
+        const bool v = threadIdx.x + s * 256 >= NG;
+        *(bf16x8 *)((v ? &vt2[0][0] : &kt[0][0]) + eoff[s]) =
+            *(const bf16x8 *)pp[s];
       }
-      // advance for the next tile (clamped re-derivation on the ragged one)
       if (kv + 2 * KVB <= Sk) {
 #pragma unroll
-        for (int s = 0; s < NST; ++s) { kp[s] += kdelta; vp[s] += vdelta; }
+        for (int s = 0; s < NST; ++s)
+          pp[s] += (threadIdx.x + s * 256 >= NG) ? vdelta : kdelta;
       } else if (kv + KVB < Sk) {
 #pragma unroll
         for (int s = 0; s < NST; ++s) {
-          const long rr = kv + KVB + srow[s];
+          int sr, sc; bool v; piece(s, sr, sc, v);
+          const long rr = kv + KVB + sr;
           const long r = (rr < Sk) ? rr : (Sk - 1);
-          const int cg = (scol[s] <= dmax) ? scol[s] : dmax;
-          kp[s] = Kb + r * st.kr + cg * 8;
-          vp[s] = Vb + r * st.vr + cg * 8;
+          const int cg = (sc <= dmax) ? sc : dmax;
+          pp[s] = (v ? Vb + r * st.vr : Kb + r * st.kr) + cg * 8;
         }
       }
     }
@@ -384,24 +392,20 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
       if (kv + KVB < Sk) {
         if (kv + 2 * KVB <= Sk) {
 #pragma unroll
-          for (int s = 0; s < NST; ++s) { kp[s] += kdelta; vp[s] += vdelta; }
+          for (int s = 0; s < NST; ++s)
+            pp[s] += (threadIdx.x + s * 256 >= NG) ? vdelta : kdelta;
         } else {
 #pragma unroll
           for (int s = 0; s < NST; ++s) {
-            const long rr = kv + KVB + srow[s];
+            int sr, sc; bool v; piece(s, sr, sc, v);
+            const long rr = kv + KVB + sr;
             const long r = (rr < Sk) ? rr : (Sk - 1);
-            const int cg = (scol[s] <= dmax) ? scol[s] : dmax;
-            kp[s] = Kb + r * st.kr + cg * 8;
-            vp[s] = Vb + r * st.vr + cg * 8;
+            const int cg = (sc <= dmax) ? sc : dmax;
+            pp[s] = (v ? Vb + r * st.vr : Kb + r * st.kr) + cg * 8;
           }
         }
 #pragma unroll
-        for (int s = 0; s < NST; ++s) {
-          if (threadIdx.x + s * 256 < NG) {
-            kst[s] = *(const bf16x8 *)kp[s];
-            vst[s] = *(const bf16x8 *)vp[s];
-          }
-        }
+        for (int s = 0; s < NST; ++s) stg[s] = *(const bf16x8 *)pp[s];
       }
     }
 
@@ -472,34 +476,61 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
 #pragma unroll
         for (int t = 0; t < 8; ++t)
           pk[t] = pack_bf16(stile[sub][2 * t], stile[sub][2 * t + 1]);
-        bf16x8 pa0, pa1;
+        bf16x8 pa[2];
         {
           auto r0 =
               __builtin_amdgcn_permlane32_swap(pk[0], pk[2], false, false);
           auto r1 =
               __builtin_amdgcn_permlane32_swap(pk[1], pk[3], false, false);
           unsigned frag[4] = {r0[0], r1[0], r0[1], r1[1]};
-          pa0 = *(bf16x8 *)frag;
+          pa[0] = *(bf16x8 *)frag;
           auto r2 =
               __builtin_amdgcn_permlane32_swap(pk[4], pk[6], false, false);
           auto r3 =
               __builtin_amdgcn_permlane32_swap(pk[5], pk[7], false, false);
           unsigned frag1[4] = {r2[0], r3[0], r2[1], r3[1]};
-          pa1 = *(bf16x8 *)frag1;
+          pa[1] = *(bf16x8 *)frag1;
         }
-        __builtin_amdgcn_s_setprio(1);
+        // PV with tr-read V fragments: per 16-key chunk, 2 transpose
+        // reads per d-tile deliver V[k..k+7][dcol] straight from the
+        // row-major tile (guide T10; mapping from tools/tr16_probe.py)
 #pragma unroll
-        for (int d = 0; d < ND; ++d) {
-          bf16x8 v0 =
-              *(const bf16x8 *)&vt[d * 32 + lq][sub * 32 + 8 * half];
-          o[qs][d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              pa0, v0, o[qs][d], 0, 0, 0);
-          bf16x8 v1 =
-              *(const bf16x8 *)&vt[d * 32 + lq][sub * 32 + 16 + 8 * half];
-          o[qs][d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              pa1, v1, o[qs][d], 0, 0, 0);
+        for (int kc = 0; kc < 2; ++kc) {
+#pragma unroll
+          for (int db = 0; db < ND; db += 2) {  // <=2 d-tiles per wait:
+            constexpr int DB = 2;               // bounds tr live ranges
+            unsigned long long tv[DB][2];
+#pragma unroll
+            for (int dd = 0; dd < DB; ++dd) {
+              const int d = db + dd;
+              if (d >= ND) break;
+              const unsigned a0 =
+                  vbase +
+                  (unsigned)(((sub * 32 + kc * 16) * RSV + d * 32) * 2);
+              asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+                           "ds_read_b64_tr_b16 %1, %3"
+                           : "=&v"(tv[dd][0]), "=&v"(tv[dd][1])
+                           : "v"(a0), "v"(a0 + (unsigned)(8 * RSV)));
+            }
+            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+            __builtin_amdgcn_sched_barrier(0);
+            __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+            for (int dd = 0; dd < DB; ++dd) {
+              const int d = db + dd;
+              if (d >= ND) break;
+              union {
+                unsigned long long u[2];
+                bf16x8 v;
+              } cv;
+              cv.u[0] = tv[dd][0];
+              cv.u[1] = tv[dd][1];
+              o[qs][d] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                  pa[kc], cv.v, o[qs][d], 0, 0, 0);
+            }
+            __builtin_amdgcn_s_setprio(0);
+          }
         }
-        __builtin_amdgcn_s_setprio(0);
       }
     }
   }
@@ -567,7 +598,7 @@ static torch::Tensor flash_attention_raw(torch::Tensor q, torch::Tensor k,
 
 #define LAUNCH_FLASH(DP_)                                                   \
   do {                                                                      \
-    if (use_v2) {                                                           \
+    if (use_v2 || DP_ > 64) { /* v3 wins measured only at DPAD<=64 */      \
       dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)(B * H));           \
       hipLaunchKernelGGL(flash_fwd_bf16_kernel<DP_>, grid, block, 0,        \
                          stream, (const __hip_bfloat16 *)q.data_ptr(),      \
@@ -576,7 +607,7 @@ static torch::Tensor flash_attention_raw(torch::Tensor q, torch::Tensor k,
                          (__hip_bfloat16 *)out.data_ptr(), (int)H, Sq, Sk,  \
                          (int)D, (float)scale, st);                         \
     } else {                                                                \
-      constexpr long ROWS_ = (DP_ <= 64 ? 256 : 128); /* QS*128 */          \
+      constexpr long ROWS_ = (DP_ <= 96 ? 256 : 128); /* QS*128 */          \
       dim3 grid((unsigned)((Sq + ROWS_ - 1) / ROWS_), (unsigned)(B * H));   \
       hipLaunchKernelGGL(flash_fwd_bf16_v3<DP_>, grid, block, 0, stream,    \
                          (const __hip_bfloat16 *)q.data_ptr(),              \
