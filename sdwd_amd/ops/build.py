@@ -20,6 +20,12 @@ OUT_PREFIX = os.path.join(HERE, "_sdwd_hip")
 def build(verbose: bool = False) -> str:
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
     os.makedirs(BUILD_DIR, exist_ok=True)
+    # torch's hipify caches hip/ext_hip.hip keyed on ext.hip alone; edits to
+    # the #included kernel files (attention.hip, conv.hip, ...) would ship a
+    # STALE build silently. Nuke the artifact so hipify + ninja re-run.
+    art = os.path.join(HERE, "hip", "ext_hip.hip")
+    if os.path.exists(art):
+        os.remove(art)
     from torch.utils.cpp_extension import load
 
     mod = load(
