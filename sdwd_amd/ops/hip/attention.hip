@@ -260,8 +260,11 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
   // the staging exactly NST pieces for EVERY thread (wave-balanced barriers;
   // the old 1.5-piece split parked half the waves at each barrier - the
   // round-1 PMC showed 52.6% SQ_WAIT_ANY at D=48).
-  __shared__ __align__(16) __bf16 kt[KVB][DPAD + PADK];
-  __shared__ __align__(16) __bf16 vt2[KVB][RSV];
+  // double-buffered tiles: tile i+1 is written into buf i^1 while buf i
+  // is computed, so ONE barrier per tile suffices (write(i) only needs
+  // compute(i-2) complete, which the barrier of iteration i-1 guarantees)
+  __shared__ __align__(16) __bf16 kt[2][KVB][DPAD + PADK];
+  __shared__ __align__(16) __bf16 vt2[2][KVB][RSV];
 
   const int bh = blockIdx.y;
   const int qblk = blockIdx.x;
@@ -321,7 +324,7 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
   // per-lane tr-read base LDS byte address (probe-decoded mapping): the
   // per-read row/col deltas are added as small unsigned offsets
   const unsigned vbase =
-      (unsigned)(unsigned long long)(const char *)&vt2[0][0] +
+      (unsigned)(unsigned long long)(const char *)&vt2[0][0][0] +
       ((((lane & 15) >> 2) + 8 * half) * RSV + 4 * (lane & 3) +
        ((lane >> 4) & 1) * 16) *
           2;
@@ -354,20 +357,21 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
     for (int s = 0; s < NST; ++s) stg[s] = *(const bf16x8 *)pp[s];
   }
 
-  for (long kv = 0; kv < Sk; kv += KVB) {
-    __syncthreads();  // previous tile's LDS reads complete
+  int cur = 0;
+  for (long kv = 0; kv < Sk; kv += KVB, cur ^= 1) {
     if constexpr (ASYNC) {
 #pragma unroll
       for (int s = 0; s < NST; ++s) {
         const bool v = threadIdx.x + s * 256 >= NG;
-        *(bf16x8 *)((v ? &vt2[0][0] : &kt[0][0]) + eoff[s]) = stg[s];
+        *(bf16x8 *)((v ? &vt2[cur][0][0] : &kt[cur][0][0]) + eoff[s]) =
+            stg[s];
       }
     } else {
       // synchronous cooperative stage (VGPR-tight big head dims)
 #pragma unroll
       for (int s = 0; s < NST; ++s) {
         const bool v = threadIdx.x + s * 256 >= NG;
-        *(bf16x8 *)((v ? &vt2[0][0] : &kt[0][0]) + eoff[s]) =
+        *(bf16x8 *)((v ? &vt2[cur][0][0] : &kt[cur][0][0]) + eoff[s]) =
             *(const bf16x8 *)pp[s];
       }
       if (kv + 2 * KVB <= Sk) {
@@ -419,7 +423,7 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
 #pragma unroll
         for (int c = 0; c < NC; ++c) {
           bf16x8 kf =
-              *(const bf16x8 *)&kt[sub * 32 + lq][c * 16 + 8 * half];
+              *(const bf16x8 *)&kt[cur][sub * 32 + lq][c * 16 + 8 * half];
           stile[sub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               kf, qf[qs][c], stile[sub], 0, 0, 0);
         }
@@ -505,7 +509,7 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
               const int d = db + dd;
               if (d >= ND) break;
               const unsigned a0 =
-                  vbase +
+                  vbase + (unsigned)(cur * (KVB * RSV * 2)) +
                   (unsigned)(((sub * 32 + kc * 16) * RSV + d * 32) * 2);
               asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
                            "ds_read_b64_tr_b16 %1, %3"
