@@ -37,9 +37,7 @@ typedef __attribute__((ext_vector_type(4))) float f32x4c;
 // read's 16-lane group touches 16 consecutive rows at one chunk; rows of
 // equal parity would collide every 2 rows at a 32-dword stride, so spread
 // by (row>>1)&7.
-__device__ __forceinline__ int cswz(int row, int chunk) {
-  return chunk ^ ((row >> 1) & 7);
-}
+// cswz() lives in conv_v4.hip (included first)
 
 // BN = 128 (default) or 64 (exact tiling for Cout % 128 == 64, e.g. the
 // SD1.5 320-channel level: 5 exact 64-col tiles instead of 3 x 128 with a
@@ -52,8 +50,9 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
     const __hip_bfloat16 *__restrict__ Res, // [N,Ho,Wo,Cout] or null
     const __hip_bfloat16 *__restrict__ CB,  // [N,Cout] or null
     const __hip_bfloat16 *__restrict__ Zero,// >=16B of zeros
-    __hip_bfloat16 *__restrict__ Y,         // [N,Ho,Wo,Cout]
-    int Nn, int H, int W, int Cin, int Cout, int Ho, int Wo, int stride) {
+    __hip_bfloat16 *__restrict__ Y,         // [N,Ho,Wo,ldY] (+col offset)
+    int Nn, int H, int W, int Cin, int Cout, int Ho, int Wo, int stride,
+    int ldY) {
   // one __shared__ object: [buf0: A(8K elems) B(8K)][buf1: A B]
   __shared__ __align__(16) __bf16 smem[4 * TILE_ELEMS];
 
@@ -202,10 +201,10 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
         float v = acc[i][j][r] + bv;
         if (HAS_CB) {
           const int ni = (int)(m / ((long)Ho * Wo));
-          v += (float)CB[(long)ni * Cout + co];
+          v += (float)CB[(long)ni * ldY + co];
         }
-        if (HAS_RES) v += (float)Res[m * Cout + co];
-        Y[m * Cout + co] = f2bf(v);
+        if (HAS_RES) v += (float)Res[m * ldY + co];
+        Y[m * ldY + co] = f2bf(v);
       }
     }
   }
@@ -293,32 +292,61 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
                        Ho, Wo, (int)stride);
     return y;
   }
-  // BN=64 exact tiling for Cout=320 measured SLOWER (460 vs 557 TF: the
-  // halved B-reuse doubles A traffic and drops per-block efficiency more
-  // than the 17% masked-FLOP tail costs); template retained, 128 always.
-  const bool bn64 = false;
-#define PICK(B_, R_, C_)                                              \
-  (bn64 ? conv3x3_nhwc_bf16_kernel<B_, R_, C_, 64>                    \
-        : conv3x3_nhwc_bf16_kernel<B_, R_, C_, 128>)
-  auto kern =
-      has_b ? (has_r ? (has_cb ? PICK(true, true, true)
-                               : PICK(true, true, false))
-                     : (has_cb ? PICK(true, false, true)
-                               : PICK(true, false, false)))
-            : (has_r ? (has_cb ? PICK(false, true, true)
-                               : PICK(false, true, false))
-                     : (has_cb ? PICK(false, false, true)
-                               : PICK(false, false, false)));
+  // v4: the 256x256 8-phase deep pipeline serves full 256-column tiles;
+  // the v2 128-tile kernel covers the Cout remainder (e.g. 320 = 256+64).
+  // SDWD_CONV=v2 forces the round-1 kernel everywhere (A/B).
+  static const bool v2only = [] {
+    const char *e = getenv("SDWD_CONV");
+    return e && strcmp(e, "v2") == 0;
+  }();
+  const long mt4 = (M + V4_BM - 1) / V4_BM;
+  const long nfull = v2only ? 0 : Cout / 256;
+  const int rem = (int)(Cout - nfull * 256);
+  if (nfull > 0) {
+    dim3 g4((unsigned)nfull, (unsigned)std::min<long>(mt4, 32768),
+            (unsigned)((mt4 + 32767) / 32768));
+    dim3 b4(512);
+#define PICK4(B_, R_, C_) conv3x3_v4_kernel<B_, R_, C_>
+    auto k4 = has_b ? (has_r ? (has_cb ? PICK4(true, true, true)
+                                       : PICK4(true, true, false))
+                             : (has_cb ? PICK4(true, false, true)
+                                       : PICK4(true, false, false)))
+                    : (has_r ? (has_cb ? PICK4(false, true, true)
+                                       : PICK4(false, true, false))
+                             : (has_cb ? PICK4(false, false, true)
+                                       : PICK4(false, false, false)));
+#undef PICK4
+    hipLaunchKernelGGL(k4, g4, b4, 0, stream,
+                       (const __hip_bfloat16 *)x.data_ptr(),
+                       (const __hip_bfloat16 *)w_prep.data_ptr(), bptr,
+                       rptr, cbptr,
+                       (const __hip_bfloat16 *)zero_page.data_ptr(),
+                       (__hip_bfloat16 *)y.data_ptr(), N, H, W, Cin,
+                       (int)(nfull * 256), Cout, Ho, Wo, (int)stride);
+  }
+  if (rem > 0) {
+    const long co0 = nfull * 256;
+#define PICK(B_, R_, C_) conv3x3_nhwc_bf16_kernel<B_, R_, C_, 128>
+    auto kern =
+        has_b ? (has_r ? (has_cb ? PICK(true, true, true)
+                                 : PICK(true, true, false))
+                       : (has_cb ? PICK(true, false, true)
+                                 : PICK(true, false, false)))
+              : (has_r ? (has_cb ? PICK(false, true, true)
+                                 : PICK(false, true, false))
+                       : (has_cb ? PICK(false, false, true)
+                                 : PICK(false, false, false)));
 #undef PICK
-  if (bn64)
-    grid.x = (unsigned)((Cout + 63) / 64);
-  hipLaunchKernelGGL(kern, grid, block, 0, stream,
-                     (const __hip_bfloat16 *)x.data_ptr(),
-                     (const __hip_bfloat16 *)w_prep.data_ptr(), bptr, rptr,
-                     cbptr,
-                     (const __hip_bfloat16 *)zero_page.data_ptr(),
-                     (__hip_bfloat16 *)y.data_ptr(), N, H, W, Cin, Cout, Ho,
-                     Wo, (int)stride);
+    dim3 g2((unsigned)((rem + CONV_BN - 1) / CONV_BN), grid.y, grid.z);
+    hipLaunchKernelGGL(
+        kern, g2, block, 0, stream, (const __hip_bfloat16 *)x.data_ptr(),
+        (const __hip_bfloat16 *)w_prep.data_ptr() + co0 * 9 * Cin,
+        bptr ? bptr + co0 : nullptr,
+        rptr ? rptr + co0 : nullptr, cbptr ? cbptr + co0 : nullptr,
+        (const __hip_bfloat16 *)zero_page.data_ptr(),
+        (__hip_bfloat16 *)y.data_ptr() + co0, N, H, W, Cin, rem, Ho, Wo,
+        (int)stride, Cout);
+  }
   return y;
 }
 #endif

@@ -1,0 +1,231 @@
+// Implicit-GEMM 3x3 conv, v4: the guide's 256x256 8-phase deep-pipeline
+// structure (cdna_hip_programming.md §5 '256² 8-phase template', T2+T3+T4+T5)
+// adapted to the im2col A operand.
+//
+//  * 512 threads (8 waves as 2M x 4N), BM=BN=256, BK=64; per-wave output
+//    128x64 = 8x4 fragments of mfma_f32_16x16x32_bf16, acc = 128 VGPRs.
+//  * TWO whole-tile LDS buffers (A 32K + B 32K each, 128 KiB total, one
+//    block/CU): while tile t streams out of buf[t&1], tile t+1 is staged
+//    into buf[~t&1] by 2 global_load_lds per thread per phase (4 phases
+//    per K-tile, one C-quadrant each = 16 MFMAs).
+//  * counted `s_waitcnt vmcnt(2)` ONCE per K-tile (phase 3), raw
+//    s_barrier everywhere (a __syncthreads would drain the glds queue);
+//    hipcc's own lgkm counting orders the ds_read->MFMA chain.
+//  * LDS rows are 128 B; the 8 16-B chunks XOR-swizzled by (row>>1)&7 on
+//    the glds SOURCE address (rule 21) - b128 fragment reads conflict-free.
+//  * bias / per-sample channel bias / residual fused in the epilogue.
+//
+// Covers full 256-column tiles only; the host launches the v2 kernel on
+// the Cout remainder (Cout=320 -> 256 here + 64 there).
+#include "common.h"
+
+// XOR-permute the 8 16-byte chunks of a 128-byte LDS row (shared with the
+// v2 kernel in conv.hip): b128 fragment reads stay bank-conflict-free.
+__device__ __forceinline__ int cswz(int row, int chunk) {
+  return chunk ^ ((row >> 1) & 7);
+}
+
+#define V4_BM 256
+#define V4_BN 256
+#define V4_BK 64
+#define V4_ATILE (V4_BM * V4_BK)  // elements per A buffer
+
+template <bool HAS_BIAS, bool HAS_RES, bool HAS_CB>
+__launch_bounds__(512, 2) __global__ void conv3x3_v4_kernel(
+    const __hip_bfloat16 *__restrict__ X,   // [N,H,W,Cin]
+    const __hip_bfloat16 *__restrict__ Wt,  // [Cout,3,3,Cin] (+col offset)
+    const float *__restrict__ bias,         // [cols] or null (+col offset)
+    const __hip_bfloat16 *__restrict__ Res, // [M,ldY] or null (+col offset)
+    const __hip_bfloat16 *__restrict__ CB,  // [N,ldY] or null (+col offset)
+    const __hip_bfloat16 *__restrict__ Zero,
+    __hip_bfloat16 *__restrict__ Y,         // [M,ldY] (+col offset)
+    int Nn, int H, int W, int Cin, int Cols, int ldY, int Ho, int Wo,
+    int stride) {
+  __shared__ __align__(16) __bf16 smem[4 * V4_ATILE];  // [buf][A|B]
+
+  const long M = (long)Nn * Ho * Wo;
+  const long m0 =
+      ((long)blockIdx.y + (long)blockIdx.z * 32768) * V4_BM;
+  const int n0 = blockIdx.x * V4_BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid % WAVE;
+  const int wid = tid / WAVE;
+  const int wm = (wid >> 2) * 128;       // wave row base (2 rows of waves)
+  const int wn = (wid & 3) * 64;         // wave col base (4 cols of waves)
+
+  // staged-piece bookkeeping: piece p = tid + 512*i covers LDS bytes
+  // p*16 of the A (i<4) or B (i-4) tile; row = p/8, chunk = p%8.
+  // A rows are t/8 + 64*i; B rows likewise.
+  const int prow = tid / 8;
+  const int schunk = tid % 8;
+  long abase[4];
+  int hs[4], ws[4];
+  int bco[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const long m = m0 + prow + 64 * i;
+    const long mm = (m < M) ? m : (M - 1);
+    const int n_img = (int)(mm / ((long)Ho * Wo));
+    const int rem = (int)(mm % ((long)Ho * Wo));
+    hs[i] = (rem / Wo) * stride;
+    ws[i] = (rem % Wo) * stride;
+    abase[i] = (((long)n_img * H + hs[i]) * W + ws[i]) * Cin;
+    bco[i] = n0 + prow + 64 * i;
+  }
+
+  const int kc_per_plane = Cin / V4_BK;
+  const int NT = 9 * kc_per_plane;
+
+  // stage 2 pieces (one phase's share) of tile t into buffer b. Piece
+  // order is chosen so the LAST-issued pieces (left in flight past the
+  // per-tile counted waits) are the LATEST-read ones: B pieces (read from
+  // the next tile's phase 0 by all waves) go first, A pieces for the
+  // mh=0 phases next, the mh=1 pieces (first read at phase 2) last.
+  //   ph0 -> B0,B1   ph1 -> B2,B3   ph2 -> A0,A2   ph3 -> A1,A3
+  auto stage2 = [&](int t, int b, int ph) {
+    const int plane = t / kc_per_plane;
+    const int kc = t % kc_per_plane;
+    const int dy = plane / 3 - 1, dx = plane % 3 - 1;
+    const long poff = ((long)dy * W + dx) * Cin + (long)kc * V4_BK;
+    __bf16 *abuf = smem + b * (2 * V4_ATILE);
+    __bf16 *bbuf = abuf + V4_ATILE;
+    const int sc = schunk;
+    constexpr int PIECES[4][2] = {{4, 5}, {6, 7}, {0, 2}, {1, 3}};
+#pragma unroll
+    for (int k = 0; k < 2; ++k) {
+      const int i = PIECES[ph][k];        // piece index 0..7
+      if (i < 4) {
+        const int row = prow + 64 * i;
+        const int swz = cswz(row, sc);
+        const bool av = (hs[i] + dy >= 0) && (hs[i] + dy < H) &&
+                        (ws[i] + dx >= 0) && (ws[i] + dx < W);
+        const __hip_bfloat16 *asrc =
+            av ? (X + abase[i] + poff + swz * 8) : Zero;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int *)asrc,
+            (__attribute__((address_space(3))) unsigned int
+                 *)(abuf + (long)i * 4096 + (long)tid * 8),
+            16, 0, 0);
+      } else {
+        const int bi = i - 4;
+        const int row = prow + 64 * bi;
+        const int swz = cswz(row, sc);
+        const bool bv = bco[bi] < Cols;
+        const __hip_bfloat16 *bsrc =
+            bv ? (Wt + (long)bco[bi] * 9 * Cin + plane * Cin +
+                  kc * V4_BK + swz * 8)
+               : Zero;
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int *)bsrc,
+            (__attribute__((address_space(3))) unsigned int
+                 *)(bbuf + (long)bi * 4096 + (long)tid * 8),
+            16, 0, 0);
+      }
+    }
+  };
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){};
+
+  // prologue: stage tile 0 fully (8 pieces), then begin the loop with
+  // tile 1 staged phase-by-phase while tile 0 is computed
+#pragma unroll
+  for (int ph = 0; ph < 4; ++ph) stage2(0, 0, ph);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  const int l16 = lane % 16;
+  const int kq4 = lane / 16;  // k quarter (8 elems each) of this lane
+
+  for (int t = 0; t < NT; ++t) {
+    const __bf16 *abuf = smem + (t & 1) * (2 * V4_ATILE);
+    const __bf16 *bbuf = abuf + V4_ATILE;
+#pragma unroll
+    for (int ph = 0; ph < 4; ++ph) {  // quadrant (mh, nh)
+      const int mh = ph >> 1, nh = ph & 1;
+      // register fragments for this quadrant (hipcc counts these reads)
+      bf16x8 af[4][2], bf[2][2];
+#pragma unroll
+      for (int s = 0; s < 2; ++s) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int ar = wm + mh * 64 + i * 16 + l16;
+          const int ck = cswz(ar, s * 4 + kq4);
+          af[i][s] = *(const bf16x8 *)((const char *)(abuf +
+                                                      (long)ar * V4_BK) +
+                                       ck * 16);
+        }
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+          const int br = wn + nh * 32 + j * 16 + l16;
+          const int ck = cswz(br, s * 4 + kq4);
+          bf[j][s] = *(const bf16x8 *)((const char *)(bbuf +
+                                                      (long)br * V4_BK) +
+                                       ck * 16);
+        }
+      }
+      // stage one phase's share of tile t+1 into the other buffer
+      if (t + 1 < NT) stage2(t + 1, (t + 1) & 1, ph);
+      if (ph == 1) {
+        // the previous tile's ph3-staged A1/A3 (left in flight past the
+        // ph3 wait) are first read at THIS tile's phase 2: wait them in
+        // (vmcnt(4) leaves this tile's ph0/ph1 issues outstanding; on the
+        // last tile nothing new was issued, so drain)
+        if (t + 1 < NT)
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      } else if (ph == 3) {
+        // once per K-tile: everything except the 2 pieces just issued
+        // (A1,A3 - not read until the next tile's phase 2) must land
+        // before the next tile's first ds_read
+        if (t + 1 < NT)
+          asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int s = 0; s < 2; ++s)
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+#pragma unroll
+          for (int j = 0; j < 2; ++j)
+            acc[mh * 4 + i][nh * 2 + j] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[i][s], bf[j][s], acc[mh * 4 + i][nh * 2 + j], 0, 0,
+                    0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // epilogue
+  const int r4 = (lane / 16) * 4;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int co = n0 + wn + (j >> 1) * 32 + (j & 1) * 16 + l16;
+      if (co >= Cols) continue;
+      const float bv = HAS_BIAS ? bias[co] : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long m = m0 + wm + (i >> 2) * 64 + (i & 3) * 16 + r4 + r;
+        if (m >= M) continue;
+        float v = acc[i][j][r] + bv;
+        if (HAS_CB) {
+          const int ni = (int)(m / ((long)Ho * Wo));
+          v += (float)CB[(long)ni * ldY + co];
+        }
+        if (HAS_RES) v += (float)Res[m * ldY + co];
+        Y[m * ldY + co] = f2bf(v);
+      }
+    }
+  }
+}

@@ -16,6 +16,7 @@ static inline hipStream_t cur_stream() {
 #include "probe.hip"
 #include "attention.hip"
 #include "conv_v3.hip"
+#include "conv_v4.hip"
 #include "conv.hip"
 #include "gemm.hip"
 
