@@ -300,7 +300,13 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
     return e && strcmp(e, "v2") == 0;
   }();
   const long mt4 = (M + V4_BM - 1) / V4_BM;
-  const long nfull = v2only ? 0 : Cout / 256;
+  // measured winners (profiles/r02_conv_v4.md): the 256x256 pipeline wins
+  // on the SD-UNet level-0/1 shapes (Cin<=960, Cout<=640, big M); the
+  // 1280-channel and VAE shapes stay on v2 (2 blocks/CU beats the deep
+  // pipeline there).
+  const bool v4_shape =
+      Cin <= 960 && Cout <= 640 && Cin != 512 && Cout >= 256;
+  const long nfull = (v2only || !v4_shape) ? 0 : Cout / 256;
   const int rem = (int)(Cout - nfull * 256);
   if (nfull > 0) {
     dim3 g4((unsigned)nfull, (unsigned)std::min<long>(mt4, 32768),
@@ -326,7 +332,10 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
   }
   if (rem > 0) {
     const long co0 = nfull * 256;
-#define PICK(B_, R_, C_) conv3x3_nhwc_bf16_kernel<B_, R_, C_, 128>
+    const bool bn64 = rem <= 64;  // exact 64-col tile for the remainder
+#define PICK(B_, R_, C_)                                              \
+  (bn64 ? conv3x3_nhwc_bf16_kernel<B_, R_, C_, 64>                    \
+        : conv3x3_nhwc_bf16_kernel<B_, R_, C_, 128>)
     auto kern =
         has_b ? (has_r ? (has_cb ? PICK(true, true, true)
                                  : PICK(true, true, false))
@@ -337,7 +346,8 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
                        : (has_cb ? PICK(false, false, true)
                                  : PICK(false, false, false)));
 #undef PICK
-    dim3 g2((unsigned)((rem + CONV_BN - 1) / CONV_BN), grid.y, grid.z);
+    dim3 g2((unsigned)((rem + (bn64 ? 63 : 127)) / (bn64 ? 64 : 128)),
+            grid.y, grid.z);
     hipLaunchKernelGGL(
         kern, g2, block, 0, stream, (const __hip_bfloat16 *)x.data_ptr(),
         (const __hip_bfloat16 *)w_prep.data_ptr() + co0 * 9 * Cin,
