@@ -116,6 +116,24 @@ class SDConv2d(nn.Conv2d):
                     xc, self._wprep(), self.bias, res, self.stride[0],
                     chan_bias,
                 )
+            if (
+                k == (3, 3)
+                and self.padding == (1, 1)
+                and self.stride[0] in (1, 2)
+                and self.stride[0] == self.stride[1]
+                and self.groups == 1
+                and residual is None
+                and chan_bias is None
+                and ops.conv3x3_small_supported(
+                    self.in_channels, self.out_channels
+                )
+            ):
+                # stem/IO convs (Cin 3/4/9): native direct kernel instead
+                # of MIOpen's fallback solvers
+                xc = x.contiguous(memory_format=torch.channels_last)
+                return ops.conv3x3_small(
+                    xc, self._wprep(), self.bias, self.stride[0]
+                )
             if k == (1, 1) and self.stride == (1, 1) and self.groups == 1:
                 xc = x.contiguous(memory_format=torch.channels_last)
                 n, c, h, w = xc.shape

@@ -253,6 +253,21 @@ def conv3x3(
     return ext().conv3x3_nhwc(x, w_prep, bias, residual, chan_bias, stride)
 
 
+def conv3x3_small(
+    x: torch.Tensor,
+    w_prep: torch.Tensor,
+    bias: Optional[torch.Tensor],
+    stride: int = 1,
+) -> torch.Tensor:
+    """Direct 3x3 conv for tiny Cin (3/4/9: the stem/IO convs) — keeps the
+    whole pipeline off MIOpen's fallback solvers. GPU-only entry."""
+    return ext().conv3x3_small(x, w_prep, bias, stride)
+
+
+def conv3x3_small_supported(cin: int, cout: int) -> bool:
+    return int(cin) in (3, 4, 9) and cout % 8 == 0 and cout <= 1536
+
+
 # ---------------------------------------------------------------------------
 # timestep embedding (sinusoidal)
 # ---------------------------------------------------------------------------

@@ -18,6 +18,7 @@ static inline hipStream_t cur_stream() {
 #include "conv_v3.hip"
 #include "conv_v4.hip"
 #include "conv.hip"
+#include "conv_small.hip"
 #include "gemm.hip"
 
 #define CHECK_IN(x)                                                     \
@@ -342,6 +343,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_supported", &flash_supported);
   m.def("conv3x3_nhwc", &conv3x3_nhwc);
   m.def("conv3x3_supported", &conv3x3_supported);
+  m.def("conv3x3_small", &conv3x3_small);
+  m.def("conv3x3_small_supported", &conv3x3_small_supported);
   m.def("linear_bf16", &linear_bf16);
   m.def("linear_supported", &linear_supported);
 }

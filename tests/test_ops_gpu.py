@@ -288,6 +288,28 @@ class TestConv3x3:
         err = relerr(out.contiguous(), ref)
         assert err < 0.05, f"conv err {err}"
 
+    @pytest.mark.parametrize("N,Cin,H,W,Cout,stride", [
+        (2, 4, 64, 64, 320, 1),    # UNet conv_in
+        (2, 9, 64, 64, 320, 1),    # inpaint-model conv_in
+        (1, 3, 96, 96, 128, 1),    # VAE encoder conv_in
+        (1, 4, 64, 64, 512, 1),    # VAE decoder conv_in
+        (1, 4, 33, 33, 320, 2),    # strided + odd spatial
+    ])
+    def test_smallcin_vs_conv2d(self, dev, N, Cin, H, W, Cout, stride):
+        torch.manual_seed(5)
+        x = torch.randn(N, Cin, H, W, device=dev, dtype=torch.bfloat16)
+        conv = torch.nn.Conv2d(Cin, Cout, 3, stride=stride, padding=1)
+        conv = conv.to(dev, torch.bfloat16)
+        ref = torch.nn.functional.conv2d(
+            x.float(), conv.weight.float(), conv.bias.float(),
+            stride=stride, padding=1,
+        )
+        xc = x.contiguous(memory_format=torch.channels_last)
+        wprep = conv.weight.permute(0, 2, 3, 1).contiguous()
+        out = ops.conv3x3_small(xc, wprep, conv.bias, stride)
+        err = relerr(out.contiguous(), ref)
+        assert err < 0.05, f"smallcin conv err {err}"
+
     def test_fused_residual(self, dev):
         x = torch.randn(2, 64, 16, 16, device=dev, dtype=torch.bfloat16)
         res = torch.randn(2, 64, 16, 16, device=dev, dtype=torch.bfloat16)
