@@ -72,6 +72,28 @@ class SDConv2d(nn.Conv2d):
             return prep
         return cache[2]
 
+    def forward_upsampled2x(self, x: torch.Tensor) -> torch.Tensor:
+        """conv(nearest2x(x)) with the upsample folded into the conv's
+        im2col addressing on the GPU path (no 4x intermediate)."""
+        if (
+            x.is_cuda
+            and x.dtype == torch.bfloat16
+            and self.kernel_size == (3, 3)
+            and self.padding == (1, 1)
+            and self.stride == (1, 1)
+            and self.groups == 1
+            and self.in_channels % 64 == 0
+            and not getattr(self, "circular", False)
+        ):
+            from .. import ops
+
+            xc = x.contiguous(memory_format=torch.channels_last)
+            return ops.ups2x_conv3x3(xc, self._wprep(), self.bias)
+        x = torch.nn.functional.interpolate(
+            x, scale_factor=2, mode="nearest"
+        )
+        return self(x)
+
     def forward(  # type: ignore[override]
         self,
         x: torch.Tensor,

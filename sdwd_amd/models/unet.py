@@ -182,8 +182,7 @@ class Upsample(nn.Module):
         self.conv = SDConv2d(ch, ch, 3, padding=1)
 
     def forward(self, x):
-        x = torch.nn.functional.interpolate(x, scale_factor=2, mode="nearest")
-        return self.conv(x)
+        return self.conv.forward_upsampled2x(x)
 
 
 class _Seq(nn.Module):

@@ -186,9 +186,7 @@ class _Up(nn.Module):
         self.conv = SDConv2d(ch, ch, 3, padding=1)
 
     def forward(self, x):
-        return self.conv(
-            torch.nn.functional.interpolate(x, scale_factor=2, mode="nearest")
-        )
+        return self.conv.forward_upsampled2x(x)
 
 
 class AutoencoderKL(nn.Module):

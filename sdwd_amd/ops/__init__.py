@@ -268,6 +268,16 @@ def conv3x3_small_supported(cin: int, cout: int) -> bool:
     return int(cin) in (3, 4, 9) and cout % 8 == 0 and cout <= 1536
 
 
+def ups2x_conv3x3(
+    x: torch.Tensor,
+    w_prep: torch.Tensor,
+    bias: Optional[torch.Tensor],
+) -> torch.Tensor:
+    """Nearest-2x upsample fused into a 3x3 conv (VAE decoder / UNet
+    Upsample): no 4x intermediate tensor. GPU-only entry."""
+    return ext().ups2x_conv3x3(x, w_prep, bias)
+
+
 # ---------------------------------------------------------------------------
 # timestep embedding (sinusoidal)
 # ---------------------------------------------------------------------------

@@ -42,7 +42,12 @@ typedef __attribute__((ext_vector_type(4))) float f32x4c;
 // BN = 128 (default) or 64 (exact tiling for Cout % 128 == 64, e.g. the
 // SD1.5 320-channel level: 5 exact 64-col tiles instead of 3 x 128 with a
 // 17% masked-FLOP tail). NJ = column fragments per wave.
-template <bool HAS_BIAS, bool HAS_RES, bool HAS_CB, int BN = 128>
+// UPS: the input is a VIRTUAL nearest-2x upsample of X (H/W are the REAL
+// input dims, Ho/Wo the upsampled-output dims): tap (h,w) of the virtual
+// image reads X[h>>1][w>>1], fusing Upsample+conv into one kernel with no
+// 4x intermediate tensor.
+template <bool HAS_BIAS, bool HAS_RES, bool HAS_CB, int BN = 128,
+          bool UPS = false>
 __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
     const __hip_bfloat16 *__restrict__ X,   // [N,H,W,Cin]
     const __hip_bfloat16 *__restrict__ Wt,  // [Cout,3,3,Cin]
@@ -80,6 +85,7 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
   int hs[4], ws[4];   // ho*stride, wo*stride per staged A row
   int bco[4];         // cout row per staged B row (NB instrs used)
   constexpr int NB = BN / 32;  // B stage instrs (rows BN over 4 waves x 8)
+  int nimg[4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     const int row = i * 32 + wid * 8 + lane / 8;
@@ -89,6 +95,7 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
     const int rem = (int)(mm % ((long)Ho * Wo));
     hs[i] = (rem / Wo) * stride;
     ws[i] = (rem % Wo) * stride;
+    nimg[i] = n_img;
     abase[i] = (((long)n_img * H + hs[i]) * W + ws[i]) * Cin;
     bco[i] = n0 + row;
   }
@@ -109,10 +116,22 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
     for (int i = 0; i < 4; ++i) {
       const int row = i * 32 + wid * 8 + lane / 8;
       const int sc = cswz(row, schunk);
-      const bool av = (hs[i] + dy >= 0) && (hs[i] + dy < H) &&
-                      (ws[i] + dx >= 0) && (ws[i] + dx < W);
-      const __hip_bfloat16 *asrc =
-          av ? (X + abase[i] + poff + sc * 8) : Zero;
+      bool av;
+      const __hip_bfloat16 *asrc;
+      if constexpr (UPS) {
+        const int vh = hs[i] + dy, vw = ws[i] + dx;  // virtual 2H x 2W
+        av = vh >= 0 && vh < 2 * H && vw >= 0 && vw < 2 * W;
+        asrc = av ? (X +
+                     (((long)nimg[i] * H + (vh >> 1)) * W + (vw >> 1)) *
+                         Cin +
+                     (long)kc * CONV_BK + sc * 8)
+                  : Zero;
+      } else {
+        av = (hs[i] + dy >= 0) && (hs[i] + dy < H) &&
+             (ws[i] + dx >= 0) && (ws[i] + dx < W);
+        asrc = av ? (X + abase[i] + poff + sc * 8) : Zero;
+      }
+      (void)poff;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int *)asrc,
           (__attribute__((address_space(3))) unsigned int
@@ -357,6 +376,49 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
         (__hip_bfloat16 *)y.data_ptr() + co0, N, H, W, Cin, rem, Ho, Wo,
         (int)stride, Cout);
   }
+  return y;
+}
+
+torch::Tensor ups2x_conv3x3(torch::Tensor x, torch::Tensor w_prep,
+                            c10::optional<torch::Tensor> bias) {
+  // nearest-2x upsample fused into the 3x3 conv (VAE decoder / UNet
+  // Upsample blocks): output is [N, Cout, 2H, 2W]
+  TORCH_CHECK(x.is_contiguous(torch::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  const int N = x.size(0), Cin = x.size(1), H = x.size(2), W = x.size(3);
+  const int Cout = w_prep.size(0);
+  TORCH_CHECK(conv3x3_supported(Cin));
+  const int Ho = 2 * H, Wo = 2 * W;
+  auto y = torch::empty({N, Cout, Ho, Wo},
+                        x.options().memory_format(
+                            torch::MemoryFormat::ChannelsLast));
+  static torch::Tensor zero_page;
+  if (!zero_page.defined() || zero_page.device() != x.device())
+    zero_page = torch::zeros({64}, x.options());
+  const long M = (long)N * Ho * Wo;
+  const long mtiles = (M + CONV_BM - 1) / CONV_BM;
+  dim3 grid((unsigned)((Cout + CONV_BN - 1) / CONV_BN),
+            (unsigned)std::min<long>(mtiles, 32768),
+            (unsigned)((mtiles + 32767) / 32768));
+  dim3 block(256);
+  const bool has_b = bias.has_value();
+  torch::Tensor bf32;
+  const float *bptr = nullptr;
+  if (has_b) {
+    bf32 = bias->to(torch::kFloat).contiguous();
+    bptr = bf32.data_ptr<float>();
+  }
+  auto kern = has_b
+                  ? conv3x3_nhwc_bf16_kernel<true, false, false, 128, true>
+                  : conv3x3_nhwc_bf16_kernel<false, false, false, 128,
+                                             true>;
+  hipLaunchKernelGGL(kern, grid, block, 0, cur_stream(),
+                     (const __hip_bfloat16 *)x.data_ptr(),
+                     (const __hip_bfloat16 *)w_prep.data_ptr(), bptr,
+                     nullptr, nullptr,
+                     (const __hip_bfloat16 *)zero_page.data_ptr(),
+                     (__hip_bfloat16 *)y.data_ptr(), N, H, W, Cin, Cout,
+                     Ho, Wo, 1, Cout);
   return y;
 }
 #endif

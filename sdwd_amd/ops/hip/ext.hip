@@ -344,6 +344,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_nhwc", &conv3x3_nhwc);
   m.def("conv3x3_supported", &conv3x3_supported);
   m.def("conv3x3_small", &conv3x3_small);
+  m.def("ups2x_conv3x3", &ups2x_conv3x3);
   m.def("conv3x3_small_supported", &conv3x3_small_supported);
   m.def("linear_bf16", &linear_bf16);
   m.def("linear_supported", &linear_supported);

@@ -288,6 +288,27 @@ class TestConv3x3:
         err = relerr(out.contiguous(), ref)
         assert err < 0.05, f"conv err {err}"
 
+    @pytest.mark.parametrize("N,Cin,H,W,Cout", [
+        (2, 64, 16, 16, 64),      # UNet upsample shape class
+        (1, 512, 24, 24, 512),    # VAE decoder up
+        (1, 256, 33, 17, 128),    # odd spatial
+    ])
+    def test_fused_upsample_conv(self, dev, N, Cin, H, W, Cout):
+        torch.manual_seed(7)
+        x = torch.randn(N, Cin, H, W, device=dev, dtype=torch.bfloat16)
+        conv = torch.nn.Conv2d(Cin, Cout, 3, padding=1).to(dev, torch.bfloat16)
+        up = torch.nn.functional.interpolate(
+            x.float(), scale_factor=2, mode="nearest"
+        )
+        ref = torch.nn.functional.conv2d(
+            up, conv.weight.float(), conv.bias.float(), padding=1
+        )
+        xc = x.contiguous(memory_format=torch.channels_last)
+        wprep = conv.weight.permute(0, 2, 3, 1).contiguous()
+        out = ops.ups2x_conv3x3(xc, wprep, conv.bias)
+        err = relerr(out.contiguous(), ref)
+        assert err < 0.05, f"ups2x conv err {err}"
+
     @pytest.mark.parametrize("N,Cin,H,W,Cout,stride", [
         (2, 4, 64, 64, 320, 1),    # UNet conv_in
         (2, 9, 64, 64, 320, 1),    # inpaint-model conv_in
