@@ -7,6 +7,19 @@ import torch.nn as nn
 from .. import ops
 
 
+def _fp32_cached(mod: nn.Module):
+    """fp32 copies of (weight, bias), re-cast only when the params change
+    (by data_ptr): saves thousands of tiny cast kernels per generation."""
+    key = (mod.weight.data_ptr(), mod.bias.data_ptr(),
+           mod.weight._version, mod.bias._version)
+    cache = getattr(mod, "_fp32_cache", None)
+    if cache is None or cache[0] != key:
+        cache = (key, mod.weight.detach().float(),
+                 mod.bias.detach().float())
+        mod._fp32_cache = cache
+    return cache[1], cache[2]
+
+
 class FusedGroupNorm(nn.Module):
     """GroupNorm with an optionally fused SiLU (one HIP kernel on GPU)."""
 
@@ -20,8 +33,10 @@ class FusedGroupNorm(nn.Module):
         self.bias = nn.Parameter(torch.zeros(channels))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        w, b = (_fp32_cached(self) if x.is_cuda
+                else (self.weight, self.bias))
         return ops.group_norm_silu(
-            x, self.weight, self.bias, self.groups, self.eps, self.silu
+            x, w, b, self.groups, self.eps, self.silu
         )
 
 
@@ -33,7 +48,9 @@ class FusedLayerNorm(nn.Module):
         self.bias = nn.Parameter(torch.zeros(dim))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return ops.layer_norm(x, self.weight, self.bias, self.eps)
+        w, b = (_fp32_cached(self) if x.is_cuda
+                else (self.weight, self.bias))
+        return ops.layer_norm(x, w, b, self.eps)
 
 
 class GEGLU(nn.Module):

@@ -104,16 +104,37 @@ class CrossAttention(nn.Module):
         self.to_v = nn.Linear(context_dim, query_dim, bias=False)
         self.to_out = nn.Linear(query_dim, query_dim)
 
+    def _wcat(self, names):
+        """Concatenated projection weight, cached by source data_ptrs:
+        one GEMM instead of 2-3 reads the (huge, M=B*S) activation once."""
+        ws = [getattr(self, n).weight for n in names]
+        key = tuple(w.data_ptr() for w in ws)
+        cache = getattr(self, "_wcat_cache", None)
+        if cache is None or cache[0] != key:
+            cat = torch.cat([w.detach() for w in ws], dim=0)
+            self._wcat_cache = cache = (key, cat)
+        return cache[1]
+
     def forward(
         self, x: torch.Tensor, context: Optional[torch.Tensor] = None
     ) -> torch.Tensor:
-        context = x if context is None else context
+        import torch.nn.functional as F
+
         b, s, d = x.shape
-        sk = context.shape[1]
-        # natural [B,S,H,D] layout end-to-end: the flash kernel reads strided
-        q = self.to_q(x).view(b, s, self.heads, self.d_head)
-        k = self.to_k(context).view(b, sk, self.heads, self.d_head)
-        v = self.to_v(context).view(b, sk, self.heads, self.d_head)
+        hd = (self.heads, self.d_head)
+        if context is None:
+            # self-attention: one fused QKV GEMM (the skinny K=channels
+            # projections are A-traffic-bound at S=4096); the q/k/v slices
+            # stay strided views - the flash kernel reads strides directly
+            qkv = F.linear(x, self._wcat(("to_q", "to_k", "to_v")))
+            q = qkv[..., :d].unflatten(-1, hd)
+            k = qkv[..., d:2 * d].unflatten(-1, hd)
+            v = qkv[..., 2 * d:].unflatten(-1, hd)
+        else:
+            q = self.to_q(x).unflatten(-1, hd)
+            kv = F.linear(context, self._wcat(("to_k", "to_v")))
+            k = kv[..., :d].unflatten(-1, hd)
+            v = kv[..., d:].unflatten(-1, hd)
         out = ops.attention_bshd(q, k, v)
         return self.to_out(out.reshape(b, s, d))
 
@@ -129,15 +150,19 @@ class BasicTransformerBlock(nn.Module):
         self.ff = nn.Sequential(GEGLU(dim, dim * 4), nn.Linear(dim * 4, dim))
 
     def forward(self, x, context):
+        from .layers import _fp32_cached
+
         h = self.attn1(self.norm1(x))
+        if x.is_cuda:
+            w2, b2 = _fp32_cached(self.norm2)
+            w3, b3 = _fp32_cached(self.norm3)
+        else:
+            w2, b2 = self.norm2.weight, self.norm2.bias
+            w3, b3 = self.norm3.weight, self.norm3.bias
         # fused residual-add + pre-norm (one HBM round trip saved per hop)
-        x, n2 = ops.add_layer_norm(
-            x, h, self.norm2.weight, self.norm2.bias, self.norm2.eps
-        )
+        x, n2 = ops.add_layer_norm(x, h, w2, b2, self.norm2.eps)
         h = self.attn2(n2, context)
-        x, n3 = ops.add_layer_norm(
-            x, h, self.norm3.weight, self.norm3.bias, self.norm3.eps
-        )
+        x, n3 = ops.add_layer_norm(x, h, w3, b3, self.norm3.eps)
         return x + self.ff(n3)
 
 
