@@ -20,7 +20,7 @@ import torch
 from ..models.registry import ModelBundle, load_model
 from ..models import tokenizer
 from ..utils import get_logger
-from .graphs import GraphedDenoiser
+from .graphs import GraphedDenoiser, GraphedModelFn, graphs_enabled
 from .samplers import build_sampler
 from .schedule import schedule_for, sigma_for_t
 
@@ -716,6 +716,42 @@ class StableDiffusionPipeline:
                 return _to_eps(
                     denoiser(x1, ts, ctx[:nb], yc), x_in, t, pred_type
                 )
+
+        # whole-step hipGraph (round-1 verdict #5): the CFG duplication,
+        # the UNet forward and the guidance combine replay as ONE graph per
+        # step; only the sampler's handful of elementwise kernels (and the
+        # deterministic per-image CPU noise) stay eager - a design choice
+        # that preserves the seed plan's bit-exact shard determinism.
+        if (
+            self.device.type == "cuda"
+            and graphs_enabled()
+            and cfg != 1.0
+            and not seg_tensors
+            and and_ws == [1.0]
+            and not cn_units
+            and s_min_uncond == 0
+            and pred_type != "v"
+        ):
+            _raw_fn = self._denoiser.fn  # eager UNet call (no graph nesting)
+            from .. import ops as _gops
+
+            def _cfg_core(x_in, t0d, ctx2, y2):
+                nb = x_in.shape[0]
+                ts = t0d.expand(nb * 2)
+                xk = torch.cat([x_in, x_in], dim=0)
+                if is_inpaint_model:
+                    ex = _inpaint_extra(x_in.shape[2], x_in.shape[3])
+                    xk = torch.cat([xk, torch.cat([ex, ex], dim=0)], dim=1)
+                eps = _raw_fn(xk, ts, ctx2, y2)
+                e_c, e_u = eps.chunk(2, dim=0)
+                return _gops.lincomb(e_c, e_u, cfg, 1.0 - cfg)
+
+            _graphed = GraphedModelFn(_cfg_core, self.device)
+
+            def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:  # noqa: F811
+                ctx2, y2, _ws = _ctx_y_for(t)
+                _graphed.bind(ctx2, y2)
+                return _graphed(x_in, t)
 
         if req.refiner_model:
             # two-model handoff: t descends through the schedule, so the

@@ -179,6 +179,31 @@ class TestPipelineGPU:
         assert res.images.shape == (2, 64, 64, 3)
         assert torch.isfinite(res.images.float()).all()
 
+    def test_whole_step_graph_matches_eager(self, dev, monkeypatch):
+        """The whole-step hipGraph (CFG + UNet + combine in one replay)
+        must reproduce the eager images bit-for-bit-ish across steps AND
+        across a second generation that re-binds new conditioning."""
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        base = dict(prompt="graphed", steps=3, width=64, height=64,
+                    seeds=[11], cfg_scale=7.0)
+        monkeypatch.setenv("SDWD_HIPGRAPH", "0")
+        eager = StableDiffusionPipeline("tiny", device=dev)
+        a = eager.generate(PipelineRequest(**base)).images
+        a2 = eager.generate(
+            PipelineRequest(**{**base, "prompt": "other"})
+        ).images
+        monkeypatch.setenv("SDWD_HIPGRAPH", "1")
+        graphed = StableDiffusionPipeline("tiny", device=dev)
+        b = graphed.generate(PipelineRequest(**base)).images
+        # second generation re-binds conditioning into the SAME graph
+        b2 = graphed.generate(
+            PipelineRequest(**{**base, "prompt": "other"})
+        ).images
+        assert (a.float() - b.float()).abs().max() <= 1.0
+        assert (a2.float() - b2.float()).abs().max() <= 1.0
+        assert not torch.equal(b, b2)  # conditioning rebind took effect
+
     def test_sd15_one_step(self, dev):
         """One real SD1.5 denoise step at 512x512 (bf16)."""
         from sdwd_amd.models import load_model

@@ -23,9 +23,36 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch  # noqa: E402
 
 
+def synthesize(arch: str, path: str) -> str:
+    """Write a FULL-KEY ldm checkpoint at the arch's real shapes (random
+    weights, fp16 like distribution files). The key set is the canonical
+    one the structural canaries in tests/test_convert.py pin against the
+    published SD1.5/SDXL layouts, so validating against this file
+    exercises the converter end-to-end at real scale without network."""
+    from safetensors.torch import save_file
+
+    from sdwd_amd.models.convert import to_ldm_state_dict
+    from sdwd_amd.models.registry import load_model
+
+    bundle = load_model(arch, device="cpu", dtype=torch.float32,
+                        cache=False)
+    state = to_ldm_state_dict(bundle)
+    state = {k: v.detach().to(torch.float16).contiguous()
+             for k, v in state.items()}
+    print(f"synthesized {len(state)} tensors for arch {arch}")
+    save_file(state, path)
+    print(f"wrote {path} "
+          f"({os.path.getsize(path) / 1e9:.2f} GB)")
+    return path
+
+
 def main() -> int:
     ap = argparse.ArgumentParser(description=__doc__)
-    ap.add_argument("checkpoint")
+    ap.add_argument("checkpoint", nargs="?", default="")
+    ap.add_argument("--synthesize", default="",
+                    help="arch (sd15/sd21/sdxl): write a full-key "
+                         "random-weight ldm file at real shapes first and "
+                         "validate against it")
     ap.add_argument("--image", default="", help="write a txt2img sample here")
     ap.add_argument("--prompt", default="a photograph of an astronaut "
                                         "riding a horse")
@@ -33,6 +60,14 @@ def main() -> int:
         "cuda:0" if torch.cuda.is_available() else "cpu"
     ))
     args = ap.parse_args()
+    if args.synthesize:
+        args.checkpoint = args.checkpoint or os.path.join(
+            os.environ.get("TMPDIR", "/tmp"),
+            f"{args.synthesize}_synth.safetensors",
+        )
+        synthesize(args.synthesize, args.checkpoint)
+    elif not args.checkpoint:
+        ap.error("checkpoint path or --synthesize required")
 
     from safetensors import safe_open
 
