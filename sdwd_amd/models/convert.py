@@ -344,21 +344,23 @@ def load_ldm_state_dict(bundle, state: Dict[str, torch.Tensor]) -> Dict[str, lis
         with torch.no_grad():
             enc_out = dict(bundle.vae.encoder.named_parameters())
             if "quant_conv.weight" in quant:
+                dev = enc_out["conv_out.weight"].device
                 w, b = _fold_output_1x1(
                     enc_out["conv_out.weight"].float(),
                     enc_out["conv_out.bias"].float(),
-                    quant["quant_conv.weight"].float(),
-                    quant["quant_conv.bias"].float(),
+                    quant["quant_conv.weight"].float().to(dev),
+                    quant["quant_conv.bias"].float().to(dev),
                 )
                 enc_out["conv_out.weight"].copy_(w.to(enc_out["conv_out.weight"].dtype))
                 enc_out["conv_out.bias"].copy_(b.to(enc_out["conv_out.bias"].dtype))
             dec = dict(bundle.vae.decoder.named_parameters())
             if "post_quant_conv.weight" in quant:
+                dev = dec["conv_in.weight"].device
                 w, b = _fold_input_1x1(
                     dec["conv_in.weight"].float(),
                     dec["conv_in.bias"].float(),
-                    quant["post_quant_conv.weight"].float(),
-                    quant["post_quant_conv.bias"].float(),
+                    quant["post_quant_conv.weight"].float().to(dev),
+                    quant["post_quant_conv.bias"].float().to(dev),
                 )
                 dec["conv_in.weight"].copy_(w.to(dec["conv_in.weight"].dtype))
                 dec["conv_in.bias"].copy_(b.to(dec["conv_in.bias"].dtype))

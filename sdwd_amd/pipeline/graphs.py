@@ -92,7 +92,9 @@ class GraphedModelFn:
                 e.y.copy_(self._y)
             e.ctx_src = id(self._ctx)
         e.graph.replay()
-        return e.out
+        # CLONE: multi-eval samplers (Heun, DPM2, UniPC...) hold the first
+        # eval's eps while the second replay overwrites the static buffer
+        return e.out.clone()
 
     def _capture(self, x, t, key) -> Optional[_FnEntry]:
         try:
@@ -158,7 +160,8 @@ class GraphedDenoiser:
         if y is not None:
             entry.y.copy_(y)
         entry.graph.replay()
-        return entry.out
+        # CLONE: see GraphedModelFn - callers may hold this across evals
+        return entry.out.clone()
 
     def _capture(self, x, ts, ctx, y, key) -> Optional[_Entry]:
         try:

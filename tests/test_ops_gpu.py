@@ -204,6 +204,25 @@ class TestPipelineGPU:
         assert (a2.float() - b2.float()).abs().max() <= 1.0
         assert not torch.equal(b, b2)  # conditioning rebind took effect
 
+    @pytest.mark.parametrize("sampler", ["Heun", "DPM2", "DPM++ 2M", "UniPC"])
+    def test_graph_multi_eval_samplers(self, dev, monkeypatch, sampler):
+        """Multi-eval samplers hold the FIRST model eval's eps across the
+        second replay - the graph out-buffer must not alias (regression:
+        the static buffer was returned uncloned)."""
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        base = dict(prompt="ms", steps=3, width=64, height=64, seeds=[3],
+                    sampler_name=sampler, cfg_scale=7.0)
+        monkeypatch.setenv("SDWD_HIPGRAPH", "0")
+        a = StableDiffusionPipeline("tiny", device=dev).generate(
+            PipelineRequest(**base)
+        ).images
+        monkeypatch.setenv("SDWD_HIPGRAPH", "1")
+        b = StableDiffusionPipeline("tiny", device=dev).generate(
+            PipelineRequest(**base)
+        ).images
+        assert (a.float() - b.float()).abs().max() <= 1.0, sampler
+
     def test_sd15_one_step(self, dev):
         """One real SD1.5 denoise step at 512x512 (bf16)."""
         from sdwd_amd.models import load_model
