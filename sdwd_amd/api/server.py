@@ -732,6 +732,8 @@ def create_app(engine: Optional[LocalEngine] = None,
                     "avg_ipm": w.eta.avg_ipm,
                     "mpe": w.eta.mpe(),
                     "pixel_cap": w.pixel_cap,
+                    "model_override": getattr(w, "model_override", "")
+                    or "",
                 }
                 for w in engine.world.workers
             ],
@@ -817,6 +819,40 @@ def create_app(engine: Optional[LocalEngine] = None,
             raise HTTPException(422, str(exc))
         engine.world.save()
         return engine.world.benchmark_payload.model_dump()
+
+    @app.post("/sdwd/worker/{label}/config")
+    def worker_config(label: str, body: Dict[str, Any]):
+        """Per-worker pixel cap + checkpoint override (ref ui.py:161-171,
+        313-319: both editable in the reference's Worker Config tab)."""
+        w = engine.world.get_worker(label)
+        if w is None:
+            raise HTTPException(404, label)
+        unknown = set(body) - {"pixel_cap", "model_override"}
+        if unknown:
+            raise HTTPException(422, f"unknown fields: {sorted(unknown)}")
+        if "pixel_cap" in body:
+            cap = body["pixel_cap"]
+            if cap in (None, "", 0, "0"):
+                w.pixel_cap = 0
+            else:
+                try:
+                    cap = int(cap)
+                except (TypeError, ValueError):
+                    raise HTTPException(422, "pixel_cap must be an int")
+                if cap < 0:
+                    raise HTTPException(422, "pixel_cap must be >= 0")
+                w.pixel_cap = cap
+        if "model_override" in body:
+            name = (body["model_override"] or "").strip()
+            if name and name not in available_models():
+                raise HTTPException(404, f"unknown model: {name}")
+            w.model_override = name or None
+        engine.world.save()
+        return {
+            "label": w.label,
+            "pixel_cap": w.pixel_cap,
+            "model_override": getattr(w, "model_override", "") or "",
+        }
 
     @app.post("/sdwd/worker/{label}/enable")
     def enable_worker(label: str):

@@ -41,6 +41,15 @@ PAGE = """<!DOCTYPE html>
 <h2>utils</h2>
 <button onclick="fetch('/sdwd/benchmark',{method:'POST'})">re-benchmark</button>
 <button onclick="fetch('/sdwd/sync-script',{method:'POST'})">run sync script</button>
+<h2>benchmark payload</h2>
+<form onsubmit="saveBench(event)" id="benchForm">
+ prompt <input id="bp_prompt" size="40"/>
+ steps <input id="bp_steps" size="3"/>
+ batch <input id="bp_batch_size" size="3"/>
+ w <input id="bp_width" size="4"/>
+ h <input id="bp_height" size="4"/>
+ <button>save payload</button> <span id="bp_saved"></span>
+</form>
 <h2>settings</h2>
 <form onsubmit="saveSettings(event)" id="settingsForm">
  job timeout <input id="s_job_timeout" size="4"/>
@@ -57,11 +66,17 @@ async function refresh(){
   try{
     const s = await (await fetch('/sdwd/status')).json();
     let h = '<table><tr><th>rank</th><th>device</th><th>state</th>'+
-            '<th>ipm</th><th>mpe%</th><th></th></tr>';
+            '<th>ipm</th><th>mpe%</th><th>pixel cap</th>'+
+            '<th>model override</th><th></th></tr>';
     for(const w of s.workers){
+      const mo = w.model_override || '';
       h += `<tr><td>${w.label}</td><td>${w.device}</td><td>${w.state}</td>`+
            `<td>${w.avg_ipm.toFixed(1)}</td><td>${w.mpe.toFixed(1)}</td>`+
-           `<td><button onclick="tog('${w.label}','${w.state}')">`+
+           `<td><input id="cap_${w.label}" size="9" value="${w.pixel_cap||''}"`+
+           ` placeholder="none"/></td>`+
+           `<td><select id="mo_${w.label}" data-v="${mo}"></select></td>`+
+           `<td><button onclick="wcfg('${w.label}')">set</button>`+
+           `<button onclick="tog('${w.label}','${w.state}')">`+
            `${w.state==='DISABLED'?'enable':'disable'}</button></td></tr>`;
     }
     h += `</table><p>model: ${s.model} — busy: ${s.busy}</p>`;
@@ -78,6 +93,45 @@ async function refresh(){
     }
     document.getElementById('log').textContent = s.log.slice(-16).join('\\n');
   }catch(e){}
+}
+let MODELS = [];
+function fillOverride(){
+  for(const sel of document.querySelectorAll('select[id^=mo_]')){
+    if (sel.options.length) continue;
+    const cur = sel.dataset.v || '';
+    sel.innerHTML = '<option value="">(follow main)</option>' +
+      MODELS.map(m => `<option ${m===cur?'selected':''}>${m}</option>`).join('');
+  }
+}
+async function wcfg(label){
+  const cap = document.getElementById('cap_'+label).value;
+  const mo = document.getElementById('mo_'+label).value;
+  await fetch(`/sdwd/worker/${label}/config`,{method:'POST',
+    headers:{'Content-Type':'application/json'},
+    body: JSON.stringify({pixel_cap: cap ? parseInt(cap) : 0,
+                          model_override: mo})});
+  refresh();
+}
+async function loadBench(){
+  try{
+    const b = await (await fetch('/sdwd/benchmark-payload')).json();
+    for(const k of ['prompt','steps','batch_size','width','height'])
+      document.getElementById('bp_'+k).value = b[k];
+  }catch(e){}
+}
+async function saveBench(ev){
+  ev.preventDefault();
+  const body = {
+    prompt: document.getElementById('bp_prompt').value,
+    steps: parseInt(document.getElementById('bp_steps').value),
+    batch_size: parseInt(document.getElementById('bp_batch_size').value),
+    width: parseInt(document.getElementById('bp_width').value),
+    height: parseInt(document.getElementById('bp_height').value),
+  };
+  const r = await fetch('/sdwd/benchmark-payload',{method:'POST',
+    headers:{'Content-Type':'application/json'}, body: JSON.stringify(body)});
+  document.getElementById('bp_saved').textContent = r.ok ? 'saved' : 'error';
+  setTimeout(()=>{document.getElementById('bp_saved').textContent='';}, 2000);
 }
 async function tog(label, state){
   const act = state==='DISABLED' ? 'enable' : 'disable';
@@ -134,8 +188,16 @@ async function loadSamplers(){
   }catch(e){}
 }
 setInterval(refresh, 1500);  // ref distributed.js:7-23 auto-refresh cadence
-refresh();
+async function loadModels(){
+  try{
+    const ms = await (await fetch('/sdapi/v1/sd-models')).json();
+    MODELS = ms.map(m => m.model_name);
+  }catch(e){}
+}
+(async () => { await loadModels(); await refresh(); fillOverride(); })();
+setInterval(fillOverride, 1600);
 loadSettings();
 loadSamplers();
+loadBench();
 </script>
 </body></html>"""

@@ -413,6 +413,39 @@ class TestControl:
         assert w1["state"] == "DISABLED"
         assert client.post("/sdwd/worker/gpu1/enable").status_code == 200
 
+    def test_worker_config_pixel_cap_and_override(self, client):
+        """Ref ui.py:161-171/313-319 parity: pixel cap + checkpoint
+        override are editable per worker through the API/UI."""
+        r = client.post(
+            "/sdwd/worker/gpu1/config",
+            json={"pixel_cap": 512 * 512 * 2, "model_override": "tiny-xl"},
+        )
+        assert r.status_code == 200, r.text
+        st = client.get("/sdwd/status").json()
+        w1 = [w for w in st["workers"] if w["label"] == "gpu1"][0]
+        assert w1["pixel_cap"] == 512 * 512 * 2
+        assert w1["model_override"] == "tiny-xl"
+        # clearing works; unknown model rejected; unknown field rejected
+        r = client.post(
+            "/sdwd/worker/gpu1/config",
+            json={"pixel_cap": 0, "model_override": ""},
+        )
+        assert r.status_code == 200
+        w1 = [
+            w for w in client.get("/sdwd/status").json()["workers"]
+            if w["label"] == "gpu1"
+        ][0]
+        assert w1["pixel_cap"] == 0 and w1["model_override"] == ""
+        assert client.post(
+            "/sdwd/worker/gpu1/config", json={"model_override": "nope"}
+        ).status_code == 404
+        assert client.post(
+            "/sdwd/worker/gpu1/config", json={"bogus": 1}
+        ).status_code == 422
+        assert client.post(
+            "/sdwd/worker/nope/config", json={"pixel_cap": 1}
+        ).status_code == 404
+
 
 class TestAlwaysonControlNet:
     def test_controlnet_unit_applied(self, client):
