@@ -159,6 +159,37 @@ def _resize_init(img: torch.Tensor, w: int, h: int, mode: int) -> torch.Tensor:
     return x[0].permute(1, 2, 0).clamp(0, 255).to(torch.uint8)
 
 
+# alwayson scripts executed natively (C18): ControlNet (full guidance
+# windows) and Dynamic Prompts (wildcard/variant expansion). Everything
+# else is logged and skipped, like the reference's compat filter when a
+# remote lacked the script - docs/usage.md documents the boundary.
+_NATIVE_ALWAYSON = ("controlnet", "dynamic prompts")
+
+
+def _dynamic_prompts_enabled(alwayson: Dict[str, Any]) -> bool:
+    """sd-dynamic-prompts payload: {"dynamic prompts": {"args": [True, ...]}}
+    (first arg = enabled, like the extension's process() signature)."""
+    for name, body in (alwayson or {}).items():
+        if name.lower().replace("-", " ") == "dynamic prompts":
+            args = (body or {}).get("args", [True])
+            return bool(args[0]) if args else True
+    return False
+
+
+def _expand_dynamic(gen) -> None:
+    """Per-image seeded expansion (determinism contract C22: image k
+    depends on seed_k only, never on shard placement)."""
+    from ..core.seeds import fix_seed
+    from ..pipeline.wildcards import expand, has_dynamic_syntax
+
+    if not has_dynamic_syntax(gen.prompt):
+        return
+    gen.seed = fix_seed(gen.seed)
+    gen.prompts = [
+        expand(gen.prompt, gen.seed + i) for i in range(gen.batch_size)
+    ]
+
+
 def _parse_controlnet(alwayson: Dict[str, Any]):
     """sdwui controlnet payload: {"controlnet": {"args": [unit, ...]}}
     (ref control_net.py:20-79 packed this; every unit is executed natively
@@ -166,8 +197,10 @@ def _parse_controlnet(alwayson: Dict[str, Any]):
     skipped (ref C18 compat filter)."""
     units = []
     for name, body in (alwayson or {}).items():
-        if name.lower() != "controlnet":
+        if name.lower() not in _NATIVE_ALWAYSON:
             log.warning("ignoring unsupported alwayson script '%s'", name)
+            continue
+        if name.lower() != "controlnet":
             continue
         for unit in (body or {}).get("args", []):
             img_b64 = unit.get("input_image") or unit.get("image")
@@ -260,7 +293,11 @@ def create_app(engine: Optional[LocalEngine] = None,
         info = {
             "all_seeds": result.seeds,
             "all_subseeds": [-1] * len(result.seeds),
-            "all_prompts": [gen.prompt] * len(result.seeds),
+            "all_prompts": (
+                list(gen.prompts)
+                if gen.prompts and len(gen.prompts) == len(result.seeds)
+                else [gen.prompt] * len(result.seeds)
+            ),
             "all_negative_prompts": [gen.negative_prompt] * len(result.seeds),
             "infotexts": result.infotexts,
             "job_summary": result.job_summary,
@@ -360,6 +397,8 @@ def create_app(engine: Optional[LocalEngine] = None,
             refiner_model=req.refiner_checkpoint,
             refiner_switch_at=req.refiner_switch_at,
         )
+        if _dynamic_prompts_enabled(req.alwayson_scripts):
+            _expand_dynamic(gen)
         return run_generation(gen, req.send_images, req.save_images)
 
     @app.post("/sdapi/v1/img2img")

@@ -447,6 +447,38 @@ class TestControl:
         ).status_code == 404
 
 
+class TestDynamicPromptsAPI:
+    def test_per_image_expansion(self, client):
+        body = {
+            "prompt": "a {red|green|blue|black|white} cow",
+            "steps": 1, "width": 64, "height": 64, "seed": 77,
+            "batch_size": 3,
+            "alwayson_scripts": {"dynamic prompts": {"args": [True]}},
+        }
+        r = client.post("/sdapi/v1/txt2img", json=body)
+        assert r.status_code == 200, r.text
+        info = r.json()["info"]
+        # per-image prompts recorded; deterministic across repeat calls
+        import json as _json
+        prompts = _json.loads(info)["all_prompts"]
+        assert len(prompts) == 3
+        assert all(" cow" in p and "{" not in p for p in prompts)
+        r2 = client.post("/sdapi/v1/txt2img", json=body)
+        assert _json.loads(r2.json()["info"])["all_prompts"] == prompts
+
+    def test_disabled_passthrough(self, client):
+        body = {
+            "prompt": "a {red|blue} cow", "steps": 1, "width": 64,
+            "height": 64, "seed": 7,
+            "alwayson_scripts": {"dynamic prompts": {"args": [False]}},
+        }
+        r = client.post("/sdapi/v1/txt2img", json=body)
+        assert r.status_code == 200
+        import json as _json
+        ps = _json.loads(r.json()["info"])["all_prompts"]
+        assert ps[0] == "a {red|blue} cow"  # literal, not expanded
+
+
 class TestAlwaysonControlNet:
     def test_controlnet_unit_applied(self, client):
         hint = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)

@@ -212,3 +212,59 @@ class TestPngMetadata:
 
         img = torch.zeros(4, 4, 3, dtype=torch.uint8)
         assert png_parameters(encode_png(img)) is None
+
+
+class TestDynamicPrompts:
+    """Native execution of the sd-dynamic-prompts syntax (C18)."""
+
+    def test_variants_deterministic(self):
+        from sdwd_amd.pipeline.wildcards import expand
+
+        a = expand("a {red|green|blue} hat", 42)
+        b = expand("a {red|green|blue} hat", 42)
+        assert a == b
+        assert a in ("a red hat", "a green hat", "a blue hat")
+        seen = {expand("a {red|green|blue} hat", s) for s in range(30)}
+        assert len(seen) > 1  # different seeds draw different variants
+
+    def test_multi_select(self):
+        from sdwd_amd.pipeline.wildcards import expand
+
+        out = expand("{2$$a|b|c}", 7)
+        parts = out.split(", ")
+        assert len(parts) == 2 and len(set(parts)) == 2
+        assert set(parts) <= {"a", "b", "c"}
+
+    def test_weights(self):
+        from sdwd_amd.pipeline.wildcards import expand
+
+        hits = sum(
+            expand("{99::heavy|1::light}", s) == "heavy" for s in range(100)
+        )
+        assert hits > 90
+
+    def test_nesting(self):
+        from sdwd_amd.pipeline.wildcards import expand
+
+        out = expand("{a {big|small} cat|a dog}", 3)
+        assert out in ("a big cat", "a small cat", "a dog")
+
+    def test_wildcard_files(self, tmp_path):
+        from sdwd_amd.pipeline.wildcards import expand
+
+        (tmp_path / "animals.txt").write_text("# comment\ncow\nhorse\n")
+        sub = tmp_path / "style"
+        sub.mkdir()
+        (sub / "mood.txt").write_text("calm\n")
+        out = expand("a __animals__, __style/mood__", 5, root=str(tmp_path))
+        assert out in ("a cow, calm", "a horse, calm")
+        # missing wildcard degrades to empty, no raise
+        assert expand("x __nope__ y", 1, root=str(tmp_path)) == "x  y"
+
+    def test_plain_text_untouched(self):
+        from sdwd_amd.pipeline.wildcards import expand, has_dynamic_syntax
+
+        assert expand("plain prompt", 1) == "plain prompt"
+        assert not has_dynamic_syntax("plain prompt")
+        assert has_dynamic_syntax("{a|b}")
+        assert has_dynamic_syntax("__cards__")
