@@ -20,6 +20,7 @@ static inline hipStream_t cur_stream() {
 #include "conv.hip"
 #include "conv_small.hip"
 #include "gemm.hip"
+#include "gemm_v2.hip"
 
 #define CHECK_IN(x)                                                     \
   TORCH_CHECK(x.is_cuda(), #x " must be on GPU");                        \
@@ -348,4 +349,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv3x3_small_supported", &conv3x3_small_supported);
   m.def("linear_bf16", &linear_bf16);
   m.def("linear_supported", &linear_supported);
+  m.def("gemm_v2", &gemm_v2);
+  m.def("gemm_v2_supported", &gemm_v2_supported);
 }
