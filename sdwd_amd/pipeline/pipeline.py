@@ -725,6 +725,7 @@ class StableDiffusionPipeline:
         if (
             self.device.type == "cuda"
             and graphs_enabled()
+            and os.environ.get("SDWD_WHOLESTEP", "1") not in ("", "0")
             and cfg != 1.0
             and not seg_tensors
             and and_ws == [1.0]
