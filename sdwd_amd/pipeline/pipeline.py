@@ -216,6 +216,10 @@ class StableDiffusionPipeline:
             lambda x, ts, ctx, y: self.model.unet(x, ts, ctx, y=y),
             self.device,
         )
+        # whole-step graphs persist ACROSS generations (a fresh wrapper per
+        # request would re-capture every generation: +3 UNet forwards)
+        self._wholestep_cache: dict = {}
+        self._canon_extra_cache: dict = {}
 
     # -- conditioning --------------------------------------------------------
     @torch.no_grad()
@@ -346,6 +350,7 @@ class StableDiffusionPipeline:
                     m.circular = req.tiling
             # captured graphs baked the old padding path
             self._denoiser.cache.clear()
+            self._wholestep_cache.clear()
             self._tiling = req.tiling
         b = req.batch_size
         f = self.model.vae.cfg.downsample_factor
@@ -732,27 +737,63 @@ class StableDiffusionPipeline:
             and not cn_units
             and s_min_uncond == 0
             and pred_type != "v"
+            and (
+                not is_inpaint_model
+                or (req.mask_image is None and req.init_latents is None)
+            )
         ):
-            _raw_fn = self._denoiser.fn  # eager UNet call (no graph nesting)
-            from .. import ops as _gops
+            # persistent per-(cfg, inpaint) graph: the core closure only
+            # captures request-INDEPENDENT state (the raw UNet call, self's
+            # canonical no-mask inpaint conditioning), so one capture
+            # serves every later generation at the same shape
+            wkey = (float(cfg), bool(is_inpaint_model))
+            _graphed = self._wholestep_cache.get(wkey)
+            if _graphed is None:
+                _raw_fn = self._denoiser.fn  # eager UNet (no graph nesting)
+                from .. import ops as _gops
 
-            def _cfg_core(x_in, t0d, ctx2, y2):
-                nb = x_in.shape[0]
-                ts = t0d.expand(nb * 2)
-                xk = torch.cat([x_in, x_in], dim=0)
-                if is_inpaint_model:
-                    ex = _inpaint_extra(x_in.shape[2], x_in.shape[3])
-                    xk = torch.cat([xk, torch.cat([ex, ex], dim=0)], dim=1)
-                eps = _raw_fn(xk, ts, ctx2, y2)
-                e_c, e_u = eps.chunk(2, dim=0)
-                return _gops.lincomb(e_c, e_u, cfg, 1.0 - cfg)
+                _lat_c = lat_c
+                _wcfg = float(cfg)
+                _inp = bool(is_inpaint_model)
 
-            _graphed = GraphedModelFn(_cfg_core, self.device)
+                def _canon_extra(h, w):
+                    key = (h, w)
+                    t = self._canon_extra_cache.get(key)
+                    if t is None:
+                        t = torch.cat(
+                            [
+                                torch.ones(1, 1, h, w, device=self.device),
+                                torch.zeros(
+                                    1, _lat_c, h, w, device=self.device
+                                ),
+                            ],
+                            dim=1,
+                        ).to(self.dtype).contiguous()
+                        self._canon_extra_cache[key] = t
+                    return t
+
+                def _cfg_core(x_in, t0d, ctx2, y2):
+                    nb = x_in.shape[0]
+                    ts = t0d.expand(nb * 2)
+                    xk = torch.cat([x_in, x_in], dim=0)
+                    if _inp:
+                        ex = _canon_extra(x_in.shape[2], x_in.shape[3])
+                        xk = torch.cat(
+                            [xk, ex.expand(nb * 2, -1, -1, -1)], dim=1
+                        )
+                    eps = _raw_fn(xk, ts, ctx2, y2)
+                    e_c, e_u = eps.chunk(2, dim=0)
+                    return _gops.lincomb(e_c, e_u, _wcfg, 1.0 - _wcfg)
+
+                _graphed = GraphedModelFn(_cfg_core, self.device)
+                self._wholestep_cache[wkey] = _graphed
+
+            _wrap = _graphed
 
             def model_fn(x_in: torch.Tensor, t: float) -> torch.Tensor:  # noqa: F811
                 ctx2, y2, _ws = _ctx_y_for(t)
-                _graphed.bind(ctx2, y2)
-                return _graphed(x_in, t)
+                _wrap.bind(ctx2, y2)
+                return _wrap(x_in, t)
 
         if req.refiner_model:
             # two-model handoff: t descends through the schedule, so the
