@@ -12,6 +12,7 @@ this is not a backend fallback).
 """
 from __future__ import annotations
 
+import gc
 import os
 from typing import Callable, Dict, Optional, Tuple
 
@@ -109,9 +110,17 @@ class GraphedModelFn:
                     self.core(sx, st, sctx, sy)
             torch.cuda.current_stream(self.device).wait_stream(side)
             torch.cuda.synchronize(self.device)
-            graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
-                out = self.core(sx, st, sctx, sy)
+            # a GC cycle-collect during capture frees CUDA memory inside
+            # the capture (hipFree -> hard abort); collect first, then
+            # hold GC off until the graph is sealed
+            gc.collect()
+            gc.disable()
+            try:
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    out = self.core(sx, st, sctx, sy)
+            finally:
+                gc.enable()
             e = _FnEntry(graph, sx, st, sctx, sy, out)
             e.ctx_src = id(self._ctx)
             self.cache[key] = e
@@ -176,9 +185,14 @@ class GraphedDenoiser:
                     self.fn(sx, sts, sctx, sy)
             torch.cuda.current_stream(self.device).wait_stream(side)
             torch.cuda.synchronize(self.device)
-            graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
-                out = self.fn(sx, sts, sctx, sy)
+            gc.collect()
+            gc.disable()  # see GraphedModelFn._capture
+            try:
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    out = self.fn(sx, sts, sctx, sy)
+            finally:
+                gc.enable()
             entry = _Entry(graph, sx, sts, sctx, sy, out)
             self.cache[key] = entry
             log.info("hipGraph captured for shape %s", key[0])
