@@ -105,7 +105,9 @@ class SDConv2d(nn.Conv2d):
             from .. import ops
 
             xc = x.contiguous(memory_format=torch.channels_last)
-            return ops.ups2x_conv3x3(xc, self._wprep(), self.bias)
+            return ops.ups2x_conv3x3(
+                xc, self._wprep(), self.bias, collect_gn=True
+            )
         x = torch.nn.functional.interpolate(
             x, scale_factor=2, mode="nearest"
         )
@@ -116,6 +118,7 @@ class SDConv2d(nn.Conv2d):
         x: torch.Tensor,
         residual: torch.Tensor | None = None,
         chan_bias: torch.Tensor | None = None,
+        collect_gn: bool = False,
     ) -> torch.Tensor:
         import torch.nn.functional as F
 
@@ -153,7 +156,7 @@ class SDConv2d(nn.Conv2d):
                     )
                 return ops.conv3x3(
                     xc, self._wprep(), self.bias, res, self.stride[0],
-                    chan_bias,
+                    chan_bias, collect_gn=collect_gn,
                 )
             if (
                 k == (3, 3)

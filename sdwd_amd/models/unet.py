@@ -89,9 +89,12 @@ class ResBlock(nn.Module):
         # the time-embedding channel bias rides conv1's epilogue; the
         # skip-residual add rides conv2's (both fused on GPU)
         emb_b = self.emb_proj(ops.silu(emb))
-        h = self.conv1(self.norm1(x), chan_bias=emb_b)
+        # collect_gn: the conv epilogues emit the NEXT GroupNorm's partial
+        # sums (norm2 consumes conv1's; the next block's norm1 consumes
+        # conv2's when the tensor reaches it unmodified)
+        h = self.conv1(self.norm1(x), chan_bias=emb_b, collect_gn=True)
         skip = self.skip(x)
-        return self.conv2(self.norm2(h), residual=skip)
+        return self.conv2(self.norm2(h), residual=skip, collect_gn=True)
 
 
 class CrossAttention(nn.Module):

@@ -59,8 +59,10 @@ class VAEResBlock(nn.Module):
         )
 
     def forward(self, x):
-        h = self.conv1(self.norm1(x))
-        return self.conv2(self.norm2(h), residual=self.skip(x))
+        h = self.conv1(self.norm1(x), collect_gn=True)
+        return self.conv2(
+            self.norm2(h), residual=self.skip(x), collect_gn=True
+        )
 
 
 class VAEAttention(nn.Module):

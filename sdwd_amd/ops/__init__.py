@@ -102,6 +102,18 @@ def group_norm_silu(
         )
         if not nhwc_ok and not x.is_contiguous():
             x = x.contiguous()
+        meta = getattr(x, "_sdwd_gnp", None)
+        if (
+            meta is not None
+            and nhwc_ok
+            and meta[2] == x._version
+            and meta[0].shape[-1] == c
+        ):
+            # stats come from the producing conv's epilogue side-channel:
+            # the full-tensor stats read is skipped entirely
+            return ext().group_norm_silu_pre(
+                x, weight, bias, groups, eps, silu, meta[0], meta[1]
+            )
         return ext().group_norm_silu(x, weight, bias, groups, eps, silu)
     out = F.group_norm(x.float(), groups, weight.float(), bias.float(), eps)
     if silu:
@@ -239,6 +251,15 @@ def scale(x: torch.Tensor, c: float) -> torch.Tensor:
     return (c * x.float()).to(x.dtype)
 
 
+def _attach_gn_partials(y: torch.Tensor, gnp: torch.Tensor) -> None:
+    """Ride the conv's per-tile (sum, sumsq) side-channel on the output
+    tensor; a downstream GroupNorm consumes it and skips its stats read.
+    The _version pin invalidates the cache if y is ever mutated in place."""
+    if gnp is not None and gnp.numel() > 0:
+        tpi = (y.shape[2] * y.shape[3]) // 128
+        y._sdwd_gnp = (gnp, tpi, y._version)
+
+
 def conv3x3(
     x: torch.Tensor,
     w_prep: torch.Tensor,
@@ -246,10 +267,19 @@ def conv3x3(
     residual: Optional[torch.Tensor],
     stride: int = 1,
     chan_bias: Optional[torch.Tensor] = None,
+    collect_gn: bool = False,
 ) -> torch.Tensor:
     """NHWC implicit-GEMM 3x3 conv (pad 1); bias, an optional residual and
     an optional per-(sample, channel) bias (the ResBlock time-embedding
-    projection) all fused into the epilogue. GPU-only entry."""
+    projection) all fused into the epilogue. With collect_gn the epilogue
+    also emits per-tile GroupNorm partial sums (see _attach_gn_partials).
+    GPU-only entry."""
+    if collect_gn:
+        y, gnp = ext().conv3x3_nhwc_gn(
+            x, w_prep, bias, residual, chan_bias, stride
+        )
+        _attach_gn_partials(y, gnp)
+        return y
     return ext().conv3x3_nhwc(x, w_prep, bias, residual, chan_bias, stride)
 
 
@@ -272,9 +302,14 @@ def ups2x_conv3x3(
     x: torch.Tensor,
     w_prep: torch.Tensor,
     bias: Optional[torch.Tensor],
+    collect_gn: bool = False,
 ) -> torch.Tensor:
     """Nearest-2x upsample fused into a 3x3 conv (VAE decoder / UNet
     Upsample): no 4x intermediate tensor. GPU-only entry."""
+    if collect_gn:
+        y, gnp = ext().ups2x_conv3x3_gn(x, w_prep, bias)
+        _attach_gn_partials(y, gnp)
+        return y
     return ext().ups2x_conv3x3(x, w_prep, bias)
 
 

@@ -57,7 +57,12 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
     const __hip_bfloat16 *__restrict__ Zero,// >=16B of zeros
     __hip_bfloat16 *__restrict__ Y,         // [N,Ho,Wo,ldY] (+col offset)
     int Nn, int H, int W, int Cin, int Cout, int Ho, int Wo, int stride,
-    int ldY) {
+    int ldY, float *__restrict__ GNP = nullptr) {
+  // GNP != null: emit per-(M-tile, channel) partial (sum, sumsq) of the
+  // STORED values into GNP[mtile][2][ldY] so GroupNorm's stats pass can
+  // skip its full-tensor read (host enables only when Ho*Wo % BM == 0,
+  // i.e. every tile lies inside one image). Deterministic: wave shfl
+  // reduce + fixed-order cross-wave combine (no atomics).
   // one __shared__ object: [buf0: A(8K elems) B(8K)][buf1: A B]
   __shared__ __align__(16) __bf16 smem[4 * TILE_ELEMS];
 
@@ -206,6 +211,9 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
 
   // ---- epilogue: bias, channel-bias, residual, store --------------------
   const int r4 = (lane / 16) * 4;
+  float gs[NJ], gq[NJ];
+#pragma unroll
+  for (int j = 0; j < NJ; ++j) gs[j] = gq[j] = 0.0f;
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
 #pragma unroll
@@ -223,8 +231,48 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
           v += (float)CB[(long)ni * ldY + co];
         }
         if (HAS_RES) v += (float)Res[m * ldY + co];
-        Y[m * ldY + co] = f2bf(v);
+        const __hip_bfloat16 vb = f2bf(v);
+        Y[m * ldY + co] = vb;
+        if (GNP) {
+          const float vr = bf2f(vb);  // stats over the STORED values
+          gs[j] += vr;
+          gq[j] += vr * vr;
+        }
       }
+    }
+  }
+  if (GNP) {
+    // wave reduce: fold the 4 lane-quarters holding one column
+    float *lds = (float *)smem;  // staging LDS is free after the k-loop
+#pragma unroll
+    for (int j = 0; j < NJ; ++j) {
+#pragma unroll
+      for (int off = 16; off < 64; off <<= 1) {
+        gs[j] += __shfl_down(gs[j], off, WAVE);
+        gq[j] += __shfl_down(gq[j], off, WAVE);
+      }
+      // lanes 0..15 hold the wave's column sums -> LDS [wid][j][l16][2]
+      if (lane < 16) {
+        float *slot = lds + ((wid * NJ + j) * 16 + lane) * 2;
+        slot[0] = gs[j];
+        slot[1] = gq[j];
+      }
+    }
+    __syncthreads();
+    // fixed-order cross-wave combine: col c's two waves are
+    // {c/(BN/2), +2} (wave grid 2Mx2N); one thread per column writes
+    const long mtile = (long)blockIdx.y + (long)blockIdx.z * 32768;
+    for (int c = tid; c < BN; c += 256) {
+      const int co = n0 + c;
+      if (co >= Cout) continue;
+      const int h = c / (BN / 2);
+      const int jj = (c % (BN / 2)) / 16;
+      const int lc = c % 16;
+      const float *s0 = lds + ((h * NJ + jj) * 16 + lc) * 2;
+      const float *s1 = lds + (((h + 2) * NJ + jj) * 16 + lc) * 2;
+      float *dst = GNP + (mtile * 2) * (long)ldY;
+      dst[co] = s0[0] + s1[0];
+      dst[ldY + co] = s0[1] + s1[1];
     }
   }
 }
@@ -233,11 +281,11 @@ __launch_bounds__(256, 2) __global__ void conv3x3_nhwc_bf16_kernel(
 #ifdef __HIP_PLATFORM_AMD__
 bool conv3x3_supported(long cin) { return cin % 64 == 0 && cin >= 64; }
 
-torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
-                           c10::optional<torch::Tensor> bias,
-                           c10::optional<torch::Tensor> residual,
-                           c10::optional<torch::Tensor> chan_bias,
-                           long stride) {
+torch::Tensor conv3x3_nhwc_impl(torch::Tensor x, torch::Tensor w_prep,
+                                c10::optional<torch::Tensor> bias,
+                                c10::optional<torch::Tensor> residual,
+                                c10::optional<torch::Tensor> chan_bias,
+                                long stride, torch::Tensor *gnp_out) {
   // x: [N,C,H,W] channels_last; w_prep: [Cout,3,3,Cin] contiguous
   TORCH_CHECK(x.is_contiguous(torch::MemoryFormat::ChannelsLast),
               "conv3x3: x must be channels_last");
@@ -281,6 +329,15 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
     cbptr = (const __hip_bfloat16 *)cbt.data_ptr();
   }
   auto stream = cur_stream();
+  // GroupNorm partial side-channel (one (sum,sumsq) pair per 128-row tile
+  // per channel): only when every 128-row tile lies inside one image
+  float *gnp_ptr = nullptr;
+  if (gnp_out != nullptr && ((long)Ho * Wo) % 128 == 0) {
+    *gnp_out = torch::empty(
+        {M / 128, 2, (long)Cout},
+        x.options().dtype(torch::kFloat));
+    gnp_ptr = gnp_out->data_ptr<float>();
+  }
   // v2 (2 blocks/CU, 2-buffer glds) measured faster than the deeper v3
   // pipeline (18.8 vs 21.1 ms on the shape set): at 2 blocks/CU the
   // block-level overlap already hides the DMA. v3 stays opt-in.
@@ -347,7 +404,8 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
                        rptr, cbptr,
                        (const __hip_bfloat16 *)zero_page.data_ptr(),
                        (__hip_bfloat16 *)y.data_ptr(), N, H, W, Cin,
-                       (int)(nfull * 256), Cout, Ho, Wo, (int)stride);
+                       (int)(nfull * 256), Cout, Ho, Wo, (int)stride,
+                       gnp_ptr);
   }
   if (rem > 0) {
     const long co0 = nfull * 256;
@@ -374,13 +432,35 @@ torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
         rptr ? rptr + co0 : nullptr, cbptr ? cbptr + co0 : nullptr,
         (const __hip_bfloat16 *)zero_page.data_ptr(),
         (__hip_bfloat16 *)y.data_ptr() + co0, N, H, W, Cin, rem, Ho, Wo,
-        (int)stride, Cout);
+        (int)stride, Cout, gnp_ptr ? gnp_ptr + co0 : nullptr);
   }
   return y;
 }
 
-torch::Tensor ups2x_conv3x3(torch::Tensor x, torch::Tensor w_prep,
-                            c10::optional<torch::Tensor> bias) {
+torch::Tensor conv3x3_nhwc(torch::Tensor x, torch::Tensor w_prep,
+                           c10::optional<torch::Tensor> bias,
+                           c10::optional<torch::Tensor> residual,
+                           c10::optional<torch::Tensor> chan_bias,
+                           long stride) {
+  return conv3x3_nhwc_impl(x, w_prep, bias, residual, chan_bias, stride,
+                           nullptr);
+}
+
+std::vector<torch::Tensor> conv3x3_nhwc_gn(
+    torch::Tensor x, torch::Tensor w_prep,
+    c10::optional<torch::Tensor> bias,
+    c10::optional<torch::Tensor> residual,
+    c10::optional<torch::Tensor> chan_bias, long stride) {
+  torch::Tensor gnp;
+  auto y = conv3x3_nhwc_impl(x, w_prep, bias, residual, chan_bias, stride,
+                             &gnp);
+  if (!gnp.defined()) gnp = torch::empty({0});
+  return {y, gnp};
+}
+
+torch::Tensor ups2x_conv3x3_impl(torch::Tensor x, torch::Tensor w_prep,
+                                 c10::optional<torch::Tensor> bias,
+                                 torch::Tensor *gnp_out) {
   // nearest-2x upsample fused into the 3x3 conv (VAE decoder / UNet
   // Upsample blocks): output is [N, Cout, 2H, 2W]
   TORCH_CHECK(x.is_contiguous(torch::MemoryFormat::ChannelsLast));
@@ -408,6 +488,12 @@ torch::Tensor ups2x_conv3x3(torch::Tensor x, torch::Tensor w_prep,
     bf32 = bias->to(torch::kFloat).contiguous();
     bptr = bf32.data_ptr<float>();
   }
+  float *gnp_ptr = nullptr;
+  if (gnp_out != nullptr && ((long)Ho * Wo) % 128 == 0) {
+    *gnp_out = torch::empty(
+        {M / 128, 2, (long)Cout}, x.options().dtype(torch::kFloat));
+    gnp_ptr = gnp_out->data_ptr<float>();
+  }
   auto kern = has_b
                   ? conv3x3_nhwc_bf16_kernel<true, false, false, 128, true>
                   : conv3x3_nhwc_bf16_kernel<false, false, false, 128,
@@ -418,7 +504,21 @@ torch::Tensor ups2x_conv3x3(torch::Tensor x, torch::Tensor w_prep,
                      nullptr, nullptr,
                      (const __hip_bfloat16 *)zero_page.data_ptr(),
                      (__hip_bfloat16 *)y.data_ptr(), N, H, W, Cin, Cout,
-                     Ho, Wo, 1, Cout);
+                     Ho, Wo, 1, Cout, gnp_ptr);
   return y;
+}
+
+torch::Tensor ups2x_conv3x3(torch::Tensor x, torch::Tensor w_prep,
+                            c10::optional<torch::Tensor> bias) {
+  return ups2x_conv3x3_impl(x, w_prep, bias, nullptr);
+}
+
+std::vector<torch::Tensor> ups2x_conv3x3_gn(
+    torch::Tensor x, torch::Tensor w_prep,
+    c10::optional<torch::Tensor> bias) {
+  torch::Tensor gnp;
+  auto y = ups2x_conv3x3_impl(x, w_prep, bias, &gnp);
+  if (!gnp.defined()) gnp = torch::empty({0});
+  return {y, gnp};
 }
 #endif

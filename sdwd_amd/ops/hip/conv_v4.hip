@@ -40,7 +40,9 @@ __launch_bounds__(512, 2) __global__ void conv3x3_v4_kernel(
     const __hip_bfloat16 *__restrict__ Zero,
     __hip_bfloat16 *__restrict__ Y,         // [M,ldY] (+col offset)
     int Nn, int H, int W, int Cin, int Cols, int ldY, int Ho, int Wo,
-    int stride) {
+    int stride, float *__restrict__ GNP = nullptr) {
+  // GNP: per-(M-tile, channel) partial (sum, sumsq) of the stored values
+  // (see conv.hip - lets GroupNorm skip its full-tensor stats read).
   __shared__ __align__(16) __bf16 smem[4 * V4_ATILE];  // [buf][A|B]
 
   const long M = (long)Nn * Ho * Wo;
@@ -207,6 +209,9 @@ __launch_bounds__(512, 2) __global__ void conv3x3_v4_kernel(
 
   // epilogue
   const int r4 = (lane / 16) * 4;
+  float gs[4], gq[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) gs[j] = gq[j] = 0.0f;
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
 #pragma unroll
@@ -224,8 +229,51 @@ __launch_bounds__(512, 2) __global__ void conv3x3_v4_kernel(
           v += (float)CB[(long)ni * ldY + co];
         }
         if (HAS_RES) v += (float)Res[m * ldY + co];
-        Y[m * ldY + co] = f2bf(v);
+        const __hip_bfloat16 vb = f2bf(v);
+        Y[m * ldY + co] = vb;
+        if (GNP) {
+          const float vr = bf2f(vb);
+          gs[j] += vr;
+          gq[j] += vr * vr;
+        }
       }
+    }
+  }
+  if (GNP) {
+    float *lds = (float *)smem;  // free after the k-loop
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+#pragma unroll
+      for (int off = 16; off < 64; off <<= 1) {
+        gs[j] += __shfl_down(gs[j], off, WAVE);
+        gq[j] += __shfl_down(gq[j], off, WAVE);
+      }
+      if (lane < 16) {
+        float *slot = lds + ((wid * 4 + j) * 16 + lane) * 2;
+        slot[0] = gs[j];
+        slot[1] = gq[j];
+      }
+    }
+    __syncthreads();
+    // GNP granularity is 128 rows (the v2 kernel's BM, so the remainder
+    // launch shares the layout): this 256-row tile's two wave M-halves
+    // write two consecutive 128-row partial slots
+    const long mt128 = m0 / 128;
+    for (int c = tid; c < V4_BN; c += 512) {
+      const int co = n0 + c;
+      if (co >= Cols) continue;
+      const int q = c / 64;                      // n-quarter wave
+      const int o = c % 64;
+      const int jj = (o / 32) * 2 + (o % 32) / 16;
+      const int lc = c % 16;
+      const float *s0 = lds + ((q * 4 + jj) * 16 + lc) * 2;       // rows 0-127
+      const float *s1 = lds + (((q + 4) * 4 + jj) * 16 + lc) * 2; // 128-255
+      float *d0 = GNP + (mt128 * 2) * (long)ldY;
+      float *d1 = GNP + ((mt128 + 1) * 2) * (long)ldY;
+      d0[co] = s0[0];
+      d0[ldY + co] = s0[1];
+      d1[co] = s1[0];
+      d1[ldY + co] = s1[1];
     }
   }
 }

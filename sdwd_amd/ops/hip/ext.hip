@@ -149,6 +149,42 @@ torch::Tensor group_norm_silu_nhwc(torch::Tensor x, torch::Tensor w,
   return out;
 }
 
+torch::Tensor group_norm_silu_pre(torch::Tensor x, torch::Tensor w,
+                                  torch::Tensor b, long groups, double eps,
+                                  bool silu_act, torch::Tensor gnp,
+                                  long tiles_per_image) {
+  // GroupNorm with conv-epilogue-emitted partials: skips the full-tensor
+  // stats read (gnp layout [n*tpi][2][C] == the stats kernel's [n][S][2][C])
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(x.is_contiguous(torch::MemoryFormat::ChannelsLast));
+  const long N = x.size(0);
+  const int C = x.size(1);
+  const long HW = x.size(2) * (long)x.size(3);
+  TORCH_CHECK(gnp.scalar_type() == torch::kFloat && gnp.is_contiguous());
+  TORCH_CHECK(gnp.numel() == N * tiles_per_image * 2 * C,
+              "gn partials shape mismatch");
+  auto wf = w.to(torch::kFloat).contiguous();
+  auto bf = b.to(torch::kFloat).contiguous();
+  auto out = torch::empty_like(x);
+  auto opts = torch::TensorOptions().dtype(torch::kFloat).device(x.device());
+  auto stats = torch::empty({N * (long)groups, 2L}, opts);
+  auto stream = cur_stream();
+  hipLaunchKernelGGL(gn_nhwc_stats, dim3((unsigned)(N * groups)),
+                     dim3(WAVE), 0, stream, gnp.data_ptr<float>(),
+                     stats.data_ptr<float>(), C, HW, (int)groups,
+                     (int)tiles_per_image, (float)eps);
+  const long total = N * HW * (C / 8);
+  auto kern =
+      silu_act ? gn_nhwc_norm_bf16<true> : gn_nhwc_norm_bf16<false>;
+  hipLaunchKernelGGL(kern, dim3(ew_grid(total)), dim3(256), 0, stream,
+                     (const __hip_bfloat16 *)x.data_ptr(),
+                     stats.data_ptr<float>(), wf.data_ptr<float>(),
+                     bf.data_ptr<float>(),
+                     (__hip_bfloat16 *)out.data_ptr(), C, HW, (int)groups,
+                     N);
+  return out;
+}
+
 torch::Tensor group_norm_silu(torch::Tensor x, torch::Tensor w,
                               torch::Tensor b, long groups, double eps,
                               bool silu_act) {
@@ -333,6 +369,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("axpby", &axpby);
   m.def("euler_step", &euler_step);
   m.def("group_norm_silu", &group_norm_silu);
+  m.def("group_norm_silu_pre", &group_norm_silu_pre);
+  m.def("conv3x3_nhwc_gn", &conv3x3_nhwc_gn);
+  m.def("ups2x_conv3x3_gn", &ups2x_conv3x3_gn);
   m.def("layer_norm", &layer_norm);
   m.def("add_layer_norm", &add_layer_norm);
   m.def("row_softmax_", &row_softmax_);

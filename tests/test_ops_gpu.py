@@ -375,6 +375,50 @@ class TestConv3x3:
         err = relerr(out.contiguous(), ref)
         assert err < 0.05, f"smallcin conv err {err}"
 
+    def test_gn_partials_match_standard(self, dev):
+        """Conv-epilogue GN partials: the fused-stats GroupNorm must match
+        the standard two-pass path (stats over the same stored values,
+        different summation order only) and the fp32 reference."""
+        torch.manual_seed(9)
+        x = torch.randn(2, 64, 32, 32, device=dev, dtype=torch.bfloat16)
+        conv = torch.nn.Conv2d(64, 64, 3, padding=1).to(dev, torch.bfloat16)
+        xc = x.contiguous(memory_format=torch.channels_last)
+        wprep = conv.weight.permute(0, 2, 3, 1).contiguous()
+        y = ops.conv3x3(xc, wprep, conv.bias, None, 1, collect_gn=True)
+        assert hasattr(y, "_sdwd_gnp"), "partials not attached"
+        w = torch.randn(64, device=dev)
+        b = torch.randn(64, device=dev)
+        fused = ops.group_norm_silu(y, w, b, 8, 1e-5, True)
+        std = ops.group_norm_silu(y.clone(), w, b, 8, 1e-5, True)
+        assert (fused.float() - std.float()).abs().max() <= 0.05
+        ref = torch.nn.functional.silu(
+            torch.nn.functional.group_norm(
+                y.float(), 8, w.float(), b.float()
+            )
+        )
+        assert relerr(fused.contiguous(), ref) < 0.03
+        # v4 path (big channels) + chan_bias/residual variants
+        x2 = torch.randn(1, 320, 32, 32, device=dev, dtype=torch.bfloat16)
+        c2 = torch.nn.Conv2d(320, 320, 3, padding=1).to(dev, torch.bfloat16)
+        res = torch.randn_like(x2).contiguous(
+            memory_format=torch.channels_last
+        )
+        y2 = ops.conv3x3(
+            x2.contiguous(memory_format=torch.channels_last),
+            c2.weight.permute(0, 2, 3, 1).contiguous(), c2.bias, res, 1,
+            collect_gn=True,
+        )
+        assert hasattr(y2, "_sdwd_gnp")
+        w2 = torch.randn(320, device=dev)
+        b2 = torch.randn(320, device=dev)
+        f2 = ops.group_norm_silu(y2, w2, b2, 32, 1e-5, True)
+        ref2 = torch.nn.functional.silu(
+            torch.nn.functional.group_norm(
+                y2.float(), 32, w2.float(), b2.float()
+            )
+        )
+        assert relerr(f2.contiguous(), ref2) < 0.03
+
     def test_fused_residual(self, dev):
         x = torch.randn(2, 64, 16, 16, device=dev, dtype=torch.bfloat16)
         res = torch.randn(2, 64, 16, 16, device=dev, dtype=torch.bfloat16)
