@@ -335,6 +335,6 @@ class UNetModel(nn.Module):
             h = h + control["mid"]
             skips = [s + c for s, c in zip(skips, control["down"])]
         for blk in self.up:
-            h = torch.cat([h, skips.pop()], dim=1)
+            h = ops.cat_channels_gn(h, skips.pop())
             h = blk(h, emb, context)
         return self.conv_out(self.norm_out(h))

@@ -260,6 +260,26 @@ def _attach_gn_partials(y: torch.Tensor, gnp: torch.Tensor) -> None:
         y._sdwd_gnp = (gnp, tpi, y._version)
 
 
+def cat_channels_gn(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Channel-concat that PRESERVES the conv-epilogue GroupNorm partials:
+    a cat along C concatenates the per-channel partial sums too, so the
+    UNet up-path's skip-concat keeps the stats side-channel alive."""
+    out = torch.cat([a, b], dim=1)
+    ma = getattr(a, "_sdwd_gnp", None)
+    mb = getattr(b, "_sdwd_gnp", None)
+    if (
+        ma is not None
+        and mb is not None
+        and ma[2] == a._version
+        and mb[2] == b._version
+        and ma[1] == mb[1]
+        and ma[0].shape[0] == mb[0].shape[0]
+    ):
+        gnp = torch.cat([ma[0], mb[0]], dim=-1)
+        out._sdwd_gnp = (gnp, ma[1], out._version)
+    return out
+
+
 def conv3x3(
     x: torch.Tensor,
     w_prep: torch.Tensor,
