@@ -599,10 +599,14 @@ static torch::Tensor flash_attention_raw(torch::Tensor q, torch::Tensor k,
     const char *e = getenv("SDWD_ATTN");
     return e && strcmp(e, "v2") == 0;
   }();
+  static const bool v3all = [] {  // measure v3 past DPAD 64 (spilly)
+    const char *e = getenv("SDWD_ATTN");
+    return e && strcmp(e, "v3all") == 0;
+  }();
 
 #define LAUNCH_FLASH(DP_)                                                   \
   do {                                                                      \
-    if (use_v2 || DP_ > 64) { /* v3 wins measured only at DPAD<=64 */      \
+    if (use_v2 || (DP_ > 64 && !v3all)) { /* v3 wins at DPAD<=64 */        \
       dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)(B * H));           \
       hipLaunchKernelGGL(flash_fwd_bf16_kernel<DP_>, grid, block, 0,        \
                          stream, (const __hip_bfloat16 *)q.data_ptr(),      \
