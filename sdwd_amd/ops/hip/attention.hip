@@ -606,7 +606,10 @@ static torch::Tensor flash_attention_raw(torch::Tensor q, torch::Tensor k,
 
 #define LAUNCH_FLASH(DP_)                                                   \
   do {                                                                      \
-    if (use_v2 || (DP_ > 64 && !v3all)) { /* v3 wins at DPAD<=64 */        \
+    /* measured dispatch: v3 wins every self-attention shape (at D>64     \
+       its prologue/epilogue spills are off the hot loop: +24% at D=80,  \
+       +7% at D=160); the short-Sk cross shapes (Sk=77) stay on v2 */    \
+    if (use_v2 || (DP_ > 64 && !v3all && Sk < 256)) {                    \
       dim3 grid((unsigned)((Sq + 255) / 256), (unsigned)(B * H));           \
       hipLaunchKernelGGL(flash_fwd_bf16_kernel<DP_>, grid, block, 0,        \
                          stream, (const __hip_bfloat16 *)q.data_ptr(),      \
