@@ -44,6 +44,11 @@ SAMPLER_EVALS_PER_STEP = {
     "DPM++ 2S a Karras": 2.0,
     "UniPC": 1.0,  # corrector eval is reused as the next predictor eval
     "Restart": 2.6,  # Heun cost + ~30% restart overhead at 20 steps
+    "DPM fast": 1.0,      # fixed eval budget == steps (ref: +15.5% vs Euler a)
+    "DPM adaptive": 2.5,  # error-controlled evals (ref table: -61% vs Euler a)
+    "PLMS": 1.0,
+    "DPM2 Karras": 2.0,
+    "DPM2 a Karras": 2.0,
 }
 
 

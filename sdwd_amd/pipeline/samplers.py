@@ -515,6 +515,176 @@ class LCM(Sampler):
         return ops.add_noise(denoised, noise_fn(), 1.0, sigma_next)
 
 
+class _DPMSolver(Sampler):
+    """Shared DPM-Solver (Lu et al. 2022) machinery in t = -log(sigma)
+    space, eps-model form (k-diffusion DPMSolver). The fast/adaptive
+    samplers below run their own t-grids, so sigma -> train-timestep goes
+    through the full sigma table rather than the step schedule."""
+
+    def __init__(self, schedule: Schedule):
+        super().__init__(schedule)
+        from .schedule import make_sigmas_full
+
+        self._table = make_sigmas_full()
+
+    def _tt(self, sigma: float) -> float:
+        from .schedule import _timesteps_for
+
+        return float(
+            _timesteps_for(torch.tensor([float(sigma)]), self._table)[0]
+        )
+
+    def _eps(self, model_fn, x, sigma: float):
+        den = _eval(model_fn, x, sigma, self._tt(sigma))
+        return ops.lincomb(x, den, 1.0 / sigma, -1.0 / sigma)
+
+    @staticmethod
+    def _sig(t: float) -> float:
+        return math.exp(-t)
+
+    def _step1(self, model_fn, x, t, tn, eps=None):
+        h = tn - t
+        if eps is None:
+            eps = self._eps(model_fn, x, self._sig(t))
+        return ops.lincomb(x, eps, 1.0, -self._sig(tn) * math.expm1(h)), eps
+
+    def _step2(self, model_fn, x, t, tn, eps=None, r1=0.5):
+        h = tn - t
+        if eps is None:
+            eps = self._eps(model_fn, x, self._sig(t))
+        s1 = t + r1 * h
+        u1 = ops.lincomb(
+            x, eps, 1.0, -self._sig(s1) * math.expm1(r1 * h)
+        )
+        eps_r1 = self._eps(model_fn, u1, self._sig(s1))
+        x2 = ops.lincomb(x, eps, 1.0, -self._sig(tn) * math.expm1(h))
+        d = ops.lincomb(eps_r1, eps, 1.0, -1.0)
+        x2 = ops.lincomb(
+            x2, d, 1.0, -self._sig(tn) / (2 * r1) * math.expm1(h)
+        )
+        return x2, eps
+
+    def _step3(self, model_fn, x, t, tn, eps=None, r1=1.0 / 3, r2=2.0 / 3):
+        h = tn - t
+        if eps is None:
+            eps = self._eps(model_fn, x, self._sig(t))
+        s1, s2 = t + r1 * h, t + r2 * h
+        u1 = ops.lincomb(
+            x, eps, 1.0, -self._sig(s1) * math.expm1(r1 * h)
+        )
+        eps_r1 = self._eps(model_fn, u1, self._sig(s1))
+        u2 = ops.lincomb(
+            x, eps, 1.0, -self._sig(s2) * math.expm1(r2 * h)
+        )
+        d1 = ops.lincomb(eps_r1, eps, 1.0, -1.0)
+        u2 = ops.lincomb(
+            u2, d1, 1.0,
+            -self._sig(s2) * (r2 / r1)
+            * (math.expm1(r2 * h) / (r2 * h) - 1.0),
+        )
+        eps_r2 = self._eps(model_fn, u2, self._sig(s2))
+        x3 = ops.lincomb(x, eps, 1.0, -self._sig(tn) * math.expm1(h))
+        d2 = ops.lincomb(eps_r2, eps, 1.0, -1.0)
+        x3 = ops.lincomb(
+            x3, d2, 1.0, -self._sig(tn) / r2 * (math.expm1(h) / h - 1.0)
+        )
+        return x3, eps
+
+
+class DPMFast(_DPMSolver):
+    """k-diffusion sample_dpm_fast: a fixed model-eval budget spent on a
+    uniform t-grid of mixed 3rd/2nd/1st-order DPM-Solver steps (the
+    reference's sampler-speed table lists it at +15.5% vs Euler a)."""
+
+    def sample(self, model_fn, x, noise_fn=None, callback=None,
+               interrupt=None, post_step=None):
+        sig = self.schedule.sigmas.tolist()
+        smax = sig[0]
+        smin = next(s for s in reversed(sig) if s > 0)
+        nfe = max(2, len(sig) - 1)
+        t0, t1 = -math.log(smax), -math.log(smin)
+        m = nfe // 3 + 1
+        ts = [t0 + (t1 - t0) * i / m for i in range(m + 1)]
+        if nfe % 3 == 0:
+            orders = [3] * (m - 2) + [2, 1]
+        else:
+            orders = [3] * (m - 1) + [nfe % 3]
+        steps = {1: self._step1, 2: self._step2, 3: self._step3}
+        for i in range(m):
+            if interrupt is not None and interrupt():
+                return x
+            x, _ = steps[orders[i]](model_fn, x, ts[i], ts[i + 1])
+            if post_step is not None:
+                x = post_step(x, self._sig(ts[i + 1]))
+            if callback is not None:
+                callback(i + 1, m)
+        return x
+
+
+class DPMAdaptive(_DPMSolver):
+    """k-diffusion sample_dpm_adaptive (order 3, I-controller defaults):
+    embedded 2nd/3rd-order error estimate drives the step size; the model
+    eval count is error-controlled, not the requested step count (the
+    reference's table lists it at -61% vs Euler a, i.e. ~2.5x the evals)."""
+
+    RTOL, ATOL = 0.05, 0.0078
+    H_INIT = 0.05
+    ACCEPT_SAFETY = 0.81
+    MAX_EVALS_FACTOR = 12  # runaway bound
+
+    def sample(self, model_fn, x, noise_fn=None, callback=None,
+               interrupt=None, post_step=None):
+        sig = self.schedule.sigmas.tolist()
+        smax = sig[0]
+        smin = next(s for s in reversed(sig) if s > 0)
+        t_end = -math.log(smin)
+        s_cur = -math.log(smax)
+        h = self.H_INIT
+        errs: list = []
+        x_prev = x
+        nominal = max(1, self.schedule.steps)
+        max_evals = self.MAX_EVALS_FACTOR * nominal
+        evals = done = 0
+        while s_cur < t_end - 1e-5 and evals < max_evals:
+            if interrupt is not None and interrupt():
+                return x
+            t = min(t_end, s_cur + h)
+            x_low, eps = self._step2(model_fn, x, s_cur, t, r1=1.0 / 3)
+            x_high, _ = self._step3(model_fn, x, s_cur, t, eps=eps)
+            evals += 5
+            delta = torch.maximum(
+                torch.full_like(x_low, self.ATOL),
+                self.RTOL
+                * torch.maximum(x_low.float().abs(), x_prev.float().abs()),
+            )
+            err = float(
+                torch.linalg.vector_norm(
+                    (x_low.float() - x_high.float()) / delta
+                )
+                / x_low.numel() ** 0.5
+            )
+            inv_err = 1.0 / (err + 1e-8)
+            if not errs:
+                errs = [inv_err] * 3
+            errs[0] = inv_err
+            factor = 1.0 + math.atan(errs[0] ** (1.0 / 3.0) - 1.0)
+            accept = factor >= self.ACCEPT_SAFETY
+            h *= factor
+            if accept:
+                errs[2], errs[1] = errs[1], errs[0]
+                x_prev = x_low
+                x = x_high
+                s_cur = t
+                if post_step is not None:
+                    x = post_step(x, self._sig(s_cur))
+                done += 1
+                if callback is not None:
+                    callback(min(done, nominal), nominal)
+        if callback is not None:
+            callback(nominal, nominal)
+        return x
+
+
 class LMS(Sampler):
     """Linear multistep (order <= 4) with exactly integrated Adams
     coefficients over each sigma interval (k-diffusion sample_lms)."""
@@ -580,6 +750,11 @@ SAMPLERS: Dict[str, type] = {
     "UniPC": UniPC,
     "LCM": LCM,
     "Restart": Restart,
+    "DPM fast": DPMFast,
+    "DPM adaptive": DPMAdaptive,
+    "PLMS": LMS,  # ldm's pseudo-LMS: the multistep family on the discrete schedule
+    "DPM2 Karras": DPM2,
+    "DPM2 a Karras": DPM2Ancestral,
 }
 
 
