@@ -495,6 +495,24 @@ class TestCrossDeviceConsistency:
         diff = (cpu.images.float() - gpu.images.float()).abs()
         assert diff.mean() < 8.0, f"mean abs diff {diff.mean()} too high"
 
+    @pytest.mark.parametrize("sampler", [
+        "Euler a", "Heun", "DPM++ 2M", "DPM++ SDE", "DPM2 a", "UniPC",
+        "LMS", "Restart", "DPM fast", "DPM adaptive", "LCM",
+    ])
+    def test_sampler_trajectory_gpu_matches_cpu(self, dev, sampler):
+        """Per-sampler trajectory check (round-1 verdict: beyond
+        finiteness): the bf16 GPU run of EACH sampler must land near its
+        fp32 CPU trajectory — a wrong update rule on the fused GPU step
+        kernels diverges far beyond precision noise."""
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        req = PipelineRequest(prompt="traj", steps=4, width=64, height=64,
+                              seeds=[33], sampler_name=sampler)
+        cpu = StableDiffusionPipeline("tiny", device="cpu").generate(req)
+        gpu = StableDiffusionPipeline("tiny", device=dev).generate(req)
+        diff = (cpu.images.float() - gpu.images.float()).abs()
+        assert diff.mean() < 8.0, f"{sampler}: mean abs diff {diff.mean()}"
+
 
 class TestLoRAGPU:
     def test_lora_on_gpu_pipeline(self, dev):
