@@ -380,8 +380,12 @@ torch::Tensor conv3x3_nhwc_impl(torch::Tensor x, torch::Tensor w_prep,
   // on the SD-UNet level-0/1 shapes (Cin<=960, Cout<=640, big M); the
   // 1280-channel and VAE shapes stay on v2 (2 blocks/CU beats the deep
   // pipeline there).
+  // ... and only when the 512-thread 1-block/CU grid actually fills the
+  // 256 CUs (small shards at N=8 under-fill 256-row tiles: the 128-row
+  // v2 kernel at 2 blocks/CU wins there)
   const bool v4_shape =
-      Cin <= 960 && Cout <= 640 && Cin != 512 && Cout >= 256;
+      Cin <= 960 && Cout <= 640 && Cin != 512 && Cout >= 256 &&
+      ((M + V4_BM - 1) / V4_BM) * (Cout / 256) >= 256;
   const long nfull = (v2only || !v4_shape) ? 0 : Cout / 256;
   const int rem = (int)(Cout - nfull * 256);
   if (nfull > 0) {
