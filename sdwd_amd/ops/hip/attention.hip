@@ -243,7 +243,10 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
   constexpr int NC = DPAD / 16;
   constexpr int DV = (DPAD + 31) / 32 * 32;
   constexpr int ND = DV / 32;
-  constexpr int PADV = (DV == 128) ? 24 : 8;  // tr-read conflict-free pad
+  // tr-read conflict pad: the residual ~6% SQ_LDS_BANK_CONFLICT is
+  // tr-instruction-intrinsic (a +16 stride variant measured identical;
+  // guide T10: addr swizzles don't move tr_read conflict classes)
+  constexpr int PADV = (DV == 128) ? 24 : 8;
   constexpr int RSV = DV + PADV;              // vt2 row stride (elements)
   constexpr int NG = KVB * (DPAD / 8);        // bf16x8 pieces per operand
   constexpr int NST = 2 * NG / 256;           // pieces/thread (exact: 16*DPAD%256==0)
