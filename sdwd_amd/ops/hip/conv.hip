@@ -413,7 +413,11 @@ torch::Tensor conv3x3_nhwc_impl(torch::Tensor x, torch::Tensor w_prep,
   }
   if (rem > 0) {
     const long co0 = nfull * 256;
-    const bool bn64 = rem <= 64;  // exact 64-col tile for the remainder
+    // BN=64 when the remainder is one 64-tile OR when the BN=128 grid
+    // would under-fill the 256 CUs (small shards: twice the blocks beat
+    // the halved B-reuse when occupancy-starved)
+    const bool bn64 =
+        rem <= 64 || mtiles * ((rem + 127) / 128) < 512;
 #define PICK(B_, R_, C_)                                              \
   (bn64 ? conv3x3_nhwc_bf16_kernel<B_, R_, C_, 64>                    \
         : conv3x3_nhwc_bf16_kernel<B_, R_, C_, 128>)
