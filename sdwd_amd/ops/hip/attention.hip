@@ -519,8 +519,15 @@ __launch_bounds__(256, 2) __global__ void flash_fwd_bf16_v3(
                            : "=&v"(tv[dd][0]), "=&v"(tv[dd][1])
                            : "v"(a0), "v"(a0 + (unsigned)(8 * RSV)));
             }
-            asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-            __builtin_amdgcn_sched_barrier(0);
+            // the wait NAMES the tr destinations (guide §5.7 form ii):
+            // the consuming MFMAs then depend on it through registers, so
+            // no sched_barrier wall is needed and independent work (the
+            // other q-subtile's chain) may schedule into the LDS shadow
+            asm volatile("s_waitcnt lgkmcnt(0)"
+                         : "+v"(tv[0][0]), "+v"(tv[0][1]), "+v"(tv[1][0]),
+                           "+v"(tv[1][1])
+                         :
+                         : "memory");
             __builtin_amdgcn_s_setprio(1);
 #pragma unroll
             for (int dd = 0; dd < DB; ++dd) {
