@@ -268,3 +268,64 @@ class TestDynamicPrompts:
         assert not has_dynamic_syntax("plain prompt")
         assert has_dynamic_syntax("{a|b}")
         assert has_dynamic_syntax("__cards__")
+
+
+class TestGnPartialHandshake:
+    """The conv->GroupNorm partials side-channel plumbing (pure logic;
+    kernel numerics are covered on GPU)."""
+
+    def _fake(self, t, tiles, c):
+        import torch
+
+        gnp = torch.zeros(tiles, 2, c)
+        t._sdwd_gnp = (gnp, 1, t._version)
+        return gnp
+
+    def test_cat_merges_valid_partials(self):
+        import torch
+
+        from sdwd_amd.ops import cat_channels_gn
+
+        a = torch.zeros(1, 8, 16, 16)
+        b = torch.zeros(1, 4, 16, 16)
+        ga = self._fake(a, 2, 8)
+        gb = self._fake(b, 2, 4)
+        out = cat_channels_gn(a, b)
+        assert out.shape[1] == 12
+        meta = getattr(out, "_sdwd_gnp", None)
+        assert meta is not None
+        assert meta[0].shape == (2, 2, 12)
+        assert torch.equal(meta[0][:, :, :8], ga)
+        assert torch.equal(meta[0][:, :, 8:], gb)
+
+    def test_cat_drops_on_missing_or_stale(self):
+        import torch
+
+        from sdwd_amd.ops import cat_channels_gn
+
+        a = torch.zeros(1, 8, 16, 16)
+        b = torch.zeros(1, 4, 16, 16)
+        self._fake(a, 2, 8)  # b has no partials
+        out = cat_channels_gn(a, b)
+        assert getattr(out, "_sdwd_gnp", None) is None
+        # stale version: in-place mutation invalidates
+        a2 = torch.zeros(1, 8, 16, 16)
+        b2 = torch.zeros(1, 4, 16, 16)
+        self._fake(a2, 2, 8)
+        self._fake(b2, 2, 4)
+        a2.add_(1.0)  # bumps _version
+        out2 = cat_channels_gn(a2, b2)
+        assert getattr(out2, "_sdwd_gnp", None) is None
+
+    def test_cat_drops_on_tile_mismatch(self):
+        import torch
+
+        from sdwd_amd.ops import cat_channels_gn
+
+        a = torch.zeros(1, 8, 16, 16)
+        b = torch.zeros(1, 4, 16, 16)
+        self._fake(a, 2, 8)
+        gb = torch.zeros(3, 2, 4)
+        b._sdwd_gnp = (gb, 1, b._version)
+        out = cat_channels_gn(a, b)
+        assert getattr(out, "_sdwd_gnp", None) is None
