@@ -315,7 +315,13 @@ def conv3x3_small(
 
 
 def conv3x3_small_supported(cin: int, cout: int) -> bool:
-    return int(cin) in (3, 4, 9) and cout % 8 == 0 and cout <= 1536
+    # LDS weight cache (9*Cin*Cout bf16) must fit the 160 KiB CU budget
+    return (
+        int(cin) in (3, 4, 9)
+        and cout % 8 == 0
+        and cout <= 1536
+        and 9 * int(cin) * int(cout) * 2 <= 160 * 1024
+    )
 
 
 def ups2x_conv3x3(

@@ -88,8 +88,9 @@ __launch_bounds__(256, 4) __global__ void conv3x3_smallcin_kernel(
 // ---------------------------------------------------------------------------
 #if defined(__HIP_PLATFORM_AMD__) && !defined(SDWD_NO_TORCH)
 bool conv3x3_small_supported(long cin, long cout) {
+  // the [tap][Cout] LDS weight cache (9*Cin*Cout bf16) must fit a CU
   return (cin == 3 || cin == 4 || cin == 9) && cout % 8 == 0 &&
-         cout <= 1536;
+         cout <= 1536 && 9 * cin * cout * 2 <= 160 * 1024;
 }
 
 torch::Tensor conv3x3_small(torch::Tensor x, torch::Tensor w_prep,
