@@ -269,11 +269,13 @@ __launch_bounds__(512, 2) __global__ void conv3x3_v4_kernel(
       const float *s0 = lds + ((q * 4 + jj) * 16 + lc) * 2;       // rows 0-127
       const float *s1 = lds + (((q + 4) * 4 + jj) * 16 + lc) * 2; // 128-255
       float *d0 = GNP + (mt128 * 2) * (long)ldY;
-      float *d1 = GNP + ((mt128 + 1) * 2) * (long)ldY;
       d0[co] = s0[0];
       d0[ldY + co] = s0[1];
-      d1[co] = s1[0];
-      d1[ldY + co] = s1[1];
+      if ((mt128 + 1) * 128 < M + 1) {  // second half exists (M%256==128
+        float *d1 = GNP + ((mt128 + 1) * 2) * (long)ldY;  // edge guard)
+        d1[co] = s1[0];
+        d1[ldY + co] = s1[1];
+      }
     }
   }
 }
