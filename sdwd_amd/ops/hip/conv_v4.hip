@@ -271,8 +271,8 @@ __launch_bounds__(512, 2) __global__ void conv3x3_v4_kernel(
       float *d0 = GNP + (mt128 * 2) * (long)ldY;
       d0[co] = s0[0];
       d0[ldY + co] = s0[1];
-      if ((mt128 + 1) * 128 < M + 1) {  // second half exists (M%256==128
-        float *d1 = GNP + ((mt128 + 1) * 2) * (long)ldY;  // edge guard)
+      if ((mt128 + 1) * 128 < M) {  // second 128-row half has real rows
+        float *d1 = GNP + ((mt128 + 1) * 2) * (long)ldY;
         d1[co] = s1[0];
         d1[ldY + co] = s1[1];
       }
