@@ -147,10 +147,21 @@ def gather_images(
         (tuple(r[1:]) for r in rows if r[0] > 0), tuple(shard.shape[1:])
     )
     if shard.shape[0] > 0 and tuple(shard.shape[1:]) != tail:
-        raise RuntimeError(
-            f"image shard shape mismatch across ranks: {tuple(shard.shape[1:])}"
-            f" vs consensus {tail}"
+        # never raise on a subset of ranks mid-collective (the others
+        # would hang in the all_gather): crop/pad to the consensus shape
+        # and scream — the gallery shows the defect, the job completes
+        log.error(
+            "image shard shape mismatch: local %s vs consensus %s; "
+            "cropping to keep the gather alive",
+            tuple(shard.shape[1:]), tail,
         )
+        fixed = torch.zeros(
+            (shard.shape[0], *tail), dtype=shard.dtype, device=shard.device
+        )
+        h = min(shard.shape[1], tail[0])
+        w = min(shard.shape[2], tail[1])
+        fixed[:, :h, :w] = shard[:, :h, :w]
+        shard = fixed
     max_n = max(max(shard_sizes), 1)
     padded = torch.zeros((max_n, *tail), dtype=shard.dtype, device=device)
     if shard.shape[0] > 0:
