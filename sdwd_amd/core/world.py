@@ -170,14 +170,15 @@ class World:
         (ref world.py:418-557)."""
         realtime = list(workers)
         # Defer ranks whose single-image ETA would stall the rest beyond
-        # job_timeout (ref job_stall, world.py:363-376).
+        # job_timeout (ref job_stall, world.py:363-376). The candidate is
+        # the rank with the largest SINGLE-IMAGE time: the proportional
+        # split hands slow ranks near-zero shares, so ranking candidates
+        # by their share's eta (as round 1 did) never surfaces them and
+        # complementary production silently never fired.
         while len(realtime) > 1:
-            shares = self._proportional(realtime, total)
-            etas = {
-                w.label: self._predict(w, shares[w.label], request)
-                for w in realtime
-            }
-            worst = max(realtime, key=lambda w: etas[w.label])
+            worst = max(
+                realtime, key=lambda w: self._predict(w, 1, request)
+            )
             one_img = self._predict(worst, 1, request)
             others = [w for w in realtime if w is not worst]
             t_without = max(

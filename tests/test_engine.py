@@ -185,6 +185,48 @@ def _spawn(world_size, tmpdir, fail_rank=-1):
     )
 
 
+def _hetero_worker(rank, world_size, port, tmpdir):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from sdwd_amd.parallel import DistributedEngine, GenerationRequest
+
+    eng = DistributedEngine(model="tiny", backend="gloo")
+    # a hopeless rank (1000x slower): the planner defers it; with step
+    # scaling on it produces one complementary image at reduced steps
+    # (ref world.py:547-557). ipm magnitudes mirror the scheduler unit
+    # test [60, 60, 0.06] scaled to the 64x64/4-step eta factors.
+    eng.world.get_worker("gpu0").eta.avg_ipm = 0.1875
+    eng.world.get_worker("gpu1").eta.avg_ipm = 0.1875
+    eng.world.get_worker("gpu2").eta.avg_ipm = 0.0001875
+    eng.world.settings.job_timeout = 1.0
+    eng.world.settings.step_scaling = True
+    assert eng.world.settings.complement_production
+    res = eng.generate(
+        GenerationRequest(
+            prompt="hetero", batch_size=6, width=64, height=64, steps=4,
+            seed=4000,
+        )
+    )
+    if rank == 0:
+        # requested batch intact AND deterministic; bonus images beyond it
+        assert res.seeds[:6] == [4000 + i for i in range(6)]
+        assert res.images.shape[0] >= 6
+        assert any("(complementary)" in s for s in res.job_summary), (
+            res.job_summary
+        )
+        assert all(
+            res.images[i].float().std() > 0
+            for i in range(res.images.shape[0])
+        )
+        torch.save(res.images, os.path.join(tmpdir, "hetero.pt"))
+    from sdwd_amd.parallel import destroy_group
+
+    destroy_group()
+
+
 def _single_rank_reference(batch=4, seed=900, prompt="dist"):
     """1-process LocalEngine gallery for image-for-image comparison."""
     from sdwd_amd.parallel import LocalEngine, GenerationRequest
@@ -216,6 +258,31 @@ class TestDistributedEngine:
         # the 2-rank gallery matches a 1-rank run image-for-image
         ref = _single_rank_reference()
         diff = (ref.images.float() - gallery.float()).abs()
+        assert diff.max() <= 1.0
+
+    def test_three_rank_complementary_production(self, tmp_path):
+        """A deferred slow rank produces bonus images through the
+        DISTRIBUTED engine (ref world.py:519-557 semantics at N>1)."""
+        import torch.multiprocessing as mp
+        import socket
+
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        mp.start_processes(
+            _hetero_worker, args=(3, port, str(tmp_path)), nprocs=3,
+            start_method="spawn", join=True,
+        )
+        gallery = torch.load(tmp_path / "hetero.pt")
+        assert gallery.shape[0] >= 6
+        # the first 6 slots match a 1-rank run image-for-image
+        from sdwd_amd.parallel import GenerationRequest, LocalEngine
+
+        one = LocalEngine(model="tiny", devices=["cpu"]).generate(
+            GenerationRequest(prompt="hetero", batch_size=6, width=64,
+                              height=64, steps=4, seed=4000)
+        )
+        diff = (one.images.float() - gallery[:6].float()).abs()
         assert diff.max() <= 1.0
 
     def test_two_rank_failure_recovery(self, tmp_path):
