@@ -163,7 +163,7 @@ def _resize_init(img: torch.Tensor, w: int, h: int, mode: int) -> torch.Tensor:
 # windows) and Dynamic Prompts (wildcard/variant expansion). Everything
 # else is logged and skipped, like the reference's compat filter when a
 # remote lacked the script - docs/usage.md documents the boundary.
-_NATIVE_ALWAYSON = ("controlnet", "dynamic prompts")
+_NATIVE_ALWAYSON = ("controlnet", "dynamic prompts", "soft inpainting")
 
 
 def _dynamic_prompts_enabled(alwayson: Dict[str, Any]) -> bool:
@@ -188,6 +188,43 @@ def _expand_dynamic(gen) -> None:
     gen.prompts = [
         expand(gen.prompt, gen.seed + i) for i in range(gen.batch_size)
     ]
+
+
+_SI_KEYS = (  # host UI labels -> GenerationRequest fields, in API arg order
+    ("Schedule bias", "si_schedule_bias"),
+    ("Preservation strength", "si_preservation_strength"),
+    ("Transition contrast boost", "si_transition_contrast_boost"),
+    ("Mask influence", "si_mask_influence"),
+    ("Difference threshold", "si_difference_threshold"),
+    ("Difference contrast", "si_difference_contrast"),
+)
+
+
+def _parse_soft_inpainting(alwayson: Dict[str, Any]) -> Dict[str, float]:
+    """Host built-in soft-inpainting payload:
+    {"soft inpainting": {"args": [{"Soft inpainting": true,
+    "Schedule bias": 1, ...}]}} — a single dict arg keyed by UI label
+    (the host's API convention), or the same values positionally with a
+    leading enabled flag. Returns GenerationRequest field overrides."""
+    for name, body in (alwayson or {}).items():
+        if name.lower().replace("-", " ") != "soft inpainting":
+            continue
+        args = (body or {}).get("args", [])
+        out: Dict[str, float] = {"soft_inpainting": True}
+        if args and isinstance(args[0], dict):
+            d = args[0]
+            if not d.get("Soft inpainting", True):
+                return {}
+            for label, field in _SI_KEYS:
+                if label in d:
+                    out[field] = float(d[label])
+        elif args:
+            if not args[0]:
+                return {}
+            for (_, field), v in zip(_SI_KEYS, args[1:]):
+                out[field] = float(v)
+        return out
+    return {}
 
 
 def _parse_controlnet(alwayson: Dict[str, Any]):
@@ -460,6 +497,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             inpainting_fill=req.inpainting_fill,
             inpainting_mask_invert=req.inpainting_mask_invert,
             color_correction=req.color_correction,
+            **_parse_soft_inpainting(req.alwayson_scripts),
         )
         return run_generation(gen, req.send_images, req.save_images)
 

@@ -92,6 +92,14 @@ class GenerationRequest:
     s_min_uncond: float = 0.0
     # sdwui img2img color correction: match output statistics to the init
     color_correction: bool = False
+    # soft inpainting (host built-in; see PipelineRequest for semantics)
+    soft_inpainting: bool = False
+    si_schedule_bias: float = 1.0
+    si_preservation_strength: float = 0.5
+    si_transition_contrast_boost: float = 4.0
+    si_mask_influence: float = 0.0
+    si_difference_threshold: float = 0.5
+    si_difference_contrast: float = 2.0
 
     def sched(self) -> GenRequest:
         return GenRequest(
@@ -172,6 +180,13 @@ def _job_pipeline_request(
         s_tmax=gen.s_tmax,
         s_noise=gen.s_noise,
         s_min_uncond=gen.s_min_uncond,
+        soft_inpainting=gen.soft_inpainting,
+        si_schedule_bias=gen.si_schedule_bias,
+        si_preservation_strength=gen.si_preservation_strength,
+        si_transition_contrast_boost=gen.si_transition_contrast_boost,
+        si_mask_influence=gen.si_mask_influence,
+        si_difference_threshold=gen.si_difference_threshold,
+        si_difference_contrast=gen.si_difference_contrast,
     )
 
 

@@ -58,6 +58,17 @@ class PipelineRequest:
     # sdwui "Masked content": 1 original (default), 2 latent noise,
     # 3 latent nothing ("fill"=0 is approximated upstream in the engine)
     inpainting_fill: int = 1
+    # Soft inpainting (the webui-host built-in the reference forwards as an
+    # alwayson-script payload): per-step SOFT latent blending instead of
+    # hard-mask pinning, plus a pixel-space composite against the decoded
+    # init at the end. Parameter names follow the host UI labels.
+    soft_inpainting: bool = False
+    si_schedule_bias: float = 1.0          # >1: repaint later / tighter band
+    si_preservation_strength: float = 0.5  # latent-magnitude restoration 0..1
+    si_transition_contrast_boost: float = 4.0  # sharpens the blend band
+    si_mask_influence: float = 0.0         # pixel composite: mask vs diff
+    si_difference_threshold: float = 0.5   # pixel diff at which gate = 0.5
+    si_difference_contrast: float = 2.0    # sharpness of the pixel gate
     # hires fix (sdwui two-pass: base gen -> latent upscale -> img2img pass)
     enable_hr: bool = False
     hr_scale: float = 2.0
@@ -123,6 +134,16 @@ def _slerp(a: torch.Tensor, b: torch.Tensor, t: float) -> torch.Tensor:
             math.sin(t * omega) / so
         ) * bf
     return out.reshape(a.shape).to(a.dtype)
+
+
+def _contrast(w: torch.Tensor, g: float) -> torch.Tensor:
+    """Monotone contrast curve w^g / (w^g + (1-w)^g): fixed points at
+    0, 0.5 and 1; g=1 is the identity, g>1 pushes values toward 0/1
+    (sharpens a soft band without moving its midline)."""
+    if g == 1.0:
+        return w
+    wg = w.clamp(0, 1).pow(g)
+    return wg / (wg + (1.0 - w.clamp(0, 1)).pow(g)).clamp_min(1e-8)
 
 
 _HR_MODES = {
@@ -321,6 +342,55 @@ class StableDiffusionPipeline:
         if lat_mask.shape[0] == 1 and b > 1:
             lat_mask = lat_mask.expand(b, -1, -1, -1)
         return lat_mask
+
+    def _soft_composite(
+        self, images: torch.Tensor, req: "PipelineRequest"
+    ) -> torch.Tensor:
+        """Soft-inpainting pixel composite: revert pixels the denoise
+        barely changed to the decoded original, gated by
+          gate = contrast(|new - orig| / (2 * difference_threshold))
+        (gate = 0.5 exactly at the threshold) and pulled toward the raw
+        mask by `mask influence`. Kills VAE round-trip drift outside the
+        repaint region while keeping genuinely repainted pixels."""
+        orig = self.model.vae.decode(
+            req.init_latents.to(self.device, self.dtype)
+        )
+        orig = (
+            ((orig.float() + 1.0) * 127.5)
+            .clamp(0, 255)
+            .to(torch.uint8)
+            .permute(0, 2, 3, 1)
+            .cpu()
+        )
+        if orig.shape != images.shape:  # defensive: odd sizes
+            return images
+        new_f = images.float()
+        diff = (new_f - orig.float()).abs().mean(dim=-1, keepdim=True) / 255.0
+        thr = max(float(req.si_difference_threshold), 1e-3)
+        gate = _contrast(
+            (diff / (2.0 * thr)).clamp(0, 1),
+            max(float(req.si_difference_contrast), 1.0),
+        )
+        mk = req.mask_image
+        if mk.dim() == 2:
+            mk = mk[None]
+        mk = torch.nn.functional.interpolate(
+            (mk.float() / 255.0)[:, None],
+            size=images.shape[1:3],
+            mode="bilinear",
+            align_corners=False,
+        ).clamp(0, 1)
+        if mk.shape[0] == 1 and images.shape[0] > 1:
+            mk = mk.expand(images.shape[0], -1, -1, -1)
+        mk = mk.permute(0, 2, 3, 1)
+        # the mask floors the weight (repainted pixels are never reverted);
+        # the difference gate only reclaims drift OUTSIDE the mask
+        w = mk + (1.0 - mk) * gate
+        mi = min(max(float(req.si_mask_influence), 0.0), 1.0)
+        if mi > 0.0:
+            w = (1.0 - mi) * w + mi * mk
+        out = w * new_f + (1.0 - w) * orig.float()
+        return out.round().clamp(0, 255).to(torch.uint8)
 
     # -- the denoise loop ----------------------------------------------------
     @torch.no_grad()
@@ -859,11 +929,46 @@ class StableDiffusionPipeline:
             init_lat = req.init_latents.to(self.device).float()
             noise_f32 = noise.float()
 
-            def post_step(xc, sigma_next):
-                keep = init_lat + noise_f32 * sigma_next
-                return (
-                    lat_mask * xc.float() + (1.0 - lat_mask) * keep
-                ).to(xc.dtype)
+            if req.soft_inpainting:
+                # Soft inpainting: the repaint weight is the soft mask
+                # value raised to a sigma-dependent exponent
+                #   w(m, s) = contrast(m ** (bias * (1 + 2*s/s_max)))
+                # so early steps (high sigma) strongly favour the original
+                # (context forms first) and the final blend keeps a soft
+                # m**bias edge instead of a hard 0/1 seam. The magnitude of
+                # the blended latent is then pulled toward the weighted mix
+                # of the input magnitudes (preservation strength), which
+                # counters the detail washout a plain lerp of latents
+                # causes in the transition band.
+                sig0 = max(float(sched.sigmas[0]), 1e-6)
+                bias = max(float(req.si_schedule_bias), 1e-3)
+                boost = max(float(req.si_transition_contrast_boost), 1.0)
+                keep_s = min(max(float(req.si_preservation_strength), 0.0), 1.0)
+
+                def post_step(xc, sigma_next):
+                    keep = init_lat + noise_f32 * sigma_next
+                    xf = xc.float()
+                    e = bias * (1.0 + 2.0 * float(sigma_next) / sig0)
+                    w = lat_mask.pow(e)
+                    w = _contrast(w, boost)
+                    mixed = w * xf + (1.0 - w) * keep
+                    if keep_s > 0.0:
+                        cur = mixed.norm(p=2, dim=1, keepdim=True)
+                        want = w * xf.norm(p=2, dim=1, keepdim=True) + (
+                            1.0 - w
+                        ) * keep.norm(p=2, dim=1, keepdim=True)
+                        ratio = (want / cur.clamp_min(1e-5)).clamp(0.25, 4.0)
+                        mixed = mixed * (
+                            1.0 + keep_s * (ratio - 1.0)
+                        )
+                    return mixed.to(xc.dtype)
+
+            else:
+                def post_step(xc, sigma_next):
+                    keep = init_lat + noise_f32 * sigma_next
+                    return (
+                        lat_mask * xc.float() + (1.0 - lat_mask) * keep
+                    ).to(xc.dtype)
 
         orig_post = post_step
 
@@ -966,6 +1071,13 @@ class StableDiffusionPipeline:
                     .cpu()
                 )
             images = torch.cat(outs) if len(outs) > 1 else outs[0]
+            if (
+                req.soft_inpainting
+                and req.mask_image is not None
+                and req.init_latents is not None
+                and not (req.enable_hr and req.hr_scale > 1.0)
+            ):
+                images = self._soft_composite(images, req)
         else:
             images = x.cpu()
 
@@ -977,6 +1089,12 @@ class StableDiffusionPipeline:
             extra += f", Clip skip: {req.clip_skip}"
         if req.init_latents is not None:
             extra += f", Denoising strength: {req.denoising_strength}"
+        if req.soft_inpainting and req.mask_image is not None:
+            extra += (
+                f", Soft inpainting: True"
+                f", Schedule bias: {req.si_schedule_bias}"
+                f", Preservation strength: {req.si_preservation_strength}"
+            )
         if req.subseed_strength > 0 and req.subseeds:
             extra += f", Variation seed strength: {req.subseed_strength}"
         if req.enable_hr and req.hr_scale > 1.0:

@@ -82,6 +82,51 @@ class TestImg2Img:
         )
         assert r.status_code == 422
 
+    def test_soft_inpainting_payload(self, client):
+        init = torch.full((64, 64, 3), 170, dtype=torch.uint8)
+        mask = torch.zeros(64, 64, 3, dtype=torch.uint8)
+        mask[:, 32:] = 255
+        body = {
+            "prompt": "re", "steps": 2, "width": 64, "height": 64,
+            "seed": 5, "denoising_strength": 1.0,
+            "init_images": [base64.b64encode(encode_png(init)).decode()],
+            "mask": base64.b64encode(encode_png(mask)).decode(),
+            "alwayson_scripts": {
+                "soft inpainting": {
+                    "args": [{
+                        "Soft inpainting": True,
+                        "Schedule bias": 1.5,
+                        "Preservation strength": 0.5,
+                    }]
+                }
+            },
+        }
+        r = client.post("/sdapi/v1/img2img", json=body)
+        assert r.status_code == 200, r.text
+        info = json.loads(r.json()["info"])
+        assert "Soft inpainting: True" in info["infotexts"][0]
+        assert "Schedule bias: 1.5" in info["infotexts"][0]
+        # same request without the script: plain hard-mask inpaint
+        body2 = dict(body)
+        body2.pop("alwayson_scripts")
+        r2 = client.post("/sdapi/v1/img2img", json=body2)
+        assert r2.status_code == 200
+        assert r.json()["images"][-1] != r2.json()["images"][-1]
+
+    def test_soft_inpainting_positional_and_disabled(self):
+        from sdwd_amd.api.server import _parse_soft_inpainting
+
+        got = _parse_soft_inpainting(
+            {"soft inpainting": {"args": [True, 2.0, 0.25]}}
+        )
+        assert got["soft_inpainting"] is True
+        assert got["si_schedule_bias"] == 2.0
+        assert got["si_preservation_strength"] == 0.25
+        assert _parse_soft_inpainting(
+            {"soft inpainting": {"args": [{"Soft inpainting": False}]}}
+        ) == {}
+        assert _parse_soft_inpainting({}) == {}
+
 
 class TestControl:
     def test_models_and_samplers(self, client):

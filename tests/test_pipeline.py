@@ -497,6 +497,64 @@ class TestInpainting:
         assert torch.equal(a, b)
 
 
+class TestSoftInpainting:
+    def _mask(self):
+        mask = torch.zeros(64, 64, dtype=torch.uint8)
+        mask[:, 32:] = 255
+        return mask
+
+    def test_soft_differs_from_hard_and_preserves_outside(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        init = torch.full((1, 64, 64, 3), 180, dtype=torch.uint8)
+        lat = pipe.encode_image(init, seeds=[9])
+        common = dict(
+            prompt="repaint", steps=4, width=64, height=64, seeds=[9],
+            init_latents=lat, denoising_strength=1.0,
+            mask_image=self._mask(),
+        )
+        hard = pipe.generate(PipelineRequest(**common)).images.float()
+        soft = pipe.generate(
+            PipelineRequest(**common, soft_inpainting=True)
+        ).images.float()
+        assert not torch.equal(hard, soft)
+        # far outside the mask both modes keep the original content
+        plain = PipelineRequest(
+            prompt="x", steps=1, width=64, height=64, seeds=[9],
+            init_latents=lat, denoising_strength=0.01,
+        )
+        base = pipe.generate(plain).images.float()
+        out_diff = (soft[:, :, :24] - base[:, :, :24]).abs().mean()
+        in_diff = (soft[:, :, 40:] - base[:, :, 40:]).abs().mean()
+        assert in_diff > out_diff * 1.5, (out_diff, in_diff)
+
+    def test_soft_deterministic_and_infotext(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        init = torch.randint(0, 255, (1, 64, 64, 3), dtype=torch.uint8)
+        lat = pipe.encode_image(init, seeds=[5])
+        req = PipelineRequest(
+            prompt="fill", steps=3, width=64, height=64, seeds=[5],
+            init_latents=lat, mask_image=self._mask(),
+            soft_inpainting=True, si_mask_influence=0.3,
+        )
+        a = pipe.generate(req)
+        b = pipe.generate(req)
+        assert torch.equal(a.images, b.images)
+        assert "Soft inpainting: True" in a.infotexts[0]
+
+    def test_contrast_curve_properties(self):
+        from sdwd_amd.pipeline.pipeline import _contrast
+
+        w = torch.linspace(0, 1, 21)
+        assert torch.allclose(_contrast(w, 1.0), w)
+        c = _contrast(w, 4.0)
+        # fixed points and monotone sharpening toward 0/1
+        assert torch.allclose(c[[0, 10, 20]], torch.tensor([0.0, 0.5, 1.0]))
+        assert (c[1:10] < w[1:10]).all() and (c[11:20] > w[11:20]).all()
+        assert (c[1:] >= c[:-1]).all()
+
+
 class TestPromptEditing:
     def test_schedule_parsing(self):
         from sdwd_amd.pipeline.prompt_schedule import (
