@@ -547,6 +547,27 @@ class TestInpaintingGPU:
         assert res.images.shape == (1, 64, 64, 3)
         assert torch.isfinite(res.images.float()).all()
 
+    def test_soft_inpainting_runs_on_gpu(self, dev):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny", device=dev)
+        init = torch.full((1, 64, 64, 3), 180, dtype=torch.uint8)
+        lat = pipe.encode_image(init, seeds=[6])
+        mask = torch.zeros(64, 64, dtype=torch.uint8)
+        mask[:, 32:] = 255
+        common = dict(prompt="soft", steps=3, width=64, height=64,
+                      seeds=[6], init_latents=lat, mask_image=mask,
+                      denoising_strength=1.0)
+        hard = pipe.generate(PipelineRequest(**common)).images
+        soft = pipe.generate(
+            PipelineRequest(**common, soft_inpainting=True)
+        ).images
+        soft2 = pipe.generate(
+            PipelineRequest(**common, soft_inpainting=True)
+        ).images
+        assert not torch.equal(hard, soft)
+        assert torch.equal(soft, soft2)
+
 
 class TestDeterminismGPU:
     def test_same_request_same_images(self, dev):
