@@ -79,13 +79,15 @@ __launch_bounds__(512, 2) __global__ void conv3x3_v4_kernel(
   const int kc_per_plane = Cin / V4_BK;
   const int NT = 9 * kc_per_plane;
 
-  // stage 2 pieces (one phase's share) of tile t into buffer b. Piece
-  // order is chosen so the LAST-issued pieces (left in flight past the
-  // per-tile counted waits) are the LATEST-read ones: B pieces (read from
-  // the next tile's phase 0 by all waves) go first, A pieces for the
-  // mh=0 phases next, the mh=1 pieces (first read at phase 2) last.
-  //   ph0 -> B0,B1   ph1 -> B2,B3   ph2 -> A0,A2   ph3 -> A1,A3
-  auto stage2 = [&](int t, int b, int ph) {
+  // stage ALL 8 pieces of tile t into buffer b at once. Issued at phase 0
+  // of the previous tile (that buffer's readers finished a barrier-pair
+  // ago), drained by ONE `s_waitcnt vmcnt(0)` at the END of phase 3 —
+  // every piece gets ~3 full phases (>1000 cycles) of slack. The earlier
+  // schedule spread the issues across the 4 phases with counted waits at
+  // ph1/ph3, which left the ph2/ph3-issued pieces only ~1 phase of slack:
+  // the waits stalled on just-issued loads (PMC: 34-42% wave-wait).
+  // B pieces first (read by every wave at the next tile's phase 0).
+  auto stage8 = [&](int t, int b) {
     const int plane = t / kc_per_plane;
     const int kc = t % kc_per_plane;
     const int dy = plane / 3 - 1, dx = plane % 3 - 1;
@@ -93,10 +95,10 @@ __launch_bounds__(512, 2) __global__ void conv3x3_v4_kernel(
     __bf16 *abuf = smem + b * (2 * V4_ATILE);
     __bf16 *bbuf = abuf + V4_ATILE;
     const int sc = schunk;
-    constexpr int PIECES[4][2] = {{4, 5}, {6, 7}, {0, 2}, {1, 3}};
+    constexpr int PIECES[8] = {4, 5, 6, 7, 0, 1, 2, 3};
 #pragma unroll
-    for (int k = 0; k < 2; ++k) {
-      const int i = PIECES[ph][k];        // piece index 0..7
+    for (int k = 0; k < 8; ++k) {
+      const int i = PIECES[k];            // piece index 0..7
       if (i < 4) {
         const int row = prow + 64 * i;
         const int swz = cswz(row, sc);
@@ -133,10 +135,9 @@ __launch_bounds__(512, 2) __global__ void conv3x3_v4_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){};
 
-  // prologue: stage tile 0 fully (8 pieces), then begin the loop with
-  // tile 1 staged phase-by-phase while tile 0 is computed
-#pragma unroll
-  for (int ph = 0; ph < 4; ++ph) stage2(0, 0, ph);
+  // prologue: stage tile 0 fully, then begin the loop with tile 1
+  // staged at tile 0's phase 0 while tile 0 is computed
+  stage8(0, 0);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
@@ -170,26 +171,9 @@ __launch_bounds__(512, 2) __global__ void conv3x3_v4_kernel(
                                        ck * 16);
         }
       }
-      // stage one phase's share of tile t+1 into the other buffer
-      if (t + 1 < NT) stage2(t + 1, (t + 1) & 1, ph);
-      if (ph == 1) {
-        // the previous tile's ph3-staged A1/A3 (left in flight past the
-        // ph3 wait) are first read at THIS tile's phase 2: wait them in
-        // (vmcnt(4) leaves this tile's ph0/ph1 issues outstanding; on the
-        // last tile nothing new was issued, so drain)
-        if (t + 1 < NT)
-          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-        else
-          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      } else if (ph == 3) {
-        // once per K-tile: everything except the 2 pieces just issued
-        // (A1,A3 - not read until the next tile's phase 2) must land
-        // before the next tile's first ds_read
-        if (t + 1 < NT)
-          asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
-        else
-          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      }
+      // stage ALL of tile t+1 into the other buffer at phase 0 (its
+      // readers in tile t-1 retired before this tile's first barrier)
+      if (ph == 0 && t + 1 < NT) stage8(t + 1, (t + 1) & 1);
       __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -203,6 +187,11 @@ __launch_bounds__(512, 2) __global__ void conv3x3_v4_kernel(
                     af[i][s], bf[j][s], acc[mh * 4 + i][nh * 2 + j], 0, 0,
                     0);
       __builtin_amdgcn_s_setprio(0);
+      // ONE drain per K-tile, ~3 phases after the issues; the final
+      // barrier below then publishes the staged LDS to every wave
+      // before the next tile's first ds_read
+      if (ph == 3 && t + 1 < NT)
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
     }
   }
