@@ -370,22 +370,30 @@ torch::Tensor conv3x3_nhwc_impl(torch::Tensor x, torch::Tensor w_prep,
   }
   // v4: the 256x256 8-phase deep pipeline serves full 256-column tiles;
   // the v2 128-tile kernel covers the Cout remainder (e.g. 320 = 256+64).
-  // SDWD_CONV=v2 forces the round-1 kernel everywhere (A/B).
-  static const bool v2only = [] {
+  // SDWD_CONV=v2 forces the round-1 kernel everywhere; =v4 forces the
+  // deep pipeline on every legal shape (A/B tooling).
+  static const int conv_force = [] {
     const char *e = getenv("SDWD_CONV");
-    return e && strcmp(e, "v2") == 0;
+    if (!e) return 0;
+    if (strcmp(e, "v2") == 0) return 2;
+    if (strcmp(e, "v4") == 0) return 4;
+    return 0;
   }();
+  const bool v2only = conv_force == 2;
   const long mt4 = (M + V4_BM - 1) / V4_BM;
-  // measured winners (profiles/r02_conv_v4.md): the 256x256 pipeline wins
-  // on the SD-UNet level-0/1 shapes (Cin<=960, Cout<=640, big M); the
-  // 1280-channel and VAE shapes stay on v2 (2 blocks/CU beats the deep
-  // pipeline there).
+  // measured winners (profiles/r02_conv_family.md): the 256x256 pipeline
+  // wins on the SD-UNet level-0/1 shapes (Cin<=960, Cout<=640, big M) and
+  // — since the phase-0 stage-8 schedule — the VAE 512-ch shapes too
+  // (+7-12%); the 1280-channel shapes stay on v2 (2 blocks/CU beats the
+  // deep pipeline there: 737-765 vs 814-822 TF/s forced).
   // ... and only when the 512-thread 1-block/CU grid actually fills the
   // 256 CUs (small shards at N=8 under-fill 256-row tiles: the 128-row
   // v2 kernel at 2 blocks/CU wins there)
   const bool v4_shape =
-      Cin <= 960 && Cout <= 640 && Cin != 512 && Cout >= 256 &&
-      ((M + V4_BM - 1) / V4_BM) * (Cout / 256) >= 256;
+      conv_force == 4
+          ? (Cin % 64 == 0 && Cout >= 256)
+          : (Cin <= 960 && Cout <= 640 && Cout >= 256 &&
+             ((M + V4_BM - 1) / V4_BM) * (Cout / 256) >= 256);
   const long nfull = (v2only || !v4_shape) ? 0 : Cout / 256;
   const int rem = (int)(Cout - nfull * 256);
   if (nfull > 0) {
