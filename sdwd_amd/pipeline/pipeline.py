@@ -1167,23 +1167,30 @@ class StableDiffusionPipeline:
             images.permute(0, 3, 1, 2).float() / 127.5 - 1.0
         ).to(self.device, self.dtype)
         if seeds:
-            # one batched encoder forward; only the sampling noise is drawn
-            # per-image (deterministic per seed, device-independent)
-            moments = self.model.vae.encoder(x)
-            mean, logvar = moments.chunk(2, dim=1)
-            std = torch.exp(0.5 * logvar.float().clamp(-30, 20))
-            noise = torch.stack(
-                [
-                    torch.randn(
-                        mean.shape[1:],
-                        generator=torch.Generator("cpu").manual_seed(
-                            (int(sd) ^ 0xE4C0DE) & 0xFFFFFFFF
-                        ),
-                        dtype=torch.float32,
-                    )
-                    for sd in seeds
-                ]
-            ).to(self.device)
-            lat = (mean.float() + std * noise) * self.model.vae.cfg.scale_factor
-            return lat.to(self.dtype)
+            # PER-IMAGE encoder forwards: batched conv numerics depend on
+            # the batch size (CPU blocking; library GEMM selection), so a
+            # shard of 2 would encode differently from the same images
+            # inside a batch of 4 and break the C22 exactness contract
+            # (N-GPU gallery == 1-GPU batch). Encoding
+            # image-by-image makes the latents shard-placement-invariant
+            # by construction; the encoder runs once per job and each
+            # image still fills the GPU (M = H*W rows at the top level).
+            # Sampling noise is seeded per image (device-independent).
+            lats = []
+            for i, sd in enumerate(seeds):
+                moments = self.model.vae.encoder(x[i : i + 1])
+                mean, logvar = moments.chunk(2, dim=1)
+                std = torch.exp(0.5 * logvar.float().clamp(-30, 20))
+                noise = torch.randn(
+                    mean.shape[1:],
+                    generator=torch.Generator("cpu").manual_seed(
+                        (int(sd) ^ 0xE4C0DE) & 0xFFFFFFFF
+                    ),
+                    dtype=torch.float32,
+                )[None].to(self.device)
+                lats.append(
+                    (mean.float() + std * noise)
+                    * self.model.vae.cfg.scale_factor
+                )
+            return torch.cat(lats).to(self.dtype)
         return self.model.vae.encode(x)

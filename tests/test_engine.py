@@ -240,6 +240,38 @@ def _single_rank_reference(batch=4, seed=900, prompt="dist"):
     )
 
 
+def _soft_inpaint_request():
+    from sdwd_amd.parallel import GenerationRequest
+
+    init = torch.full((1, 64, 64, 3), 180, dtype=torch.uint8).expand(
+        4, -1, -1, -1
+    )
+    mask = torch.zeros(64, 64, dtype=torch.uint8)
+    mask[:, 32:] = 255
+    return GenerationRequest(
+        prompt="soft-dist", batch_size=4, width=64, height=64, steps=2,
+        seed=770, init_images=init.contiguous(), mask_image=mask,
+        denoising_strength=1.0, soft_inpainting=True, si_mask_influence=0.2,
+    )
+
+
+def _soft_worker(rank, world_size, port, tmpdir):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from sdwd_amd.parallel import DistributedEngine
+
+    eng = DistributedEngine(model="tiny", backend="gloo")
+    res = eng.generate(_soft_inpaint_request())
+    if rank == 0:
+        torch.save(res.images, os.path.join(tmpdir, "soft.pt"))
+    from sdwd_amd.parallel import destroy_group
+
+    destroy_group()
+
+
 @pytest.mark.timeout(300)
 class TestDistributedEngine:
     def test_two_rank_gloo(self, tmp_path):
@@ -259,6 +291,29 @@ class TestDistributedEngine:
         ref = _single_rank_reference()
         diff = (ref.images.float() - gallery.float()).abs()
         assert diff.max() <= 1.0
+
+    def test_two_rank_soft_inpainting_matches_single(self, tmp_path):
+        """Soft inpainting shards image-for-image like any other request:
+        the per-step soft blend and the pixel composite are pure functions
+        of each image's own seed/latents, never of shard placement."""
+        import torch.multiprocessing as mp
+        import socket
+
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        mp.start_processes(
+            _soft_worker, args=(2, port, str(tmp_path)), nprocs=2,
+            start_method="spawn", join=True,
+        )
+        gallery = torch.load(tmp_path / "soft.pt")
+        assert gallery.shape == (4, 64, 64, 3)
+        from sdwd_amd.parallel import LocalEngine
+
+        one = LocalEngine(model="tiny", devices=["cpu"]).generate(
+            _soft_inpaint_request()
+        )
+        assert torch.equal(gallery, one.images)
 
     def test_three_rank_complementary_production(self, tmp_path):
         """A deferred slow rank produces bonus images through the
