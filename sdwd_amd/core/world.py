@@ -18,7 +18,6 @@ from __future__ import annotations
 
 import concurrent.futures
 import threading
-import time
 from dataclasses import dataclass
 from typing import Dict, List, Optional
 
