@@ -163,7 +163,38 @@ def _resize_init(img: torch.Tensor, w: int, h: int, mode: int) -> torch.Tensor:
 # windows) and Dynamic Prompts (wildcard/variant expansion). Everything
 # else is logged and skipped, like the reference's compat filter when a
 # remote lacked the script - docs/usage.md documents the boundary.
-_NATIVE_ALWAYSON = ("controlnet", "dynamic prompts", "soft inpainting")
+_NATIVE_ALWAYSON = (
+    "controlnet", "dynamic prompts", "soft inpainting", "regional prompter",
+)
+
+
+def _parse_regional_prompter(alwayson: Dict[str, Any]) -> Dict[str, Any]:
+    """Regional Prompter matrix-mode payload, dict form:
+    {"regional prompter": {"args": [{"active": true, "mode": "Columns",
+    "ratios": "1,1", "base_ratio": 0.2, "use_base": true}]}}. Only the
+    matrix (columns/rows) mode executes natively; mask/prompt modes are
+    logged and skipped like any unsupported script."""
+    for name, body in (alwayson or {}).items():
+        if name.lower().replace("-", " ") != "regional prompter":
+            continue
+        args = (body or {}).get("args", [])
+        d = args[0] if args and isinstance(args[0], dict) else {}
+        if not d or not d.get("active", True):
+            return {}
+        mode = str(d.get("mode", d.get("Matrix mode", "columns"))).lower()
+        if mode not in ("columns", "rows"):
+            log.warning("regional prompter: mode %r unsupported", mode)
+            return {}
+        out: Dict[str, Any] = {
+            "regional_mode": mode,
+            "regional_ratios": str(d.get("ratios", "1,1")),
+        }
+        base = d.get("base_ratio", 0.2)
+        if not d.get("use_base", True):
+            base = 0.0
+        out["regional_base_ratio"] = float(base)
+        return out
+    return {}
 
 
 def _dynamic_prompts_enabled(alwayson: Dict[str, Any]) -> bool:
@@ -433,6 +464,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             model=model,
             refiner_model=req.refiner_checkpoint,
             refiner_switch_at=req.refiner_switch_at,
+            **_parse_regional_prompter(req.alwayson_scripts),
         )
         if _dynamic_prompts_enabled(req.alwayson_scripts):
             _expand_dynamic(gen)
@@ -498,6 +530,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             inpainting_mask_invert=req.inpainting_mask_invert,
             color_correction=req.color_correction,
             **_parse_soft_inpainting(req.alwayson_scripts),
+            **_parse_regional_prompter(req.alwayson_scripts),
         )
         return run_generation(gen, req.send_images, req.save_images)
 

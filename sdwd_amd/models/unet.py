@@ -97,6 +97,80 @@ class ResBlock(nn.Module):
         return self.conv2(self.norm2(h), residual=skip, collect_gn=True)
 
 
+class RegionalContext:
+    """Attention-couple regional prompting (the Regional Prompter
+    extension's matrix mode, executed natively — the reference only
+    FORWARDED its payload, distributed.py:199-234).
+
+    Wraps the normal conditioning rows (`plain`, [N,77,C]) plus R
+    per-region contexts ([R,77,C]) and R spatial masks on the latent
+    grid. Cross-attention evaluates `plain` for every row as usual, then
+    re-blends the first `rows` rows (the cond rows) from per-region
+    attention outputs weighted by the masks pooled to that resolution:
+
+        out = (base + (1-base)*(1-cover)) * attn(base_ctx)
+              + (1-base) * sum_r mask_r * attn(ctx_r)
+
+    so uncovered pixels and `base_ratio` keep the shared base prompt.
+    Only cross-attention interprets the wrapper; everything else treats
+    conditioning opaquely, and hipGraph wrappers fall back to eager."""
+
+    def __init__(self, plain, region_ctx, masks, rows, base_ratio, lat_hw):
+        self.plain = plain            # [N, 77, C]
+        self.region_ctx = region_ctx  # [R, 77, C]
+        self.masks = masks            # [R, lat_h, lat_w] float, sum<=1/pixel
+        self.rows = int(rows)         # leading rows blended regionally
+        self.base_ratio = float(base_ratio)
+        self.lat_hw = lat_hw          # latent grid the masks are drawn on
+        self._cache = {}              # seq_len -> [R, s] pooled masks
+
+    def __getitem__(self, sl):
+        plain = self.plain[sl]
+        out = RegionalContext(
+            plain, self.region_ctx, self.masks,
+            min(self.rows, plain.shape[0]), self.base_ratio, self.lat_hw,
+        )
+        out._cache = self._cache
+        return out
+
+    @property
+    def shape(self):
+        return self.plain.shape
+
+    def to(self, *a, **kw):
+        out = RegionalContext(
+            self.plain.to(*a, **kw), self.region_ctx.to(*a, **kw),
+            self.masks, self.rows, self.base_ratio, self.lat_hw,
+        )
+        out._cache = self._cache
+        return out
+
+    def masks_for(self, s: int, device, dtype) -> torch.Tensor:
+        """[R, s] masks at the attention resolution with h*w == s (the
+        UNet halves the even latent grid per level)."""
+        got = self._cache.get(s)
+        if got is not None and got.device == device:
+            return got
+        lh, lw = self.lat_hw
+        k = 0
+        while k < 6 and (lh >> k) * (lw >> k) != s:
+            k += 1
+        h, w = max(1, lh >> k), max(1, lw >> k)
+        if h * w != s:  # non-halving grid: nearest square-ish fallback
+            import math
+
+            h = max(1, int(round(math.sqrt(s * lh / max(lw, 1)))))
+            while s % h:
+                h -= 1
+            w = s // h
+        m = torch.nn.functional.adaptive_avg_pool2d(
+            self.masks[None].float(), (h, w)
+        )[0]
+        m = m.reshape(m.shape[0], s).to(device=device, dtype=torch.float32)
+        self._cache[s] = m
+        return m
+
+
 class CrossAttention(nn.Module):
     def __init__(self, query_dim: int, context_dim: int, heads: int):
         super().__init__()
@@ -133,11 +207,42 @@ class CrossAttention(nn.Module):
             q = qkv[..., :d].unflatten(-1, hd)
             k = qkv[..., d:2 * d].unflatten(-1, hd)
             v = qkv[..., 2 * d:].unflatten(-1, hd)
-        else:
+        elif torch.is_tensor(context):
             q = self.to_q(x).unflatten(-1, hd)
             kv = F.linear(context, self._wcat(("to_k", "to_v")))
             k = kv[..., :d].unflatten(-1, hd)
             v = kv[..., d:].unflatten(-1, hd)
+        else:  # RegionalContext: base attention + masked per-region blend
+            rc = context
+            wkv = self._wcat(("to_k", "to_v"))
+            q = self.to_q(x).unflatten(-1, hd)
+            kv = F.linear(rc.plain, wkv)
+            k = kv[..., :d].unflatten(-1, hd)
+            v = kv[..., d:].unflatten(-1, hd)
+            out = ops.attention_bshd(q, k, v).reshape(b, s, d)
+            nr = rc.rows
+            if nr > 0 and rc.region_ctx is not None:
+                m = rc.masks_for(s, x.device, x.dtype)  # [R, s]
+                qr = q[:nr]
+                acc = None
+                for r in range(rc.region_ctx.shape[0]):
+                    kvr = F.linear(
+                        rc.region_ctx[r : r + 1].expand(nr, -1, -1), wkv
+                    )
+                    kr = kvr[..., :d].unflatten(-1, hd)
+                    vr = kvr[..., d:].unflatten(-1, hd)
+                    orr = ops.attention_bshd(qr, kr, vr).reshape(nr, s, d)
+                    term = orr * m[r][None, :, None]
+                    acc = term if acc is None else acc + term
+                bw = rc.base_ratio
+                keep = (bw + (1.0 - bw) * (1.0 - m.sum(0).clamp(0, 1)))[
+                    None, :, None
+                ]
+                blended = (
+                    out[:nr].float() * keep + (1.0 - bw) * acc.float()
+                ).to(out.dtype)
+                out = torch.cat([blended, out[nr:]], dim=0)
+            return self.to_out(out)
         out = ops.attention_bshd(q, k, v)
         return self.to_out(out.reshape(b, s, d))
 

@@ -92,6 +92,10 @@ class GenerationRequest:
     s_min_uncond: float = 0.0
     # sdwui img2img color correction: match output statistics to the init
     color_correction: bool = False
+    # Regional Prompter matrix mode (see PipelineRequest for semantics)
+    regional_mode: str = ""
+    regional_ratios: str = "1,1"
+    regional_base_ratio: float = 0.2
     # soft inpainting (host built-in; see PipelineRequest for semantics)
     soft_inpainting: bool = False
     si_schedule_bias: float = 1.0
@@ -180,6 +184,9 @@ def _job_pipeline_request(
         s_tmax=gen.s_tmax,
         s_noise=gen.s_noise,
         s_min_uncond=gen.s_min_uncond,
+        regional_mode=gen.regional_mode,
+        regional_ratios=gen.regional_ratios,
+        regional_base_ratio=gen.regional_base_ratio,
         soft_inpainting=gen.soft_inpainting,
         si_schedule_bias=gen.si_schedule_bias,
         si_preservation_strength=gen.si_preservation_strength,

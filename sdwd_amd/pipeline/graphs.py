@@ -151,7 +151,9 @@ class GraphedDenoiser:
         ctx: torch.Tensor,
         y: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
-        if not self.enabled or self.failed:
+        if not self.enabled or self.failed or not torch.is_tensor(ctx):
+            # non-tensor ctx (RegionalContext) carries python-side control
+            # flow the capture cannot freeze — run eager
             return self.fn(x, ts, ctx, y)
         key = (
             tuple(x.shape),

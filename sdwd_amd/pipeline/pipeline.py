@@ -105,6 +105,16 @@ class PipelineRequest:
     control_scale: float = 1.0
     control_units: Optional[List[dict]] = None
     clip_skip: int = 1  # 1 = final layer; 2 = penultimate (sdwui setting)
+    # Regional Prompter matrix mode, executed natively (the reference
+    # forwarded the extension's payload, C18): the prompt is split on
+    # "BREAK" into [base?, region1, region2, ...]; regions tile the
+    # canvas as columns or rows with the given ratios and blend through
+    # masked cross-attention (attention couple). base_ratio > 0 keeps
+    # that weight of the shared base prompt everywhere (and expects the
+    # first BREAK chunk to BE the base prompt).
+    regional_mode: str = ""        # "" off | "columns" | "rows"
+    regional_ratios: str = "1,1"
+    regional_base_ratio: float = 0.2
 
     @property
     def batch_size(self) -> int:
@@ -134,6 +144,70 @@ def _slerp(a: torch.Tensor, b: torch.Tensor, t: float) -> torch.Tensor:
             math.sin(t * omega) / so
         ) * bf
     return out.reshape(a.shape).to(a.dtype)
+
+
+def _parse_regional(req: "PipelineRequest"):
+    """Split the prompt on BREAK into (base, regions) per the ratios.
+
+    With base_ratio > 0 the first chunk is the shared base prompt and the
+    remaining chunks are the regions; with base_ratio == 0 every chunk is
+    a region (base attention falls back to an empty prompt for uncovered
+    pixels only). Returns None (regional off, warning logged) when the
+    chunk count does not match the ratio count."""
+    import re as _re
+
+    chunks = [c.strip() for c in _re.split(r"\bBREAK\b", req.prompt)]
+    try:
+        ratios = [
+            float(v) for v in str(req.regional_ratios).split(",") if v.strip()
+        ]
+    except ValueError:
+        ratios = []
+    if not ratios or any(r <= 0 for r in ratios):
+        log.warning("regional: bad ratios %r — disabled", req.regional_ratios)
+        return None
+    base_w = min(max(float(req.regional_base_ratio), 0.0), 1.0)
+    if base_w > 0 and len(chunks) == len(ratios) + 1:
+        return chunks[0], chunks[1:], ratios, base_w
+    if len(chunks) == len(ratios):
+        return ("", chunks, ratios, base_w) if base_w == 0 else (
+            chunks[0], chunks, ratios, base_w
+        )
+    log.warning(
+        "regional: %d BREAK chunks vs %d ratios — disabled",
+        len(chunks), len(ratios),
+    )
+    return None
+
+
+def _region_masks(
+    mode: str, ratios: List[float], lat_h: int, lat_w: int
+) -> torch.Tensor:
+    """[R, lat_h, lat_w] 0/1 masks tiling the canvas as columns or rows
+    proportionally to the ratios (matrix-mode Regional Prompter)."""
+    total = sum(ratios)
+    extent = lat_w if mode == "columns" else lat_h
+    nr = len(ratios)
+    bounds = [0]
+    acc = 0.0
+    for r in ratios:
+        acc += r
+        bounds.append(int(round(extent * acc / total)))
+    bounds[-1] = extent
+    # every region keeps at least one line of the grid despite rounding
+    for i in range(1, nr + 1):
+        bounds[i] = max(bounds[i], bounds[i - 1] + 1)
+    bounds[-1] = extent
+    for i in range(nr, 0, -1):
+        bounds[i - 1] = min(bounds[i - 1], bounds[i] - 1)
+    masks = torch.zeros(nr, lat_h, lat_w)
+    for i in range(nr):
+        lo, hi = max(bounds[i], 0), bounds[i + 1]
+        if mode == "columns":
+            masks[i, :, lo:hi] = 1.0
+        else:
+            masks[i, lo:hi, :] = 1.0
+    return masks
 
 
 def _contrast(w: torch.Tensor, g: float) -> torch.Tensor:
@@ -475,6 +549,40 @@ class StableDiffusionPipeline:
                 req, b,
             )
 
+        # Regional Prompter matrix mode: rebuild the cond rows from the
+        # base chunk and wrap the conditioning so cross-attention blends
+        # per-region contexts under columns/rows masks. Composes with CFG
+        # and samplers; AND composition / prompt editing / per-image
+        # prompts keep their own conditioning machinery and win.
+        regional = (
+            req.regional_mode in ("columns", "rows")
+            and not per_image
+            and and_ws == [1.0]
+            and len(p_segs) == 1
+            and len(n_segs) == 1
+        )
+        if regional:
+            parsed = _parse_regional(req)
+            if parsed is None:
+                regional = False
+            else:
+                base_text, region_texts, ratios, base_w = parsed
+                ctx, y, and_ws = self._build_ctx(
+                    base_text, req.negative_prompt, req, b
+                )
+                region_rows, _, _ = self.encode_prompts(
+                    region_texts, [req.negative_prompt], req.clip_skip
+                )
+                masks = _region_masks(
+                    req.regional_mode, ratios, lat_h, lat_w
+                )
+                from ..models.unet import RegionalContext
+
+                ctx = RegionalContext(
+                    ctx, region_rows.to(ctx.dtype), masks.to(self.device),
+                    rows=b, base_ratio=base_w, lat_hw=(lat_h, lat_w),
+                )
+
         sched = schedule_for(req.sampler_name, req.steps, req.scheduler)
         sampler = build_sampler(req.sampler_name, sched)
         _apply_sampler_params(sampler, req)
@@ -635,6 +743,8 @@ class StableDiffusionPipeline:
 
         def _control_residuals(xk, ts, c_ctx, t):
             """Sum the active units' residuals at timestep t."""
+            if not torch.is_tensor(c_ctx):  # RegionalContext: CN is a
+                c_ctx = c_ctx.plain         # base-UNet copy, feed base rows
             total = None
             for mod, hint_u, scale, t_begin, t_end in cn_units:
                 if not (t <= t_begin + 1e-6 and t > t_end):
@@ -807,6 +917,7 @@ class StableDiffusionPipeline:
             and not cn_units
             and s_min_uncond == 0
             and pred_type != "v"
+            and torch.is_tensor(ctx)  # RegionalContext runs eager
             and (
                 not is_inpaint_model
                 or (req.mask_image is None and req.init_latents is None)
@@ -1094,6 +1205,12 @@ class StableDiffusionPipeline:
                 f", Soft inpainting: True"
                 f", Schedule bias: {req.si_schedule_bias}"
                 f", Preservation strength: {req.si_preservation_strength}"
+            )
+        if regional:
+            extra += (
+                f", RP Active: True, RP Matrix submode: {req.regional_mode}"
+                f", RP Ratios: \"{req.regional_ratios}\""
+                f", RP Base Ratios: {req.regional_base_ratio}"
             )
         if req.subseed_strength > 0 and req.subseeds:
             extra += f", Variation seed strength: {req.subseed_strength}"
