@@ -569,6 +569,27 @@ class TestInpaintingGPU:
         assert torch.equal(soft, soft2)
 
 
+class TestRegionalGPU:
+    def test_regional_runs_and_differs(self, dev):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny", device=dev)
+        base = dict(steps=3, width=64, height=64, seeds=[21])
+        plain = pipe.generate(
+            PipelineRequest(prompt="sky", **base)
+        ).images
+        req = PipelineRequest(
+            prompt="sky BREAK red tree BREAK blue lake",
+            regional_mode="columns", regional_ratios="1,1",
+            regional_base_ratio=0.2, **base,
+        )
+        a = pipe.generate(req)
+        b = pipe.generate(req)
+        assert torch.equal(a.images, b.images)
+        assert not torch.equal(plain, a.images)
+        assert "RP Active: True" in a.infotexts[0]
+
+
 class TestDeterminismGPU:
     def test_same_request_same_images(self, dev):
         from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
