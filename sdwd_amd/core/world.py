@@ -248,6 +248,30 @@ class World:
                 overflow -= 1
             i += 1
             spins += 1
+        # every realtime rank capped out: spill onto deferred ranks (their
+        # jobs were marked complementary) and PROMOTE them — these images
+        # are part of the requested batch and must own real gallery slots,
+        # arriving late beats not arriving (batch conservation is what the
+        # seed plan and the gather math are built on)
+        if overflow > 0:
+            for w in sorted(workers, key=lambda w: -w.eta.avg_ipm):
+                job = plan[w.label]
+                if not job.complementary:
+                    continue
+                while overflow > 0:
+                    if (
+                        w.pixel_cap
+                        and (job.batch_size + 1)
+                        * request.width * request.height > w.pixel_cap
+                    ):
+                        break
+                    job.batch_size += 1
+                    overflow -= 1
+                if job.batch_size > 0:
+                    job.complementary = False
+                    job.step_override = None
+                if overflow == 0:
+                    break
         if overflow > 0:
             log.warning("pixel caps too tight: %d images dropped", overflow)
 
