@@ -15,7 +15,7 @@ except ImportError:  # pragma: no cover
 from sdwd_amd.core.eta import EtaPredictor
 from sdwd_amd.core.seeds import shard_seeds
 
-common = settings(max_examples=60, deadline=None)
+common = settings(max_examples=60, deadline=None, derandomize=True)
 ipm = st.floats(min_value=0.01, max_value=500.0,
                 allow_nan=False, allow_infinity=False)
 

@@ -19,7 +19,7 @@ import torch
 from sdwd_amd.pipeline.samplers import sampler_names
 from sdwd_amd.pipeline.schedule import TRAIN_STEPS, schedule_for, scheduler_names
 
-common = settings(max_examples=40, deadline=None)
+common = settings(max_examples=40, deadline=None, derandomize=True)
 
 
 class TestScheduleInvariants:

@@ -51,6 +51,7 @@ def plan_signature(jobs):
 
 
 common = settings(
+    derandomize=True,
     max_examples=60,
     deadline=None,
     suppress_health_check=[HealthCheck.function_scoped_fixture],
