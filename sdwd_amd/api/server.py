@@ -673,11 +673,25 @@ def create_app(engine: Optional[LocalEngine] = None,
 
     @app.get("/sdapi/v1/scripts")
     def scripts():
-        return {"txt2img": [], "img2img": []}
+        # the natively-executed alwayson set (C18) — the reference PROBED
+        # each remote's script list through exactly this surface
+        # (worker.py:375-404), so report what this engine runs in-process
+        return {
+            "txt2img": list(_NATIVE_ALWAYSON),
+            "img2img": list(_NATIVE_ALWAYSON),
+        }
 
     @app.get("/sdapi/v1/script-info")
     def script_info():
-        return []
+        return [
+            {"name": n, "is_alwayson": True, "is_img2img": True,
+             "args": []}
+            for n in _NATIVE_ALWAYSON
+        ] + [
+            {"name": n, "is_alwayson": True, "is_img2img": False,
+             "args": []}
+            for n in _NATIVE_ALWAYSON
+        ]
 
     @app.get("/sdapi/v1/cmd-flags")
     def cmd_flags():
