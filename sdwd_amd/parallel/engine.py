@@ -369,6 +369,32 @@ def _paste_inpaint_full_res(result: "GalleryResult", paste_ctx) -> None:
 
 
 class _EngineBase:
+    # Master-side postprocess hooks (ref 2.3.0 "compatibility for
+    # extensions which mostly only do postprocessing", e.g. ADetailer:
+    # the reference ran those on the MASTER after the gather, never on
+    # the remotes). Each hook is called on the fully assembled
+    # GalleryResult (rank 0 in torchrun mode) and may edit it in place;
+    # a hook that raises is logged and skipped so a broken postprocessor
+    # cannot lose the batch.
+    @property
+    def postprocess_hooks(self) -> List:
+        hooks = self.__dict__.get("_pp_hooks")
+        if hooks is None:
+            hooks = self.__dict__["_pp_hooks"] = []
+        return hooks
+
+    def _run_postprocess_hooks(self, result: "GalleryResult") -> None:
+        hooks = self.__dict__.get("_pp_hooks") or []
+        if not hooks:
+            return
+        for hook in list(hooks):
+            try:
+                hook(result)
+            except Exception:
+                log.exception("postprocess hook %r failed; skipped", hook)
+        if result.images is not None and result.images.shape[0] > 1:
+            result.grid = make_grid(result.images)
+
     def _assemble(
         self,
         gen: GenerationRequest,
@@ -628,6 +654,7 @@ class LocalEngine(_EngineBase):
         )
         _paste_inpaint_full_res(result, paste_ctx)
         _apply_color_correction(result, gen)
+        self._run_postprocess_hooks(result)
         self._autosave()
         return result
 
@@ -1084,4 +1111,6 @@ class DistributedEngine(_EngineBase):
         )
         _paste_inpaint_full_res(result, paste_ctx)
         _apply_color_correction(result, gen)
+        if self.rank == 0:
+            self._run_postprocess_hooks(result)
         return result

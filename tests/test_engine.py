@@ -767,6 +767,34 @@ class TestAutosave:
         assert not (tmp_path / "no.json").exists()
 
 
+class TestPostprocessHooks:
+    def test_hooks_run_after_gather_and_bad_hook_skipped(self):
+        """ref 2.3.0: postprocessing-only extensions (ADetailer et al.) ran
+        on the MASTER after the gather — the native analogue is a hook on
+        the assembled gallery; a raising hook must not lose the batch."""
+        eng = make_engine(2)
+        seen = []
+
+        def hook(res):
+            seen.append(res.images.shape[0])
+            res.images = res.images.flip(2)
+
+        def bad(res):
+            raise RuntimeError("boom")
+
+        eng.postprocess_hooks.extend([bad, hook])
+        plain_eng = make_engine(2)
+        req = dict(prompt="pp", batch_size=2, width=64, height=64,
+                   steps=1, seed=5)
+        plain = plain_eng.generate(GenerationRequest(**req))
+        hooked = eng.generate(GenerationRequest(**req))
+        assert seen == [2]
+        assert torch.equal(hooked.images, plain.images.flip(2))
+        # the grid reflects the postprocessed images
+        assert hooked.grid is not None
+        assert not torch.equal(hooked.grid, plain.grid)
+
+
 class TestMaskInvert:
     def test_invert_swaps_preserved_region(self):
         eng = make_engine(1)
