@@ -543,6 +543,23 @@ class TestSoftInpainting:
         assert torch.equal(a.images, b.images)
         assert "Soft inpainting: True" in a.infotexts[0]
 
+    def test_soft_inpainting_with_hires_skips_composite(self, pipe):
+        """With hires the output resolution no longer matches the decoded
+        init, so the pixel composite is skipped (latent soft blending
+        still applies) — the request must succeed at the hires size."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        init = torch.full((1, 64, 64, 3), 180, dtype=torch.uint8)
+        lat = pipe.encode_image(init, seeds=[3])
+        out = pipe.generate(PipelineRequest(
+            prompt="soft-hr", steps=2, width=64, height=64, seeds=[3],
+            init_latents=lat, mask_image=self._mask(),
+            soft_inpainting=True, enable_hr=True, hr_scale=2.0,
+            hr_steps=2, denoising_strength=0.6,
+        ))
+        assert out.images.shape == (1, 128, 128, 3)
+        assert torch.isfinite(out.images.float()).all()
+
     def test_contrast_curve_properties(self):
         from sdwd_amd.pipeline.pipeline import _contrast
 

@@ -159,6 +159,43 @@ class TestRegionalPipeline:
         out = pipe.generate(req)
         assert out.images.shape == (1, 64, 64, 3)
 
+    def test_regional_img2img_with_mask(self, pipe):
+        # regional conditioning composes with the img2img/inpaint path
+        init = torch.full((1, 64, 64, 3), 190, dtype=torch.uint8)
+        lat = pipe.encode_image(init, seeds=[11])
+        mask = torch.zeros(64, 64, dtype=torch.uint8)
+        mask[:, 32:] = 255
+        req = self._req(
+            "base BREAK left BREAK right",
+            init_latents=lat, mask_image=mask, denoising_strength=0.9,
+        )
+        a = pipe.generate(req)
+        b = pipe.generate(req)
+        assert torch.equal(a.images, b.images)
+        assert "RP Active: True" in a.infotexts[0]
+
+    def test_regional_with_controlnet_unit(self, pipe):
+        # the ControlNet copy receives the BASE rows (it is a 4ch
+        # base-UNet clone and knows nothing of region masks)
+        hint = torch.randint(0, 255, (1, 64, 64, 3), dtype=torch.uint8)
+        req = self._req(
+            "base BREAK a BREAK b",
+            control_image=hint, control_model="controlnet-tiny",
+            control_scale=0.7,
+        )
+        out = pipe.generate(req)
+        no_cn = pipe.generate(self._req("base BREAK a BREAK b"))
+        assert not torch.equal(out.images, no_cn.images)
+
+    def test_regional_with_hires(self, pipe):
+        req = self._req(
+            "base BREAK a BREAK b", enable_hr=True, hr_scale=2.0,
+            hr_steps=2, denoising_strength=0.6,
+        )
+        out = pipe.generate(req)
+        assert out.images.shape == (1, 128, 128, 3)
+        assert torch.isfinite(out.images.float()).all()
+
 
 class TestRegionalAPIParse:
     def test_dict_payload(self):
