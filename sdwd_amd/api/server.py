@@ -920,6 +920,26 @@ def create_app(engine: Optional[LocalEngine] = None,
         log.warning("generation lock force-released")
         return {}
 
+    @app.post("/sdwd/restart-workers")
+    def restart_workers():
+        """The Utils tab's 'restart all remotes' (ref 2.1.0,
+        worker.py:690-717 POSTed /server-restart to every remote): for
+        in-node ranks a restart is a state reset — interrupt flags clear,
+        UNAVAILABLE/INTERRUPTED ranks return to IDLE and rejoin
+        scheduling (their next liveness probe re-verifies them)."""
+        from ..core.state import State as WState
+
+        eng = app.state.engine
+        eng.world.interrupted.clear()
+        restarted = []
+        for w in eng.world.workers:
+            if w.state in (WState.UNAVAILABLE, WState.INTERRUPTED,
+                           WState.WORKING):
+                w.set_state(WState.IDLE)
+                restarted.append(w.label)
+        log.info("restart-workers: %s", restarted or "none needed")
+        return {"restarted": restarted}
+
     @app.get("/sdwd/benchmark-payload")
     def get_benchmark_payload():
         """The canonical benchmark payload (ref shared.py:63-77 constants,

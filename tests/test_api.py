@@ -424,6 +424,20 @@ class TestControl:
         )
         assert r.status_code == 200
 
+    def test_restart_workers(self, client):
+        from sdwd_amd.core.state import State
+
+        eng = client.app.state.engine
+        eng.world.get_worker("gpu0").set_state(State.UNAVAILABLE)
+        eng.world.interrupted.set()
+        r = client.post("/sdwd/restart-workers")
+        assert r.status_code == 200
+        assert "gpu0" in r.json()["restarted"]
+        assert eng.world.get_worker("gpu0").state is State.IDLE
+        assert not eng.world.interrupted.is_set()
+        # idempotent: nothing left to restart
+        assert client.post("/sdwd/restart-workers").json()["restarted"] == []
+
     def test_benchmark_payload_round_trip(self, client):
         before = client.get("/sdwd/benchmark-payload").json()
         assert before["steps"] == 20
