@@ -65,7 +65,7 @@ class GEGLU(nn.Module):
 class SDConv2d(nn.Conv2d):
     """Conv2d dispatching to the MI355X-native paths:
 
-    * 3x3 pad-1 stride-1/2 with Cin % 32 == 0, bf16 channels_last on GPU ->
+    * 3x3 pad-1 stride-1/2 with Cin % 64 == 0, bf16 channels_last on GPU ->
       the implicit-GEMM MFMA kernel (ops/hip/conv.hip), with bias and an
       optional residual add fused into the epilogue;
     * 1x1 stride-1 on GPU -> a hipBLASLt GEMM over the NHWC rows;
