@@ -32,6 +32,9 @@ SAMPLER_EVALS_PER_STEP = {
     "DPM++ 2M Karras": 1.0,
     "DPM++ 2M SDE": 1.0,
     "DPM++ 2M SDE Karras": 1.0,
+    "DPM++ 2M SDE Heun": 1.0,
+    "DPM++ 2M SDE Heun Karras": 1.0,
+    "DPM++ 2M SDE Heun Exponential": 1.0,
     "DPM++ 3M SDE": 1.0,
     "DPM++ 3M SDE Karras": 1.0,
     "LCM": 1.0,

@@ -235,13 +235,30 @@ class DPMpp2MSDE(Sampler):
         if self.old_denoised is not None and self.h_last is not None:
             r = self.h_last / h
             d = ops.lincomb(denoised, self.old_denoised, 1.0, -1.0)
-            x = ops.lincomb(x, d, 1.0, 0.5 * phi / r)
+            x = ops.lincomb(x, d, 1.0, self._ms_coeff(phi, h, eta_h) / r)
         if self.eta > 0 and noise_fn is not None:
             amp = sigma_next * math.sqrt(max(0.0, -math.expm1(-2 * eta_h)))
             x = ops.add_noise(x, noise_fn(), 1.0, amp)
         self.old_denoised = denoised
         self.h_last = h
         return x
+
+    @staticmethod
+    def _ms_coeff(phi: float, h: float, eta_h: float) -> float:
+        # midpoint solver: half the decayed step weight
+        return 0.5 * phi
+
+
+class DPMpp2MSDEHeun(DPMpp2MSDE):
+    """DPM++ 2M SDE with the Heun ('improved Euler') solver: the
+    multistep difference term carries the exact second-order weight
+    phi_2 = 1 - phi/(h + eta*h) instead of the midpoint 0.5*phi; both
+    agree to O(h) as h -> 0 (phi ~ h + ...) and share the decay/noise
+    schedule. sdwui's 'DPM++ 2M SDE Heun' sampler family."""
+
+    @staticmethod
+    def _ms_coeff(phi: float, h: float, eta_h: float) -> float:
+        return 1.0 - phi / (h + eta_h)
 
 
 class DPMpp3MSDE(Sampler):
@@ -738,6 +755,9 @@ SAMPLERS: Dict[str, type] = {
     "DPM++ SDE Karras": DPMppSDE,
     "DPM++ 2M SDE": DPMpp2MSDE,
     "DPM++ 2M SDE Karras": DPMpp2MSDE,
+    "DPM++ 2M SDE Heun": DPMpp2MSDEHeun,
+    "DPM++ 2M SDE Heun Karras": DPMpp2MSDEHeun,
+    "DPM++ 2M SDE Heun Exponential": DPMpp2MSDEHeun,
     "DPM++ 3M SDE": DPMpp3MSDE,
     "DPM++ 3M SDE Karras": DPMpp3MSDE,
     "LMS": LMS,

@@ -171,4 +171,6 @@ def schedule_for(
         raise KeyError(f"unknown scheduler '{scheduler}'")
     if "Karras" in sampler_name:
         return karras_schedule(steps)
+    if "Exponential" in sampler_name:
+        return exponential_schedule(steps)
     return discrete_schedule(steps)

@@ -319,6 +319,7 @@ class TestSamplerOracle:
                                       "DPM2", "DPM2 a", "DDPM",
                                       "DPM++ 2S a", "UniPC",
                                       "DPM++ 2M SDE", "DPM++ 3M SDE",
+                                      "DPM++ 2M SDE Heun",
                                       "Restart"])
     def test_converges_to_point_mass(self, name):
         from sdwd_amd.pipeline.samplers import build_sampler
@@ -352,6 +353,7 @@ class TestKarrasOracle:
     @pytest.mark.parametrize("name", ["DPM++ 2M Karras", "DPM++ SDE Karras",
                                       "DPM++ 2S a Karras",
                                       "DPM++ 2M SDE Karras",
+                                      "DPM++ 2M SDE Heun Karras",
                                       "DPM++ 3M SDE Karras"])
     def test_karras_also_converges(self, name):
         """Same point-mass oracle through the Karras sigma schedule (uses
