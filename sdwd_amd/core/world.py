@@ -270,6 +270,9 @@ class World:
                 if job.batch_size > 0:
                     job.complementary = False
                     job.step_override = None
+                    job.predicted_eta = self._predict(
+                        w, job.batch_size, request
+                    )
                 if overflow == 0:
                     break
         if overflow > 0:
