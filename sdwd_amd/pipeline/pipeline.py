@@ -426,16 +426,21 @@ class StableDiffusionPipeline:
         (gate = 0.5 exactly at the threshold) and pulled toward the raw
         mask by `mask influence`. Kills VAE round-trip drift outside the
         repaint region while keeping genuinely repainted pixels."""
-        orig = self.model.vae.decode(
-            req.init_latents.to(self.device, self.dtype)
-        )
-        orig = (
-            ((orig.float() + 1.0) * 127.5)
+        lat = req.init_latents.to(self.device, self.dtype)
+        # per-image decode: like encode_image, keeps the reference pixels
+        # identical under any shard split (batched conv numerics vary
+        # with batch size at the ulp level)
+        orig = torch.cat([
+            (
+                (self.model.vae.decode(lat[i : i + 1]).float() + 1.0)
+                * 127.5
+            )
             .clamp(0, 255)
             .to(torch.uint8)
             .permute(0, 2, 3, 1)
             .cpu()
-        )
+            for i in range(lat.shape[0])
+        ])
         if orig.shape != images.shape:  # defensive: odd sizes
             return images
         new_f = images.float()
