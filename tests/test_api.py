@@ -721,3 +721,70 @@ class TestExtrasBatch:
             dec = decode_png(base64.b64decode(b))
             assert dec.shape == (16, 16, 3)
             assert torch.equal(dec[::2, ::2], src)
+
+
+class TestGoldenWorkflow:
+    """One realistic session end to end through the HTTP surface: the
+    flows a reference user strings together (benchmark -> generate with
+    hires -> iterate with soft inpaint -> inspect parameters -> utils)."""
+
+    def test_session(self, client):
+        import json as _json
+
+        # 1. status + a re-benchmark on an edited (tiny) payload
+        st = client.get("/sdwd/status").json()
+        assert len(st["workers"]) == 2
+        assert client.post("/sdwd/benchmark-payload", json={
+            "width": 64, "height": 64, "steps": 1,
+        }).status_code == 200
+        r0 = client.post("/sdwd/benchmark")
+        assert r0.status_code == 200
+        assert all(v > 0 for v in r0.json()["speeds"].values())
+
+        # 2. txt2img with hires fix + per-request model override fields
+        r = client.post("/sdapi/v1/txt2img", json={
+            "prompt": "a golden cow", "steps": 2, "width": 64,
+            "height": 64, "seed": 41, "batch_size": 2,
+            "enable_hr": True, "hr_scale": 2.0, "hr_second_pass_steps": 2,
+            "denoising_strength": 0.6,
+        })
+        assert r.status_code == 200, r.text
+        body = r.json()
+        info = _json.loads(body["info"])
+        assert info["all_seeds"] == [41, 42]
+        gen_png = body["images"][-1]  # last = an image (first may be grid)
+
+        # 3. png-info round-trips the parameters line
+        pi = client.post("/sdapi/v1/png-info", json={"image": gen_png})
+        assert pi.status_code == 200
+        assert "a golden cow" in pi.json()["info"]
+        assert "Hires upscale: 2.0" in pi.json()["info"]
+
+        # 4. img2img soft-inpaint iteration on the generated image
+        mask = torch.zeros(64, 64, 3, dtype=torch.uint8)
+        mask[:, :32] = 255
+        # the hires output is 128x128; send it back at 64x64
+        from sdwd_amd.utils.images import decode_png, encode_png
+
+        img = decode_png(base64.b64decode(gen_png))[::2, ::2].contiguous()
+        r2 = client.post("/sdapi/v1/img2img", json={
+            "prompt": "a silver cow", "steps": 2, "width": 64,
+            "height": 64, "seed": 43, "denoising_strength": 0.9,
+            "init_images": [base64.b64encode(encode_png(img)).decode()],
+            "mask": base64.b64encode(encode_png(mask)).decode(),
+            "alwayson_scripts": {
+                "soft inpainting": {"args": [{"Soft inpainting": True}]}
+            },
+        })
+        assert r2.status_code == 200, r2.text
+        info2 = _json.loads(r2.json()["info"])
+        assert "Soft inpainting: True" in info2["infotexts"][0]
+
+        # 5. utils: interrupt (idle no-op) + restart-workers + progress
+        assert client.post("/sdapi/v1/interrupt").status_code == 200
+        assert client.post("/sdwd/restart-workers").status_code == 200
+        pr = client.get("/sdapi/v1/progress").json()
+        assert pr["progress"] in (0, 0.0) or pr["progress"] <= 1.0
+        # the session leaves every rank schedulable
+        st2 = client.get("/sdwd/status").json()
+        assert all(w["state"] in ("IDLE",) for w in st2["workers"])
