@@ -788,3 +788,26 @@ class TestGoldenWorkflow:
         # the session leaves every rank schedulable
         st2 = client.get("/sdwd/status").json()
         assert all(w["state"] in ("IDLE",) for w in st2["workers"])
+
+
+class TestWebUIWiring:
+    def test_every_ui_fetch_target_exists(self, client):
+        """The single-page UI's JS calls must map to real routes — catches
+        a renamed endpoint breaking the page silently (no browser in CI)."""
+        import re
+
+        html = client.get("/").text
+        static = set(re.findall(r"fetch\('([^']+)'", html))
+        dynamic = set(re.findall(r"fetch\(`([^`]+)`", html))
+        assert static, "UI lost its JS wiring?"
+        known_paths = {r.path for r in client.app.routes}
+        for ep in static:
+            path = ep.split("?")[0]
+            assert path in known_paths, f"UI references missing route {path}"
+        for ep in dynamic:
+            # template routes: substitute the parameters the JS uses
+            path = (ep.split("?")[0]
+                    .replace("${label}", "gpu0")
+                    .replace("${act}", "enable"))
+            r = client.post(path, json={})
+            assert r.status_code < 500, (ep, r.status_code)
