@@ -81,11 +81,11 @@ class SDConv2d(nn.Conv2d):
         cache = self._wprep_cache
         if (
             cache is None
-            or cache[0] != w.data_ptr()
+            or cache[0] != (w.data_ptr(), w._version)
             or cache[1] != w.dtype
         ):
             prep = w.permute(0, 2, 3, 1).contiguous()  # [Cout,3,3,Cin]
-            self._wprep_cache = (w.data_ptr(), w.dtype, prep)
+            self._wprep_cache = ((w.data_ptr(), w._version), w.dtype, prep)
             return prep
         return cache[2]
 

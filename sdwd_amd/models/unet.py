@@ -182,10 +182,13 @@ class CrossAttention(nn.Module):
         self.to_out = nn.Linear(query_dim, query_dim)
 
     def _wcat(self, names):
-        """Concatenated projection weight, cached by source data_ptrs:
-        one GEMM instead of 2-3 reads the (huge, M=B*S) activation once."""
+        """Concatenated projection weight, cached by source data_ptrs AND
+        in-place versions (LoRA applies with add_/copy_, which keeps the
+        pointer — keying on the pointer alone served a stale concat and
+        silently dropped LoRA deltas on q/k/v): one GEMM instead of 2-3
+        reads the (huge, M=B*S) activation once."""
         ws = [getattr(self, n).weight for n in names]
-        key = tuple(w.data_ptr() for w in ws)
+        key = tuple((w.data_ptr(), w._version) for w in ws)
         cache = getattr(self, "_wcat_cache", None)
         if cache is None or cache[0] != key:
             cat = torch.cat([w.detach() for w in ws], dim=0)
