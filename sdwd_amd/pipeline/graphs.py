@@ -45,7 +45,7 @@ class _FnEntry:
         self.ctx = ctx
         self.y = y
         self.out = out
-        self.ctx_src: Optional[int] = None  # id() of last-bound ctx
+        self.ctx_src: Optional[int] = None  # bind-token of last copy
 
 
 class GraphedModelFn:
@@ -64,9 +64,14 @@ class GraphedModelFn:
         self.failed = False
         self._ctx: Optional[torch.Tensor] = None
         self._y: Optional[torch.Tensor] = None
+        self._bind_gen = 0  # bumped per bind(): keys the ctx re-copy
 
     def bind(self, ctx: torch.Tensor, y: Optional[torch.Tensor]) -> None:
+        # a monotone token, NOT id(ctx): a freed ctx object's id can be
+        # reused by the next generation's tensor, which would skip the
+        # static-buffer copy and replay the previous conditioning
         self._ctx, self._y = ctx, y
+        self._bind_gen += 1
 
     def _t0d(self, t: float) -> torch.Tensor:
         return torch.tensor(float(t), device=self.device,
@@ -87,11 +92,11 @@ class GraphedModelFn:
                 return self.core(x, self._t0d(t), self._ctx, self._y)
         e.x.copy_(x)
         e.t.fill_(float(t))
-        if e.ctx_src != id(self._ctx):
+        if e.ctx_src != self._bind_gen:
             e.ctx.copy_(self._ctx)
             if self._y is not None:
                 e.y.copy_(self._y)
-            e.ctx_src = id(self._ctx)
+            e.ctx_src = self._bind_gen
         e.graph.replay()
         # CLONE: multi-eval samplers (Heun, DPM2, UniPC...) hold the first
         # eval's eps while the second replay overwrites the static buffer
@@ -122,7 +127,7 @@ class GraphedModelFn:
             finally:
                 gc.enable()
             e = _FnEntry(graph, sx, st, sctx, sy, out)
-            e.ctx_src = id(self._ctx)
+            e.ctx_src = self._bind_gen  # capture cloned the current ctx
             self.cache[key] = e
             log.info("whole-step hipGraph captured for shape %s", key[0])
             return e
