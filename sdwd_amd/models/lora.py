@@ -184,6 +184,12 @@ class LoraManager:
             for name, scale in wanted:
                 self._apply(name, scale, +1.0)
         self.active = list(wanted)
+        # the fused q/k/v projection concats must be refreshed IN PLACE so
+        # captured hipGraphs (which baked the concat buffers' addresses)
+        # replay the new weights (unet.refresh_fused_projections docstring)
+        from .unet import refresh_fused_projections
+
+        refresh_fused_projections(self.unet)
         if wanted:
             log.info("active loras: %s", wanted)
 

@@ -191,9 +191,19 @@ class CrossAttention(nn.Module):
         key = tuple((w.data_ptr(), w._version) for w in ws)
         cache = getattr(self, "_wcat_cache", None)
         if cache is None or cache[0] != key:
-            cat = torch.cat([w.detach() for w in ws], dim=0)
-            self._wcat_cache = cache = (key, cat)
+            if cache is not None and cache[1].shape[0] == sum(
+                w.shape[0] for w in ws
+            ):
+                # refresh IN PLACE: captured hipGraphs baked this buffer's
+                # address, so a reallocation would leave graph replays on
+                # the pre-update weights while eager code moved on
+                cat = cache[1]
+                torch.cat([w.detach() for w in ws], dim=0, out=cat)
+            else:
+                cat = torch.cat([w.detach() for w in ws], dim=0)
+            self._wcat_cache = cache = (key, cat, tuple(names))
         return cache[1]
+
 
     def forward(
         self, x: torch.Tensor, context: Optional[torch.Tensor] = None
@@ -248,6 +258,19 @@ class CrossAttention(nn.Module):
             return self.to_out(out)
         out = ops.attention_bshd(q, k, v)
         return self.to_out(out.reshape(b, s, d))
+
+
+
+def refresh_fused_projections(module: torch.nn.Module) -> None:
+    """Re-run every CrossAttention's fused-projection concat after an
+    in-place weight update (LoRA set_active). The refresh writes into the
+    SAME storage, so hipGraphs that baked the buffer's address replay the
+    updated weights; without this, graphed generations would keep the
+    previous LoRA state on q/k/v."""
+    for mod in module.modules():
+        cache = getattr(mod, "_wcat_cache", None)
+        if cache is not None:
+            mod._wcat(cache[2])
 
 
 class BasicTransformerBlock(nn.Module):
