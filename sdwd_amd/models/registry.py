@@ -115,6 +115,25 @@ def _build_sd15_inpaint(name: str) -> ModelBundle:
     return ModelBundle(name, te, None, unet, vae, context_dim=768)
 
 
+def _build_sdxl_inpaint(name: str) -> ModelBundle:
+    """SDXL-inpainting lineage (diffusers sd-xl-inpainting layout): the
+    XL UNet with a 9-channel input (4 latent + 1 mask + 4 masked-image
+    latent); the same generic inpaint-conditioning path as sd15-inpaint
+    (pipeline in_channels == 2*lat_c+1 detection)."""
+    te = CLIPTextEncoder(d_model=768, layers=12, heads=12)
+    te2 = CLIPTextEncoder(d_model=1280, layers=32, heads=20)
+    cfg = UNetConfig.sdxl()
+    cfg.in_channels = 9
+    unet = UNetModel(cfg)
+    vae = AutoencoderKL(VAEConfig.sd())
+    vae.cfg.scale_factor = 0.13025
+    for seed_off, m in enumerate((te, te2, unet, vae)):
+        _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
+    return ModelBundle(
+        name, te, te2, unet, vae, context_dim=2048, is_sdxl=True
+    )
+
+
 def _build_tiny_inpaint(name: str) -> ModelBundle:
     te = CLIPTextEncoder(d_model=64, layers=2, heads=2, max_len=77)
     cfg = UNetConfig.tiny()
@@ -185,6 +204,22 @@ def _build_tiny_xl(name: str) -> ModelBundle:
     return ModelBundle(name, te, te2, unet, vae, context_dim=64, is_sdxl=True)
 
 
+def _build_tiny_xl_inpaint(name: str) -> ModelBundle:
+    """CPU-test XL + 9-channel inpaint interaction (the sdxl-inpaint
+    lineage at tiny scale)."""
+    b = _build_tiny_xl(name)
+    from .unet import UNetModel as _UM
+
+    cfg = b.unet.cfg
+    cfg.in_channels = 9
+    unet = _UM(cfg)
+    _seeded_init(unet, zlib.crc32(name.encode()) % (2**31) + 7)
+    return ModelBundle(
+        name, b.text_encoder, b.text_encoder_2, unet, b.vae,
+        context_dim=64, is_sdxl=True,
+    )
+
+
 def _build_tiny(name: str) -> ModelBundle:
     te = CLIPTextEncoder(d_model=64, layers=2, heads=2, max_len=77)
     unet = UNetModel(UNetConfig.tiny())
@@ -199,12 +234,14 @@ _BUILDERS: Dict[str, Callable[[str], ModelBundle]] = {
     "sd21": _build_sd21,
     "sd21v": _build_sd21v,
     "sd15-inpaint": _build_sd15_inpaint,
+    "sdxl-inpaint": _build_sdxl_inpaint,
     "sdxl": _build_sdxl,
     "tiny-v": _build_tiny_v,
     "tiny-inpaint": _build_tiny_inpaint,
     "tiny-inpaint-v": _build_tiny_inpaint_v,
     "tiny": _build_tiny,
     "tiny-xl": _build_tiny_xl,
+    "tiny-xl-inpaint": _build_tiny_xl_inpaint,
 }
 
 _cache: Dict[tuple, ModelBundle] = {}

@@ -1663,3 +1663,31 @@ class TestSoftMask:
 
         d_none, d_gray, d_full = dev(0), dev(128), dev(255)
         assert d_none < d_gray < d_full, (d_none, d_gray, d_full)
+
+
+class TestXLInpaint:
+    def test_tiny_xl_inpaint_masked_generation(self):
+        """XL (dual encoders + ADM vector) x 9-channel inpaint
+        conditioning — the sdxl-inpaint lineage at CPU-test scale."""
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny-xl-inpaint", device="cpu")
+        assert pipe.model.unet.cfg.in_channels == 9
+        assert pipe.model.is_sdxl
+        init = torch.full((1, 64, 64, 3), 120, dtype=torch.uint8)
+        lat = pipe.encode_image(init, seeds=[2])
+        mask = torch.zeros(64, 64, dtype=torch.uint8)
+        mask[:, 32:] = 255
+        req = PipelineRequest(
+            prompt="xl inpaint", steps=2, width=64, height=64, seeds=[2],
+            init_latents=lat, mask_image=mask, denoising_strength=0.8,
+        )
+        a = pipe.generate(req)
+        b = pipe.generate(req)
+        assert torch.equal(a.images, b.images)
+        assert torch.isfinite(a.images.float()).all()
+
+    def test_registry_lists_sdxl_inpaint(self):
+        from sdwd_amd.models.registry import available_models
+
+        assert "sdxl-inpaint" in available_models()
