@@ -115,6 +115,18 @@ def _build_sd15_inpaint(name: str) -> ModelBundle:
     return ModelBundle(name, te, None, unet, vae, context_dim=768)
 
 
+def _build_sd21_inpaint(name: str) -> ModelBundle:
+    """SD2-inpainting lineage: the sd21 stack with a 9-channel UNet."""
+    te = CLIPTextEncoder(d_model=1024, layers=24, heads=16)
+    unet = UNetModel(
+        UNetConfig(in_channels=9, context_dim=1024, num_heads=0)
+    )
+    vae = AutoencoderKL(VAEConfig.sd())
+    for seed_off, m in enumerate((te, unet, vae)):
+        _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
+    return ModelBundle(name, te, None, unet, vae, context_dim=1024)
+
+
 def _build_sdxl_inpaint(name: str) -> ModelBundle:
     """SDXL-inpainting lineage (diffusers sd-xl-inpainting layout): the
     XL UNet with a 9-channel input (4 latent + 1 mask + 4 masked-image
@@ -235,6 +247,7 @@ _BUILDERS: Dict[str, Callable[[str], ModelBundle]] = {
     "sd21v": _build_sd21v,
     "sd15-inpaint": _build_sd15_inpaint,
     "sdxl-inpaint": _build_sdxl_inpaint,
+    "sd21-inpaint": _build_sd21_inpaint,
     "sdxl": _build_sdxl,
     "tiny-v": _build_tiny_v,
     "tiny-inpaint": _build_tiny_inpaint,
@@ -381,8 +394,8 @@ def load_checkpoint(path: str, device="cpu", dtype=None) -> ModelBundle:
             is_sd2 = any(k.startswith("cond_stage_model.model.") for k in keys)
             if ch == 320:
                 arch = "sdxl" if is_xl else ("sd21" if is_sd2 else "sd15")
-                if in_ch == 9 and arch == "sd15":
-                    arch = "sd15-inpaint"
+                if in_ch == 9:  # the 9ch inpainting lineages
+                    arch += "-inpaint"
             else:
                 arch = "tiny-xl" if is_xl else (
                     "tiny-inpaint" if in_ch == 9 else "tiny"
