@@ -569,6 +569,23 @@ class TestInpaintingGPU:
         assert torch.equal(soft, soft2)
 
 
+class TestXLInpaintGPU:
+    def test_tiny_xl_inpaint_generates(self, dev):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny-xl-inpaint", device=dev)
+        init = torch.full((1, 64, 64, 3), 120, dtype=torch.uint8)
+        lat = pipe.encode_image(init, seeds=[2])
+        mask = torch.zeros(64, 64, dtype=torch.uint8)
+        mask[:, 32:] = 255
+        res = pipe.generate(PipelineRequest(
+            prompt="xl inpaint", steps=2, width=64, height=64, seeds=[2],
+            init_latents=lat, mask_image=mask, denoising_strength=0.8,
+        ))
+        assert res.images.shape == (1, 64, 64, 3)
+        assert torch.isfinite(res.images.float()).all()
+
+
 class TestRegionalGPU:
     def test_regional_runs_and_differs(self, dev):
         from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
