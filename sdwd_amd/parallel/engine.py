@@ -15,6 +15,7 @@ Two engines share the scheduler (core.World) and pipeline:
 """
 from __future__ import annotations
 
+import os
 import threading
 import time
 from dataclasses import dataclass, field, replace as replace_dc
@@ -461,10 +462,21 @@ class LocalEngine(_EngineBase):
         dtype: Optional[torch.dtype] = None,
     ) -> None:
         if devices is None:
+            # --sdwd-devices / SDWD_DEVICES: ordinal subset (the flag that
+            # replaces the reference's --distributed-remotes host list)
+            sel = os.environ.get("SDWD_DEVICES", "").strip()
             if torch.cuda.is_available():
-                devices = [f"cuda:{i}" for i in range(torch.cuda.device_count())]
+                if sel:
+                    devices = [
+                        f"cuda:{int(o)}" for o in sel.split(",") if o.strip()
+                    ]
+                else:
+                    devices = [
+                        f"cuda:{i}" for i in range(torch.cuda.device_count())
+                    ]
             else:
-                devices = ["cpu"]
+                devices = ["cpu"] * max(1, len(sel.split(","))) if sel \
+                    else ["cpu"]
         self.devices = devices
         self.model_name = model
         self.world = world or World.from_devices(len(devices))
