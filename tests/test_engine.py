@@ -822,3 +822,16 @@ class TestMaskInvert:
         right_i = (inverted.float()[:, :, 40:] - ref[:, :, 40:]).abs().mean()
         assert left_n < right_n, (left_n, right_n)
         assert right_i < left_i, (left_i, right_i)
+
+
+class TestDeviceSelection:
+    def test_sdwd_devices_env_selects_ranks(self, monkeypatch):
+        """--sdwd-devices / SDWD_DEVICES (the --distributed-remotes
+        replacement) bounds the rank set."""
+        from sdwd_amd.parallel import LocalEngine
+
+        monkeypatch.setenv("SDWD_DEVICES", "0,0,0")
+        eng = LocalEngine(model="tiny")
+        assert len(eng.devices) == 3
+        monkeypatch.delenv("SDWD_DEVICES")
+        assert len(LocalEngine(model="tiny").devices) == 1
