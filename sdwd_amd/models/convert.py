@@ -38,6 +38,8 @@ CLIP_PREFIX = "cond_stage_model.transformer.text_model."
 # embedders.0 is SD-style CLIP-L, embedders.1 is open_clip (bigG) layout
 XL_CLIP_L_PREFIX = "conditioner.embedders.0.transformer.text_model."
 XL_CLIP_G_PREFIX = "conditioner.embedders.1.model."
+# the SDXL REFINER has no CLIP-L: its CLIP-G sits at embedder index 0
+XL_REFINER_G_PREFIX = "conditioner.embedders.0.model."
 # SD2.x: single open_clip (ViT-H) text tower
 SD2_CLIP_PREFIX = "cond_stage_model.model."
 
@@ -382,6 +384,8 @@ def load_ldm_state_dict(bundle, state: Dict[str, torch.Tensor]) -> Dict[str, lis
         _load_sd_clip(bundle.text_encoder, xl_l, report)
         report["unexpected"] += [XL_CLIP_L_PREFIX + k for k in xl_l]
     xl_g = take(XL_CLIP_G_PREFIX)
+    if not xl_g and getattr(bundle, "is_refiner", False):
+        xl_g = take(XL_REFINER_G_PREFIX)
     if xl_g and getattr(bundle, "text_encoder_2", None) is not None:
         enc2 = bundle.text_encoder_2
         _load_part(enc2, xl_g, openclip_key_map(enc2), report)
@@ -460,10 +464,15 @@ def to_ldm_state_dict(bundle) -> Dict[str, torch.Tensor]:
     if bundle.is_sdxl and getattr(bundle, "text_encoder_2", None) is not None:
         enc2 = bundle.text_encoder_2
         sd2 = enc2.state_dict()
+        g_prefix = (
+            XL_REFINER_G_PREFIX
+            if getattr(bundle, "is_refiner", False)
+            else XL_CLIP_G_PREFIX
+        )
         for ldm_key, our_key in openclip_key_map(enc2).items():
-            out[XL_CLIP_G_PREFIX + ldm_key] = sd2[our_key]
+            out[g_prefix + ldm_key] = sd2[our_key]
         tp = sd2.get("text_proj")
         if tp is None:
             tp = torch.eye(enc2.d_model)
-        out[XL_CLIP_G_PREFIX + "text_projection"] = tp
+        out[g_prefix + "text_projection"] = tp
     return out

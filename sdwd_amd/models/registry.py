@@ -35,6 +35,9 @@ class ModelBundle:
     vae: AutoencoderKL
     context_dim: int
     is_sdxl: bool = False
+    # SDXL-refiner lineage: CLIP-G only (text_encoder is None), context
+    # 1280, ADM = pooled + [orig_h, orig_w, crop_t, crop_l, aesthetic]
+    is_refiner: bool = False
     # "eps" (noise prediction) or "v" (velocity, SD2.x-768 lineage);
     # the pipeline converts v -> eps algebraically before the sampler
     prediction_type: str = "eps"
@@ -113,6 +116,30 @@ def _build_sd15_inpaint(name: str) -> ModelBundle:
     for seed_off, m in enumerate((te, unet, vae)):
         _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
     return ModelBundle(name, te, None, unet, vae, context_dim=768)
+
+
+def _build_sdxl_refiner(name: str) -> ModelBundle:
+    """SDXL-refiner lineage (sgm refiner config): CLIP-G text tower ONLY
+    (the refiner has no CLIP-L), context 1280, model_channels 384 with
+    transformer depth 4 at the inner levels, ADM vector = pooled(1280) +
+    5 x 256 fourier conds (orig size, crop, aesthetic score)."""
+    te2 = CLIPTextEncoder(d_model=1280, layers=32, heads=20)
+    unet = UNetModel(UNetConfig(
+        model_channels=384,
+        channel_mult=[1, 2, 4, 4],
+        transformer_depth=[0, 4, 4, 0],
+        context_dim=1280,
+        num_heads=0,  # head_dim 64
+        adm_in_channels=1280 + 5 * 256,
+    ))
+    vae = AutoencoderKL(VAEConfig.sd())
+    vae.cfg.scale_factor = 0.13025
+    for seed_off, m in enumerate((te2, unet, vae)):
+        _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
+    return ModelBundle(
+        name, None, te2, unet, vae, context_dim=1280, is_sdxl=True,
+        is_refiner=True,
+    )
 
 
 def _build_sd21_inpaint(name: str) -> ModelBundle:
@@ -232,6 +259,30 @@ def _build_tiny_xl_inpaint(name: str) -> ModelBundle:
     )
 
 
+def _build_tiny_xl_refiner(name: str) -> ModelBundle:
+    """CPU-test refiner lineage (CLIP-G-only + aesthetic ADM at tiny
+    scale; shares tiny's VAE so it can refine tiny/tiny-xl latents)."""
+    te2 = CLIPTextEncoder(d_model=32, layers=2, heads=2)
+    cfg = UNetConfig(
+        model_channels=32,
+        channel_mult=[1, 2],
+        num_res_blocks=1,
+        transformer_depth=[0, 1],
+        context_dim=32,
+        num_heads=0,
+        groups=8,
+        adm_in_channels=32 + 5 * 256,
+    )
+    unet = UNetModel(cfg)
+    vae = AutoencoderKL(VAEConfig.tiny())
+    for seed_off, m in enumerate((te2, unet, vae)):
+        _seeded_init(m, zlib.crc32(name.encode()) % (2**31) + seed_off)
+    return ModelBundle(
+        name, None, te2, unet, vae, context_dim=32, is_sdxl=True,
+        is_refiner=True,
+    )
+
+
 def _build_tiny(name: str) -> ModelBundle:
     te = CLIPTextEncoder(d_model=64, layers=2, heads=2, max_len=77)
     unet = UNetModel(UNetConfig.tiny())
@@ -247,6 +298,7 @@ _BUILDERS: Dict[str, Callable[[str], ModelBundle]] = {
     "sd21v": _build_sd21v,
     "sd15-inpaint": _build_sd15_inpaint,
     "sdxl-inpaint": _build_sdxl_inpaint,
+    "sdxl-refiner": _build_sdxl_refiner,
     "sd21-inpaint": _build_sd21_inpaint,
     "sdxl": _build_sdxl,
     "tiny-v": _build_tiny_v,
@@ -255,6 +307,7 @@ _BUILDERS: Dict[str, Callable[[str], ModelBundle]] = {
     "tiny": _build_tiny,
     "tiny-xl": _build_tiny_xl,
     "tiny-xl-inpaint": _build_tiny_xl_inpaint,
+    "tiny-xl-refiner": _build_tiny_xl_refiner,
 }
 
 _cache: Dict[tuple, ModelBundle] = {}
@@ -392,7 +445,9 @@ def load_checkpoint(path: str, device="cpu", dtype=None) -> ModelBundle:
             ch, in_ch = int(w_in.shape[0]), int(w_in.shape[1])
             is_xl = any(k.startswith("conditioner.") for k in keys)
             is_sd2 = any(k.startswith("cond_stage_model.model.") for k in keys)
-            if ch == 320:
+            if ch == 384 and is_xl:
+                arch = "sdxl-refiner"
+            elif ch == 320:
                 arch = "sdxl" if is_xl else ("sd21" if is_sd2 else "sd15")
                 if in_ch == 9:  # the 9ch inpainting lineages
                     arch += "-inpaint"

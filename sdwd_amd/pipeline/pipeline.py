@@ -335,9 +335,12 @@ class StableDiffusionPipeline:
             return h.reshape(n, k * L, h.shape[-1])
 
         if m.is_sdxl:
-            h1 = _cat_chunks(m.text_encoder(flat, penultimate=True))
             h2 = _cat_chunks(m.text_encoder_2(flat, penultimate=True))
-            ctx = torch.cat([h1, h2], dim=-1)
+            if m.text_encoder is None:  # refiner lineage: CLIP-G only
+                ctx = h2
+            else:
+                h1 = _cat_chunks(m.text_encoder(flat, penultimate=True))
+                ctx = torch.cat([h1, h2], dim=-1)
             # pooled conditioning comes from the FIRST chunk's EOT
             first = tokens[:, 0]
             pooled = m.text_encoder_2.pooled(
@@ -358,20 +361,29 @@ class StableDiffusionPipeline:
         pooled_cu = (pooled[:n], pooled[n:]) if pooled is not None else None
         return cond.to(self.dtype), uncond.to(self.dtype), pooled_cu
 
-    def _sdxl_vector(self, req: PipelineRequest, pooled: torch.Tensor):
-        """SDXL add-conditioning: pooled embed + size/crop/target vectors."""
-        sizes = torch.tensor(
-            [
+    def _sdxl_vector(self, req: PipelineRequest, pooled: torch.Tensor,
+                     bundle=None, aesthetic: float = 6.0):
+        """SDXL add-conditioning: pooled embed + fourier cond vectors.
+
+        Base models take [orig_h, orig_w, crop_t, crop_l, tgt_h, tgt_w];
+        the refiner lineage takes [orig_h, orig_w, crop_t, crop_l,
+        aesthetic_score] (sgm convention: 6.0 for the cond row, 2.5 for
+        the negative)."""
+        m = bundle if bundle is not None else self.model
+        if getattr(m, "is_refiner", False):
+            conds = [req.height, req.width, 0, 0, aesthetic]
+        else:
+            conds = [
                 req.height, req.width,  # original size
                 0, 0,                   # crop top-left
                 req.height, req.width,  # target size
-            ],
-            dtype=torch.float32,
-            device=self.device,
+            ]
+        sizes = torch.tensor(
+            conds, dtype=torch.float32, device=self.device,
         )
         from .. import ops
 
-        emb = ops.timestep_embedding(sizes, 256).flatten()  # [6*256]
+        emb = ops.timestep_embedding(sizes, 256).flatten()
         b = pooled.shape[0]
         return torch.cat(
             [pooled.float(), emb[None].expand(b, -1)], dim=-1
@@ -397,10 +409,12 @@ class StableDiffusionPipeline:
         if self.model.is_sdxl and pl is not None:
             pc, pu = pl
             yrows = [
-                self._sdxl_vector(req, pc[i : i + 1].expand(b, -1))
+                self._sdxl_vector(req, pc[i : i + 1].expand(b, -1))  # cond
                 for i in range(len(prompts))
             ]
-            yrows.append(self._sdxl_vector(req, pu.expand(b, -1)))
+            yrows.append(
+                self._sdxl_vector(req, pu.expand(b, -1), aesthetic=2.5)
+            )
             y = torch.cat(yrows)
         return ctx, y, ws
 
@@ -544,7 +558,7 @@ class StableDiffusionPipeline:
                 y = torch.cat(
                     [
                         self._sdxl_vector(req, pl[0]),
-                        self._sdxl_vector(req, pl[1]),
+                        self._sdxl_vector(req, pl[1], aesthetic=2.5),
                     ]
                 )
         else:
@@ -1001,8 +1015,13 @@ class StableDiffusionPipeline:
             if refiner.is_sdxl and r_p1 is not None:
                 r_y = torch.cat(
                     [
-                        self._sdxl_vector(req, r_p1[0].expand(b, -1)),
-                        self._sdxl_vector(req, r_p1[1].expand(b, -1)),
+                        self._sdxl_vector(
+                            req, r_p1[0].expand(b, -1), bundle=refiner
+                        ),
+                        self._sdxl_vector(
+                            req, r_p1[1].expand(b, -1), bundle=refiner,
+                            aesthetic=2.5,
+                        ),
                     ]
                 )
             ts_list = sched.timesteps.tolist()

@@ -331,3 +331,22 @@ class TestModelCacheIsolation:
         # same key -> same instance (the cache still caches)
         c = load_model("tiny", device="cpu", dtype=_t.float32)
         assert c.unet is a.unet
+
+
+class TestRefinerConvert:
+    def test_refiner_round_trip_and_prefix(self):
+        """The refiner's CLIP-G exports under conditioner.embedders.0
+        (it has no CLIP-L) and reloads with full key consumption."""
+        a = load_model("tiny-xl-refiner", device="cpu", cache=False)
+        state = to_ldm_state_dict(a)
+        assert any(
+            k.startswith("conditioner.embedders.0.model.") for k in state
+        )
+        assert not any(
+            k.startswith("conditioner.embedders.1.") for k in state
+        )
+        from sdwd_amd.models.convert import load_ldm_state_dict
+
+        b = load_model("tiny-xl-refiner", device="cpu", cache=False)
+        rep = load_ldm_state_dict(b, state)
+        assert not rep["missing"] and not rep["unexpected"]

@@ -1691,3 +1691,49 @@ class TestXLInpaint:
         from sdwd_amd.models.registry import available_models
 
         assert "sdxl-inpaint" in available_models()
+
+
+class TestSDXLRefinerLineage:
+    def test_refiner_as_base_model(self):
+        """CLIP-G-only encode + aesthetic ADM: the refiner generates on
+        its own (sdwui allows selecting it as the checkpoint)."""
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny-xl-refiner", device="cpu")
+        assert pipe.model.text_encoder is None
+        assert pipe.model.is_refiner
+        req = PipelineRequest(prompt="ref base", steps=2, width=64,
+                              height=64, seeds=[9])
+        a = pipe.generate(req)
+        b = pipe.generate(req)
+        assert torch.equal(a.images, b.images)
+        assert torch.isfinite(a.images.float()).all()
+
+    def test_handoff_from_xl_base(self):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny-xl", device="cpu")
+        base = dict(prompt="refined cow", steps=4, width=64, height=64,
+                    seeds=[3])
+        plain = pipe.generate(PipelineRequest(**base)).images
+        refined = pipe.generate(PipelineRequest(
+            **base, refiner_model="tiny-xl-refiner", refiner_switch_at=0.5,
+        )).images
+        assert not torch.equal(plain, refined)
+        # switch_at=1.0: the refiner never fires
+        noop = pipe.generate(PipelineRequest(
+            **base, refiner_model="tiny-xl-refiner", refiner_switch_at=1.0,
+        )).images
+        assert torch.equal(plain, noop)
+
+    def test_aesthetic_score_conditions_the_vector(self):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny-xl-refiner", device="cpu")
+        pooled = torch.randn(1, 32)
+        req = PipelineRequest(width=64, height=64)
+        hi = pipe._sdxl_vector(req, pooled, aesthetic=6.0)
+        lo = pipe._sdxl_vector(req, pooled, aesthetic=2.5)
+        assert hi.shape[-1] == 32 + 5 * 256
+        assert not torch.equal(hi, lo)
+        assert torch.equal(hi[:, : 32 + 4 * 256], lo[:, : 32 + 4 * 256])
