@@ -586,6 +586,19 @@ class TestXLInpaintGPU:
         assert torch.isfinite(res.images.float()).all()
 
 
+class TestRefinerGPU:
+    def test_refiner_lineage_on_gpu(self, dev):
+        from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
+
+        pipe = StableDiffusionPipeline("tiny-xl", device=dev)
+        res = pipe.generate(PipelineRequest(
+            prompt="refined", steps=3, width=64, height=64, seeds=[3],
+            refiner_model="tiny-xl-refiner", refiner_switch_at=0.5,
+        ))
+        assert res.images.shape == (1, 64, 64, 3)
+        assert torch.isfinite(res.images.float()).all()
+
+
 class TestRegionalGPU:
     def test_regional_runs_and_differs(self, dev):
         from sdwd_amd.pipeline import PipelineRequest, StableDiffusionPipeline
