@@ -250,6 +250,19 @@ def _upscale_latent(x: torch.Tensor, scale: float, upscaler: str,
     )
 
 
+# sdwui's NON-"Latent" hr_upscalers work in pixel space (decode ->
+# upscale -> re-encode); the model-free kernels map to torch modes and
+# model-based upscalers (ESRGAN/SwinIR/LDSR — weights can't ship
+# offline) fall back to bicubic-antialiased with a warning.
+_PIXEL_HR_MODES = {
+    "none": ("nearest", False),
+    "lanczos": ("bicubic", True),  # closest torch kernel
+    "pixel nearest": ("nearest", False),
+    "pixel bilinear": ("bilinear", True),
+    "pixel bicubic": ("bicubic", True),
+}
+
+
 def _apply_sampler_params(sampler, req: "PipelineRequest") -> None:
     sampler.s_churn = float(req.s_churn)
     sampler.s_tmin = float(req.s_tmin)
@@ -484,6 +497,45 @@ class StableDiffusionPipeline:
             w = (1.0 - mi) * w + mi * mk
         out = w * new_f + (1.0 - w) * orig.float()
         return out.round().clamp(0, 255).to(torch.uint8)
+
+    def _pixel_hires(
+        self, x: torch.Tensor, req: "PipelineRequest", hr_size, f: int
+    ) -> torch.Tensor:
+        """Pixel-space hires handoff (sdwui non-"Latent" hr_upscalers):
+        decode the base latents, upscale the uint8 pixels (the PIL round
+        trip sdwui performs), re-encode per image (shard-invariant,
+        seeded). Model-based upscaler names fall back to
+        bicubic-antialiased with a warning."""
+        key = (req.hr_upscaler or "").lower()
+        m, aa = _PIXEL_HR_MODES.get(key, ("bicubic", True))
+        if key not in _PIXEL_HR_MODES:
+            log.warning(
+                "hr upscaler %r needs model weights that cannot ship "
+                "offline; using bicubic-antialiased pixels", req.hr_upscaler,
+            )
+        if hr_size is not None:
+            th, tw = hr_size
+        else:
+            th = int(x.shape[2] * req.hr_scale)
+            tw = int(x.shape[3] * req.hr_scale)
+        kwargs = {"antialias": True} if aa else {}
+        outs = []
+        for i in range(x.shape[0]):
+            img = self.model.vae.decode(x[i : i + 1].to(self.dtype))
+            img8 = (
+                ((img.float() + 1.0) * 127.5).clamp(0, 255).to(torch.uint8)
+            )
+            up = torch.nn.functional.interpolate(
+                img8.float(), size=(th * f, tw * f), mode=m, **kwargs
+            )
+            outs.append(
+                up.clamp(0, 255).to(torch.uint8)[0].permute(1, 2, 0).cpu()
+            )
+        lat = self.encode_image(
+            torch.stack(outs),
+            seeds=[(int(s) ^ 0x9C51) & 0xFFFFFFFF for s in req.seeds],
+        )
+        return lat.to(self.dtype)
 
     # -- the denoise loop ----------------------------------------------------
     @torch.no_grad()
@@ -1136,9 +1188,14 @@ class StableDiffusionPipeline:
             hr_size = None
             if req.hr_resize_x > 0 and req.hr_resize_y > 0:
                 hr_size = (req.hr_resize_y // f, req.hr_resize_x // f)
-            x = _upscale_latent(
-                x.float(), req.hr_scale, req.hr_upscaler, size=hr_size
-            ).to(self.dtype)
+            up_key = (req.hr_upscaler or "nearest").lower()
+            if up_key in _HR_MODES:
+                x = _upscale_latent(
+                    x.float(), req.hr_scale, req.hr_upscaler, size=hr_size
+                ).to(self.dtype)
+            else:
+                # sdwui non-"Latent" upscaler: pixel space round trip
+                x = self._pixel_hires(x, req, hr_size, f)
             hr_sampler = req.hr_sampler_name or req.sampler_name
             hsched = schedule_for(hr_sampler, hr_steps, req.scheduler)
             start = max(
