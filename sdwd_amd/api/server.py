@@ -878,6 +878,7 @@ def create_app(engine: Optional[LocalEngine] = None,
     @app.post("/sdwd/benchmark")
     def benchmark(rebenchmark: bool = True):
         speeds = engine.benchmark(rebenchmark=rebenchmark)
+        engine.world.save()  # ref world.py:278 persisted post-benchmark
         return {"speeds": speeds}
 
     @app.get("/sdwd/settings")
