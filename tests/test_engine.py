@@ -835,3 +835,19 @@ class TestDeviceSelection:
         assert len(eng.devices) == 3
         monkeypatch.delenv("SDWD_DEVICES")
         assert len(LocalEngine(model="tiny").devices) == 1
+
+
+class TestThinClientEngine:
+    def test_master_orchestrates_without_a_shard(self):
+        """ref C10: in thin-client mode rank 0 only schedules/assembles;
+        the batch is produced entirely by the other ranks."""
+        eng = make_engine(3)
+        eng.world.settings.thin_client = True
+        res = eng.generate(GenerationRequest(
+            prompt="tc", batch_size=4, width=64, height=64, steps=1,
+            seed=2,
+        ))
+        assert res.images.shape == (4, 64, 64, 3)
+        assert res.seeds == [2, 3, 4, 5]
+        assert not any("gpu0" in s for s in res.job_summary)
+        assert all(res.images[i].float().std() > 0 for i in range(4))
