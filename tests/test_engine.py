@@ -851,3 +851,18 @@ class TestThinClientEngine:
         assert res.seeds == [2, 3, 4, 5]
         assert not any("gpu0" in s for s in res.job_summary)
         assert all(res.images[i].float().std() > 0 for i in range(4))
+
+
+class TestSubseedSharding:
+    def test_variation_seeds_shard_like_plain_seeds(self):
+        """Subseed variation (shared base seed + per-slot subseeds) keeps
+        the C22 contract: a 2-rank gallery matches 1-rank to library-GEMM
+        rounding (±1 uint8 LSB on a handful of pixels)."""
+        req = dict(prompt="var", batch_size=4, width=64, height=64,
+                   steps=2, seed=70, subseed=99, subseed_strength=0.5)
+        one = make_engine(1).generate(GenerationRequest(**req))
+        two = make_engine(2).generate(GenerationRequest(**req))
+        assert one.seeds == two.seeds
+        diff = (one.images.float() - two.images.float()).abs()
+        assert diff.max() <= 1.0
+        assert (diff > 0).float().mean() < 1e-3
