@@ -866,3 +866,16 @@ class TestSubseedSharding:
         diff = (one.images.float() - two.images.float()).abs()
         assert diff.max() <= 1.0
         assert (diff > 0).float().mean() < 1e-3
+
+    def test_hires_shards_like_base(self):
+        """The hires second pass (per-image re-noise seeds) keeps shard
+        parity at the upscaled resolution."""
+        req = dict(prompt="hr", batch_size=4, width=64, height=64,
+                   steps=2, seed=31, enable_hr=True, hr_scale=2.0,
+                   hr_steps=2, denoising_strength=0.6)
+        one = make_engine(1).generate(GenerationRequest(**req))
+        two = make_engine(2).generate(GenerationRequest(**req))
+        assert one.images.shape == (4, 128, 128, 3)
+        diff = (one.images.float() - two.images.float()).abs()
+        assert diff.max() <= 1.0
+        assert (diff > 0).float().mean() < 1e-3
