@@ -879,3 +879,19 @@ class TestSubseedSharding:
         diff = (one.images.float() - two.images.float()).abs()
         assert diff.max() <= 1.0
         assert (diff > 0).float().mean() < 1e-3
+
+    def test_img2img_color_correction_and_ensd_shard_exact(self):
+        """img2img + color correction + eta_noise_seed_delta: bit-exact
+        across shard splits (per-image encode + per-image ancestral
+        generators + post-gather per-image color matching)."""
+        init = torch.randint(
+            0, 255, (4, 64, 64, 3), dtype=torch.uint8,
+            generator=torch.Generator().manual_seed(3),
+        )
+        req = dict(prompt="cc", batch_size=4, width=64, height=64,
+                   steps=2, seed=44, init_images=init,
+                   denoising_strength=0.7, color_correction=True,
+                   eta_noise_seed_delta=31337)
+        one = make_engine(1).generate(GenerationRequest(**req))
+        two = make_engine(2).generate(GenerationRequest(**req))
+        assert torch.equal(one.images, two.images)
