@@ -25,11 +25,29 @@ import torch
 
 from ..core import GenRequest, Job, State, World, Worker
 from ..pipeline import PipelineRequest, StableDiffusionPipeline
+from ..pipeline.pipeline import hires_active, hr_target_resolution
 from ..utils import get_logger
 from ..utils.images import make_grid
 from . import group as pg
 
 log = get_logger("engine")
+
+
+def _final_output_hw(gen) -> "tuple[int, int]":
+    """Pixel (H, W) the pipeline actually decodes for this request — the
+    latent floor math, including sdwui hires resize-to/truncate semantics
+    (pipeline.hr_target_resolution). Placeholder/empty-shard tensors sized
+    from this always match generating ranks' output shapes."""
+    f = 8
+    if hires_active(
+        gen.enable_hr, gen.hr_scale, gen.hr_resize_x, gen.hr_resize_y
+    ):
+        uh, uw, ty, tx = hr_target_resolution(
+            gen.width, gen.height, gen.hr_scale,
+            gen.hr_resize_x, gen.hr_resize_y, f,
+        )
+        return (uh - ty) * f, (uw - tx) * f
+    return (gen.height // f) * f, (gen.width // f) * f
 
 
 @dataclass
@@ -107,6 +125,18 @@ class GenerationRequest:
     si_difference_contrast: float = 2.0
 
     def sched(self) -> GenRequest:
+        hr_on = hires_active(
+            self.enable_hr, self.hr_scale, self.hr_resize_x, self.hr_resize_y
+        )
+        eff_scale = 0.0
+        if hr_on:
+            # ETA model wants one linear scale factor; "resize to" mode
+            # (hr_resize_x/y) implies it from the actual output pixels
+            oh, ow = _final_output_hw(self)
+            eff_scale = (
+                (oh * ow)
+                / max(1, ((self.height // 8) * 8) * ((self.width // 8) * 8))
+            ) ** 0.5
         return GenRequest(
             task="img2img" if self.init_images is not None else "txt2img",
             batch_size=self.batch_size,
@@ -117,8 +147,8 @@ class GenerationRequest:
             seed=self.seed,
             subseed=self.subseed,
             subseed_strength=self.subseed_strength,
-            hr_scale=self.hr_scale if self.enable_hr else 0.0,
-            hr_steps=(self.hr_steps or self.steps) if self.enable_hr else 0,
+            hr_scale=eff_scale,
+            hr_steps=(self.hr_steps or self.steps) if hr_on else 0,
         )
 
 
@@ -408,8 +438,7 @@ class _EngineBase:
         """Order shards by gallery offset, fix up infotexts with the worker
         label (ref distributed.py:343-349), build the grid."""
         total = sum(j.batch_size for j in jobs)
-        f = gen.hr_scale if gen.enable_hr else 1.0
-        h, w = int(gen.height * f), int(gen.width * f)
+        h, w = _final_output_hw(gen)
         for sh in shards.values():  # trust actual shard shape (fractional hr)
             if sh is not None and sh.numel():
                 h, w = sh.shape[1], sh.shape[2]
@@ -879,8 +908,7 @@ class DistributedEngine(_EngineBase):
         """Run one shard on this rank. Returns images/infos/status; never
         raises (failures are reported through ok=0 so the collectives that
         follow stay symmetric across ranks)."""
-        hf = gen.hr_scale if gen.enable_hr else 1.0
-        h, w = int(gen.height * hf), int(gen.width * hf)
+        h, w = _final_output_hw(gen)
         out = {
             "images": torch.zeros(0, h, w, 3, dtype=torch.uint8),
             "infos": [],

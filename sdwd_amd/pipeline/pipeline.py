@@ -250,6 +250,49 @@ def _upscale_latent(x: torch.Tensor, scale: float, upscaler: str,
     )
 
 
+def hires_active(enable_hr: bool, hr_scale: float, rx: int, ry: int) -> bool:
+    """True when the hires second pass runs: sdwui's "resize to" mode
+    (hr_resize_x/y set) enables the pass even with hr_scale<=1 (the UI
+    sends hr_scale=0 in that mode)."""
+    return bool(enable_hr) and (hr_scale > 1.0 or rx > 0 or ry > 0)
+
+
+def hr_target_resolution(
+    width: int, height: int, hr_scale: float, rx: int, ry: int, f: int = 8
+):
+    """sdwui hires target-resolution semantics
+    (processing.py calculate_target_resolution):
+
+    - rx == ry == 0: upscale by hr_scale;
+    - one of rx/ry set: the other follows the source aspect ratio;
+    - both set: upscale preserving aspect until the target is covered,
+      then center-crop ("truncate") the latent down to the target.
+
+    Returns (up_lat_h, up_lat_w, crop_lat_y, crop_lat_x): the latent
+    dims to upscale to, and how many latent rows/cols to crop off
+    (split top/bottom, left/right) before the second denoise pass.
+    """
+    if rx <= 0 and ry <= 0:
+        ux, uy = int(width * hr_scale), int(height * hr_scale)
+        tx = ty = 0
+    elif ry <= 0:
+        ux, uy = rx, rx * height // width
+        tx = ty = 0
+    elif rx <= 0:
+        ux, uy = ry * width // height, ry
+        tx = ty = 0
+    else:
+        src_ratio = width / height
+        dst_ratio = rx / ry
+        if src_ratio < dst_ratio:
+            ux, uy = rx, rx * height // width
+        else:
+            ux, uy = ry * width // height, ry
+        tx = (ux - rx) // f
+        ty = (uy - ry) // f
+    return uy // f, ux // f, max(0, ty), max(0, tx)
+
+
 # sdwui's NON-"Latent" hr_upscalers work in pixel space (decode ->
 # upscale -> re-encode); the model-free kernels map to torch modes and
 # model-based upscalers (ESRGAN/SwinIR/LDSR — weights can't ship
@@ -1173,7 +1216,12 @@ class StableDiffusionPipeline:
         # hires fix: latent-upscale the base result and run a second,
         # strength-limited denoise pass at the scaled resolution
         # (ref eta_hr, worker.py:205-228 predicts exactly this shape).
-        if req.enable_hr and req.hr_scale > 1.0 and not was_interrupted:
+        if (
+            hires_active(
+                req.enable_hr, req.hr_scale, req.hr_resize_x, req.hr_resize_y
+            )
+            and not was_interrupted
+        ):
             hr_steps = req.hr_steps or req.steps
             if req.hr_prompt or req.hr_negative_prompt:
                 # the hires pass denoises under its own conditioning;
@@ -1185,9 +1233,11 @@ class StableDiffusionPipeline:
                 hn = req.hr_negative_prompt or req.negative_prompt
                 ctx, y, and_ws = self._build_ctx(hp, hn, req, b)
                 seg_tensors.clear()
-            hr_size = None
-            if req.hr_resize_x > 0 and req.hr_resize_y > 0:
-                hr_size = (req.hr_resize_y // f, req.hr_resize_x // f)
+            uh, uw, crop_y, crop_x = hr_target_resolution(
+                req.width, req.height, req.hr_scale,
+                req.hr_resize_x, req.hr_resize_y, f,
+            )
+            hr_size = (uh, uw)
             up_key = (req.hr_upscaler or "nearest").lower()
             if up_key in _HR_MODES:
                 x = _upscale_latent(
@@ -1196,6 +1246,15 @@ class StableDiffusionPipeline:
             else:
                 # sdwui non-"Latent" upscaler: pixel space round trip
                 x = self._pixel_hires(x, req, hr_size, f)
+            if crop_y or crop_x:
+                # sdwui "truncate": both hr_resize dims set -> the
+                # aspect-covering upscale is center-cropped to the target
+                # BEFORE the second pass denoises at that size
+                x = x[
+                    :, :,
+                    crop_y // 2 : x.shape[2] - (crop_y - crop_y // 2),
+                    crop_x // 2 : x.shape[3] - (crop_x - crop_x // 2),
+                ]
             hr_sampler = req.hr_sampler_name or req.sampler_name
             hsched = schedule_for(hr_sampler, hr_steps, req.scheduler)
             start = max(
@@ -1247,9 +1306,7 @@ class StableDiffusionPipeline:
             # chunked decode: bounds decoder activation memory at large
             # batches (and sidesteps a torch NHWC-upsample grid-size limit
             # seen at batch 64 x 512^2); scale the chunk down with pixels
-            px = req.width * req.height * (
-                req.hr_scale**2 if req.enable_hr else 1.0
-            )
+            px = x.shape[2] * x.shape[3] * f * f  # actual final latent dims
             default = max(1, int(16 * (512 * 512) / max(px, 1)))
             chunk = int(os.environ.get("SDWD_DECODE_CHUNK", str(default)))
             outs = []
@@ -1267,7 +1324,10 @@ class StableDiffusionPipeline:
                 req.soft_inpainting
                 and req.mask_image is not None
                 and req.init_latents is not None
-                and not (req.enable_hr and req.hr_scale > 1.0)
+                and not hires_active(
+                    req.enable_hr, req.hr_scale,
+                    req.hr_resize_x, req.hr_resize_y,
+                )
             ):
                 images = self._soft_composite(images, req)
         else:
@@ -1295,9 +1355,17 @@ class StableDiffusionPipeline:
             )
         if req.subseed_strength > 0 and req.subseeds:
             extra += f", Variation seed strength: {req.subseed_strength}"
-        if req.enable_hr and req.hr_scale > 1.0:
+        if hires_active(
+            req.enable_hr, req.hr_scale, req.hr_resize_x, req.hr_resize_y
+        ):
+            if req.hr_resize_x > 0 or req.hr_resize_y > 0:
+                # sdwui "resize to" mode reports the target, not a scale
+                extra += (
+                    f", Hires resize: {req.hr_resize_x}x{req.hr_resize_y}"
+                )
+            else:
+                extra += f", Hires upscale: {req.hr_scale}"
             extra += (
-                f", Hires upscale: {req.hr_scale}"
                 f", Hires steps: {req.hr_steps or req.steps}"
                 f", Hires upscaler: {req.hr_upscaler}"
             )

@@ -56,6 +56,30 @@ class TestTxt2Img:
         assert len(info["infotexts"]) == 2
 
 
+class TestBatchCount:
+    def test_n_iter_folds_into_gallery(self, client):
+        """sdwui batch_count (n_iter): N sequential batches; here the sharder
+        runs them as one contiguous gallery with the same seed plan."""
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={
+                "prompt": "a cow",
+                "steps": 2,
+                "width": 64,
+                "height": 64,
+                "batch_size": 2,
+                "n_iter": 2,
+                "seed": 300,
+            },
+        )
+        assert r.status_code == 200
+        body = r.json()
+        assert len(body["images"]) == 5  # grid + 4
+        info = json.loads(body["info"])
+        assert info["all_seeds"] == [300, 301, 302, 303]
+        assert len(info["infotexts"]) == 4
+
+
 class TestImg2Img:
     def test_round_trip(self, client):
         init = torch.randint(0, 255, (64, 64, 3), dtype=torch.uint8)

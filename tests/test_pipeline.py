@@ -256,6 +256,65 @@ class TestHiresFix:
         assert res.images.shape == (2, 128, 128, 3)
 
 
+class TestHiresResizeTo:
+    """sdwui 'resize to' hires mode: hr_resize_x/y drive the target (the
+    UI sends hr_scale=0 in that mode), one-sided resize preserves aspect,
+    and a both-sided aspect mismatch upscales-to-cover then center-crops
+    the latent ("truncate") before the second pass."""
+
+    def test_target_resolution_math(self):
+        from sdwd_amd.pipeline.pipeline import hr_target_resolution
+
+        # pure scale mode
+        assert hr_target_resolution(512, 512, 2.0, 0, 0) == (128, 128, 0, 0)
+        # both set, same aspect: no crop
+        assert hr_target_resolution(512, 512, 0.0, 768, 768) == (96, 96, 0, 0)
+        # both set, wider target: cover by width, crop rows
+        assert hr_target_resolution(512, 512, 0.0, 768, 512) == (96, 96, 32, 0)
+        # one-sided: the other dim follows the source aspect
+        assert hr_target_resolution(512, 256, 0.0, 1024, 0) == (64, 128, 0, 0)
+        assert hr_target_resolution(512, 256, 0.0, 0, 512) == (64, 128, 0, 0)
+
+    def test_resize_to_with_zero_scale(self, pipe):
+        res = pipe.generate(PipelineRequest(
+            prompt="hr", steps=2, width=64, height=64, seeds=[5],
+            enable_hr=True, hr_scale=0.0, hr_steps=2,
+            hr_resize_x=96, hr_resize_y=96, denoising_strength=0.6,
+        ))
+        assert res.images.shape == (1, 96, 96, 3)
+        assert ", Hires resize: 96x96" in res.infotexts[0]
+
+    def test_one_sided_resize_keeps_aspect(self, pipe):
+        res = pipe.generate(PipelineRequest(
+            prompt="hr", steps=2, width=64, height=32, seeds=[5],
+            enable_hr=True, hr_scale=0.0, hr_steps=2,
+            hr_resize_x=128, denoising_strength=0.6,
+        ))
+        assert res.images.shape == (1, 64, 128, 3)
+
+    def test_cover_then_truncate(self, pipe):
+        # 64x64 -> target 96x64: aspect-cover to 96x96, crop rows to 64
+        res = pipe.generate(PipelineRequest(
+            prompt="hr", steps=2, width=64, height=64, seeds=[5],
+            enable_hr=True, hr_scale=0.0, hr_steps=2,
+            hr_resize_x=96, hr_resize_y=64, denoising_strength=0.6,
+        ))
+        assert res.images.shape == (1, 64, 96, 3)
+
+    def test_engine_resize_to(self):
+        from sdwd_amd.parallel import GenerationRequest, LocalEngine
+
+        eng = LocalEngine(model="tiny", devices=["cpu", "cpu"])
+        for w in eng.world.workers:
+            w.eta.avg_ipm = 60.0
+        res = eng.generate(GenerationRequest(
+            prompt="hr", batch_size=2, width=64, height=64, steps=2,
+            seed=4, enable_hr=True, hr_scale=0.0, hr_steps=2,
+            hr_resize_x=96, hr_resize_y=96,
+        ))
+        assert res.images.shape == (2, 96, 96, 3)
+
+
 class TestControlNet:
     def test_control_changes_output(self, pipe):
         from sdwd_amd.pipeline import PipelineRequest
