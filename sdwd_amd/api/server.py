@@ -77,6 +77,10 @@ class Txt2ImgRequest(BaseModel):
     # sdwui per-request overrides (sd_model_checkpoint,
     # CLIP_stop_at_last_layers are honored; the rest are ignored)
     override_settings: Dict[str, Any] = Field(default_factory=dict)
+    # sdwui selectable script (X/Y/Z plot runs natively; any other
+    # script_name is rejected loudly — parallel/xyz.py docstring)
+    script_name: Optional[str] = ""  # sdwui clients send null for "none"
+    script_args: List[Any] = Field(default_factory=list)
     send_images: bool = True   # omit base64 images from the response
     save_images: bool = False  # persist PNGs server-side (SDWD_OUTPUT_DIR)
 
@@ -387,6 +391,112 @@ def create_app(engine: Optional[LocalEngine] = None,
             "info": json.dumps(info),
         }
 
+    _XYZ_NAMES = {"x/y/z plot", "xyz plot", "xyz grid", "x/y/z", "xyz"}
+
+    def _parse_xyz_args(args: List[Any]) -> Dict[str, Any]:
+        """Positional sdwui xyz_grid script_args, both layouts:
+
+        new (dropdown) — [x_type, x_values, x_values_dropdown, y_type,
+            y_values, y_values_dropdown, z_type, z_values,
+            z_values_dropdown, draw_legend, no_fixed_seeds,
+            include_lone_images, include_sub_grids, margin, csv_mode];
+        old — [x_type, x_values, y_type, y_values, z_type, z_values,
+            draw_legend, include_lone_images, include_sub_grids,
+            no_fixed_seeds].
+
+        Axis types are int indices into our published AXIS_OPTIONS or
+        axis names (version-proof). A dropdown list, when non-empty,
+        takes precedence over the paired values string.
+        """
+        if not args:
+            raise HTTPException(
+                422, "X/Y/Z plot needs script_args (x_type, x_values, ...)"
+            )
+
+        def _vals(raw, drop):
+            return drop if isinstance(drop, (list, tuple)) and drop else raw
+
+        dropdown = len(args) >= 9 and (
+            isinstance(args[2], (list, tuple))
+            or isinstance(args[5], (list, tuple))
+            or isinstance(args[8], (list, tuple))
+        )
+        # heuristic: 15-arg payloads are always the dropdown layout
+        dropdown = dropdown or len(args) >= 13
+        g = lambda i, d=None: args[i] if i < len(args) else d  # noqa: E731
+        if dropdown:
+            out = dict(
+                x_axis=g(0, 0), x_values=_vals(g(1, ""), g(2)),
+                y_axis=g(3, 0), y_values=_vals(g(4, ""), g(5)),
+                z_axis=g(6, 0), z_values=_vals(g(7, ""), g(8)),
+                no_fixed_seeds=bool(g(10, False)),
+                include_lone_images=bool(g(11, False)),
+                include_sub_grids=bool(g(12, False)),
+            )
+        else:
+            out = dict(
+                x_axis=g(0, 0), x_values=g(1, ""),
+                y_axis=g(2, 0), y_values=g(3, ""),
+                z_axis=g(4, 0), z_values=g(5, ""),
+                include_lone_images=bool(g(7, False)),
+                include_sub_grids=bool(g(8, False)),
+                no_fixed_seeds=bool(g(9, False)),
+            )
+        return out
+
+    def run_script(gen: GenerationRequest, req: Txt2ImgRequest
+                   ) -> Dict[str, Any]:
+        """Dispatch a sdwui selectable script (script_name). Only the
+        natively-implemented X/Y/Z plot runs; anything else is rejected
+        loudly (same degradation the reference applied when a remote
+        lacked a script, distributed.py:199-234)."""
+        from ..parallel.xyz import run_xyz
+
+        name = req.script_name.strip().lower()
+        if name not in _XYZ_NAMES:
+            raise HTTPException(
+                422,
+                f"selectable script {req.script_name!r} is not available "
+                "natively; supported: X/Y/Z plot",
+            )
+        kw = _parse_xyz_args(req.script_args)
+        try:
+            with state.lock:
+                state.busy = True
+                state.started_at = time.time()
+                try:
+                    out = run_xyz(engine, gen, **kw)
+                finally:
+                    state.busy = False
+        except ValueError as exc:
+            raise HTTPException(422, str(exc))
+        images = []
+        if req.send_images:
+            if out["grid"] is not None:
+                images.append(_b64_png(out["grid"]))
+            for sg in out["sub_grids"]:
+                images.append(_b64_png(sg))
+            images.extend(
+                base64.b64encode(
+                    encode_png(img, info)
+                ).decode()
+                for img, info in zip(out["images"], out["infotexts"])
+            )
+        info = {
+            "all_seeds": out["seeds"],
+            "all_subseeds": [-1] * len(out["seeds"]),
+            "all_prompts": [gen.prompt] * len(out["seeds"]),
+            "all_negative_prompts": [gen.negative_prompt] * len(out["seeds"]),
+            "infotexts": out["infotexts"],
+            "xyz_plot": out["labels"],
+            "interrupted": out["interrupted"],
+        }
+        return {
+            "images": images,
+            "parameters": {"script_name": req.script_name},
+            "info": json.dumps(info),
+        }
+
     @app.get("/")
     def index():
         """Built-in control surface (ref C19)."""
@@ -468,6 +578,8 @@ def create_app(engine: Optional[LocalEngine] = None,
         )
         if _dynamic_prompts_enabled(req.alwayson_scripts):
             _expand_dynamic(gen)
+        if req.script_name:
+            return run_script(gen, req)
         return run_generation(gen, req.send_images, req.save_images)
 
     @app.post("/sdapi/v1/img2img")
@@ -532,6 +644,8 @@ def create_app(engine: Optional[LocalEngine] = None,
             **_parse_soft_inpainting(req.alwayson_scripts),
             **_parse_regional_prompter(req.alwayson_scripts),
         )
+        if req.script_name:
+            return run_script(gen, req)
         return run_generation(gen, req.send_images, req.save_images)
 
     @app.post("/sdapi/v1/options")
@@ -677,12 +791,18 @@ def create_app(engine: Optional[LocalEngine] = None,
         # each remote's script list through exactly this surface
         # (worker.py:375-404), so report what this engine runs in-process
         return {
-            "txt2img": list(_NATIVE_ALWAYSON),
-            "img2img": list(_NATIVE_ALWAYSON),
+            "txt2img": list(_NATIVE_ALWAYSON) + ["x/y/z plot"],
+            "img2img": list(_NATIVE_ALWAYSON) + ["x/y/z plot"],
         }
 
     @app.get("/sdapi/v1/script-info")
     def script_info():
+        from ..parallel.xyz import AXIS_OPTIONS
+
+        xyz_args = [
+            {"label": f"{i}={o.name}", "kind": o.kind}
+            for i, o in enumerate(AXIS_OPTIONS)
+        ]
         return [
             {"name": n, "is_alwayson": True, "is_img2img": True,
              "args": []}
@@ -691,6 +811,10 @@ def create_app(engine: Optional[LocalEngine] = None,
             {"name": n, "is_alwayson": True, "is_img2img": False,
              "args": []}
             for n in _NATIVE_ALWAYSON
+        ] + [
+            {"name": "x/y/z plot", "is_alwayson": False,
+             "is_img2img": im, "args": xyz_args}
+            for im in (True, False)
         ]
 
     @app.get("/sdapi/v1/cmd-flags")
@@ -729,6 +853,14 @@ def create_app(engine: Optional[LocalEngine] = None,
 
     @app.post("/sdapi/v1/interrupt")
     def interrupt():
+        engine.interrupt()
+        return {}
+
+    @app.post("/sdapi/v1/skip")
+    def skip():
+        """sdwui 'skip' ends the current iteration and continues the job;
+        with n_iter folded into one sharded batch the nearest semantic is
+        interrupting the in-flight generation (partial gallery returns)."""
         engine.interrupt()
         return {}
 
