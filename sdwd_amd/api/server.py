@@ -460,6 +460,30 @@ def create_app(engine: Optional[LocalEngine] = None,
                 "natively; supported: X/Y/Z plot",
             )
         kw = _parse_xyz_args(req.script_args)
+        # pre-validate axes whose bad values would otherwise produce a
+        # misleading grid (sampler falls back to Euler a by design) or
+        # fail mid-sweep (unknown checkpoint)
+        from ..parallel.xyz import parse_axis_values, resolve_axis
+        from ..pipeline import sampler_names
+
+        for axis_key, vals_key in (
+            ("x_axis", "x_values"), ("y_axis", "y_values"),
+            ("z_axis", "z_values"),
+        ):
+            try:
+                ax = resolve_axis(kw[axis_key])
+                vals = parse_axis_values(ax.kind, kw[vals_key])
+            except ValueError as exc:
+                raise HTTPException(422, str(exc))
+            if ax.name == "Checkpoint name":
+                for v in vals:
+                    if v not in available_models():
+                        raise HTTPException(422, f"unknown model {v!r}")
+            if ax.name == "Sampler":
+                known = {s.lower() for s in sampler_names()}
+                for v in vals:
+                    if str(v).lower() not in known:
+                        raise HTTPException(422, f"unknown sampler {v!r}")
         try:
             with state.lock:
                 state.busy = True

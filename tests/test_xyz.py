@@ -267,3 +267,23 @@ class TestXYZApi:
 
     def test_skip_route(self, client):
         assert client.post("/sdapi/v1/skip").status_code == 200
+
+    def test_unknown_sampler_axis_value_rejected(self, client):
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={"prompt": "x", "script_name": "x/y/z plot",
+                  "script_args": ["Sampler", "Euler a, NotASampler",
+                                   0, "", 0, ""]},
+        )
+        assert r.status_code == 422
+        assert "NotASampler" in r.json()["detail"]
+
+    def test_unknown_checkpoint_axis_value_rejected(self, client):
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={"prompt": "x", "script_name": "x/y/z plot",
+                  "script_args": ["Checkpoint name", "no-such-model",
+                                   0, "", 0, ""]},
+        )
+        assert r.status_code == 422
+        assert "no-such-model" in r.json()["detail"]
