@@ -56,6 +56,7 @@ class Txt2ImgRequest(BaseModel):
     s_tmax: float = 0.0
     s_noise: float = 1.0
     s_min_uncond: float = 0.0
+    eta: float = -1.0  # sdwui Eta for ancestral/DDIM; -1/unset = default
     # hires fix (sdwui fields)
     enable_hr: bool = False
     hr_scale: float = 2.0
@@ -583,6 +584,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             s_tmax=req.s_tmax,
             s_noise=req.s_noise,
             s_min_uncond=req.s_min_uncond,
+            eta=req.eta,
             enable_hr=req.enable_hr,
             hr_scale=req.hr_scale,
             hr_steps=req.hr_second_pass_steps,
@@ -654,6 +656,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             s_tmax=req.s_tmax,
             s_noise=req.s_noise,
             s_min_uncond=req.s_min_uncond,
+            eta=req.eta,
             init_images=inits,
             denoising_strength=req.denoising_strength,
             mask_image=mask_image,

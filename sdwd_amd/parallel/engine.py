@@ -109,6 +109,7 @@ class GenerationRequest:
     s_tmax: float = 0.0
     s_noise: float = 1.0
     s_min_uncond: float = 0.0
+    eta: float = -1.0  # sdwui Eta (ancestral/SDE); -1 = sampler default
     # sdwui img2img color correction: match output statistics to the init
     color_correction: bool = False
     # Regional Prompter matrix mode (see PipelineRequest for semantics)
@@ -215,6 +216,7 @@ def _job_pipeline_request(
         s_tmax=gen.s_tmax,
         s_noise=gen.s_noise,
         s_min_uncond=gen.s_min_uncond,
+        eta=gen.eta,
         regional_mode=gen.regional_mode,
         regional_ratios=gen.regional_ratios,
         regional_base_ratio=gen.regional_base_ratio,

@@ -160,6 +160,7 @@ AXIS_OPTIONS: List[AxisOption] = [
         "ENSD", "int",
         lambda g, v: replace(g, eta_noise_seed_delta=int(v)),
     ),
+    AxisOption("Eta", "float", lambda g, v: replace(g, eta=float(v))),
     AxisOption("Width", "int", lambda g, v: replace(g, width=int(v))),
     AxisOption("Height", "int", lambda g, v: replace(g, height=int(v))),
 ]

@@ -97,6 +97,8 @@ class PipelineRequest:
     s_noise: float = 1.0
     # skip the uncond eval below this sigma (sdwui s_min_uncond perf knob)
     s_min_uncond: float = 0.0
+    # sdwui Eta: ancestral/SDE noise multiplier; None = sampler default
+    eta: float = -1.0
     # controlnet (ref C17 executed natively); either the single-unit
     # legacy fields or a list of unit dicts with
     # {image, model, scale, guidance_start, guidance_end}
@@ -311,6 +313,8 @@ def _apply_sampler_params(sampler, req: "PipelineRequest") -> None:
     sampler.s_tmin = float(req.s_tmin)
     sampler.s_tmax = float(req.s_tmax) if req.s_tmax > 0 else float("inf")
     sampler.s_noise = float(req.s_noise)
+    if req.eta >= 0:  # sdwui Eta; -1 keeps the sampler's own default
+        sampler.eta = float(req.eta)
 
 
 def _image_noise(
@@ -1355,6 +1359,8 @@ class StableDiffusionPipeline:
             )
         if req.subseed_strength > 0 and req.subseeds:
             extra += f", Variation seed strength: {req.subseed_strength}"
+        if req.eta >= 0:
+            extra += f", Eta: {req.eta}"
         if hires_active(
             req.enable_hr, req.hr_scale, req.hr_resize_x, req.hr_resize_y
         ):

@@ -55,6 +55,10 @@ class Sampler:
     """Base: walks the schedule, delegates the per-step update."""
 
     order = 1  # model evals per step
+    # sdwui "Eta" (ancestral-noise multiplier, opts eta_ancestral /
+    # eta_ddim): consumed by the ancestral and SDE families; 0 makes
+    # them deterministic. Overridden per request via the API `eta` field.
+    eta = 1.0
 
     def __init__(self, schedule: Schedule):
         self.schedule = schedule
@@ -137,11 +141,22 @@ class EulerAncestral(Sampler):
     def step(self, model_fn, x, sigma, sigma_next, t, noise_fn,
              t_next=None):
         denoised = _eval(model_fn, x, sigma, t)
-        sd, su = _ancestral_sigmas(sigma, sigma_next)
+        if self.eta == 0:
+            # deterministic path is bit-exact Euler (no sqrt round trip)
+            return ops.euler_step(x, denoised, sigma, sigma_next)
+        sd, su = _ancestral_sigmas(sigma, sigma_next, self.eta)
         x = ops.euler_step(x, denoised, sigma, sd)
         if su > 0 and noise_fn is not None:
             x = ops.add_noise(x, noise_fn(), 1.0, su)
         return x
+
+
+class DDIMSampler(EulerAncestral):
+    """Deterministic DDIM == Euler in sigma space (eta 0, sdwui's
+    eta_ddim default); a request-level eta > 0 restores the stochastic
+    DDIM update (ancestral noise scaled by eta)."""
+
+    eta = 0.0
 
 
 class Heun(Sampler):
@@ -203,7 +218,7 @@ class DPMppSDE(Sampler):
         x_mid = ops.euler_step(x, denoised, sigma, sigma_mid)
         t_mid = self._t_for(sigma_mid, sigma, sigma_next, t, t_next)
         denoised2 = _eval(model_fn, x_mid, sigma_mid, t_mid)
-        sd, su = _ancestral_sigmas(sigma, sigma_next)
+        sd, su = _ancestral_sigmas(sigma, sigma_next, self.eta)
         x = ops.euler_step(x, denoised2, sigma, sd)
         if su > 0 and noise_fn is not None:
             x = ops.add_noise(x, noise_fn(), 1.0, su)
@@ -357,7 +372,7 @@ class DPMpp2SAncestral(Sampler):
         denoised = _eval(model_fn, x, sigma, t)
         if sigma_next <= 0:
             return denoised
-        sd, su = _ancestral_sigmas(sigma, sigma_next)
+        sd, su = _ancestral_sigmas(sigma, sigma_next, self.eta)
         tt, tn = -math.log(sigma), -math.log(sd)
         h = tn - tt
         s = tt + 0.5 * h
@@ -747,7 +762,7 @@ class LMS(Sampler):
 SAMPLERS: Dict[str, type] = {
     "Euler": Euler,
     "Euler a": EulerAncestral,
-    "DDIM": Euler,  # deterministic DDIM == Euler in sigma space
+    "DDIM": DDIMSampler,  # eta 0 == Euler in sigma space; eta>0 noises
     "Heun": Heun,
     "DPM++ 2M": DPMpp2M,
     "DPM++ 2M Karras": DPMpp2M,

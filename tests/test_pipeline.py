@@ -913,6 +913,53 @@ class TestSamplerParams:
         b = pipe.generate(PipelineRequest(**base, s_churn=0.0)).images
         assert torch.equal(a, b)
 
+    def test_eta_zero_ancestral_equals_euler(self, pipe):
+        """Euler a with eta=0 IS Euler (su=0, the deterministic
+        short-circuit makes it bit-exact)."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="e", steps=4, width=64, height=64, seeds=[9])
+        a = pipe.generate(
+            PipelineRequest(**base, sampler_name="Euler a", eta=0.0)
+        ).images
+        b = pipe.generate(
+            PipelineRequest(**base, sampler_name="Euler")
+        ).images
+        assert torch.equal(a, b)
+        assert ", Eta: 0.0" in pipe.generate(
+            PipelineRequest(**base, sampler_name="Euler a", eta=0.0)
+        ).infotexts[0]
+
+    def test_eta_scales_ancestral_noise(self, pipe):
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="e", steps=4, width=64, height=64, seeds=[9],
+                    sampler_name="Euler a")
+        dflt = pipe.generate(PipelineRequest(**base)).images
+        one = pipe.generate(PipelineRequest(**base, eta=1.0)).images
+        half = pipe.generate(PipelineRequest(**base, eta=0.5)).images
+        assert torch.equal(dflt, one)  # sampler default is eta=1
+        assert not torch.equal(dflt, half)
+
+    def test_ddim_eta_stochastic(self, pipe):
+        """DDIM defaults deterministic (eta_ddim=0, == Euler); a request
+        eta>0 restores the stochastic DDIM update."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="d", steps=4, width=64, height=64, seeds=[9])
+        a = pipe.generate(
+            PipelineRequest(**base, sampler_name="DDIM")
+        ).images
+        e = pipe.generate(
+            PipelineRequest(**base, sampler_name="Euler")
+        ).images
+        base["sampler_name"] = "DDIM"
+        assert torch.equal(a, e)
+        s = pipe.generate(PipelineRequest(**base, eta=1.0)).images
+        assert not torch.equal(a, s)
+        s2 = pipe.generate(PipelineRequest(**base, eta=1.0)).images
+        assert torch.equal(s, s2)  # seeded noise: deterministic
+
 
 class TestInpaintModel9ch:
     def test_txt2img_with_inpaint_model(self):
