@@ -393,6 +393,95 @@ def create_app(engine: Optional[LocalEngine] = None,
         }
 
     _XYZ_NAMES = {"x/y/z plot", "xyz plot", "xyz grid", "x/y/z", "xyz"}
+    _MATRIX_NAMES = {"prompt matrix"}
+    _FROMFILE_NAMES = {"prompts from file or textbox", "prompts from file"}
+
+    def _script_response(out: Dict[str, Any], gen: GenerationRequest,
+                         req: Txt2ImgRequest,
+                         extra_info: Optional[Dict[str, Any]] = None
+                         ) -> Dict[str, Any]:
+        """Shared response assembly for selectable-script runs."""
+        images = []
+        if req.send_images:
+            if out.get("grid") is not None:
+                images.append(_b64_png(out["grid"]))
+            for sg in out.get("sub_grids", []):
+                images.append(_b64_png(sg))
+            images.extend(
+                base64.b64encode(encode_png(img, info)).decode()
+                for img, info in zip(out["images"], out["infotexts"])
+            )
+        info = {
+            "all_seeds": out["seeds"],
+            "all_subseeds": [-1] * len(out["seeds"]),
+            "all_prompts": out.get(
+                "prompts", [gen.prompt] * len(out["seeds"])
+            ),
+            "all_negative_prompts": [gen.negative_prompt] * len(out["seeds"]),
+            "infotexts": out["infotexts"],
+            "interrupted": out["interrupted"],
+        }
+        if extra_info:
+            info.update(extra_info)
+        return {
+            "images": images,
+            "parameters": {"script_name": req.script_name},
+            "info": json.dumps(info),
+        }
+
+    def _run_prompt_matrix_script(gen: GenerationRequest,
+                                  req: Txt2ImgRequest) -> Dict[str, Any]:
+        """sdwui Prompt matrix script_args:
+        [put_at_start, different_seeds, prompt_type, variations_delimiter,
+        margin_size] (trailing args optional)."""
+        from ..parallel.builtin_scripts import run_prompt_matrix
+
+        a = req.script_args
+        g = lambda i, d=None: a[i] if i < len(a) else d  # noqa: E731
+        try:
+            with state.lock:
+                state.busy = True
+                state.started_at = time.time()
+                try:
+                    out = run_prompt_matrix(
+                        engine, gen,
+                        put_at_start=bool(g(0, False)),
+                        different_seeds=bool(g(1, False)),
+                        prompt_type=str(g(2, "positive") or "positive").lower(),
+                        variations_delimiter=str(g(3, "comma") or "comma"),
+                    )
+                finally:
+                    state.busy = False
+        except ValueError as exc:
+            raise HTTPException(422, str(exc))
+        return _script_response(out, gen, req)
+
+    def _run_prompts_file_script(gen: GenerationRequest,
+                                 req: Txt2ImgRequest) -> Dict[str, Any]:
+        """sdwui Prompts-from-file script_args:
+        [checkbox_iterate, checkbox_iterate_batches, prompt_txt]; the
+        text may also arrive as the sole string argument."""
+        from ..parallel.builtin_scripts import run_prompts_from_file
+
+        a = req.script_args
+        text = next((v for v in a if isinstance(v, str) and v.strip()), "")
+        bools = [v for v in a if isinstance(v, bool)]
+        it = bools[0] if bools else False
+        itb = bools[1] if len(bools) > 1 else False
+        try:
+            with state.lock:
+                state.busy = True
+                state.started_at = time.time()
+                try:
+                    out = run_prompts_from_file(
+                        engine, gen, text,
+                        checkbox_iterate=it, checkbox_iterate_batches=itb,
+                    )
+                finally:
+                    state.busy = False
+        except ValueError as exc:
+            raise HTTPException(422, str(exc))
+        return _script_response(out, gen, req)
 
     def _parse_xyz_args(args: List[Any]) -> Dict[str, Any]:
         """Positional sdwui xyz_grid script_args, both layouts:
@@ -454,11 +543,16 @@ def create_app(engine: Optional[LocalEngine] = None,
         from ..parallel.xyz import run_xyz
 
         name = req.script_name.strip().lower()
+        if name in _MATRIX_NAMES:
+            return _run_prompt_matrix_script(gen, req)
+        if name in _FROMFILE_NAMES:
+            return _run_prompts_file_script(gen, req)
         if name not in _XYZ_NAMES:
             raise HTTPException(
                 422,
                 f"selectable script {req.script_name!r} is not available "
-                "natively; supported: X/Y/Z plot",
+                "natively; supported: X/Y/Z plot, Prompt matrix, "
+                "Prompts from file or textbox",
             )
         kw = _parse_xyz_args(req.script_args)
         # pre-validate axes whose bad values would otherwise produce a
@@ -495,32 +589,9 @@ def create_app(engine: Optional[LocalEngine] = None,
                     state.busy = False
         except ValueError as exc:
             raise HTTPException(422, str(exc))
-        images = []
-        if req.send_images:
-            if out["grid"] is not None:
-                images.append(_b64_png(out["grid"]))
-            for sg in out["sub_grids"]:
-                images.append(_b64_png(sg))
-            images.extend(
-                base64.b64encode(
-                    encode_png(img, info)
-                ).decode()
-                for img, info in zip(out["images"], out["infotexts"])
-            )
-        info = {
-            "all_seeds": out["seeds"],
-            "all_subseeds": [-1] * len(out["seeds"]),
-            "all_prompts": [gen.prompt] * len(out["seeds"]),
-            "all_negative_prompts": [gen.negative_prompt] * len(out["seeds"]),
-            "infotexts": out["infotexts"],
-            "xyz_plot": out["labels"],
-            "interrupted": out["interrupted"],
-        }
-        return {
-            "images": images,
-            "parameters": {"script_name": req.script_name},
-            "info": json.dumps(info),
-        }
+        return _script_response(
+            out, gen, req, extra_info={"xyz_plot": out["labels"]}
+        )
 
     @app.get("/")
     def index():
@@ -817,9 +888,12 @@ def create_app(engine: Optional[LocalEngine] = None,
         # the natively-executed alwayson set (C18) — the reference PROBED
         # each remote's script list through exactly this surface
         # (worker.py:375-404), so report what this engine runs in-process
+        selectable = [
+            "x/y/z plot", "prompt matrix", "prompts from file or textbox",
+        ]
         return {
-            "txt2img": list(_NATIVE_ALWAYSON) + ["x/y/z plot"],
-            "img2img": list(_NATIVE_ALWAYSON) + ["x/y/z plot"],
+            "txt2img": list(_NATIVE_ALWAYSON) + selectable,
+            "img2img": list(_NATIVE_ALWAYSON) + selectable,
         }
 
     @app.get("/sdapi/v1/script-info")
@@ -841,6 +915,10 @@ def create_app(engine: Optional[LocalEngine] = None,
         ] + [
             {"name": "x/y/z plot", "is_alwayson": False,
              "is_img2img": im, "args": xyz_args}
+            for im in (True, False)
+        ] + [
+            {"name": n, "is_alwayson": False, "is_img2img": im, "args": []}
+            for n in ("prompt matrix", "prompts from file or textbox")
             for im in (True, False)
         ]
 
