@@ -880,6 +880,20 @@ class TestSubseedSharding:
         assert diff.max() <= 1.0
         assert (diff > 0).float().mean() < 1e-3
 
+    def test_hires_resize_to_shards_like_base(self):
+        """sdwui 'resize to' mode (hr_scale=0, cover+truncate crop) keeps
+        shard parity and the correct cropped output size."""
+        req = dict(prompt="hr", batch_size=4, width=64, height=64,
+                   steps=2, seed=31, enable_hr=True, hr_scale=0.0,
+                   hr_steps=2, hr_resize_x=96, hr_resize_y=64,
+                   denoising_strength=0.6)
+        one = make_engine(1).generate(GenerationRequest(**req))
+        two = make_engine(2).generate(GenerationRequest(**req))
+        assert one.images.shape == (4, 64, 96, 3)
+        diff = (one.images.float() - two.images.float()).abs()
+        assert diff.max() <= 1.0
+        assert (diff > 0).float().mean() < 1e-3
+
     def test_img2img_color_correction_and_ensd_shard_exact(self):
         """img2img + color correction + eta_noise_seed_delta: bit-exact
         across shard splits (per-image encode + per-image ancestral
