@@ -761,13 +761,27 @@ def create_app(engine: Optional[LocalEngine] = None,
                 raise HTTPException(404, f"unknown model {name}")
             engine.set_model(name)
             state.current_model = name
+        if req.sd_vae:
+            from ..models.registry import available_vaes, refresh_vae_files
+
+            vae = req.sd_vae
+            if vae in ("Automatic", "None"):  # sdwui dropdown spellings
+                vae = "auto"
+            if vae not in available_vaes():
+                refresh_vae_files()
+            if vae not in available_vaes():
+                raise HTTPException(404, f"unknown VAE {vae}")
+            try:
+                engine.set_vae(vae)
+            except (KeyError, ValueError) as exc:
+                raise HTTPException(422, str(exc))
         return {}
 
     @app.get("/sdapi/v1/options")
     def get_options():
         return {
             "sd_model_checkpoint": state.current_model,
-            "sd_vae": "auto",
+            "sd_vae": getattr(engine, "vae_name", "auto"),
             "CLIP_stop_at_last_layers": state.default_clip_skip,
             "eta_noise_seed_delta": state.ensd,
         }
@@ -936,7 +950,20 @@ def create_app(engine: Optional[LocalEngine] = None,
 
     @app.get("/sdapi/v1/sd-vae")
     def sd_vae():
-        return [{"model_name": "auto", "filename": ""}]
+        from ..models.registry import available_vaes, refresh_vae_files
+
+        refresh_vae_files()
+        return [
+            {"model_name": n,
+             "filename": "" if n == "auto" else f"{n}.safetensors"}
+            for n in available_vaes()
+        ]
+
+    @app.post("/sdapi/v1/refresh-vae")
+    def refresh_vae():
+        from ..models.registry import refresh_vae_files
+
+        return {"found": refresh_vae_files()}
 
     @app.get("/sdapi/v1/schedulers")
     def schedulers():

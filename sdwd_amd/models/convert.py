@@ -317,6 +317,62 @@ def _load_sd_clip(enc, clip_sd: Dict[str, torch.Tensor],
                 report["loaded"].append(f"encoder.layers.{i}.self_attn.qkv")
 
 
+def _load_vae_part(vae, vae_sd: Dict[str, torch.Tensor],
+                   report: Dict[str, list]) -> None:
+    """Fill ``vae`` from bare ldm VAE keys (no first_stage_model. prefix),
+    folding the file's quant_conv/post_quant_conv 1x1s into the native
+    encoder.conv_out / decoder.conv_in (this module's docstring).
+    Consumed keys are popped from ``vae_sd``."""
+    quant = {k: vae_sd.pop(k) for k in list(vae_sd) if "quant_conv" in k}
+    _load_part(vae, vae_sd, vae_key_map(vae), report)
+    with torch.no_grad():
+        enc_out = dict(vae.encoder.named_parameters())
+        if "quant_conv.weight" in quant:
+            dev = enc_out["conv_out.weight"].device
+            w, b = _fold_output_1x1(
+                enc_out["conv_out.weight"].float(),
+                enc_out["conv_out.bias"].float(),
+                quant["quant_conv.weight"].float().to(dev),
+                quant["quant_conv.bias"].float().to(dev),
+            )
+            enc_out["conv_out.weight"].copy_(w.to(enc_out["conv_out.weight"].dtype))
+            enc_out["conv_out.bias"].copy_(b.to(enc_out["conv_out.bias"].dtype))
+        dec = dict(vae.decoder.named_parameters())
+        if "post_quant_conv.weight" in quant:
+            dev = dec["conv_in.weight"].device
+            w, b = _fold_input_1x1(
+                dec["conv_in.weight"].float(),
+                dec["conv_in.bias"].float(),
+                quant["post_quant_conv.weight"].float().to(dev),
+                quant["post_quant_conv.bias"].float().to(dev),
+            )
+            dec["conv_in.weight"].copy_(w.to(dec["conv_in.weight"].dtype))
+            dec["conv_in.bias"].copy_(b.to(dec["conv_in.bias"].dtype))
+
+
+def load_vae_state_dict(vae, state: Dict[str, torch.Tensor]
+                        ) -> Dict[str, list]:
+    """Load a STANDALONE VAE file (the sdwui "SD VAE" dropdown files the
+    reference synced by name through load_options — ref worker.py:646-688):
+    keys are either ``first_stage_model.*`` (as in full checkpoints) or the
+    bare ldm VAE naming used by published vae-ft-* / anime VAE files.
+    EMA/loss bookkeeping keys are tolerated."""
+    report: Dict[str, list] = {"loaded": [], "missing": [], "unexpected": []}
+    if any(k.startswith(VAE_PREFIX) for k in state):
+        sub = {
+            k[len(VAE_PREFIX):]: v
+            for k, v in state.items() if k.startswith(VAE_PREFIX)
+        }
+    else:
+        sub = {
+            k: v for k, v in state.items()
+            if not k.startswith(("loss.", "model_ema.", "ema."))
+        }
+    _load_vae_part(vae, sub, report)
+    report["unexpected"] += list(sub)
+    return report
+
+
 def load_ldm_state_dict(bundle, state: Dict[str, torch.Tensor]) -> Dict[str, list]:
     """Fill ``bundle`` (unet/vae/text_encoder) from an sdwui/ldm state dict.
 
@@ -341,31 +397,7 @@ def load_ldm_state_dict(bundle, state: Dict[str, torch.Tensor]) -> Dict[str, lis
         _load_part(bundle.unet, unet_sd, unet_key_map(bundle.unet), report)
         report["unexpected"] += [UNET_PREFIX + k for k in unet_sd]
     if vae_sd:
-        quant = {k: vae_sd.pop(k) for k in list(vae_sd) if "quant_conv" in k}
-        _load_part(bundle.vae, vae_sd, vae_key_map(bundle.vae), report)
-        with torch.no_grad():
-            enc_out = dict(bundle.vae.encoder.named_parameters())
-            if "quant_conv.weight" in quant:
-                dev = enc_out["conv_out.weight"].device
-                w, b = _fold_output_1x1(
-                    enc_out["conv_out.weight"].float(),
-                    enc_out["conv_out.bias"].float(),
-                    quant["quant_conv.weight"].float().to(dev),
-                    quant["quant_conv.bias"].float().to(dev),
-                )
-                enc_out["conv_out.weight"].copy_(w.to(enc_out["conv_out.weight"].dtype))
-                enc_out["conv_out.bias"].copy_(b.to(enc_out["conv_out.bias"].dtype))
-            dec = dict(bundle.vae.decoder.named_parameters())
-            if "post_quant_conv.weight" in quant:
-                dev = dec["conv_in.weight"].device
-                w, b = _fold_input_1x1(
-                    dec["conv_in.weight"].float(),
-                    dec["conv_in.bias"].float(),
-                    quant["post_quant_conv.weight"].float().to(dev),
-                    quant["post_quant_conv.bias"].float().to(dev),
-                )
-                dec["conv_in.weight"].copy_(w.to(dec["conv_in.weight"].dtype))
-                dec["conv_in.bias"].copy_(b.to(dec["conv_in.bias"].dtype))
+        _load_vae_part(bundle.vae, vae_sd, report)
         report["unexpected"] += [VAE_PREFIX + k for k in vae_sd]
     if clip_sd and bundle.text_encoder is not None:
         _load_sd_clip(bundle.text_encoder, clip_sd, report)

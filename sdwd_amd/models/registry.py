@@ -345,6 +345,61 @@ def available_models() -> list:
     return sorted(set(_BUILDERS) | set(_FILE_MODELS))
 
 
+_VAE_FILES: Dict[str, str] = {}
+
+
+def vae_dir() -> str:
+    import os
+
+    return os.environ.get("SDWD_VAE_DIR", "vae")
+
+
+def refresh_vae_files(dirpath: Optional[str] = None) -> list:
+    """Scan the VAE directory for standalone *.safetensors VAE files
+    (the sdwui "SD VAE" dropdown the reference synced by name through
+    load_options, ref worker.py:646-688). Each file's stem becomes a
+    selectable VAE name; "auto" is the checkpoint's own VAE."""
+    import os
+
+    global _VAE_FILES
+    d = dirpath or vae_dir()
+    found: Dict[str, str] = {}
+    if os.path.isdir(d):
+        for fn in sorted(os.listdir(d)):
+            if fn.endswith(".safetensors"):
+                found[os.path.splitext(fn)[0]] = os.path.join(d, fn)
+    _VAE_FILES = found
+    return sorted(found)
+
+
+def available_vaes() -> list:
+    return ["auto"] + sorted(_VAE_FILES)
+
+
+def load_vae_into(bundle, name: str) -> Dict[str, list]:
+    """Swap ``bundle``'s VAE weights to the named standalone VAE file
+    ("auto" restores the checkpoint's own VAE by rebuilding from its
+    source). Returns the converter report."""
+    from safetensors.torch import load_file
+
+    from .convert import load_vae_state_dict
+
+    if name not in _VAE_FILES:
+        refresh_vae_files()
+    if name not in _VAE_FILES:
+        raise KeyError(f"unknown VAE '{name}'")
+    state = load_file(_VAE_FILES[name])
+    dev = next(bundle.vae.parameters()).device
+    state = {k: v.to(dev) for k, v in state.items()}
+    report = load_vae_state_dict(bundle.vae, state)
+    if report["missing"]:
+        raise ValueError(
+            f"VAE '{name}' is missing {len(report['missing'])} tensors "
+            f"(first: {report['missing'][:3]}) — wrong architecture?"
+        )
+    return report
+
+
 def load_model(
     name: str, device="cpu", dtype: Optional[torch.dtype] = None,
     cache: bool = True,

@@ -510,6 +510,7 @@ class LocalEngine(_EngineBase):
                     else ["cpu"]
         self.devices = devices
         self.model_name = model
+        self.vae_name = "auto"
         self.world = world or World.from_devices(len(devices))
         self.pipes: Dict[str, StableDiffusionPipeline] = {}
         self._live_jobs = []
@@ -537,6 +538,32 @@ class LocalEngine(_EngineBase):
                 name, device=pipe.device, dtype=pipe.dtype
             )
         self.model_name = name
+        self.vae_name = "auto"  # fresh bundles carry their own VAE
+
+    def set_vae(self, name: str) -> None:
+        """sdwui "SD VAE" selection (ref C13: the reference synced the VAE
+        by name to every remote through load_options, worker.py:646-688):
+        swap every rank's VAE weights to a standalone file from
+        SDWD_VAE_DIR; "auto" restores each checkpoint's own VAE from a
+        snapshot taken before the first swap."""
+        from ..models.registry import load_vae_into
+
+        name = name or "auto"
+        if name == getattr(self, "vae_name", "auto"):
+            return
+        log.info("switching VAE to %s on all ranks", name)
+        for pipe in self.pipes.values():
+            vae = pipe.model.vae
+            if not hasattr(pipe.model, "_own_vae_sd"):
+                pipe.model._own_vae_sd = {
+                    k: v.detach().cpu().clone()
+                    for k, v in vae.state_dict().items()
+                }
+            if name == "auto":
+                vae.load_state_dict(pipe.model._own_vae_sd)
+            else:
+                load_vae_into(pipe.model, name)
+        self.vae_name = name
 
     # benchmark runner wired into core.Worker.benchmark (ref C8)
     def _bench_runner(self, worker: Worker, payload) -> float:

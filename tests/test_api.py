@@ -656,6 +656,40 @@ class TestGlobalOptions:
             json={"CLIP_stop_at_last_layers": 1, "eta_noise_seed_delta": 0},
         )
 
+    def test_sd_vae_selection(self, client, tmp_path, monkeypatch):
+        """sdwui SD-VAE dropdown flow: list, select, restore (ref C13:
+        load_options synced vae by name, worker.py:646-688)."""
+        from safetensors.torch import save_file
+
+        from sdwd_amd.models.convert import to_ldm_state_dict
+        from sdwd_amd.models.registry import load_model
+
+        a = load_model("tiny", device="cpu", cache=False)
+        with torch.no_grad():
+            for p in a.vae.parameters():
+                p.add_(torch.randn_like(p) * 0.05)
+        vae_sd = {
+            k: v.contiguous().clone()
+            for k, v in to_ldm_state_dict(a).items()
+            if k.startswith("first_stage_model.")
+        }
+        save_file(vae_sd, str(tmp_path / "anime.safetensors"))
+        monkeypatch.setenv("SDWD_VAE_DIR", str(tmp_path))
+        names = [e["model_name"] for e in client.get("/sdapi/v1/sd-vae").json()]
+        assert names == ["auto", "anime"]
+        assert client.post(
+            "/sdapi/v1/options", json={"sd_vae": "anime"}
+        ).status_code == 200
+        assert client.get("/sdapi/v1/options").json()["sd_vae"] == "anime"
+        assert client.post(
+            "/sdapi/v1/options", json={"sd_vae": "missing-vae"}
+        ).status_code == 404
+        # "Automatic" restores the checkpoint's own VAE
+        assert client.post(
+            "/sdapi/v1/options", json={"sd_vae": "Automatic"}
+        ).status_code == 200
+        assert client.get("/sdapi/v1/options").json()["sd_vae"] == "auto"
+
 
 class TestResponseHygiene:
     def test_parameters_never_echo_tensors(self, client):

@@ -350,3 +350,73 @@ class TestRefinerConvert:
         b = load_model("tiny-xl-refiner", device="cpu", cache=False)
         rep = load_ldm_state_dict(b, state)
         assert not rep["missing"] and not rep["unexpected"]
+
+
+class TestStandaloneVae:
+    """sdwui "SD VAE" dropdown files (ref C13 synced VAE by name through
+    load_options, worker.py:646-688) — standalone-file load + engine swap."""
+
+    def _vae_file(self, tmp_path, bare=False):
+        from safetensors.torch import save_file
+
+        a = load_model("tiny", device="cpu", cache=False)
+        with torch.no_grad():
+            for p in a.vae.parameters():
+                p.add_(torch.randn_like(p) * 0.05)
+        full = to_ldm_state_dict(a)
+        vae_sd = {
+            (k[len("first_stage_model."):] if bare else k): v.contiguous().clone()
+            for k, v in full.items()
+            if k.startswith("first_stage_model.")
+        }
+        path = str(tmp_path / ("bare.safetensors" if bare else "pfx.safetensors"))
+        save_file(vae_sd, path)
+        return a, path
+
+    @pytest.mark.parametrize("bare", [False, True])
+    def test_load_vae_state_dict(self, tmp_path, bare):
+        """Both key spellings (first_stage_model.* and bare ldm) load and
+        reproduce the source VAE's decode exactly."""
+        from safetensors.torch import load_file
+
+        from sdwd_amd.models.convert import load_vae_state_dict
+
+        a, path = self._vae_file(tmp_path, bare=bare)
+        b = load_model("tiny", device="cpu", cache=False)
+        z = torch.randn(1, 4, 8, 8, generator=torch.Generator().manual_seed(5))
+        with torch.no_grad():
+            before = b.vae.decode(z)
+            report = load_vae_state_dict(b.vae, load_file(path))
+            after = b.vae.decode(z)
+            want = a.vae.decode(z)
+        assert not report["missing"], report["missing"][:5]
+        assert not report["unexpected"], report["unexpected"][:5]
+        assert not torch.equal(before, after)
+        assert torch.equal(after, want)
+
+    def test_engine_set_vae_and_restore(self, tmp_path, monkeypatch):
+        from sdwd_amd.models.registry import refresh_vae_files
+        from sdwd_amd.parallel import GenerationRequest, LocalEngine
+
+        _, path = self._vae_file(tmp_path)
+        monkeypatch.setenv("SDWD_VAE_DIR", str(tmp_path))
+        refresh_vae_files()
+        eng = LocalEngine(model="tiny", devices=["cpu"])
+        eng.world.workers[0].eta.avg_ipm = 60.0
+        req = GenerationRequest(
+            prompt="v", batch_size=1, width=64, height=64, steps=1, seed=8
+        )
+        auto = eng.generate(req).images
+        eng.set_vae("pfx")
+        swapped = eng.generate(req).images
+        assert not torch.equal(auto, swapped)
+        eng.set_vae("auto")
+        restored = eng.generate(req).images
+        assert torch.equal(auto, restored)
+
+    def test_unknown_vae_raises(self):
+        from sdwd_amd.parallel import LocalEngine
+
+        eng = LocalEngine(model="tiny", devices=["cpu"])
+        with pytest.raises(KeyError):
+            eng.set_vae("no-such-vae")
