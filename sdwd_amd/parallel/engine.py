@@ -459,7 +459,13 @@ class _EngineBase:
             for i in range(n):
                 seeds[o + i] = job.seeds[i] if i < len(job.seeds) else 0
                 base = job_infos[i] if i < len(job_infos) else ""
-                infotexts[o + i] = f"{base}\nWorker: {job.worker_label}"
+                # same line as the param list, comma-joined — exactly the
+                # reference's fixup (distributed.py:347-349) and the only
+                # placement sdwui's infotext parser round-trips
+                infotexts[o + i] = (
+                    f"{base}, Worker Label: {job.worker_label}"
+                    if base else f"Worker Label: {job.worker_label}"
+                )
             tag = " (complementary)" if job.complementary else ""
             summary.append(
                 f"{job.worker_label}: {n} image(s){tag}, "

@@ -33,7 +33,7 @@ class TestLocalEngine:
         assert res.seeds == [123, 124, 125, 126]
         assert res.grid is not None
         assert len(res.job_summary) == 2
-        assert all("Worker: gpu" in t for t in res.infotexts)
+        assert all(", Worker Label: gpu" in t for t in res.infotexts)
 
     def test_matches_single_rank(self):
         """2-rank gallery ~= 1-rank gallery (<=1 uint8 LSB, see pipeline
@@ -312,7 +312,7 @@ class TestDistributedEngine:
         infos = (tmp_path / "infotexts.txt").read_text().split("\x1e")
         assert len(infos) == 4
         for t in infos:
-            base, _, worker = t.rpartition("\nWorker: ")
+            base, _, worker = t.rpartition(", Worker Label: ")
             assert worker.startswith("gpu"), t
             assert "dist" in base, f"lost base infotext: {t!r}"
         # the 2-rank gallery matches a 1-rank run image-for-image
@@ -411,7 +411,7 @@ class TestDistributedEngine:
         infos = (tmp_path / "infotexts.txt").read_text().split("\x1e")
         assert len(infos) == 4
         for t in infos:
-            base, _, worker = t.rpartition("\nWorker: ")
+            base, _, worker = t.rpartition(", Worker Label: ")
             assert worker.startswith("gpu"), t
             assert "dist" in base, f"lost infotext after recovery: {t!r}"
 

@@ -329,3 +329,63 @@ class TestGnPartialHandshake:
         b._sdwd_gnp = (gb, 1, b._version)
         out = cat_channels_gn(a, b)
         assert getattr(out, "_sdwd_gnp", None) is None
+
+
+class TestInfotextParse:
+    """parse_infotext inverts the pipeline's emitted 'parameters' text —
+    the grammar sdwui's own parser (and every 'send to txt2img' tool)
+    expects."""
+
+    def test_basic_grammar(self):
+        from sdwd_amd.utils.infotext import parse_infotext
+
+        out = parse_infotext(
+            "a cow\nover two lines\nNegative prompt: blurry\n"
+            "Steps: 20, Sampler: Euler a, CFG scale: 7.0, Seed: 5, "
+            'Size: 512x512, RP Ratios: "1,2", Worker Label: gpu1'
+        )
+        assert out["prompt"] == "a cow\nover two lines"
+        assert out["negative_prompt"] == "blurry"
+        assert out["Steps"] == "20"
+        assert out["Sampler"] == "Euler a"
+        assert out["RP Ratios"] == "1,2"  # quoted comma survives
+        assert out["Worker Label"] == "gpu1"
+
+    def test_no_negative(self):
+        from sdwd_amd.utils.infotext import parse_infotext
+
+        out = parse_infotext("just a cow\nSteps: 4, Seed: 1, Size: 64x64")
+        assert out["prompt"] == "just a cow"
+        assert out["negative_prompt"] == ""
+        assert out["Seed"] == "1"
+
+    def test_engine_round_trip(self):
+        """A fully-optioned engine run emits infotext this parser (and so
+        sdwui's) reads back field-for-field."""
+        import torch
+
+        from sdwd_amd.parallel import GenerationRequest, LocalEngine
+        from sdwd_amd.utils.infotext import parse_infotext
+
+        eng = LocalEngine(model="tiny", devices=["cpu", "cpu"])
+        for w in eng.world.workers:
+            w.eta.avg_ipm = 60.0
+        res = eng.generate(GenerationRequest(
+            prompt="a cow", negative_prompt="blurry", batch_size=2,
+            width=64, height=64, steps=2, seed=9, cfg_scale=4.5,
+            sampler_name="Heun", clip_skip=2, eta=0.5, tiling=True,
+            enable_hr=True, hr_scale=0.0, hr_resize_x=96, hr_resize_y=96,
+            hr_steps=1, denoising_strength=0.6,
+        ))
+        out = parse_infotext(res.infotexts[1])
+        assert out["prompt"] == "a cow"
+        assert out["negative_prompt"] == "blurry"
+        assert out["Steps"] == "2"
+        assert out["Sampler"] == "Heun"
+        assert out["CFG scale"] == "4.5"
+        assert out["Seed"] == "10"
+        assert out["Clip skip"] == "2"
+        assert out["Eta"] == "0.5"
+        assert out["Tiling"] == "True"
+        assert out["Hires resize"] == "96x96"
+        assert out["Worker Label"].startswith("gpu")
