@@ -94,40 +94,26 @@ def color_correct(out: torch.Tensor, ref: torch.Tensor) -> torch.Tensor:
 
 
 def parse_infotext(text: str) -> dict:
-    """Parse an sdwui 'parameters' infotext back into request fields
-    (inverse of the generator in pipeline.py; sdwui
-    parse_generation_parameters shape)."""
-    import re
+    """Parse an sdwui 'parameters' infotext back into TYPED request fields
+    (snake_case keys, ints/floats coerced) — built on the raw grammar
+    parser in utils/infotext.py, which preserves sdwui key spelling and
+    handles double-quoted values."""
+    from .infotext import parse_infotext as parse_raw
 
-    out: dict = {}
-    lines = text.split("\n")
-    kv_line = ""
-    prompt_lines = []
-    for ln in lines:
-        if re.match(r"^Steps: \d+", ln.strip()):
-            kv_line = ln
-            break
-        prompt_lines.append(ln)
-    neg = ""
-    prompts = []
-    for ln in prompt_lines:
-        if ln.startswith("Negative prompt: "):
-            neg = ln[len("Negative prompt: "):]
-        else:
-            prompts.append(ln)
-    out["prompt"] = "\n".join(prompts).strip()
-    out["negative_prompt"] = neg
-    for m in re.finditer(r"([A-Za-z ]+): ([^,]+)(?:, |$)", kv_line):
-        key = m.group(1).strip().lower().replace(" ", "_")
-        val = m.group(2).strip()
-        out[key] = val
-    if "size" in out and "x" in out["size"]:
+    raw = parse_raw(text)
+    out: dict = {
+        "prompt": raw.pop("prompt", ""),
+        "negative_prompt": raw.pop("negative_prompt", ""),
+    }
+    for key, val in raw.items():
+        out[key.strip().lower().replace(" ", "_")] = val
+    if "size" in out and "x" in str(out["size"]):
         w, h = out.pop("size").split("x")
         out["width"], out["height"] = int(w), int(h)
     for k in ("steps", "seed", "clip_skip"):
         if k in out:
             out[k] = int(out[k])
-    for k in ("cfg_scale", "denoising_strength"):
+    for k in ("cfg_scale", "denoising_strength", "eta"):
         if k in out:
             out[k] = float(out[k])
     return out
