@@ -38,6 +38,10 @@ PAGE = """<!DOCTYPE html>
 </form>
 <div id="gallery"></div>
 <div>live: <img id="preview" style="max-width:128px;display:none"/></div>
+<h2>model</h2>
+checkpoint <select id="mainModel"></select>
+vae <select id="vaeSel"></select>
+<button onclick="applyModel()">apply</button> <span id="m_saved"></span>
 <h2>utils</h2>
 <button onclick="fetch('/sdwd/benchmark',{method:'POST'})">re-benchmark</button>
 <button onclick="fetch('/sdwd/sync-script',{method:'POST'})">run sync script</button>
@@ -194,7 +198,31 @@ async function loadModels(){
     MODELS = ms.map(m => m.model_name);
   }catch(e){}
 }
-(async () => { await loadModels(); await refresh(); fillOverride(); })();
+async function loadModelSelectors(){
+  try{
+    const vs = await (await fetch('/sdapi/v1/sd-vae')).json();
+    const opts = await (await fetch('/sdapi/v1/options')).json();
+    document.getElementById('mainModel').innerHTML = MODELS.map(m =>
+      `<option ${m===opts.sd_model_checkpoint?'selected':''}>${m}</option>`
+    ).join('');
+    document.getElementById('vaeSel').innerHTML = vs.map(v =>
+      `<option ${v.model_name===opts.sd_vae?'selected':''}>`+
+      `${v.model_name}</option>`).join('');
+  }catch(e){}
+}
+async function applyModel(){
+  const body = {
+    sd_model_checkpoint: document.getElementById('mainModel').value,
+    sd_vae: document.getElementById('vaeSel').value,
+  };
+  const r = await fetch('/sdapi/v1/options',{method:'POST',
+    headers:{'Content-Type':'application/json'}, body: JSON.stringify(body)});
+  document.getElementById('m_saved').textContent = r.ok ? 'applied' : 'error';
+  setTimeout(()=>{document.getElementById('m_saved').textContent='';}, 2000);
+  refresh();
+}
+(async () => { await loadModels(); await refresh(); fillOverride();
+               loadModelSelectors(); })();
 setInterval(fillOverride, 1600);
 loadSettings();
 loadSamplers();
