@@ -328,16 +328,23 @@ def create_app(engine: Optional[LocalEngine] = None,
             return await call_next(request)
 
     def run_generation(gen: GenerationRequest, send_images: bool = True,
-                       save_images: bool = False) -> Dict[str, Any]:
+                       save_images: bool = False,
+                       vae_override: str = "") -> Dict[str, Any]:
         # one generation at a time (the reference serialized on the host's
         # queue_lock, world.py:244,273); concurrent requests queue here
         with state.lock:
             state.busy = True
             state.started_at = time.time()
+            prev_vae = getattr(engine, "vae_name", "auto")
+            swap = bool(vae_override) and vae_override != prev_vae
             try:
+                if swap:  # sdwui override_settings.sd_vae: per-request
+                    engine.set_vae(vae_override)
                 result = engine.generate(gen)
                 state.current_model = engine.model_name
             finally:
+                if swap:
+                    engine.set_vae(prev_vae)
                 state.busy = False
         if save_images:
             from ..utils.images import save_png
@@ -627,12 +634,22 @@ def create_app(engine: Optional[LocalEngine] = None,
             req.refiner_checkpoint not in available_models()
         ):
             raise HTTPException(404, f"unknown refiner {req.refiner_checkpoint}")
-        return model, clip_skip, ensd
+        vae = str(ov.get("sd_vae") or "")
+        if vae in ("Automatic", "None"):
+            vae = "auto"
+        if vae and vae != "auto":
+            from ..models.registry import available_vaes, refresh_vae_files
+
+            if vae not in available_vaes():
+                refresh_vae_files()
+            if vae not in available_vaes():
+                raise HTTPException(404, f"unknown VAE {vae}")
+        return model, clip_skip, ensd, vae
 
     @app.post("/sdapi/v1/txt2img")
     def txt2img(req: Txt2ImgRequest):
         control_units = _parse_controlnet(req.alwayson_scripts)
-        model, clip_skip, ensd = _overrides(req)
+        model, clip_skip, ensd, vae_ov = _overrides(req)
         gen = GenerationRequest(
             prompt=req.prompt,
             negative_prompt=req.negative_prompt,
@@ -677,13 +694,14 @@ def create_app(engine: Optional[LocalEngine] = None,
             _expand_dynamic(gen)
         if req.script_name:
             return run_script(gen, req)
-        return run_generation(gen, req.send_images, req.save_images)
+        return run_generation(gen, req.send_images, req.save_images,
+                              vae_override=vae_ov)
 
     @app.post("/sdapi/v1/img2img")
     def img2img(req: Img2ImgRequest):
         if not req.init_images:
             raise HTTPException(422, "init_images required")
-        model, clip_skip, ensd = _overrides(req)
+        model, clip_skip, ensd, vae_ov = _overrides(req)
         try:
             inits = torch.stack(
                 [
@@ -744,7 +762,8 @@ def create_app(engine: Optional[LocalEngine] = None,
         )
         if req.script_name:
             return run_script(gen, req)
-        return run_generation(gen, req.send_images, req.save_images)
+        return run_generation(gen, req.send_images, req.save_images,
+                              vae_override=vae_ov)
 
     @app.post("/sdapi/v1/options")
     def set_options(req: OptionsRequest):

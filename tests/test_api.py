@@ -689,6 +689,23 @@ class TestGlobalOptions:
             "/sdapi/v1/options", json={"sd_vae": "Automatic"}
         ).status_code == 200
         assert client.get("/sdapi/v1/options").json()["sd_vae"] == "auto"
+        # per-request override: applies for one generation, restores after
+        body = {"prompt": "v", "steps": 1, "width": 64, "height": 64,
+                "seed": 6}
+        plain = client.post("/sdapi/v1/txt2img", json=body).json()["images"]
+        over = client.post(
+            "/sdapi/v1/txt2img",
+            json={**body, "override_settings": {"sd_vae": "anime"}},
+        ).json()["images"]
+        assert over[-1] != plain[-1]  # different VAE decodes differently
+        assert client.get("/sdapi/v1/options").json()["sd_vae"] == "auto"
+        again = client.post("/sdapi/v1/txt2img", json=body).json()["images"]
+        assert again[-1] == plain[-1]  # global VAE untouched
+        r = client.post(
+            "/sdapi/v1/txt2img",
+            json={**body, "override_settings": {"sd_vae": "nope"}},
+        )
+        assert r.status_code == 404
 
 
 class TestResponseHygiene:
