@@ -78,8 +78,9 @@ class Txt2ImgRequest(BaseModel):
     # sdwui per-request overrides (sd_model_checkpoint,
     # CLIP_stop_at_last_layers are honored; the rest are ignored)
     override_settings: Dict[str, Any] = Field(default_factory=dict)
-    # sdwui selectable script (X/Y/Z plot runs natively; any other
-    # script_name is rejected loudly — parallel/xyz.py docstring)
+    # sdwui selectable script: X/Y/Z plot, Prompt matrix and
+    # Prompts-from-file run natively (parallel/xyz.py,
+    # parallel/builtin_scripts.py); any other name is rejected loudly
     script_name: Optional[str] = ""  # sdwui clients send null for "none"
     script_args: List[Any] = Field(default_factory=list)
     send_images: bool = True   # omit base64 images from the response
