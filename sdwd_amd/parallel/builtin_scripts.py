@@ -54,7 +54,11 @@ def run_prompt_matrix(
     if len(combos) > 256:
         raise ValueError(f"prompt matrix of {len(combos)} combinations (max 256)")
     if gen.seed == -1:
-        gen = replace(gen, seed=int(torch.randint(0, 2**31 - 1, (1,)).item()))
+        from . import group as pg
+
+        gen = replace(gen, seed=pg.broadcast_object(
+            int(torch.randint(0, 2**31 - 1, (1,)).item())
+        ))
     images: List[torch.Tensor] = []
     seeds: List[int] = []
     infos: List[str] = []
@@ -160,7 +164,11 @@ def run_prompts_from_file(
         raise ValueError(f"prompts from file: {len(lines)} lines (max 1024)")
     jobs = [parse_prompt_line(ln) for ln in lines]
     if (checkbox_iterate or checkbox_iterate_batches) and gen.seed == -1:
-        gen = replace(gen, seed=int(torch.randint(0, 2**31 - 1, (1,)).item()))
+        from . import group as pg
+
+        gen = replace(gen, seed=pg.broadcast_object(
+            int(torch.randint(0, 2**31 - 1, (1,)).item())
+        ))
     images: List[torch.Tensor] = []
     seeds: List[int] = []
     infos: List[str] = []

@@ -254,11 +254,15 @@ def run_xyz(
             sr_search[axis] = str(vals[0])
 
     # sdwui fixes the seed ONCE so cells are comparable, unless the user
-    # asked for free seeds or sweeps the seed axis itself
+    # asked for free seeds or sweeps the seed axis itself. Broadcast from
+    # rank 0 so DistributedEngine ranks build identical cell requests.
     if not no_fixed_seeds and gen.seed == -1:
-        gen = replace(
-            gen, seed=int(torch.randint(0, 2**31 - 1, (1,)).item())
+        from . import group as pg
+
+        seed = pg.broadcast_object(
+            int(torch.randint(0, 2**31 - 1, (1,)).item())
         )
+        gen = replace(gen, seed=seed)
 
     n_cells = len(xs) * len(ys) * len(zs)
     if n_cells > 1024:
