@@ -43,6 +43,12 @@ def main() -> int:
     ap.add_argument("--strength", type=float, default=0.75)
     ap.add_argument("--hires", action="store_true")
     ap.add_argument("--hr-scale", type=float, default=2.0)
+    ap.add_argument("--hr-resize", default="",
+                    help="WxH hires target (sdwui resize-to mode)")
+    ap.add_argument("--eta", type=float, default=-1.0,
+                    help="ancestral/SDE noise multiplier (sdwui Eta)")
+    ap.add_argument("--vae", default="",
+                    help="standalone VAE name from SDWD_VAE_DIR")
     ap.add_argument("--out", default="outputs")
     ap.add_argument("--benchmark", action="store_true",
                     help="re-benchmark ranks before generating")
@@ -59,6 +65,8 @@ def main() -> int:
         if not args.prompt:
             args.prompt = prompts[0]
     engine = LocalEngine(model=args.model)
+    if args.vae:
+        engine.set_vae(args.vae)
     print(f"{len(engine.devices)} rank(s): {engine.devices}")
     if args.benchmark:
         print("benchmarking...", engine.benchmark(rebenchmark=True))
@@ -85,8 +93,11 @@ def main() -> int:
             seed=args.seed,
             init_images=init_images,
             denoising_strength=args.strength,
-            enable_hr=args.hires,
-            hr_scale=args.hr_scale,
+            enable_hr=args.hires or bool(args.hr_resize),
+            hr_scale=0.0 if args.hr_resize else args.hr_scale,
+            hr_resize_x=int(args.hr_resize.split('x')[0]) if args.hr_resize else 0,
+            hr_resize_y=int(args.hr_resize.split('x')[1]) if args.hr_resize else 0,
+            eta=args.eta,
             hr_upscaler=args.hr_upscaler,
             refiner_model=args.refiner,
             refiner_switch_at=args.refiner_switch_at,
