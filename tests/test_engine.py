@@ -169,6 +169,33 @@ def _dist_worker(rank, world_size, port, tmpdir, fail_rank):
     destroy_group()
 
 
+def _xyz_worker(rank, world_size, port, tmpdir):
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world_size)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    from sdwd_amd.parallel import DistributedEngine, GenerationRequest
+    from sdwd_amd.parallel.xyz import run_xyz
+
+    eng = DistributedEngine(model="tiny", backend="gloo")
+    out = run_xyz(
+        eng,
+        GenerationRequest(
+            prompt="sweep", batch_size=2, width=64, height=64, steps=2,
+            seed=-1,  # broadcast-fixed once; must agree across ranks
+        ),
+        "Steps", "1,2", "CFG Scale", "5,7",
+    )
+    if rank == 0:
+        torch.save(out["grid"], os.path.join(tmpdir, "xyz_grid.pt"))
+        with open(os.path.join(tmpdir, "xyz_seed.txt"), "w") as fh:
+            fh.write(",".join(str(s) for s in out["seeds"]))
+    from sdwd_amd.parallel import destroy_group
+
+    destroy_group()
+
+
 def _spawn(world_size, tmpdir, fail_rank=-1):
     import torch.multiprocessing as mp
     import socket
@@ -371,6 +398,26 @@ class TestDistributedEngine:
         assert diff.max() <= 1.0, diff.max()
         assert (diff > 0).float().mean() < 1e-3
         assert "RP Active: True" in one.infotexts[0]
+
+    def test_two_rank_xyz_sweep(self, tmp_path):
+        """X/Y/Z plot through DistributedEngine: every cell is a
+        collective generate; the broadcast-fixed seed keeps ranks'
+        control flow identical (seed=-1 start)."""
+        import socket
+
+        import torch.multiprocessing as mp
+
+        with socket.socket() as s:
+            s.bind(("127.0.0.1", 0))
+            port = s.getsockname()[1]
+        mp.start_processes(
+            _xyz_worker, args=(2, port, str(tmp_path)), nprocs=2,
+            start_method="spawn", join=True,
+        )
+        grid = torch.load(tmp_path / "xyz_grid.pt")
+        assert grid.shape == (128, 128, 3)  # 2x2 cells of 64x64
+        seeds = (tmp_path / "xyz_seed.txt").read_text().split(",")
+        assert len(set(seeds)) == 1 and seeds[0] != "-1"
 
     def test_three_rank_complementary_production(self, tmp_path):
         """A deferred slow rank produces bonus images through the
