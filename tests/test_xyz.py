@@ -287,3 +287,44 @@ class TestXYZApi:
         )
         assert r.status_code == 422
         assert "no-such-model" in r.json()["detail"]
+
+
+class TestValueParsingProperties:
+    """Hypothesis laws for the sdwui range grammar."""
+
+    from hypothesis import given, settings as hsettings
+    from hypothesis import strategies as st
+
+    @given(a=st.integers(-50, 50), b=st.integers(-50, 50))
+    @hsettings(max_examples=60, deadline=None)
+    def test_int_range_is_inclusive_and_monotone(self, a, b):
+        if a > b:
+            a, b = b, a
+        vals = parse_axis_values("int", f"{a}-{b}")
+        assert vals[0] == a and vals[-1] == b
+        assert vals == sorted(vals)
+        assert len(vals) == b - a + 1
+
+    @given(a=st.integers(-20, 20), b=st.integers(-20, 20),
+           n=st.integers(2, 9))
+    @hsettings(max_examples=60, deadline=None)
+    def test_count_form_hits_endpoints(self, a, b, n):
+        vals = parse_axis_values("int", f"{a}-{b} [{n}]")
+        assert len(vals) == n
+        assert vals[0] == a and vals[-1] == b
+
+    @given(a=st.integers(-20, 20), b=st.integers(0, 40),
+           s=st.integers(1, 7))
+    @hsettings(max_examples=60, deadline=None)
+    def test_step_form_never_overshoots(self, a, b, s):
+        hi = a + b
+        vals = parse_axis_values("int", f"{a}-{hi} (+{s})")
+        assert vals[0] == a
+        assert all(v <= hi for v in vals)
+        assert all(y - x == s for x, y in zip(vals, vals[1:]))
+
+    @given(xs=st.lists(st.integers(-99, 99), min_size=1, max_size=6))
+    @hsettings(max_examples=60, deadline=None)
+    def test_comma_list_round_trips(self, xs):
+        vals = parse_axis_values("int", ", ".join(str(x) for x in xs))
+        assert vals == xs
