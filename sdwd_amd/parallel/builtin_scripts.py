@@ -172,6 +172,7 @@ def run_prompts_from_file(
     images: List[torch.Tensor] = []
     seeds: List[int] = []
     infos: List[str] = []
+    prompts: List[str] = []
     interrupted = False
     seed_cursor: Optional[int] = gen.seed if gen.seed != -1 else None
     for overrides in jobs:
@@ -185,6 +186,7 @@ def run_prompts_from_file(
             images.append(res.images[i])
             seeds.append(res.seeds[i])
             infos.append(res.infotexts[i])
+            prompts.append(cell.prompt)
         if checkbox_iterate and seed_cursor is not None:
             seed_cursor += cell.batch_size
         if res.interrupted:
@@ -201,5 +203,6 @@ def run_prompts_from_file(
         "images": images,
         "seeds": seeds,
         "infotexts": infos,
+        "prompts": prompts,
         "interrupted": interrupted,
     }
