@@ -191,6 +191,7 @@ class TestScriptsApi:
         assert r.status_code == 200, r.text
         info = json.loads(r.json()["info"])
         assert info["all_seeds"] == [5, 9]
+        assert info["all_prompts"] == ["a cow", "x"]  # per-line prompts
 
     def test_bad_line_rejected(self, client):
         r = client.post(
