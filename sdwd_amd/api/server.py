@@ -1148,6 +1148,7 @@ def create_app(engine: Optional[LocalEngine] = None,
             "speed_summary": engine.world.speed_summary(),
             "log": ring_buffer(),
             "model": state.current_model,
+            "vae": getattr(engine, "vae_name", "auto"),
             "busy": state.busy,
         }
 

@@ -83,7 +83,7 @@ async function refresh(){
            `<button onclick="tog('${w.label}','${w.state}')">`+
            `${w.state==='DISABLED'?'enable':'disable'}</button></td></tr>`;
     }
-    h += `</table><p>model: ${s.model} — busy: ${s.busy}</p>`;
+    h += `</table><p>model: ${s.model} — vae: ${s.vae} — busy: ${s.busy}</p>`;
     document.getElementById('status').innerHTML = h;
     if (s.busy) {
       const pr = await (await fetch('/sdapi/v1/progress')).json();

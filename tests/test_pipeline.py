@@ -941,6 +941,22 @@ class TestSamplerParams:
         assert torch.equal(dflt, one)  # sampler default is eta=1
         assert not torch.equal(dflt, half)
 
+    @pytest.mark.parametrize("name", [
+        "DPM++ SDE", "DPM++ 2S a", "DPM2 a", "DPM++ 2M SDE",
+    ])
+    def test_eta_family_deterministic_and_distinct(self, pipe, name):
+        """Every eta-consuming sampler: a given eta is reproducible, and
+        eta=0.5 diverges from the default eta=1 trajectory."""
+        from sdwd_amd.pipeline import PipelineRequest
+
+        base = dict(prompt="e", steps=4, width=64, height=64, seeds=[9],
+                    sampler_name=name)
+        dflt = pipe.generate(PipelineRequest(**base)).images
+        half = pipe.generate(PipelineRequest(**base, eta=0.5)).images
+        half2 = pipe.generate(PipelineRequest(**base, eta=0.5)).images
+        assert torch.equal(half, half2), name
+        assert not torch.equal(dflt, half), name
+
     def test_ddim_eta_stochastic(self, pipe):
         """DDIM defaults deterministic (eta_ddim=0, == Euler); a request
         eta>0 restores the stochastic DDIM update."""
